@@ -1,0 +1,179 @@
+/* hipframe.h — C-ABI of libhipframe.so: the MI355X-native partition-operator
+ * engine behind the modin_amd backend.
+ *
+ * This boundary replaces Modin's per-partition "engine deploy" call: in the
+ * reference, every operator template executes a pandas kernel on a partition's
+ * pandas.DataFrame through
+ *   modin/core/execution/python/common/engine_wrapper.py:17-42 (PythonWrapper.deploy)
+ *   modin/core/dataframe/pandas/partitioning/partition.py:114   (apply)
+ * Here the partition payload is a set of device-resident column buffers
+ * (hf_col) on one MI355X, and each operator template dispatches one of the
+ * entry points below (hand-written gfx950 HIP kernels) instead of a pandas
+ * call.  Host-side callers bind via ctypes (see modin_amd/core/lib.py and
+ * INTEGRATION.md for the stub a Modin maintainer would add).
+ *
+ * Conventions:
+ *   - every function returns int status (0 = HF_OK, nonzero = error) unless
+ *     stated; hf_last_error() gives a thread-local message for the last
+ *     failure.
+ *   - all kernels run on one module-owned HIP stream per process;
+ *     hf_sync() drains it.  Host calls are not thread-safe (Modin's L4/L5
+ *     calls all arrive on the user thread — partition_manager.py semantics).
+ *   - ownership: every hf_col* returned by the library is owned by the caller
+ *     and freed with hf_col_free().
+ *   - there is NO CPU fallback behind any entry point: on a machine without a
+ *     visible GPU, hf_init fails and every compute call after a failed init
+ *     fails loudly.
+ */
+#ifndef HIPFRAME_H
+#define HIPFRAME_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- status ---- */
+enum {
+  HF_OK = 0,
+  HF_ERR_HIP = 1,        /* HIP runtime error; hf_last_error() has detail   */
+  HF_ERR_ARG = 2,        /* bad argument (dtype/len mismatch, null, range)  */
+  HF_ERR_NOINIT = 3,     /* hf_init not called / failed                     */
+  HF_ERR_UNSUPPORTED = 4 /* op/dtype combination not implemented            */
+};
+
+/* ---- dtypes (column element types; SoA device layout) ---- */
+enum {
+  HF_INT64 = 0,
+  HF_FLOAT64 = 1
+};
+
+/* Opaque column: device pointer + dtype + length + owning GPU.
+ * Replaces the per-partition pandas block of
+ * modin/core/execution/python/.../partition.py:22 (partition wraps a concrete
+ * pandas.DataFrame); here a partition wraps a set of hf_col. */
+typedef struct hf_col hf_col;
+
+/* ---- lifecycle ---- */
+int  hf_init(int gpu);              /* select device, create stream + pool   */
+int  hf_shutdown(void);
+int  hf_device_count(int* out);     /* works without init                    */
+const char* hf_last_error(void);
+int  hf_sync(void);                 /* drain the module stream               */
+
+/* ---- column memory / transfer ----
+ * Replace PandasDataframePartition.put / get:
+ *   modin/core/dataframe/pandas/partitioning/partition.py:277 (put)
+ *   partition_manager.py:1070 from_pandas -> H2D columnar upload. */
+int  hf_put(const void* host, int64_t len, int dtype, hf_col** out); /* H2D  */
+int  hf_get(const hf_col* col, void* host);                          /* D2H  */
+int  hf_col_alloc(int64_t len, int dtype, hf_col** out);  /* uninitialised   */
+int  hf_col_free(hf_col* col);
+int64_t   hf_col_len(const hf_col* col);
+int       hf_col_dtype(const hf_col* col);
+uintptr_t hf_col_dptr(const hf_col* col);  /* raw device ptr (RCCL/torch
+                                              interop plumbing only)         */
+
+/* Raw device buffer alloc/free for dense groupby tables when the caller does
+ * not hand in an externally allocated (e.g. torch) buffer. */
+int  hf_alloc_raw(int64_t bytes, uintptr_t* dptr);
+int  hf_free_raw(uintptr_t dptr);
+int  hf_memset_raw(uintptr_t dptr, int value, int64_t bytes);
+
+/* ---- Map: elementwise scalar ops ----
+ * Device form of the Map operator template
+ * (modin/core/dataframe/algebra/map.py:28-70 + query_compiler.py:2036 abs,
+ * :2710 fillna, binary.py:449 scalar branch): one coalesced HBM scan,
+ * out[i] = op(in[i], scalar).  f64 and i64 columns. */
+enum {
+  HF_MAP_ADD = 0,   /* x + s       */
+  HF_MAP_SUB = 1,   /* x - s       */
+  HF_MAP_RSUB = 2,  /* s - x       */
+  HF_MAP_MUL = 3,   /* x * s       */
+  HF_MAP_DIV = 4,   /* x / s  (f64 only)                        */
+  HF_MAP_RDIV = 5,  /* s / x  (f64 only)                        */
+  HF_MAP_FILLNA = 6,/* isnan(x) ? s : x  (f64 only)             */
+  HF_MAP_ABS = 7,   /* |x|         */
+  HF_MAP_NEG = 8,   /* -x          */
+  HF_MAP_CAST_F64 = 9 /* (double)x : i64 -> f64; scalar ignored */
+};
+int hf_map_scalar(int op, const hf_col* in, double scalar, hf_col** out);
+/* i64 column with an exact int64 scalar (double cannot hold all int64). */
+int hf_map_scalar_i64(int op, const hf_col* in, int64_t scalar, hf_col** out);
+
+/* ---- Binary: elementwise column<op>column ----
+ * Device form of algebra/binary.py:293-459 frame branch (n_ary_op zip-apply,
+ * dataframe.py:3851) for two row-aligned columns of equal length. */
+enum {
+  HF_BIN_ADD = 0, HF_BIN_SUB = 1, HF_BIN_MUL = 2, HF_BIN_DIV = 3
+};
+int hf_binary(int op, const hf_col* a, const hf_col* b, hf_col** out);
+
+/* ---- TreeReduce: full-column reductions ----
+ * Device form of TreeReduce map+reduce (algebra/tree_reduce.py:29-82,
+ * dataframe.py:2208-2250; pandas nan-skipping semantics of
+ * DataFrame.sum/count/min/max, query_compiler.py:984).
+ * One pass returns all four partials so mean/count need no extra scan:
+ *   sum   = sum of non-NaN elements (0.0 if none — pandas sum min_count=0)
+ *   count = number of non-NaN elements
+ *   mn/mx = min/max over non-NaN (caller maps count==0 -> NaN)
+ * For HF_INT64 columns the partials are exact (sum may wrap like pandas/numpy
+ * int64); mn/mx returned via the f64 slots losslessly for |x| < 2^53 and via
+ * imn/imx exactly. */
+typedef struct {
+  double  sum;
+  int64_t count;
+  double  mn, mx;     /* valid when count > 0 */
+  int64_t isum, imn, imx; /* exact int64 partials for HF_INT64 columns */
+} hf_reduce_result;
+int hf_reduce(const hf_col* in, hf_reduce_result* out); /* syncs the stream */
+
+/* ---- GroupByReduce: dense-key hash/array aggregation ----
+ * Device form of GroupByReduce.map / .reduce
+ * (modin/core/dataframe/algebra/groupby.py:124-208 map = per-partition
+ *  groupby(...).sum(); :211-300 reduce = concat partials + groupby(level=0))
+ * redesigned MI355X-first: the "map" phase accumulates every partition into
+ * ONE dense key-indexed table per GPU (keys must lie in
+ * [key_min, key_min + n_slots)); the "reduce" phase is a table merge (RCCL
+ * all-reduce across GPUs — done by the caller over the raw table buffers) +
+ * on-device compaction to sorted present keys.
+ *
+ * Table layout (all device memory, caller-allocated so it can live in an
+ * RCCL-reducible tensor):
+ *   sums   : double[nvals][n_slots]   per-value-column NaN-skipping sums
+ *   rowcnt : int64 [n_slots]          rows seen per key (NaN rows included —
+ *                                     defines group presence like pandas)
+ *   counts : int64 [nvals][n_slots]   non-NaN count per value column
+ *                                     (optional: pass 0 to skip; needed for
+ *                                     count/mean)
+ * All buffers must be zeroed by the caller before the first accumulate. */
+int hf_groupby_accum(const hf_col* keys,            /* HF_INT64, len n        */
+                     const hf_col* const* vals,     /* nvals HF_FLOAT64 cols  */
+                     int nvals,
+                     int64_t key_min, int64_t n_slots,
+                     uintptr_t sums, uintptr_t rowcnt, uintptr_t counts);
+
+/* Compact a (merged) table to pandas-groupby-shaped output: ascending present
+ * keys (rowcnt>0), per-column sums, optional counts.  Returns n_groups and
+ * caller-owned output columns (out_counts may be NULL if counts==0). */
+int hf_groupby_compact(uintptr_t sums, uintptr_t rowcnt, uintptr_t counts,
+                       int nvals, int64_t key_min, int64_t n_slots,
+                       hf_col** out_keys,          /* HF_INT64 [n_groups]     */
+                       hf_col** out_sums,          /* nvals cols, caller array*/
+                       hf_col** out_counts,        /* nvals cols or NULL      */
+                       int64_t* n_groups);
+
+/* ---- profiling (bench.py roofline leg) ----
+ * When enabled, every kernel launch is bracketed by HIP events on the module
+ * stream; hf_kernel_stats returns the accumulated count and total ms for the
+ * named kernel since the last hf_kernel_stats_reset. */
+int hf_profiling(int enable);
+int hf_kernel_stats(const char* name, int64_t* launches, double* total_ms);
+int hf_kernel_stats_reset(void);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* HIPFRAME_H */

@@ -1,0 +1,1058 @@
+// hipframe.hip — gfx950 (MI355X, CDNA4) implementation of the hipframe C-ABI.
+//
+// Every op here is the device form of one Modin operator template (the
+// reference executes these as pandas calls inside each partition — see the
+// per-function citations in include/hipframe.h).  All paths are
+// HBM-bandwidth-bound (no MFMA anywhere on this path), so the
+// kernels are built around the CDNA4 streaming rules:
+//   - 16 B/lane vectorized loads (double2 / longlong2) — the coalescing sweet
+//     spot on gfx950,
+//   - grid-stride loops capped at a few thousand 256-thread workgroups
+//     (≫256 so all 8 XCDs fill),
+//   - wave64 shuffle reductions + LDS block reductions, one atomic per block,
+//   - dense-key groupby via hardware global_atomic_add_f64 into an
+//     HBM/L3-resident key-indexed table (the table for the north-star config,
+//     1e6 keys, is 8–24 MB: Infinity-Cache resident).
+//
+// Build: hipcc --offload-arch=gfx950 -O3 -shared -fPIC (see Makefile).
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+#include <string>
+#include <type_traits>
+#include <unordered_map>
+#include <vector>
+
+#include "../../include/hipframe.h"
+
+// ---------------------------------------------------------------------------
+// module state & error plumbing
+// ---------------------------------------------------------------------------
+
+struct hf_col {
+  void*   dptr;
+  int64_t len;
+  int     dtype;
+  int     gpu;
+};
+
+namespace {
+
+thread_local std::string g_err;
+
+struct TimedPair { hipEvent_t a, b; std::string name; };
+
+struct State {
+  bool        inited = false;
+  int         gpu = -1;
+  hipStream_t stream = nullptr;
+  // small persistent device scratch: reduce accumulators + error word +
+  // compact bookkeeping
+  void*       d_scratch = nullptr;   // see layout below
+  // profiling
+  bool        profiling = false;
+  std::vector<TimedPair> pending;
+  std::unordered_map<std::string, std::pair<int64_t, double>> stats;
+};
+State g;
+
+// d_scratch layout (bytes):
+//   [0..128)    reduce accumulator block (hf_reduce)
+//   [128..136)  groupby error word (u64: count of out-of-range keys)
+//   [136..144)  compact total (i64 n_groups)
+constexpr int64_t SCRATCH_BYTES = 4096;
+constexpr int64_t SCRATCH_GB_ERR = 128;
+constexpr int64_t SCRATCH_NGROUPS = 136;
+
+int set_err(int code, const char* where, const char* what) {
+  g_err = std::string(where) + ": " + what;
+  return code;
+}
+
+int set_hip_err(const char* where, hipError_t e) {
+  return set_err(HF_ERR_HIP, where, hipGetErrorString(e));
+}
+
+#define HF_HIP(where, call)                                   \
+  do {                                                        \
+    hipError_t _e = (call);                                   \
+    if (_e != hipSuccess) return set_hip_err(where, _e);      \
+  } while (0)
+
+#define HF_NEED_INIT(where)                                   \
+  if (!g.inited) return set_err(HF_ERR_NOINIT, where, "hf_init not called")
+
+int64_t dtype_size(int dt) {
+  switch (dt) {
+    case HF_INT64:   return 8;
+    case HF_FLOAT64: return 8;
+    default:         return 0;
+  }
+}
+
+// resolve pending profiling events into stats (syncs the stream)
+int resolve_stats(const char* where) {
+  if (g.pending.empty()) return HF_OK;
+  HF_HIP(where, hipStreamSynchronize(g.stream));
+  for (auto& p : g.pending) {
+    float ms = 0.f;
+    hipEventElapsedTime(&ms, p.a, p.b);
+    auto& s = g.stats[p.name];
+    s.first += 1;
+    s.second += ms;
+    hipEventDestroy(p.a);
+    hipEventDestroy(p.b);
+  }
+  g.pending.clear();
+  return HF_OK;
+}
+
+// launch helper with optional event bracketing
+template <typename F>
+int timed_launch(const char* name, F&& launch) {
+  if (!g.profiling) {
+    launch();
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess) return set_hip_err(name, e);
+    return HF_OK;
+  }
+  TimedPair p;
+  p.name = name;
+  HF_HIP(name, hipEventCreate(&p.a));
+  HF_HIP(name, hipEventCreate(&p.b));
+  HF_HIP(name, hipEventRecord(p.a, g.stream));
+  launch();
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) return set_hip_err(name, e);
+  HF_HIP(name, hipEventRecord(p.b, g.stream));
+  g.pending.push_back(p);
+  if (g.pending.size() > 4096) return resolve_stats(name);
+  return HF_OK;
+}
+
+constexpr int BLOCK = 256;
+// memory-bound grid cap: ≫256 workgroups to fill 8 XCDs, grid-stride the rest
+// (cdna_hip_programming.md §6 Guideline 11)
+constexpr int64_t GRID_CAP = 4096;
+
+int64_t grid_for(int64_t work_items) {
+  int64_t b = (work_items + BLOCK - 1) / BLOCK;
+  if (b < 1) b = 1;
+  return b < GRID_CAP ? b : GRID_CAP;
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// kernels: Map (elementwise scalar) — algebra/map.py:28 device form
+// ---------------------------------------------------------------------------
+
+namespace {
+
+template <int OP>
+__device__ __forceinline__ double map1_f64(double x, double s) {
+  switch (OP) {
+    case HF_MAP_ADD:    return x + s;
+    case HF_MAP_SUB:    return x - s;
+    case HF_MAP_RSUB:   return s - x;
+    case HF_MAP_MUL:    return x * s;
+    case HF_MAP_DIV:    return x / s;
+    case HF_MAP_RDIV:   return s / x;
+    case HF_MAP_FILLNA: return (x != x) ? s : x;
+    case HF_MAP_ABS:    return fabs(x);
+    case HF_MAP_NEG:    return -x;
+  }
+  return x;
+}
+
+template <int OP>
+__global__ void __launch_bounds__(BLOCK) k_map_f64(const double* __restrict__ in,
+                                                   double* __restrict__ out,
+                                                   double s, int64_t n) {
+  // 16 B/lane double2 stream; tail element handled by thread 0 of block 0.
+  const int64_t npair = n >> 1;
+  const double2* in2 = reinterpret_cast<const double2*>(in);
+  double2* out2 = reinterpret_cast<double2*>(out);
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < npair; i += stride) {
+    double2 v = in2[i];
+    v.x = map1_f64<OP>(v.x, s);
+    v.y = map1_f64<OP>(v.y, s);
+    out2[i] = v;
+  }
+  if ((n & 1) && blockIdx.x == 0 && threadIdx.x == 0)
+    out[n - 1] = map1_f64<OP>(in[n - 1], s);
+}
+
+template <int OP>
+__device__ __forceinline__ int64_t map1_i64(int64_t x, int64_t s) {
+  switch (OP) {
+    case HF_MAP_ADD:  return x + s;
+    case HF_MAP_SUB:  return x - s;
+    case HF_MAP_RSUB: return s - x;
+    case HF_MAP_MUL:  return x * s;
+    case HF_MAP_ABS:  return x < 0 ? -x : x;
+    case HF_MAP_NEG:  return -x;
+  }
+  return x;
+}
+
+template <int OP>
+__global__ void __launch_bounds__(BLOCK) k_map_i64(const int64_t* __restrict__ in,
+                                                   int64_t* __restrict__ out,
+                                                   int64_t s, int64_t n) {
+  const int64_t npair = n >> 1;
+  const longlong2* in2 = reinterpret_cast<const longlong2*>(in);
+  longlong2* out2 = reinterpret_cast<longlong2*>(out);
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < npair; i += stride) {
+    longlong2 v = in2[i];
+    v.x = map1_i64<OP>(v.x, s);
+    v.y = map1_i64<OP>(v.y, s);
+    out2[i] = v;
+  }
+  if ((n & 1) && blockIdx.x == 0 && threadIdx.x == 0)
+    out[n - 1] = map1_i64<OP>(in[n - 1], s);
+}
+
+__global__ void __launch_bounds__(BLOCK) k_cast_i64_f64(const int64_t* __restrict__ in,
+                                                        double* __restrict__ out,
+                                                        int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = (double)in[i];
+}
+
+// ---------------------------------------------------------------------------
+// kernels: Binary (elementwise column op column) — algebra/binary.py device form
+// ---------------------------------------------------------------------------
+
+template <int OP, typename T>
+__device__ __forceinline__ T bin1(T a, T b) {
+  switch (OP) {
+    case HF_BIN_ADD: return a + b;
+    case HF_BIN_SUB: return a - b;
+    case HF_BIN_MUL: return a * b;
+    case HF_BIN_DIV: return a / b;
+  }
+  return a;
+}
+
+template <int OP, typename T, typename T2>
+__global__ void __launch_bounds__(BLOCK) k_bin(const T* __restrict__ a,
+                                               const T* __restrict__ b,
+                                               T* __restrict__ out, int64_t n) {
+  const int64_t npair = n >> 1;
+  const T2* a2 = reinterpret_cast<const T2*>(a);
+  const T2* b2 = reinterpret_cast<const T2*>(b);
+  T2* o2 = reinterpret_cast<T2*>(out);
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < npair; i += stride) {
+    T2 va = a2[i], vb = b2[i];
+    va.x = bin1<OP, T>(va.x, vb.x);
+    va.y = bin1<OP, T>(va.y, vb.y);
+    o2[i] = va;
+  }
+  if ((n & 1) && blockIdx.x == 0 && threadIdx.x == 0)
+    out[n - 1] = bin1<OP, T>(a[n - 1], b[n - 1]);
+}
+
+// ---------------------------------------------------------------------------
+// kernels: TreeReduce — algebra/tree_reduce.py device form
+// pandas nan-skipping sum/count/min/max in one pass.
+// Wave64 shuffle reduce -> LDS across waves -> one CAS/atomic per block
+// (cdna_hip_programming.md Appendix B "Reduction").
+// ---------------------------------------------------------------------------
+
+struct ReduceAccF64 {  // lives in d_scratch[0..128)
+  double sum;
+  unsigned long long count;
+  double mn, mx;
+};
+struct ReduceAccI64 {
+  long long sum;
+  unsigned long long count;
+  long long mn, mx;
+};
+
+__device__ void atomic_min_f64(double* addr, double v) {
+  unsigned long long* p = reinterpret_cast<unsigned long long*>(addr);
+  unsigned long long old = *p, assumed;
+  while (v < __longlong_as_double(old)) {
+    assumed = old;
+    old = atomicCAS(p, assumed, __double_as_longlong(v));
+    if (old == assumed) break;
+  }
+}
+__device__ void atomic_max_f64(double* addr, double v) {
+  unsigned long long* p = reinterpret_cast<unsigned long long*>(addr);
+  unsigned long long old = *p, assumed;
+  while (v > __longlong_as_double(old)) {
+    assumed = old;
+    old = atomicCAS(p, assumed, __double_as_longlong(v));
+    if (old == assumed) break;
+  }
+}
+__device__ void atomic_min_i64(long long* addr, long long v) {
+  unsigned long long* p = reinterpret_cast<unsigned long long*>(addr);
+  unsigned long long old = *p, assumed;
+  while (v < (long long)old) {
+    assumed = old;
+    old = atomicCAS(p, assumed, (unsigned long long)v);
+    if (old == assumed) break;
+  }
+}
+__device__ void atomic_max_i64(long long* addr, long long v) {
+  unsigned long long* p = reinterpret_cast<unsigned long long*>(addr);
+  unsigned long long old = *p, assumed;
+  while (v > (long long)old) {
+    assumed = old;
+    old = atomicCAS(p, assumed, (unsigned long long)v);
+    if (old == assumed) break;
+  }
+}
+
+__global__ void k_reduce_init(ReduceAccF64* f, ReduceAccI64* i64a) {
+  f->sum = 0.0; f->count = 0;
+  f->mn = __longlong_as_double(0x7FF0000000000000LL);   // +inf
+  f->mx = __longlong_as_double(0xFFF0000000000000LL);   // -inf
+  i64a->sum = 0; i64a->count = 0;
+  i64a->mn = 0x7FFFFFFFFFFFFFFFLL;
+  i64a->mx = 0x8000000000000000LL;
+}
+
+__global__ void __launch_bounds__(BLOCK) k_reduce_f64(const double* __restrict__ in,
+                                                      int64_t n, ReduceAccF64* acc) {
+  double sum = 0.0, mn = __longlong_as_double(0x7FF0000000000000LL),
+         mx = __longlong_as_double(0xFFF0000000000000LL);
+  unsigned long long cnt = 0;
+  const int64_t npair = n >> 1;
+  const double2* in2 = reinterpret_cast<const double2*>(in);
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < npair; i += stride) {
+    double2 v = in2[i];
+    if (v.x == v.x) { sum += v.x; ++cnt; mn = fmin(mn, v.x); mx = fmax(mx, v.x); }
+    if (v.y == v.y) { sum += v.y; ++cnt; mn = fmin(mn, v.y); mx = fmax(mx, v.y); }
+  }
+  if ((n & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
+    double v = in[n - 1];
+    if (v == v) { sum += v; ++cnt; mn = fmin(mn, v); mx = fmax(mx, v); }
+  }
+  // wave64 shuffle reduce
+  for (int off = 32; off > 0; off >>= 1) {
+    sum += __shfl_down(sum, off);
+    cnt += __shfl_down(cnt, off);
+    mn = fmin(mn, __shfl_down(mn, off));
+    mx = fmax(mx, __shfl_down(mx, off));
+  }
+  __shared__ double s_sum[BLOCK / 64], s_mn[BLOCK / 64], s_mx[BLOCK / 64];
+  __shared__ unsigned long long s_cnt[BLOCK / 64];
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) { s_sum[wave] = sum; s_cnt[wave] = cnt; s_mn[wave] = mn; s_mx[wave] = mx; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < BLOCK / 64; ++w) {
+      sum += s_sum[w]; cnt += s_cnt[w];
+      mn = fmin(mn, s_mn[w]); mx = fmax(mx, s_mx[w]);
+    }
+    if (cnt) {
+      unsafeAtomicAdd(&acc->sum, sum);
+      atomicAdd(&acc->count, cnt);
+      atomic_min_f64(&acc->mn, mn);
+      atomic_max_f64(&acc->mx, mx);
+    }
+  }
+}
+
+__device__ __forceinline__ long long llmin(long long a, long long b) { return a < b ? a : b; }
+__device__ __forceinline__ long long llmax(long long a, long long b) { return a > b ? a : b; }
+
+__global__ void __launch_bounds__(BLOCK) k_reduce_i64(const int64_t* __restrict__ in,
+                                                      int64_t n, ReduceAccI64* acc) {
+  long long sum = 0, mn = 0x7FFFFFFFFFFFFFFFLL, mx = 0x8000000000000000LL;
+  unsigned long long cnt = 0;
+  const int64_t npair = n >> 1;
+  const longlong2* in2 = reinterpret_cast<const longlong2*>(in);
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < npair; i += stride) {
+    longlong2 v = in2[i];
+    sum += v.x + v.y; cnt += 2;
+    mn = llmin(mn, llmin(v.x, v.y));
+    mx = llmax(mx, llmax(v.x, v.y));
+  }
+  if ((n & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
+    long long v = in[n - 1];
+    sum += v; ++cnt; mn = llmin(mn, v); mx = llmax(mx, v);
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    sum += __shfl_down(sum, off);
+    cnt += __shfl_down(cnt, off);
+    mn = llmin(mn, (long long)__shfl_down(mn, off));
+    mx = llmax(mx, (long long)__shfl_down(mx, off));
+  }
+  __shared__ long long s_sum[BLOCK / 64], s_mn[BLOCK / 64], s_mx[BLOCK / 64];
+  __shared__ unsigned long long s_cnt[BLOCK / 64];
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) { s_sum[wave] = sum; s_cnt[wave] = cnt; s_mn[wave] = mn; s_mx[wave] = mx; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < BLOCK / 64; ++w) {
+      sum += s_sum[w]; cnt += s_cnt[w];
+      mn = llmin(mn, s_mn[w]); mx = llmax(mx, s_mx[w]);
+    }
+    if (cnt) {
+      atomicAdd((unsigned long long*)&acc->sum, (unsigned long long)sum);
+      atomicAdd(&acc->count, cnt);
+      atomic_min_i64(&acc->mn, mn);
+      atomic_max_i64(&acc->mx, mx);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// kernels: GroupByReduce — algebra/groupby.py:124/:211 device form.
+// Dense key-indexed table accumulation: keys in [key_min, key_min+n_slots).
+// One u32-slot rowcnt atomic per row + one hardware f64 atomic add per
+// non-NaN value.  The 1e6-key north-star table (8 MB sums + 8 MB rowcnt)
+// is Infinity-Cache resident; atomics execute memory-side so the per-XCD L2
+// incoherence is not in play.
+// ---------------------------------------------------------------------------
+
+constexpr int GB_MAX_VALS = 8;
+
+struct GbPtrs {
+  const double* vals[GB_MAX_VALS];
+};
+
+template <int NVALS, bool COUNTS>
+__global__ void __launch_bounds__(BLOCK) k_gb_accum(
+    const int64_t* __restrict__ keys, GbPtrs ptrs, int64_t n,
+    int64_t key_min, int64_t n_slots,
+    double* __restrict__ sums,            // [NVALS][n_slots]
+    unsigned long long* __restrict__ rowcnt,  // [n_slots]
+    unsigned long long* __restrict__ counts,  // [NVALS][n_slots] or null
+    unsigned long long* __restrict__ err) {
+  const int64_t npair = n >> 1;
+  const longlong2* keys2 = reinterpret_cast<const longlong2*>(keys);
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < npair; i += stride) {
+    longlong2 kk = keys2[i];
+    const int64_t k0 = kk.x - key_min, k1 = kk.y - key_min;
+    const bool ok0 = (uint64_t)k0 < (uint64_t)n_slots;
+    const bool ok1 = (uint64_t)k1 < (uint64_t)n_slots;
+    if (!ok0 || !ok1) atomicAdd(err, (unsigned long long)(!ok0 + !ok1));
+    if (ok0) atomicAdd(&rowcnt[k0], 1ULL);
+    if (ok1) atomicAdd(&rowcnt[k1], 1ULL);
+#pragma unroll
+    for (int c = 0; c < NVALS; ++c) {
+      const double2 v = reinterpret_cast<const double2*>(ptrs.vals[c])[i];
+      if (ok0 && v.x == v.x) {
+        unsafeAtomicAdd(&sums[(int64_t)c * n_slots + k0], v.x);
+        if (COUNTS) atomicAdd(&counts[(int64_t)c * n_slots + k0], 1ULL);
+      }
+      if (ok1 && v.y == v.y) {
+        unsafeAtomicAdd(&sums[(int64_t)c * n_slots + k1], v.y);
+        if (COUNTS) atomicAdd(&counts[(int64_t)c * n_slots + k1], 1ULL);
+      }
+    }
+  }
+  if ((n & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
+    const int64_t k = keys[n - 1] - key_min;
+    if ((uint64_t)k < (uint64_t)n_slots) {
+      atomicAdd(&rowcnt[k], 1ULL);
+      for (int c = 0; c < NVALS; ++c) {
+        const double v = ptrs.vals[c][n - 1];
+        if (v == v) {
+          unsafeAtomicAdd(&sums[(int64_t)c * n_slots + k], v);
+          if (COUNTS) atomicAdd(&counts[(int64_t)c * n_slots + k], 1ULL);
+        }
+      }
+    } else {
+      atomicAdd(err, 1ULL);
+    }
+  }
+}
+
+// ---- compaction: present slots (rowcnt>0) -> ascending keys + columns ----
+// Fixed tile partitioning so prefix order == slot order.
+constexpr int COMPACT_TILE = 4096;  // slots per tile, one 256-thread block/tile
+
+__global__ void __launch_bounds__(BLOCK) k_compact_count(
+    const unsigned long long* __restrict__ rowcnt, int64_t n_slots,
+    int64_t* __restrict__ tile_counts) {
+  const int64_t t0 = (int64_t)blockIdx.x * COMPACT_TILE;
+  const int64_t t1 = min(t0 + (int64_t)COMPACT_TILE, n_slots);
+  int local = 0;
+  for (int64_t s = t0 + threadIdx.x; s < t1; s += blockDim.x)
+    local += (rowcnt[s] != 0);
+  // block reduce
+  for (int off = 32; off > 0; off >>= 1) local += __shfl_down(local, off);
+  __shared__ int s_c[BLOCK / 64];
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) s_c[wave] = local;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int tot = 0;
+    for (int w = 0; w < BLOCK / 64; ++w) tot += s_c[w];
+    tile_counts[blockIdx.x] = tot;
+  }
+}
+
+// single-workgroup exclusive scan of tile_counts (ntiles <= 1M/4096*32 ~ small)
+__global__ void __launch_bounds__(1024) k_compact_scan(
+    int64_t* __restrict__ tile_counts, int64_t ntiles, int64_t* __restrict__ total) {
+  __shared__ int64_t carry;
+  if (threadIdx.x == 0) carry = 0;
+  __syncthreads();
+  __shared__ int64_t buf[1024];
+  for (int64_t base = 0; base < ntiles; base += 1024) {
+    const int64_t i = base + threadIdx.x;
+    int64_t v = (i < ntiles) ? tile_counts[i] : 0;
+    // Hillis–Steele inclusive scan in LDS
+    buf[threadIdx.x] = v;
+    __syncthreads();
+    for (int off = 1; off < 1024; off <<= 1) {
+      int64_t add = (threadIdx.x >= off) ? buf[threadIdx.x - off] : 0;
+      __syncthreads();
+      buf[threadIdx.x] += add;
+      __syncthreads();
+    }
+    const int64_t incl = buf[threadIdx.x];
+    if (i < ntiles) tile_counts[i] = carry + incl - v;  // exclusive
+    __syncthreads();
+    if (threadIdx.x == 1023) carry += incl;
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) *total = carry;
+}
+
+template <bool COUNTS>
+__global__ void __launch_bounds__(BLOCK) k_compact_scatter(
+    const double* __restrict__ sums, const unsigned long long* __restrict__ rowcnt,
+    const unsigned long long* __restrict__ counts, int nvals,
+    int64_t key_min, int64_t n_slots,
+    const int64_t* __restrict__ tile_offsets,
+    int64_t* __restrict__ out_keys, double* const* __restrict__ out_sums,
+    int64_t* const* __restrict__ out_counts) {
+  const int64_t t0 = (int64_t)blockIdx.x * COMPACT_TILE;
+  const int64_t t1 = min(t0 + (int64_t)COMPACT_TILE, n_slots);
+  __shared__ int64_t s_base;
+  __shared__ int s_wave_cnt[BLOCK / 64];
+  if (threadIdx.x == 0) s_base = tile_offsets[blockIdx.x];
+  __syncthreads();
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  for (int64_t chunk = t0; chunk < t1; chunk += blockDim.x) {
+    const int64_t s = chunk + threadIdx.x;
+    const bool pred = (s < t1) && (rowcnt[s] != 0);
+    const uint64_t ballot = __ballot(pred);
+    if (lane == 0) s_wave_cnt[wave] = __popcll(ballot);
+    __syncthreads();
+    int64_t wave_base = 0;
+    for (int w = 0; w < wave; ++w) wave_base += s_wave_cnt[w];
+    if (pred) {
+      const int64_t pos = s_base + wave_base +
+          __popcll(ballot & ((lane == 63) ? ~0ULL >> 1 : ((1ULL << lane) - 1)));
+      out_keys[pos] = key_min + s;
+      for (int c = 0; c < nvals; ++c) {
+        out_sums[c][pos] = sums[(int64_t)c * n_slots + s];
+        if (COUNTS) out_counts[c][pos] = (int64_t)counts[(int64_t)c * n_slots + s];
+      }
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      int64_t tot = 0;
+      for (int w = 0; w < BLOCK / 64; ++w) tot += s_wave_cnt[w];
+      s_base += tot;
+    }
+    __syncthreads();
+  }
+}
+
+}  // namespace
+
+// forward decls used by the map launch helpers (defined in the C ABI below)
+extern "C" int hf_col_alloc(int64_t len, int dtype, hf_col** out);
+extern "C" int hf_col_free(hf_col* col);
+extern "C" int hf_map_scalar_i64(int op, const hf_col* in, int64_t scalar,
+                                 hf_col** out);
+
+namespace {
+template <int OP>
+int launch_map_f64(const hf_col* in, double s, hf_col* out) {
+  const int64_t n = in->len;
+  return timed_launch("map_f64", [&] {
+    hipLaunchKernelGGL(k_map_f64<OP>, dim3(grid_for((n >> 1) + 1)), dim3(BLOCK), 0,
+                       g.stream, (const double*)in->dptr, (double*)out->dptr, s, n);
+  });
+}
+template <int OP>
+int launch_map_i64(const hf_col* in, int64_t s, hf_col* out) {
+  const int64_t n = in->len;
+  return timed_launch("map_i64", [&] {
+    hipLaunchKernelGGL(k_map_i64<OP>, dim3(grid_for((n >> 1) + 1)), dim3(BLOCK), 0,
+                       g.stream, (const int64_t*)in->dptr, (int64_t*)out->dptr, s, n);
+  });
+}
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// C ABI
+// ---------------------------------------------------------------------------
+
+extern "C" {
+
+int hf_init(int gpu) {
+  if (g.inited && g.gpu == gpu) return HF_OK;
+  if (g.inited) hf_shutdown();
+  int n = 0;
+  hipError_t e = hipGetDeviceCount(&n);
+  if (e != hipSuccess) return set_hip_err("hf_init", e);
+  if (gpu < 0 || gpu >= n)
+    return set_err(HF_ERR_ARG, "hf_init", "gpu index out of range");
+  HF_HIP("hf_init", hipSetDevice(gpu));
+  HF_HIP("hf_init", hipStreamCreate(&g.stream));
+  HF_HIP("hf_init", hipMalloc(&g.d_scratch, SCRATCH_BYTES));
+  HF_HIP("hf_init", hipMemset(g.d_scratch, 0, SCRATCH_BYTES));
+  g.gpu = gpu;
+  g.inited = true;
+  return HF_OK;
+}
+
+int hf_shutdown(void) {
+  if (!g.inited) return HF_OK;
+  hipStreamSynchronize(g.stream);
+  for (auto& p : g.pending) { hipEventDestroy(p.a); hipEventDestroy(p.b); }
+  g.pending.clear();
+  g.stats.clear();
+  if (g.d_scratch) hipFree(g.d_scratch);
+  hipStreamDestroy(g.stream);
+  g = State{};
+  return HF_OK;
+}
+
+int hf_device_count(int* out) {
+  int n = 0;
+  hipError_t e = hipGetDeviceCount(&n);
+  if (e != hipSuccess) { *out = 0; return set_hip_err("hf_device_count", e); }
+  *out = n;
+  return HF_OK;
+}
+
+const char* hf_last_error(void) { return g_err.c_str(); }
+
+int hf_sync(void) {
+  HF_NEED_INIT("hf_sync");
+  HF_HIP("hf_sync", hipStreamSynchronize(g.stream));
+  return HF_OK;
+}
+
+// ---- memory ----
+
+int hf_col_alloc(int64_t len, int dtype, hf_col** out) {
+  HF_NEED_INIT("hf_col_alloc");
+  if (len < 0 || dtype_size(dtype) == 0 || !out)
+    return set_err(HF_ERR_ARG, "hf_col_alloc", "bad len/dtype");
+  void* d = nullptr;
+  int64_t bytes = len * dtype_size(dtype);
+  if (bytes == 0) bytes = 8;  // keep zero-length columns addressable
+  HF_HIP("hf_col_alloc", hipMallocAsync(&d, bytes, g.stream));
+  hf_col* c = new hf_col{d, len, dtype, g.gpu};
+  *out = c;
+  return HF_OK;
+}
+
+int hf_put(const void* host, int64_t len, int dtype, hf_col** out) {
+  HF_NEED_INIT("hf_put");
+  int rc = hf_col_alloc(len, dtype, out);
+  if (rc != HF_OK) return rc;
+  if (len > 0)
+    HF_HIP("hf_put", hipMemcpyAsync((*out)->dptr, host, len * dtype_size(dtype),
+                                    hipMemcpyHostToDevice, g.stream));
+  return HF_OK;
+}
+
+int hf_get(const hf_col* col, void* host) {
+  HF_NEED_INIT("hf_get");
+  if (!col || !host) return set_err(HF_ERR_ARG, "hf_get", "null");
+  if (col->len > 0)
+    HF_HIP("hf_get", hipMemcpyAsync(host, col->dptr, col->len * dtype_size(col->dtype),
+                                    hipMemcpyDeviceToHost, g.stream));
+  HF_HIP("hf_get", hipStreamSynchronize(g.stream));
+  return HF_OK;
+}
+
+int hf_col_free(hf_col* col) {
+  if (!col) return HF_OK;
+  if (g.inited && col->dptr) hipFreeAsync(col->dptr, g.stream);
+  delete col;
+  return HF_OK;
+}
+
+int64_t hf_col_len(const hf_col* c) { return c ? c->len : -1; }
+int hf_col_dtype(const hf_col* c) { return c ? c->dtype : -1; }
+uintptr_t hf_col_dptr(const hf_col* c) { return c ? (uintptr_t)c->dptr : 0; }
+
+int hf_alloc_raw(int64_t bytes, uintptr_t* dptr) {
+  HF_NEED_INIT("hf_alloc_raw");
+  void* d = nullptr;
+  HF_HIP("hf_alloc_raw", hipMallocAsync(&d, bytes, g.stream));
+  *dptr = (uintptr_t)d;
+  return HF_OK;
+}
+int hf_free_raw(uintptr_t dptr) {
+  HF_NEED_INIT("hf_free_raw");
+  HF_HIP("hf_free_raw", hipFreeAsync((void*)dptr, g.stream));
+  return HF_OK;
+}
+int hf_memset_raw(uintptr_t dptr, int value, int64_t bytes) {
+  HF_NEED_INIT("hf_memset_raw");
+  HF_HIP("hf_memset_raw", hipMemsetAsync((void*)dptr, value, bytes, g.stream));
+  return HF_OK;
+}
+
+// ---- map ----
+
+int hf_map_scalar(int op, const hf_col* in, double scalar, hf_col** out) {
+  HF_NEED_INIT("hf_map_scalar");
+  if (!in || !out) return set_err(HF_ERR_ARG, "hf_map_scalar", "null");
+  if (op == HF_MAP_CAST_F64) {
+    if (in->dtype != HF_INT64)
+      return set_err(HF_ERR_ARG, "hf_map_scalar", "cast_f64 needs int64 input");
+    int rc = hf_col_alloc(in->len, HF_FLOAT64, out);
+    if (rc != HF_OK) return rc;
+    return timed_launch("cast_i64_f64", [&] {
+      hipLaunchKernelGGL(k_cast_i64_f64, dim3(grid_for(in->len)), dim3(BLOCK), 0,
+                         g.stream, (const int64_t*)in->dptr, (double*)(*out)->dptr,
+                         in->len);
+    });
+  }
+  if (in->dtype == HF_INT64) return hf_map_scalar_i64(op, in, (int64_t)scalar, out);
+  int rc = hf_col_alloc(in->len, HF_FLOAT64, out);
+  if (rc != HF_OK) return rc;
+  switch (op) {
+    case HF_MAP_ADD:    rc = launch_map_f64<HF_MAP_ADD>(in, scalar, *out); break;
+    case HF_MAP_SUB:    rc = launch_map_f64<HF_MAP_SUB>(in, scalar, *out); break;
+    case HF_MAP_RSUB:   rc = launch_map_f64<HF_MAP_RSUB>(in, scalar, *out); break;
+    case HF_MAP_MUL:    rc = launch_map_f64<HF_MAP_MUL>(in, scalar, *out); break;
+    case HF_MAP_DIV:    rc = launch_map_f64<HF_MAP_DIV>(in, scalar, *out); break;
+    case HF_MAP_RDIV:   rc = launch_map_f64<HF_MAP_RDIV>(in, scalar, *out); break;
+    case HF_MAP_FILLNA: rc = launch_map_f64<HF_MAP_FILLNA>(in, scalar, *out); break;
+    case HF_MAP_ABS:    rc = launch_map_f64<HF_MAP_ABS>(in, scalar, *out); break;
+    case HF_MAP_NEG:    rc = launch_map_f64<HF_MAP_NEG>(in, scalar, *out); break;
+    default:
+      hf_col_free(*out);
+      *out = nullptr;
+      return set_err(HF_ERR_UNSUPPORTED, "hf_map_scalar", "unknown op");
+  }
+  if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  return rc;
+}
+
+int hf_map_scalar_i64(int op, const hf_col* in, int64_t scalar, hf_col** out) {
+  HF_NEED_INIT("hf_map_scalar_i64");
+  if (!in || !out) return set_err(HF_ERR_ARG, "hf_map_scalar_i64", "null");
+  if (in->dtype != HF_INT64)
+    return set_err(HF_ERR_ARG, "hf_map_scalar_i64", "int64 column required");
+  int rc = hf_col_alloc(in->len, HF_INT64, out);
+  if (rc != HF_OK) return rc;
+  switch (op) {
+    case HF_MAP_ADD:  rc = launch_map_i64<HF_MAP_ADD>(in, scalar, *out); break;
+    case HF_MAP_SUB:  rc = launch_map_i64<HF_MAP_SUB>(in, scalar, *out); break;
+    case HF_MAP_RSUB: rc = launch_map_i64<HF_MAP_RSUB>(in, scalar, *out); break;
+    case HF_MAP_MUL:  rc = launch_map_i64<HF_MAP_MUL>(in, scalar, *out); break;
+    case HF_MAP_ABS:  rc = launch_map_i64<HF_MAP_ABS>(in, scalar, *out); break;
+    case HF_MAP_NEG:  rc = launch_map_i64<HF_MAP_NEG>(in, scalar, *out); break;
+    default:
+      hf_col_free(*out);
+      *out = nullptr;
+      return set_err(HF_ERR_UNSUPPORTED, "hf_map_scalar_i64",
+                     "op not defined for int64 (div promotes via cast_f64)");
+  }
+  if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  return rc;
+}
+
+// ---- binary ----
+
+int hf_binary(int op, const hf_col* a, const hf_col* b, hf_col** out) {
+  HF_NEED_INIT("hf_binary");
+  if (!a || !b || !out) return set_err(HF_ERR_ARG, "hf_binary", "null");
+  if (a->len != b->len)
+    return set_err(HF_ERR_ARG, "hf_binary", "length mismatch");
+  if (a->dtype != b->dtype)
+    return set_err(HF_ERR_ARG, "hf_binary", "dtype mismatch (cast first)");
+  const bool f64 = a->dtype == HF_FLOAT64;
+  if (!f64 && op == HF_BIN_DIV)
+    return set_err(HF_ERR_UNSUPPORTED, "hf_binary", "int64 div promotes via cast_f64");
+  int rc = hf_col_alloc(a->len, a->dtype, out);
+  if (rc != HF_OK) return rc;
+  const int64_t n = a->len;
+  // dispatch with typed wrappers
+  auto Lf = [&](auto opTag) {
+    constexpr int O = decltype(opTag)::value;
+    return timed_launch("bin_f64", [&] {
+      hipLaunchKernelGGL((k_bin<O, double, double2>), dim3(grid_for((n >> 1) + 1)),
+                         dim3(BLOCK), 0, g.stream, (const double*)a->dptr,
+                         (const double*)b->dptr, (double*)(*out)->dptr, n);
+    });
+  };
+  auto Li = [&](auto opTag) {
+    constexpr int O = decltype(opTag)::value;
+    return timed_launch("bin_i64", [&] {
+      hipLaunchKernelGGL((k_bin<O, int64_t, longlong2>), dim3(grid_for((n >> 1) + 1)),
+                         dim3(BLOCK), 0, g.stream, (const int64_t*)a->dptr,
+                         (const int64_t*)b->dptr, (int64_t*)(*out)->dptr, n);
+    });
+  };
+  switch (op) {
+    case HF_BIN_ADD: rc = f64 ? Lf(std::integral_constant<int, HF_BIN_ADD>{})
+                              : Li(std::integral_constant<int, HF_BIN_ADD>{}); break;
+    case HF_BIN_SUB: rc = f64 ? Lf(std::integral_constant<int, HF_BIN_SUB>{})
+                              : Li(std::integral_constant<int, HF_BIN_SUB>{}); break;
+    case HF_BIN_MUL: rc = f64 ? Lf(std::integral_constant<int, HF_BIN_MUL>{})
+                              : Li(std::integral_constant<int, HF_BIN_MUL>{}); break;
+    case HF_BIN_DIV: rc = Lf(std::integral_constant<int, HF_BIN_DIV>{}); break;
+    default:
+      hf_col_free(*out);
+      *out = nullptr;
+      return set_err(HF_ERR_UNSUPPORTED, "hf_binary", "unknown op");
+  }
+  if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  return rc;
+}
+
+// ---- reduce ----
+
+int hf_reduce(const hf_col* in, hf_reduce_result* out) {
+  HF_NEED_INIT("hf_reduce");
+  if (!in || !out) return set_err(HF_ERR_ARG, "hf_reduce", "null");
+  ReduceAccF64* dF = (ReduceAccF64*)g.d_scratch;
+  ReduceAccI64* dI = (ReduceAccI64*)((char*)g.d_scratch + 64);
+  hipLaunchKernelGGL(k_reduce_init, dim3(1), dim3(1), 0, g.stream, dF, dI);
+  const int64_t n = in->len;
+  int rc;
+  if (in->dtype == HF_FLOAT64) {
+    rc = timed_launch("reduce_f64", [&] {
+      hipLaunchKernelGGL(k_reduce_f64, dim3(grid_for((n >> 1) + 1)), dim3(BLOCK), 0,
+                         g.stream, (const double*)in->dptr, n, dF);
+    });
+  } else {
+    rc = timed_launch("reduce_i64", [&] {
+      hipLaunchKernelGGL(k_reduce_i64, dim3(grid_for((n >> 1) + 1)), dim3(BLOCK), 0,
+                         g.stream, (const int64_t*)in->dptr, n, dI);
+    });
+  }
+  if (rc != HF_OK) return rc;
+  ReduceAccF64 hF;
+  ReduceAccI64 hI;
+  HF_HIP("hf_reduce", hipMemcpyAsync(&hF, dF, sizeof(hF), hipMemcpyDeviceToHost, g.stream));
+  HF_HIP("hf_reduce", hipMemcpyAsync(&hI, dI, sizeof(hI), hipMemcpyDeviceToHost, g.stream));
+  HF_HIP("hf_reduce", hipStreamSynchronize(g.stream));
+  if (in->dtype == HF_FLOAT64) {
+    out->sum = hF.sum;
+    out->count = (int64_t)hF.count;
+    out->mn = hF.mn;
+    out->mx = hF.mx;
+    out->isum = (int64_t)hF.sum;
+    out->imn = (int64_t)hF.mn;
+    out->imx = (int64_t)hF.mx;
+  } else {
+    out->sum = (double)hI.sum;
+    out->count = (int64_t)hI.count;
+    out->mn = (double)hI.mn;
+    out->mx = (double)hI.mx;
+    out->isum = hI.sum;
+    out->imn = hI.mn;
+    out->imx = hI.mx;
+  }
+  return HF_OK;
+}
+
+// ---- groupby ----
+
+int hf_groupby_accum(const hf_col* keys, const hf_col* const* vals, int nvals,
+                     int64_t key_min, int64_t n_slots,
+                     uintptr_t sums, uintptr_t rowcnt, uintptr_t counts) {
+  HF_NEED_INIT("hf_groupby_accum");
+  if (!keys || !vals || nvals < 0 || nvals > GB_MAX_VALS)
+    return set_err(HF_ERR_ARG, "hf_groupby_accum", "bad args (nvals<=8)");
+  if (keys->dtype != HF_INT64)
+    return set_err(HF_ERR_ARG, "hf_groupby_accum", "keys must be int64");
+  if (n_slots <= 0) return set_err(HF_ERR_ARG, "hf_groupby_accum", "n_slots<=0");
+  GbPtrs ptrs{};
+  for (int c = 0; c < nvals; ++c) {
+    if (!vals[c] || vals[c]->dtype != HF_FLOAT64 || vals[c]->len != keys->len)
+      return set_err(HF_ERR_ARG, "hf_groupby_accum",
+                     "vals must be float64 columns of keys' length");
+    ptrs.vals[c] = (const double*)vals[c]->dptr;
+  }
+  const int64_t n = keys->len;
+  unsigned long long* d_err =
+      (unsigned long long*)((char*)g.d_scratch + SCRATCH_GB_ERR);
+  auto launch = [&](auto nvTag, auto cntTag) {
+    constexpr int NV = decltype(nvTag)::value;
+    constexpr bool CNT = decltype(cntTag)::value;
+    return timed_launch("gb_accum", [&] {
+      hipLaunchKernelGGL((k_gb_accum<NV, CNT>), dim3(grid_for((n >> 1) + 1)),
+                         dim3(BLOCK), 0, g.stream, (const int64_t*)keys->dptr, ptrs,
+                         n, key_min, n_slots, (double*)sums,
+                         (unsigned long long*)rowcnt, (unsigned long long*)counts,
+                         d_err);
+    });
+  };
+  const bool cnt = counts != 0;
+  switch (nvals) {
+#define HF_GB_CASE(NV)                                                        \
+  case NV:                                                                    \
+    return cnt ? launch(std::integral_constant<int, NV>{},                    \
+                        std::integral_constant<bool, true>{})                 \
+               : launch(std::integral_constant<int, NV>{},                    \
+                        std::integral_constant<bool, false>{});
+    HF_GB_CASE(1) HF_GB_CASE(2) HF_GB_CASE(3) HF_GB_CASE(4)
+    HF_GB_CASE(5) HF_GB_CASE(6) HF_GB_CASE(7) HF_GB_CASE(8)
+#undef HF_GB_CASE
+    case 0:
+      return cnt ? launch(std::integral_constant<int, 0>{},
+                          std::integral_constant<bool, true>{})
+                 : launch(std::integral_constant<int, 0>{},
+                          std::integral_constant<bool, false>{});
+  }
+  return set_err(HF_ERR_ARG, "hf_groupby_accum", "nvals out of range");
+}
+
+int hf_groupby_compact(uintptr_t sums, uintptr_t rowcnt, uintptr_t counts,
+                       int nvals, int64_t key_min, int64_t n_slots,
+                       hf_col** out_keys, hf_col** out_sums, hf_col** out_counts,
+                       int64_t* n_groups) {
+  HF_NEED_INIT("hf_groupby_compact");
+  if (!rowcnt || nvals < 0 || nvals > GB_MAX_VALS || !out_keys || !n_groups)
+    return set_err(HF_ERR_ARG, "hf_groupby_compact", "bad args");
+  const int64_t ntiles = (n_slots + COMPACT_TILE - 1) / COMPACT_TILE;
+  // check the accumulate-phase error word
+  unsigned long long h_err = 0;
+  unsigned long long* d_err =
+      (unsigned long long*)((char*)g.d_scratch + SCRATCH_GB_ERR);
+  HF_HIP("hf_groupby_compact",
+         hipMemcpyAsync(&h_err, d_err, 8, hipMemcpyDeviceToHost, g.stream));
+  // tile counts
+  int64_t* d_tiles = nullptr;
+  HF_HIP("hf_groupby_compact",
+         hipMallocAsync((void**)&d_tiles, (ntiles + 1) * 8, g.stream));
+  int64_t* d_total = (int64_t*)((char*)g.d_scratch + SCRATCH_NGROUPS);
+  int rc = timed_launch("gb_compact_count", [&] {
+    hipLaunchKernelGGL(k_compact_count, dim3((uint32_t)ntiles), dim3(BLOCK), 0,
+                       g.stream, (const unsigned long long*)rowcnt, n_slots, d_tiles);
+  });
+  if (rc != HF_OK) return rc;
+  rc = timed_launch("gb_compact_scan", [&] {
+    hipLaunchKernelGGL(k_compact_scan, dim3(1), dim3(1024), 0, g.stream, d_tiles,
+                       ntiles, d_total);
+  });
+  if (rc != HF_OK) return rc;
+  int64_t total = 0;
+  HF_HIP("hf_groupby_compact",
+         hipMemcpyAsync(&total, d_total, 8, hipMemcpyDeviceToHost, g.stream));
+  HF_HIP("hf_groupby_compact", hipStreamSynchronize(g.stream));
+  if (h_err != 0) {
+    hipFreeAsync(d_tiles, g.stream);
+    char buf[128];
+    snprintf(buf, sizeof buf,
+             "%llu keys fell outside [key_min, key_min+n_slots) during accumulate",
+             h_err);
+    hipMemsetAsync(d_err, 0, 8, g.stream);
+    return set_err(HF_ERR_ARG, "hf_groupby_compact", buf);
+  }
+  // allocate outputs
+  rc = hf_col_alloc(total, HF_INT64, out_keys);
+  if (rc != HF_OK) return rc;
+  std::vector<double*> h_sum_ptrs(nvals ? nvals : 1);
+  std::vector<int64_t*> h_cnt_ptrs(nvals ? nvals : 1);
+  for (int c = 0; c < nvals; ++c) {
+    rc = hf_col_alloc(total, HF_FLOAT64, &out_sums[c]);
+    if (rc != HF_OK) return rc;
+    h_sum_ptrs[c] = (double*)out_sums[c]->dptr;
+    if (counts && out_counts) {
+      rc = hf_col_alloc(total, HF_INT64, &out_counts[c]);
+      if (rc != HF_OK) return rc;
+      h_cnt_ptrs[c] = (int64_t*)out_counts[c]->dptr;
+    }
+  }
+  // ship the per-column output pointer arrays to the device
+  double** d_sum_ptrs = nullptr;
+  int64_t** d_cnt_ptrs = nullptr;
+  HF_HIP("hf_groupby_compact",
+         hipMallocAsync((void**)&d_sum_ptrs, sizeof(double*) * (nvals ? nvals : 1),
+                        g.stream));
+  HF_HIP("hf_groupby_compact",
+         hipMallocAsync((void**)&d_cnt_ptrs, sizeof(int64_t*) * (nvals ? nvals : 1),
+                        g.stream));
+  HF_HIP("hf_groupby_compact",
+         hipMemcpyAsync(d_sum_ptrs, h_sum_ptrs.data(), sizeof(double*) * nvals,
+                        hipMemcpyHostToDevice, g.stream));
+  if (counts && out_counts)
+    HF_HIP("hf_groupby_compact",
+           hipMemcpyAsync(d_cnt_ptrs, h_cnt_ptrs.data(), sizeof(int64_t*) * nvals,
+                          hipMemcpyHostToDevice, g.stream));
+  const bool want_counts = counts && out_counts;
+  rc = timed_launch("gb_compact_scatter", [&] {
+    if (want_counts)
+      hipLaunchKernelGGL(k_compact_scatter<true>, dim3((uint32_t)ntiles), dim3(BLOCK),
+                         0, g.stream, (const double*)sums,
+                         (const unsigned long long*)rowcnt,
+                         (const unsigned long long*)counts, nvals, key_min, n_slots,
+                         d_tiles, (int64_t*)(*out_keys)->dptr, d_sum_ptrs, d_cnt_ptrs);
+    else
+      hipLaunchKernelGGL(k_compact_scatter<false>, dim3((uint32_t)ntiles), dim3(BLOCK),
+                         0, g.stream, (const double*)sums,
+                         (const unsigned long long*)rowcnt,
+                         (const unsigned long long*)counts, nvals, key_min, n_slots,
+                         d_tiles, (int64_t*)(*out_keys)->dptr, d_sum_ptrs, d_cnt_ptrs);
+  });
+  hipFreeAsync(d_tiles, g.stream);
+  hipFreeAsync(d_sum_ptrs, g.stream);
+  hipFreeAsync(d_cnt_ptrs, g.stream);
+  if (rc != HF_OK) return rc;
+  *n_groups = total;
+  return HF_OK;
+}
+
+// ---- profiling ----
+
+int hf_profiling(int enable) {
+  HF_NEED_INIT("hf_profiling");
+  g.profiling = enable != 0;
+  return HF_OK;
+}
+
+int hf_kernel_stats(const char* name, int64_t* launches, double* total_ms) {
+  HF_NEED_INIT("hf_kernel_stats");
+  int rc = resolve_stats("hf_kernel_stats");
+  if (rc != HF_OK) return rc;
+  auto it = g.stats.find(name);
+  if (it == g.stats.end()) {
+    *launches = 0;
+    *total_ms = 0.0;
+    return HF_OK;
+  }
+  *launches = it->second.first;
+  *total_ms = it->second.second;
+  return HF_OK;
+}
+
+int hf_kernel_stats_reset(void) {
+  HF_NEED_INIT("hf_kernel_stats_reset");
+  int rc = resolve_stats("hf_kernel_stats_reset");
+  if (rc != HF_OK) return rc;
+  g.stats.clear();
+  return HF_OK;
+}
+
+}  // extern "C"
