@@ -1,0 +1,147 @@
+"""HipDataframe — the core dataframe (reference L4) on device partitions.
+
+Mirrors ``modin/core/dataframe/pandas/dataframe/dataframe.py``:
+``PandasDataframe(partitions, index, columns, row_lengths, ..., dtypes)``
+(:82,161) with operators ``map`` (:2253), ``tree_reduce`` (:2208),
+``n_ary_op`` (:3851), ``groupby_reduce`` (:4530), ``from_pandas`` (:4592),
+and the lazy-metadata pattern of ``modin/core/dataframe/pandas/metadata/``
+(ModinIndex): a groupby result's index is a device key column materialized
+to a pandas.Index only on demand (and cached) — the D2H copy the reference
+performs in ``PartitionManager.get_indices`` (partition_manager.py:1220).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import pandas
+
+from .. import config
+from . import lib
+from .partition import DeviceBlock, HipDataframePartition
+from .partition_manager import HipDataframePartitionManager
+
+
+class DeviceIndex:
+    """Lazy index backed by a device int64 column (groupby keys)."""
+
+    def __init__(self, col: lib.ColumnRef, name=None):
+        self.col = col
+        self.name = name
+        self._cache = None
+
+    def materialize(self) -> pandas.Index:
+        if self._cache is None:
+            self._cache = pandas.Index(lib.get(self.col), name=self.name)
+        return self._cache
+
+    def __len__(self):
+        return self.col.length
+
+
+class HipDataframe:
+    _partition_mgr_cls = HipDataframePartitionManager
+
+    def __init__(self, partitions, index, columns, row_lengths, dtypes):
+        self._partitions = partitions          # list[HipDataframePartition], p×1
+        self._index = index                    # pandas.Index | DeviceIndex
+        self.columns = pandas.Index(columns)
+        self._row_lengths = row_lengths
+        self.dtypes = dtypes                   # pandas.Series name -> np.dtype
+
+    # ---- metadata ----
+    @property
+    def index(self) -> pandas.Index:
+        if isinstance(self._index, DeviceIndex):
+            return self._index.materialize()
+        return self._index
+
+    def __len__(self):
+        return sum(self._row_lengths)
+
+    # ---- ingestion (dataframe.py:4592) ----
+    @classmethod
+    def from_pandas(cls, df: pandas.DataFrame) -> "HipDataframe":
+        parts, row_lengths = cls._partition_mgr_cls.from_pandas(df)
+        return cls(parts, df.index, df.columns, row_lengths,
+                   df.dtypes.copy())
+
+    def to_pandas(self) -> pandas.DataFrame:
+        out = self._partition_mgr_cls.to_pandas(self._partitions)
+        out.index = self.index
+        out = out[list(self.columns)]
+        return out.astype(dict(self.dtypes))
+
+    # ---- Map (dataframe.py:2253) ----
+    def map(self, block_fn, lazy: bool = False, dtypes=None) -> "HipDataframe":
+        mgr = self._partition_mgr_cls
+        parts = (mgr.lazy_map_partitions(self._partitions, block_fn) if lazy
+                 else mgr.map_partitions(self._partitions, block_fn))
+        new_dtypes = dtypes if dtypes is not None else _peek_dtypes(parts, self)
+        return HipDataframe(parts, self._index, self.columns,
+                            self._row_lengths, new_dtypes)
+
+    # ---- Binary zip (dataframe.py:3851) ----
+    def n_ary_op(self, zip_fn, other: "HipDataframe") -> "HipDataframe":
+        if self._row_lengths != other._row_lengths:
+            if len(self) != len(other):
+                raise lib.HfError("n_ary_op: length mismatch")
+            # co-partition (dataframe.py:3709 _copartition): re-split the rhs
+            # to the lhs row splits via host round trip is NOT offered — the
+            # deterministic from_pandas chunking makes equal-length frames
+            # align; anything else is a later round.
+            raise lib.HfError("n_ary_op: frames are not co-partitioned")
+        parts = self._partition_mgr_cls.binary_partitions(
+            self._partitions, other._partitions, zip_fn
+        )
+        return HipDataframe(parts, self._index, self.columns,
+                            self._row_lengths, _peek_dtypes(parts, self))
+
+    # ---- TreeReduce (dataframe.py:2208) ----
+    def tree_reduce(self, col_names):
+        partials = self._partition_mgr_cls.reduce_partitions(
+            self._partitions, col_names
+        )
+        from ..distributed import allreduce_partials, is_active
+        if is_active():
+            partials = allreduce_partials(partials, list(col_names))
+        return partials
+
+    # ---- GroupByReduce (dataframe.py:4530) ----
+    def groupby_reduce(self, by: str, agg: str) -> "HipDataframe":
+        val_names = [c for c in self.columns if c != by]
+        want_counts = agg in ("count", "mean")
+        keys, sums, counts, n = self._partition_mgr_cls.groupby_reduce(
+            self._partitions, by, val_names, want_counts
+        )
+        if agg == "sum":
+            cols = {name: sums[i] for i, name in enumerate(val_names)}
+            dtypes = pandas.Series({n_: np.dtype(np.float64) for n_ in val_names})
+        elif agg == "count":
+            cols = {name: counts[i] for i, name in enumerate(val_names)}
+            dtypes = pandas.Series({n_: np.dtype(np.int64) for n_ in val_names})
+        else:  # mean = sums / counts (GroupbyReduceImpl mean shape, groupby.py:87)
+            cols = {}
+            for i, name in enumerate(val_names):
+                cnt_f = lib.cast_f64(counts[i])
+                cols[name] = lib.binary(lib.BIN_DIV, sums[i], cnt_f)
+            dtypes = pandas.Series({n_: np.dtype(np.float64) for n_ in val_names})
+        block = DeviceBlock(cols, n)
+        part = HipDataframePartition(block)
+        return HipDataframe([part], DeviceIndex(keys, name=by), val_names,
+                            [n], dtypes)
+
+    # ---- column selection (getitem_column_array device form) ----
+    def take_columns(self, names) -> "HipDataframe":
+        def sel(block: DeviceBlock) -> DeviceBlock:
+            return block.select(names)
+        parts = [p.add_to_apply_calls(sel) for p in self._partitions]
+        return HipDataframe(parts, self._index, names, self._row_lengths,
+                            self.dtypes[list(names)])
+
+
+def _peek_dtypes(parts, frame) -> pandas.Series:
+    """Derive result dtypes from the first partition's (drained) block."""
+    block = parts[0].block()
+    return pandas.Series({name: col.np_dtype for name, col in block.columns.items()})
+
+

@@ -1,0 +1,242 @@
+"""Partition manager: classmethods over the partition list.
+
+Mirrors ``modin/core/dataframe/pandas/partitioning/partition_manager.py``:
+``map_partitions`` (:708), ``lazy_map_partitions`` (:773),
+``groupby_reduce`` (:303 — map phase per partition, reduce phase across),
+``from_pandas``/``split_pandas_df_into_partitions`` (:1070/:1029) with the
+reference's chunking rule (``compute_chunksize``,
+modin/core/storage_formats/pandas/utils.py:28), and the BenchmarkMode
+barrier (:52-92, here a hipStreamSynchronize).
+
+MI355X redesign of GroupByReduce: the reference's map phase
+runs pandas ``groupby().sum()`` per partition and its reduce phase concats
+partials and groupbys again (algebra/groupby.py:124/:211).  Here the map
+phase accumulates every partition into ONE dense key-indexed device table
+(hardware f64 atomics), the cross-GPU reduce is an RCCL all-reduce of that
+table (distributed.py), and the final "reduce" is an on-device compaction to
+ascending present keys — no concat copy, no second hash pass.
+"""
+
+from __future__ import annotations
+
+import math
+
+import numpy as np
+import pandas
+
+from .. import config
+from ..distributed import maybe_allreduce_table
+from . import lib
+from .partition import DeviceBlock, HipDataframePartition
+
+
+def compute_chunksize(n: int, num_splits: int, min_size: int) -> int:
+    """Reference: modin/core/storage_formats/pandas/utils.py:28."""
+    chunk = math.ceil(n / num_splits) if num_splits else n
+    return max(chunk, min_size)
+
+
+def wait_if_benchmark_mode(fn):
+    """Reference: partition_manager.py:52-92 decorator."""
+    def wrapper(*args, **kwargs):
+        out = fn(*args, **kwargs)
+        if config.BenchmarkMode.get():
+            lib.sync()
+        return out
+    return wrapper
+
+
+class GroupbyTable:
+    """Dense key-indexed aggregation table (device buffers).
+
+    Layout matches hf_groupby_accum: sums f64[nvals][n_slots],
+    rowcnt u64[n_slots], counts u64[nvals][n_slots] (optional).
+    Buffers may be library-allocated (single process) or torch-allocated
+    (multi-GPU, so RCCL can all-reduce them in place) — see distributed.py.
+    """
+
+    def __init__(self, nvals: int, key_min: int, n_slots: int, want_counts: bool):
+        self.nvals = nvals
+        self.key_min = key_min
+        self.n_slots = n_slots
+        self.want_counts = want_counts
+        self._torch_tensors = None
+        from ..distributed import is_active, alloc_table_torch
+        if is_active():
+            self._torch_tensors, self.sums, self.rowcnt, self.counts = \
+                alloc_table_torch(nvals, n_slots, want_counts)
+        else:
+            self.sums = lib.alloc_raw(8 * nvals * n_slots)
+            self.rowcnt = lib.alloc_raw(8 * n_slots)
+            self.counts = lib.alloc_raw(8 * nvals * n_slots) if want_counts else 0
+            lib.memset_raw(self.sums, 0, 8 * nvals * n_slots)
+            lib.memset_raw(self.rowcnt, 0, 8 * n_slots)
+            if want_counts:
+                lib.memset_raw(self.counts, 0, 8 * nvals * n_slots)
+
+    def free(self):
+        if self._torch_tensors is not None:
+            self._torch_tensors = None
+        else:
+            lib.free_raw(self.sums)
+            lib.free_raw(self.rowcnt)
+            if self.counts:
+                lib.free_raw(self.counts)
+        self.sums = self.rowcnt = self.counts = 0
+
+
+class HipDataframePartitionManager:
+    """Classmethod namespace over lists of HipDataframePartition (p×1 grid)."""
+
+    _partition_class = HipDataframePartition
+
+    # ---- ingestion (partition_manager.py:1070) ----
+    @classmethod
+    def from_pandas(cls, df: pandas.DataFrame, num_splits=None):
+        if num_splits is None:
+            num_splits = config.NPartitions.get()
+        n = len(df)
+        chunk = compute_chunksize(n, num_splits, config.MinRowPartitionSize.get())
+        parts = []
+        row_lengths = []
+        start = 0
+        while start < n:
+            stop = min(start + chunk, n)
+            parts.append(cls._partition_class.put(df.iloc[start:stop]))
+            row_lengths.append(stop - start)
+            start = stop
+        if not parts:  # empty frame: keep one empty partition
+            parts = [cls._partition_class.put(df)]
+            row_lengths = [0]
+        return parts, row_lengths
+
+    @classmethod
+    def to_pandas(cls, parts, index=None, columns=None):
+        frames = [p.get() for p in parts]
+        out = pandas.concat(frames, ignore_index=True) if len(frames) > 1 else frames[0]
+        if index is not None:
+            out.index = index
+        if columns is not None:
+            out = out[list(columns)]
+        return out
+
+    # ---- Map (partition_manager.py:708/:773) ----
+    @classmethod
+    @wait_if_benchmark_mode
+    def map_partitions(cls, parts, func):
+        return [p.apply(func) for p in parts]
+
+    @classmethod
+    def lazy_map_partitions(cls, parts, func):
+        return [p.add_to_apply_calls(func) for p in parts]
+
+    # ---- Binary / n-ary zip (dataframe.py:3851 n_ary_op device form) ----
+    @classmethod
+    @wait_if_benchmark_mode
+    def binary_partitions(cls, left, right, func):
+        if len(left) != len(right):
+            raise lib.HfError(
+                "n_ary_op requires co-partitioned frames (same row splits) — "
+                "repartition first"
+            )
+        out = []
+        for lp, rp in zip(left, right):
+            lp.drain_call_queue()
+            rp.drain_call_queue()
+            out.append(HipDataframePartition(func(lp.block(), rp.block())))
+        return out
+
+    # ---- TreeReduce (partition_manager.py map+map_axis_partitions shape,
+    #      dataframe.py:2244-2247) ----
+    @classmethod
+    @wait_if_benchmark_mode
+    def reduce_partitions(cls, parts, col_names):
+        """Per-partition single-pass partials, combined on host.
+
+        Returns dict name -> dict(sum, count, mn, mx, isum, imn, imx) of the
+        across-partition combination (the reference's reduce phase over a
+        p×1 grid is a p-way combine of 1-row partials).
+        """
+        partials = {name: [] for name in col_names}
+        for p in parts:
+            block = p.block()
+            for name in col_names:
+                partials[name].append(lib.reduce(block.columns[name]))
+        out = {}
+        for name, rs in partials.items():
+            tot_cnt = sum(r.count for r in rs)
+            have = [r for r in rs if r.count > 0]
+            dtype = None
+            out[name] = {
+                "sum": float(sum(r.sum for r in have)) if have else 0.0,
+                "isum": int(sum(r.isum for r in have)) if have else 0,
+                "count": int(tot_cnt),
+                "mn": min((r.mn for r in have), default=float("nan")),
+                "mx": max((r.mx for r in have), default=float("nan")),
+                "imn": min((r.imn for r in have), default=0),
+                "imx": max((r.imx for r in have), default=0),
+            }
+        return out
+
+    # ---- GroupByReduce (partition_manager.py:303) ----
+    @classmethod
+    @wait_if_benchmark_mode
+    def groupby_reduce(cls, parts, by_name, val_names, want_counts):
+        """Dense-table groupby: returns (keys_col, sum_cols, count_cols, n).
+
+        Map phase (algebra/groupby.py:124 device form): hf_groupby_accum per
+        partition into one shared table.  Reduce phase (:211 device form):
+        RCCL all-reduce across ranks (if distributed) + on-device compaction.
+        """
+        # key range via the i64 reduce kernel
+        kmin, kmax, total_rows = None, None, 0
+        key_cols = []
+        val_cols_per_part = []
+        for p in parts:
+            block = p.block()
+            kcol = block.columns[by_name]
+            if kcol.dtype_code != lib.HF_INT64:
+                raise lib.HfError(
+                    f"groupby key column {by_name!r} must be int64 (dense-key "
+                    "path); hashed/float keys are a later round"
+                )
+            key_cols.append(kcol)
+            vals = [lib.cast_f64(block.columns[v]) for v in val_names]
+            val_cols_per_part.append(vals)
+            if kcol.length:
+                r = lib.reduce(kcol)
+                kmin = r.imn if kmin is None else min(kmin, r.imn)
+                kmax = r.imx if kmax is None else max(kmax, r.imx)
+            total_rows += kcol.length
+
+        kmin, kmax = maybe_allreduce_keyrange(kmin, kmax)
+        if kmin is None:  # globally empty
+            kmin, kmax = 0, -1
+        n_slots = kmax - kmin + 1
+        if n_slots < 0:
+            n_slots = 0
+        if n_slots > config.MaxGroupbySlots.get():
+            raise lib.HfError(
+                f"groupby key range {n_slots} exceeds MaxGroupbySlots "
+                f"({config.MaxGroupbySlots.get()}): the dense-key table does "
+                "not apply; hash aggregation is a later round"
+            )
+        n_slots = max(n_slots, 1)
+        table = GroupbyTable(len(val_names), kmin, n_slots, want_counts)
+        for kcol, vals in zip(key_cols, val_cols_per_part):
+            if kcol.length:
+                lib.groupby_accum(kcol, vals, kmin, n_slots, table.sums,
+                                  table.rowcnt, table.counts)
+        maybe_allreduce_table(table)
+        keys, sums, counts, n = lib.groupby_compact(
+            table.sums, table.rowcnt, table.counts, len(val_names), kmin, n_slots
+        )
+        table.free()
+        return keys, sums, counts, n
+
+
+def maybe_allreduce_keyrange(kmin, kmax):
+    from ..distributed import allreduce_minmax, is_active
+    if not is_active():
+        return kmin, kmax
+    return allreduce_minmax(kmin, kmax)

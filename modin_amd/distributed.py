@@ -1,0 +1,181 @@
+"""Multi-GPU execution: one process per GPU, torch.distributed over RCCL.
+
+The reference has no explicit collectives — data movement is Ray/Dask futures
+(SURVEY.md §5.8).  The MI355X-native equivalent: every rank owns the row
+shard of the frame it built (SPMD), map/binary/filter run with zero exchange
+("weak" scaling), and the two exchange points are
+
+  * TreeReduce: host-side combine of tiny per-rank partials (gloo/nccl
+    all-reduce of a few doubles),
+  * GroupByReduce: RCCL all-reduce of the dense key-indexed table
+    (sums f64 + rowcnt/counts u64) over xGMI — the device form of the
+    reference's reduce phase (algebra/groupby.py:211), chosen per SURVEY §8e
+    option (i) for bounded key spaces (1e6 keys → 8–24 MB buffers).
+
+torch is plumbing here: device allocation for RCCL-visible buffers and the
+process group.  Kernels remain the hipframe HIP kernels, which write into the
+torch-allocated table via raw device pointers on the hipframe stream
+(hf_sync before the collective, torch.cuda.synchronize after).
+"""
+
+from __future__ import annotations
+
+import os
+
+_state = {"active": False, "rank": 0, "world": 1, "device": None, "backend": None}
+
+
+def is_active() -> bool:
+    return _state["active"]
+
+
+def rank() -> int:
+    return _state["rank"]
+
+
+def world_size() -> int:
+    return _state["world"]
+
+
+def init_from_env(backend=None, gpu: bool = True):
+    """Initialise from torchrun env (RANK/WORLD_SIZE/LOCAL_RANK)."""
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world <= 1:
+        return False
+    import torch
+    import torch.distributed as dist
+    r = int(os.environ.get("RANK", "0"))
+    local = int(os.environ.get("LOCAL_RANK", str(r)))
+    if backend is None:
+        backend = "nccl" if (gpu and torch.cuda.is_available()) else "gloo"
+    if not dist.is_initialized():
+        dist.init_process_group(backend=backend)
+    _state.update(active=True, rank=r, world=world, backend=backend)
+    if backend == "nccl":
+        torch.cuda.set_device(local)
+        _state["device"] = f"cuda:{local}"
+        os.environ["MODIN_AMD_GPU"] = str(local)
+    else:
+        _state["device"] = "cpu"
+    return True
+
+
+def shutdown():
+    import torch.distributed as dist
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    _state.update(active=False, rank=0, world=1, device=None, backend=None)
+
+
+def barrier():
+    if not is_active():
+        return
+    import torch.distributed as dist
+    dist.barrier()
+
+
+def allreduce_minmax(kmin, kmax):
+    """Global [min, max] of per-rank key ranges (None on empty shards)."""
+    import torch
+    import torch.distributed as dist
+    dev = _state["device"]
+    big = 2**62
+    t = torch.tensor(
+        [kmin if kmin is not None else big, -(kmax if kmax is not None else -big)],
+        dtype=torch.int64, device=dev,
+    )
+    dist.all_reduce(t, op=dist.ReduceOp.MIN)
+    lo, hi = int(t[0].item()), -int(t[1].item())
+    if lo == big:
+        return None, None
+    return lo, hi
+
+
+def allreduce_partials(partials: dict, names: list) -> dict:
+    """Cross-rank reduce phase of TreeReduce: combine the per-rank 1-row
+    partial dicts (sum/count/min/max) — the device grid's p-way combine
+    (dataframe.py:2244-2247) extended over ranks with three tiny collectives.
+    Empty-shard ranks contribute identity elements."""
+    import torch
+    import torch.distributed as dist
+    dev = _state["device"]
+    inf = float("inf")
+    big = 2**62
+    add, mns, mxs = [], [], []
+    for n in names:
+        p = partials[n]
+        add += [p["sum"], float(p["count"])]
+        mns += [p["mn"] if p["count"] else inf,
+                float(p["imn"]) if p["count"] else inf]
+        mxs += [p["mx"] if p["count"] else -inf,
+                float(p["imx"]) if p["count"] else -inf]
+    t_add = torch.tensor(add, dtype=torch.float64, device=dev)
+    t_mn = torch.tensor(mns, dtype=torch.float64, device=dev)
+    t_mx = torch.tensor(mxs, dtype=torch.float64, device=dev)
+    t_isum = torch.tensor([partials[n]["isum"] for n in names],
+                          dtype=torch.int64, device=dev)
+    t_iminmax = torch.tensor(
+        [(partials[n]["imn"] if partials[n]["count"] else big) for n in names]
+        + [-(partials[n]["imx"] if partials[n]["count"] else -big) for n in names],
+        dtype=torch.int64, device=dev)
+    dist.all_reduce(t_add)
+    dist.all_reduce(t_isum)
+    dist.all_reduce(t_mn, op=dist.ReduceOp.MIN)
+    dist.all_reduce(t_mx, op=dist.ReduceOp.MAX)
+    dist.all_reduce(t_iminmax, op=dist.ReduceOp.MIN)
+    out = {}
+    for i, n in enumerate(names):
+        cnt = int(t_add[2 * i + 1].item())
+        out[n] = {
+            "sum": float(t_add[2 * i].item()),
+            "count": cnt,
+            "isum": int(t_isum[i].item()),
+            "mn": float(t_mn[2 * i].item()) if cnt else float("nan"),
+            "mx": float(t_mx[2 * i].item()) if cnt else float("nan"),
+            "imn": int(t_iminmax[i].item()) if cnt else 0,
+            "imx": -int(t_iminmax[len(names) + i].item()) if cnt else 0,
+        }
+    return out
+
+
+def allreduce_scalars(values):
+    """Sum-all-reduce a small list of floats (TreeReduce partial combine)."""
+    import torch
+    import torch.distributed as dist
+    t = torch.tensor(values, dtype=torch.float64, device=_state["device"])
+    dist.all_reduce(t)
+    return t.tolist()
+
+
+def alloc_table_torch(nvals: int, n_slots: int, want_counts: bool):
+    """Zeroed RCCL-reducible table buffers on this rank's GPU.
+
+    Returns (keepalive, sums_ptr, rowcnt_ptr, counts_ptr).
+    """
+    import torch
+    dev = _state["device"]
+    sums = torch.zeros(nvals * n_slots, dtype=torch.float64, device=dev)
+    rowcnt = torch.zeros(n_slots, dtype=torch.int64, device=dev)
+    counts = (torch.zeros(nvals * n_slots, dtype=torch.int64, device=dev)
+              if want_counts else None)
+    torch.cuda.synchronize()
+    keep = (sums, rowcnt, counts)
+    return keep, sums.data_ptr(), rowcnt.data_ptr(), \
+        (counts.data_ptr() if counts is not None else 0)
+
+
+def maybe_allreduce_table(table) -> None:
+    """RCCL all-reduce of the dense groupby table (reduce phase across GPUs)."""
+    if not is_active():
+        return
+    import torch
+    import torch.distributed as dist
+    from .core import lib
+    lib.sync()  # hipframe-stream accumulation must be visible to RCCL
+    sums, rowcnt, counts = table._torch_tensors
+    dist.all_reduce(sums)
+    dist.all_reduce(rowcnt)
+    if counts is not None:
+        dist.all_reduce(counts)
+    if _state["backend"] == "nccl":
+        torch.cuda.synchronize()  # collective must land before compaction reads

@@ -1,0 +1,236 @@
+"""modin_amd.pandas — the drop-in user API (reference L1) for the hot path.
+
+Mirrors ``modin/pandas``'s DataFrame/Series/GroupBy surface for the operators
+the HipNative backend accelerates: construction/from_pandas,
+elementwise arithmetic (+ - * /), fillna/abs, sum/mean/count/min/max, column
+projection and ``groupby(key).sum()/count()/mean()/agg(...)``
+(modin/pandas/dataframe.py:2188 sum; modin/pandas/groupby.py:1330
+DataFrameGroupBy.sum -> _wrap_aggregation).  Everything else the reference's
+API offers is out of scope per SURVEY.md §8 and raises loudly.
+
+Reductions return pandas.Series (the reference's API layer also lowers
+1×N reduce results into Series via ``_reduce_dimension``).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import pandas
+
+from ..core import lib
+from ..query_compiler import HipQueryCompiler
+
+__all__ = ["DataFrame", "Series", "from_pandas"]
+
+
+def from_pandas(df: pandas.DataFrame) -> "DataFrame":
+    return DataFrame(query_compiler=HipQueryCompiler.from_pandas(df))
+
+
+class _HipPandasBase:
+    _query_compiler: HipQueryCompiler
+
+    # ---- reductions ----
+    def sum(self, **kwargs):
+        return self._lower(self._query_compiler.sum(**kwargs))
+
+    def mean(self, **kwargs):
+        return self._lower(self._query_compiler.mean(**kwargs))
+
+    def count(self, **kwargs):
+        return self._lower(self._query_compiler.count(**kwargs))
+
+    def min(self, **kwargs):
+        return self._lower(self._query_compiler.min(**kwargs))
+
+    def max(self, **kwargs):
+        return self._lower(self._query_compiler.max(**kwargs))
+
+    # ---- arithmetic ----
+    def __add__(self, other):
+        return self._rewrap(type(self._query_compiler).add(self._query_compiler,
+                                                           _unwrap(other)))
+
+    def __radd__(self, other):
+        return self.__add__(other)
+
+    def __sub__(self, other):
+        return self._rewrap(type(self._query_compiler).sub(self._query_compiler,
+                                                           _unwrap(other)))
+
+    def __rsub__(self, other):
+        return self._rewrap(type(self._query_compiler).rsub(self._query_compiler,
+                                                            _unwrap(other)))
+
+    def __mul__(self, other):
+        return self._rewrap(type(self._query_compiler).mul(self._query_compiler,
+                                                           _unwrap(other)))
+
+    def __rmul__(self, other):
+        return self.__mul__(other)
+
+    def __truediv__(self, other):
+        return self._rewrap(type(self._query_compiler).truediv(self._query_compiler,
+                                                               _unwrap(other)))
+
+    def __rtruediv__(self, other):
+        return self._rewrap(type(self._query_compiler).rtruediv(self._query_compiler,
+                                                                _unwrap(other)))
+
+    def add(self, other):
+        return self.__add__(other)
+
+    def sub(self, other):
+        return self.__sub__(other)
+
+    def mul(self, other):
+        return self.__mul__(other)
+
+    def truediv(self, other):
+        return self.__truediv__(other)
+
+    def fillna(self, value):
+        return self._rewrap(type(self._query_compiler).fillna(self._query_compiler,
+                                                              float(value)))
+
+    def abs(self):
+        return self._rewrap(type(self._query_compiler).abs(self._query_compiler,
+                                                           None))
+
+    def __getattr__(self, name):
+        raise AttributeError(
+            f"{type(self).__name__}.{name} is outside the HipNative hot-path "
+            "scope (SURVEY.md §8) — not implemented in this round"
+        )
+
+
+def _unwrap(other):
+    return other._query_compiler if isinstance(other, _HipPandasBase) else other
+
+
+class DataFrame(_HipPandasBase):
+    def __init__(self, data=None, query_compiler=None):
+        if query_compiler is not None:
+            self._query_compiler = query_compiler
+            return
+        if isinstance(data, pandas.DataFrame):
+            pdf = data
+        elif isinstance(data, dict):
+            pdf = pandas.DataFrame(data)
+        else:
+            raise lib.HfError(
+                "DataFrame accepts a dict of columns or a pandas.DataFrame"
+            )
+        self._query_compiler = HipQueryCompiler.from_pandas(pdf)
+
+    def _rewrap(self, qc):
+        return DataFrame(query_compiler=qc)
+
+    def _lower(self, series):
+        return series  # reductions come back as pandas.Series already
+
+    @property
+    def columns(self):
+        return self._query_compiler.columns
+
+    @property
+    def index(self):
+        return self._query_compiler.index
+
+    @property
+    def dtypes(self):
+        return self._query_compiler.dtypes
+
+    @property
+    def shape(self):
+        return (len(self._query_compiler), len(self.columns))
+
+    def __len__(self):
+        return len(self._query_compiler)
+
+    def __getitem__(self, key):
+        if isinstance(key, str):
+            return Series(query_compiler=self._query_compiler.getitem_column_array(
+                [key]), name=key)
+        if isinstance(key, (list, tuple, pandas.Index)):
+            return DataFrame(
+                query_compiler=self._query_compiler.getitem_column_array(list(key))
+            )
+        raise lib.HfError("only column selection is supported")
+
+    def groupby(self, by: str) -> "DataFrameGroupBy":
+        if not isinstance(by, str) or by not in list(self.columns):
+            raise lib.HfError("groupby(by=<column name>) only")
+        return DataFrameGroupBy(self, by)
+
+    def to_pandas(self) -> pandas.DataFrame:
+        return self._query_compiler.to_pandas()
+
+    def _to_pandas(self) -> pandas.DataFrame:  # reference-compatible alias
+        return self.to_pandas()
+
+    def __repr__(self):
+        return f"modin_amd.DataFrame({self.shape[0]}x{self.shape[1]} on device)"
+
+
+class Series(_HipPandasBase):
+    def __init__(self, data=None, query_compiler=None, name=None):
+        self.name = name
+        if query_compiler is not None:
+            self._query_compiler = query_compiler
+            return
+        if isinstance(data, pandas.Series):
+            pdf = data.to_frame(name=data.name or 0)
+            self.name = data.name
+        else:
+            raise lib.HfError("Series accepts a pandas.Series")
+        self._query_compiler = HipQueryCompiler.from_pandas(pdf)
+
+    def _rewrap(self, qc):
+        return Series(query_compiler=qc, name=self.name)
+
+    def _lower(self, series):
+        # a Series reduction is a scalar
+        return series.iloc[0]
+
+    def __len__(self):
+        return len(self._query_compiler)
+
+    def to_pandas(self) -> pandas.Series:
+        df = self._query_compiler.to_pandas()
+        s = df[df.columns[0]]
+        s.name = self.name
+        return s
+
+    def _to_pandas(self) -> pandas.Series:
+        return self.to_pandas()
+
+    def __repr__(self):
+        return f"modin_amd.Series(len={len(self)} on device)"
+
+
+class DataFrameGroupBy:
+    """Mirrors modin/pandas/groupby.py DataFrameGroupBy for the reduce aggs
+    (sum :1330, count, mean -> _wrap_aggregation -> qc.groupby_<agg>)."""
+
+    def __init__(self, df: DataFrame, by: str):
+        self._df = df
+        self._by = by
+
+    def _agg(self, how: str) -> DataFrame:
+        qc = self._df._query_compiler.groupby_agg(self._by, how)
+        return DataFrame(query_compiler=qc)
+
+    def sum(self):
+        return self._agg("sum")
+
+    def count(self):
+        return self._agg("count")
+
+    def mean(self):
+        return self._agg("mean")
+
+    def agg(self, how):
+        if isinstance(how, str):
+            return self._agg(how)
+        raise lib.HfError("groupby.agg accepts a single agg name this round")

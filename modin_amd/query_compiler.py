@@ -1,0 +1,94 @@
+"""HipQueryCompiler — L2 of the stack, reusing the reference's method names.
+
+Mirrors ``modin/core/storage_formats/pandas/query_compiler.py``
+(``PandasQueryCompiler`` :279,297 — wraps one ``_modin_frame``): every hot op
+is bound to an operator template exactly as the reference binds them:
+  sum   = TreeReduce.register(...)        (query_compiler.py:984)
+  add   = Binary.register(...)            (:535)
+  fillna = Map.register(...)              (:2710)
+  groupby_sum = GroupByReduce agg table   (:3747 via
+                storage_formats/pandas/groupby.py:75 build_qc_method)
+The registered kernels are hipframe device ops, not pandas callables.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import pandas
+
+from .algebra import Binary, GroupByReduce, Map, TreeReduce
+from .core import lib
+from .core.dataframe import HipDataframe
+
+
+class HipQueryCompiler:
+    def __init__(self, modin_frame: HipDataframe):
+        self._modin_frame = modin_frame
+
+    @property
+    def __constructor__(self):
+        return type(self)
+
+    # ---- ingestion ----
+    @classmethod
+    def from_pandas(cls, df: pandas.DataFrame) -> "HipQueryCompiler":
+        return cls(HipDataframe.from_pandas(df))
+
+    def to_pandas(self) -> pandas.DataFrame:
+        return self._modin_frame.to_pandas()
+
+    @property
+    def columns(self):
+        return self._modin_frame.columns
+
+    @property
+    def index(self):
+        return self._modin_frame.index
+
+    @property
+    def dtypes(self):
+        return self._modin_frame.dtypes
+
+    def __len__(self):
+        return len(self._modin_frame)
+
+    # ---- Map ops (query_compiler.py:2036 abs, :2710 fillna) ----
+    abs = Map.register(lib.MAP_ABS)
+    neg = Map.register(lib.MAP_NEG)
+    fillna = Map.register(lib.MAP_FILLNA, f64_only=True)
+
+    # ---- Binary ops (query_compiler.py:535 add & friends) ----
+    add = Binary.register(lib.MAP_ADD, lib.BIN_ADD)
+    sub = Binary.register(lib.MAP_SUB, lib.BIN_SUB)
+    rsub = Binary.register(lib.MAP_RSUB, lib.BIN_SUB)
+    mul = Binary.register(lib.MAP_MUL, lib.BIN_MUL)
+    truediv = Binary.register(lib.MAP_DIV, lib.BIN_DIV)
+    rtruediv = Binary.register(lib.MAP_RDIV, lib.BIN_DIV)
+
+    # ---- TreeReduce (query_compiler.py:984 sum etc.) ----
+    sum = TreeReduce.register("sum")
+    count = TreeReduce.register("count")
+    mean = TreeReduce.register("mean")
+    min = TreeReduce.register("min")
+    max = TreeReduce.register("max")
+
+    # ---- GroupByReduce (query_compiler.py:3747 groupby_sum) ----
+    groupby_sum = GroupByReduce.register("sum")
+    groupby_count = GroupByReduce.register("count")
+    groupby_mean = GroupByReduce.register("mean")
+
+    def groupby_agg(self, by: str, agg: str) -> "HipQueryCompiler":
+        fn = {
+            "sum": type(self).groupby_sum,
+            "count": type(self).groupby_count,
+            "mean": type(self).groupby_mean,
+        }.get(agg)
+        if fn is None:
+            raise lib.HfError(
+                f"groupby agg {agg!r} not implemented on the HipNative backend"
+            )
+        return fn(self, by)
+
+    # ---- projection ----
+    def getitem_column_array(self, names) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.take_columns(list(names)))
