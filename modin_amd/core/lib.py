@@ -57,7 +57,7 @@ def so_path() -> str:
     return os.path.join(os.path.dirname(__file__), "..", "csrc", _LIB_NAME)
 
 
-_lock = threading.Lock()
+_lock = threading.RLock()
 _dll = None
 _inited_gpu = None
 

@@ -170,8 +170,9 @@ def maybe_allreduce_table(table) -> None:
         return
     import torch
     import torch.distributed as dist
-    from .core import lib
-    lib.sync()  # hipframe-stream accumulation must be visible to RCCL
+    if _state["backend"] == "nccl":
+        from .core import lib
+        lib.sync()  # hipframe-stream accumulation must be visible to RCCL
     sums, rowcnt, counts = table._torch_tensors
     dist.all_reduce(sums)
     dist.all_reduce(rowcnt)

@@ -1,0 +1,201 @@
+"""ORACLE — TEST INFRASTRUCTURE ONLY (see oracle/__init__.py header).
+
+Numpy restatement of the reference's partition-operator semantics, each
+function citing the reference code it follows.  No pandas in the compute
+path of the oracle itself (pandas appears only in golden-vector generation,
+which runs the real reference); numpy is the arithmetic substrate, matching
+where the reference's own arithmetic lives (pandas/numpy inside each
+partition — SURVEY.md §2 "Implementation language").
+"""
+
+from __future__ import annotations
+
+import math
+
+import numpy as np
+
+
+# ---------------------------------------------------------------------------
+# ingestion / chunking — modin/core/storage_formats/pandas/utils.py:28
+# (compute_chunksize) + partition_manager.py:1029
+# (split_pandas_df_into_partitions): chunk = ceil(n / num_splits), floored at
+# MinRowPartitionSize; row slices [i*chunk, (i+1)*chunk).
+# ---------------------------------------------------------------------------
+
+def split_row_counts(n: int, num_splits: int, min_size: int = 32) -> list:
+    chunk = max(math.ceil(n / num_splits) if num_splits else n, min_size)
+    out = []
+    start = 0
+    while start < n:
+        stop = min(start + chunk, n)
+        out.append(stop - start)
+        start = stop
+    return out or [0]
+
+
+# ---------------------------------------------------------------------------
+# Map — modin/core/dataframe/algebra/map.py:28 (function applied blockwise;
+# blockwise == global for elementwise ops).  Ops per
+# query_compiler.py bindings: add/sub/mul/div scalars (:535 Binary scalar
+# branch -> lazy map), fillna (:2710), abs (:2036).
+# ---------------------------------------------------------------------------
+
+def map_op(op: str, x: np.ndarray, scalar=None) -> np.ndarray:
+    if op == "add":
+        return x + scalar
+    if op == "sub":
+        return x - scalar
+    if op == "rsub":
+        return scalar - x
+    if op == "mul":
+        return x * scalar
+    if op == "div":
+        return x / scalar
+    if op == "rdiv":
+        return scalar / x
+    if op == "fillna":
+        y = x.astype(np.float64, copy=True)
+        y[np.isnan(y)] = scalar
+        return y
+    if op == "abs":
+        return np.abs(x)
+    if op == "neg":
+        return -x
+    raise ValueError(op)
+
+
+# ---------------------------------------------------------------------------
+# Binary frame op — algebra/binary.py:420 frame branch -> n_ary_op
+# (dataframe.py:3851): row-aligned elementwise zip.
+# ---------------------------------------------------------------------------
+
+def binary_op(op: str, a: np.ndarray, b: np.ndarray) -> np.ndarray:
+    if op == "add":
+        return a + b
+    if op == "sub":
+        return a - b
+    if op == "mul":
+        return a * b
+    if op == "div":
+        return a.astype(np.float64) / b.astype(np.float64)
+    raise ValueError(op)
+
+
+# ---------------------------------------------------------------------------
+# TreeReduce — algebra/tree_reduce.py:63 -> dataframe.py:2208: per-partition
+# map (pandas DataFrame.sum & friends: NaN-skipping), then axis reduce.
+# pandas semantics restated:
+#   sum:   nansum, 0.0 for all-NaN/empty (min_count=0)
+#   count: non-NaN count
+#   mean:  nansum/count, NaN if count==0
+#   min/max: NaN-skipping, NaN if count==0
+# int64 columns have no NaN; sum wraps like numpy int64.
+# ---------------------------------------------------------------------------
+
+def reduce_op(op: str, x: np.ndarray):
+    if x.dtype == np.int64:
+        if op == "sum":
+            return np.add.reduce(x, dtype=np.int64) if x.size else np.int64(0)
+        if op == "count":
+            return int(x.size)
+        if op == "mean":
+            return float(np.add.reduce(x, dtype=np.float64) / x.size) if x.size \
+                else float("nan")
+        if op == "min":
+            return np.min(x) if x.size else float("nan")
+        if op == "max":
+            return np.max(x) if x.size else float("nan")
+        raise ValueError(op)
+    valid = ~np.isnan(x)
+    cnt = int(valid.sum())
+    if op == "count":
+        return cnt
+    if op == "sum":
+        return float(x[valid].sum()) if cnt else 0.0
+    if op == "mean":
+        return float(x[valid].sum() / cnt) if cnt else float("nan")
+    if op == "min":
+        return float(x[valid].min()) if cnt else float("nan")
+    if op == "max":
+        return float(x[valid].max()) if cnt else float("nan")
+    raise ValueError(op)
+
+
+# ---------------------------------------------------------------------------
+# GroupByReduce — algebra/groupby.py:124 (map: per-partition
+# groupby(by, as_index=True, observed=True).<agg>()) and :211 (reduce:
+# concat partials, groupby(level=0).<agg'>()), with the map/reduce pairing
+# of storage_formats/pandas/groupby.py:237-248:
+#   sum  -> (sum, sum);  count -> (count, sum);
+#   mean -> map {sum,count}, reduce sum then divide (:87-113).
+# pandas result shape: ascending unique keys as the index; a key whose
+# values are all NaN is PRESENT with sum 0.0 / count 0.
+# Restated over dense int64 keys with np.bincount (the arithmetic is the
+# same reassociated per-key summation).
+# ---------------------------------------------------------------------------
+
+def groupby_agg(keys: np.ndarray, vals: dict, agg: str):
+    """Global groupby over int64 keys.  Returns (unique_keys, {name: agg})."""
+    assert keys.dtype == np.int64
+    if keys.size == 0:
+        return np.empty(0, np.int64), {
+            n: np.empty(0, np.int64 if agg == "count" else np.float64)
+            for n in vals
+        }
+    kmin = keys.min()
+    shifted = keys - kmin
+    n_slots = int(shifted.max()) + 1
+    rowcnt = np.bincount(shifted, minlength=n_slots)
+    present = rowcnt > 0
+    out_keys = (np.nonzero(present)[0] + kmin).astype(np.int64)
+    out = {}
+    for name, v in vals.items():
+        v = np.asarray(v, dtype=np.float64)
+        valid = ~np.isnan(v)
+        sums = np.bincount(shifted[valid], weights=v[valid], minlength=n_slots)
+        cnts = np.bincount(shifted[valid], minlength=n_slots)
+        if agg == "sum":
+            out[name] = sums[present]
+        elif agg == "count":
+            out[name] = cnts[present].astype(np.int64)
+        elif agg == "mean":
+            with np.errstate(invalid="ignore", divide="ignore"):
+                out[name] = sums[present] / cnts[present]
+        else:
+            raise ValueError(agg)
+    return out_keys, out
+
+
+def partitioned_groupby_agg(keys: np.ndarray, vals: dict, agg: str,
+                            num_splits: int, min_size: int = 32):
+    """The two-phase form the reference actually runs (map per partition,
+    reduce across partitions) — used to check that partitioning does not
+    change the result beyond fp reassociation."""
+    counts = split_row_counts(len(keys), num_splits, min_size)
+    offs = np.cumsum([0] + counts)
+    # map phase: per-partition partials (sum & count cover all three aggs)
+    partial = {}
+    for i in range(len(counts)):
+        sl = slice(offs[i], offs[i + 1])
+        uk, psums = groupby_agg(keys[sl], {n: v[sl] for n, v in vals.items()}, "sum")
+        _, pcnts = groupby_agg(keys[sl], {n: v[sl] for n, v in vals.items()}, "count")
+        for j, k in enumerate(uk):
+            acc = partial.setdefault(int(k), {n: [0.0, 0] for n in vals})
+            for n in vals:
+                acc[n][0] += psums[n][j]
+                acc[n][1] += pcnts[n][j]
+    out_keys = np.array(sorted(partial.keys()), dtype=np.int64)
+    out = {}
+    for n in vals:
+        if agg == "sum":
+            out[n] = np.array([partial[int(k)][n][0] for k in out_keys])
+        elif agg == "count":
+            out[n] = np.array([partial[int(k)][n][1] for k in out_keys],
+                              dtype=np.int64)
+        else:
+            out[n] = np.array([
+                partial[int(k)][n][0] / partial[int(k)][n][1]
+                if partial[int(k)][n][1] else float("nan")
+                for k in out_keys
+            ])
+    return out_keys, out

@@ -1,0 +1,118 @@
+"""CPU tier: the multi-GPU exchange logic over gloo, world_size=2.
+
+Covers the two collectives of the distributed path
+without HIP: (a) the dense groupby-table all-reduce — per-rank shard tables
+built by the oracle must merge to the global oracle table; (b) the
+TreeReduce partial combine; (c) the key-range min/max.
+"""
+
+import multiprocessing as mp
+import os
+
+import numpy as np
+import pytest
+
+import oracle
+
+
+def _worker(rank, world, port, fail_q):
+    try:
+        os.environ.update({
+            "RANK": str(rank), "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        })
+        import torch
+        import modin_amd.distributed as dist_mod
+        assert dist_mod.init_from_env(backend="gloo", gpu=False)
+
+        rng = np.random.default_rng(123)  # same on both ranks
+        n = 20_000
+        keys = rng.integers(-10, 90, n).astype(np.int64)
+        vals = rng.random(n)
+        vals[rng.random(n) < 0.1] = np.nan
+        counts = oracle.split_row_counts(n, world, 32)
+        offs = np.cumsum([0] + counts)
+        sl = slice(offs[rank], offs[rank + 1])
+        k_loc, v_loc = keys[sl], vals[sl]
+
+        # (c) key-range all-reduce
+        lo, hi = dist_mod.allreduce_minmax(
+            int(k_loc.min()) if k_loc.size else None,
+            int(k_loc.max()) if k_loc.size else None,
+        )
+        assert lo == int(keys.min()) and hi == int(keys.max())
+
+        # (a) dense-table all-reduce == global oracle
+        n_slots = hi - lo + 1
+        shifted = k_loc - lo
+        valid = ~np.isnan(v_loc)
+        sums = np.bincount(shifted[valid], weights=v_loc[valid], minlength=n_slots)
+        rowcnt = np.bincount(shifted, minlength=n_slots)
+        cnts = np.bincount(shifted[valid], minlength=n_slots)
+
+        class FakeTable:
+            _torch_tensors = (
+                torch.tensor(sums, dtype=torch.float64),
+                torch.tensor(rowcnt, dtype=torch.int64),
+                torch.tensor(cnts, dtype=torch.int64),
+            )
+
+        dist_mod.maybe_allreduce_table(FakeTable)
+        g_sums, g_rowcnt, g_cnts = (t.numpy() for t in FakeTable._torch_tensors)
+        ok, osum = oracle.groupby_agg(keys, {"v": vals}, "sum")
+        _, ocnt = oracle.groupby_agg(keys, {"v": vals}, "count")
+        present = g_rowcnt > 0
+        np.testing.assert_array_equal(np.nonzero(present)[0] + lo, ok)
+        np.testing.assert_allclose(g_sums[present], osum["v"], rtol=1e-12)
+        np.testing.assert_array_equal(g_cnts[present], ocnt["v"])
+
+        # (b) TreeReduce partial combine
+        loc = {
+            "v": {
+                "sum": oracle.reduce_op("sum", v_loc),
+                "count": oracle.reduce_op("count", v_loc),
+                "mn": oracle.reduce_op("min", v_loc),
+                "mx": oracle.reduce_op("max", v_loc),
+                "isum": 0, "imn": 0, "imx": 0,
+            },
+            "k": {
+                "sum": float(k_loc.sum()),
+                "count": int(k_loc.size),
+                "mn": float(k_loc.min()), "mx": float(k_loc.max()),
+                "isum": int(k_loc.sum()), "imn": int(k_loc.min()),
+                "imx": int(k_loc.max()),
+            },
+        }
+        out = dist_mod.allreduce_partials(loc, ["v", "k"])
+        np.testing.assert_allclose(out["v"]["sum"],
+                                   oracle.reduce_op("sum", vals), rtol=1e-12)
+        assert out["v"]["count"] == oracle.reduce_op("count", vals)
+        np.testing.assert_allclose(out["v"]["mn"],
+                                   oracle.reduce_op("min", vals), rtol=0)
+        np.testing.assert_allclose(out["v"]["mx"],
+                                   oracle.reduce_op("max", vals), rtol=0)
+        assert out["k"]["isum"] == int(keys.sum())
+        assert out["k"]["imn"] == int(keys.min())
+        assert out["k"]["imx"] == int(keys.max())
+        dist_mod.shutdown()
+    except Exception as e:  # pragma: no cover
+        import traceback
+        fail_q.put(f"rank {rank}: {e}\n{traceback.format_exc()}")
+
+
+@pytest.mark.timeout(120)
+def test_gloo_world2_exchange_logic():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29511
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=110)
+    errs = []
+    while not q.empty():
+        errs.append(q.get())
+    assert not errs, "\n".join(errs)
+    assert all(p.exitcode == 0 for p in procs)

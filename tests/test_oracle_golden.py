@@ -1,0 +1,97 @@
+"""CPU tier: pin the oracle against the reference-generated golden vectors.
+
+The vectors were produced by oracle/make_golden.py running the REAL reference
+(Modin PandasOnPython, NPartitions=3) in the build container and are the
+parity anchor on the GPU box where /root/reference does not exist.
+"""
+
+import numpy as np
+import pytest
+
+import oracle
+from tests.conftest import golden_cases, load_golden
+
+GB_CASES = golden_cases("gb_")
+RED_CASES = golden_cases("red_")
+
+
+def _in_cols(g):
+    return {k[3:]: v for k, v in g.items() if k.startswith("in_") and k != "in_k"}
+
+
+@pytest.mark.parametrize("case", GB_CASES)
+@pytest.mark.parametrize("agg", ["sum", "count", "mean"])
+def test_groupby_vs_golden(case, agg):
+    g = load_golden(case)
+    keys, out = oracle.groupby_agg(g["in_k"], _in_cols(g), agg)
+    np.testing.assert_array_equal(keys, g[f"out_{agg}_keys"])
+    for name in _in_cols(g):
+        expect = g[f"out_{agg}_{name}"]
+        if agg == "count":
+            np.testing.assert_array_equal(out[name], expect.astype(np.int64))
+        else:
+            np.testing.assert_allclose(out[name], expect, rtol=1e-12, equal_nan=True)
+
+
+@pytest.mark.parametrize("case", GB_CASES)
+def test_partitioned_groupby_matches_global(case):
+    """The reference's two-phase (map per partition / reduce) path must agree
+    with the one-shot oracle up to fp reassociation."""
+    g = load_golden(case)
+    cols = _in_cols(g)
+    k1, o1 = oracle.groupby_agg(g["in_k"], cols, "sum")
+    k2, o2 = oracle.partitioned_groupby_agg(g["in_k"], cols, "sum", num_splits=4)
+    np.testing.assert_array_equal(k1, k2)
+    for name in cols:
+        np.testing.assert_allclose(o1[name], o2[name], rtol=1e-12)
+
+
+@pytest.mark.parametrize("case", RED_CASES)
+@pytest.mark.parametrize("agg", ["sum", "count", "mean", "min", "max"])
+def test_reduce_vs_golden(case, agg):
+    g = load_golden(case)
+    names = [k[3:] for k in g if k.startswith("in_")]
+    got = np.array([float(oracle.reduce_op(agg, g[f"in_{n}"])) for n in names])
+    np.testing.assert_allclose(got, g[f"out_{agg}"], rtol=1e-12, equal_nan=True)
+
+
+def test_map_binary_vs_golden():
+    g = load_golden("map_binary")
+    v, w, i = g["in_v"], g["in_w"], g["in_i"]
+    checks = {
+        "add1": lambda x: oracle.map_op("add", x, 1),
+        "mul2": lambda x: oracle.map_op("mul", x, 2.5),
+        "sub3": lambda x: oracle.map_op("sub", x, 3.25),
+        "div2": lambda x: oracle.map_op("div", x, 2.0),
+        "rsub": lambda x: oracle.map_op("rsub", x, 1.0),
+        "fill0": lambda x: oracle.map_op("fillna", x, 0.0),
+        "fillm1": lambda x: oracle.map_op("fillna", x, -1.5),
+        "abs": lambda x: oracle.map_op("abs", x),
+        "frame_add": lambda x: oracle.binary_op("add", x, x),
+        "frame_mul": lambda x: oracle.binary_op("mul", x, x),
+        "frame_div": lambda x: oracle.binary_op("div", x, oracle.map_op("add", x, 10.0)),
+    }
+    for tag, fn in checks.items():
+        np.testing.assert_array_equal(fn(v), g[f"out_{tag}_v"], err_msg=tag)
+        np.testing.assert_array_equal(fn(w), g[f"out_{tag}_w"], err_msg=tag)
+    np.testing.assert_array_equal(oracle.map_op("add", i, 7), g["out_iadd_i"])
+    np.testing.assert_array_equal(oracle.map_op("mul", i, -3), g["out_imul_i"])
+    np.testing.assert_array_equal(oracle.map_op("abs", i), g["out_iabs_i"])
+
+
+def test_split_row_counts_matches_reference_rule():
+    """compute_chunksize (storage_formats/pandas/utils.py:28): ceil division,
+    floored at MinRowPartitionSize."""
+    # chunk = max(ceil(n/num_splits), min_size)
+    assert oracle.split_row_counts(100, 4, 32) == [32, 32, 32, 4]
+    assert oracle.split_row_counts(100, 3, 32) == [34, 34, 32]
+    assert oracle.split_row_counts(100, 3, 1) == [34, 34, 32]
+    assert oracle.split_row_counts(10, 4, 32) == [10]
+    assert oracle.split_row_counts(0, 4, 32) == [0]
+    assert sum(oracle.split_row_counts(10**6, 8, 32)) == 10**6
+    assert len(oracle.split_row_counts(10**6, 8, 32)) == 8
+
+
+def test_groupby_empty():
+    keys, out = oracle.groupby_agg(np.empty(0, np.int64), {"v": np.empty(0)}, "sum")
+    assert keys.size == 0 and out["v"].size == 0
