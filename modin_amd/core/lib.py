@@ -173,14 +173,21 @@ def device_count() -> int:
 
 
 class ColumnRef:
-    """Owner of one hf_col device column (frees it on GC)."""
+    """Owner of one hf_col device column (frees it on GC).
 
-    __slots__ = ("handle", "length", "dtype_code")
+    Device columns are immutable (every op writes a new column), so per-column
+    reduce partials are cached here — the lazy-metadata pattern of the
+    reference (modin/core/dataframe/pandas/metadata/, ModinDtypes/index
+    caches): a groupby's key-range scan runs once per column, not per query.
+    """
+
+    __slots__ = ("handle", "length", "dtype_code", "_reduce_cache")
 
     def __init__(self, handle: ct.c_void_p, length: int, dtype_code: int):
         self.handle = handle
         self.length = length
         self.dtype_code = dtype_code
+        self._reduce_cache = None
 
     @property
     def np_dtype(self):
@@ -269,9 +276,12 @@ def binary(op: int, a: ColumnRef, b: ColumnRef) -> ColumnRef:
 
 
 def reduce(col: ColumnRef) -> HfReduceResult:
+    if col._reduce_cache is not None:
+        return col._reduce_cache
     ensure_ready()
     res = HfReduceResult()
     _check(load().hf_reduce(col.handle, ct.byref(res)), "hf_reduce")
+    col._reduce_cache = res
     return res
 
 
