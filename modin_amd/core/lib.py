@@ -246,8 +246,8 @@ def map_scalar(op: int, col: ColumnRef, scalar) -> ColumnRef:
     ensure_ready()
     out = ct.c_void_p()
     if col.dtype_code == HF_INT64 and op != MAP_CAST_F64:
-        _check(load().hf_map_scalar_i64(op, col.handle, int(scalar), ct.byref(out)),
-               "hf_map_scalar_i64")
+        _check(load().hf_map_scalar_i64(op, col.handle, int(scalar or 0),
+                                        ct.byref(out)), "hf_map_scalar_i64")
         return _wrap(out, col.length, HF_INT64)
     _check(load().hf_map_scalar(op, col.handle, float(scalar or 0.0), ct.byref(out)),
            "hf_map_scalar")

@@ -59,7 +59,7 @@ def test_groupby_vs_golden(case, agg, npartitions):
                                           expect.astype(np.int64))
         else:
             np.testing.assert_allclose(out[name].to_numpy(), expect, rtol=RTOL,
-                                       equal_nan=True)
+                                       atol=1e-9, equal_nan=True)
 
 
 @pytest.mark.parametrize("case", golden_cases("red_"))
