@@ -150,19 +150,28 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    n_launch, total_ms = lib.kernel_stats("gb_accum")
-    rows_per_launch = (local_n / max(args.parts, 1)) if local_n else 0
+    # dominant kernel of the groupby path this run (radix: gb_scatter;
+    # dense: gb_dense; fallback: gb_accum)
+    stats = {name: lib.kernel_stats(name)
+             for name in ("gb_scatter", "gb_bucket_agg", "gb_dense",
+                          "gb_accum", "gb_hist", "gb_compact_scatter")}
+    dom = max(stats, key=lambda k: stats[k][1])
+    n_launch, total_ms = stats[dom]
+    rows_per_launch = local_n * args.steps / n_launch if n_launch else 0
     avg_ms = total_ms / n_launch if n_launch else float("nan")
     achieved = (ALG_BYTES_PER_ROW * rows_per_launch) / (avg_ms / 1e3) \
         if n_launch and avg_ms > 0 else 0.0
     traffic_env = os.environ.get("HF_TRAFFIC_BYTES_PER_LAUNCH")
     roofline = {
         "bound": "hbm",
-        "achieved": achieved / 1e9,          # GB/s
+        "kernel": dom,
+        "achieved": achieved / 1e9,          # GB/s (algorithmic 16 B/row)
         "peak": HBM_PEAK / 1e9,
         "unit": "GB/s",
         "frac": achieved / HBM_PEAK,
         "traffic": float(traffic_env) if traffic_env else None,
+        "kernel_ms": {k: [v[0], round(v[1], 3)] for k, v in stats.items()
+                      if v[0]},
     }
 
     value = args.rows * args.steps / elapsed if elapsed > 0 else 0.0

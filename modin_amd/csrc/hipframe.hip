@@ -17,6 +17,7 @@
 // Build: hipcc --offload-arch=gfx950 -O3 -shared -fPIC (see Makefile).
 
 #include <hip/hip_runtime.h>
+#include <algorithm>
 #include <cstdint>
 #include <cstdio>
 #include <cstring>
@@ -36,6 +37,15 @@ struct hf_col {
   int64_t len;
   int     dtype;
   int     gpu;
+  // cached per-bucket histogram for the radix groupby path (device u64
+  // array of n_buckets entries).  Valid when hist_kmin/hist_nb match the
+  // current query.  The column is immutable, so this is the device analog
+  // of the reference's lazy-metadata caches (modin/core/dataframe/pandas/
+  // metadata/) — a repeated groupby on the same key column skips the
+  // histogram pass.
+  void*   d_hist = nullptr;
+  int64_t hist_kmin = 0;
+  int64_t hist_nb = 0;
 };
 
 namespace {
@@ -481,6 +491,204 @@ __global__ void __launch_bounds__(BLOCK) k_gb_accum(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Radix groupby path (the north-star kernel set).
+//
+// Global f64/u64 atomics saturate at ~23-27 G ops/s on gfx950 (measured,
+// tools/atomic_probe.hip; privatizing tables per XCD changes nothing), while
+// LDS ds_add_f64 aggregation runs within 14% of the streaming ceiling.  So
+// for key ranges beyond LDS capacity the kernel radix-partitions rows into
+// buckets of GB_RANGE=8192 keys (u16 local key + f64 value, packed SoA),
+// then aggregates each bucket in an LDS-resident dense table:
+//   P0 k_gb_hist       : per-bucket row counts (8 B/row read; cached per key
+//                        column — immutable columns make this the device
+//                        analog of the reference's lazy metadata caches)
+//   P1 k_gb_scatter    : read 16 B/row, write 8 B val + 2 B lowkey into
+//                        exact per-bucket regions (LDS ranks, one global
+//                        cursor atomic per block-tile per bucket)
+//   P2 k_gb_bucket_agg : stream each bucket chunk into an LDS table
+//                        (ds_add_f64), merge non-empty slots into the global
+//                        dense table (atomics only per slot, not per row)
+// n_slots <= GB_RANGE skips P0/P1 entirely (k_gb_dense: one 16 B/row pass).
+// ---------------------------------------------------------------------------
+
+constexpr int GB_RANGE_LOG = 13;               // 8192 keys per bucket
+constexpr int GB_RANGE = 1 << GB_RANGE_LOG;
+constexpr int64_t GB_MAX_BUCKETS = 1024;       // radix path: n_slots <= 8.4M
+constexpr int SCAT_RPT = 16;                   // rows per thread per tile
+constexpr int64_t SCAT_TILE = (int64_t)BLOCK * SCAT_RPT;  // 4096 rows
+constexpr int64_t AGG_CHUNK = 1 << 21;         // rows per P2 work item
+
+struct GbWorkItem {
+  int64_t start;   // absolute row in the scatter regions
+  int32_t bucket;
+  int32_t len;
+};
+
+__global__ void __launch_bounds__(BLOCK) k_gb_hist(
+    const int64_t* __restrict__ keys, int64_t n, int64_t key_min,
+    int64_t n_slots, int nb, unsigned long long* __restrict__ hist) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  unsigned* lhist = reinterpret_cast<unsigned*>(smem_raw);
+  for (int t = threadIdx.x; t < nb; t += blockDim.x) lhist[t] = 0;
+  __syncthreads();
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const int64_t k = keys[i] - key_min;
+    if ((uint64_t)k < (uint64_t)n_slots)
+      atomicAdd(&lhist[k >> GB_RANGE_LOG], 1u);
+  }
+  __syncthreads();
+  for (int t = threadIdx.x; t < nb; t += blockDim.x)
+    if (lhist[t]) atomicAdd(&hist[t], (unsigned long long)lhist[t]);
+}
+
+template <int NV>
+__global__ void __launch_bounds__(BLOCK) k_gb_scatter(
+    const int64_t* __restrict__ keys, const double* __restrict__ v0,
+    const double* __restrict__ v1, int64_t n, int64_t key_min, int64_t n_slots,
+    int nb, unsigned* __restrict__ cursors,
+    double* __restrict__ r0, double* __restrict__ r1,
+    unsigned short* __restrict__ rk, unsigned long long* __restrict__ err) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  unsigned* it_cnt = reinterpret_cast<unsigned*>(smem_raw);          // [nb]
+  unsigned* it_base = it_cnt + nb;                                   // [nb]
+  const int64_t ntiles = (n + SCAT_TILE - 1) / SCAT_TILE;
+  for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+    const int64_t t0 = tile * SCAT_TILE;
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) it_cnt[t] = 0;
+    // stage rows in registers
+    int lb[SCAT_RPT];
+    unsigned short lk[SCAT_RPT];
+    unsigned lr[SCAT_RPT];
+    double lv0[SCAT_RPT], lv1[SCAT_RPT];
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < SCAT_RPT; ++j) {
+      const int64_t row = t0 + (int64_t)j * blockDim.x + threadIdx.x;
+      lb[j] = -1;
+      if (row < n) {
+        const int64_t k = keys[row] - key_min;
+        if ((uint64_t)k < (uint64_t)n_slots) {
+          lb[j] = (int)(k >> GB_RANGE_LOG);
+          lk[j] = (unsigned short)(k & (GB_RANGE - 1));
+          lv0[j] = NV > 0 ? v0[row] : 0.0;
+          if (NV > 1) lv1[j] = v1[row];
+        } else {
+          atomicAdd(err, 1ULL);
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < SCAT_RPT; ++j)
+      if (lb[j] >= 0) lr[j] = atomicAdd(&it_cnt[lb[j]], 1u);
+    __syncthreads();
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) {
+      const unsigned c = it_cnt[t];
+      if (c) it_base[t] = atomicAdd(&cursors[t], c);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < SCAT_RPT; ++j) {
+      if (lb[j] >= 0) {
+        const int64_t pos = (int64_t)it_base[lb[j]] + lr[j];
+        rk[pos] = lk[j];
+        if (NV > 0) r0[pos] = lv0[j];
+        if (NV > 1) r1[pos] = lv1[j];
+      }
+    }
+    __syncthreads();  // it_cnt reused next tile
+  }
+}
+
+// P2: one work item per (bucket, chunk); one value column per launch.
+// ROWCNT: also merge per-slot row counts (first column / keys-only pass).
+// CNT: also merge per-slot non-NaN counts for this column.
+template <bool ROWCNT, bool CNT, bool HAVE_VAL>
+__global__ void __launch_bounds__(512) k_gb_bucket_agg(
+    const double* __restrict__ vals, const unsigned short* __restrict__ lowkeys,
+    const GbWorkItem* __restrict__ work, int64_t n_slots,
+    double* __restrict__ gsums, unsigned long long* __restrict__ growcnt,
+    unsigned long long* __restrict__ gcounts) {
+  __shared__ double lsums[GB_RANGE];            // 64 KB
+  __shared__ unsigned ltouch[GB_RANGE];         // 32 KB (rowcnt / touch marks)
+  __shared__ unsigned lcnt[CNT ? GB_RANGE : 1]; // 32 KB when counting
+  const GbWorkItem w = work[blockIdx.x];
+  for (int s = threadIdx.x; s < GB_RANGE; s += blockDim.x) {
+    if (HAVE_VAL) lsums[s] = 0.0;
+    ltouch[s] = 0;
+    if (CNT) lcnt[s] = 0;
+  }
+  __syncthreads();
+  const int64_t end = w.start + w.len;
+  for (int64_t i = w.start + threadIdx.x; i < end; i += blockDim.x) {
+    const int slot = lowkeys[i];
+    atomicAdd(&ltouch[slot], 1u);
+    if (HAVE_VAL) {
+      const double v = vals[i];
+      if (v == v) {
+        unsafeAtomicAdd(&lsums[slot], v);
+        if (CNT) atomicAdd(&lcnt[slot], 1u);
+      }
+    }
+  }
+  __syncthreads();
+  const int64_t gbase = (int64_t)w.bucket << GB_RANGE_LOG;
+  for (int s = threadIdx.x; s < GB_RANGE; s += blockDim.x) {
+    const unsigned t = ltouch[s];
+    if (!t || gbase + s >= n_slots) continue;
+    if (HAVE_VAL) unsafeAtomicAdd(&gsums[gbase + s], lsums[s]);
+    if (ROWCNT) atomicAdd(&growcnt[gbase + s], (unsigned long long)t);
+    if (CNT) atomicAdd(&gcounts[gbase + s], (unsigned long long)lcnt[s]);
+  }
+}
+
+// Dense direct path: n_slots <= GB_RANGE, one 16 B/row pass per column.
+template <bool ROWCNT, bool CNT, bool HAVE_VAL>
+__global__ void __launch_bounds__(512) k_gb_dense(
+    const int64_t* __restrict__ keys, const double* __restrict__ vals,
+    int64_t n, int64_t key_min, int64_t n_slots,
+    double* __restrict__ gsums, unsigned long long* __restrict__ growcnt,
+    unsigned long long* __restrict__ gcounts,
+    unsigned long long* __restrict__ err) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  double* lsums = reinterpret_cast<double*>(smem_raw);       // [n_slots]
+  unsigned* ltouch = reinterpret_cast<unsigned*>(lsums + (HAVE_VAL ? n_slots : 0));
+  unsigned* lcnt = ltouch + n_slots;                         // when CNT
+  for (int64_t s = threadIdx.x; s < n_slots; s += blockDim.x) {
+    if (HAVE_VAL) lsums[s] = 0.0;
+    ltouch[s] = 0;
+    if (CNT) lcnt[s] = 0;
+  }
+  __syncthreads();
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const int64_t k = keys[i] - key_min;
+    if ((uint64_t)k >= (uint64_t)n_slots) {
+      atomicAdd(err, 1ULL);
+      continue;
+    }
+    atomicAdd(&ltouch[k], 1u);
+    if (HAVE_VAL) {
+      const double v = vals[i];
+      if (v == v) {
+        unsafeAtomicAdd(&lsums[k], v);
+        if (CNT) atomicAdd(&lcnt[k], 1u);
+      }
+    }
+  }
+  __syncthreads();
+  for (int64_t s = threadIdx.x; s < n_slots; s += blockDim.x) {
+    const unsigned t = ltouch[s];
+    if (!t) continue;
+    if (HAVE_VAL) unsafeAtomicAdd(&gsums[s], lsums[s]);
+    if (ROWCNT) atomicAdd(&growcnt[s], (unsigned long long)t);
+    if (CNT) atomicAdd(&gcounts[s], (unsigned long long)lcnt[s]);
+  }
+}
+
 // ---- compaction: present slots (rowcnt>0) -> ascending keys + columns ----
 // Fixed tile partitioning so prefix order == slot order.
 constexpr int COMPACT_TILE = 4096;  // slots per tile, one 256-thread block/tile
@@ -692,6 +900,7 @@ int hf_get(const hf_col* col, void* host) {
 int hf_col_free(hf_col* col) {
   if (!col) return HF_OK;
   if (g.inited && col->dptr) hipFreeAsync(col->dptr, g.stream);
+  if (col->d_hist) free(col->d_hist);  // host-side cached histogram
   delete col;
   return HF_OK;
 }
@@ -878,6 +1087,184 @@ int hf_reduce(const hf_col* in, hf_reduce_result* out) {
 
 // ---- groupby ----
 
+namespace {
+
+// Per-bucket histogram of an (immutable) key column, cached host-side on the
+// hf_col — repeated groupbys on the same keys skip the P0 pass entirely.
+int ensure_host_hist(hf_col* keys, int64_t key_min, int64_t n_slots,
+                     int64_t nb) {
+  if (keys->d_hist && keys->hist_kmin == key_min && keys->hist_nb == nb)
+    return HF_OK;
+  if (keys->d_hist) { free(keys->d_hist); keys->d_hist = nullptr; }
+  unsigned long long* d_h = nullptr;
+  HF_HIP("gb_hist", hipMallocAsync((void**)&d_h, nb * 8, g.stream));
+  HF_HIP("gb_hist", hipMemsetAsync(d_h, 0, nb * 8, g.stream));
+  const int64_t n = keys->len;
+  int rc = timed_launch("gb_hist", [&] {
+    hipLaunchKernelGGL(k_gb_hist, dim3((uint32_t)grid_for(n)), dim3(BLOCK),
+                       (uint32_t)(nb * 4), g.stream, (const int64_t*)keys->dptr,
+                       n, key_min, n_slots, (int)nb, d_h);
+  });
+  if (rc != HF_OK) return rc;
+  int64_t* h = (int64_t*)malloc(nb * 8);
+  HF_HIP("gb_hist", hipMemcpyAsync(h, d_h, nb * 8, hipMemcpyDeviceToHost,
+                                   g.stream));
+  HF_HIP("gb_hist", hipStreamSynchronize(g.stream));
+  HF_HIP("gb_hist", hipFreeAsync(d_h, g.stream));
+  keys->d_hist = h;
+  keys->hist_kmin = key_min;
+  keys->hist_nb = nb;
+  return HF_OK;
+}
+
+int gb_dense_path(const hf_col* keys, const GbPtrs& ptrs, int nvals,
+                  int64_t key_min, int64_t n_slots, uintptr_t sums,
+                  uintptr_t rowcnt, uintptr_t counts,
+                  unsigned long long* d_err) {
+  const int64_t n = keys->len;
+  const bool cnt = counts != 0;
+  int64_t grid = n / 65536 + 1;
+  if (grid > 512) grid = 512;
+  auto L = [&](auto rTag, auto cTag, auto vTag, const double* v, double* gs,
+               unsigned long long* gc) {
+    constexpr bool R = decltype(rTag)::value, C = decltype(cTag)::value,
+                   V = decltype(vTag)::value;
+    const uint32_t lds =
+        (uint32_t)(n_slots * ((V ? 8 : 0) + 4 + (C ? 4 : 0)));
+    return timed_launch("gb_dense", [&] {
+      hipLaunchKernelGGL((k_gb_dense<R, C, V>), dim3((uint32_t)grid), dim3(512),
+                         lds, g.stream, (const int64_t*)keys->dptr, v, n,
+                         key_min, n_slots, gs, (unsigned long long*)rowcnt, gc,
+                         d_err);
+    });
+  };
+  using T = std::true_type;
+  using F = std::false_type;
+  if (nvals == 0) return L(T{}, F{}, F{}, nullptr, nullptr, nullptr);
+  for (int c = 0; c < nvals; ++c) {
+    double* gs = (double*)sums + (int64_t)c * n_slots;
+    unsigned long long* gc =
+        cnt ? (unsigned long long*)counts + (int64_t)c * n_slots : nullptr;
+    int rc;
+    if (c == 0)
+      rc = cnt ? L(T{}, T{}, T{}, ptrs.vals[c], gs, gc)
+               : L(T{}, F{}, T{}, ptrs.vals[c], gs, gc);
+    else
+      rc = cnt ? L(F{}, T{}, T{}, ptrs.vals[c], gs, gc)
+               : L(F{}, F{}, T{}, ptrs.vals[c], gs, gc);
+    if (rc != HF_OK) return rc;
+  }
+  return HF_OK;
+}
+
+int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
+                  int64_t n_slots, uintptr_t sums, uintptr_t rowcnt,
+                  uintptr_t counts, unsigned long long* d_err) {
+  const int64_t n = keys->len;
+  const int64_t nb = (n_slots + GB_RANGE - 1) >> GB_RANGE_LOG;
+  int rc = ensure_host_hist(keys, key_min, n_slots, nb);
+  if (rc != HF_OK) return rc;
+  const int64_t* h = keys->d_hist ? (const int64_t*)keys->d_hist : nullptr;
+  // exact per-bucket regions, 64-row aligned; u32 cursors cap one partition
+  // at ~4.29e9 rows (shard larger frames)
+  std::vector<unsigned> cur_init((size_t)nb);
+  std::vector<GbWorkItem> work;
+  int64_t off = 0;
+  for (int64_t b = 0; b < nb; ++b) {
+    cur_init[b] = (unsigned)off;
+    const int64_t rows_b = h[b];
+    int64_t done = 0;
+    while (done < rows_b) {
+      const int64_t len = std::min(AGG_CHUNK, rows_b - done);
+      work.push_back(GbWorkItem{off + done, (int32_t)b, (int32_t)len});
+      done += len;
+    }
+    off += (rows_b + 63) & ~63LL;
+  }
+  if (off > 0xFFFFFFF0LL)
+    return set_err(HF_ERR_UNSUPPORTED, "hf_groupby_accum",
+                   "partition too large for radix scatter (shard it)");
+  double* r0 = nullptr;
+  double* r1 = nullptr;
+  unsigned short* rk = nullptr;
+  unsigned* d_cur = nullptr;
+  GbWorkItem* d_work = nullptr;
+  const int64_t alloc_rows = off > 0 ? off : 64;
+  if (nvals > 0)
+    HF_HIP("gb_radix", hipMallocAsync((void**)&r0, alloc_rows * 8, g.stream));
+  if (nvals > 1)
+    HF_HIP("gb_radix", hipMallocAsync((void**)&r1, alloc_rows * 8, g.stream));
+  HF_HIP("gb_radix", hipMallocAsync((void**)&rk, alloc_rows * 2, g.stream));
+  HF_HIP("gb_radix", hipMallocAsync((void**)&d_cur, nb * 4, g.stream));
+  HF_HIP("gb_radix",
+         hipMallocAsync((void**)&d_work, work.size() * sizeof(GbWorkItem),
+                        g.stream));
+  HF_HIP("gb_radix", hipMemcpyAsync(d_cur, cur_init.data(), nb * 4,
+                                    hipMemcpyHostToDevice, g.stream));
+  HF_HIP("gb_radix",
+         hipMemcpyAsync(d_work, work.data(), work.size() * sizeof(GbWorkItem),
+                        hipMemcpyHostToDevice, g.stream));
+  // host vectors must outlive the async H2D of pageable memory
+  HF_HIP("gb_radix", hipStreamSynchronize(g.stream));
+  // P1 scatter
+  const int64_t ntiles = (n + SCAT_TILE - 1) / SCAT_TILE;
+  const uint32_t sgrid = (uint32_t)std::min<int64_t>(ntiles, 1024);
+  auto scat = [&](auto nvTag) {
+    constexpr int NVv = decltype(nvTag)::value;
+    return timed_launch("gb_scatter", [&] {
+      hipLaunchKernelGGL((k_gb_scatter<NVv>), dim3(sgrid), dim3(BLOCK),
+                         (uint32_t)(nb * 8), g.stream,
+                         (const int64_t*)keys->dptr, ptrs.vals[0], ptrs.vals[1],
+                         n, key_min, n_slots, (int)nb, d_cur, r0, r1, rk, d_err);
+    });
+  };
+  rc = nvals == 0   ? scat(std::integral_constant<int, 0>{})
+       : nvals == 1 ? scat(std::integral_constant<int, 1>{})
+                    : scat(std::integral_constant<int, 2>{});
+  if (rc != HF_OK) return rc;
+  // P2 aggregate, one column per launch
+  const bool cnt = counts != 0;
+  const uint32_t agrid = (uint32_t)work.size();
+  auto agg = [&](auto rTag, auto cTag, auto vTag, const double* v, double* gs,
+                 unsigned long long* gc) {
+    constexpr bool R = decltype(rTag)::value, C = decltype(cTag)::value,
+                   V = decltype(vTag)::value;
+    return timed_launch("gb_bucket_agg", [&] {
+      hipLaunchKernelGGL((k_gb_bucket_agg<R, C, V>), dim3(agrid), dim3(512), 0,
+                         g.stream, v, rk, d_work, n_slots, gs,
+                         (unsigned long long*)rowcnt, gc);
+    });
+  };
+  using T = std::true_type;
+  using F = std::false_type;
+  if (agrid > 0) {
+    if (nvals == 0) {
+      rc = agg(T{}, F{}, F{}, nullptr, nullptr, nullptr);
+    } else {
+      for (int c = 0; c < nvals && rc == HF_OK; ++c) {
+        const double* v = c == 0 ? r0 : r1;
+        double* gs = (double*)sums + (int64_t)c * n_slots;
+        unsigned long long* gc =
+            cnt ? (unsigned long long*)counts + (int64_t)c * n_slots : nullptr;
+        if (c == 0)
+          rc = cnt ? agg(T{}, T{}, T{}, v, gs, gc)
+                   : agg(T{}, F{}, T{}, v, gs, gc);
+        else
+          rc = cnt ? agg(F{}, T{}, T{}, v, gs, gc)
+                   : agg(F{}, F{}, T{}, v, gs, gc);
+      }
+    }
+  }
+  if (r0) hipFreeAsync(r0, g.stream);
+  if (r1) hipFreeAsync(r1, g.stream);
+  hipFreeAsync(rk, g.stream);
+  hipFreeAsync(d_cur, g.stream);
+  hipFreeAsync(d_work, g.stream);
+  return rc;
+}
+
+}  // namespace
+
 int hf_groupby_accum(const hf_col* keys, const hf_col* const* vals, int nvals,
                      int64_t key_min, int64_t n_slots,
                      uintptr_t sums, uintptr_t rowcnt, uintptr_t counts) {
@@ -897,6 +1284,16 @@ int hf_groupby_accum(const hf_col* keys, const hf_col* const* vals, int nvals,
   const int64_t n = keys->len;
   unsigned long long* d_err =
       (unsigned long long*)((char*)g.d_scratch + SCRATCH_GB_ERR);
+  // path selection (DESIGN.md §GroupBy kernels): LDS-dense for small ranges,
+  // radix partition for the north-star range, global atomics as the wide
+  // fallback (slow but correct for any range below the slot cap)
+  if (n > 0 && n_slots <= GB_RANGE)
+    return gb_dense_path(keys, ptrs, nvals, key_min, n_slots, sums, rowcnt,
+                         counts, d_err);
+  const int64_t nb = (n_slots + GB_RANGE - 1) >> GB_RANGE_LOG;
+  if (n > 0 && nb <= GB_MAX_BUCKETS && nvals <= 2)
+    return gb_radix_path(const_cast<hf_col*>(keys), ptrs, nvals, key_min,
+                         n_slots, sums, rowcnt, counts, d_err);
   auto launch = [&](auto nvTag, auto cntTag) {
     constexpr int NV = decltype(nvTag)::value;
     constexpr bool CNT = decltype(cntTag)::value;
