@@ -1,0 +1,338 @@
+// radix_probe — variant sweep for the radix-groupby scatter/aggregate
+// kernels (north-star shape: 1e9 rows, 1e6 keys).  Winners get folded into
+// modin_amd/csrc/hipframe.hip.
+// Build: hipcc --offload-arch=gfx950 -O3 tools/radix_probe.hip -o tools/radix_probe
+#include <hip/hip_runtime.h>
+#include <cstdio>
+#include <cstdint>
+#include <cstring>
+#include <functional>
+#include <vector>
+#include <random>
+#include <algorithm>
+
+#define CHECK(x) do { hipError_t e = (x); if (e != hipSuccess) { \
+  printf("HIP error %s at %d\n", hipGetErrorString(e), __LINE__); exit(1); } } while (0)
+
+constexpr int BLOCK = 256;
+
+struct Work { int64_t start; int32_t bucket; int32_t len; };
+
+// ---------------- scatter variants ----------------
+
+// A: register-staged (current production shape), params RPT/RL
+template <int RPT, int RL>
+__global__ void __launch_bounds__(BLOCK) k_scat_reg(
+    const int64_t* __restrict__ keys, const double* __restrict__ v0, int64_t n,
+    int64_t n_slots, int nb, unsigned* __restrict__ cursors,
+    double* __restrict__ r0, unsigned short* __restrict__ rk) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  unsigned* it_cnt = reinterpret_cast<unsigned*>(smem_raw);
+  unsigned* it_base = it_cnt + nb;
+  const int64_t TILE = (int64_t)BLOCK * RPT;
+  const int64_t ntiles = (n + TILE - 1) / TILE;
+  for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+    const int64_t t0 = tile * TILE;
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) it_cnt[t] = 0;
+    int lb[RPT];
+    unsigned short lk[RPT];
+    unsigned lr[RPT];
+    double lv[RPT];
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < RPT; ++j) {
+      const int64_t row = t0 + (int64_t)j * blockDim.x + threadIdx.x;
+      lb[j] = -1;
+      if (row < n) {
+        const int64_t k = keys[row];
+        lb[j] = (int)(k >> RL);
+        lk[j] = (unsigned short)(k & ((1 << RL) - 1));
+        lv[j] = v0[row];
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < RPT; ++j)
+      if (lb[j] >= 0) lr[j] = atomicAdd(&it_cnt[lb[j]], 1u);
+    __syncthreads();
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) {
+      const unsigned c = it_cnt[t];
+      if (c) it_base[t] = atomicAdd(&cursors[t], c);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < RPT; ++j) {
+      if (lb[j] >= 0) {
+        const int64_t pos = (int64_t)it_base[lb[j]] + lr[j];
+        rk[pos] = lk[j];
+        r0[pos] = lv[j];
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// B: LDS-staged bucket-sorted tile, coalesced writes.  TILE rows staged in
+// LDS (val 8B + packed (bucket<<RL)|lowkey u32), written out in sorted order.
+template <int RPT, int RL>
+__global__ void __launch_bounds__(BLOCK) k_scat_lds(
+    const int64_t* __restrict__ keys, const double* __restrict__ v0, int64_t n,
+    int64_t n_slots, int nb, unsigned* __restrict__ cursors,
+    double* __restrict__ r0, unsigned short* __restrict__ rk) {
+  constexpr int TILE = BLOCK * RPT;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  double* sval = reinterpret_cast<double*>(smem_raw);                 // TILE*8
+  unsigned* skey = reinterpret_cast<unsigned*>(sval + TILE);          // TILE*4
+  unsigned* it_cnt = skey + TILE;                                     // nb
+  unsigned* it_off = it_cnt + nb;                                     // nb (excl scan)
+  unsigned* it_gbase = it_off + nb;                                   // nb
+  const int64_t ntiles = (n + TILE - 1) / TILE;
+  for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+    const int64_t t0 = tile * TILE;
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) it_cnt[t] = 0;
+    int lb[RPT];
+    unsigned lk[RPT];
+    unsigned lr[RPT];
+    double lv[RPT];
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < RPT; ++j) {
+      const int64_t row = t0 + (int64_t)j * blockDim.x + threadIdx.x;
+      lb[j] = -1;
+      if (row < n) {
+        const int64_t k = keys[row];
+        lb[j] = (int)(k >> RL);
+        lk[j] = (unsigned)(k & ((1 << RL) - 1));
+        lv[j] = v0[row];
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < RPT; ++j)
+      if (lb[j] >= 0) lr[j] = atomicAdd(&it_cnt[lb[j]], 1u);
+    __syncthreads();
+    // exclusive scan of it_cnt (single lane; nb <= 1024 — cheap vs the tile)
+    if (threadIdx.x == 0) {
+      unsigned run = 0;
+      for (int t = 0; t < nb; ++t) { it_off[t] = run; run += it_cnt[t]; }
+    }
+    __syncthreads();
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) {
+      const unsigned c = it_cnt[t];
+      if (c) it_gbase[t] = atomicAdd(&cursors[t], c);
+    }
+    // stage bucket-sorted
+#pragma unroll
+    for (int j = 0; j < RPT; ++j) {
+      if (lb[j] >= 0) {
+        const unsigned p = it_off[lb[j]] + lr[j];
+        sval[p] = lv[j];
+        skey[p] = ((unsigned)lb[j] << 16) | lk[j];
+      }
+    }
+    __syncthreads();
+    // write out in sorted order: consecutive threads -> consecutive global
+    // positions within each bucket segment
+    const int valid = (int)min((int64_t)TILE, n - t0);
+    for (int p = threadIdx.x; p < valid; p += blockDim.x) {
+      const unsigned b = skey[p] >> 16;
+      const int64_t pos = (int64_t)it_gbase[b] + (p - it_off[b]);
+      r0[pos] = sval[p];
+      rk[pos] = (unsigned short)(skey[p] & 0xFFFF);
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------- aggregate variants ----------------
+
+template <int RL, int VEC>
+__global__ void __launch_bounds__(512) k_agg(
+    const double* __restrict__ vals, const unsigned short* __restrict__ lowkeys,
+    const Work* __restrict__ work, int64_t n_slots, double* __restrict__ gsums,
+    unsigned long long* __restrict__ growcnt) {
+  constexpr int RANGE = 1 << RL;
+  __shared__ double lsums[RANGE];
+  __shared__ unsigned ltouch[RANGE];
+  const Work w = work[blockIdx.x];
+  for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
+    lsums[s] = 0.0;
+    ltouch[s] = 0;
+  }
+  __syncthreads();
+  const int64_t end = w.start + w.len;
+  if (VEC == 1) {
+    for (int64_t i = w.start + threadIdx.x; i < end; i += blockDim.x) {
+      const int slot = lowkeys[i];
+      atomicAdd(&ltouch[slot], 1u);
+      const double v = vals[i];
+      if (v == v) unsafeAtomicAdd(&lsums[slot], v);
+    }
+  } else {
+    // 2 rows per lane: ushort2 + double2 (start is 64-aligned)
+    const int64_t npair = (int64_t)(w.len) >> 1;
+    const ushort2* k2 = reinterpret_cast<const ushort2*>(lowkeys + w.start);
+    const double2* v2 = reinterpret_cast<const double2*>(vals + w.start);
+    for (int64_t i = threadIdx.x; i < npair; i += blockDim.x) {
+      const ushort2 kk = k2[i];
+      const double2 vv = v2[i];
+      atomicAdd(&ltouch[kk.x], 1u);
+      atomicAdd(&ltouch[kk.y], 1u);
+      if (vv.x == vv.x) unsafeAtomicAdd(&lsums[kk.x], vv.x);
+      if (vv.y == vv.y) unsafeAtomicAdd(&lsums[kk.y], vv.y);
+    }
+    if ((w.len & 1) && threadIdx.x == 0) {
+      const int64_t i = end - 1;
+      const int slot = lowkeys[i];
+      atomicAdd(&ltouch[slot], 1u);
+      const double v = vals[i];
+      if (v == v) unsafeAtomicAdd(&lsums[slot], v);
+    }
+  }
+  __syncthreads();
+  const int64_t gbase = (int64_t)w.bucket << RL;
+  for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
+    const unsigned t = ltouch[s];
+    if (!t || gbase + s >= n_slots) continue;
+    unsafeAtomicAdd(&gsums[gbase + s], lsums[s]);
+    atomicAdd(&growcnt[gbase + s], (unsigned long long)t);
+  }
+}
+
+// ---------------- harness ----------------
+
+static double run(const char* name, int reps, const std::function<void()>& pre,
+                  const std::function<void()>& fn, double alg_bytes) {
+  pre();
+  fn();
+  CHECK(hipDeviceSynchronize());
+  hipEvent_t a, b;
+  CHECK(hipEventCreate(&a));
+  CHECK(hipEventCreate(&b));
+  float total = 0;
+  for (int r = 0; r < reps; ++r) {
+    pre();
+    CHECK(hipDeviceSynchronize());
+    CHECK(hipEventRecord(a));
+    fn();
+    CHECK(hipEventRecord(b));
+    CHECK(hipDeviceSynchronize());
+    float ms;
+    CHECK(hipEventElapsedTime(&ms, a, b));
+    total += ms;
+  }
+  double per = total / reps;
+  printf("%-34s %8.2f ms  alg16B %7.1f GB/s  %7.1f Grows/s\n", name, per,
+         alg_bytes / per / 1e6, (alg_bytes / 16.0) / per / 1e6);
+  fflush(stdout);
+  return per;
+}
+
+int main(int argc, char** argv) {
+  int64_t n = argc > 1 ? atoll(argv[1]) : 1000000000LL;
+  const int64_t n_slots = 1000000;
+  printf("n=%lld slots=%lld\n", (long long)n, (long long)n_slots);
+  int64_t* keys;
+  double* v0;
+  CHECK(hipMalloc(&keys, n * 8));
+  CHECK(hipMalloc(&v0, n * 8));
+  {
+    std::mt19937_64 gen(42);
+    std::vector<int64_t> hk(1 << 24);
+    std::vector<double> hv(1 << 24);
+    for (int64_t off = 0; off < n; off += (int64_t)hk.size()) {
+      int64_t m = std::min<int64_t>(hk.size(), n - off);
+      for (int64_t i = 0; i < m; ++i) {
+        hk[i] = gen() % n_slots;
+        hv[i] = (double)(gen() % 1000) / 1000.0;
+      }
+      CHECK(hipMemcpy(keys + off, hk.data(), m * 8, hipMemcpyHostToDevice));
+      CHECK(hipMemcpy(v0 + off, hv.data(), m * 8, hipMemcpyHostToDevice));
+    }
+  }
+  double* r0;
+  unsigned short* rk;
+  unsigned* d_cur;
+  double* gsums;
+  unsigned long long* growcnt;
+  const int64_t alloc_rows = n + 64 * 2048;
+  CHECK(hipMalloc(&r0, alloc_rows * 8));
+  CHECK(hipMalloc(&rk, alloc_rows * 2));
+  CHECK(hipMalloc(&d_cur, 2048 * 4));
+  CHECK(hipMalloc(&gsums, n_slots * 8));
+  CHECK(hipMalloc(&growcnt, n_slots * 8));
+
+  auto sweep = [&](auto rlTag) {
+    constexpr int RL = decltype(rlTag)::value;
+    const int nb = (int)((n_slots + (1 << RL) - 1) >> RL);
+    // uniform keys: exact histogram close to n/nb; compute on host for regions
+    std::vector<int64_t> h((size_t)nb, 0);
+    {
+      // approximate: uniform keys; region = n/nb * 1.1 margin, aligned
+      // (probe only; production uses the exact histogram)
+      int64_t per = (int64_t)((double)n / nb * 1.15) + 4096;
+      per = (per + 63) & ~63LL;
+      for (int b = 0; b < nb; ++b) h[b] = per;
+    }
+    std::vector<unsigned> cur((size_t)nb);
+    std::vector<Work> work;
+    int64_t off = 0;
+    for (int b = 0; b < nb; ++b) {
+      cur[b] = (unsigned)off;
+      for (int64_t done = 0; done < h[b]; done += (1 << 21))
+        work.push_back(Work{off + done, b,
+                            (int32_t)std::min<int64_t>(1 << 21, h[b] - done)});
+      off += h[b];
+    }
+    if (off > alloc_rows) { printf("alloc overflow\n"); exit(1); }
+    Work* d_work;
+    CHECK(hipMalloc(&d_work, work.size() * sizeof(Work)));
+    CHECK(hipMemcpy(d_work, work.data(), work.size() * sizeof(Work),
+                    hipMemcpyHostToDevice));
+    auto reset_cur = [&] {
+      CHECK(hipMemcpyAsync(d_cur, cur.data(), nb * 4, hipMemcpyHostToDevice, 0));
+    };
+    char nm[128];
+    auto scat_reg = [&](auto rptTag) {
+      constexpr int RPT = decltype(rptTag)::value;
+      snprintf(nm, sizeof nm, "scat_reg RPT=%d RL=%d nb=%d", RPT, RL, nb);
+      const int64_t ntiles = (n + BLOCK * RPT - 1) / (BLOCK * RPT);
+      const uint32_t grid = (uint32_t)std::min<int64_t>(ntiles, 2048);
+      run(nm, 2, reset_cur, [&] {
+        hipLaunchKernelGGL((k_scat_reg<RPT, RL>), dim3(grid), dim3(BLOCK),
+                           nb * 8, 0, keys, v0, n, n_slots, nb, d_cur, r0, rk);
+      }, 26.0 * n);
+    };
+    scat_reg(std::integral_constant<int, 8>{});
+    scat_reg(std::integral_constant<int, 16>{});
+    scat_reg(std::integral_constant<int, 32>{});
+    {
+      constexpr int RPT = 8;  // TILE=2048: LDS = 2048*12 + nb*16 ~ 28-56KB
+      snprintf(nm, sizeof nm, "scat_lds RPT=%d RL=%d nb=%d", RPT, RL, nb);
+      const int64_t ntiles = (n + BLOCK * RPT - 1) / (BLOCK * RPT);
+      const uint32_t grid = (uint32_t)std::min<int64_t>(ntiles, 2048);
+      const uint32_t lds = BLOCK * RPT * 12 + nb * 16;
+      run(nm, 2, reset_cur, [&] {
+        hipLaunchKernelGGL((k_scat_lds<RPT, RL>), dim3(grid), dim3(BLOCK), lds,
+                           0, keys, v0, n, n_slots, nb, d_cur, r0, rk);
+      }, 26.0 * n);
+    }
+    // aggregate variants (consume whatever the last scatter left; perf-only)
+    auto agg = [&](auto vecTag) {
+      constexpr int VEC = decltype(vecTag)::value;
+      snprintf(nm, sizeof nm, "agg VEC=%d RL=%d (%zu wi)", VEC, RL, work.size());
+      auto nop = [] {};
+      run(nm, 3, nop, [&] {
+        hipLaunchKernelGGL((k_agg<RL, VEC>), dim3((uint32_t)work.size()),
+                           dim3(512), 0, 0, r0, rk, d_work, n_slots, gsums,
+                           growcnt);
+      }, 10.0 * n);
+    };
+    agg(std::integral_constant<int, 1>{});
+    agg(std::integral_constant<int, 2>{});
+    CHECK(hipFree(d_work));
+  };
+  sweep(std::integral_constant<int, 13>{});
+  sweep(std::integral_constant<int, 12>{});
+  sweep(std::integral_constant<int, 11>{});
+  printf("done\n");
+  return 0;
+}
