@@ -253,7 +253,7 @@ int main(int argc, char** argv) {
   unsigned* d_cur;
   double* gsums;
   unsigned long long* growcnt;
-  const int64_t alloc_rows = n + 64 * 2048;
+  const int64_t alloc_rows = (int64_t)(n * 1.25) + 4096 * 2048;
   CHECK(hipMalloc(&r0, alloc_rows * 8));
   CHECK(hipMalloc(&rk, alloc_rows * 2));
   CHECK(hipMalloc(&d_cur, 2048 * 4));
