@@ -71,6 +71,85 @@ __global__ void __launch_bounds__(BLOCK) k_scat_reg(
   }
 }
 
+// B2: LDS-staged with wave-parallel exclusive scan of it_cnt
+template <int RPT, int RL>
+__global__ void __launch_bounds__(BLOCK) k_scat_lds2(
+    const int64_t* __restrict__ keys, const double* __restrict__ v0, int64_t n,
+    int64_t n_slots, int nb, unsigned* __restrict__ cursors,
+    double* __restrict__ r0, unsigned short* __restrict__ rk) {
+  constexpr int TILE = BLOCK * RPT;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  double* sval = reinterpret_cast<double*>(smem_raw);
+  unsigned* skey = reinterpret_cast<unsigned*>(sval + TILE);
+  unsigned* it_cnt = skey + TILE;
+  unsigned* it_off = it_cnt + nb;
+  unsigned* it_gbase = it_off + nb;
+  const int64_t ntiles = (n + TILE - 1) / TILE;
+  for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+    const int64_t t0 = tile * TILE;
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) it_cnt[t] = 0;
+    int lb[RPT];
+    unsigned lk[RPT];
+    unsigned lr[RPT];
+    double lv[RPT];
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < RPT; ++j) {
+      const int64_t row = t0 + (int64_t)j * blockDim.x + threadIdx.x;
+      lb[j] = -1;
+      if (row < n) {
+        const int64_t k = keys[row];
+        lb[j] = (int)(k >> RL);
+        lk[j] = (unsigned)(k & ((1 << RL) - 1));
+        lv[j] = v0[row];
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < RPT; ++j)
+      if (lb[j] >= 0) lr[j] = atomicAdd(&it_cnt[lb[j]], 1u);
+    __syncthreads();
+    // wave 0: parallel exclusive scan over nb counters (chunks of 64 lanes)
+    if (threadIdx.x < 64) {
+      const int lane = threadIdx.x;
+      unsigned carry = 0;
+      for (int base = 0; base < nb; base += 64) {
+        const int t = base + lane;
+        unsigned v = (t < nb) ? it_cnt[t] : 0;
+        unsigned incl = v;
+#pragma unroll
+        for (int d = 1; d < 64; d <<= 1) {
+          unsigned up = __shfl_up(incl, d);
+          if (lane >= d) incl += up;
+        }
+        if (t < nb) it_off[t] = carry + incl - v;
+        carry += __shfl(incl, 63);
+      }
+    }
+    __syncthreads();
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) {
+      const unsigned c = it_cnt[t];
+      if (c) it_gbase[t] = atomicAdd(&cursors[t], c);
+    }
+#pragma unroll
+    for (int j = 0; j < RPT; ++j) {
+      if (lb[j] >= 0) {
+        const unsigned p = it_off[lb[j]] + lr[j];
+        sval[p] = lv[j];
+        skey[p] = ((unsigned)lb[j] << 16) | lk[j];
+      }
+    }
+    __syncthreads();
+    const int valid = (int)min((int64_t)TILE, n - t0);
+    for (int p = threadIdx.x; p < valid; p += blockDim.x) {
+      const unsigned b = skey[p] >> 16;
+      const int64_t pos = (int64_t)it_gbase[b] + (p - it_off[b]);
+      r0[pos] = sval[p];
+      rk[pos] = (unsigned short)(skey[p] & 0xFFFF);
+    }
+    __syncthreads();
+  }
+}
+
 // B: LDS-staged bucket-sorted tile, coalesced writes.  TILE rows staged in
 // LDS (val 8B + packed (bucket<<RL)|lowkey u32), written out in sorted order.
 template <int RPT, int RL>
@@ -302,19 +381,26 @@ int main(int argc, char** argv) {
       }, 26.0 * n);
     };
     scat_reg(std::integral_constant<int, 8>{});
-    scat_reg(std::integral_constant<int, 16>{});
-    scat_reg(std::integral_constant<int, 32>{});
-    {
-      constexpr int RPT = 8;  // TILE=2048: LDS = 2048*12 + nb*16 ~ 28-56KB
-      snprintf(nm, sizeof nm, "scat_lds RPT=%d RL=%d nb=%d", RPT, RL, nb);
+    auto scat_lds = [&](auto rptTag, bool pscan) {
+      constexpr int RPT = decltype(rptTag)::value;
+      snprintf(nm, sizeof nm, "scat_lds%s RPT=%d RL=%d nb=%d",
+               pscan ? "2" : "", RPT, RL, nb);
       const int64_t ntiles = (n + BLOCK * RPT - 1) / (BLOCK * RPT);
       const uint32_t grid = (uint32_t)std::min<int64_t>(ntiles, 2048);
       const uint32_t lds = BLOCK * RPT * 12 + nb * 16;
       run(nm, 2, reset_cur, [&] {
-        hipLaunchKernelGGL((k_scat_lds<RPT, RL>), dim3(grid), dim3(BLOCK), lds,
-                           0, keys, v0, n, n_slots, nb, d_cur, r0, rk);
+        if (pscan)
+          hipLaunchKernelGGL((k_scat_lds2<RPT, RL>), dim3(grid), dim3(BLOCK),
+                             lds, 0, keys, v0, n, n_slots, nb, d_cur, r0, rk);
+        else
+          hipLaunchKernelGGL((k_scat_lds<RPT, RL>), dim3(grid), dim3(BLOCK),
+                             lds, 0, keys, v0, n, n_slots, nb, d_cur, r0, rk);
       }, 26.0 * n);
-    }
+    };
+    scat_lds(std::integral_constant<int, 8>{}, false);
+    scat_lds(std::integral_constant<int, 8>{}, true);
+    scat_lds(std::integral_constant<int, 16>{}, true);
+    scat_lds(std::integral_constant<int, 24>{}, true);
     // aggregate variants (consume whatever the last scatter left; perf-only)
     auto agg = [&](auto vecTag) {
       constexpr int VEC = decltype(vecTag)::value;
@@ -332,7 +418,6 @@ int main(int argc, char** argv) {
   };
   sweep(std::integral_constant<int, 13>{});
   sweep(std::integral_constant<int, 12>{});
-  sweep(std::integral_constant<int, 11>{});
   printf("done\n");
   return 0;
 }
