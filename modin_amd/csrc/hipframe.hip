@@ -512,11 +512,13 @@ __global__ void __launch_bounds__(BLOCK) k_gb_accum(
 // n_slots <= GB_RANGE skips P0/P1 entirely (k_gb_dense: one 16 B/row pass).
 // ---------------------------------------------------------------------------
 
-constexpr int GB_RANGE_LOG = 13;               // 8192 keys per bucket
+constexpr int GB_RANGE_LOG = 12;               // 4096 keys per bucket: the
+                                               // P2 LDS table is 48 KB -> 3
+                                               // blocks/CU (probe: RL12 agg
+                                               // 5.0 ms vs RL13 7.2)
 constexpr int GB_RANGE = 1 << GB_RANGE_LOG;
-constexpr int64_t GB_MAX_BUCKETS = 1024;       // radix path: n_slots <= 8.4M
-constexpr int SCAT_RPT = 16;                   // rows per thread per tile
-constexpr int64_t SCAT_TILE = (int64_t)BLOCK * SCAT_RPT;  // 4096 rows
+constexpr int64_t GB_MAX_BUCKETS = 1024;       // radix path: n_slots <= 4.2M
+constexpr int64_t GB_DENSE_MAX = 8192;         // direct-LDS path bound
 constexpr int64_t AGG_CHUNK = 1 << 21;         // rows per P2 work item
 
 struct GbWorkItem {
@@ -544,36 +546,48 @@ __global__ void __launch_bounds__(BLOCK) k_gb_hist(
     if (lhist[t]) atomicAdd(&hist[t], (unsigned long long)lhist[t]);
 }
 
-template <int NV>
+// P1: LDS-staged bucket-sorted tiles.  Rows are ranked into a per-tile LDS
+// histogram, staged bucket-sorted in LDS, and written out coalesced — each
+// tile contributes one contiguous chunk per bucket stream (probe: 2.4x the
+// register-staged direct scatter).  SCAT_RPT rows/thread; NV value columns.
+template <int NV, int RPT>
 __global__ void __launch_bounds__(BLOCK) k_gb_scatter(
     const int64_t* __restrict__ keys, const double* __restrict__ v0,
     const double* __restrict__ v1, int64_t n, int64_t key_min, int64_t n_slots,
     int nb, unsigned* __restrict__ cursors,
     double* __restrict__ r0, double* __restrict__ r1,
     unsigned short* __restrict__ rk, unsigned long long* __restrict__ err) {
+  constexpr int TILE = BLOCK * RPT;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  unsigned* it_cnt = reinterpret_cast<unsigned*>(smem_raw);          // [nb]
-  unsigned* it_base = it_cnt + nb;                                   // [nb]
-  const int64_t ntiles = (n + SCAT_TILE - 1) / SCAT_TILE;
+  double* sval0 = reinterpret_cast<double*>(smem_raw);            // [TILE]
+  double* sval1 = sval0 + (NV > 1 ? TILE : 0);
+  unsigned* skey = reinterpret_cast<unsigned*>(
+      sval0 + (NV > 1 ? 2 * (int64_t)TILE : (NV > 0 ? TILE : 0)));  // [TILE]
+  unsigned* it_cnt = skey + TILE;                                 // [nb]
+  unsigned* it_off = it_cnt + nb;                                 // [nb]
+  unsigned* it_gbase = it_off + nb;                               // [nb]
+  unsigned* s_total = it_gbase + nb;  // scalar; keep ALL LDS in the dynamic
+                                      // region (a static __shared__ would
+                                      // shift the base off 16B — G17)
+  const int64_t ntiles = (n + TILE - 1) / TILE;
   for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
-    const int64_t t0 = tile * SCAT_TILE;
+    const int64_t t0 = tile * TILE;
     for (int t = threadIdx.x; t < nb; t += blockDim.x) it_cnt[t] = 0;
-    // stage rows in registers
-    int lb[SCAT_RPT];
-    unsigned short lk[SCAT_RPT];
-    unsigned lr[SCAT_RPT];
-    double lv0[SCAT_RPT], lv1[SCAT_RPT];
+    int lb[RPT];
+    unsigned lk[RPT];
+    unsigned lr[RPT];
+    double lv0[RPT], lv1[RPT];
     __syncthreads();
 #pragma unroll
-    for (int j = 0; j < SCAT_RPT; ++j) {
+    for (int j = 0; j < RPT; ++j) {
       const int64_t row = t0 + (int64_t)j * blockDim.x + threadIdx.x;
       lb[j] = -1;
       if (row < n) {
         const int64_t k = keys[row] - key_min;
         if ((uint64_t)k < (uint64_t)n_slots) {
           lb[j] = (int)(k >> GB_RANGE_LOG);
-          lk[j] = (unsigned short)(k & (GB_RANGE - 1));
-          lv0[j] = NV > 0 ? v0[row] : 0.0;
+          lk[j] = (unsigned)(k & (GB_RANGE - 1));
+          if (NV > 0) lv0[j] = v0[row];
           if (NV > 1) lv1[j] = v1[row];
         } else {
           atomicAdd(err, 1ULL);
@@ -581,24 +595,51 @@ __global__ void __launch_bounds__(BLOCK) k_gb_scatter(
       }
     }
 #pragma unroll
-    for (int j = 0; j < SCAT_RPT; ++j)
+    for (int j = 0; j < RPT; ++j)
       if (lb[j] >= 0) lr[j] = atomicAdd(&it_cnt[lb[j]], 1u);
+    __syncthreads();
+    // wave 0: parallel exclusive scan over the nb tile counters
+    if (threadIdx.x < 64) {
+      const int lane = threadIdx.x;
+      unsigned carry = 0;
+      for (int base = 0; base < nb; base += 64) {
+        const int t = base + lane;
+        unsigned v = (t < nb) ? it_cnt[t] : 0;
+        unsigned incl = v;
+#pragma unroll
+        for (int d = 1; d < 64; d <<= 1) {
+          unsigned up = __shfl_up(incl, d);
+          if (lane >= d) incl += up;
+        }
+        if (t < nb) it_off[t] = carry + incl - v;
+        carry += __shfl(incl, 63);
+      }
+      if (lane == 0) *s_total = carry;
+    }
     __syncthreads();
     for (int t = threadIdx.x; t < nb; t += blockDim.x) {
       const unsigned c = it_cnt[t];
-      if (c) it_base[t] = atomicAdd(&cursors[t], c);
+      if (c) it_gbase[t] = atomicAdd(&cursors[t], c);
     }
-    __syncthreads();
 #pragma unroll
-    for (int j = 0; j < SCAT_RPT; ++j) {
+    for (int j = 0; j < RPT; ++j) {
       if (lb[j] >= 0) {
-        const int64_t pos = (int64_t)it_base[lb[j]] + lr[j];
-        rk[pos] = lk[j];
-        if (NV > 0) r0[pos] = lv0[j];
-        if (NV > 1) r1[pos] = lv1[j];
+        const unsigned p = it_off[lb[j]] + lr[j];
+        skey[p] = ((unsigned)lb[j] << 16) | lk[j];
+        if (NV > 0) sval0[p] = lv0[j];
+        if (NV > 1) sval1[p] = lv1[j];
       }
     }
-    __syncthreads();  // it_cnt reused next tile
+    __syncthreads();
+    const int staged = (int)*s_total;
+    for (int p = threadIdx.x; p < staged; p += blockDim.x) {
+      const unsigned b = skey[p] >> 16;
+      const int64_t pos = (int64_t)it_gbase[b] + (p - it_off[b]);
+      rk[pos] = (unsigned short)(skey[p] & 0xFFFF);
+      if (NV > 0) r0[pos] = sval0[p];
+      if (NV > 1) r1[pos] = sval1[p];
+    }
+    __syncthreads();
   }
 }
 
@@ -1216,21 +1257,29 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
                         hipMemcpyHostToDevice, g.stream));
   // host vectors must outlive the async H2D of pageable memory
   HF_HIP("gb_radix", hipStreamSynchronize(g.stream));
-  // P1 scatter
-  const int64_t ntiles = (n + SCAT_TILE - 1) / SCAT_TILE;
-  const uint32_t sgrid = (uint32_t)std::min<int64_t>(ntiles, 1024);
-  auto scat = [&](auto nvTag) {
+  // P1 scatter: RPT=16 (4096-row tiles) for <=1 value column, RPT=8 for 2
+  // (LDS staging budget)
+  auto scat = [&](auto nvTag, auto rptTag) {
     constexpr int NVv = decltype(nvTag)::value;
+    constexpr int RPTv = decltype(rptTag)::value;
+    const int64_t tile_sz = (int64_t)BLOCK * RPTv;
+    const int64_t ntiles = (n + tile_sz - 1) / tile_sz;
+    const uint32_t sgrid = (uint32_t)std::min<int64_t>(ntiles, 2048);
+    const uint32_t lds =
+        (uint32_t)(tile_sz * (8 * NVv + 4) + nb * 12 + 16);
     return timed_launch("gb_scatter", [&] {
-      hipLaunchKernelGGL((k_gb_scatter<NVv>), dim3(sgrid), dim3(BLOCK),
-                         (uint32_t)(nb * 8), g.stream,
+      hipLaunchKernelGGL((k_gb_scatter<NVv, RPTv>), dim3(sgrid), dim3(BLOCK),
+                         lds, g.stream,
                          (const int64_t*)keys->dptr, ptrs.vals[0], ptrs.vals[1],
                          n, key_min, n_slots, (int)nb, d_cur, r0, r1, rk, d_err);
     });
   };
-  rc = nvals == 0   ? scat(std::integral_constant<int, 0>{})
-       : nvals == 1 ? scat(std::integral_constant<int, 1>{})
-                    : scat(std::integral_constant<int, 2>{});
+  rc = nvals == 0 ? scat(std::integral_constant<int, 0>{},
+                         std::integral_constant<int, 16>{})
+       : nvals == 1 ? scat(std::integral_constant<int, 1>{},
+                           std::integral_constant<int, 16>{})
+                    : scat(std::integral_constant<int, 2>{},
+                           std::integral_constant<int, 8>{});
   if (rc != HF_OK) return rc;
   // P2 aggregate, one column per launch
   const bool cnt = counts != 0;
@@ -1297,7 +1346,7 @@ int hf_groupby_accum(const hf_col* keys, const hf_col* const* vals, int nvals,
   // path selection (DESIGN.md §GroupBy kernels): LDS-dense for small ranges,
   // radix partition for the north-star range, global atomics as the wide
   // fallback (slow but correct for any range below the slot cap)
-  if (n > 0 && n_slots <= GB_RANGE)
+  if (n > 0 && n_slots <= GB_DENSE_MAX)
     return gb_dense_path(keys, ptrs, nvals, key_min, n_slots, sums, rowcnt,
                          counts, d_err);
   const int64_t nb = (n_slots + GB_RANGE - 1) >> GB_RANGE_LOG;
