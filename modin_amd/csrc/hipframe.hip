@@ -546,18 +546,36 @@ __global__ void __launch_bounds__(BLOCK) k_gb_hist(
     if (lhist[t]) atomicAdd(&hist[t], (unsigned long long)lhist[t]);
 }
 
-// P1: LDS-staged bucket-sorted tiles.  Rows are ranked into a per-tile LDS
-// histogram, staged bucket-sorted in LDS, and written out coalesced — each
-// tile contributes one contiguous chunk per bucket stream (probe: 2.4x the
-// register-staged direct scatter).  SCAT_RPT rows/thread; NV value columns.
-template <int NV, int RPT>
-__global__ void __launch_bounds__(BLOCK) k_gb_scatter(
+// P1: LDS-staged bucket-sorted tiles.  Row pairs load 16 B-vectorized
+// (longlong2/double2), are ranked into a per-tile LDS histogram (`ds_add`),
+// prefix-scanned wave-parallel, staged bucket-SORTED in LDS, and written out
+// coalesced — each tile emits one contiguous chunk per bucket stream
+// (probe: 2.6x the register-staged direct scatter).  RPT rows/thread
+// (even); NV value columns; odd tail row handled by block 0 up front.
+template <int NV, int RPT, int BLK>
+__global__ void __launch_bounds__(BLK) k_gb_scatter(
     const int64_t* __restrict__ keys, const double* __restrict__ v0,
     const double* __restrict__ v1, int64_t n, int64_t key_min, int64_t n_slots,
     int nb, unsigned* __restrict__ cursors,
     double* __restrict__ r0, double* __restrict__ r1,
     unsigned short* __restrict__ rk, unsigned long long* __restrict__ err) {
-  constexpr int TILE = BLOCK * RPT;
+  constexpr int TILE = BLK * RPT;
+  constexpr int PAIRS = RPT / 2;
+  if ((n & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
+    // odd tail row: direct single-row reservation + write
+    const int64_t k = keys[n - 1] - key_min;
+    if ((uint64_t)k < (uint64_t)n_slots) {
+      const int b = (int)(k >> GB_RANGE_LOG);
+      const int64_t pos = (int64_t)atomicAdd(&cursors[b], 1u);
+      rk[pos] = (unsigned short)(k & (GB_RANGE - 1));
+      if (NV > 0) r0[pos] = v0[n - 1];
+      if (NV > 1) r1[pos] = v1[n - 1];
+    } else {
+      atomicAdd(err, 1ULL);
+    }
+  }
+  const int64_t neven = n & ~1LL;
+  const int64_t npair_total = neven >> 1;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   double* sval0 = reinterpret_cast<double*>(smem_raw);            // [TILE]
   double* sval1 = sval0 + (NV > 1 ? TILE : 0);
@@ -569,7 +587,7 @@ __global__ void __launch_bounds__(BLOCK) k_gb_scatter(
   unsigned* s_total = it_gbase + nb;  // scalar; keep ALL LDS in the dynamic
                                       // region (a static __shared__ would
                                       // shift the base off 16B — G17)
-  const int64_t ntiles = (n + TILE - 1) / TILE;
+  const int64_t ntiles = (neven + TILE - 1) / TILE;
   for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
     const int64_t t0 = tile * TILE;
     for (int t = threadIdx.x; t < nb; t += blockDim.x) it_cnt[t] = 0;
@@ -579,16 +597,29 @@ __global__ void __launch_bounds__(BLOCK) k_gb_scatter(
     double lv0[RPT], lv1[RPT];
     __syncthreads();
 #pragma unroll
-    for (int j = 0; j < RPT; ++j) {
-      const int64_t row = t0 + (int64_t)j * blockDim.x + threadIdx.x;
-      lb[j] = -1;
-      if (row < n) {
-        const int64_t k = keys[row] - key_min;
-        if ((uint64_t)k < (uint64_t)n_slots) {
-          lb[j] = (int)(k >> GB_RANGE_LOG);
-          lk[j] = (unsigned)(k & (GB_RANGE - 1));
-          if (NV > 0) lv0[j] = v0[row];
-          if (NV > 1) lv1[j] = v1[row];
+    for (int j = 0; j < PAIRS; ++j) {
+      const int64_t pr = (t0 >> 1) + (int64_t)j * BLK + threadIdx.x;
+      const int a = 2 * j, bslot = 2 * j + 1;
+      lb[a] = lb[bslot] = -1;
+      if (pr < npair_total) {
+        const longlong2 kk = reinterpret_cast<const longlong2*>(keys)[pr];
+        const int64_t ka = kk.x - key_min, kb = kk.y - key_min;
+        double2 vv0{}, vv1{};
+        if (NV > 0) vv0 = reinterpret_cast<const double2*>(v0)[pr];
+        if (NV > 1) vv1 = reinterpret_cast<const double2*>(v1)[pr];
+        if ((uint64_t)ka < (uint64_t)n_slots) {
+          lb[a] = (int)(ka >> GB_RANGE_LOG);
+          lk[a] = (unsigned)(ka & (GB_RANGE - 1));
+          if (NV > 0) lv0[a] = vv0.x;
+          if (NV > 1) lv1[a] = vv1.x;
+        } else {
+          atomicAdd(err, 1ULL);
+        }
+        if ((uint64_t)kb < (uint64_t)n_slots) {
+          lb[bslot] = (int)(kb >> GB_RANGE_LOG);
+          lk[bslot] = (unsigned)(kb & (GB_RANGE - 1));
+          if (NV > 0) lv0[bslot] = vv0.y;
+          if (NV > 1) lv1[bslot] = vv1.y;
         } else {
           atomicAdd(err, 1ULL);
         }
@@ -1257,19 +1288,21 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
                         hipMemcpyHostToDevice, g.stream));
   // host vectors must outlive the async H2D of pageable memory
   HF_HIP("gb_radix", hipStreamSynchronize(g.stream));
-  // P1 scatter: RPT=16 (4096-row tiles) for <=1 value column, RPT=8 for 2
-  // (LDS staging budget)
+  // P1 scatter: 512-thread blocks; 8192-row tiles (RPT=16) for <=1 value
+  // column, 4096-row tiles (RPT=8) for 2 (LDS staging budget)
   auto scat = [&](auto nvTag, auto rptTag) {
     constexpr int NVv = decltype(nvTag)::value;
     constexpr int RPTv = decltype(rptTag)::value;
-    const int64_t tile_sz = (int64_t)BLOCK * RPTv;
-    const int64_t ntiles = (n + tile_sz - 1) / tile_sz;
-    const uint32_t sgrid = (uint32_t)std::min<int64_t>(ntiles, 2048);
+    constexpr int BLKv = 512;
+    const int64_t tile_sz = (int64_t)BLKv * RPTv;
+    const int64_t ntiles = ((n & ~1LL) + tile_sz - 1) / tile_sz;
+    const uint32_t sgrid =
+        (uint32_t)std::min<int64_t>(std::max<int64_t>(ntiles, 1), 2048);
     const uint32_t lds =
         (uint32_t)(tile_sz * (8 * NVv + 4) + nb * 12 + 16);
     return timed_launch("gb_scatter", [&] {
-      hipLaunchKernelGGL((k_gb_scatter<NVv, RPTv>), dim3(sgrid), dim3(BLOCK),
-                         lds, g.stream,
+      hipLaunchKernelGGL((k_gb_scatter<NVv, RPTv, BLKv>), dim3(sgrid),
+                         dim3(BLKv), lds, g.stream,
                          (const int64_t*)keys->dptr, ptrs.vals[0], ptrs.vals[1],
                          n, key_min, n_slots, (int)nb, d_cur, r0, r1, rk, d_err);
     });
