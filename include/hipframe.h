@@ -165,6 +165,42 @@ int hf_groupby_compact(uintptr_t sums, uintptr_t rowcnt, uintptr_t counts,
                        hf_col** out_counts,        /* nvals cols or NULL      */
                        int64_t* n_groups);
 
+/* ---- Join: broadcast-right hash (dense-range CSR) inner join ----
+ * Device form of MergeImpl.row_axis_merge
+ * (modin/core/storage_formats/pandas/merge.py:104-178: materialize the right
+ * frame once, broadcast it to every left partition, per-partition
+ * pandas.merge).  Here the "broadcast right" is a device-resident CSR over
+ * the dense key range [key_min, key_min+n_slots): per-key row lists in
+ * right-row order (pandas inner-merge match order), with the right value
+ * columns gathered into CSR order so probing streams sequentially.
+ * The probe preserves pandas semantics: output rows ordered by left row,
+ * then right row within a key; result index is a fresh RangeIndex. */
+typedef struct hf_join hf_join;
+
+int hf_join_build(const hf_col* rkeys,          /* HF_INT64                 */
+                  const hf_col* const* rvals,   /* nr HF_FLOAT64 columns    */
+                  int nr, int64_t key_min, int64_t n_slots, hf_join** out);
+int hf_join_free(hf_join* j);
+
+/* Probe with a left partition.  Outputs: out_keys (match keys), out_lidx
+ * (left row index per match, for gathering left payload columns), and the
+ * nr right columns in match order.  Caller owns all returned columns. */
+int hf_join_probe(const hf_join* j, const hf_col* lkeys,
+                  hf_col** out_keys,            /* HF_INT64 [n_out]         */
+                  hf_col** out_lidx,            /* HF_INT64 [n_out]         */
+                  hf_col** out_rcols,           /* nr cols, caller array    */
+                  int64_t* n_out);
+
+/* Gather col[idx[i]] -> out[i] (materialize left payload columns of a join
+ * result; also the generic `take` kernel). */
+int hf_gather(const hf_col* col, const hf_col* idx, hf_col** out);
+
+/* Concatenate columns device-to-device (the device form of the partition
+ * concat in deploy_axis_func, axis_partition.py:449 — used to materialize a
+ * multi-partition right frame for the broadcast join, combine() at
+ * dataframe.py:2918). */
+int hf_col_concat(const hf_col* const* cols, int ncols, hf_col** out);
+
 /* ---- profiling (bench.py roofline leg) ----
  * When enabled, every kernel launch is bracketed by HIP events on the module
  * stream; hf_kernel_stats returns the accumulated count and total ms for the

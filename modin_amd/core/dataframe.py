@@ -130,6 +130,63 @@ class HipDataframe:
         return HipDataframe([part], DeviceIndex(keys, name=by), val_names,
                             [n], dtypes)
 
+    # ---- broadcast inner join (MergeImpl.row_axis_merge device form,
+    #      merge.py:104-178: combine() the right frame once, probe per left
+    #      partition; pandas suffix rules "_x"/"_y" on collisions) ----
+    def broadcast_join(self, other: "HipDataframe", on: str) -> "HipDataframe":
+        if on not in self.columns or on not in other.columns:
+            raise lib.HfError(f"merge: key column {on!r} missing")
+        left_names = [c for c in self.columns if c != on]
+        right_names = [c for c in other.columns if c != on]
+        common = set(left_names) & set(right_names)
+        lout = {n: (n + "_x" if n in common else n) for n in left_names}
+        rout = {n: (n + "_y" if n in common else n) for n in right_names}
+
+        # materialize right device-side (combine(), dataframe.py:2918)
+        def concat_col(frame, name):
+            cols = [p.block().columns[name] for p in frame._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        rkeys = concat_col(other, on)
+        if rkeys.dtype_code != lib.HF_INT64:
+            raise lib.HfError("merge: key column must be int64 (dense-range "
+                              "CSR join; hashed keys are a later round)")
+        rvals = [concat_col(other, n) for n in right_names]
+        if rkeys.length:
+            r = lib.reduce(rkeys)
+            kmin, n_slots = r.imn, r.imx - r.imn + 1
+        else:
+            kmin, n_slots = 0, 1
+        j = lib.join_build(rkeys, rvals, kmin, n_slots)
+
+        out_parts, lengths = [], []
+        for p in self._partitions:
+            block = p.block()
+            lkeys = block.columns[on]
+            if lkeys.dtype_code != lib.HF_INT64:
+                raise lib.HfError("merge: key column must be int64")
+            keys_c, lidx, rcols, nout = lib.join_probe(j, lkeys)
+            cols = {}
+            for name in self.columns:  # left column order, key in place
+                if name == on:
+                    cols[on] = keys_c
+                else:
+                    cols[lout[name]] = lib.gather(block.columns[name], lidx)
+            for i, rn in enumerate(right_names):
+                cols[rout[rn]] = rcols[i]
+            out_parts.append(HipDataframePartition(DeviceBlock(cols, nout)))
+            lengths.append(nout)
+        out_columns = ([on if c == on else lout[c] for c in self.columns]
+                       + [rout[c] for c in right_names])
+        dtypes = {}
+        for c in self.columns:
+            dtypes[on if c == on else lout[c]] = self.dtypes[c]
+        for c in right_names:
+            dtypes[rout[c]] = other.dtypes[c]
+        total = sum(lengths)
+        return HipDataframe(out_parts, pandas.RangeIndex(total), out_columns,
+                            lengths, pandas.Series(dtypes))
+
     # ---- column selection (getitem_column_array device form) ----
     def take_columns(self, names) -> "HipDataframe":
         def sel(block: DeviceBlock) -> DeviceBlock:

@@ -110,6 +110,19 @@ def load() -> ct.CDLL:
                                               ct.POINTER(ct.c_void_p),
                                               ct.POINTER(ct.c_void_p),
                                               ct.POINTER(ct.c_int64)]),
+            "hf_col_concat": (ct.c_int, [ct.POINTER(ct.c_void_p), ct.c_int,
+                                         ct.POINTER(ct.c_void_p)]),
+            "hf_join_build": (ct.c_int, [ct.c_void_p, ct.POINTER(ct.c_void_p),
+                                         ct.c_int, ct.c_int64, ct.c_int64,
+                                         ct.POINTER(ct.c_void_p)]),
+            "hf_join_free": (ct.c_int, [ct.c_void_p]),
+            "hf_join_probe": (ct.c_int, [ct.c_void_p, ct.c_void_p,
+                                         ct.POINTER(ct.c_void_p),
+                                         ct.POINTER(ct.c_void_p),
+                                         ct.POINTER(ct.c_void_p),
+                                         ct.POINTER(ct.c_int64)]),
+            "hf_gather": (ct.c_int, [ct.c_void_p, ct.c_void_p,
+                                     ct.POINTER(ct.c_void_p)]),
             "hf_profiling": (ct.c_int, [ct.c_int]),
             "hf_kernel_stats": (ct.c_int, [ct.c_char_p, ct.POINTER(ct.c_int64),
                                            ct.POINTER(ct.c_double)]),
@@ -130,8 +143,9 @@ def exported_symbols():
         "hf_put", "hf_get", "hf_col_alloc", "hf_col_free", "hf_col_len",
         "hf_col_dtype", "hf_col_dptr", "hf_alloc_raw", "hf_free_raw",
         "hf_memset_raw", "hf_map_scalar", "hf_map_scalar_i64", "hf_binary",
-        "hf_reduce", "hf_groupby_accum", "hf_groupby_compact", "hf_profiling",
-        "hf_kernel_stats", "hf_kernel_stats_reset",
+        "hf_reduce", "hf_groupby_accum", "hf_groupby_compact",
+        "hf_col_concat", "hf_join_build", "hf_join_free", "hf_join_probe",
+        "hf_gather", "hf_profiling", "hf_kernel_stats", "hf_kernel_stats_reset",
     ]
 
 
@@ -327,6 +341,65 @@ def groupby_compact(sums: int, rowcnt: int, counts: int, nvals: int,
     ccols = ([_wrap(ct.c_void_p(out_counts[c]), n, HF_INT64) for c in range(nvals)]
              if counts else None)
     return kcol, scols, ccols, n
+
+
+def concat(cols: list) -> ColumnRef:
+    ensure_ready()
+    arr = (ct.c_void_p * len(cols))(*[c.handle for c in cols])
+    out = ct.c_void_p()
+    _check(load().hf_col_concat(arr, len(cols), ct.byref(out)), "hf_col_concat")
+    return _wrap(out, sum(c.length for c in cols), cols[0].dtype_code)
+
+
+class JoinRef:
+    """Owner of a built broadcast-right join structure (hf_join)."""
+
+    __slots__ = ("handle", "nr", "rdtypes")
+
+    def __init__(self, handle, nr, rdtypes):
+        self.handle = handle
+        self.nr = nr
+        self.rdtypes = rdtypes
+
+    def __del__(self):
+        try:
+            if _dll is not None and _inited_gpu is not None and self.handle:
+                _dll.hf_join_free(self.handle)
+        except Exception:
+            pass
+
+
+def join_build(rkeys: ColumnRef, rvals: list, key_min: int,
+               n_slots: int) -> JoinRef:
+    ensure_ready()
+    arr = (ct.c_void_p * max(len(rvals), 1))(*[v.handle for v in rvals])
+    out = ct.c_void_p()
+    _check(load().hf_join_build(rkeys.handle, arr, len(rvals), key_min,
+                                n_slots, ct.byref(out)), "hf_join_build")
+    return JoinRef(out, len(rvals), [v.dtype_code for v in rvals])
+
+
+def join_probe(j: JoinRef, lkeys: ColumnRef):
+    """Returns (keys_col, lidx_col, [right_cols], n_out)."""
+    ensure_ready()
+    out_keys = ct.c_void_p()
+    out_lidx = ct.c_void_p()
+    out_r = (ct.c_void_p * max(j.nr, 1))()
+    n_out = ct.c_int64(0)
+    _check(load().hf_join_probe(j.handle, lkeys.handle, ct.byref(out_keys),
+                                ct.byref(out_lidx), out_r, ct.byref(n_out)),
+           "hf_join_probe")
+    n = n_out.value
+    rcols = [_wrap(ct.c_void_p(out_r[c]), n, j.rdtypes[c]) for c in range(j.nr)]
+    return (_wrap(out_keys, n, HF_INT64), _wrap(out_lidx, n, HF_INT64),
+            rcols, n)
+
+
+def gather(col: ColumnRef, idx: ColumnRef) -> ColumnRef:
+    ensure_ready()
+    out = ct.c_void_p()
+    _check(load().hf_gather(col.handle, idx.handle, ct.byref(out)), "hf_gather")
+    return _wrap(out, idx.length, col.dtype_code)
 
 
 def sync() -> None:

@@ -856,7 +856,261 @@ __global__ void __launch_bounds__(BLOCK) k_compact_scatter(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Join kernels — broadcast-right dense-range CSR inner join
+// (device form of MergeImpl.row_axis_merge, merge.py:104-178; see
+// include/hipframe.h).  Build: count -> scan -> fill -> order fixup.
+// Probe: two passes so the output preserves pandas' match order exactly
+// (left row order, right row order within key): pass A stores per-left-row
+// (offset, count) and per-tile totals; a scan of tile totals gives exact
+// output positions; pass B emits.
+// ---------------------------------------------------------------------------
+
+constexpr int JOIN_TILE = 4096;           // left rows per probe tile
+constexpr int JOIN_MAX_DUP = 4096;        // right rows per key cap (fixup sort)
+
+__global__ void __launch_bounds__(BLOCK) k_hist_u32(
+    const int64_t* __restrict__ keys, int64_t n, int64_t key_min,
+    int64_t n_slots, unsigned* __restrict__ cnt,
+    unsigned long long* __restrict__ err) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const int64_t k = keys[i] - key_min;
+    if ((uint64_t)k < (uint64_t)n_slots)
+      atomicAdd(&cnt[k], 1u);
+    else
+      atomicAdd(err, 1ULL);
+  }
+}
+
+// tile sums of a u32 array (for the big exclusive scan building the CSR)
+__global__ void __launch_bounds__(BLOCK) k_tile_sums_u32(
+    const unsigned* __restrict__ v, int64_t n, int64_t* __restrict__ sums) {
+  const int64_t t0 = (int64_t)blockIdx.x * JOIN_TILE;
+  const int64_t t1 = min(t0 + (int64_t)JOIN_TILE, n);
+  long long local = 0;
+  for (int64_t s = t0 + threadIdx.x; s < t1; s += blockDim.x) local += v[s];
+  for (int off = 32; off > 0; off >>= 1) local += __shfl_down(local, off);
+  __shared__ long long sc[BLOCK / 64];
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) sc[wave] = local;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    long long tot = 0;
+    for (int w = 0; w < BLOCK / 64; ++w) tot += sc[w];
+    sums[blockIdx.x] = tot;
+  }
+}
+
+// apply tile bases: csr[s] = tile_base + local exclusive prefix; also writes
+// csr[n] = grand total (u64 offsets)
+__global__ void __launch_bounds__(BLOCK) k_scan_apply_u32(
+    const unsigned* __restrict__ v, int64_t n,
+    const int64_t* __restrict__ tile_bases, const int64_t* __restrict__ total,
+    unsigned long long* __restrict__ csr) {
+  __shared__ unsigned long long s_run;
+  const int64_t t0 = (int64_t)blockIdx.x * JOIN_TILE;
+  const int64_t t1 = min(t0 + (int64_t)JOIN_TILE, n);
+  if (threadIdx.x == 0) s_run = (unsigned long long)tile_bases[blockIdx.x];
+  __shared__ int s_wave_sum[BLOCK / 64];
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  __syncthreads();
+  for (int64_t chunk = t0; chunk < t1; chunk += blockDim.x) {
+    const int64_t s = chunk + threadIdx.x;
+    const unsigned x = (s < t1) ? v[s] : 0;
+    // wave-inclusive scan
+    unsigned incl = x;
+    for (int d = 1; d < 64; d <<= 1) {
+      unsigned up = __shfl_up(incl, d);
+      if (lane >= d) incl += up;
+    }
+    if (lane == 63) s_wave_sum[wave] = (int)incl;
+    __syncthreads();
+    unsigned wave_base = 0;
+    for (int w = 0; w < wave; ++w) wave_base += (unsigned)s_wave_sum[w];
+    if (s < t1) csr[s] = s_run + wave_base + incl - x;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      unsigned tot = 0;
+      for (int w = 0; w < BLOCK / 64; ++w) tot += (unsigned)s_wave_sum[w];
+      s_run += tot;
+    }
+    __syncthreads();
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0) csr[n] = (unsigned long long)*total;
+}
+
+struct JoinPtrs {
+  double* vals[GB_MAX_VALS];
+};
+struct JoinConstPtrs {
+  const double* vals[GB_MAX_VALS];
+};
+
+template <int NR>
+__global__ void __launch_bounds__(BLOCK) k_join_fill(
+    const int64_t* __restrict__ keys, JoinConstPtrs rv, int64_t n,
+    int64_t key_min, int64_t n_slots, const unsigned long long* __restrict__ csr,
+    unsigned* __restrict__ cursor, unsigned* __restrict__ jidx, JoinPtrs jv) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const int64_t k = keys[i] - key_min;
+    if ((uint64_t)k >= (uint64_t)n_slots) continue;  // counted in hist err
+    const unsigned r = atomicAdd(&cursor[k], 1u);
+    const int64_t pos = (int64_t)csr[k] + r;
+    jidx[pos] = (unsigned)i;
+#pragma unroll
+    for (int c = 0; c < NR; ++c) jv.vals[c][pos] = rv.vals[c][i];
+  }
+}
+
+// restore right-row order within each key (pandas match order): tiny
+// insertion sort per multi-occupancy slot, one thread per slot
+template <int NR>
+__global__ void __launch_bounds__(BLOCK) k_join_fixup(
+    const unsigned long long* __restrict__ csr, int64_t n_slots,
+    unsigned* __restrict__ jidx, JoinPtrs jv,
+    unsigned long long* __restrict__ err) {
+  int64_t s = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; s < n_slots; s += stride) {
+    const int64_t lo = (int64_t)csr[s], hi = (int64_t)csr[s + 1];
+    const int64_t cnt = hi - lo;
+    if (cnt <= 1) continue;
+    if (cnt > JOIN_MAX_DUP) {
+      atomicAdd(err, 1ULL);
+      continue;
+    }
+    for (int64_t a = lo + 1; a < hi; ++a) {
+      const unsigned ki = jidx[a];
+      double kv[NR > 0 ? NR : 1];
+#pragma unroll
+      for (int c = 0; c < NR; ++c) kv[c] = jv.vals[c][a];
+      int64_t b = a - 1;
+      while (b >= lo && jidx[b] > ki) {
+        jidx[b + 1] = jidx[b];
+#pragma unroll
+        for (int c = 0; c < NR; ++c) jv.vals[c][b + 1] = jv.vals[c][b];
+        --b;
+      }
+      jidx[b + 1] = ki;
+#pragma unroll
+      for (int c = 0; c < NR; ++c) jv.vals[c][b + 1] = kv[c];
+    }
+  }
+}
+
+// probe pass A: per-left-row offset+count, per-tile match totals
+__global__ void __launch_bounds__(BLOCK) k_probe_count(
+    const int64_t* __restrict__ lkeys, int64_t n, int64_t key_min,
+    int64_t n_slots, const unsigned long long* __restrict__ csr,
+    unsigned long long* __restrict__ offs, unsigned* __restrict__ cnts,
+    int64_t* __restrict__ tile_sums) {
+  const int64_t t0 = (int64_t)blockIdx.x * JOIN_TILE;
+  const int64_t t1 = min(t0 + (int64_t)JOIN_TILE, n);
+  long long local = 0;
+  for (int64_t i = t0 + threadIdx.x; i < t1; i += blockDim.x) {
+    const int64_t k = lkeys[i] - key_min;
+    unsigned c = 0;
+    unsigned long long o = 0;
+    if ((uint64_t)k < (uint64_t)n_slots) {
+      o = csr[k];
+      c = (unsigned)(csr[k + 1] - o);
+    }
+    offs[i] = o;
+    cnts[i] = c;
+    local += c;
+  }
+  for (int off = 32; off > 0; off >>= 1) local += __shfl_down(local, off);
+  __shared__ long long sc[BLOCK / 64];
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) sc[wave] = local;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    long long tot = 0;
+    for (int w = 0; w < BLOCK / 64; ++w) tot += sc[w];
+    tile_sums[blockIdx.x] = tot;
+  }
+}
+
+// probe pass B: emit matches at exact positions
+template <int NR>
+__global__ void __launch_bounds__(BLOCK) k_probe_emit(
+    const int64_t* __restrict__ lkeys, int64_t n, int64_t lbase,
+    const unsigned long long* __restrict__ offs,
+    const unsigned* __restrict__ cnts, const int64_t* __restrict__ tile_bases,
+    const unsigned* __restrict__ jidx_unused, JoinConstPtrs jv,
+    int64_t* __restrict__ out_keys, int64_t* __restrict__ out_lidx,
+    JoinPtrs out_r) {
+  __shared__ unsigned long long s_run;
+  __shared__ int s_wave_sum[BLOCK / 64];
+  const int64_t t0 = (int64_t)blockIdx.x * JOIN_TILE;
+  const int64_t t1 = min(t0 + (int64_t)JOIN_TILE, n);
+  if (threadIdx.x == 0) s_run = (unsigned long long)tile_bases[blockIdx.x];
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  __syncthreads();
+  for (int64_t chunk = t0; chunk < t1; chunk += blockDim.x) {
+    const int64_t i = chunk + threadIdx.x;
+    const unsigned c = (i < t1) ? cnts[i] : 0;
+    unsigned incl = c;
+    for (int d = 1; d < 64; d <<= 1) {
+      unsigned up = __shfl_up(incl, d);
+      if (lane >= d) incl += up;
+    }
+    if (lane == 63) s_wave_sum[wave] = (int)incl;
+    __syncthreads();
+    unsigned wave_base = 0;
+    for (int w = 0; w < wave; ++w) wave_base += (unsigned)s_wave_sum[w];
+    if (i < t1 && c) {
+      int64_t dst = (int64_t)(s_run + wave_base + incl - c);
+      const int64_t k = lkeys[i];
+      const unsigned long long o = offs[i];
+      for (unsigned j = 0; j < c; ++j) {
+        out_keys[dst + j] = k;
+        out_lidx[dst + j] = lbase + i;
+#pragma unroll
+        for (int cc = 0; cc < NR; ++cc)
+          out_r.vals[cc][dst + j] = jv.vals[cc][o + j];
+      }
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      unsigned tot = 0;
+      for (int w = 0; w < BLOCK / 64; ++w) tot += (unsigned)s_wave_sum[w];
+      s_run += tot;
+    }
+    __syncthreads();
+  }
+}
+
+__global__ void __launch_bounds__(BLOCK) k_gather_f64(
+    const double* __restrict__ src, const int64_t* __restrict__ idx,
+    double* __restrict__ out, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = src[idx[i]];
+}
+__global__ void __launch_bounds__(BLOCK) k_gather_i64(
+    const int64_t* __restrict__ src, const int64_t* __restrict__ idx,
+    int64_t* __restrict__ out, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = src[idx[i]];
+}
+
 }  // namespace
+
+struct hf_join {
+  int64_t key_min, n_slots, n_right;
+  int nr;
+  unsigned long long* d_csr;   // [n_slots+1]
+  unsigned* d_jidx;            // [n_right] right row index, key-grouped
+  double* d_jval[GB_MAX_VALS]; // [nr][n_right] right values in CSR order
+                               // (opaque 8 B payloads — i64 or f64)
+  int dtypes[GB_MAX_VALS];     // per right column
+};
 
 // forward decls used by the map launch helpers (defined in the C ABI below)
 extern "C" int hf_col_alloc(int64_t len, int dtype, hf_col** out);
@@ -1511,6 +1765,248 @@ int hf_groupby_compact(uintptr_t sums, uintptr_t rowcnt, uintptr_t counts,
   if (rc != HF_OK) return rc;
   *n_groups = total;
   return HF_OK;
+}
+
+int hf_col_concat(const hf_col* const* cols, int ncols, hf_col** out) {
+  HF_NEED_INIT("hf_col_concat");
+  if (!cols || ncols <= 0 || !out)
+    return set_err(HF_ERR_ARG, "hf_col_concat", "bad args");
+  const int dt = cols[0]->dtype;
+  int64_t total = 0;
+  for (int i = 0; i < ncols; ++i) {
+    if (!cols[i] || cols[i]->dtype != dt)
+      return set_err(HF_ERR_ARG, "hf_col_concat", "dtype mismatch");
+    total += cols[i]->len;
+  }
+  int rc = hf_col_alloc(total, dt, out);
+  if (rc != HF_OK) return rc;
+  char* dst = (char*)(*out)->dptr;
+  const int64_t esz = dtype_size(dt);
+  for (int i = 0; i < ncols; ++i) {
+    const int64_t bytes = cols[i]->len * esz;
+    if (bytes > 0)
+      HF_HIP("hf_col_concat",
+             hipMemcpyAsync(dst, cols[i]->dptr, bytes,
+                            hipMemcpyDeviceToDevice, g.stream));
+    dst += bytes;
+  }
+  return HF_OK;
+}
+
+// ---- join ----
+
+int hf_join_build(const hf_col* rkeys, const hf_col* const* rvals, int nr,
+                  int64_t key_min, int64_t n_slots, hf_join** out) {
+  HF_NEED_INIT("hf_join_build");
+  if (!rkeys || nr < 0 || nr > GB_MAX_VALS || !out || n_slots <= 0)
+    return set_err(HF_ERR_ARG, "hf_join_build", "bad args (nr<=8)");
+  if (rkeys->dtype != HF_INT64)
+    return set_err(HF_ERR_ARG, "hf_join_build", "right keys must be int64");
+  if (n_slots > (1LL << 27))
+    return set_err(HF_ERR_UNSUPPORTED, "hf_join_build",
+                   "right key range too large for the dense-range CSR "
+                   "(hash build is a later round)");
+  JoinConstPtrs rv{};
+  for (int c = 0; c < nr; ++c) {
+    // the join only MOVES right payloads (8 B opaque), so both dtypes pass
+    if (!rvals[c] || dtype_size(rvals[c]->dtype) != 8 ||
+        rvals[c]->len != rkeys->len)
+      return set_err(HF_ERR_ARG, "hf_join_build",
+                     "right vals must be 8-byte columns of keys' length");
+    rv.vals[c] = (const double*)rvals[c]->dptr;
+  }
+  const int64_t n = rkeys->len;
+  unsigned long long* d_err =
+      (unsigned long long*)((char*)g.d_scratch + SCRATCH_GB_ERR);
+  int64_t* d_total = (int64_t*)((char*)g.d_scratch + SCRATCH_NGROUPS);
+  unsigned* d_cnt = nullptr;
+  int64_t* d_tiles = nullptr;
+  unsigned long long* d_csr = nullptr;
+  const int64_t ntiles = (n_slots + JOIN_TILE - 1) / JOIN_TILE;
+  HF_HIP("hf_join_build", hipMallocAsync((void**)&d_cnt, n_slots * 4, g.stream));
+  HF_HIP("hf_join_build", hipMemsetAsync(d_cnt, 0, n_slots * 4, g.stream));
+  HF_HIP("hf_join_build",
+         hipMallocAsync((void**)&d_tiles, ntiles * 8, g.stream));
+  HF_HIP("hf_join_build",
+         hipMallocAsync((void**)&d_csr, (n_slots + 1) * 8, g.stream));
+  int rc = timed_launch("join_hist", [&] {
+    hipLaunchKernelGGL(k_hist_u32, dim3((uint32_t)grid_for(n)), dim3(BLOCK), 0,
+                       g.stream, (const int64_t*)rkeys->dptr, n, key_min,
+                       n_slots, d_cnt, d_err);
+  });
+  if (rc != HF_OK) return rc;
+  rc = timed_launch("join_scan", [&] {
+    hipLaunchKernelGGL(k_tile_sums_u32, dim3((uint32_t)ntiles), dim3(BLOCK), 0,
+                       g.stream, d_cnt, n_slots, d_tiles);
+    hipLaunchKernelGGL(k_compact_scan, dim3(1), dim3(1024), 0, g.stream,
+                       d_tiles, ntiles, d_total);
+    hipLaunchKernelGGL(k_scan_apply_u32, dim3((uint32_t)ntiles), dim3(BLOCK),
+                       0, g.stream, d_cnt, n_slots, d_tiles, d_total, d_csr);
+  });
+  if (rc != HF_OK) return rc;
+  // reuse d_cnt as the fill cursor
+  HF_HIP("hf_join_build", hipMemsetAsync(d_cnt, 0, n_slots * 4, g.stream));
+  hf_join* j = new hf_join{};
+  j->key_min = key_min;
+  j->n_slots = n_slots;
+  j->n_right = n;
+  j->nr = nr;
+  j->d_csr = d_csr;
+  for (int c = 0; c < nr; ++c) j->dtypes[c] = rvals[c]->dtype;
+  const int64_t alloc_n = n > 0 ? n : 1;
+  HF_HIP("hf_join_build",
+         hipMallocAsync((void**)&j->d_jidx, alloc_n * 4, g.stream));
+  JoinPtrs jv{};
+  for (int c = 0; c < nr; ++c) {
+    HF_HIP("hf_join_build",
+           hipMallocAsync((void**)&j->d_jval[c], alloc_n * 8, g.stream));
+    jv.vals[c] = j->d_jval[c];
+  }
+  auto fill = [&](auto nrTag) {
+    constexpr int NR = decltype(nrTag)::value;
+    int r2 = timed_launch("join_fill", [&] {
+      hipLaunchKernelGGL((k_join_fill<NR>), dim3((uint32_t)grid_for(n)),
+                         dim3(BLOCK), 0, g.stream, (const int64_t*)rkeys->dptr,
+                         rv, n, key_min, n_slots, d_csr, d_cnt, j->d_jidx, jv);
+    });
+    if (r2 != HF_OK) return r2;
+    return timed_launch("join_fixup", [&] {
+      hipLaunchKernelGGL((k_join_fixup<NR>), dim3(2048), dim3(BLOCK), 0,
+                         g.stream, d_csr, n_slots, j->d_jidx, jv, d_err);
+    });
+  };
+  switch (nr) {
+#define HF_JB_CASE(NR) case NR: rc = fill(std::integral_constant<int, NR>{}); break;
+    HF_JB_CASE(0) HF_JB_CASE(1) HF_JB_CASE(2) HF_JB_CASE(3) HF_JB_CASE(4)
+    HF_JB_CASE(5) HF_JB_CASE(6) HF_JB_CASE(7) HF_JB_CASE(8)
+#undef HF_JB_CASE
+  }
+  hipFreeAsync(d_cnt, g.stream);
+  hipFreeAsync(d_tiles, g.stream);
+  if (rc != HF_OK) { hf_join_free(j); return rc; }
+  // surface hist/fixup errors
+  unsigned long long h_err = 0;
+  HF_HIP("hf_join_build",
+         hipMemcpyAsync(&h_err, d_err, 8, hipMemcpyDeviceToHost, g.stream));
+  HF_HIP("hf_join_build", hipStreamSynchronize(g.stream));
+  if (h_err) {
+    hipMemsetAsync(d_err, 0, 8, g.stream);
+    hf_join_free(j);
+    return set_err(HF_ERR_ARG, "hf_join_build",
+                   "right keys outside [key_min, key_min+n_slots) or a key "
+                   "exceeded the 4096-duplicate cap");
+  }
+  *out = j;
+  return HF_OK;
+}
+
+int hf_join_free(hf_join* j) {
+  if (!j) return HF_OK;
+  if (g.inited) {
+    if (j->d_csr) hipFreeAsync(j->d_csr, g.stream);
+    if (j->d_jidx) hipFreeAsync(j->d_jidx, g.stream);
+    for (int c = 0; c < j->nr; ++c)
+      if (j->d_jval[c]) hipFreeAsync(j->d_jval[c], g.stream);
+  }
+  delete j;
+  return HF_OK;
+}
+
+int hf_join_probe(const hf_join* j, const hf_col* lkeys, hf_col** out_keys,
+                  hf_col** out_lidx, hf_col** out_rcols, int64_t* n_out) {
+  HF_NEED_INIT("hf_join_probe");
+  if (!j || !lkeys || !out_keys || !out_lidx || !n_out)
+    return set_err(HF_ERR_ARG, "hf_join_probe", "null");
+  if (lkeys->dtype != HF_INT64)
+    return set_err(HF_ERR_ARG, "hf_join_probe", "left keys must be int64");
+  const int64_t n = lkeys->len;
+  const int64_t ntiles = n > 0 ? (n + JOIN_TILE - 1) / JOIN_TILE : 1;
+  unsigned long long* d_offs = nullptr;
+  unsigned* d_cnts = nullptr;
+  int64_t* d_tiles = nullptr;
+  int64_t* d_total = (int64_t*)((char*)g.d_scratch + SCRATCH_NGROUPS);
+  const int64_t alloc_n = n > 0 ? n : 1;
+  HF_HIP("hf_join_probe",
+         hipMallocAsync((void**)&d_offs, alloc_n * 8, g.stream));
+  HF_HIP("hf_join_probe",
+         hipMallocAsync((void**)&d_cnts, alloc_n * 4, g.stream));
+  HF_HIP("hf_join_probe", hipMallocAsync((void**)&d_tiles, ntiles * 8, g.stream));
+  HF_HIP("hf_join_probe", hipMemsetAsync(d_tiles, 0, ntiles * 8, g.stream));
+  int rc = HF_OK;
+  if (n > 0) {
+    rc = timed_launch("join_probe_count", [&] {
+      hipLaunchKernelGGL(k_probe_count, dim3((uint32_t)ntiles), dim3(BLOCK), 0,
+                         g.stream, (const int64_t*)lkeys->dptr, n, j->key_min,
+                         j->n_slots, j->d_csr, d_offs, d_cnts, d_tiles);
+    });
+    if (rc != HF_OK) return rc;
+  }
+  hipLaunchKernelGGL(k_compact_scan, dim3(1), dim3(1024), 0, g.stream, d_tiles,
+                     ntiles, d_total);
+  int64_t total = 0;
+  HF_HIP("hf_join_probe",
+         hipMemcpyAsync(&total, d_total, 8, hipMemcpyDeviceToHost, g.stream));
+  HF_HIP("hf_join_probe", hipStreamSynchronize(g.stream));
+  rc = hf_col_alloc(total, HF_INT64, out_keys);
+  if (rc != HF_OK) return rc;
+  rc = hf_col_alloc(total, HF_INT64, out_lidx);
+  if (rc != HF_OK) return rc;
+  JoinPtrs outr{};
+  JoinConstPtrs jvc{};
+  for (int c = 0; c < j->nr; ++c) {
+    rc = hf_col_alloc(total, j->dtypes[c], &out_rcols[c]);
+    if (rc != HF_OK) return rc;
+    outr.vals[c] = (double*)out_rcols[c]->dptr;
+    jvc.vals[c] = j->d_jval[c];
+  }
+  if (n > 0 && total > 0) {
+    auto emit = [&](auto nrTag) {
+      constexpr int NR = decltype(nrTag)::value;
+      return timed_launch("join_probe_emit", [&] {
+        hipLaunchKernelGGL((k_probe_emit<NR>), dim3((uint32_t)ntiles),
+                           dim3(BLOCK), 0, g.stream,
+                           (const int64_t*)lkeys->dptr, n, (int64_t)0, d_offs,
+                           d_cnts, d_tiles, j->d_jidx, jvc,
+                           (int64_t*)(*out_keys)->dptr,
+                           (int64_t*)(*out_lidx)->dptr, outr);
+      });
+    };
+    switch (j->nr) {
+#define HF_JP_CASE(NR) case NR: rc = emit(std::integral_constant<int, NR>{}); break;
+      HF_JP_CASE(0) HF_JP_CASE(1) HF_JP_CASE(2) HF_JP_CASE(3) HF_JP_CASE(4)
+      HF_JP_CASE(5) HF_JP_CASE(6) HF_JP_CASE(7) HF_JP_CASE(8)
+#undef HF_JP_CASE
+    }
+  }
+  hipFreeAsync(d_offs, g.stream);
+  hipFreeAsync(d_cnts, g.stream);
+  hipFreeAsync(d_tiles, g.stream);
+  if (rc != HF_OK) return rc;
+  *n_out = total;
+  return HF_OK;
+}
+
+int hf_gather(const hf_col* col, const hf_col* idx, hf_col** out) {
+  HF_NEED_INIT("hf_gather");
+  if (!col || !idx || !out) return set_err(HF_ERR_ARG, "hf_gather", "null");
+  if (idx->dtype != HF_INT64)
+    return set_err(HF_ERR_ARG, "hf_gather", "index must be int64");
+  int rc = hf_col_alloc(idx->len, col->dtype, out);
+  if (rc != HF_OK) return rc;
+  const int64_t n = idx->len;
+  if (n == 0) return HF_OK;
+  rc = timed_launch("gather", [&] {
+    if (col->dtype == HF_FLOAT64)
+      hipLaunchKernelGGL(k_gather_f64, dim3((uint32_t)grid_for(n)), dim3(BLOCK),
+                         0, g.stream, (const double*)col->dptr,
+                         (const int64_t*)idx->dptr, (double*)(*out)->dptr, n);
+    else
+      hipLaunchKernelGGL(k_gather_i64, dim3((uint32_t)grid_for(n)), dim3(BLOCK),
+                         0, g.stream, (const int64_t*)col->dptr,
+                         (const int64_t*)idx->dptr, (int64_t*)(*out)->dptr, n);
+  });
+  if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  return rc;
 }
 
 // ---- profiling ----

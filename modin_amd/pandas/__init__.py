@@ -158,6 +158,12 @@ class DataFrame(_HipPandasBase):
             )
         raise lib.HfError("only column selection is supported")
 
+    def merge(self, other: "DataFrame", on: str, how: str = "inner"):
+        """Inner merge on an int64 key column (modin/pandas API ->
+        qc.merge -> broadcast-right device join)."""
+        return DataFrame(query_compiler=self._query_compiler.merge(
+            other._query_compiler, on=on, how=how))
+
     def groupby(self, by: str) -> "DataFrameGroupBy":
         if not isinstance(by, str) or by not in list(self.columns):
             raise lib.HfError("groupby(by=<column name>) only")

@@ -89,6 +89,18 @@ class HipQueryCompiler:
             )
         return fn(self, by)
 
+    # ---- merge (query_compiler merge -> MergeImpl.row_axis_merge,
+    #      storage_formats/pandas/merge.py:104) ----
+    def merge(self, right: "HipQueryCompiler", on: str,
+              how: str = "inner") -> "HipQueryCompiler":
+        if how != "inner":
+            raise lib.HfError(
+                f"merge how={how!r} not implemented (inner broadcast join "
+                "this round)")
+        return self.__constructor__(
+            self._modin_frame.broadcast_join(right._modin_frame, on)
+        )
+
     # ---- projection ----
     def getitem_column_array(self, names) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.take_columns(list(names)))

@@ -18,6 +18,7 @@ path against this oracle and the same vectors.
 from .ops import (  # noqa: F401
     binary_op,
     groupby_agg,
+    inner_join,
     map_op,
     partitioned_groupby_agg,
     reduce_op,

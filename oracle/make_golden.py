@@ -176,6 +176,52 @@ def gen_map_binary_cases(mpd, rng):
     return cases
 
 
+def gen_merge_cases(mpd, rng):
+    import pandas
+    cases = {}
+
+    def make(name, lk, lcols, rk, rcols):
+        ldata = {"k": lk.astype(np.int64), **lcols}
+        rdata = {"k": rk.astype(np.int64), **rcols}
+        mL, mR = mpd.DataFrame(dict(ldata)), mpd.DataFrame(dict(rdata))
+        pL, pR = pandas.DataFrame(dict(ldata)), pandas.DataFrame(dict(rdata))
+        mres = mL.merge(mR, on="k")._to_pandas()
+        pres = pL.merge(pR, on="k")
+        assert list(mres.columns) == list(pres.columns)
+        assert (mres.index == pres.index).all()
+        for c in pres.columns:
+            np.testing.assert_array_equal(mres[c].to_numpy(), pres[c].to_numpy())
+        arrays = {"in_lk": ldata["k"], "in_rk": rdata["k"]}
+        for n, v in lcols.items():
+            arrays[f"in_l_{n}"] = v
+        for n, v in rcols.items():
+            arrays[f"in_r_{n}"] = v
+        arrays["out_columns"] = np.array(list(pres.columns), dtype="U32")
+        for c in pres.columns:
+            arrays[f"out_{c}"] = pres[c].to_numpy()
+        cases[name] = arrays
+
+    nl, nr_ = 3000, 500
+    make("mg_basic",
+         rng.integers(0, 100, nl), {"lv": rng.random(nl),
+                                    "li": rng.integers(-9, 9, nl).astype(np.int64)},
+         rng.integers(0, 120, nr_), {"rv": rng.random(nr_) * 10})
+    make("mg_dup_right",
+         rng.integers(0, 40, nl), {"lv": rng.random(nl)},
+         rng.integers(0, 40, nr_), {"rv": rng.random(nr_),
+                                    "ri": rng.integers(0, 5, nr_).astype(np.int64)})
+    make("mg_collide",
+         rng.integers(0, 50, nl), {"v": rng.random(nl)},
+         rng.integers(0, 50, nr_), {"v": rng.random(nr_)})
+    make("mg_disjoint",
+         rng.integers(0, 50, nl), {"lv": rng.random(nl)},
+         rng.integers(1000, 1050, nr_), {"rv": rng.random(nr_)})
+    make("mg_negative",
+         rng.integers(-30, 30, nl), {"lv": rng.random(nl)},
+         rng.integers(-30, 30, nr_), {"rv": rng.random(nr_)})
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -184,6 +230,7 @@ def main():
     all_cases.update(gen_groupby_cases(mpd, rng))
     all_cases.update(gen_reduce_cases(mpd, rng))
     all_cases.update(gen_map_binary_cases(mpd, rng))
+    all_cases.update(gen_merge_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)

@@ -166,6 +166,45 @@ def groupby_agg(keys: np.ndarray, vals: dict, agg: str):
     return out_keys, out
 
 
+# ---------------------------------------------------------------------------
+# Merge — MergeImpl.row_axis_merge (storage_formats/pandas/merge.py:104-178):
+# materialize the right frame once (combine(), dataframe.py:2918), broadcast
+# to every left partition, per-partition pandas.merge(how="inner").
+# pandas inner-merge semantics restated: output rows ordered by left row,
+# then right row within a key; result index is a fresh RangeIndex; key
+# column keeps its left position; colliding value columns get _x/_y.
+# ---------------------------------------------------------------------------
+
+def inner_join(lk: np.ndarray, lvals: dict, rk: np.ndarray, rvals: dict):
+    """Returns (keys, lidx, out_lvals, out_rvals) in pandas match order."""
+    lk = np.asarray(lk)
+    rk = np.asarray(rk)
+    if rk.size == 0 or lk.size == 0:
+        z = np.empty(0, np.int64)
+        return (z, z, {n: np.asarray(v)[:0] for n, v in lvals.items()},
+                {n: np.asarray(v)[:0] for n, v in rvals.items()})
+    kmin = int(rk.min())
+    slots = int(rk.max()) - kmin + 1
+    cnt = np.bincount(rk - kmin, minlength=slots)
+    csr = np.zeros(slots + 1, dtype=np.int64)
+    np.cumsum(cnt, out=csr[1:])
+    order = np.argsort(rk, kind="stable")  # right rows grouped, row-ordered
+    lkk = lk - kmin
+    inr = (lkk >= 0) & (lkk < slots)
+    clipped = np.clip(lkk, 0, slots - 1)
+    lcnt = np.where(inr, cnt[clipped], 0)
+    total = int(lcnt.sum())
+    lidx = np.repeat(np.arange(lk.size, dtype=np.int64), lcnt)
+    starts = np.repeat(csr[clipped], lcnt)
+    base = np.repeat(np.cumsum(lcnt) - lcnt, lcnt)
+    rpos = starts + (np.arange(total, dtype=np.int64) - base)
+    ridx = order[rpos]
+    keys = lk[lidx]
+    return (keys, lidx,
+            {n: np.asarray(v)[lidx] for n, v in lvals.items()},
+            {n: np.asarray(v)[ridx] for n, v in rvals.items()})
+
+
 def partitioned_groupby_agg(keys: np.ndarray, vals: dict, agg: str,
                             num_splits: int, min_size: int = 32):
     """The two-phase form the reference actually runs (map per partition,

@@ -178,6 +178,36 @@ def test_error_surfaces():
         config.MaxGroupbySlots.put(old)
 
 
+@pytest.mark.parametrize("case", golden_cases("mg_"))
+def test_merge_vs_golden(case, npartitions):
+    g = load_golden(case)
+    lcols = {k[len("in_l_"):]: v for k, v in g.items() if k.startswith("in_l_")}
+    rcols = {k[len("in_r_"):]: v for k, v in g.items() if k.startswith("in_r_")}
+    left = mpd.DataFrame({"k": g["in_lk"], **lcols})
+    right = mpd.DataFrame({"k": g["in_rk"], **rcols})
+    out = left.merge(right, on="k").to_pandas()
+    expect_cols = [str(c) for c in g["out_columns"]]
+    assert list(out.columns) == expect_cols
+    assert (out.index == pandas.RangeIndex(len(out))).all()
+    for c in expect_cols:
+        np.testing.assert_array_equal(out[c].to_numpy(), g[f"out_{c}"],
+                                      err_msg=c)
+        assert out[c].dtype == g[f"out_{c}"].dtype
+
+
+def test_merge_error_surfaces():
+    rng = np.random.default_rng(21)
+    left = mpd.DataFrame({"k": rng.random(10), "v": rng.random(10)})
+    right = mpd.DataFrame({"k": rng.integers(0, 5, 10).astype(np.int64),
+                           "w": rng.random(10)})
+    with pytest.raises(lib.HfError, match="int64"):
+        left.merge(right, on="k")
+    ok = mpd.DataFrame({"k": rng.integers(0, 5, 10).astype(np.int64),
+                        "v": rng.random(10)})
+    with pytest.raises(lib.HfError, match="not implemented"):
+        ok.merge(right, on="k", how="left")
+
+
 def test_native_extension_is_loaded():
     """Guard against a silent eager/pandas fallback: the in-tree .so must be
     mapped into this process."""

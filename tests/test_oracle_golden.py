@@ -92,6 +92,36 @@ def test_split_row_counts_matches_reference_rule():
     assert len(oracle.split_row_counts(10**6, 8, 32)) == 8
 
 
+def _merge_inputs(g):
+    lcols = {k[len("in_l_"):]: v for k, v in g.items() if k.startswith("in_l_")}
+    rcols = {k[len("in_r_"):]: v for k, v in g.items() if k.startswith("in_r_")}
+    return g["in_lk"], lcols, g["in_rk"], rcols
+
+
+def oracle_merge_columns(lk, lcols, rk, rcols):
+    """Assemble the full pandas-shaped merge output from the oracle join
+    (suffix rules of pandas merge: colliding names get _x/_y)."""
+    keys, lidx, out_l, out_r = oracle.inner_join(lk, lcols, rk, rcols)
+    common = set(lcols) & set(rcols)
+    out = {"k": keys}
+    for n in lcols:
+        out[n + "_x" if n in common else n] = out_l[n]
+    for n in rcols:
+        out[n + "_y" if n in common else n] = out_r[n]
+    return out
+
+
+@pytest.mark.parametrize("case", golden_cases("mg_"))
+def test_merge_vs_golden(case):
+    g = load_golden(case)
+    lk, lcols, rk, rcols = _merge_inputs(g)
+    out = oracle_merge_columns(lk, lcols, rk, rcols)
+    expect_cols = [str(c) for c in g["out_columns"]]
+    assert set(out) == set(expect_cols)
+    for c in expect_cols:
+        np.testing.assert_array_equal(out[c], g[f"out_{c}"], err_msg=c)
+
+
 def test_groupby_empty():
     keys, out = oracle.groupby_agg(np.empty(0, np.int64), {"v": np.empty(0)}, "sum")
     assert keys.size == 0 and out["v"].size == 0
