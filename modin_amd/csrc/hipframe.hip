@@ -75,6 +75,8 @@ State g;
 constexpr int64_t SCRATCH_BYTES = 4096;
 constexpr int64_t SCRATCH_GB_ERR = 128;
 constexpr int64_t SCRATCH_NGROUPS = 136;
+constexpr int64_t SCRATCH_JOIN_HIST_ERR = 144;
+constexpr int64_t SCRATCH_JOIN_FIXUP_ERR = 152;
 
 int set_err(int code, const char* where, const char* what) {
   g_err = std::string(where) + ": " + what;
@@ -1816,8 +1818,10 @@ int hf_join_build(const hf_col* rkeys, const hf_col* const* rvals, int nr,
     rv.vals[c] = (const double*)rvals[c]->dptr;
   }
   const int64_t n = rkeys->len;
-  unsigned long long* d_err =
-      (unsigned long long*)((char*)g.d_scratch + SCRATCH_GB_ERR);
+  unsigned long long* d_hist_err =
+      (unsigned long long*)((char*)g.d_scratch + SCRATCH_JOIN_HIST_ERR);
+  unsigned long long* d_fix_err =
+      (unsigned long long*)((char*)g.d_scratch + SCRATCH_JOIN_FIXUP_ERR);
   int64_t* d_total = (int64_t*)((char*)g.d_scratch + SCRATCH_NGROUPS);
   unsigned* d_cnt = nullptr;
   int64_t* d_tiles = nullptr;
@@ -1832,7 +1836,7 @@ int hf_join_build(const hf_col* rkeys, const hf_col* const* rvals, int nr,
   int rc = timed_launch("join_hist", [&] {
     hipLaunchKernelGGL(k_hist_u32, dim3((uint32_t)grid_for(n)), dim3(BLOCK), 0,
                        g.stream, (const int64_t*)rkeys->dptr, n, key_min,
-                       n_slots, d_cnt, d_err);
+                       n_slots, d_cnt, d_hist_err);
   });
   if (rc != HF_OK) return rc;
   rc = timed_launch("join_scan", [&] {
@@ -1872,7 +1876,7 @@ int hf_join_build(const hf_col* rkeys, const hf_col* const* rvals, int nr,
     if (r2 != HF_OK) return r2;
     return timed_launch("join_fixup", [&] {
       hipLaunchKernelGGL((k_join_fixup<NR>), dim3(2048), dim3(BLOCK), 0,
-                         g.stream, d_csr, n_slots, j->d_jidx, jv, d_err);
+                         g.stream, d_csr, n_slots, j->d_jidx, jv, d_fix_err);
     });
   };
   switch (nr) {
@@ -1885,16 +1889,24 @@ int hf_join_build(const hf_col* rkeys, const hf_col* const* rvals, int nr,
   hipFreeAsync(d_tiles, g.stream);
   if (rc != HF_OK) { hf_join_free(j); return rc; }
   // surface hist/fixup errors
-  unsigned long long h_err = 0;
+  unsigned long long h_err[2] = {0, 0};
   HF_HIP("hf_join_build",
-         hipMemcpyAsync(&h_err, d_err, 8, hipMemcpyDeviceToHost, g.stream));
+         hipMemcpyAsync(&h_err[0], d_hist_err, 8, hipMemcpyDeviceToHost,
+                        g.stream));
+  HF_HIP("hf_join_build",
+         hipMemcpyAsync(&h_err[1], d_fix_err, 8, hipMemcpyDeviceToHost,
+                        g.stream));
   HF_HIP("hf_join_build", hipStreamSynchronize(g.stream));
-  if (h_err) {
-    hipMemsetAsync(d_err, 0, 8, g.stream);
+  if (h_err[0] || h_err[1]) {
+    hipMemsetAsync(d_hist_err, 0, 8, g.stream);
+    hipMemsetAsync(d_fix_err, 0, 8, g.stream);
     hf_join_free(j);
-    return set_err(HF_ERR_ARG, "hf_join_build",
-                   "right keys outside [key_min, key_min+n_slots) or a key "
-                   "exceeded the 4096-duplicate cap");
+    char buf[160];
+    snprintf(buf, sizeof buf,
+             "%llu right keys outside [key_min, key_min+n_slots); %llu keys "
+             "exceeded the 4096-duplicate cap",
+             h_err[0], h_err[1]);
+    return set_err(HF_ERR_ARG, "hf_join_build", buf);
   }
   *out = j;
   return HF_OK;
