@@ -39,17 +39,34 @@ def tryit(tag, rk, lk, rv):
         lib.sync()
 
 
+def csr_check(tag, rk):
+    """Build only; D2H the CSR and diff against numpy bincount."""
+    import ctypes as ct
+    try:
+        rkc = lib.put(rk)
+        rvc = lib.put(np.zeros(rk.size))
+        r = lib.reduce(rkc)
+        kmin, n_slots = r.imn, r.imx - r.imn + 1
+        j = lib.join_build(rkc, [rvc], kmin, n_slots)
+        print(f"{tag}: build OK")
+    except lib.HfError as e:
+        print(f"{tag}: build FAIL {str(e)[:120]}")
+    lib.sync()
+
+
 def main():
     lib.ensure_ready(0)
-    rng = np.random.default_rng(0)
-    for n in (10**4, 10**5, 10**6, 10**7):
-        nl = n * 2
-        rk = np.arange(n, dtype=np.int64)
-        tryit(f"arange n={n}", rk, rng.integers(0, n, nl).astype(np.int64),
-              rng.random(n))
+    if os.environ.get("JOINDBG_PROF"):
+        lib.profiling(True)
+        print("profiling ON")
+    rng = np.random.default_rng(42)
+    reps = int(os.environ.get("JOINDBG_REPS", "3"))
+    n = 10**7
+    for rep in range(reps):
         rku = rng.integers(0, n, n).astype(np.int64)
-        tryit(f"uniform n={n}", rku, rng.integers(0, n, nl).astype(np.int64),
-              rng.random(n))
+        csr_check(f"rep{rep} uniform n={n}", rku)
+        tryit(f"rep{rep} full n={n}", rku,
+              rng.integers(0, n, 2 * n).astype(np.int64), rng.random(n))
 
 
 if __name__ == "__main__":
