@@ -54,11 +54,51 @@ def csr_check(tag, rk):
     lib.sync()
 
 
+def replicate_joinbench(nl, nr):
+    """Exact joinbench flow: left frame first (k+lv), right (k+rv), qc.merge."""
+    import pandas
+    from modin_amd.core.dataframe import HipDataframe
+    from modin_amd.core.partition import DeviceBlock, HipDataframePartition
+    from modin_amd.query_compiler import HipQueryCompiler
+
+    lib.profiling(True)
+    rng = np.random.default_rng(42)
+    keyspace = nr
+
+    def frame(n, cols):
+        block = {"k": lib.put(rng.integers(0, keyspace, n).astype(np.int64))}
+        for name in cols:
+            block[name] = lib.put(rng.random(n))
+        names = list(block)
+        return HipQueryCompiler(HipDataframe(
+            [HipDataframePartition(DeviceBlock(block, n))],
+            pandas.RangeIndex(n), names, [n],
+            pandas.Series({"k": np.dtype(np.int64),
+                           **{c: np.dtype(np.float64) for c in cols}})))
+
+    qL = frame(nl, ["lv"])
+    qR = frame(nr, ["rv"])
+    try:
+        out = qL.merge(qR, on="k")
+        lib.sync()
+        print(f"replicate nl={nl} nr={nr}: OK out={len(out)}")
+    except lib.HfError as e:
+        print(f"replicate nl={nl} nr={nr}: FAIL {str(e)[:120]}")
+        # diff the histogram vs numpy to see whether the counts were wrong
+        rk = lib.get(qR._modin_frame._partitions[0].block().columns["k"])
+        cnt = np.bincount(rk, minlength=keyspace)
+        print("  numpy max multiplicity:", cnt.max())
+
+
 def main():
     lib.ensure_ready(0)
     if os.environ.get("JOINDBG_PROF"):
         lib.profiling(True)
         print("profiling ON")
+    mode = os.environ.get("JOINDBG_MODE", "replicate")
+    if mode == "replicate":
+        replicate_joinbench(2 * 10**7, 10**7)
+        return
     rng = np.random.default_rng(42)
     reps = int(os.environ.get("JOINDBG_REPS", "3"))
     n = 10**7
