@@ -1848,6 +1848,37 @@ int hf_join_build(const hf_col* rkeys, const hf_col* const* rvals, int nr,
                        0, g.stream, d_cnt, n_slots, d_tiles, d_total, d_csr);
   });
   if (rc != HF_OK) return rc;
+  if (getenv("HF_JOIN_DEBUG")) {
+    // host-side CSR verification (debug only): monotone, steps match the
+    // histogram, total == rows
+    std::vector<unsigned> h_cnt((size_t)n_slots);
+    std::vector<unsigned long long> h_csr((size_t)n_slots + 1);
+    hipMemcpyAsync(h_cnt.data(), d_cnt, n_slots * 4, hipMemcpyDeviceToHost,
+                   g.stream);
+    hipMemcpyAsync(h_csr.data(), d_csr, (n_slots + 1) * 8,
+                   hipMemcpyDeviceToHost, g.stream);
+    hipStreamSynchronize(g.stream);
+    unsigned long long run = 0;
+    int bad = 0;
+    for (int64_t s = 0; s < n_slots && bad < 5; ++s) {
+      if (h_csr[s] != run) {
+        fprintf(stderr,
+                "[join_debug] csr[%lld]=%llu expect %llu (cnt[s]=%u, "
+                "tile=%lld)\n",
+                (long long)s, h_csr[s], run, h_cnt[s], (long long)(s / 4096));
+        ++bad;
+      }
+      run += h_cnt[s];
+    }
+    if (h_csr[n_slots] != run)
+      fprintf(stderr, "[join_debug] csr[n]=%llu expect %llu\n", h_csr[n_slots],
+              run);
+    unsigned long long mx = 0;
+    for (int64_t s = 0; s < n_slots; ++s)
+      if (h_cnt[s] > mx) mx = h_cnt[s];
+    fprintf(stderr, "[join_debug] hist total=%llu rows=%lld max_mult=%llu "
+            "bad=%d\n", run, (long long)n, mx, bad);
+  }
   // reuse d_cnt as the fill cursor
   HF_HIP("hf_join_build", hipMemsetAsync(d_cnt, 0, n_slots * 4, g.stream));
   hf_join* j = new hf_join{};
