@@ -21,6 +21,7 @@
 #include <cstdint>
 #include <cstdio>
 #include <cstring>
+#include <cstdlib>
 #include <string>
 #include <type_traits>
 #include <unordered_map>
@@ -155,6 +156,25 @@ int64_t grid_for(int64_t work_items) {
   return b < GRID_CAP ? b : GRID_CAP;
 }
 
+}  // namespace
+
+
+// Env-gated allocator: HF_SYNC_ALLOC=1 swaps the stream-ordered pool for
+// plain hipMalloc/hipFree (debug: isolates pool-reuse bugs).
+namespace {
+inline bool sync_alloc() {
+  static int v = -1;
+  if (v < 0) v = getenv("HF_SYNC_ALLOC") ? 1 : 0;
+  return v == 1;
+}
+inline hipError_t dev_alloc(void** p, int64_t bytes, hipStream_t s) {
+  if (sync_alloc()) { hipStreamSynchronize(s); return hipMalloc(p, bytes); }
+  return hipMallocAsync(p, bytes, s);
+}
+inline hipError_t dev_free(void* p, hipStream_t s) {
+  if (sync_alloc()) { hipStreamSynchronize(s); return hipFree(p); }
+  return hipFreeAsync(p, s);
+}
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -1209,7 +1229,7 @@ int hf_col_alloc(int64_t len, int dtype, hf_col** out) {
   void* d = nullptr;
   int64_t bytes = len * dtype_size(dtype);
   if (bytes == 0) bytes = 8;  // keep zero-length columns addressable
-  HF_HIP("hf_col_alloc", hipMallocAsync(&d, bytes, g.stream));
+  HF_HIP("hf_col_alloc", dev_alloc((void**)&d, bytes, g.stream));
   hf_col* c = new hf_col{d, len, dtype, g.gpu};
   *out = c;
   return HF_OK;
@@ -1237,7 +1257,7 @@ int hf_get(const hf_col* col, void* host) {
 
 int hf_col_free(hf_col* col) {
   if (!col) return HF_OK;
-  if (g.inited && col->dptr) hipFreeAsync(col->dptr, g.stream);
+  if (g.inited && col->dptr) dev_free(col->dptr, g.stream);
   if (col->d_hist) free(col->d_hist);  // host-side cached histogram
   delete col;
   return HF_OK;
@@ -1250,13 +1270,13 @@ uintptr_t hf_col_dptr(const hf_col* c) { return c ? (uintptr_t)c->dptr : 0; }
 int hf_alloc_raw(int64_t bytes, uintptr_t* dptr) {
   HF_NEED_INIT("hf_alloc_raw");
   void* d = nullptr;
-  HF_HIP("hf_alloc_raw", hipMallocAsync(&d, bytes, g.stream));
+  HF_HIP("hf_alloc_raw", dev_alloc((void**)&d, bytes, g.stream));
   *dptr = (uintptr_t)d;
   return HF_OK;
 }
 int hf_free_raw(uintptr_t dptr) {
   HF_NEED_INIT("hf_free_raw");
-  HF_HIP("hf_free_raw", hipFreeAsync((void*)dptr, g.stream));
+  HF_HIP("hf_free_raw", dev_free((void*)dptr, g.stream));
   return HF_OK;
 }
 int hf_memset_raw(uintptr_t dptr, int value, int64_t bytes) {
@@ -1435,7 +1455,7 @@ int ensure_host_hist(hf_col* keys, int64_t key_min, int64_t n_slots,
     return HF_OK;
   if (keys->d_hist) { free(keys->d_hist); keys->d_hist = nullptr; }
   unsigned long long* d_h = nullptr;
-  HF_HIP("gb_hist", hipMallocAsync((void**)&d_h, nb * 8, g.stream));
+  HF_HIP("gb_hist", dev_alloc((void**)&d_h, nb * 8, g.stream));
   HF_HIP("gb_hist", hipMemsetAsync(d_h, 0, nb * 8, g.stream));
   const int64_t n = keys->len;
   int rc = timed_launch("gb_hist", [&] {
@@ -1448,7 +1468,7 @@ int ensure_host_hist(hf_col* keys, int64_t key_min, int64_t n_slots,
   HF_HIP("gb_hist", hipMemcpyAsync(h, d_h, nb * 8, hipMemcpyDeviceToHost,
                                    g.stream));
   HF_HIP("gb_hist", hipStreamSynchronize(g.stream));
-  HF_HIP("gb_hist", hipFreeAsync(d_h, g.stream));
+  HF_HIP("gb_hist", dev_free(d_h, g.stream));
   keys->d_hist = h;
   keys->hist_kmin = key_min;
   keys->hist_nb = nb;
@@ -1529,13 +1549,13 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
   GbWorkItem* d_work = nullptr;
   const int64_t alloc_rows = off > 0 ? off : 64;
   if (nvals > 0)
-    HF_HIP("gb_radix", hipMallocAsync((void**)&r0, alloc_rows * 8, g.stream));
+    HF_HIP("gb_radix", dev_alloc((void**)&r0, alloc_rows * 8, g.stream));
   if (nvals > 1)
-    HF_HIP("gb_radix", hipMallocAsync((void**)&r1, alloc_rows * 8, g.stream));
-  HF_HIP("gb_radix", hipMallocAsync((void**)&rk, alloc_rows * 2, g.stream));
-  HF_HIP("gb_radix", hipMallocAsync((void**)&d_cur, nb * 4, g.stream));
+    HF_HIP("gb_radix", dev_alloc((void**)&r1, alloc_rows * 8, g.stream));
+  HF_HIP("gb_radix", dev_alloc((void**)&rk, alloc_rows * 2, g.stream));
+  HF_HIP("gb_radix", dev_alloc((void**)&d_cur, nb * 4, g.stream));
   HF_HIP("gb_radix",
-         hipMallocAsync((void**)&d_work, work.size() * sizeof(GbWorkItem),
+         dev_alloc((void**)&d_work, work.size() * sizeof(GbWorkItem),
                         g.stream));
   HF_HIP("gb_radix", hipMemcpyAsync(d_cur, cur_init.data(), nb * 4,
                                     hipMemcpyHostToDevice, g.stream));
@@ -1603,11 +1623,11 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
       }
     }
   }
-  if (r0) hipFreeAsync(r0, g.stream);
-  if (r1) hipFreeAsync(r1, g.stream);
-  hipFreeAsync(rk, g.stream);
-  hipFreeAsync(d_cur, g.stream);
-  hipFreeAsync(d_work, g.stream);
+  if (r0) dev_free(r0, g.stream);
+  if (r1) dev_free(r1, g.stream);
+  dev_free(rk, g.stream);
+  dev_free(d_cur, g.stream);
+  dev_free(d_work, g.stream);
   return rc;
 }
 
@@ -1690,7 +1710,7 @@ int hf_groupby_compact(uintptr_t sums, uintptr_t rowcnt, uintptr_t counts,
   // tile counts
   int64_t* d_tiles = nullptr;
   HF_HIP("hf_groupby_compact",
-         hipMallocAsync((void**)&d_tiles, (ntiles + 1) * 8, g.stream));
+         dev_alloc((void**)&d_tiles, (ntiles + 1) * 8, g.stream));
   int64_t* d_total = (int64_t*)((char*)g.d_scratch + SCRATCH_NGROUPS);
   int rc = timed_launch("gb_compact_count", [&] {
     hipLaunchKernelGGL(k_compact_count, dim3((uint32_t)ntiles), dim3(BLOCK), 0,
@@ -1707,7 +1727,7 @@ int hf_groupby_compact(uintptr_t sums, uintptr_t rowcnt, uintptr_t counts,
          hipMemcpyAsync(&total, d_total, 8, hipMemcpyDeviceToHost, g.stream));
   HF_HIP("hf_groupby_compact", hipStreamSynchronize(g.stream));
   if (h_err != 0) {
-    hipFreeAsync(d_tiles, g.stream);
+    dev_free(d_tiles, g.stream);
     char buf[128];
     snprintf(buf, sizeof buf,
              "%llu keys fell outside [key_min, key_min+n_slots) during accumulate",
@@ -1734,10 +1754,10 @@ int hf_groupby_compact(uintptr_t sums, uintptr_t rowcnt, uintptr_t counts,
   double** d_sum_ptrs = nullptr;
   int64_t** d_cnt_ptrs = nullptr;
   HF_HIP("hf_groupby_compact",
-         hipMallocAsync((void**)&d_sum_ptrs, sizeof(double*) * (nvals ? nvals : 1),
+         dev_alloc((void**)&d_sum_ptrs, sizeof(double*) * (nvals ? nvals : 1),
                         g.stream));
   HF_HIP("hf_groupby_compact",
-         hipMallocAsync((void**)&d_cnt_ptrs, sizeof(int64_t*) * (nvals ? nvals : 1),
+         dev_alloc((void**)&d_cnt_ptrs, sizeof(int64_t*) * (nvals ? nvals : 1),
                         g.stream));
   HF_HIP("hf_groupby_compact",
          hipMemcpyAsync(d_sum_ptrs, h_sum_ptrs.data(), sizeof(double*) * nvals,
@@ -1761,9 +1781,9 @@ int hf_groupby_compact(uintptr_t sums, uintptr_t rowcnt, uintptr_t counts,
                          (const unsigned long long*)counts, nvals, key_min, n_slots,
                          d_tiles, (int64_t*)(*out_keys)->dptr, d_sum_ptrs, d_cnt_ptrs);
   });
-  hipFreeAsync(d_tiles, g.stream);
-  hipFreeAsync(d_sum_ptrs, g.stream);
-  hipFreeAsync(d_cnt_ptrs, g.stream);
+  dev_free(d_tiles, g.stream);
+  dev_free(d_sum_ptrs, g.stream);
+  dev_free(d_cnt_ptrs, g.stream);
   if (rc != HF_OK) return rc;
   *n_groups = total;
   return HF_OK;
@@ -1827,12 +1847,12 @@ int hf_join_build(const hf_col* rkeys, const hf_col* const* rvals, int nr,
   int64_t* d_tiles = nullptr;
   unsigned long long* d_csr = nullptr;
   const int64_t ntiles = (n_slots + JOIN_TILE - 1) / JOIN_TILE;
-  HF_HIP("hf_join_build", hipMallocAsync((void**)&d_cnt, n_slots * 4, g.stream));
+  HF_HIP("hf_join_build", dev_alloc((void**)&d_cnt, n_slots * 4, g.stream));
   HF_HIP("hf_join_build", hipMemsetAsync(d_cnt, 0, n_slots * 4, g.stream));
   HF_HIP("hf_join_build",
-         hipMallocAsync((void**)&d_tiles, ntiles * 8, g.stream));
+         dev_alloc((void**)&d_tiles, ntiles * 8, g.stream));
   HF_HIP("hf_join_build",
-         hipMallocAsync((void**)&d_csr, (n_slots + 1) * 8, g.stream));
+         dev_alloc((void**)&d_csr, (n_slots + 1) * 8, g.stream));
   int rc = timed_launch("join_hist", [&] {
     hipLaunchKernelGGL(k_hist_u32, dim3((uint32_t)grid_for(n)), dim3(BLOCK), 0,
                        g.stream, (const int64_t*)rkeys->dptr, n, key_min,
@@ -1890,11 +1910,11 @@ int hf_join_build(const hf_col* rkeys, const hf_col* const* rvals, int nr,
   for (int c = 0; c < nr; ++c) j->dtypes[c] = rvals[c]->dtype;
   const int64_t alloc_n = n > 0 ? n : 1;
   HF_HIP("hf_join_build",
-         hipMallocAsync((void**)&j->d_jidx, alloc_n * 4, g.stream));
+         dev_alloc((void**)&j->d_jidx, alloc_n * 4, g.stream));
   JoinPtrs jv{};
   for (int c = 0; c < nr; ++c) {
     HF_HIP("hf_join_build",
-           hipMallocAsync((void**)&j->d_jval[c], alloc_n * 8, g.stream));
+           dev_alloc((void**)&j->d_jval[c], alloc_n * 8, g.stream));
     jv.vals[c] = j->d_jval[c];
   }
   auto fill = [&](auto nrTag) {
@@ -1916,8 +1936,8 @@ int hf_join_build(const hf_col* rkeys, const hf_col* const* rvals, int nr,
     HF_JB_CASE(5) HF_JB_CASE(6) HF_JB_CASE(7) HF_JB_CASE(8)
 #undef HF_JB_CASE
   }
-  hipFreeAsync(d_cnt, g.stream);
-  hipFreeAsync(d_tiles, g.stream);
+  dev_free(d_cnt, g.stream);
+  dev_free(d_tiles, g.stream);
   if (rc != HF_OK) { hf_join_free(j); return rc; }
   // surface hist/fixup errors
   unsigned long long h_err[2] = {0, 0};
@@ -1946,10 +1966,10 @@ int hf_join_build(const hf_col* rkeys, const hf_col* const* rvals, int nr,
 int hf_join_free(hf_join* j) {
   if (!j) return HF_OK;
   if (g.inited) {
-    if (j->d_csr) hipFreeAsync(j->d_csr, g.stream);
-    if (j->d_jidx) hipFreeAsync(j->d_jidx, g.stream);
+    if (j->d_csr) dev_free(j->d_csr, g.stream);
+    if (j->d_jidx) dev_free(j->d_jidx, g.stream);
     for (int c = 0; c < j->nr; ++c)
-      if (j->d_jval[c]) hipFreeAsync(j->d_jval[c], g.stream);
+      if (j->d_jval[c]) dev_free(j->d_jval[c], g.stream);
   }
   delete j;
   return HF_OK;
@@ -1970,10 +1990,10 @@ int hf_join_probe(const hf_join* j, const hf_col* lkeys, hf_col** out_keys,
   int64_t* d_total = (int64_t*)((char*)g.d_scratch + SCRATCH_NGROUPS);
   const int64_t alloc_n = n > 0 ? n : 1;
   HF_HIP("hf_join_probe",
-         hipMallocAsync((void**)&d_offs, alloc_n * 8, g.stream));
+         dev_alloc((void**)&d_offs, alloc_n * 8, g.stream));
   HF_HIP("hf_join_probe",
-         hipMallocAsync((void**)&d_cnts, alloc_n * 4, g.stream));
-  HF_HIP("hf_join_probe", hipMallocAsync((void**)&d_tiles, ntiles * 8, g.stream));
+         dev_alloc((void**)&d_cnts, alloc_n * 4, g.stream));
+  HF_HIP("hf_join_probe", dev_alloc((void**)&d_tiles, ntiles * 8, g.stream));
   HF_HIP("hf_join_probe", hipMemsetAsync(d_tiles, 0, ntiles * 8, g.stream));
   int rc = HF_OK;
   if (n > 0) {
@@ -2021,9 +2041,9 @@ int hf_join_probe(const hf_join* j, const hf_col* lkeys, hf_col** out_keys,
 #undef HF_JP_CASE
     }
   }
-  hipFreeAsync(d_offs, g.stream);
-  hipFreeAsync(d_cnts, g.stream);
-  hipFreeAsync(d_tiles, g.stream);
+  dev_free(d_offs, g.stream);
+  dev_free(d_cnts, g.stream);
+  dev_free(d_tiles, g.stream);
   if (rc != HF_OK) return rc;
   *n_out = total;
   return HF_OK;
