@@ -159,21 +159,68 @@ int64_t grid_for(int64_t work_items) {
 }  // namespace
 
 
-// Env-gated allocator: HF_SYNC_ALLOC=1 swaps the stream-ordered pool for
-// plain hipMalloc/hipFree (debug: isolates pool-reuse bugs).
+// Device allocator: a size-bucketed cache over hipMalloc.  All compute runs
+// on ONE module stream, so reusing a cached block is ordered-correct by
+// construction (every prior user's work precedes the next user's on the
+// stream).  ROCm 7.2's hipMallocAsync default pool showed reuse corruption
+// under the join's varied alloc/free pattern (partial histograms + "write
+// access to a read-only page" faults, joinbench vs joindbg bisect) — this
+// cache replaces it.  Steady-state workloads allocate nothing.
+// HF_SYNC_ALLOC=1 bypasses the cache (plain hipMalloc/hipFree, debug).
 namespace {
 inline bool sync_alloc() {
   static int v = -1;
   if (v < 0) v = getenv("HF_SYNC_ALLOC") ? 1 : 0;
   return v == 1;
 }
+
+struct DevCache {
+  std::unordered_map<int64_t, std::vector<void*>> free_by_size;
+  std::unordered_map<void*, int64_t> size_of;
+  int64_t cached_bytes = 0;
+
+  hipError_t alloc(void** p, int64_t bytes) {
+    bytes = (bytes + 255) & ~255LL;
+    auto it = free_by_size.find(bytes);
+    if (it != free_by_size.end() && !it->second.empty()) {
+      *p = it->second.back();
+      it->second.pop_back();
+      cached_bytes -= bytes;
+      return hipSuccess;
+    }
+    hipError_t e = hipMalloc(p, bytes);
+    if (e == hipErrorOutOfMemory) {
+      trim();
+      e = hipMalloc(p, bytes);
+    }
+    if (e == hipSuccess) size_of[*p] = bytes;
+    return e;
+  }
+  void free(void* p) {
+    auto it = size_of.find(p);
+    if (it == size_of.end()) { hipFree(p); return; }
+    free_by_size[it->second].push_back(p);
+    cached_bytes += it->second;
+  }
+  void trim() {  // OOM fallback: release every cached block after draining
+                 // outstanding users
+    hipDeviceSynchronize();
+    for (auto& kv : free_by_size)
+      for (void* p : kv.second) { size_of.erase(p); hipFree(p); }
+    free_by_size.clear();
+    cached_bytes = 0;
+  }
+};
+DevCache g_cache;
+
 inline hipError_t dev_alloc(void** p, int64_t bytes, hipStream_t s) {
   if (sync_alloc()) { hipStreamSynchronize(s); return hipMalloc(p, bytes); }
-  return hipMallocAsync(p, bytes, s);
+  return g_cache.alloc(p, bytes);
 }
 inline hipError_t dev_free(void* p, hipStream_t s) {
   if (sync_alloc()) { hipStreamSynchronize(s); return hipFree(p); }
-  return hipFreeAsync(p, s);
+  g_cache.free(p);
+  return hipSuccess;
 }
 }  // namespace
 
@@ -1195,6 +1242,7 @@ int hf_init(int gpu) {
 int hf_shutdown(void) {
   if (!g.inited) return HF_OK;
   hipStreamSynchronize(g.stream);
+  g_cache.trim();
   for (auto& p : g.pending) { hipEventDestroy(p.a); hipEventDestroy(p.b); }
   g.pending.clear();
   g.stats.clear();
