@@ -92,9 +92,10 @@ def main():
     report("map_fillna (df.fillna(0))", "map_f64", ne, 16,
            lambda: HipQueryCompiler.fillna(qc, 0.0)._modin_frame._partitions[0]
            .drain_call_queue())
-    report("binary_add (df+df)", "bin_f64", ne, 24,
-           lambda: HipQueryCompiler.add(qc, qc))
-    del qc
+    qc2 = frame(args.map_rows, args.map_cols)  # distinct frame: honest 24 B/elem
+    report("binary_add (df+df2)", "bin_f64", ne, 24,
+           lambda: HipQueryCompiler.add(qc, qc2))
+    del qc, qc2
 
     # config 3: TreeReduce — 1e9 x 4 f64 (fresh columns each op; reduce
     # results are cached per immutable column, so bypass the cache by

@@ -501,10 +501,12 @@ int main(int argc, char** argv) {
                            lds, 0, keys, v0, n, n_slots, nb, d_cur, r0, rk);
       }, 26.0 * n);
     };
-    scat_lds3(std::integral_constant<int, 16>{}, std::integral_constant<int, 256>{});
-    scat_lds3(std::integral_constant<int, 8>{}, std::integral_constant<int, 512>{});
+    scat_lds3(std::integral_constant<int, 10>{}, std::integral_constant<int, 512>{});
+    scat_lds3(std::integral_constant<int, 12>{}, std::integral_constant<int, 512>{});
+    scat_lds3(std::integral_constant<int, 14>{}, std::integral_constant<int, 512>{});
     scat_lds3(std::integral_constant<int, 16>{}, std::integral_constant<int, 512>{});
-    scat_lds3(std::integral_constant<int, 32>{}, std::integral_constant<int, 256>{});
+    scat_lds3(std::integral_constant<int, 20>{}, std::integral_constant<int, 512>{});
+    scat_lds3(std::integral_constant<int, 24>{}, std::integral_constant<int, 256>{});
     // aggregate variants (consume whatever the last scatter left; perf-only)
     auto agg = [&](auto vecTag, auto blkTag) {
       constexpr int VEC = decltype(vecTag)::value;
