@@ -744,17 +744,22 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
 }
 
 // P2: one work item per (bucket, chunk); one value column per launch.
-// ROWCNT: also merge per-slot row counts (first column / keys-only pass).
+// ROWCNT: also mark group presence (first column / keys-only pass).  The
+// global rowcnt table is PRESENCE-ONLY everywhere (compaction tests >0;
+// per-column counts live in the counts table), so the per-row mark is a
+// plain LDS byte store (last-writer-wins, LDS is CU-local) — no atomic.
 // CNT: also merge per-slot non-NaN counts for this column.
+// 2 rows/lane vectorized (regions are 64-row aligned; odd chunk tails are
+// only ever the last chunk of a bucket).
 template <bool ROWCNT, bool CNT, bool HAVE_VAL>
 __global__ void __launch_bounds__(512) k_gb_bucket_agg(
     const double* __restrict__ vals, const unsigned short* __restrict__ lowkeys,
     const GbWorkItem* __restrict__ work, int64_t n_slots,
     double* __restrict__ gsums, unsigned long long* __restrict__ growcnt,
     unsigned long long* __restrict__ gcounts) {
-  __shared__ double lsums[GB_RANGE];            // 64 KB
-  __shared__ unsigned ltouch[GB_RANGE];         // 32 KB (rowcnt / touch marks)
-  __shared__ unsigned lcnt[CNT ? GB_RANGE : 1]; // 32 KB when counting
+  __shared__ double lsums[HAVE_VAL ? GB_RANGE : 1];        // 32 KB
+  __shared__ unsigned lcnt[CNT ? GB_RANGE : 1];            // 16 KB when CNT
+  __shared__ unsigned char ltouch[GB_RANGE];               // 4 KB
   const GbWorkItem w = work[blockIdx.x];
   for (int s = threadIdx.x; s < GB_RANGE; s += blockDim.x) {
     if (HAVE_VAL) lsums[s] = 0.0;
@@ -763,11 +768,30 @@ __global__ void __launch_bounds__(512) k_gb_bucket_agg(
   }
   __syncthreads();
   const int64_t end = w.start + w.len;
-  for (int64_t i = w.start + threadIdx.x; i < end; i += blockDim.x) {
-    const int slot = lowkeys[i];
-    atomicAdd(&ltouch[slot], 1u);
+  const int64_t npair = (int64_t)w.len >> 1;
+  const ushort2* k2 = reinterpret_cast<const ushort2*>(lowkeys + w.start);
+  const double2* v2 = reinterpret_cast<const double2*>(vals + w.start);
+  for (int64_t i = threadIdx.x; i < npair; i += blockDim.x) {
+    const ushort2 kk = k2[i];
+    ltouch[kk.x] = 1;
+    ltouch[kk.y] = 1;
     if (HAVE_VAL) {
-      const double v = vals[i];
+      const double2 vv = v2[i];
+      if (vv.x == vv.x) {
+        unsafeAtomicAdd(&lsums[kk.x], vv.x);
+        if (CNT) atomicAdd(&lcnt[kk.x], 1u);
+      }
+      if (vv.y == vv.y) {
+        unsafeAtomicAdd(&lsums[kk.y], vv.y);
+        if (CNT) atomicAdd(&lcnt[kk.y], 1u);
+      }
+    }
+  }
+  if ((w.len & 1) && threadIdx.x == 0) {
+    const int slot = lowkeys[end - 1];
+    ltouch[slot] = 1;
+    if (HAVE_VAL) {
+      const double v = vals[end - 1];
       if (v == v) {
         unsafeAtomicAdd(&lsums[slot], v);
         if (CNT) atomicAdd(&lcnt[slot], 1u);
@@ -777,10 +801,9 @@ __global__ void __launch_bounds__(512) k_gb_bucket_agg(
   __syncthreads();
   const int64_t gbase = (int64_t)w.bucket << GB_RANGE_LOG;
   for (int s = threadIdx.x; s < GB_RANGE; s += blockDim.x) {
-    const unsigned t = ltouch[s];
-    if (!t || gbase + s >= n_slots) continue;
+    if (!ltouch[s] || gbase + s >= n_slots) continue;
     if (HAVE_VAL) unsafeAtomicAdd(&gsums[gbase + s], lsums[s]);
-    if (ROWCNT) atomicAdd(&growcnt[gbase + s], (unsigned long long)t);
+    if (ROWCNT) atomicAdd(&growcnt[gbase + s], 1ULL);
     if (CNT) atomicAdd(&gcounts[gbase + s], (unsigned long long)lcnt[s]);
   }
 }
@@ -795,8 +818,10 @@ __global__ void __launch_bounds__(512) k_gb_dense(
     unsigned long long* __restrict__ err) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   double* lsums = reinterpret_cast<double*>(smem_raw);       // [n_slots]
-  unsigned* ltouch = reinterpret_cast<unsigned*>(lsums + (HAVE_VAL ? n_slots : 0));
-  unsigned* lcnt = ltouch + n_slots;                         // when CNT
+  unsigned* lcnt = reinterpret_cast<unsigned*>(
+      lsums + (HAVE_VAL ? n_slots : 0));                     // when CNT
+  unsigned char* ltouch = reinterpret_cast<unsigned char*>(
+      lcnt + (CNT ? n_slots : 0));                           // presence bytes
   for (int64_t s = threadIdx.x; s < n_slots; s += blockDim.x) {
     if (HAVE_VAL) lsums[s] = 0.0;
     ltouch[s] = 0;
@@ -811,7 +836,7 @@ __global__ void __launch_bounds__(512) k_gb_dense(
       atomicAdd(err, 1ULL);
       continue;
     }
-    atomicAdd(&ltouch[k], 1u);
+    ltouch[k] = 1;  // presence-only (plain LDS byte store, CU-local)
     if (HAVE_VAL) {
       const double v = vals[i];
       if (v == v) {
@@ -822,10 +847,9 @@ __global__ void __launch_bounds__(512) k_gb_dense(
   }
   __syncthreads();
   for (int64_t s = threadIdx.x; s < n_slots; s += blockDim.x) {
-    const unsigned t = ltouch[s];
-    if (!t) continue;
+    if (!ltouch[s]) continue;
     if (HAVE_VAL) unsafeAtomicAdd(&gsums[s], lsums[s]);
-    if (ROWCNT) atomicAdd(&growcnt[s], (unsigned long long)t);
+    if (ROWCNT) atomicAdd(&growcnt[s], 1ULL);
     if (CNT) atomicAdd(&gcounts[s], (unsigned long long)lcnt[s]);
   }
 }
@@ -1536,7 +1560,7 @@ int gb_dense_path(const hf_col* keys, const GbPtrs& ptrs, int nvals,
     constexpr bool R = decltype(rTag)::value, C = decltype(cTag)::value,
                    V = decltype(vTag)::value;
     const uint32_t lds =
-        (uint32_t)(n_slots * ((V ? 8 : 0) + 4 + (C ? 4 : 0)));
+        (uint32_t)(n_slots * ((V ? 8 : 0) + 1 + (C ? 4 : 0)));
     return timed_launch("gb_dense", [&] {
       hipLaunchKernelGGL((k_gb_dense<R, C, V>), dim3((uint32_t)grid), dim3(512),
                          lds, g.stream, (const int64_t*)keys->dptr, v, n,
@@ -1612,12 +1636,12 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
                         hipMemcpyHostToDevice, g.stream));
   // host vectors must outlive the async H2D of pageable memory
   HF_HIP("gb_radix", hipStreamSynchronize(g.stream));
-  // P1 scatter: 512-thread blocks; 8192-row tiles (RPT=16) for <=1 value
-  // column, 4096-row tiles (RPT=8) for 2 (LDS staging budget)
+  // P1 scatter: 6144-row tiles (256x24) for <=1 value column — 74 KB LDS,
+  // 2 blocks/CU (probe winner); 3072-row tiles for 2 columns
   auto scat = [&](auto nvTag, auto rptTag) {
     constexpr int NVv = decltype(nvTag)::value;
     constexpr int RPTv = decltype(rptTag)::value;
-    constexpr int BLKv = 512;
+    constexpr int BLKv = 256;
     const int64_t tile_sz = (int64_t)BLKv * RPTv;
     const int64_t ntiles = ((n & ~1LL) + tile_sz - 1) / tile_sz;
     const uint32_t sgrid =
@@ -1632,11 +1656,11 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
     });
   };
   rc = nvals == 0 ? scat(std::integral_constant<int, 0>{},
-                         std::integral_constant<int, 16>{})
+                         std::integral_constant<int, 24>{})
        : nvals == 1 ? scat(std::integral_constant<int, 1>{},
-                           std::integral_constant<int, 16>{})
+                           std::integral_constant<int, 24>{})
                     : scat(std::integral_constant<int, 2>{},
-                           std::integral_constant<int, 8>{});
+                           std::integral_constant<int, 12>{});
   if (rc != HF_OK) return rc;
   // P2 aggregate, one column per launch
   const bool cnt = counts != 0;
