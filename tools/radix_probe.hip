@@ -313,6 +313,41 @@ __global__ void __launch_bounds__(BLOCK) k_scat_lds(
 
 // ---------------- aggregate variants ----------------
 
+// presence-byte agg (production shape after the rowcnt->presence change)
+template <int RL, int BLK>
+__global__ void __launch_bounds__(BLK) k_agg_u8(
+    const double* __restrict__ vals, const unsigned short* __restrict__ lowkeys,
+    const Work* __restrict__ work, int64_t n_slots, double* __restrict__ gsums,
+    unsigned long long* __restrict__ growcnt) {
+  constexpr int RANGE = 1 << RL;
+  __shared__ double lsums[RANGE];
+  __shared__ unsigned char ltouch[RANGE];
+  const Work w = work[blockIdx.x];
+  for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
+    lsums[s] = 0.0;
+    ltouch[s] = 0;
+  }
+  __syncthreads();
+  const int64_t npair = (int64_t)(w.len) >> 1;
+  const ushort2* k2 = reinterpret_cast<const ushort2*>(lowkeys + w.start);
+  const double2* v2 = reinterpret_cast<const double2*>(vals + w.start);
+  for (int64_t i = threadIdx.x; i < npair; i += blockDim.x) {
+    const ushort2 kk = k2[i];
+    const double2 vv = v2[i];
+    ltouch[kk.x] = 1;
+    ltouch[kk.y] = 1;
+    if (vv.x == vv.x) unsafeAtomicAdd(&lsums[kk.x], vv.x);
+    if (vv.y == vv.y) unsafeAtomicAdd(&lsums[kk.y], vv.y);
+  }
+  __syncthreads();
+  const int64_t gbase = (int64_t)w.bucket << RL;
+  for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
+    if (!ltouch[s] || gbase + s >= n_slots) continue;
+    unsafeAtomicAdd(&gsums[gbase + s], lsums[s]);
+    atomicAdd(&growcnt[gbase + s], 1ULL);
+  }
+}
+
 template <int RL, int VEC, int BLK = 512>
 __global__ void __launch_bounds__(BLK) k_agg(
     const double* __restrict__ vals, const unsigned short* __restrict__ lowkeys,
@@ -520,10 +555,20 @@ int main(int argc, char** argv) {
                            growcnt);
       }, 10.0 * n);
     };
-    agg(std::integral_constant<int, 1>{}, std::integral_constant<int, 512>{});
+    auto agg8 = [&](auto blkTag) {
+      constexpr int BLK = decltype(blkTag)::value;
+      snprintf(nm, sizeof nm, "agg_u8 BLK=%d RL=%d (%zu wi)", BLK, RL,
+               work.size());
+      auto nop = [] {};
+      run(nm, 3, nop, [&] {
+        hipLaunchKernelGGL((k_agg_u8<RL, BLK>), dim3((uint32_t)work.size()),
+                           dim3(BLK), 0, 0, r0, rk, d_work, n_slots, gsums,
+                           growcnt);
+      }, 10.0 * n);
+    };
+    agg8(std::integral_constant<int, 512>{});
+    agg8(std::integral_constant<int, 256>{});
     agg(std::integral_constant<int, 2>{}, std::integral_constant<int, 512>{});
-    agg(std::integral_constant<int, 2>{}, std::integral_constant<int, 256>{});
-    agg(std::integral_constant<int, 1>{}, std::integral_constant<int, 1024>{});
     CHECK(hipFree(d_work));
   };
   sweep(std::integral_constant<int, 13>{});
