@@ -311,6 +311,124 @@ __global__ void __launch_bounds__(BLOCK) k_scat_lds(
   }
 }
 
+// B4: lds3 + (a) scan by wave0 WHILE other waves reserve cursors, (b)
+// paired writeout (2 entries/lane: b128 LDS val reads, fused stores when the
+// pair shares a bucket).
+template <int RPT, int RL, int BLK>
+__global__ void __launch_bounds__(BLK) k_scat_lds4(
+    const int64_t* __restrict__ keys, const double* __restrict__ v0, int64_t n,
+    int64_t n_slots, int nb, unsigned* __restrict__ cursors,
+    double* __restrict__ r0, unsigned short* __restrict__ rk) {
+  constexpr int TILE = BLK * RPT;
+  constexpr int PAIRS = RPT / 2;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  double* sval = reinterpret_cast<double*>(smem_raw);
+  unsigned* skey = reinterpret_cast<unsigned*>(sval + TILE);
+  unsigned* it_cnt = skey + TILE;
+  unsigned* it_off = it_cnt + nb;
+  unsigned* it_gbase = it_off + nb;
+  unsigned* s_total = it_gbase + nb;
+  const int64_t ntiles = (n + TILE - 1) / TILE;
+  const int64_t npair_total = n >> 1;
+  for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+    const int64_t t0 = tile * TILE;
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) it_cnt[t] = 0;
+    int lb[RPT];
+    unsigned lk[RPT];
+    unsigned lr[RPT];
+    double lv[RPT];
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < PAIRS; ++j) {
+      const int64_t pr = (t0 >> 1) + (int64_t)j * BLK + threadIdx.x;
+      const int a = 2 * j, b = 2 * j + 1;
+      lb[a] = lb[b] = -1;
+      if (pr < npair_total) {
+        const longlong2 kk = reinterpret_cast<const longlong2*>(keys)[pr];
+        const double2 vv = reinterpret_cast<const double2*>(v0)[pr];
+        lb[a] = (int)(kk.x >> RL);
+        lk[a] = (unsigned)(kk.x & ((1 << RL) - 1));
+        lv[a] = vv.x;
+        lb[b] = (int)(kk.y >> RL);
+        lk[b] = (unsigned)(kk.y & ((1 << RL) - 1));
+        lv[b] = vv.y;
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < RPT; ++j)
+      if (lb[j] >= 0) lr[j] = atomicAdd(&it_cnt[lb[j]], 1u);
+    __syncthreads();
+    // wave 0 scans; the OTHER waves reserve global cursors concurrently
+    if (threadIdx.x < 64) {
+      const int lane = threadIdx.x;
+      unsigned carry = 0;
+      for (int base = 0; base < nb; base += 64) {
+        const int t = base + lane;
+        unsigned v = (t < nb) ? it_cnt[t] : 0;
+        unsigned incl = v;
+#pragma unroll
+        for (int d = 1; d < 64; d <<= 1) {
+          unsigned up = __shfl_up(incl, d);
+          if (lane >= d) incl += up;
+        }
+        if (t < nb) it_off[t] = carry + incl - v;
+        carry += __shfl(incl, 63);
+      }
+      if (lane == 0) *s_total = carry;
+    } else {
+      for (int t = (int)threadIdx.x - 64; t < nb; t += blockDim.x - 64) {
+        const unsigned c = it_cnt[t];
+        if (c) it_gbase[t] = atomicAdd(&cursors[t], c);
+      }
+    }
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < RPT; ++j) {
+      if (lb[j] >= 0) {
+        const unsigned p = it_off[lb[j]] + lr[j];
+        sval[p] = lv[j];
+        skey[p] = ((unsigned)lb[j] << 16) | lk[j];
+      }
+    }
+    __syncthreads();
+    // paired writeout
+    const int staged = (int)*s_total;
+    const int half = (staged + 1) >> 1;
+    for (int q = threadIdx.x; q < half; q += blockDim.x) {
+      const int p = 2 * q;
+      const unsigned ka = skey[p];
+      const unsigned ba = ka >> 16;
+      const int64_t pa = (int64_t)it_gbase[ba] + (p - it_off[ba]);
+      if (p + 1 < staged) {
+        const unsigned kb2 = skey[p + 1];
+        const unsigned bb = kb2 >> 16;
+        const double2 vv = *reinterpret_cast<const double2*>(&sval[p]);
+        if (bb == ba) {
+          r0[pa] = vv.x;
+          r0[pa + 1] = vv.y;
+          const unsigned pack = (ka & 0xFFFF) | ((kb2 & 0xFFFF) << 16);
+          if ((pa & 1) == 0)
+            *reinterpret_cast<unsigned*>(&rk[pa]) = pack;
+          else {
+            rk[pa] = (unsigned short)(ka & 0xFFFF);
+            rk[pa + 1] = (unsigned short)(kb2 & 0xFFFF);
+          }
+        } else {
+          const int64_t pb = (int64_t)it_gbase[bb] + (p + 1 - it_off[bb]);
+          r0[pa] = vv.x;
+          r0[pb] = vv.y;
+          rk[pa] = (unsigned short)(ka & 0xFFFF);
+          rk[pb] = (unsigned short)(kb2 & 0xFFFF);
+        }
+      } else {
+        r0[pa] = sval[p];
+        rk[pa] = (unsigned short)(ka & 0xFFFF);
+      }
+    }
+    __syncthreads();
+  }
+}
+
 // ---------------- aggregate variants ----------------
 
 // presence-byte agg (production shape after the rowcnt->presence change)
@@ -536,12 +654,22 @@ int main(int argc, char** argv) {
                            lds, 0, keys, v0, n, n_slots, nb, d_cur, r0, rk);
       }, 26.0 * n);
     };
-    scat_lds3(std::integral_constant<int, 10>{}, std::integral_constant<int, 512>{});
-    scat_lds3(std::integral_constant<int, 12>{}, std::integral_constant<int, 512>{});
-    scat_lds3(std::integral_constant<int, 14>{}, std::integral_constant<int, 512>{});
-    scat_lds3(std::integral_constant<int, 16>{}, std::integral_constant<int, 512>{});
-    scat_lds3(std::integral_constant<int, 20>{}, std::integral_constant<int, 512>{});
     scat_lds3(std::integral_constant<int, 24>{}, std::integral_constant<int, 256>{});
+    auto scat_lds4 = [&](auto rptTag, auto blkTag) {
+      constexpr int RPT = decltype(rptTag)::value;
+      constexpr int BLK = decltype(blkTag)::value;
+      snprintf(nm, sizeof nm, "scat_lds4 RPT=%d BLK=%d RL=%d", RPT, BLK, RL);
+      const int64_t tile_sz = (int64_t)BLK * RPT;
+      const int64_t ntiles = (n + tile_sz - 1) / tile_sz;
+      const uint32_t grid = (uint32_t)std::min<int64_t>(ntiles, 2048);
+      const uint32_t lds = tile_sz * 12 + nb * 16 + 16;
+      run(nm, 2, reset_cur, [&] {
+        hipLaunchKernelGGL((k_scat_lds4<RPT, RL, BLK>), dim3(grid), dim3(BLK),
+                           lds, 0, keys, v0, n, n_slots, nb, d_cur, r0, rk);
+      }, 26.0 * n);
+    };
+    scat_lds4(std::integral_constant<int, 24>{}, std::integral_constant<int, 256>{});
+    scat_lds4(std::integral_constant<int, 16>{}, std::integral_constant<int, 512>{});
     // aggregate variants (consume whatever the last scatter left; perf-only)
     auto agg = [&](auto vecTag, auto blkTag) {
       constexpr int VEC = decltype(vecTag)::value;
