@@ -201,6 +201,30 @@ int hf_gather(const hf_col* col, const hf_col* idx, hf_col** out);
  * dataframe.py:2918). */
 int hf_col_concat(const hf_col* const* cols, int ncols, hf_col** out);
 
+/* ---- Compare + Filter (df[df.v > x], dropna) ----
+ * Compare is a Map-shaped elementwise kernel producing an int64 0/1 mask
+ * (query_compiler comparison bindings feed Binary.register in the
+ * reference; the mask column is the device form of the boolean Series).
+ * Filter is the device form of `PandasDataframe.filter`/`mask`
+ * (partition.py:224): a plan (per-tile kept-row offsets from one mask scan)
+ * applied per column as a ballot-ranked compaction, preserving row order;
+ * hf_filter_iota materializes the kept original positions (the pandas
+ * result index). */
+enum {
+  HF_CMP_GT = 0, HF_CMP_GE = 1, HF_CMP_LT = 2, HF_CMP_LE = 3,
+  HF_CMP_EQ = 4, HF_CMP_NE = 5
+};
+int hf_compare_scalar(int op, const hf_col* col, double scalar,
+                      hf_col** out);  /* int64 0/1 mask; NaN compares false
+                                         except NE (pandas semantics) */
+
+typedef struct hf_filterplan hf_filterplan;
+int hf_filter_plan(const hf_col* mask, hf_filterplan** out, int64_t* n_kept);
+int hf_filter_apply(const hf_filterplan* plan, const hf_col* col,
+                    hf_col** out);
+int hf_filter_iota(const hf_filterplan* plan, int64_t base, hf_col** out);
+int hf_filter_plan_free(hf_filterplan* plan);
+
 /* ---- profiling (bench.py roofline leg) ----
  * When enabled, every kernel launch is bracketed by HIP events on the module
  * stream; hf_kernel_stats returns the accumulated count and total ms for the

@@ -187,6 +187,46 @@ class HipDataframe:
         return HipDataframe(out_parts, pandas.RangeIndex(total), out_columns,
                             lengths, pandas.Series(dtypes))
 
+    # ---- row filter (PandasDataframe.filter / mask device form,
+    #      partition.py:224) ----
+    def filter_rows(self, mask_frame: "HipDataframe") -> "HipDataframe":
+        """Keep rows where the 1-column int64 mask is nonzero.  Result index
+        = kept original positions (pandas boolean-mask semantics over the
+        RangeIndex these frames carry)."""
+        if self._row_lengths != mask_frame._row_lengths:
+            raise lib.HfError("filter: mask is not co-partitioned with frame")
+        if not isinstance(self._index, pandas.RangeIndex) or \
+                self._index.start != 0 or self._index.step != 1:
+            raise lib.HfError(
+                "filter: only RangeIndex frames this round (groupby results "
+                "etc. need index gather — later round)")
+        out_parts, lengths, idx_cols = [], [], []
+        base = 0
+        for p, mp, length in zip(self._partitions, mask_frame._partitions,
+                                 self._row_lengths):
+            mblock = mp.block()
+            (mask_col,) = mblock.columns.values()
+            plan = lib.filter_plan(mask_col)
+            block = p.block()
+            cols = {name: lib.filter_apply(plan, col)
+                    for name, col in block.columns.items()}
+            idx_cols.append(lib.filter_iota(plan, base))
+            out_parts.append(HipDataframePartition(DeviceBlock(cols,
+                                                               plan.n_kept)))
+            lengths.append(plan.n_kept)
+            base += length
+        idx_col = idx_cols[0] if len(idx_cols) == 1 else lib.concat(idx_cols)
+        return HipDataframe(out_parts, DeviceIndex(idx_col, name=None),
+                            self.columns, lengths, self.dtypes)
+
+    # ---- comparison map (mask column) ----
+    def compare_scalar(self, op_code: int, scalar) -> "HipDataframe":
+        def block_fn(block: DeviceBlock) -> DeviceBlock:
+            out = {name: lib.compare_scalar(op_code, col, scalar)
+                   for name, col in block.columns.items()}
+            return DeviceBlock(out, block.length)
+        return self.map(block_fn)
+
     # ---- column selection (getitem_column_array device form) ----
     def take_columns(self, names) -> "HipDataframe":
         def sel(block: DeviceBlock) -> DeviceBlock:

@@ -123,6 +123,16 @@ def load() -> ct.CDLL:
                                          ct.POINTER(ct.c_int64)]),
             "hf_gather": (ct.c_int, [ct.c_void_p, ct.c_void_p,
                                      ct.POINTER(ct.c_void_p)]),
+            "hf_compare_scalar": (ct.c_int, [ct.c_int, ct.c_void_p,
+                                             ct.c_double,
+                                             ct.POINTER(ct.c_void_p)]),
+            "hf_filter_plan": (ct.c_int, [ct.c_void_p, ct.POINTER(ct.c_void_p),
+                                          ct.POINTER(ct.c_int64)]),
+            "hf_filter_apply": (ct.c_int, [ct.c_void_p, ct.c_void_p,
+                                           ct.POINTER(ct.c_void_p)]),
+            "hf_filter_iota": (ct.c_int, [ct.c_void_p, ct.c_int64,
+                                          ct.POINTER(ct.c_void_p)]),
+            "hf_filter_plan_free": (ct.c_int, [ct.c_void_p]),
             "hf_profiling": (ct.c_int, [ct.c_int]),
             "hf_kernel_stats": (ct.c_int, [ct.c_char_p, ct.POINTER(ct.c_int64),
                                            ct.POINTER(ct.c_double)]),
@@ -145,7 +155,9 @@ def exported_symbols():
         "hf_memset_raw", "hf_map_scalar", "hf_map_scalar_i64", "hf_binary",
         "hf_reduce", "hf_groupby_accum", "hf_groupby_compact",
         "hf_col_concat", "hf_join_build", "hf_join_free", "hf_join_probe",
-        "hf_gather", "hf_profiling", "hf_kernel_stats", "hf_kernel_stats_reset",
+        "hf_gather", "hf_compare_scalar", "hf_filter_plan", "hf_filter_apply",
+        "hf_filter_iota", "hf_filter_plan_free", "hf_profiling",
+        "hf_kernel_stats", "hf_kernel_stats_reset",
     ]
 
 
@@ -393,6 +405,62 @@ def join_probe(j: JoinRef, lkeys: ColumnRef):
     rcols = [_wrap(ct.c_void_p(out_r[c]), n, j.rdtypes[c]) for c in range(j.nr)]
     return (_wrap(out_keys, n, HF_INT64), _wrap(out_lidx, n, HF_INT64),
             rcols, n)
+
+
+# compare ops (include/hipframe.h)
+CMP_GT, CMP_GE, CMP_LT, CMP_LE, CMP_EQ, CMP_NE = range(6)
+
+
+def compare_scalar(op: int, col: ColumnRef, scalar: float) -> ColumnRef:
+    ensure_ready()
+    out = ct.c_void_p()
+    _check(load().hf_compare_scalar(op, col.handle, float(scalar),
+                                    ct.byref(out)), "hf_compare_scalar")
+    return _wrap(out, col.length, HF_INT64)
+
+
+class FilterPlan:
+    """Owner of an hf_filterplan; keeps the mask column alive (the plan
+    borrows its device pointer)."""
+
+    __slots__ = ("handle", "mask", "n_kept")
+
+    def __init__(self, handle, mask, n_kept):
+        self.handle = handle
+        self.mask = mask
+        self.n_kept = n_kept
+
+    def __del__(self):
+        try:
+            if _dll is not None and _inited_gpu is not None and self.handle:
+                _dll.hf_filter_plan_free(self.handle)
+        except Exception:
+            pass
+
+
+def filter_plan(mask: ColumnRef) -> FilterPlan:
+    ensure_ready()
+    out = ct.c_void_p()
+    n = ct.c_int64(0)
+    _check(load().hf_filter_plan(mask.handle, ct.byref(out), ct.byref(n)),
+           "hf_filter_plan")
+    return FilterPlan(out, mask, n.value)
+
+
+def filter_apply(plan: FilterPlan, col: ColumnRef) -> ColumnRef:
+    ensure_ready()
+    out = ct.c_void_p()
+    _check(load().hf_filter_apply(plan.handle, col.handle, ct.byref(out)),
+           "hf_filter_apply")
+    return _wrap(out, plan.n_kept, col.dtype_code)
+
+
+def filter_iota(plan: FilterPlan, base: int) -> ColumnRef:
+    ensure_ready()
+    out = ct.c_void_p()
+    _check(load().hf_filter_iota(plan.handle, base, ct.byref(out)),
+           "hf_filter_iota")
+    return _wrap(out, plan.n_kept, HF_INT64)
 
 
 def gather(col: ColumnRef, idx: ColumnRef) -> ColumnRef:

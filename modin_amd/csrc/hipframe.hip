@@ -1178,6 +1178,88 @@ __global__ void __launch_bounds__(BLOCK) k_probe_emit(
   }
 }
 
+// ---- compare + filter kernels ----
+
+template <int OP, typename T>
+__device__ __forceinline__ long long cmp1(T x, double s) {
+  switch (OP) {
+    case HF_CMP_GT: return x > s;
+    case HF_CMP_GE: return x >= s;
+    case HF_CMP_LT: return x < s;
+    case HF_CMP_LE: return x <= s;
+    case HF_CMP_EQ: return x == s;
+    case HF_CMP_NE: return !(x == s);  // NaN != s -> true (pandas)
+  }
+  return 0;
+}
+
+template <int OP, typename T>
+__global__ void __launch_bounds__(BLOCK) k_compare(const T* __restrict__ in,
+                                                   long long* __restrict__ out,
+                                                   double s, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = cmp1<OP, T>(in[i], s);
+}
+
+constexpr int FILT_TILE = 4096;
+
+__global__ void __launch_bounds__(BLOCK) k_filter_count(
+    const long long* __restrict__ mask, int64_t n, int64_t* __restrict__ sums) {
+  const int64_t t0 = (int64_t)blockIdx.x * FILT_TILE;
+  const int64_t t1 = min(t0 + (int64_t)FILT_TILE, n);
+  long long local = 0;
+  for (int64_t i = t0 + threadIdx.x; i < t1; i += blockDim.x)
+    local += (mask[i] != 0);
+  for (int off = 32; off > 0; off >>= 1) local += __shfl_down(local, off);
+  __shared__ long long sc[BLOCK / 64];
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) sc[wave] = local;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    long long tot = 0;
+    for (int w = 0; w < BLOCK / 64; ++w) tot += sc[w];
+    sums[blockIdx.x] = tot;
+  }
+}
+
+// order-preserving compaction: ballot-ranked within tile (the same pattern
+// as k_compact_scatter).  IOTA: emit base + row index instead of data.
+template <typename T, bool IOTA>
+__global__ void __launch_bounds__(BLOCK) k_filter_scatter(
+    const long long* __restrict__ mask, const T* __restrict__ in, int64_t n,
+    const int64_t* __restrict__ tile_bases, int64_t iota_base,
+    T* __restrict__ out) {
+  const int64_t t0 = (int64_t)blockIdx.x * FILT_TILE;
+  const int64_t t1 = min(t0 + (int64_t)FILT_TILE, n);
+  __shared__ int64_t s_base;
+  __shared__ int s_wave_cnt[BLOCK / 64];
+  if (threadIdx.x == 0) s_base = tile_bases[blockIdx.x];
+  __syncthreads();
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  for (int64_t chunk = t0; chunk < t1; chunk += blockDim.x) {
+    const int64_t i = chunk + threadIdx.x;
+    const bool pred = (i < t1) && (mask[i] != 0);
+    const uint64_t ballot = __ballot(pred);
+    if (lane == 0) s_wave_cnt[wave] = __popcll(ballot);
+    __syncthreads();
+    int64_t wave_base = 0;
+    for (int w = 0; w < wave; ++w) wave_base += s_wave_cnt[w];
+    if (pred) {
+      const int64_t pos = s_base + wave_base +
+          __popcll(ballot & ((lane == 63) ? ~0ULL >> 1 : ((1ULL << lane) - 1)));
+      out[pos] = IOTA ? (T)(iota_base + i) : in[i];
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      int64_t tot = 0;
+      for (int w = 0; w < BLOCK / 64; ++w) tot += s_wave_cnt[w];
+      s_base += tot;
+    }
+    __syncthreads();
+  }
+}
+
 __global__ void __launch_bounds__(BLOCK) k_gather_f64(
     const double* __restrict__ src, const int64_t* __restrict__ idx,
     double* __restrict__ out, int64_t n) {
@@ -2139,6 +2221,129 @@ int hf_gather(const hf_col* col, const hf_col* idx, hf_col** out) {
       hipLaunchKernelGGL(k_gather_i64, dim3((uint32_t)grid_for(n)), dim3(BLOCK),
                          0, g.stream, (const int64_t*)col->dptr,
                          (const int64_t*)idx->dptr, (int64_t*)(*out)->dptr, n);
+  });
+  if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  return rc;
+}
+
+// ---- compare + filter ----
+
+struct hf_filterplan {
+  const long long* mask;   // borrowed: caller keeps the mask column alive
+  int64_t n, total, ntiles;
+  int64_t* d_tiles;        // exclusive per-tile kept offsets
+};
+
+int hf_compare_scalar(int op, const hf_col* col, double scalar, hf_col** out) {
+  HF_NEED_INIT("hf_compare_scalar");
+  if (!col || !out) return set_err(HF_ERR_ARG, "hf_compare_scalar", "null");
+  if (op < HF_CMP_GT || op > HF_CMP_NE)
+    return set_err(HF_ERR_ARG, "hf_compare_scalar", "unknown op");
+  int rc = hf_col_alloc(col->len, HF_INT64, out);
+  if (rc != HF_OK) return rc;
+  const int64_t n = col->len;
+  auto L = [&](auto opTag, auto tTag) {
+    constexpr int O = decltype(opTag)::value;
+    using T = typename decltype(tTag)::type;
+    return timed_launch("compare", [&] {
+      hipLaunchKernelGGL((k_compare<O, T>), dim3((uint32_t)grid_for(n)),
+                         dim3(BLOCK), 0, g.stream, (const T*)col->dptr,
+                         (long long*)(*out)->dptr, scalar, n);
+    });
+  };
+  struct F64 { using type = double; };
+  struct I64 { using type = long long; };
+  const bool f = col->dtype == HF_FLOAT64;
+  switch (op) {
+#define HF_CMP_CASE(O)                                                         \
+  case O:                                                                      \
+    rc = f ? L(std::integral_constant<int, O>{}, F64{})                        \
+           : L(std::integral_constant<int, O>{}, I64{});                       \
+    break;
+    HF_CMP_CASE(HF_CMP_GT) HF_CMP_CASE(HF_CMP_GE) HF_CMP_CASE(HF_CMP_LT)
+    HF_CMP_CASE(HF_CMP_LE) HF_CMP_CASE(HF_CMP_EQ) HF_CMP_CASE(HF_CMP_NE)
+#undef HF_CMP_CASE
+  }
+  if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  return rc;
+}
+
+int hf_filter_plan(const hf_col* mask, hf_filterplan** out, int64_t* n_kept) {
+  HF_NEED_INIT("hf_filter_plan");
+  if (!mask || !out || !n_kept)
+    return set_err(HF_ERR_ARG, "hf_filter_plan", "null");
+  if (mask->dtype != HF_INT64)
+    return set_err(HF_ERR_ARG, "hf_filter_plan", "mask must be int64 0/1");
+  const int64_t n = mask->len;
+  const int64_t ntiles = n > 0 ? (n + FILT_TILE - 1) / FILT_TILE : 1;
+  hf_filterplan* p = new hf_filterplan{};
+  p->mask = (const long long*)mask->dptr;
+  p->n = n;
+  p->ntiles = ntiles;
+  HF_HIP("hf_filter_plan",
+         dev_alloc((void**)&p->d_tiles, ntiles * 8, g.stream));
+  HF_HIP("hf_filter_plan", hipMemsetAsync(p->d_tiles, 0, ntiles * 8, g.stream));
+  int64_t* d_total = (int64_t*)((char*)g.d_scratch + SCRATCH_NGROUPS);
+  int rc = HF_OK;
+  if (n > 0) {
+    rc = timed_launch("filter_count", [&] {
+      hipLaunchKernelGGL(k_filter_count, dim3((uint32_t)ntiles), dim3(BLOCK), 0,
+                         g.stream, p->mask, n, p->d_tiles);
+    });
+    if (rc != HF_OK) { hf_filter_plan_free(p); return rc; }
+  }
+  hipLaunchKernelGGL(k_compact_scan, dim3(1), dim3(1024), 0, g.stream,
+                     p->d_tiles, ntiles, d_total);
+  HF_HIP("hf_filter_plan",
+         hipMemcpyAsync(&p->total, d_total, 8, hipMemcpyDeviceToHost, g.stream));
+  HF_HIP("hf_filter_plan", hipStreamSynchronize(g.stream));
+  *n_kept = p->total;
+  *out = p;
+  return HF_OK;
+}
+
+int hf_filter_plan_free(hf_filterplan* p) {
+  if (!p) return HF_OK;
+  if (g.inited && p->d_tiles) dev_free(p->d_tiles, g.stream);
+  delete p;
+  return HF_OK;
+}
+
+int hf_filter_apply(const hf_filterplan* p, const hf_col* col, hf_col** out) {
+  HF_NEED_INIT("hf_filter_apply");
+  if (!p || !col || !out) return set_err(HF_ERR_ARG, "hf_filter_apply", "null");
+  if (col->len != p->n)
+    return set_err(HF_ERR_ARG, "hf_filter_apply", "length mismatch");
+  int rc = hf_col_alloc(p->total, col->dtype, out);
+  if (rc != HF_OK) return rc;
+  if (p->n == 0 || p->total == 0) return HF_OK;
+  rc = timed_launch("filter_scatter", [&] {
+    if (col->dtype == HF_FLOAT64)
+      hipLaunchKernelGGL((k_filter_scatter<double, false>),
+                         dim3((uint32_t)p->ntiles), dim3(BLOCK), 0, g.stream,
+                         p->mask, (const double*)col->dptr, p->n, p->d_tiles,
+                         (int64_t)0, (double*)(*out)->dptr);
+    else
+      hipLaunchKernelGGL((k_filter_scatter<long long, false>),
+                         dim3((uint32_t)p->ntiles), dim3(BLOCK), 0, g.stream,
+                         p->mask, (const long long*)col->dptr, p->n, p->d_tiles,
+                         (int64_t)0, (long long*)(*out)->dptr);
+  });
+  if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  return rc;
+}
+
+int hf_filter_iota(const hf_filterplan* p, int64_t base, hf_col** out) {
+  HF_NEED_INIT("hf_filter_iota");
+  if (!p || !out) return set_err(HF_ERR_ARG, "hf_filter_iota", "null");
+  int rc = hf_col_alloc(p->total, HF_INT64, out);
+  if (rc != HF_OK) return rc;
+  if (p->n == 0 || p->total == 0) return HF_OK;
+  rc = timed_launch("filter_iota", [&] {
+    hipLaunchKernelGGL((k_filter_scatter<long long, true>),
+                       dim3((uint32_t)p->ntiles), dim3(BLOCK), 0, g.stream,
+                       p->mask, (const long long*)nullptr, p->n, p->d_tiles,
+                       base, (long long*)(*out)->dptr);
   });
   if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
   return rc;

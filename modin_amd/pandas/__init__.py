@@ -93,6 +93,31 @@ class _HipPandasBase:
         return self._rewrap(type(self._query_compiler).fillna(self._query_compiler,
                                                               float(value)))
 
+    # ---- comparisons -> boolean mask (int64 0/1 device column; to_pandas
+    #      lowers to bool dtype) ----
+    def _cmp(self, name, other):
+        out = self._rewrap(getattr(self._query_compiler, name)(other))
+        out._bool_mask = True
+        return out
+
+    def __gt__(self, other):
+        return self._cmp("gt", other)
+
+    def __ge__(self, other):
+        return self._cmp("ge", other)
+
+    def __lt__(self, other):
+        return self._cmp("lt", other)
+
+    def __le__(self, other):
+        return self._cmp("le", other)
+
+    def __eq__(self, other):  # noqa: A003 — pandas-style elementwise eq
+        return self._cmp("eq", other)
+
+    def __ne__(self, other):  # noqa: A003
+        return self._cmp("ne", other)
+
     def abs(self):
         return self._rewrap(type(self._query_compiler).abs(self._query_compiler,
                                                            None))
@@ -152,11 +177,16 @@ class DataFrame(_HipPandasBase):
         if isinstance(key, str):
             return Series(query_compiler=self._query_compiler.getitem_column_array(
                 [key]), name=key)
+        if isinstance(key, Series):  # boolean row mask: df[df.v > x]
+            return DataFrame(
+                query_compiler=self._query_compiler.getitem_array(
+                    key._query_compiler)
+            )
         if isinstance(key, (list, tuple, pandas.Index)):
             return DataFrame(
                 query_compiler=self._query_compiler.getitem_column_array(list(key))
             )
-        raise lib.HfError("only column selection is supported")
+        raise lib.HfError("only column selection / boolean masks are supported")
 
     def merge(self, other: "DataFrame", on: str, how: str = "inner"):
         """Inner merge on an int64 key column (modin/pandas API ->
@@ -180,6 +210,8 @@ class DataFrame(_HipPandasBase):
 
 
 class Series(_HipPandasBase):
+    _bool_mask = False  # comparisons set this: to_pandas lowers int64->bool
+
     def __init__(self, data=None, query_compiler=None, name=None):
         self.name = name
         if query_compiler is not None:
@@ -206,6 +238,8 @@ class Series(_HipPandasBase):
         df = self._query_compiler.to_pandas()
         s = df[df.columns[0]]
         s.name = self.name
+        if self._bool_mask:
+            s = s.astype(bool)
         return s
 
     def _to_pandas(self) -> pandas.Series:

@@ -89,6 +89,36 @@ class HipQueryCompiler:
             )
         return fn(self, by)
 
+    # ---- comparisons (query_compiler gt/lt/eq bindings) -> int64 0/1 mask
+    def _compare(self, op_code, other):
+        if not np.isscalar(other):
+            raise lib.HfError("comparisons support scalars this round")
+        return self.__constructor__(
+            self._modin_frame.compare_scalar(op_code, float(other)))
+
+    def gt(self, other):
+        return self._compare(lib.CMP_GT, other)
+
+    def ge(self, other):
+        return self._compare(lib.CMP_GE, other)
+
+    def lt(self, other):
+        return self._compare(lib.CMP_LT, other)
+
+    def le(self, other):
+        return self._compare(lib.CMP_LE, other)
+
+    def eq(self, other):
+        return self._compare(lib.CMP_EQ, other)
+
+    def ne(self, other):
+        return self._compare(lib.CMP_NE, other)
+
+    # ---- boolean row mask (qc.getitem_array device form) ----
+    def getitem_array(self, mask_qc: "HipQueryCompiler") -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.filter_rows(mask_qc._modin_frame))
+
     # ---- merge (query_compiler merge -> MergeImpl.row_axis_merge,
     #      storage_formats/pandas/merge.py:104) ----
     def merge(self, right: "HipQueryCompiler", on: str,

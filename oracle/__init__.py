@@ -17,6 +17,8 @@ path against this oracle and the same vectors.
 
 from .ops import (  # noqa: F401
     binary_op,
+    compare_op,
+    filter_rows,
     groupby_agg,
     inner_join,
     map_op,

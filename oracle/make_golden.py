@@ -222,6 +222,41 @@ def gen_merge_cases(mpd, rng):
     return cases
 
 
+def gen_filter_cases(mpd, rng):
+    import pandas
+    cases = {}
+    n = 5000
+    v = rng.random(n)
+    v[rng.random(n) < 0.1] = np.nan
+    w = rng.standard_normal(n)
+    i = rng.integers(0, 10, n).astype(np.int64)
+    mdf = mpd.DataFrame({"v": v, "w": w, "i": i})
+    pdf = pandas.DataFrame({"v": v, "w": w, "i": i})
+    arrays = {"in_v": v, "in_w": w, "in_i": i}
+    for tag, mmask, pmask in [
+        ("gt", mdf["v"] > 0.25, pdf["v"] > 0.25),
+        ("le", mdf["v"] <= 0.5, pdf["v"] <= 0.5),
+        ("eq", mdf["i"] == 3, pdf["i"] == 3),
+        ("ne", mdf["v"] != 0.0, pdf["v"] != 0.0),  # NaN != 0 -> kept
+        ("none", mdf["v"] > 2.0, pdf["v"] > 2.0),  # empty result
+    ]:
+        m_np = mmask._to_pandas().to_numpy()
+        p_np = pmask.to_numpy()
+        np.testing.assert_array_equal(m_np, p_np)
+        arrays[f"out_mask_{tag}"] = p_np
+        mres = mdf[mmask]._to_pandas()
+        pres = pdf[pmask]
+        np.testing.assert_array_equal(mres.index.to_numpy(),
+                                      pres.index.to_numpy())
+        arrays[f"out_idx_{tag}"] = pres.index.to_numpy().astype(np.int64)
+        for c in ("v", "w", "i"):
+            np.testing.assert_array_equal(mres[c].to_numpy(),
+                                          pres[c].to_numpy())
+            arrays[f"out_{tag}_{c}"] = pres[c].to_numpy()
+    cases["flt_basic"] = arrays
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -231,6 +266,7 @@ def main():
     all_cases.update(gen_reduce_cases(mpd, rng))
     all_cases.update(gen_map_binary_cases(mpd, rng))
     all_cases.update(gen_merge_cases(mpd, rng))
+    all_cases.update(gen_filter_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)

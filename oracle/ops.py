@@ -167,6 +167,35 @@ def groupby_agg(keys: np.ndarray, vals: dict, agg: str):
 
 
 # ---------------------------------------------------------------------------
+# Compare + boolean filter — query_compiler comparison bindings and
+# `df[mask]` (`PandasDataframe.filter`/`mask`, partition.py:224).
+# pandas/numpy agree: NaN compares False for every op except != (True).
+# ---------------------------------------------------------------------------
+
+def compare_op(op: str, x: np.ndarray, s) -> np.ndarray:
+    if op == "gt":
+        return x > s
+    if op == "ge":
+        return x >= s
+    if op == "lt":
+        return x < s
+    if op == "le":
+        return x <= s
+    if op == "eq":
+        return x == s
+    if op == "ne":
+        return x != s
+    raise ValueError(op)
+
+
+def filter_rows(mask: np.ndarray, cols: dict):
+    """Returns (kept_positions, filtered cols) preserving row order."""
+    keep = np.asarray(mask).astype(bool)
+    pos = np.nonzero(keep)[0].astype(np.int64)
+    return pos, {n: np.asarray(v)[keep] for n, v in cols.items()}
+
+
+# ---------------------------------------------------------------------------
 # Merge — MergeImpl.row_axis_merge (storage_formats/pandas/merge.py:104-178):
 # materialize the right frame once (combine(), dataframe.py:2918), broadcast
 # to every left partition, per-partition pandas.merge(how="inner").

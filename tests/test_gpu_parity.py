@@ -208,6 +208,42 @@ def test_merge_error_surfaces():
         ok.merge(right, on="k", how="left")
 
 
+def test_filter_vs_golden(npartitions):
+    g = load_golden("flt_basic")
+    df = mpd.DataFrame({"v": g["in_v"], "w": g["in_w"], "i": g["in_i"]})
+    for tag, mask in [
+        ("gt", df["v"] > 0.25), ("le", df["v"] <= 0.5),
+        ("eq", df["i"] == 3), ("ne", df["v"] != 0.0),
+        ("none", df["v"] > 2.0),
+    ]:
+        m = mask.to_pandas()
+        assert m.dtype == bool
+        np.testing.assert_array_equal(m.to_numpy(), g[f"out_mask_{tag}"],
+                                      err_msg=tag)
+        out = df[mask].to_pandas()
+        np.testing.assert_array_equal(out.index.to_numpy(),
+                                      g[f"out_idx_{tag}"], err_msg=tag)
+        for c in ("v", "w", "i"):
+            np.testing.assert_array_equal(out[c].to_numpy(),
+                                          g[f"out_{tag}_{c}"],
+                                          err_msg=f"{tag}/{c}")
+            assert out[c].dtype == g[f"out_{tag}_{c}"].dtype
+
+
+def test_filter_then_groupby(npartitions):
+    """Composition: df[df.v > .5].groupby('k').sum() against the oracle."""
+    rng = np.random.default_rng(33)
+    n = 300_000
+    k = rng.integers(0, 500, n).astype(np.int64)
+    v = rng.random(n)
+    df = mpd.DataFrame({"k": k, "v": v})
+    out = df[df["v"] > 0.5].groupby("k").sum().to_pandas()
+    keep = v > 0.5
+    ok, osums = oracle.groupby_agg(k[keep], {"v": v[keep]}, "sum")
+    np.testing.assert_array_equal(out.index.to_numpy(), ok)
+    np.testing.assert_allclose(out["v"].to_numpy(), osums["v"], rtol=RTOL)
+
+
 def test_native_extension_is_loaded():
     """Guard against a silent eager/pandas fallback: the in-tree .so must be
     mapped into this process."""

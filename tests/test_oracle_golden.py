@@ -122,6 +122,25 @@ def test_merge_vs_golden(case):
         np.testing.assert_array_equal(out[c], g[f"out_{c}"], err_msg=c)
 
 
+def test_filter_vs_golden():
+    g = load_golden("flt_basic")
+    cols = {"v": g["in_v"], "w": g["in_w"], "i": g["in_i"]}
+    masks = {
+        "gt": oracle.compare_op("gt", g["in_v"], 0.25),
+        "le": oracle.compare_op("le", g["in_v"], 0.5),
+        "eq": oracle.compare_op("eq", g["in_i"], 3),
+        "ne": oracle.compare_op("ne", g["in_v"], 0.0),
+        "none": oracle.compare_op("gt", g["in_v"], 2.0),
+    }
+    for tag, mask in masks.items():
+        np.testing.assert_array_equal(mask, g[f"out_mask_{tag}"], err_msg=tag)
+        pos, out = oracle.filter_rows(mask, cols)
+        np.testing.assert_array_equal(pos, g[f"out_idx_{tag}"], err_msg=tag)
+        for c in cols:
+            np.testing.assert_array_equal(out[c], g[f"out_{tag}_{c}"],
+                                          err_msg=f"{tag}/{c}")
+
+
 def test_groupby_empty():
     keys, out = oracle.groupby_agg(np.empty(0, np.int64), {"v": np.empty(0)}, "sum")
     assert keys.size == 0 and out["v"].size == 0
