@@ -429,6 +429,102 @@ __global__ void __launch_bounds__(BLK) k_scat_lds4(
   }
 }
 
+// B5: lds3 with SPLIT loads — keys load + ranks first, value loads issued
+// AFTER ranks (latency hides under scan/reserve/barriers; plain loads are
+// not drained by __syncthreads).
+template <int RPT, int RL, int BLK>
+__global__ void __launch_bounds__(BLK) k_scat_lds5(
+    const int64_t* __restrict__ keys, const double* __restrict__ v0, int64_t n,
+    int64_t n_slots, int nb, unsigned* __restrict__ cursors,
+    double* __restrict__ r0, unsigned short* __restrict__ rk) {
+  constexpr int TILE = BLK * RPT;
+  constexpr int PAIRS = RPT / 2;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  double* sval = reinterpret_cast<double*>(smem_raw);
+  unsigned* skey = reinterpret_cast<unsigned*>(sval + TILE);
+  unsigned* it_cnt = skey + TILE;
+  unsigned* it_off = it_cnt + nb;
+  unsigned* it_gbase = it_off + nb;
+  unsigned* s_total = it_gbase + nb;
+  const int64_t ntiles = (n + TILE - 1) / TILE;
+  const int64_t npair_total = n >> 1;
+  for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+    const int64_t t0 = tile * TILE;
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) it_cnt[t] = 0;
+    int lb[RPT];
+    unsigned lk[RPT];
+    unsigned lr[RPT];
+    double lv[RPT];
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < PAIRS; ++j) {
+      const int64_t pr = (t0 >> 1) + (int64_t)j * BLK + threadIdx.x;
+      const int a = 2 * j, b = 2 * j + 1;
+      lb[a] = lb[b] = -1;
+      if (pr < npair_total) {
+        const longlong2 kk = reinterpret_cast<const longlong2*>(keys)[pr];
+        lb[a] = (int)(kk.x >> RL);
+        lk[a] = (unsigned)(kk.x & ((1 << RL) - 1));
+        lb[b] = (int)(kk.y >> RL);
+        lk[b] = (unsigned)(kk.y & ((1 << RL) - 1));
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < RPT; ++j)
+      if (lb[j] >= 0) lr[j] = atomicAdd(&it_cnt[lb[j]], 1u);
+    // value loads issued NOW: latency hides under scan + reserve + barriers
+#pragma unroll
+    for (int j = 0; j < PAIRS; ++j) {
+      const int64_t pr = (t0 >> 1) + (int64_t)j * BLK + threadIdx.x;
+      if (pr < npair_total) {
+        const double2 vv = reinterpret_cast<const double2*>(v0)[pr];
+        lv[2 * j] = vv.x;
+        lv[2 * j + 1] = vv.y;
+      }
+    }
+    __syncthreads();
+    if (threadIdx.x < 64) {
+      const int lane = threadIdx.x;
+      unsigned carry = 0;
+      for (int base = 0; base < nb; base += 64) {
+        const int t = base + lane;
+        unsigned v = (t < nb) ? it_cnt[t] : 0;
+        unsigned incl = v;
+#pragma unroll
+        for (int d = 1; d < 64; d <<= 1) {
+          unsigned up = __shfl_up(incl, d);
+          if (lane >= d) incl += up;
+        }
+        if (t < nb) it_off[t] = carry + incl - v;
+        carry += __shfl(incl, 63);
+      }
+      if (lane == 0) *s_total = carry;
+    }
+    __syncthreads();
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) {
+      const unsigned c = it_cnt[t];
+      if (c) it_gbase[t] = atomicAdd(&cursors[t], c);
+    }
+#pragma unroll
+    for (int j = 0; j < RPT; ++j) {
+      if (lb[j] >= 0) {
+        const unsigned p = it_off[lb[j]] + lr[j];
+        sval[p] = lv[j];
+        skey[p] = ((unsigned)lb[j] << 16) | lk[j];
+      }
+    }
+    __syncthreads();
+    const int staged = (int)*s_total;
+    for (int p = threadIdx.x; p < staged; p += blockDim.x) {
+      const unsigned b = skey[p] >> 16;
+      const int64_t pos = (int64_t)it_gbase[b] + (p - it_off[b]);
+      r0[pos] = sval[p];
+      rk[pos] = (unsigned short)(skey[p] & 0xFFFF);
+    }
+    __syncthreads();
+  }
+}
+
 // ---------------- aggregate variants ----------------
 
 // presence-byte agg (production shape after the rowcnt->presence change)
@@ -668,8 +764,23 @@ int main(int argc, char** argv) {
                            lds, 0, keys, v0, n, n_slots, nb, d_cur, r0, rk);
       }, 26.0 * n);
     };
-    scat_lds4(std::integral_constant<int, 24>{}, std::integral_constant<int, 256>{});
-    scat_lds4(std::integral_constant<int, 16>{}, std::integral_constant<int, 512>{});
+    auto scat_lds5 = [&](auto rptTag, auto blkTag) {
+      constexpr int RPT = decltype(rptTag)::value;
+      constexpr int BLK = decltype(blkTag)::value;
+      snprintf(nm, sizeof nm, "scat_lds5 RPT=%d BLK=%d RL=%d", RPT, BLK, RL);
+      const int64_t tile_sz = (int64_t)BLK * RPT;
+      const int64_t ntiles = (n + tile_sz - 1) / tile_sz;
+      const uint32_t grid = (uint32_t)std::min<int64_t>(ntiles, 2048);
+      const uint32_t lds = tile_sz * 12 + nb * 16 + 16;
+      run(nm, 2, reset_cur, [&] {
+        hipLaunchKernelGGL((k_scat_lds5<RPT, RL, BLK>), dim3(grid), dim3(BLK),
+                           lds, 0, keys, v0, n, n_slots, nb, d_cur, r0, rk);
+      }, 26.0 * n);
+    };
+    scat_lds5(std::integral_constant<int, 24>{}, std::integral_constant<int, 256>{});
+    scat_lds5(std::integral_constant<int, 16>{}, std::integral_constant<int, 256>{});
+    scat_lds5(std::integral_constant<int, 16>{}, std::integral_constant<int, 512>{});
+    scat_lds5(std::integral_constant<int, 32>{}, std::integral_constant<int, 256>{});
     // aggregate variants (consume whatever the last scatter left; perf-only)
     auto agg = [&](auto vecTag, auto blkTag) {
       constexpr int VEC = decltype(vecTag)::value;
