@@ -212,7 +212,8 @@ int hf_col_concat(const hf_col* const* cols, int ncols, hf_col** out);
  * result index). */
 enum {
   HF_CMP_GT = 0, HF_CMP_GE = 1, HF_CMP_LT = 2, HF_CMP_LE = 3,
-  HF_CMP_EQ = 4, HF_CMP_NE = 5
+  HF_CMP_EQ = 4, HF_CMP_NE = 5,
+  HF_CMP_NOTNA = 6  /* x == x (dropna/notna; scalar ignored) */
 };
 int hf_compare_scalar(int op, const hf_col* col, double scalar,
                       hf_col** out);  /* int64 0/1 mask; NaN compares false

@@ -227,6 +227,16 @@ class HipDataframe:
             return DeviceBlock(out, block.length)
         return self.map(block_fn)
 
+    # ---- dropna mask: AND of per-column notna (pandas dropna(how="any")) ----
+    def notna_all_mask(self) -> "HipDataframe":
+        def block_fn(block: DeviceBlock) -> DeviceBlock:
+            acc = None
+            for col in block.columns.values():
+                m = lib.compare_scalar(lib.CMP_NOTNA, col, 0.0)
+                acc = m if acc is None else lib.binary(lib.BIN_MUL, acc, m)
+            return DeviceBlock({"mask": acc}, block.length)
+        return self.map(block_fn)
+
     # ---- column selection (getitem_column_array device form) ----
     def take_columns(self, names) -> "HipDataframe":
         def sel(block: DeviceBlock) -> DeviceBlock:

@@ -271,6 +271,8 @@ def alloc(length: int, dtype_code: int) -> ColumnRef:
 def map_scalar(op: int, col: ColumnRef, scalar) -> ColumnRef:
     ensure_ready()
     out = ct.c_void_p()
+    if col.dtype_code == HF_INT64 and op in (MAP_DIV, MAP_RDIV, MAP_FILLNA):
+        col = cast_f64(col)  # pandas promotes int div to float; fillna no-ops
     if col.dtype_code == HF_INT64 and op != MAP_CAST_F64:
         _check(load().hf_map_scalar_i64(op, col.handle, int(scalar or 0),
                                         ct.byref(out)), "hf_map_scalar_i64")
@@ -408,7 +410,7 @@ def join_probe(j: JoinRef, lkeys: ColumnRef):
 
 
 # compare ops (include/hipframe.h)
-CMP_GT, CMP_GE, CMP_LT, CMP_LE, CMP_EQ, CMP_NE = range(6)
+CMP_GT, CMP_GE, CMP_LT, CMP_LE, CMP_EQ, CMP_NE, CMP_NOTNA = range(7)
 
 
 def compare_scalar(op: int, col: ColumnRef, scalar: float) -> ColumnRef:

@@ -1189,6 +1189,7 @@ __device__ __forceinline__ long long cmp1(T x, double s) {
     case HF_CMP_LE: return x <= s;
     case HF_CMP_EQ: return x == s;
     case HF_CMP_NE: return !(x == s);  // NaN != s -> true (pandas)
+    case HF_CMP_NOTNA: return x == x;  // int64: always 1
   }
   return 0;
 }
@@ -2237,7 +2238,7 @@ struct hf_filterplan {
 int hf_compare_scalar(int op, const hf_col* col, double scalar, hf_col** out) {
   HF_NEED_INIT("hf_compare_scalar");
   if (!col || !out) return set_err(HF_ERR_ARG, "hf_compare_scalar", "null");
-  if (op < HF_CMP_GT || op > HF_CMP_NE)
+  if (op < HF_CMP_GT || op > HF_CMP_NOTNA)
     return set_err(HF_ERR_ARG, "hf_compare_scalar", "unknown op");
   int rc = hf_col_alloc(col->len, HF_INT64, out);
   if (rc != HF_OK) return rc;
@@ -2262,6 +2263,7 @@ int hf_compare_scalar(int op, const hf_col* col, double scalar, hf_col** out) {
     break;
     HF_CMP_CASE(HF_CMP_GT) HF_CMP_CASE(HF_CMP_GE) HF_CMP_CASE(HF_CMP_LT)
     HF_CMP_CASE(HF_CMP_LE) HF_CMP_CASE(HF_CMP_EQ) HF_CMP_CASE(HF_CMP_NE)
+    HF_CMP_CASE(HF_CMP_NOTNA)
 #undef HF_CMP_CASE
   }
   if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }

@@ -118,6 +118,40 @@ class _HipPandasBase:
     def __ne__(self, other):  # noqa: A003
         return self._cmp("ne", other)
 
+    __hash__ = None  # elementwise __eq__ makes these unhashable, like pandas
+
+    def notna(self):
+        out = self._rewrap(self._query_compiler.notna())
+        out._bool_mask = True
+        return out
+
+    def isna(self):
+        return ~self.notna()
+
+    # ---- boolean mask algebra (int64 0/1 masks) ----
+    def __and__(self, other):
+        out = self._rewrap(type(self._query_compiler).mul(
+            self._query_compiler, _unwrap(other)))
+        out._bool_mask = True
+        return out
+
+    def __or__(self, other):
+        s = type(self._query_compiler).add(self._query_compiler, _unwrap(other))
+        out = self._rewrap(s.ge(1))
+        out._bool_mask = True
+        return out
+
+    def __invert__(self):
+        out = self._rewrap(type(self._query_compiler).rsub(
+            self._query_compiler, 1))
+        out._bool_mask = True
+        return out
+
+    def dropna(self):
+        """pandas dropna(how='any', axis=0): keep rows with no NaN."""
+        mask_qc = self._query_compiler.dropna_mask()
+        return self._rewrap(self._query_compiler.getitem_array(mask_qc))
+
     def abs(self):
         return self._rewrap(type(self._query_compiler).abs(self._query_compiler,
                                                            None))

@@ -114,6 +114,12 @@ class HipQueryCompiler:
     def ne(self, other):
         return self._compare(lib.CMP_NE, other)
 
+    def notna(self):
+        return self._compare(lib.CMP_NOTNA, 0.0)
+
+    def dropna_mask(self) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.notna_all_mask())
+
     # ---- boolean row mask (qc.getitem_array device form) ----
     def getitem_array(self, mask_qc: "HipQueryCompiler") -> "HipQueryCompiler":
         return self.__constructor__(

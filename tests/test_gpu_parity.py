@@ -244,6 +244,55 @@ def test_filter_then_groupby(npartitions):
     np.testing.assert_allclose(out["v"].to_numpy(), osums["v"], rtol=RTOL)
 
 
+def test_int_div_promotes():
+    rng = np.random.default_rng(40)
+    i = rng.integers(1, 100, 1000).astype(np.int64)
+    df = mpd.DataFrame({"i": i})
+    out = (df / 4).to_pandas()
+    np.testing.assert_array_equal(out["i"].to_numpy(), i / 4)
+    assert out["i"].dtype == np.float64
+
+
+@pytest.mark.parametrize("span", [8191, 8192, 8193, 10001, 4096 * 3 + 1])
+def test_groupby_dense_radix_boundary(span):
+    """Exercise both sides of the dense/radix switch and bucket-edge keys
+    (range not a multiple of GB_RANGE; negative key_min)."""
+    rng = np.random.default_rng(41)
+    n = 400_000
+    k = (rng.integers(0, span, n) - span // 2).astype(np.int64)
+    # pin the exact range ends so n_slots == span
+    k[0], k[1] = -(span // 2), span - 1 - span // 2
+    v = rng.random(n)
+    df = mpd.DataFrame({"k": k, "v": v})
+    out = df.groupby("k").sum().to_pandas()
+    ok, osums = oracle.groupby_agg(k, {"v": v}, "sum")
+    np.testing.assert_array_equal(out.index.to_numpy(), ok)
+    np.testing.assert_allclose(out["v"].to_numpy(), osums["v"], rtol=RTOL,
+                               atol=1e-9)
+
+
+def test_dropna_and_mask_algebra(npartitions):
+    rng = np.random.default_rng(42)
+    n = 20_000
+    v = rng.random(n)
+    v[rng.random(n) < 0.15] = np.nan
+    w = rng.random(n)
+    k = rng.integers(0, 7, n).astype(np.int64)
+    df = mpd.DataFrame({"k": k, "v": v, "w": w})
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w})
+    pandas.testing.assert_frame_equal(df.dropna().to_pandas(), pdf.dropna())
+    m = (df["v"] > 0.3) & (df["w"] <= 0.9)
+    pm = (pdf["v"] > 0.3) & (pdf["w"] <= 0.9)
+    np.testing.assert_array_equal(m.to_pandas().to_numpy(), pm.to_numpy())
+    m2 = (df["v"] > 0.5) | ~(df["w"] > 0.2)
+    pm2 = (pdf["v"] > 0.5) | ~(pdf["w"] > 0.2)
+    np.testing.assert_array_equal(m2.to_pandas().to_numpy(), pm2.to_numpy())
+    pandas.testing.assert_frame_equal(df[m2].to_pandas(), pdf[pm2])
+    s = df["v"].isna()
+    np.testing.assert_array_equal(s.to_pandas().to_numpy(),
+                                  pdf["v"].isna().to_numpy())
+
+
 def test_native_extension_is_loaded():
     """Guard against a silent eager/pandas fallback: the in-tree .so must be
     mapped into this process."""
