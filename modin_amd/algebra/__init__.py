@@ -85,8 +85,18 @@ class Binary(Operator):
 
             def zip_fn(lblock: DeviceBlock, rblock: DeviceBlock) -> DeviceBlock:
                 out = {}
+                if lblock.width == 1 and rblock.width == 1:
+                    # Series op Series: positional pairing, left name wins
+                    (lname, lcol), = lblock.columns.items()
+                    (rcol,) = rblock.columns.values()
+                    out[lname] = lib.binary(bin_op_code, lcol, rcol)
+                    return DeviceBlock(out, lblock.length)
                 for name, lcol in lblock.columns.items():
-                    rcol = rblock.columns[name]
+                    rcol = rblock.columns.get(name)
+                    if rcol is None:
+                        raise lib.HfError(
+                            f"binary frame op: column {name!r} missing on the "
+                            "right (NaN-fill alignment is a later round)")
                     out[name] = lib.binary(bin_op_code, lcol, rcol)
                 return DeviceBlock(out, lblock.length)
 
