@@ -598,7 +598,8 @@ struct GbWorkItem {
 
 __global__ void __launch_bounds__(BLOCK) k_gb_hist(
     const int64_t* __restrict__ keys, int64_t n, int64_t key_min,
-    int64_t n_slots, int nb, unsigned long long* __restrict__ hist) {
+    int64_t n_slots, int nb, int range_log,
+    unsigned long long* __restrict__ hist) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   unsigned* lhist = reinterpret_cast<unsigned*>(smem_raw);
   for (int t = threadIdx.x; t < nb; t += blockDim.x) lhist[t] = 0;
@@ -608,7 +609,7 @@ __global__ void __launch_bounds__(BLOCK) k_gb_hist(
   for (; i < n; i += stride) {
     const int64_t k = keys[i] - key_min;
     if ((uint64_t)k < (uint64_t)n_slots)
-      atomicAdd(&lhist[k >> GB_RANGE_LOG], 1u);
+      atomicAdd(&lhist[k >> range_log], 1u);
   }
   __syncthreads();
   for (int t = threadIdx.x; t < nb; t += blockDim.x)
@@ -621,7 +622,7 @@ __global__ void __launch_bounds__(BLOCK) k_gb_hist(
 // coalesced — each tile emits one contiguous chunk per bucket stream
 // (probe: 2.6x the register-staged direct scatter).  RPT rows/thread
 // (even); NV value columns; odd tail row handled by block 0 up front.
-template <int NV, int RPT, int BLK>
+template <int NV, int RPT, int BLK, int RL>
 __global__ void __launch_bounds__(BLK) k_gb_scatter(
     const int64_t* __restrict__ keys, const double* __restrict__ v0,
     const double* __restrict__ v1, int64_t n, int64_t key_min, int64_t n_slots,
@@ -634,9 +635,9 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
     // odd tail row: direct single-row reservation + write
     const int64_t k = keys[n - 1] - key_min;
     if ((uint64_t)k < (uint64_t)n_slots) {
-      const int b = (int)(k >> GB_RANGE_LOG);
+      const int b = (int)(k >> RL);
       const int64_t pos = (int64_t)atomicAdd(&cursors[b], 1u);
-      rk[pos] = (unsigned short)(k & (GB_RANGE - 1));
+      rk[pos] = (unsigned short)(k & ((1 << RL) - 1));
       if (NV > 0) r0[pos] = v0[n - 1];
       if (NV > 1) r1[pos] = v1[n - 1];
     } else {
@@ -677,16 +678,16 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
         if (NV > 0) vv0 = reinterpret_cast<const double2*>(v0)[pr];
         if (NV > 1) vv1 = reinterpret_cast<const double2*>(v1)[pr];
         if ((uint64_t)ka < (uint64_t)n_slots) {
-          lb[a] = (int)(ka >> GB_RANGE_LOG);
-          lk[a] = (unsigned)(ka & (GB_RANGE - 1));
+          lb[a] = (int)(ka >> RL);
+          lk[a] = (unsigned)(ka & ((1 << RL) - 1));
           if (NV > 0) lv0[a] = vv0.x;
           if (NV > 1) lv1[a] = vv1.x;
         } else {
           atomicAdd(err, 1ULL);
         }
         if ((uint64_t)kb < (uint64_t)n_slots) {
-          lb[bslot] = (int)(kb >> GB_RANGE_LOG);
-          lk[bslot] = (unsigned)(kb & (GB_RANGE - 1));
+          lb[bslot] = (int)(kb >> RL);
+          lk[bslot] = (unsigned)(kb & ((1 << RL) - 1));
           if (NV > 0) lv0[bslot] = vv0.y;
           if (NV > 1) lv1[bslot] = vv1.y;
         } else {
@@ -751,17 +752,18 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
 // CNT: also merge per-slot non-NaN counts for this column.
 // 2 rows/lane vectorized (regions are 64-row aligned; odd chunk tails are
 // only ever the last chunk of a bucket).
-template <bool ROWCNT, bool CNT, bool HAVE_VAL>
+template <bool ROWCNT, bool CNT, bool HAVE_VAL, int RL>
 __global__ void __launch_bounds__(512) k_gb_bucket_agg(
     const double* __restrict__ vals, const unsigned short* __restrict__ lowkeys,
     const GbWorkItem* __restrict__ work, int64_t n_slots,
     double* __restrict__ gsums, unsigned long long* __restrict__ growcnt,
     unsigned long long* __restrict__ gcounts) {
-  __shared__ double lsums[HAVE_VAL ? GB_RANGE : 1];        // 32 KB
-  __shared__ unsigned lcnt[CNT ? GB_RANGE : 1];            // 16 KB when CNT
-  __shared__ unsigned char ltouch[GB_RANGE];               // 4 KB
+  constexpr int RANGE = 1 << RL;
+  __shared__ double lsums[HAVE_VAL ? RANGE : 1];
+  __shared__ unsigned lcnt[CNT ? RANGE : 1];
+  __shared__ unsigned char ltouch[RANGE];
   const GbWorkItem w = work[blockIdx.x];
-  for (int s = threadIdx.x; s < GB_RANGE; s += blockDim.x) {
+  for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
     if (HAVE_VAL) lsums[s] = 0.0;
     ltouch[s] = 0;
     if (CNT) lcnt[s] = 0;
@@ -799,8 +801,8 @@ __global__ void __launch_bounds__(512) k_gb_bucket_agg(
     }
   }
   __syncthreads();
-  const int64_t gbase = (int64_t)w.bucket << GB_RANGE_LOG;
-  for (int s = threadIdx.x; s < GB_RANGE; s += blockDim.x) {
+  const int64_t gbase = (int64_t)w.bucket << RL;
+  for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
     if (!ltouch[s] || gbase + s >= n_slots) continue;
     if (HAVE_VAL) unsafeAtomicAdd(&gsums[gbase + s], lsums[s]);
     if (ROWCNT) atomicAdd(&growcnt[gbase + s], 1ULL);
@@ -1600,12 +1602,14 @@ int hf_reduce(const hf_col* in, hf_reduce_result* out) {
 
 // ---- groupby ----
 
+}  // extern "C" (the radix-path helpers below are C++ templates)
+
 namespace {
 
 // Per-bucket histogram of an (immutable) key column, cached host-side on the
 // hf_col — repeated groupbys on the same keys skip the P0 pass entirely.
 int ensure_host_hist(hf_col* keys, int64_t key_min, int64_t n_slots,
-                     int64_t nb) {
+                     int64_t nb, int range_log) {
   if (keys->d_hist && keys->hist_kmin == key_min && keys->hist_nb == nb)
     return HF_OK;
   if (keys->d_hist) { free(keys->d_hist); keys->d_hist = nullptr; }
@@ -1616,7 +1620,7 @@ int ensure_host_hist(hf_col* keys, int64_t key_min, int64_t n_slots,
   int rc = timed_launch("gb_hist", [&] {
     hipLaunchKernelGGL(k_gb_hist, dim3((uint32_t)grid_for(n)), dim3(BLOCK),
                        (uint32_t)(nb * 4), g.stream, (const int64_t*)keys->dptr,
-                       n, key_min, n_slots, (int)nb, d_h);
+                       n, key_min, n_slots, (int)nb, range_log, d_h);
   });
   if (rc != HF_OK) return rc;
   int64_t* h = (int64_t*)malloc(nb * 8);
@@ -1670,12 +1674,13 @@ int gb_dense_path(const hf_col* keys, const GbPtrs& ptrs, int nvals,
   return HF_OK;
 }
 
+template <int RL>
 int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
                   int64_t n_slots, uintptr_t sums, uintptr_t rowcnt,
                   uintptr_t counts, unsigned long long* d_err) {
   const int64_t n = keys->len;
-  const int64_t nb = (n_slots + GB_RANGE - 1) >> GB_RANGE_LOG;
-  int rc = ensure_host_hist(keys, key_min, n_slots, nb);
+  const int64_t nb = (n_slots + (1 << RL) - 1) >> RL;
+  int rc = ensure_host_hist(keys, key_min, n_slots, nb, RL);
   if (rc != HF_OK) return rc;
   const int64_t* h = keys->d_hist ? (const int64_t*)keys->d_hist : nullptr;
   // exact per-bucket regions, 64-row aligned; u32 cursors cap one partition
@@ -1732,7 +1737,7 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
     const uint32_t lds =
         (uint32_t)(tile_sz * (8 * NVv + 4) + nb * 12 + 16);
     return timed_launch("gb_scatter", [&] {
-      hipLaunchKernelGGL((k_gb_scatter<NVv, RPTv, BLKv>), dim3(sgrid),
+      hipLaunchKernelGGL((k_gb_scatter<NVv, RPTv, BLKv, RL>), dim3(sgrid),
                          dim3(BLKv), lds, g.stream,
                          (const int64_t*)keys->dptr, ptrs.vals[0], ptrs.vals[1],
                          n, key_min, n_slots, (int)nb, d_cur, r0, r1, rk, d_err);
@@ -1753,8 +1758,8 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
     constexpr bool R = decltype(rTag)::value, C = decltype(cTag)::value,
                    V = decltype(vTag)::value;
     return timed_launch("gb_bucket_agg", [&] {
-      hipLaunchKernelGGL((k_gb_bucket_agg<R, C, V>), dim3(agrid), dim3(512), 0,
-                         g.stream, v, rk, d_work, n_slots, gs,
+      hipLaunchKernelGGL((k_gb_bucket_agg<R, C, V, RL>), dim3(agrid), dim3(512),
+                         0, g.stream, v, rk, d_work, n_slots, gs,
                          (unsigned long long*)rowcnt, gc);
     });
   };
@@ -1788,6 +1793,8 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
 
 }  // namespace
 
+extern "C" {
+
 int hf_groupby_accum(const hf_col* keys, const hf_col* const* vals, int nvals,
                      int64_t key_min, int64_t n_slots,
                      uintptr_t sums, uintptr_t rowcnt, uintptr_t counts) {
@@ -1813,10 +1820,23 @@ int hf_groupby_accum(const hf_col* keys, const hf_col* const* vals, int nvals,
   if (n > 0 && n_slots <= GB_DENSE_MAX)
     return gb_dense_path(keys, ptrs, nvals, key_min, n_slots, sums, rowcnt,
                          counts, d_err);
-  const int64_t nb = (n_slots + GB_RANGE - 1) >> GB_RANGE_LOG;
-  if (n > 0 && nb <= GB_MAX_BUCKETS && nvals <= 2)
-    return gb_radix_path(const_cast<hf_col*>(keys), ptrs, nvals, key_min,
-                         n_slots, sums, rowcnt, counts, d_err);
+  // sum-only uses 8192-key buckets (bigger scatter chunks; the u8-touch agg
+  // table still fits 2 blocks/CU); count/mean use 4096 (3 blocks/CU with the
+  // counts table)
+  const bool want_cnt = counts != 0;
+  const int64_t nb13 = (n_slots + (1 << 13) - 1) >> 13;
+  const int64_t nb12 = (n_slots + (1 << 12) - 1) >> 12;
+  if (n > 0 && nvals <= 2) {
+    if (!want_cnt && nb13 <= GB_MAX_BUCKETS)
+      return gb_radix_path<13>(const_cast<hf_col*>(keys), ptrs, nvals, key_min,
+                               n_slots, sums, rowcnt, counts, d_err);
+    if (want_cnt && nb12 <= GB_MAX_BUCKETS)
+      return gb_radix_path<12>(const_cast<hf_col*>(keys), ptrs, nvals, key_min,
+                               n_slots, sums, rowcnt, counts, d_err);
+    if (nb13 <= GB_MAX_BUCKETS)  // counts wanted but range too big for RL12
+      return gb_radix_path<13>(const_cast<hf_col*>(keys), ptrs, nvals, key_min,
+                               n_slots, sums, rowcnt, counts, d_err);
+  }
   auto launch = [&](auto nvTag, auto cntTag) {
     constexpr int NV = decltype(nvTag)::value;
     constexpr bool CNT = decltype(cntTag)::value;
