@@ -149,11 +149,22 @@ int hf_reduce(const hf_col* in, hf_reduce_result* out); /* syncs the stream */
  *                                     (optional: pass 0 to skip; needed for
  *                                     count/mean)
  * All buffers must be zeroed by the caller before the first accumulate. */
+enum { HF_AGG_SUM = 0, HF_AGG_MIN = 1, HF_AGG_MAX = 2 };
+/* agg_op picks the per-slot combine for the value table (the "sums" buffer
+ * doubles as the min/max table; initialize it to 0 / +inf / -inf with
+ * hf_fill_f64 before the first accumulate — GroupbyReduceImpl's other
+ * map/reduce pairs, storage_formats/pandas/groupby.py:237-248). */
 int hf_groupby_accum(const hf_col* keys,            /* HF_INT64, len n        */
                      const hf_col* const* vals,     /* nvals HF_FLOAT64 cols  */
-                     int nvals,
+                     int nvals, int agg_op,
                      int64_t key_min, int64_t n_slots,
                      uintptr_t sums, uintptr_t rowcnt, uintptr_t counts);
+
+int hf_fill_f64(uintptr_t dptr, double value, int64_t n);
+
+/* out[i] = count[i] != 0 ? val[i] : NaN — min/max of an empty (all-NaN)
+ * group is NaN in pandas. */
+int hf_fixup_empty(const hf_col* val, const hf_col* cnt, hf_col** out);
 
 /* Compact a (merged) table to pandas-groupby-shaped output: ascending present
  * keys (rowcnt>0), per-column sums, optional counts.  Returns n_groups and

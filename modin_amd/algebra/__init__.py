@@ -158,16 +158,17 @@ class GroupByReduce(Operator):
       sum   -> map "accumulate", reduce "table-merge", value sums
       count -> same kernels, value counts
       mean  -> sums/counts division on the compacted columns (:87-113 shape)
+      min/max -> ds/global f64 min/max tables; empty groups fixed to NaN
     """
 
-    SUPPORTED = ("sum", "count", "mean")
+    SUPPORTED = ("sum", "count", "mean", "min", "max")
 
     @classmethod
     def register(cls, agg: str):
         if agg not in cls.SUPPORTED:
             raise lib.HfError(
-                f"groupby agg {agg!r} not implemented (round-1 dense-key table "
-                "supports sum/count/mean)"
+                f"groupby agg {agg!r} not implemented (round-1 supports "
+                "sum/count/mean/min/max)"
             )
 
         def caller(query_compiler, by: str, **kwargs):

@@ -109,9 +109,10 @@ class HipDataframe:
     # ---- GroupByReduce (dataframe.py:4530) ----
     def groupby_reduce(self, by: str, agg: str) -> "HipDataframe":
         val_names = [c for c in self.columns if c != by]
-        want_counts = agg in ("count", "mean")
+        want_counts = agg in ("count", "mean", "min", "max")
+        agg_op = lib.AGG_OP_OF[agg]
         keys, sums, counts, n = self._partition_mgr_cls.groupby_reduce(
-            self._partitions, by, val_names, want_counts
+            self._partitions, by, val_names, want_counts, agg_op
         )
         if agg == "sum":
             cols = {name: sums[i] for i, name in enumerate(val_names)}
@@ -119,6 +120,11 @@ class HipDataframe:
         elif agg == "count":
             cols = {name: counts[i] for i, name in enumerate(val_names)}
             dtypes = pandas.Series({n_: np.dtype(np.int64) for n_ in val_names})
+        elif agg in ("min", "max"):
+            # empty (all-NaN) groups hold the agg identity; pandas says NaN
+            cols = {name: lib.fixup_empty(sums[i], counts[i])
+                    for i, name in enumerate(val_names)}
+            dtypes = pandas.Series({n_: np.dtype(np.float64) for n_ in val_names})
         else:  # mean = sums / counts (GroupbyReduceImpl mean shape, groupby.py:87)
             cols = {}
             for i, name in enumerate(val_names):

@@ -102,8 +102,12 @@ def load() -> ct.CDLL:
                                      ct.POINTER(ct.c_void_p)]),
             "hf_reduce": (ct.c_int, [ct.c_void_p, ct.POINTER(HfReduceResult)]),
             "hf_groupby_accum": (ct.c_int, [ct.c_void_p, ct.POINTER(ct.c_void_p),
-                                            ct.c_int, ct.c_int64, ct.c_int64,
-                                            ct.c_size_t, ct.c_size_t, ct.c_size_t]),
+                                            ct.c_int, ct.c_int, ct.c_int64,
+                                            ct.c_int64, ct.c_size_t,
+                                            ct.c_size_t, ct.c_size_t]),
+            "hf_fill_f64": (ct.c_int, [ct.c_size_t, ct.c_double, ct.c_int64]),
+            "hf_fixup_empty": (ct.c_int, [ct.c_void_p, ct.c_void_p,
+                                          ct.POINTER(ct.c_void_p)]),
             "hf_groupby_compact": (ct.c_int, [ct.c_size_t, ct.c_size_t, ct.c_size_t,
                                               ct.c_int, ct.c_int64, ct.c_int64,
                                               ct.POINTER(ct.c_void_p),
@@ -153,7 +157,8 @@ def exported_symbols():
         "hf_put", "hf_get", "hf_col_alloc", "hf_col_free", "hf_col_len",
         "hf_col_dtype", "hf_col_dptr", "hf_alloc_raw", "hf_free_raw",
         "hf_memset_raw", "hf_map_scalar", "hf_map_scalar_i64", "hf_binary",
-        "hf_reduce", "hf_groupby_accum", "hf_groupby_compact",
+        "hf_reduce", "hf_groupby_accum", "hf_groupby_compact", "hf_fill_f64",
+        "hf_fixup_empty",
         "hf_col_concat", "hf_join_build", "hf_join_free", "hf_join_probe",
         "hf_gather", "hf_compare_scalar", "hf_filter_plan", "hf_filter_apply",
         "hf_filter_iota", "hf_filter_plan_free", "hf_profiling",

@@ -55,21 +55,27 @@ class GroupbyTable:
     (multi-GPU, so RCCL can all-reduce them in place) — see distributed.py.
     """
 
-    def __init__(self, nvals: int, key_min: int, n_slots: int, want_counts: bool):
+    def __init__(self, nvals: int, key_min: int, n_slots: int,
+                 want_counts: bool, agg_op: int = 0):
         self.nvals = nvals
         self.key_min = key_min
         self.n_slots = n_slots
         self.want_counts = want_counts
+        self.agg_op = agg_op
+        init = lib.AGG_IDENTITY[agg_op]
         self._torch_tensors = None
         from ..distributed import is_active, alloc_table_torch
         if is_active():
             self._torch_tensors, self.sums, self.rowcnt, self.counts = \
-                alloc_table_torch(nvals, n_slots, want_counts)
+                alloc_table_torch(nvals, n_slots, want_counts, init)
         else:
             self.sums = lib.alloc_raw(8 * nvals * n_slots)
             self.rowcnt = lib.alloc_raw(8 * n_slots)
             self.counts = lib.alloc_raw(8 * nvals * n_slots) if want_counts else 0
-            lib.memset_raw(self.sums, 0, 8 * nvals * n_slots)
+            if init == 0.0:
+                lib.memset_raw(self.sums, 0, 8 * nvals * n_slots)
+            else:
+                lib.fill_f64(self.sums, init, nvals * n_slots)
             lib.memset_raw(self.rowcnt, 0, 8 * n_slots)
             if want_counts:
                 lib.memset_raw(self.counts, 0, 8 * nvals * n_slots)
@@ -181,7 +187,8 @@ class HipDataframePartitionManager:
     # ---- GroupByReduce (partition_manager.py:303) ----
     @classmethod
     @wait_if_benchmark_mode
-    def groupby_reduce(cls, parts, by_name, val_names, want_counts):
+    def groupby_reduce(cls, parts, by_name, val_names, want_counts,
+                       agg_op=0):
         """Dense-table groupby: returns (keys_col, sum_cols, count_cols, n).
 
         Map phase (algebra/groupby.py:124 device form): hf_groupby_accum per
@@ -222,11 +229,12 @@ class HipDataframePartitionManager:
                 "not apply; hash aggregation is a later round"
             )
         n_slots = max(n_slots, 1)
-        table = GroupbyTable(len(val_names), kmin, n_slots, want_counts)
+        table = GroupbyTable(len(val_names), kmin, n_slots, want_counts,
+                             agg_op)
         for kcol, vals in zip(key_cols, val_cols_per_part):
             if kcol.length:
-                lib.groupby_accum(kcol, vals, kmin, n_slots, table.sums,
-                                  table.rowcnt, table.counts)
+                lib.groupby_accum(kcol, vals, agg_op, kmin, n_slots,
+                                  table.sums, table.rowcnt, table.counts)
         maybe_allreduce_table(table)
         keys, sums, counts, n = lib.groupby_compact(
             table.sums, table.rowcnt, table.counts, len(val_names), kmin, n_slots

@@ -506,11 +506,38 @@ __global__ void __launch_bounds__(BLOCK) k_reduce_i64(const int64_t* __restrict_
 
 constexpr int GB_MAX_VALS = 8;
 
+// per-slot combine for the value table (HF_AGG_*); LDS and global forms use
+// hardware ds_*_f64 / global_atomic_*_f64
+template <int AOP>
+__device__ __forceinline__ void lds_slot_agg(double* a, double v) {
+  if (AOP == HF_AGG_SUM)
+    unsafeAtomicAdd(a, v);
+  else if (AOP == HF_AGG_MIN)
+    __hip_atomic_fetch_min(a, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_WORKGROUP);
+  else
+    __hip_atomic_fetch_max(a, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_WORKGROUP);
+}
+template <int AOP>
+__device__ __forceinline__ void glob_slot_agg(double* a, double v) {
+  if (AOP == HF_AGG_SUM)
+    unsafeAtomicAdd(a, v);
+  else if (AOP == HF_AGG_MIN)
+    __hip_atomic_fetch_min(a, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  else
+    __hip_atomic_fetch_max(a, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+template <int AOP>
+__device__ __forceinline__ double agg_identity() {
+  return AOP == HF_AGG_SUM ? 0.0
+         : AOP == HF_AGG_MIN ? __longlong_as_double(0x7FF0000000000000LL)
+                             : __longlong_as_double(0xFFF0000000000000LL);
+}
+
 struct GbPtrs {
   const double* vals[GB_MAX_VALS];
 };
 
-template <int NVALS, bool COUNTS>
+template <int NVALS, bool COUNTS, int AOP>
 __global__ void __launch_bounds__(BLOCK) k_gb_accum(
     const int64_t* __restrict__ keys, GbPtrs ptrs, int64_t n,
     int64_t key_min, int64_t n_slots,
@@ -534,11 +561,11 @@ __global__ void __launch_bounds__(BLOCK) k_gb_accum(
     for (int c = 0; c < NVALS; ++c) {
       const double2 v = reinterpret_cast<const double2*>(ptrs.vals[c])[i];
       if (ok0 && v.x == v.x) {
-        unsafeAtomicAdd(&sums[(int64_t)c * n_slots + k0], v.x);
+        glob_slot_agg<AOP>(&sums[(int64_t)c * n_slots + k0], v.x);
         if (COUNTS) atomicAdd(&counts[(int64_t)c * n_slots + k0], 1ULL);
       }
       if (ok1 && v.y == v.y) {
-        unsafeAtomicAdd(&sums[(int64_t)c * n_slots + k1], v.y);
+        glob_slot_agg<AOP>(&sums[(int64_t)c * n_slots + k1], v.y);
         if (COUNTS) atomicAdd(&counts[(int64_t)c * n_slots + k1], 1ULL);
       }
     }
@@ -550,7 +577,7 @@ __global__ void __launch_bounds__(BLOCK) k_gb_accum(
       for (int c = 0; c < NVALS; ++c) {
         const double v = ptrs.vals[c][n - 1];
         if (v == v) {
-          unsafeAtomicAdd(&sums[(int64_t)c * n_slots + k], v);
+          glob_slot_agg<AOP>(&sums[(int64_t)c * n_slots + k], v);
           if (COUNTS) atomicAdd(&counts[(int64_t)c * n_slots + k], 1ULL);
         }
       }
@@ -752,7 +779,7 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
 // CNT: also merge per-slot non-NaN counts for this column.
 // 2 rows/lane vectorized (regions are 64-row aligned; odd chunk tails are
 // only ever the last chunk of a bucket).
-template <bool ROWCNT, bool CNT, bool HAVE_VAL, int RL>
+template <bool ROWCNT, bool CNT, bool HAVE_VAL, int RL, int AOP>
 __global__ void __launch_bounds__(512) k_gb_bucket_agg(
     const double* __restrict__ vals, const unsigned short* __restrict__ lowkeys,
     const GbWorkItem* __restrict__ work, int64_t n_slots,
@@ -764,7 +791,7 @@ __global__ void __launch_bounds__(512) k_gb_bucket_agg(
   __shared__ unsigned char ltouch[RANGE];
   const GbWorkItem w = work[blockIdx.x];
   for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
-    if (HAVE_VAL) lsums[s] = 0.0;
+    if (HAVE_VAL) lsums[s] = agg_identity<AOP>();
     ltouch[s] = 0;
     if (CNT) lcnt[s] = 0;
   }
@@ -780,11 +807,11 @@ __global__ void __launch_bounds__(512) k_gb_bucket_agg(
     if (HAVE_VAL) {
       const double2 vv = v2[i];
       if (vv.x == vv.x) {
-        unsafeAtomicAdd(&lsums[kk.x], vv.x);
+        lds_slot_agg<AOP>(&lsums[kk.x], vv.x);
         if (CNT) atomicAdd(&lcnt[kk.x], 1u);
       }
       if (vv.y == vv.y) {
-        unsafeAtomicAdd(&lsums[kk.y], vv.y);
+        lds_slot_agg<AOP>(&lsums[kk.y], vv.y);
         if (CNT) atomicAdd(&lcnt[kk.y], 1u);
       }
     }
@@ -795,7 +822,7 @@ __global__ void __launch_bounds__(512) k_gb_bucket_agg(
     if (HAVE_VAL) {
       const double v = vals[end - 1];
       if (v == v) {
-        unsafeAtomicAdd(&lsums[slot], v);
+        lds_slot_agg<AOP>(&lsums[slot], v);
         if (CNT) atomicAdd(&lcnt[slot], 1u);
       }
     }
@@ -804,14 +831,14 @@ __global__ void __launch_bounds__(512) k_gb_bucket_agg(
   const int64_t gbase = (int64_t)w.bucket << RL;
   for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
     if (!ltouch[s] || gbase + s >= n_slots) continue;
-    if (HAVE_VAL) unsafeAtomicAdd(&gsums[gbase + s], lsums[s]);
+    if (HAVE_VAL) glob_slot_agg<AOP>(&gsums[gbase + s], lsums[s]);
     if (ROWCNT) atomicAdd(&growcnt[gbase + s], 1ULL);
     if (CNT) atomicAdd(&gcounts[gbase + s], (unsigned long long)lcnt[s]);
   }
 }
 
 // Dense direct path: n_slots <= GB_RANGE, one 16 B/row pass per column.
-template <bool ROWCNT, bool CNT, bool HAVE_VAL>
+template <bool ROWCNT, bool CNT, bool HAVE_VAL, int AOP>
 __global__ void __launch_bounds__(512) k_gb_dense(
     const int64_t* __restrict__ keys, const double* __restrict__ vals,
     int64_t n, int64_t key_min, int64_t n_slots,
@@ -825,7 +852,7 @@ __global__ void __launch_bounds__(512) k_gb_dense(
   unsigned char* ltouch = reinterpret_cast<unsigned char*>(
       lcnt + (CNT ? n_slots : 0));                           // presence bytes
   for (int64_t s = threadIdx.x; s < n_slots; s += blockDim.x) {
-    if (HAVE_VAL) lsums[s] = 0.0;
+    if (HAVE_VAL) lsums[s] = agg_identity<AOP>();
     ltouch[s] = 0;
     if (CNT) lcnt[s] = 0;
   }
@@ -842,7 +869,7 @@ __global__ void __launch_bounds__(512) k_gb_dense(
     if (HAVE_VAL) {
       const double v = vals[i];
       if (v == v) {
-        unsafeAtomicAdd(&lsums[k], v);
+        lds_slot_agg<AOP>(&lsums[k], v);
         if (CNT) atomicAdd(&lcnt[k], 1u);
       }
     }
@@ -850,7 +877,7 @@ __global__ void __launch_bounds__(512) k_gb_dense(
   __syncthreads();
   for (int64_t s = threadIdx.x; s < n_slots; s += blockDim.x) {
     if (!ltouch[s]) continue;
-    if (HAVE_VAL) unsafeAtomicAdd(&gsums[s], lsums[s]);
+    if (HAVE_VAL) glob_slot_agg<AOP>(&gsums[s], lsums[s]);
     if (ROWCNT) atomicAdd(&growcnt[s], 1ULL);
     if (CNT) atomicAdd(&gcounts[s], (unsigned long long)lcnt[s]);
   }
@@ -1178,6 +1205,22 @@ __global__ void __launch_bounds__(BLOCK) k_probe_emit(
     }
     __syncthreads();
   }
+}
+
+__global__ void __launch_bounds__(BLOCK) k_fill_f64(double* __restrict__ p,
+                                                    double v, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) p[i] = v;
+}
+
+__global__ void __launch_bounds__(BLOCK) k_fixup_empty(
+    const double* __restrict__ val, const int64_t* __restrict__ cnt,
+    double* __restrict__ out, int64_t n) {
+  const double nanv = __longlong_as_double(0x7FF8000000000000LL);
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = cnt[i] ? val[i] : nanv;
 }
 
 // ---- compare + filter kernels ----
@@ -1634,6 +1677,7 @@ int ensure_host_hist(hf_col* keys, int64_t key_min, int64_t n_slots,
   return HF_OK;
 }
 
+template <int AOP>
 int gb_dense_path(const hf_col* keys, const GbPtrs& ptrs, int nvals,
                   int64_t key_min, int64_t n_slots, uintptr_t sums,
                   uintptr_t rowcnt, uintptr_t counts,
@@ -1649,10 +1693,10 @@ int gb_dense_path(const hf_col* keys, const GbPtrs& ptrs, int nvals,
     const uint32_t lds =
         (uint32_t)(n_slots * ((V ? 8 : 0) + 1 + (C ? 4 : 0)));
     return timed_launch("gb_dense", [&] {
-      hipLaunchKernelGGL((k_gb_dense<R, C, V>), dim3((uint32_t)grid), dim3(512),
-                         lds, g.stream, (const int64_t*)keys->dptr, v, n,
-                         key_min, n_slots, gs, (unsigned long long*)rowcnt, gc,
-                         d_err);
+      hipLaunchKernelGGL((k_gb_dense<R, C, V, AOP>), dim3((uint32_t)grid),
+                         dim3(512), lds, g.stream, (const int64_t*)keys->dptr,
+                         v, n, key_min, n_slots, gs,
+                         (unsigned long long*)rowcnt, gc, d_err);
     });
   };
   using T = std::true_type;
@@ -1674,7 +1718,7 @@ int gb_dense_path(const hf_col* keys, const GbPtrs& ptrs, int nvals,
   return HF_OK;
 }
 
-template <int RL>
+template <int RL, int AOP>
 int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
                   int64_t n_slots, uintptr_t sums, uintptr_t rowcnt,
                   uintptr_t counts, unsigned long long* d_err) {
@@ -1758,8 +1802,8 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
     constexpr bool R = decltype(rTag)::value, C = decltype(cTag)::value,
                    V = decltype(vTag)::value;
     return timed_launch("gb_bucket_agg", [&] {
-      hipLaunchKernelGGL((k_gb_bucket_agg<R, C, V, RL>), dim3(agrid), dim3(512),
-                         0, g.stream, v, rk, d_work, n_slots, gs,
+      hipLaunchKernelGGL((k_gb_bucket_agg<R, C, V, RL, AOP>), dim3(agrid),
+                         dim3(512), 0, g.stream, v, rk, d_work, n_slots, gs,
                          (unsigned long long*)rowcnt, gc);
     });
   };
@@ -1796,11 +1840,13 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
 extern "C" {
 
 int hf_groupby_accum(const hf_col* keys, const hf_col* const* vals, int nvals,
-                     int64_t key_min, int64_t n_slots,
+                     int agg_op, int64_t key_min, int64_t n_slots,
                      uintptr_t sums, uintptr_t rowcnt, uintptr_t counts) {
   HF_NEED_INIT("hf_groupby_accum");
   if (!keys || !vals || nvals < 0 || nvals > GB_MAX_VALS)
     return set_err(HF_ERR_ARG, "hf_groupby_accum", "bad args (nvals<=8)");
+  if (agg_op < HF_AGG_SUM || agg_op > HF_AGG_MAX)
+    return set_err(HF_ERR_ARG, "hf_groupby_accum", "bad agg_op");
   if (keys->dtype != HF_INT64)
     return set_err(HF_ERR_ARG, "hf_groupby_accum", "keys must be int64");
   if (n_slots <= 0) return set_err(HF_ERR_ARG, "hf_groupby_accum", "n_slots<=0");
@@ -1817,36 +1863,54 @@ int hf_groupby_accum(const hf_col* keys, const hf_col* const* vals, int nvals,
   // path selection (DESIGN.md §GroupBy kernels): LDS-dense for small ranges,
   // radix partition for the north-star range, global atomics as the wide
   // fallback (slow but correct for any range below the slot cap)
-  if (n > 0 && n_slots <= GB_DENSE_MAX)
-    return gb_dense_path(keys, ptrs, nvals, key_min, n_slots, sums, rowcnt,
-                         counts, d_err);
-  // sum-only uses 8192-key buckets (bigger scatter chunks; the u8-touch agg
-  // table still fits 2 blocks/CU); count/mean use 4096 (3 blocks/CU with the
-  // counts table)
-  const bool want_cnt = counts != 0;
-  const int64_t nb13 = (n_slots + (1 << 13) - 1) >> 13;
-  const int64_t nb12 = (n_slots + (1 << 12) - 1) >> 12;
-  if (n > 0 && nvals <= 2) {
-    if (!want_cnt && nb13 <= GB_MAX_BUCKETS)
-      return gb_radix_path<13>(const_cast<hf_col*>(keys), ptrs, nvals, key_min,
-                               n_slots, sums, rowcnt, counts, d_err);
-    if (want_cnt && nb12 <= GB_MAX_BUCKETS)
-      return gb_radix_path<12>(const_cast<hf_col*>(keys), ptrs, nvals, key_min,
-                               n_slots, sums, rowcnt, counts, d_err);
-    if (nb13 <= GB_MAX_BUCKETS)  // counts wanted but range too big for RL12
-      return gb_radix_path<13>(const_cast<hf_col*>(keys), ptrs, nvals, key_min,
-                               n_slots, sums, rowcnt, counts, d_err);
-  }
+  auto route = [&](auto aopTag) -> int {
+    constexpr int AOP = decltype(aopTag)::value;
+    if (n > 0 && n_slots <= GB_DENSE_MAX)
+      return gb_dense_path<AOP>(keys, ptrs, nvals, key_min, n_slots, sums,
+                                rowcnt, counts, d_err);
+    // sum-only uses 8192-key buckets (bigger scatter chunks; the u8-touch agg
+    // table still fits 2 blocks/CU); count/mean use 4096 (3 blocks/CU with
+    // the counts table)
+    const bool want_cnt = counts != 0;
+    const int64_t nb13 = (n_slots + (1 << 13) - 1) >> 13;
+    const int64_t nb12 = (n_slots + (1 << 12) - 1) >> 12;
+    if (n > 0 && nvals <= 2) {
+      if (!want_cnt && nb13 <= GB_MAX_BUCKETS)
+        return gb_radix_path<13, AOP>(const_cast<hf_col*>(keys), ptrs, nvals,
+                                      key_min, n_slots, sums, rowcnt, counts,
+                                      d_err);
+      if (want_cnt && nb12 <= GB_MAX_BUCKETS)
+        return gb_radix_path<12, AOP>(const_cast<hf_col*>(keys), ptrs, nvals,
+                                      key_min, n_slots, sums, rowcnt, counts,
+                                      d_err);
+      if (nb13 <= GB_MAX_BUCKETS)
+        return gb_radix_path<13, AOP>(const_cast<hf_col*>(keys), ptrs, nvals,
+                                      key_min, n_slots, sums, rowcnt, counts,
+                                      d_err);
+    }
+    return -1;  // fall through to the atomic path
+  };
+  int routed = agg_op == HF_AGG_SUM ? route(std::integral_constant<int, HF_AGG_SUM>{})
+               : agg_op == HF_AGG_MIN ? route(std::integral_constant<int, HF_AGG_MIN>{})
+                                      : route(std::integral_constant<int, HF_AGG_MAX>{});
+  if (routed != -1) return routed;
   auto launch = [&](auto nvTag, auto cntTag) {
     constexpr int NV = decltype(nvTag)::value;
     constexpr bool CNT = decltype(cntTag)::value;
-    return timed_launch("gb_accum", [&] {
-      hipLaunchKernelGGL((k_gb_accum<NV, CNT>), dim3(grid_for((n >> 1) + 1)),
-                         dim3(BLOCK), 0, g.stream, (const int64_t*)keys->dptr, ptrs,
-                         n, key_min, n_slots, (double*)sums,
-                         (unsigned long long*)rowcnt, (unsigned long long*)counts,
-                         d_err);
-    });
+    auto go = [&](auto aopTag) {
+      constexpr int AOP = decltype(aopTag)::value;
+      return timed_launch("gb_accum", [&] {
+        hipLaunchKernelGGL((k_gb_accum<NV, CNT, AOP>),
+                           dim3(grid_for((n >> 1) + 1)), dim3(BLOCK), 0,
+                           g.stream, (const int64_t*)keys->dptr, ptrs, n,
+                           key_min, n_slots, (double*)sums,
+                           (unsigned long long*)rowcnt,
+                           (unsigned long long*)counts, d_err);
+      });
+    };
+    return agg_op == HF_AGG_SUM ? go(std::integral_constant<int, HF_AGG_SUM>{})
+           : agg_op == HF_AGG_MIN ? go(std::integral_constant<int, HF_AGG_MIN>{})
+                                  : go(std::integral_constant<int, HF_AGG_MAX>{});
   };
   const bool cnt = counts != 0;
   switch (nvals) {
@@ -2242,6 +2306,33 @@ int hf_gather(const hf_col* col, const hf_col* idx, hf_col** out) {
       hipLaunchKernelGGL(k_gather_i64, dim3((uint32_t)grid_for(n)), dim3(BLOCK),
                          0, g.stream, (const int64_t*)col->dptr,
                          (const int64_t*)idx->dptr, (int64_t*)(*out)->dptr, n);
+  });
+  if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  return rc;
+}
+
+int hf_fill_f64(uintptr_t dptr, double value, int64_t n) {
+  HF_NEED_INIT("hf_fill_f64");
+  if (n <= 0) return HF_OK;
+  return timed_launch("fill_f64", [&] {
+    hipLaunchKernelGGL(k_fill_f64, dim3((uint32_t)grid_for(n)), dim3(BLOCK), 0,
+                       g.stream, (double*)dptr, value, n);
+  });
+}
+
+int hf_fixup_empty(const hf_col* val, const hf_col* cnt, hf_col** out) {
+  HF_NEED_INIT("hf_fixup_empty");
+  if (!val || !cnt || !out || val->len != cnt->len ||
+      val->dtype != HF_FLOAT64 || cnt->dtype != HF_INT64)
+    return set_err(HF_ERR_ARG, "hf_fixup_empty", "bad args");
+  int rc = hf_col_alloc(val->len, HF_FLOAT64, out);
+  if (rc != HF_OK) return rc;
+  if (val->len == 0) return HF_OK;
+  rc = timed_launch("fixup_empty", [&] {
+    hipLaunchKernelGGL(k_fixup_empty, dim3((uint32_t)grid_for(val->len)),
+                       dim3(BLOCK), 0, g.stream, (const double*)val->dptr,
+                       (const int64_t*)cnt->dptr, (double*)(*out)->dptr,
+                       val->len);
   });
   if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
   return rc;

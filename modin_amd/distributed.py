@@ -147,14 +147,17 @@ def allreduce_scalars(values):
     return t.tolist()
 
 
-def alloc_table_torch(nvals: int, n_slots: int, want_counts: bool):
-    """Zeroed RCCL-reducible table buffers on this rank's GPU.
+def alloc_table_torch(nvals: int, n_slots: int, want_counts: bool,
+                      init: float = 0.0):
+    """RCCL-reducible table buffers on this rank's GPU (value table filled
+    with the agg identity: 0 / +inf / -inf).
 
     Returns (keepalive, sums_ptr, rowcnt_ptr, counts_ptr).
     """
     import torch
     dev = _state["device"]
-    sums = torch.zeros(nvals * n_slots, dtype=torch.float64, device=dev)
+    sums = torch.full((nvals * n_slots,), init, dtype=torch.float64,
+                      device=dev)
     rowcnt = torch.zeros(n_slots, dtype=torch.int64, device=dev)
     counts = (torch.zeros(nvals * n_slots, dtype=torch.int64, device=dev)
               if want_counts else None)
@@ -174,7 +177,10 @@ def maybe_allreduce_table(table) -> None:
         from .core import lib
         lib.sync()  # hipframe-stream accumulation must be visible to RCCL
     sums, rowcnt, counts = table._torch_tensors
-    dist.all_reduce(sums)
+    agg_op = getattr(table, "agg_op", 0)
+    op = (dist.ReduceOp.MIN if agg_op == 1
+          else dist.ReduceOp.MAX if agg_op == 2 else dist.ReduceOp.SUM)
+    dist.all_reduce(sums, op=op)
     dist.all_reduce(rowcnt)
     if counts is not None:
         dist.all_reduce(counts)

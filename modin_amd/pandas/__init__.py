@@ -304,6 +304,12 @@ class DataFrameGroupBy:
     def mean(self):
         return self._agg("mean")
 
+    def min(self):
+        return self._agg("min")
+
+    def max(self):
+        return self._agg("max")
+
     def agg(self, how):
         if isinstance(how, str):
             return self._agg(how)

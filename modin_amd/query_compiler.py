@@ -76,12 +76,16 @@ class HipQueryCompiler:
     groupby_sum = GroupByReduce.register("sum")
     groupby_count = GroupByReduce.register("count")
     groupby_mean = GroupByReduce.register("mean")
+    groupby_min = GroupByReduce.register("min")
+    groupby_max = GroupByReduce.register("max")
 
     def groupby_agg(self, by: str, agg: str) -> "HipQueryCompiler":
         fn = {
             "sum": type(self).groupby_sum,
             "count": type(self).groupby_count,
             "mean": type(self).groupby_mean,
+            "min": type(self).groupby_min,
+            "max": type(self).groupby_max,
         }.get(agg)
         if fn is None:
             raise lib.HfError(

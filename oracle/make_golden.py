@@ -61,7 +61,7 @@ def gen_groupby_cases(mpd, rng):
         arrays = {"in_k": data["k"]}
         for cn, cv in cols.items():
             arrays[f"in_{cn}"] = cv
-        for agg in ("sum", "count", "mean"):
+        for agg in ("sum", "count", "mean", "min", "max"):
             mres = getattr(mdf.groupby("k"), agg)()
             pres = getattr(pdf.groupby("k"), agg)()
             got = _check_vs_pandas(mres, pres)

@@ -161,6 +161,13 @@ def groupby_agg(keys: np.ndarray, vals: dict, agg: str):
         elif agg == "mean":
             with np.errstate(invalid="ignore", divide="ignore"):
                 out[name] = sums[present] / cnts[present]
+        elif agg in ("min", "max"):
+            acc = np.full(n_slots, np.inf if agg == "min" else -np.inf)
+            ufunc = np.minimum if agg == "min" else np.maximum
+            ufunc.at(acc, shifted[valid], v[valid])
+            res = acc[present]
+            res[cnts[present] == 0] = np.nan  # all-NaN group -> NaN
+            out[name] = res
         else:
             raise ValueError(agg)
     return out_keys, out

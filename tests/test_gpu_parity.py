@@ -45,7 +45,7 @@ def _gb_inputs(g):
 
 
 @pytest.mark.parametrize("case", golden_cases("gb_"))
-@pytest.mark.parametrize("agg", ["sum", "count", "mean"])
+@pytest.mark.parametrize("agg", ["sum", "count", "mean", "min", "max"])
 def test_groupby_vs_golden(case, agg, npartitions):
     g = load_golden(case)
     df = mpd.DataFrame(_gb_inputs(g))

@@ -47,7 +47,7 @@ def test_groupby_reduce_agg_table():
     """The algebra layer's agg table mirrors GroupbyReduceImpl's supported
     map/reduce pairs (storage_formats/pandas/groupby.py:237-248 subset)."""
     from modin_amd.algebra import GroupByReduce
-    assert set(GroupByReduce.SUPPORTED) == {"sum", "count", "mean"}
+    assert set(GroupByReduce.SUPPORTED) == {"sum", "count", "mean", "min", "max"}
     with pytest.raises(lib.HfError, match="not implemented"):
         GroupByReduce.register("median")
 

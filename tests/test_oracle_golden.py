@@ -20,7 +20,7 @@ def _in_cols(g):
 
 
 @pytest.mark.parametrize("case", GB_CASES)
-@pytest.mark.parametrize("agg", ["sum", "count", "mean"])
+@pytest.mark.parametrize("agg", ["sum", "count", "mean", "min", "max"])
 def test_groupby_vs_golden(case, agg):
     g = load_golden(case)
     keys, out = oracle.groupby_agg(g["in_k"], _in_cols(g), agg)
