@@ -334,12 +334,32 @@ def memset_raw(dptr: int, value: int, nbytes: int) -> None:
     _check(load().hf_memset_raw(dptr, value, nbytes), "hf_memset_raw")
 
 
-def groupby_accum(keys: ColumnRef, vals: list[ColumnRef], key_min: int,
+AGG_SUM, AGG_MIN, AGG_MAX = range(3)
+AGG_OP_OF = {"sum": AGG_SUM, "count": AGG_SUM, "mean": AGG_SUM,
+             "min": AGG_MIN, "max": AGG_MAX}
+AGG_IDENTITY = {AGG_SUM: 0.0, AGG_MIN: float("inf"), AGG_MAX: float("-inf")}
+
+
+def groupby_accum(keys: ColumnRef, vals: list, agg_op: int, key_min: int,
                   n_slots: int, sums: int, rowcnt: int, counts: int) -> None:
     ensure_ready()
     arr = (ct.c_void_p * max(len(vals), 1))(*[v.handle for v in vals])
-    _check(load().hf_groupby_accum(keys.handle, arr, len(vals), key_min, n_slots,
-                                   sums, rowcnt, counts), "hf_groupby_accum")
+    _check(load().hf_groupby_accum(keys.handle, arr, len(vals), agg_op,
+                                   key_min, n_slots, sums, rowcnt, counts),
+           "hf_groupby_accum")
+
+
+def fill_f64(dptr: int, value: float, n: int) -> None:
+    ensure_ready()
+    _check(load().hf_fill_f64(dptr, value, n), "hf_fill_f64")
+
+
+def fixup_empty(val: ColumnRef, cnt: ColumnRef) -> ColumnRef:
+    ensure_ready()
+    out = ct.c_void_p()
+    _check(load().hf_fixup_empty(val.handle, cnt.handle, ct.byref(out)),
+           "hf_fixup_empty")
+    return _wrap(out, val.length, HF_FLOAT64)
 
 
 def groupby_compact(sums: int, rowcnt: int, counts: int, nvals: int,

@@ -111,7 +111,8 @@ def _gb_via_cabi(keys, vals_dict, want_counts):
     lib.memset_raw(rowcnt, 0, 8 * n_slots)
     if want_counts:
         lib.memset_raw(counts, 0, 8 * nv * n_slots)
-    lib.groupby_accum(key_col, val_cols, kmin, n_slots, sums, rowcnt, counts)
+    lib.groupby_accum(key_col, val_cols, lib.AGG_SUM, kmin, n_slots, sums,
+                      rowcnt, counts)
     keys_out, sums_out, counts_out, n = lib.groupby_compact(
         sums, rowcnt, counts, nv, kmin, n_slots)
     got_keys = lib.get(keys_out)
@@ -154,7 +155,7 @@ def test_groupby_out_of_range_key_detected():
     rowcnt = lib.alloc_raw(8 * 10)
     lib.memset_raw(sums, 0, 80)
     lib.memset_raw(rowcnt, 0, 80)
-    lib.groupby_accum(key_col, [val_col], 0, 10, sums, rowcnt, 0)
+    lib.groupby_accum(key_col, [val_col], lib.AGG_SUM, 0, 10, sums, rowcnt, 0)
     with pytest.raises(lib.HfError, match="outside"):
         lib.groupby_compact(sums, rowcnt, 0, 1, 0, 10)
     lib.free_raw(sums)
