@@ -47,16 +47,25 @@ def init_from_env(backend=None, gpu: bool = True):
     r = int(os.environ.get("RANK", "0"))
     local = int(os.environ.get("LOCAL_RANK", str(r)))
     if backend is None:
+        backend = os.environ.get("MODIN_AMD_DIST_BACKEND")
+    if backend is None:
         backend = "nccl" if (gpu and torch.cuda.is_available()) else "gloo"
     if not dist.is_initialized():
         dist.init_process_group(backend=backend)
     _state.update(active=True, rank=r, world=world, backend=backend)
-    if backend == "nccl":
-        torch.cuda.set_device(local)
-        _state["device"] = f"cuda:{local}"
-        os.environ["MODIN_AMD_GPU"] = str(local)
+    if gpu and torch.cuda.is_available():
+        # oversubscription fallback (testing N ranks on fewer GPUs)
+        dev = local % torch.cuda.device_count()
+        torch.cuda.set_device(dev)
+        # small collectives run on the backend's native device; the groupby
+        # TABLE always lives on the GPU (hipframe kernels write it) and is
+        # CPU-bounced for gloo collectives in maybe_allreduce_table
+        _state["device"] = f"cuda:{dev}" if backend == "nccl" else "cpu"
+        _state["table_device"] = f"cuda:{dev}"
+        os.environ["MODIN_AMD_GPU"] = str(dev)
     else:
         _state["device"] = "cpu"
+        _state["table_device"] = "cpu"
     return True
 
 
@@ -155,13 +164,14 @@ def alloc_table_torch(nvals: int, n_slots: int, want_counts: bool,
     Returns (keepalive, sums_ptr, rowcnt_ptr, counts_ptr).
     """
     import torch
-    dev = _state["device"]
+    dev = _state.get("table_device") or _state["device"]
     sums = torch.full((nvals * n_slots,), init, dtype=torch.float64,
                       device=dev)
     rowcnt = torch.zeros(n_slots, dtype=torch.int64, device=dev)
     counts = (torch.zeros(nvals * n_slots, dtype=torch.int64, device=dev)
               if want_counts else None)
-    torch.cuda.synchronize()
+    if dev != "cpu":
+        torch.cuda.synchronize()
     keep = (sums, rowcnt, counts)
     return keep, sums.data_ptr(), rowcnt.data_ptr(), \
         (counts.data_ptr() if counts is not None else 0)
@@ -173,16 +183,26 @@ def maybe_allreduce_table(table) -> None:
         return
     import torch
     import torch.distributed as dist
-    if _state["backend"] == "nccl":
+    sums, rowcnt, counts = table._torch_tensors
+    if getattr(sums, "is_cuda", False):
         from .core import lib
         lib.sync()  # hipframe-stream accumulation must be visible to RCCL
-    sums, rowcnt, counts = table._torch_tensors
     agg_op = getattr(table, "agg_op", 0)
     op = (dist.ReduceOp.MIN if agg_op == 1
           else dist.ReduceOp.MAX if agg_op == 2 else dist.ReduceOp.SUM)
-    dist.all_reduce(sums, op=op)
-    dist.all_reduce(rowcnt)
+
+    def reduce_(t, o):
+        if t.is_cuda and _state["backend"] != "nccl":
+            # gloo cannot reduce device tensors: bounce via host (test path)
+            h = t.cpu()
+            dist.all_reduce(h, op=o)
+            t.copy_(h)
+        else:
+            dist.all_reduce(t, op=o)
+
+    reduce_(sums, op)
+    reduce_(rowcnt, dist.ReduceOp.SUM)
     if counts is not None:
-        dist.all_reduce(counts)
-    if _state["backend"] == "nccl":
+        reduce_(counts, dist.ReduceOp.SUM)
+    if sums.is_cuda:
         torch.cuda.synchronize()  # collective must land before compaction reads
