@@ -237,6 +237,18 @@ int hf_filter_apply(const hf_filterplan* plan, const hf_col* col,
 int hf_filter_iota(const hf_filterplan* plan, int64_t base, hf_col** out);
 int hf_filter_plan_free(hf_filterplan* plan);
 
+/* ---- Sort: stable permutation by an int64 key column ----
+ * Device form of PandasDataframe.sort_by (dataframe.py:2742 ->
+ * _apply_func_to_range_partitioning), restricted to one
+ * int64 key with bounded range (key range <= 2^27) this round: an LSD radix
+ * sort over packed (u32 shifted-key, u32 origin) pairs, 8-bit digits,
+ * wave-private tiles (in-order lanes make each pass stable, so the result
+ * equals pandas sort_values(kind="stable")).  ascending=0 sorts by
+ * (key_max - key), which preserves pandas' stable-descending tie order.
+ * Returns the permutation as an int64 column of original positions; the
+ * caller gathers payload columns and the index with hf_gather. */
+int hf_sort_perm(const hf_col* keys, int ascending, hf_col** out_perm);
+
 /* ---- profiling (bench.py roofline leg) ----
  * When enabled, every kernel launch is bracketed by HIP events on the module
  * stream; hf_kernel_stats returns the accumulated count and total ms for the

@@ -233,6 +233,31 @@ class HipDataframe:
             return DeviceBlock(out, block.length)
         return self.map(block_fn)
 
+    # ---- sort (PandasDataframe.sort_by device form, dataframe.py:2742;
+    #      SURVEY §8f.2): stable radix permutation + column gathers ----
+    def sort_rows(self, by: str, ascending: bool = True) -> "HipDataframe":
+        if by not in self.columns:
+            raise lib.HfError(f"sort_values: column {by!r} missing")
+        if not isinstance(self._index, pandas.RangeIndex) or \
+                self._index.start != 0 or self._index.step != 1:
+            raise lib.HfError(
+                "sort_values: only RangeIndex frames this round")
+
+        def concat_col(name):
+            cols = [p.block().columns[name] for p in self._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        kcol = concat_col(by)
+        perm = lib.sort_perm(kcol, ascending)
+        cols = {}
+        for name in self.columns:
+            src = kcol if name == by else concat_col(name)
+            cols[name] = lib.gather(src, perm)
+        n = perm.length
+        part = HipDataframePartition(DeviceBlock(cols, n))
+        return HipDataframe([part], DeviceIndex(perm, name=None),
+                            self.columns, [n], self.dtypes)
+
     # ---- dropna mask: AND of per-column notna (pandas dropna(how="any")) ----
     def notna_all_mask(self) -> "HipDataframe":
         def block_fn(block: DeviceBlock) -> DeviceBlock:

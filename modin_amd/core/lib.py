@@ -106,6 +106,8 @@ def load() -> ct.CDLL:
                                             ct.c_int64, ct.c_size_t,
                                             ct.c_size_t, ct.c_size_t]),
             "hf_fill_f64": (ct.c_int, [ct.c_size_t, ct.c_double, ct.c_int64]),
+            "hf_sort_perm": (ct.c_int, [ct.c_void_p, ct.c_int,
+                                        ct.POINTER(ct.c_void_p)]),
             "hf_fixup_empty": (ct.c_int, [ct.c_void_p, ct.c_void_p,
                                           ct.POINTER(ct.c_void_p)]),
             "hf_groupby_compact": (ct.c_int, [ct.c_size_t, ct.c_size_t, ct.c_size_t,
@@ -158,7 +160,7 @@ def exported_symbols():
         "hf_col_dtype", "hf_col_dptr", "hf_alloc_raw", "hf_free_raw",
         "hf_memset_raw", "hf_map_scalar", "hf_map_scalar_i64", "hf_binary",
         "hf_reduce", "hf_groupby_accum", "hf_groupby_compact", "hf_fill_f64",
-        "hf_fixup_empty",
+        "hf_fixup_empty", "hf_sort_perm",
         "hf_col_concat", "hf_join_build", "hf_join_free", "hf_join_probe",
         "hf_gather", "hf_compare_scalar", "hf_filter_plan", "hf_filter_apply",
         "hf_filter_iota", "hf_filter_plan_free", "hf_profiling",
@@ -488,6 +490,14 @@ def filter_iota(plan: FilterPlan, base: int) -> ColumnRef:
     _check(load().hf_filter_iota(plan.handle, base, ct.byref(out)),
            "hf_filter_iota")
     return _wrap(out, plan.n_kept, HF_INT64)
+
+
+def sort_perm(keys: ColumnRef, ascending: bool = True) -> ColumnRef:
+    ensure_ready()
+    out = ct.c_void_p()
+    _check(load().hf_sort_perm(keys.handle, 1 if ascending else 0,
+                               ct.byref(out)), "hf_sort_perm")
+    return _wrap(out, keys.length, HF_INT64)
 
 
 def gather(col: ColumnRef, idx: ColumnRef) -> ColumnRef:

@@ -1207,6 +1207,117 @@ __global__ void __launch_bounds__(BLOCK) k_probe_emit(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Sort kernels — stable LSD radix over (u32 key, u32 idx) pairs.
+// A "tile" is 64 lanes x SORT_RPT rows owned by ONE wave: lanes are in
+// order and the per-step leader loop ranks same-digit lanes by lane id, so
+// every pass is stable without cross-wave coordination.
+// ---------------------------------------------------------------------------
+
+constexpr int SORT_RPT = 32;                 // rows per lane per tile
+constexpr int SORT_TILE = 64 * SORT_RPT;     // 2048 rows per wave-tile
+constexpr int SORT_WPB = 4;                  // waves per block
+
+__global__ void __launch_bounds__(BLOCK) k_sort_pack(
+    const int64_t* __restrict__ keys, int64_t n, int64_t key_min,
+    int64_t key_span, int ascending, unsigned long long* __restrict__ pairs) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const unsigned k = (unsigned)(keys[i] - key_min);
+    const unsigned kk = ascending ? k : (unsigned)key_span - k;
+    pairs[i] = ((unsigned long long)(unsigned)i << 32) | kk;
+  }
+}
+
+// per-wave-tile digit histogram -> C[tile][256] (row-major, coalesced)
+__global__ void __launch_bounds__(BLOCK) k_sort_count(
+    const unsigned long long* __restrict__ pairs, int64_t n, int shift,
+    unsigned* __restrict__ C, int64_t ntiles) {
+  __shared__ unsigned hist[SORT_WPB][256];
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int64_t wtiles_per_block = SORT_WPB;
+  for (int64_t tile = (int64_t)blockIdx.x * wtiles_per_block + wave;
+       tile < ntiles; tile += (int64_t)gridDim.x * wtiles_per_block) {
+    for (int d = lane; d < 256; d += 64) hist[wave][d] = 0;
+    __builtin_amdgcn_wave_barrier();
+    const int64_t t0 = tile * SORT_TILE;
+#pragma unroll 4
+    for (int j = 0; j < SORT_RPT; ++j) {
+      const int64_t row = t0 + (int64_t)j * 64 + lane;
+      if (row < n) {
+        const unsigned d = ((unsigned)pairs[row] >> shift) & 255u;
+        atomicAdd(&hist[wave][d], 1u);  // LDS, same-wave only
+      }
+    }
+    __builtin_amdgcn_wave_barrier();
+    for (int d = lane; d < 256; d += 64) C[tile * 256 + d] = hist[wave][d];
+  }
+}
+
+// 32x32 tiled transpose: C[ntiles][256] -> CT[256][ntiles]
+__global__ void __launch_bounds__(1024) k_transpose256(
+    const unsigned* __restrict__ C, unsigned* __restrict__ CT,
+    int64_t ntiles) {
+  __shared__ unsigned t[32][33];
+  const int64_t tx0 = (int64_t)blockIdx.x * 32;       // tile-index block
+  const int dy = (int)(blockIdx.y * 32);              // digit block (0..224)
+  const int lx = threadIdx.x & 31, ly = threadIdx.x >> 5;  // 32x32 threads
+  const int64_t src_row = tx0 + ly;
+  if (src_row < ntiles) t[ly][lx] = C[src_row * 256 + dy + lx];
+  __syncthreads();
+  const int64_t dst_col = tx0 + lx;
+  if (dst_col < ntiles) CT[(int64_t)(dy + ly) * ntiles + dst_col] = t[lx][ly];
+}
+
+// stable scatter: each wave re-walks its tile in order; per step the leader
+// loop hands out in-order ranks per digit from the wave's offset registers
+__global__ void __launch_bounds__(BLOCK) k_sort_scatter(
+    const unsigned long long* __restrict__ pairs, int64_t n, int shift,
+    const unsigned long long* __restrict__ offs,   // [256][ntiles] exclusive
+    int64_t ntiles, unsigned long long* __restrict__ out) {
+  __shared__ unsigned long long base[SORT_WPB][256];
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  for (int64_t tile = (int64_t)blockIdx.x * SORT_WPB + wave; tile < ntiles;
+       tile += (int64_t)gridDim.x * SORT_WPB) {
+    for (int d = lane; d < 256; d += 64)
+      base[wave][d] = offs[(int64_t)d * ntiles + tile];
+    __builtin_amdgcn_wave_barrier();
+    const int64_t t0 = tile * SORT_TILE;
+    for (int j = 0; j < SORT_RPT; ++j) {
+      const int64_t row = t0 + (int64_t)j * 64 + lane;
+      const bool valid = row < n;
+      const unsigned long long p = valid ? pairs[row] : 0;
+      const unsigned d = ((unsigned)p >> shift) & 255u;
+      // leader loop over the distinct digits present in this step
+      unsigned long long exec = __ballot(valid);
+      unsigned long long pos = 0;
+      while (exec) {
+        const int leader = __ffsll((long long)exec) - 1;
+        const unsigned dl = (unsigned)__shfl((int)d, leader);
+        const unsigned long long members = __ballot(valid && d == dl);
+        if (valid && d == dl) {
+          const unsigned rank =
+              (unsigned)__popcll(members & ((1ULL << lane) - 1ULL));
+          pos = base[wave][dl] + rank;
+        }
+        if (lane == leader) base[wave][dl] += __popcll(members);
+        exec &= ~members;
+      }
+      if (valid) out[pos] = p;
+      __builtin_amdgcn_wave_barrier();
+    }
+  }
+}
+
+__global__ void __launch_bounds__(BLOCK) k_sort_unpack(
+    const unsigned long long* __restrict__ pairs, int64_t n,
+    int64_t* __restrict__ perm) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) perm[i] = (int64_t)(pairs[i] >> 32);
+}
+
 __global__ void __launch_bounds__(BLOCK) k_fill_f64(double* __restrict__ p,
                                                     double v, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -2308,6 +2419,88 @@ int hf_gather(const hf_col* col, const hf_col* idx, hf_col** out) {
                          (const int64_t*)idx->dptr, (int64_t*)(*out)->dptr, n);
   });
   if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  return rc;
+}
+
+int hf_sort_perm(const hf_col* keys, int ascending, hf_col** out_perm) {
+  HF_NEED_INIT("hf_sort_perm");
+  if (!keys || !out_perm) return set_err(HF_ERR_ARG, "hf_sort_perm", "null");
+  if (keys->dtype != HF_INT64)
+    return set_err(HF_ERR_ARG, "hf_sort_perm", "sort key must be int64");
+  const int64_t n = keys->len;
+  int rc = hf_col_alloc(n, HF_INT64, out_perm);
+  if (rc != HF_OK) return rc;
+  if (n == 0) return HF_OK;
+  if (n >= (1LL << 31))
+    return set_err(HF_ERR_UNSUPPORTED, "hf_sort_perm",
+                   "partition too large (u32 origin indices)");
+  hf_reduce_result r;
+  rc = hf_reduce(keys, &r);
+  if (rc != HF_OK) return rc;
+  const int64_t key_min = r.imn, span = r.imx - r.imn;
+  if (span >= (1LL << 27))
+    return set_err(HF_ERR_UNSUPPORTED, "hf_sort_perm",
+                   "key range too large for the bounded-range radix sort "
+                   "(general 64-bit keys are a later round)");
+  int bits = 0;
+  while ((span >> bits) != 0) ++bits;
+  const int passes = (bits + 7) / 8;
+  const int64_t ntiles = (n + SORT_TILE - 1) / SORT_TILE;
+  const int64_t L = ntiles * 256;
+  unsigned long long *bufA = nullptr, *bufB = nullptr, *offs = nullptr;
+  unsigned *C = nullptr, *CT = nullptr;
+  int64_t* scan_tiles = nullptr;
+  const int64_t scan_tiles_n = (L + JOIN_TILE - 1) / JOIN_TILE;
+  HF_HIP("hf_sort_perm", dev_alloc((void**)&bufA, n * 8, g.stream));
+  HF_HIP("hf_sort_perm", dev_alloc((void**)&bufB, n * 8, g.stream));
+  HF_HIP("hf_sort_perm", dev_alloc((void**)&C, L * 4, g.stream));
+  HF_HIP("hf_sort_perm", dev_alloc((void**)&CT, L * 4, g.stream));
+  HF_HIP("hf_sort_perm", dev_alloc((void**)&offs, (L + 1) * 8, g.stream));
+  HF_HIP("hf_sort_perm", dev_alloc((void**)&scan_tiles, scan_tiles_n * 8,
+                                   g.stream));
+  int64_t* d_total = (int64_t*)((char*)g.d_scratch + SCRATCH_NGROUPS);
+  rc = timed_launch("sort_pack", [&] {
+    hipLaunchKernelGGL(k_sort_pack, dim3((uint32_t)grid_for(n)), dim3(BLOCK),
+                       0, g.stream, (const int64_t*)keys->dptr, n, key_min,
+                       span, ascending, bufA);
+  });
+  unsigned long long* cur = bufA;
+  unsigned long long* alt = bufB;
+  const uint32_t wgrid =
+      (uint32_t)std::min<int64_t>((ntiles + SORT_WPB - 1) / SORT_WPB, 2048);
+  for (int p = 0; p < passes && rc == HF_OK; ++p) {
+    const int shift = 8 * p;
+    rc = timed_launch("sort_pass", [&] {
+      hipLaunchKernelGGL(k_sort_count, dim3(wgrid), dim3(BLOCK), 0, g.stream,
+                         cur, n, shift, C, ntiles);
+      hipLaunchKernelGGL(k_transpose256,
+                         dim3((uint32_t)((ntiles + 31) / 32), 8), dim3(1024),
+                         0, g.stream, C, CT, ntiles);
+      hipLaunchKernelGGL(k_tile_sums_u32, dim3((uint32_t)scan_tiles_n),
+                         dim3(BLOCK), 0, g.stream, CT, L, scan_tiles);
+      hipLaunchKernelGGL(k_compact_scan, dim3(1), dim3(1024), 0, g.stream,
+                         scan_tiles, scan_tiles_n, d_total);
+      hipLaunchKernelGGL(k_scan_apply_u32, dim3((uint32_t)scan_tiles_n),
+                         dim3(BLOCK), 0, g.stream, CT, L, scan_tiles, d_total,
+                         offs);
+      hipLaunchKernelGGL(k_sort_scatter, dim3(wgrid), dim3(BLOCK), 0, g.stream,
+                         cur, n, shift, offs, ntiles, alt);
+    });
+    std::swap(cur, alt);
+  }
+  if (rc == HF_OK)
+    rc = timed_launch("sort_unpack", [&] {
+      hipLaunchKernelGGL(k_sort_unpack, dim3((uint32_t)grid_for(n)),
+                         dim3(BLOCK), 0, g.stream, cur,
+                         n, (int64_t*)(*out_perm)->dptr);
+    });
+  dev_free(bufA, g.stream);
+  dev_free(bufB, g.stream);
+  dev_free(C, g.stream);
+  dev_free(CT, g.stream);
+  dev_free(offs, g.stream);
+  dev_free(scan_tiles, g.stream);
+  if (rc != HF_OK) { hf_col_free(*out_perm); *out_perm = nullptr; }
   return rc;
 }
 

@@ -222,6 +222,14 @@ class DataFrame(_HipPandasBase):
             )
         raise lib.HfError("only column selection / boolean masks are supported")
 
+    def sort_values(self, by: str, ascending: bool = True,
+                    kind: str = "stable"):
+        """Always stable (equals pandas sort_values(kind='stable'), a
+        stronger guarantee than the default quicksort)."""
+        return DataFrame(
+            query_compiler=self._query_compiler.sort_rows_by_column_values(
+                by, ascending))
+
     def merge(self, other: "DataFrame", on: str, how: str = "inner"):
         """Inner merge on an int64 key column (modin/pandas API ->
         qc.merge -> broadcast-right device join)."""

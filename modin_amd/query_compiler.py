@@ -129,6 +129,13 @@ class HipQueryCompiler:
         return self.__constructor__(
             self._modin_frame.filter_rows(mask_qc._modin_frame))
 
+    # ---- sort (reference qc.sort_rows_by_column_values) ----
+    def sort_rows_by_column_values(self, by: str,
+                                   ascending: bool = True
+                                   ) -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.sort_rows(by, ascending))
+
     # ---- merge (query_compiler merge -> MergeImpl.row_axis_merge,
     #      storage_formats/pandas/merge.py:104) ----
     def merge(self, right: "HipQueryCompiler", on: str,

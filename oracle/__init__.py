@@ -24,5 +24,6 @@ from .ops import (  # noqa: F401
     map_op,
     partitioned_groupby_agg,
     reduce_op,
+    sort_perm,
     split_row_counts,
 )

@@ -257,6 +257,42 @@ def gen_filter_cases(mpd, rng):
     return cases
 
 
+def gen_sort_cases(mpd, rng):
+    import pandas
+    cases = {}
+    n = 6000
+    k = rng.integers(0, 50, n).astype(np.int64)  # heavy duplication: tie order
+    v = rng.random(n)
+    i = rng.integers(-5, 5, n).astype(np.int64)
+    mdf = mpd.DataFrame({"k": k, "v": v, "i": i})
+    pdf = pandas.DataFrame({"k": k, "v": v, "i": i})
+    arrays = {"in_k": k, "in_v": v, "in_i": i}
+    for tag, asc in [("asc", True), ("desc", False)]:
+        mres = mdf.sort_values("k", ascending=asc, kind="stable")._to_pandas()
+        pres = pdf.sort_values("k", ascending=asc, kind="stable")
+        np.testing.assert_array_equal(mres.index.to_numpy(),
+                                      pres.index.to_numpy())
+        for c in ("k", "v", "i"):
+            np.testing.assert_array_equal(mres[c].to_numpy(),
+                                          pres[c].to_numpy())
+        arrays[f"out_idx_{tag}"] = pres.index.to_numpy().astype(np.int64)
+        for c in ("k", "v", "i"):
+            arrays[f"out_{tag}_{c}"] = pres[c].to_numpy()
+    # negative keys
+    kn = rng.integers(-1000, 1000, n).astype(np.int64)
+    mdf2 = mpd.DataFrame({"k": kn, "v": v})
+    pdf2 = pandas.DataFrame({"k": kn, "v": v})
+    mres = mdf2.sort_values("k", kind="stable")._to_pandas()
+    pres = pdf2.sort_values("k", kind="stable")
+    np.testing.assert_array_equal(mres.index.to_numpy(), pres.index.to_numpy())
+    arrays["in_kn"] = kn
+    arrays["out_neg_idx"] = pres.index.to_numpy().astype(np.int64)
+    arrays["out_neg_k"] = pres["k"].to_numpy()
+    arrays["out_neg_v"] = pres["v"].to_numpy()
+    cases["srt_basic"] = arrays
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -267,6 +303,7 @@ def main():
     all_cases.update(gen_map_binary_cases(mpd, rng))
     all_cases.update(gen_merge_cases(mpd, rng))
     all_cases.update(gen_filter_cases(mpd, rng))
+    all_cases.update(gen_sort_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)

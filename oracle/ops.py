@@ -203,6 +203,21 @@ def filter_rows(mask: np.ndarray, cols: dict):
 
 
 # ---------------------------------------------------------------------------
+# Sort — PandasDataframe.sort_by (dataframe.py:2742; stable order ==
+# pandas sort_values(kind="stable"); descending keeps ties in original
+# order, i.e. stable on the negated key).
+# ---------------------------------------------------------------------------
+
+def sort_perm(keys: np.ndarray, ascending: bool = True) -> np.ndarray:
+    keys = np.asarray(keys)
+    if keys.size == 0:
+        return np.empty(0, np.int64)
+    if ascending:
+        return np.argsort(keys, kind="stable").astype(np.int64)
+    return np.argsort(keys.max() - keys, kind="stable").astype(np.int64)
+
+
+# ---------------------------------------------------------------------------
 # Merge — MergeImpl.row_axis_merge (storage_formats/pandas/merge.py:104-178):
 # materialize the right frame once (combine(), dataframe.py:2918), broadcast
 # to every left partition, per-partition pandas.merge(how="inner").

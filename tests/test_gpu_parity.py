@@ -293,6 +293,44 @@ def test_dropna_and_mask_algebra(npartitions):
                                   pdf["v"].isna().to_numpy())
 
 
+def test_sort_vs_golden(npartitions):
+    g = load_golden("srt_basic")
+    df = mpd.DataFrame({"k": g["in_k"], "v": g["in_v"], "i": g["in_i"]})
+    for tag, asc in [("asc", True), ("desc", False)]:
+        out = df.sort_values("k", ascending=asc).to_pandas()
+        np.testing.assert_array_equal(out.index.to_numpy(),
+                                      g[f"out_idx_{tag}"], err_msg=tag)
+        for c in ("k", "v", "i"):
+            np.testing.assert_array_equal(out[c].to_numpy(),
+                                          g[f"out_{tag}_{c}"],
+                                          err_msg=f"{tag}/{c}")
+            assert out[c].dtype == g[f"out_{tag}_{c}"].dtype
+    dfn = mpd.DataFrame({"k": g["in_kn"], "v": g["in_v"]})
+    out = dfn.sort_values("k").to_pandas()
+    np.testing.assert_array_equal(out.index.to_numpy(), g["out_neg_idx"])
+    np.testing.assert_array_equal(out["k"].to_numpy(), g["out_neg_k"])
+
+
+def test_sort_property_large():
+    """Stability + correctness at a multi-pass size (3 radix passes) against
+    the oracle, exact."""
+    rng = np.random.default_rng(50)
+    n = 2_000_000
+    k = rng.integers(0, 10**6, n).astype(np.int64)
+    v = rng.random(n)
+    df = mpd.DataFrame({"k": k, "v": v})
+    out = df.sort_values("k").to_pandas()
+    perm = oracle.sort_perm(k)
+    np.testing.assert_array_equal(out.index.to_numpy(), perm)
+    np.testing.assert_array_equal(out["k"].to_numpy(), k[perm])
+    np.testing.assert_array_equal(out["v"].to_numpy(), v[perm])
+    # single-key edge (zero radix passes)
+    one = mpd.DataFrame({"k": np.full(1000, 7, dtype=np.int64),
+                         "v": rng.random(1000)})
+    res = one.sort_values("k").to_pandas()
+    np.testing.assert_array_equal(res.index.to_numpy(), np.arange(1000))
+
+
 def test_native_extension_is_loaded():
     """Guard against a silent eager/pandas fallback: the in-tree .so must be
     mapped into this process."""

@@ -141,6 +141,20 @@ def test_filter_vs_golden():
                                           err_msg=f"{tag}/{c}")
 
 
+def test_sort_vs_golden():
+    g = load_golden("srt_basic")
+    for tag, asc in [("asc", True), ("desc", False)]:
+        perm = oracle.sort_perm(g["in_k"], ascending=asc)
+        np.testing.assert_array_equal(perm, g[f"out_idx_{tag}"], err_msg=tag)
+        for c in ("k", "v", "i"):
+            np.testing.assert_array_equal(g[f"in_{c}"][perm],
+                                          g[f"out_{tag}_{c}"],
+                                          err_msg=f"{tag}/{c}")
+    perm = oracle.sort_perm(g["in_kn"])
+    np.testing.assert_array_equal(perm, g["out_neg_idx"])
+    np.testing.assert_array_equal(g["in_kn"][perm], g["out_neg_k"])
+
+
 def test_groupby_empty():
     keys, out = oracle.groupby_agg(np.empty(0, np.int64), {"v": np.empty(0)}, "sum")
     assert keys.size == 0 and out["v"].size == 0
