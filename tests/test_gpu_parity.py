@@ -331,6 +331,38 @@ def test_sort_property_large():
     np.testing.assert_array_equal(res.index.to_numpy(), np.arange(1000))
 
 
+def test_pipeline_filter_merge_groupby_sort(npartitions):
+    """Integration chain: filter -> merge -> groupby -> sort, checked against
+    the same chain on pandas."""
+    rng = np.random.default_rng(60)
+    n = 120_000
+    left = {"k": rng.integers(0, 300, n).astype(np.int64),
+            "v": rng.random(n)}
+    right = {"k": rng.integers(0, 300, 700).astype(np.int64),
+             "w": rng.random(700)}
+    df = mpd.DataFrame(left)
+    rf = mpd.DataFrame(right)
+    out = (df[df["v"] > 0.25]
+           .merge(rf, on="k")
+           .groupby("k").sum()
+           .to_pandas())
+    pdf = pandas.DataFrame(left)
+    prf = pandas.DataFrame(right)
+    expect = (pdf[pdf["v"] > 0.25]
+              .merge(prf, on="k")
+              .groupby("k").sum())
+    np.testing.assert_array_equal(out.index.to_numpy(),
+                                  expect.index.to_numpy())
+    for c in ("v", "w"):
+        np.testing.assert_allclose(out[c].to_numpy(), expect[c].to_numpy(),
+                                   rtol=RTOL, atol=1e-9)
+    # and a sort on top of a merge result
+    m = df.merge(rf, on="k").sort_values("k").to_pandas()
+    pm = pdf.merge(prf, on="k").sort_values("k", kind="stable")
+    np.testing.assert_array_equal(m["k"].to_numpy(), pm["k"].to_numpy())
+    np.testing.assert_array_equal(m["w"].to_numpy(), pm["w"].to_numpy())
+
+
 def test_native_extension_is_loaded():
     """Guard against a silent eager/pandas fallback: the in-tree .so must be
     mapped into this process."""

@@ -113,6 +113,30 @@ def main():
         HipQueryCompiler.sum(qc)
     report("tree_reduce_sum (df.sum)", "reduce_f64", ne, 8, do_sum)
 
+    # sort_values: 1e9 rows, 1e6-key range (3 radix passes + gathers)
+    del qc, block
+    import oracle  # noqa: F401  (only to mirror bench imports)
+    rng2 = np.random.default_rng(43)
+    n = args.red_rows
+    kcol = lib.put(rng2.integers(0, 10**6, n).astype(np.int64))
+    lib.reduce(kcol)  # key-range metadata (cached)
+    lib.sync()
+    lib.kernel_stats_reset()
+    t0 = time.perf_counter()
+    for _ in range(3):
+        perm = lib.sort_perm(kcol, True)
+    lib.sync()
+    dt = (time.perf_counter() - t0) / 3
+    ks = {}
+    for k in ("sort_pack", "sort_pass", "sort_unpack"):
+        nl, ms = lib.kernel_stats(k)
+        if nl:
+            ks[k] = [nl, round(ms / 3, 3)]
+    line = {"op": "sort_perm (1e9 rows, 1e6-key range)", "elems": n,
+            "ms_per_op": dt * 1e3, "rows_per_s": n / dt, "kernels": ks}
+    results.append(line)
+    print(json.dumps(line), flush=True)
+
     with open(os.path.join(REPO, "gpurun_out", "opbench.json"), "w") as f:
         for line in results:
             f.write(json.dumps(line) + "\n")
