@@ -163,7 +163,17 @@ class HipDataframe:
             kmin, n_slots = r.imn, r.imx - r.imn + 1
         else:
             kmin, n_slots = 0, 1
-        j = lib.join_build(rkeys, rvals, kmin, n_slots)
+        # cache the build side on the right frame: its columns are immutable,
+        # so repeated merges with the same right frame skip hist/scan/fill
+        # (the lazy-metadata pattern again; a real broadcast join caches its
+        # build side)
+        cache_key = (on, tuple(right_names), kmin, n_slots)
+        cached = getattr(other, "_join_build_cache", None)
+        if cached is not None and cached[0] == cache_key:
+            j = cached[1]
+        else:
+            j = lib.join_build(rkeys, rvals, kmin, n_slots)
+            other._join_build_cache = (cache_key, j)
 
         out_parts, lengths = [], []
         for p in self._partitions:
