@@ -162,10 +162,11 @@ def main():
     achieved = (ALG_BYTES_PER_ROW * rows_per_launch) / (avg_ms / 1e3) \
         if n_launch and avg_ms > 0 else 0.0
     # HBM traffic per launch of the dominant kernel, from rocprofv3 PMC
-    # (profiles/r01/pmc_traffic.txt: FETCH_SIZE x2-corrected + WRITE_SIZE =
-    # 28.05 B/row for gb_scatter — reads exactly algorithmic, writes 1.2x
-    # from chunk-boundary partial lines).  Env override wins.
-    PMC_TRAFFIC_B_PER_ROW = {"gb_scatter": 28.05}
+    # (profiles/r01c/pmc_traffic.txt: FETCH_SIZE x2-corrected + WRITE_SIZE =
+    # 27.0 B/row for gb_scatter — reads exactly the algorithmic 16 B/row,
+    # writes 1.1x of the 10 B/row payload from chunk-boundary partial
+    # lines).  Env override wins.
+    PMC_TRAFFIC_B_PER_ROW = {"gb_scatter": 27.0}
     traffic_env = os.environ.get("HF_TRAFFIC_BYTES_PER_LAUNCH")
     if not traffic_env and dom in PMC_TRAFFIC_B_PER_ROW:
         traffic_env = PMC_TRAFFIC_B_PER_ROW[dom] * rows_per_launch
