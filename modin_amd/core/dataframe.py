@@ -243,6 +243,22 @@ class HipDataframe:
             return DeviceBlock(out, block.length)
         return self.map(block_fn)
 
+    # ---- row concat (PartitionManager.concat device form,
+    #      partition_manager.py:943: stack the partition lists; no device
+    #      copy — partitions are immutable) ----
+    def concat_rows(self, others: list) -> "HipDataframe":
+        cols = list(self.columns)
+        for o in others:
+            if list(o.columns) != cols:
+                raise lib.HfError(
+                    "concat: all frames must share the same columns this "
+                    "round (NaN-fill alignment is a later round)")
+        frames = [self] + list(others)
+        parts = [p for f in frames for p in f._partitions]
+        lengths = [ln for f in frames for ln in f._row_lengths]
+        idx = pandas.Index(np.concatenate([np.asarray(f.index) for f in frames]))
+        return HipDataframe(parts, idx, cols, lengths, self.dtypes)
+
     # ---- sort (PandasDataframe.sort_by device form, dataframe.py:2742;
     #      SURVEY §8f.2): stable radix permutation + column gathers ----
     def sort_rows(self, by: str, ascending: bool = True) -> "HipDataframe":

@@ -18,9 +18,23 @@ import numpy as np
 import pandas
 
 from ..core import lib
+from ..core.lib import HfError as HfErrorProxy
 from ..query_compiler import HipQueryCompiler
 
-__all__ = ["DataFrame", "Series", "from_pandas"]
+__all__ = ["DataFrame", "Series", "concat", "from_pandas"]
+
+
+def concat(objs, ignore_index: bool = False):
+    """pandas.concat(axis=0) over DataFrames with identical columns."""
+    objs = list(objs)
+    if not objs:
+        raise HfErrorProxy("concat of empty list")
+    out = DataFrame(query_compiler=objs[0]._query_compiler.concat(
+        [o._query_compiler for o in objs[1:]]))
+    if ignore_index:
+        out._query_compiler._modin_frame._index = __import__(
+            "pandas").RangeIndex(len(out))
+    return out
 
 
 def from_pandas(df: pandas.DataFrame) -> "DataFrame":

@@ -129,6 +129,11 @@ class HipQueryCompiler:
         return self.__constructor__(
             self._modin_frame.filter_rows(mask_qc._modin_frame))
 
+    # ---- concat (reference qc.concat -> PartitionManager.concat :943) ----
+    def concat(self, others: list) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.concat_rows(
+            [o._modin_frame for o in others]))
+
     # ---- sort (reference qc.sort_rows_by_column_values) ----
     def sort_rows_by_column_values(self, by: str,
                                    ascending: bool = True

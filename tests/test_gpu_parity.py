@@ -363,6 +363,25 @@ def test_pipeline_filter_merge_groupby_sort(npartitions):
     np.testing.assert_array_equal(m["w"].to_numpy(), pm["w"].to_numpy())
 
 
+def test_concat_vs_pandas(npartitions):
+    rng = np.random.default_rng(70)
+    a = {"k": rng.integers(0, 9, 1000).astype(np.int64), "v": rng.random(1000)}
+    b = {"k": rng.integers(0, 9, 500).astype(np.int64), "v": rng.random(500)}
+    out = mpd.concat([mpd.DataFrame(a), mpd.DataFrame(b)]).to_pandas()
+    expect = pandas.concat([pandas.DataFrame(a), pandas.DataFrame(b)])
+    pandas.testing.assert_frame_equal(out, expect)
+    out2 = mpd.concat([mpd.DataFrame(a), mpd.DataFrame(b)],
+                      ignore_index=True)
+    expect2 = pandas.concat([pandas.DataFrame(a), pandas.DataFrame(b)],
+                            ignore_index=True)
+    pandas.testing.assert_frame_equal(out2.to_pandas(), expect2)
+    # concat result feeds the groupby path
+    g = out2.groupby("k").sum().to_pandas()
+    ge = expect2.groupby("k").sum()
+    np.testing.assert_allclose(g["v"].to_numpy(), ge["v"].to_numpy(),
+                               rtol=RTOL)
+
+
 def test_native_extension_is_loaded():
     """Guard against a silent eager/pandas fallback: the in-tree .so must be
     mapped into this process."""
