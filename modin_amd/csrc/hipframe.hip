@@ -1484,16 +1484,6 @@ int hf_init(int gpu) {
   if (gpu < 0 || gpu >= n)
     return set_err(HF_ERR_ARG, "hf_init", "gpu index out of range");
   HF_HIP("hf_init", hipSetDevice(gpu));
-  // keep the stream-ordered allocator's pool resident: the default release
-  // threshold (0) returns freed memory to the OS at every sync, making each
-  // step's 10 GB radix scratch a fresh ~300 ms mapping
-  {
-    hipMemPool_t pool = nullptr;
-    if (hipDeviceGetDefaultMemPool(&pool, gpu) == hipSuccess && pool) {
-      uint64_t thresh = UINT64_MAX;
-      hipMemPoolSetAttribute(pool, hipMemPoolAttrReleaseThreshold, &thresh);
-    }
-  }
   HF_HIP("hf_init", hipStreamCreate(&g.stream));
   HF_HIP("hf_init", hipMalloc(&g.d_scratch, SCRATCH_BYTES));
   HF_HIP("hf_init", hipMemset(g.d_scratch, 0, SCRATCH_BYTES));
