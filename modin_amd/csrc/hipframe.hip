@@ -47,6 +47,11 @@ struct hf_col {
   void*   d_hist = nullptr;
   int64_t hist_kmin = 0;
   int64_t hist_nb = 0;
+  // cached u32 shifted copy of an int64 key column (key - key_min), the
+  // narrow-key fast path for the radix scatter (12 B/row instead of 16).
+  // Same immutable-column caching rationale as d_hist.
+  void*   d_k32 = nullptr;
+  int64_t k32_min = 0;
 };
 
 namespace {
@@ -649,9 +654,21 @@ __global__ void __launch_bounds__(BLOCK) k_gb_hist(
 // coalesced — each tile emits one contiguous chunk per bucket stream
 // (probe: 2.6x the register-staged direct scatter).  RPT rows/thread
 // (even); NV value columns; odd tail row handled by block 0 up front.
-template <int NV, int RPT, int BLK, int RL>
+__global__ void __launch_bounds__(BLOCK) k_conv_keys32(
+    const int64_t* __restrict__ keys, int64_t n, int64_t key_min,
+    unsigned* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const int64_t k = keys[i] - key_min;
+    out[i] = ((uint64_t)k < 0xFFFFFFFFull) ? (unsigned)k : 0xFFFFFFFFu;
+  }
+}
+
+template <int NV, int RPT, int BLK, int RL, bool K32>
 __global__ void __launch_bounds__(BLK) k_gb_scatter(
-    const int64_t* __restrict__ keys, const double* __restrict__ v0,
+    const int64_t* __restrict__ keys, const unsigned* __restrict__ keys32,
+    const double* __restrict__ v0,
     const double* __restrict__ v1, int64_t n, int64_t key_min, int64_t n_slots,
     int nb, unsigned* __restrict__ cursors,
     double* __restrict__ r0, double* __restrict__ r1,
@@ -660,7 +677,7 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
   constexpr int PAIRS = RPT / 2;
   if ((n & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
     // odd tail row: direct single-row reservation + write
-    const int64_t k = keys[n - 1] - key_min;
+    const int64_t k = K32 ? (int64_t)keys32[n - 1] : keys[n - 1] - key_min;
     if ((uint64_t)k < (uint64_t)n_slots) {
       const int b = (int)(k >> RL);
       const int64_t pos = (int64_t)atomicAdd(&cursors[b], 1u);
@@ -699,8 +716,16 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
       const int a = 2 * j, bslot = 2 * j + 1;
       lb[a] = lb[bslot] = -1;
       if (pr < npair_total) {
-        const longlong2 kk = reinterpret_cast<const longlong2*>(keys)[pr];
-        const int64_t ka = kk.x - key_min, kb = kk.y - key_min;
+        int64_t ka, kb;
+        if (K32) {
+          const uint2 kk32 = reinterpret_cast<const uint2*>(keys32)[pr];
+          ka = (int64_t)kk32.x;
+          kb = (int64_t)kk32.y;
+        } else {
+          const longlong2 kk = reinterpret_cast<const longlong2*>(keys)[pr];
+          ka = kk.x - key_min;
+          kb = kk.y - key_min;
+        }
         double2 vv0{}, vv1{};
         if (NV > 0) vv0 = reinterpret_cast<const double2*>(v0)[pr];
         if (NV > 1) vv1 = reinterpret_cast<const double2*>(v1)[pr];
@@ -1560,6 +1585,7 @@ int hf_col_free(hf_col* col) {
   if (!col) return HF_OK;
   if (g.inited && col->dptr) dev_free(col->dptr, g.stream);
   if (col->d_hist) free(col->d_hist);  // host-side cached histogram
+  if (g.inited && col->d_k32) dev_free(col->d_k32, g.stream);
   delete col;
   return HF_OK;
 }
@@ -1819,6 +1845,21 @@ int gb_dense_path(const hf_col* keys, const GbPtrs& ptrs, int nvals,
   return HF_OK;
 }
 
+int ensure_keys32(hf_col* keys, int64_t key_min) {
+  if (keys->d_k32 && keys->k32_min == key_min) return HF_OK;
+  if (keys->d_k32) { dev_free(keys->d_k32, g.stream); keys->d_k32 = nullptr; }
+  const int64_t n = keys->len;
+  HF_HIP("keys32", dev_alloc(&keys->d_k32, (n > 0 ? n : 1) * 4, g.stream));
+  int rc = timed_launch("gb_keys32", [&] {
+    hipLaunchKernelGGL(k_conv_keys32, dim3((uint32_t)grid_for(n)), dim3(BLOCK),
+                       0, g.stream, (const int64_t*)keys->dptr, n, key_min,
+                       (unsigned*)keys->d_k32);
+  });
+  if (rc != HF_OK) return rc;
+  keys->k32_min = key_min;
+  return HF_OK;
+}
+
 template <int RL, int AOP>
 int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
                   int64_t n_slots, uintptr_t sums, uintptr_t rowcnt,
@@ -1826,6 +1867,8 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
   const int64_t n = keys->len;
   const int64_t nb = (n_slots + (1 << RL) - 1) >> RL;
   int rc = ensure_host_hist(keys, key_min, n_slots, nb, RL);
+  if (rc != HF_OK) return rc;
+  rc = ensure_keys32(keys, key_min);
   if (rc != HF_OK) return rc;
   const int64_t* h = keys->d_hist ? (const int64_t*)keys->d_hist : nullptr;
   // exact per-bucket regions, 64-row aligned; u32 cursors cap one partition
@@ -1882,10 +1925,12 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
     const uint32_t lds =
         (uint32_t)(tile_sz * (8 * NVv + 4) + nb * 12 + 16);
     return timed_launch("gb_scatter", [&] {
-      hipLaunchKernelGGL((k_gb_scatter<NVv, RPTv, BLKv, RL>), dim3(sgrid),
-                         dim3(BLKv), lds, g.stream,
-                         (const int64_t*)keys->dptr, ptrs.vals[0], ptrs.vals[1],
-                         n, key_min, n_slots, (int)nb, d_cur, r0, r1, rk, d_err);
+      hipLaunchKernelGGL((k_gb_scatter<NVv, RPTv, BLKv, RL, true>),
+                         dim3(sgrid), dim3(BLKv), lds, g.stream,
+                         (const int64_t*)keys->dptr,
+                         (const unsigned*)keys->d_k32, ptrs.vals[0],
+                         ptrs.vals[1], n, key_min, n_slots, (int)nb, d_cur, r0,
+                         r1, rk, d_err);
     });
   };
   rc = nvals == 0 ? scat(std::integral_constant<int, 0>{},
