@@ -525,6 +525,115 @@ __global__ void __launch_bounds__(BLK) k_scat_lds5(
   }
 }
 
+// B6: ablation — template-disable phases of the lds3 structure to find the
+// dominant cost.  PH bitmask: 1=ranks, 2=scan, 4=reserve, 8=stage, 16=writeout
+template <int RPT, int RL, int BLK, int PH>
+__global__ void __launch_bounds__(BLK) k_scat_abl(
+    const int64_t* __restrict__ keys, const double* __restrict__ v0, int64_t n,
+    int64_t n_slots, int nb, unsigned* __restrict__ cursors,
+    double* __restrict__ r0, unsigned short* __restrict__ rk) {
+  constexpr int TILE = BLK * RPT;
+  constexpr int PAIRS = RPT / 2;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  double* sval = reinterpret_cast<double*>(smem_raw);
+  unsigned* skey = reinterpret_cast<unsigned*>(sval + TILE);
+  unsigned* it_cnt = skey + TILE;
+  unsigned* it_off = it_cnt + nb;
+  unsigned* it_gbase = it_off + nb;
+  unsigned* s_total = it_gbase + nb;
+  const int64_t ntiles = (n + TILE - 1) / TILE;
+  const int64_t npair_total = n >> 1;
+  for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+    const int64_t t0 = tile * TILE;
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) {
+      it_cnt[t] = 0;
+      if (!(PH & 4)) it_gbase[t] = (unsigned)(t * 64);  // fake bases
+    }
+    int lb[RPT];
+    unsigned lk[RPT];
+    unsigned lr[RPT];
+    double lv[RPT];
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < PAIRS; ++j) {
+      const int64_t pr = (t0 >> 1) + (int64_t)j * BLK + threadIdx.x;
+      const int a = 2 * j, b = 2 * j + 1;
+      lb[a] = lb[b] = -1;
+      if (pr < npair_total) {
+        const longlong2 kk = reinterpret_cast<const longlong2*>(keys)[pr];
+        const double2 vv = reinterpret_cast<const double2*>(v0)[pr];
+        lb[a] = (int)(kk.x >> RL);
+        lk[a] = (unsigned)(kk.x & ((1 << RL) - 1));
+        lv[a] = vv.x;
+        lb[b] = (int)(kk.y >> RL);
+        lk[b] = (unsigned)(kk.y & ((1 << RL) - 1));
+        lv[b] = vv.y;
+      }
+    }
+    if (PH & 1) {
+#pragma unroll
+      for (int j = 0; j < RPT; ++j)
+        if (lb[j] >= 0) lr[j] = atomicAdd(&it_cnt[lb[j]], 1u);
+    } else {
+#pragma unroll
+      for (int j = 0; j < RPT; ++j) lr[j] = (unsigned)(threadIdx.x + j) & 63;
+    }
+    __syncthreads();
+    if (PH & 2) {
+      if (threadIdx.x < 64) {
+        const int lane = threadIdx.x;
+        unsigned carry = 0;
+        for (int base = 0; base < nb; base += 64) {
+          const int t = base + lane;
+          unsigned v = (t < nb) ? it_cnt[t] : 0;
+          unsigned incl = v;
+#pragma unroll
+          for (int d = 1; d < 64; d <<= 1) {
+            unsigned up = __shfl_up(incl, d);
+            if (lane >= d) incl += up;
+          }
+          if (t < nb) it_off[t] = carry + incl - v;
+          carry += __shfl(incl, 63);
+        }
+        if (lane == 0) *s_total = carry;
+      }
+    } else if (threadIdx.x == 0) {
+      *s_total = (unsigned)min((int64_t)TILE, n - t0);
+    }
+    __syncthreads();
+    if (PH & 4) {
+      for (int t = threadIdx.x; t < nb; t += blockDim.x) {
+        const unsigned c = it_cnt[t];
+        if (c) it_gbase[t] = atomicAdd(&cursors[t], c);
+      }
+    }
+    if (PH & 8) {
+#pragma unroll
+      for (int j = 0; j < RPT; ++j) {
+        if (lb[j] >= 0) {
+          unsigned p = (PH & 2) ? it_off[lb[j]] + lr[j]
+                                : (unsigned)(j * BLK + threadIdx.x);
+          if (p >= TILE) p = TILE - 1;  // ablation safety
+          sval[p] = lv[j];
+          skey[p] = ((unsigned)lb[j] << 16) | lk[j];
+        }
+      }
+    }
+    __syncthreads();
+    if (PH & 16) {
+      const int staged = (int)*s_total;
+      for (int p = threadIdx.x; p < staged; p += blockDim.x) {
+        const unsigned b = skey[p] >> 16;
+        int64_t pos = (int64_t)it_gbase[b] + (p - it_off[b]);
+        pos &= (1LL << 30) - 1;  // ablation safety (alloc > 2^30 rows)
+        r0[pos] = sval[p];
+        rk[pos] = (unsigned short)(skey[p] & 0xFFFF);
+      }
+    }
+    __syncthreads();
+  }
+}
+
 // ---------------- aggregate variants ----------------
 
 // presence-byte agg (production shape after the rowcnt->presence change)
@@ -777,10 +886,28 @@ int main(int argc, char** argv) {
                            lds, 0, keys, v0, n, n_slots, nb, d_cur, r0, rk);
       }, 26.0 * n);
     };
-    scat_lds5(std::integral_constant<int, 24>{}, std::integral_constant<int, 256>{});
-    scat_lds5(std::integral_constant<int, 16>{}, std::integral_constant<int, 256>{});
-    scat_lds5(std::integral_constant<int, 16>{}, std::integral_constant<int, 512>{});
-    scat_lds5(std::integral_constant<int, 32>{}, std::integral_constant<int, 256>{});
+    auto abl = [&](auto phTag) {
+      constexpr int PH = decltype(phTag)::value;
+      constexpr int RPT = 24, BLK = 256;
+      snprintf(nm, sizeof nm, "scat_abl PH=%d RL=%d", PH, RL);
+      const int64_t tile_sz = (int64_t)BLK * RPT;
+      const int64_t ntiles = (n + tile_sz - 1) / tile_sz;
+      const uint32_t grid = (uint32_t)std::min<int64_t>(ntiles, 2048);
+      const uint32_t lds = tile_sz * 12 + nb * 16 + 16;
+      run(nm, 2, reset_cur, [&] {
+        hipLaunchKernelGGL((k_scat_abl<RPT, RL, BLK, PH>), dim3(grid),
+                           dim3(BLK), lds, 0, keys, v0, n, n_slots, nb, d_cur,
+                           r0, rk);
+      }, 26.0 * n);
+    };
+    abl(std::integral_constant<int, 31>{});   // full
+    abl(std::integral_constant<int, 15>{});   // no writeout
+    abl(std::integral_constant<int, 27>{});   // no reserve
+    abl(std::integral_constant<int, 30>{});   // no ranks
+    abl(std::integral_constant<int, 23>{});   // no stage
+    abl(std::integral_constant<int, 29>{});   // no scan
+    abl(std::integral_constant<int, 3>{});    // loads+ranks+scan only
+    abl(std::integral_constant<int, 0>{});    // loads only
     // aggregate variants (consume whatever the last scatter left; perf-only)
     auto agg = [&](auto vecTag, auto blkTag) {
       constexpr int VEC = decltype(vecTag)::value;
