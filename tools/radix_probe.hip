@@ -525,6 +525,118 @@ __global__ void __launch_bounds__(BLK) k_scat_lds5(
   }
 }
 
+// B7: per-wave rank counters — it_cnt is [WPB][nb] so rank atomics never
+// contend across waves; the scan folds wave offsets into per-(wave,bucket)
+// bases.  WPB = BLK/64.
+template <int RPT, int RL, int BLK>
+__global__ void __launch_bounds__(BLK) k_scat_lds7(
+    const int64_t* __restrict__ keys, const double* __restrict__ v0, int64_t n,
+    int64_t n_slots, int nb, unsigned* __restrict__ cursors,
+    double* __restrict__ r0, unsigned short* __restrict__ rk) {
+  constexpr int TILE = BLK * RPT;
+  constexpr int PAIRS = RPT / 2;
+  constexpr int WPB = BLK / 64;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  double* sval = reinterpret_cast<double*>(smem_raw);
+  unsigned* skey = reinterpret_cast<unsigned*>(sval + TILE);
+  unsigned* it_cnt = skey + TILE;        // [WPB][nb]
+  unsigned* it_base = it_cnt + WPB * nb; // [WPB][nb] per-(wave,bucket) base
+  unsigned* it_off = it_base + WPB * nb; // [nb] block-level offsets
+  unsigned* it_gbase = it_off + nb;      // [nb]
+  unsigned* s_total = it_gbase + nb;
+  const int wave = threadIdx.x >> 6;
+  const int64_t ntiles = (n + TILE - 1) / TILE;
+  const int64_t npair_total = n >> 1;
+  for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+    const int64_t t0 = tile * TILE;
+    for (int t = threadIdx.x; t < WPB * nb; t += blockDim.x) it_cnt[t] = 0;
+    int lb[RPT];
+    unsigned lk[RPT];
+    unsigned lr[RPT];
+    double lv[RPT];
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < PAIRS; ++j) {
+      const int64_t pr = (t0 >> 1) + (int64_t)j * BLK + threadIdx.x;
+      const int a = 2 * j, b = 2 * j + 1;
+      lb[a] = lb[b] = -1;
+      if (pr < npair_total) {
+        const longlong2 kk = reinterpret_cast<const longlong2*>(keys)[pr];
+        const double2 vv = reinterpret_cast<const double2*>(v0)[pr];
+        lb[a] = (int)(kk.x >> RL);
+        lk[a] = (unsigned)(kk.x & ((1 << RL) - 1));
+        lv[a] = vv.x;
+        lb[b] = (int)(kk.y >> RL);
+        lk[b] = (unsigned)(kk.y & ((1 << RL) - 1));
+        lv[b] = vv.y;
+      }
+    }
+    unsigned* mycnt = it_cnt + wave * nb;
+#pragma unroll
+    for (int j = 0; j < RPT; ++j)
+      if (lb[j] >= 0) lr[j] = atomicAdd(&mycnt[lb[j]], 1u);
+    __syncthreads();
+    // wave 0: per-bucket totals + block prefix + per-wave bases
+    if (threadIdx.x < 64) {
+      const int lane = threadIdx.x;
+      unsigned carry = 0;
+      for (int base = 0; base < nb; base += 64) {
+        const int t = base + lane;
+        unsigned tot = 0, wsum[WPB];
+        if (t < nb) {
+#pragma unroll
+          for (int w = 0; w < WPB; ++w) {
+            wsum[w] = it_cnt[w * nb + t];
+            tot += wsum[w];
+          }
+        }
+        unsigned incl = tot;
+#pragma unroll
+        for (int d = 1; d < 64; d <<= 1) {
+          unsigned up = __shfl_up(incl, d);
+          if (lane >= d) incl += up;
+        }
+        if (t < nb) {
+          unsigned off = carry + incl - tot;
+          it_off[t] = off;
+#pragma unroll
+          for (int w = 0; w < WPB; ++w) {
+            it_base[w * nb + t] = off;
+            off += wsum[w];
+          }
+        }
+        carry += __shfl(incl, 63);
+      }
+      if (lane == 0) *s_total = carry;
+    }
+    __syncthreads();
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) {
+      unsigned c = 0;
+#pragma unroll
+      for (int w = 0; w < WPB; ++w) c += it_cnt[w * nb + t];
+      if (c) it_gbase[t] = atomicAdd(&cursors[t], c);
+    }
+    unsigned* mybase = it_base + wave * nb;
+#pragma unroll
+    for (int j = 0; j < RPT; ++j) {
+      if (lb[j] >= 0) {
+        const unsigned p = mybase[lb[j]] + lr[j];
+        sval[p] = lv[j];
+        skey[p] = ((unsigned)lb[j] << 16) | lk[j];
+      }
+    }
+    __syncthreads();
+    const int staged = (int)*s_total;
+    for (int p = threadIdx.x; p < staged; p += blockDim.x) {
+      const unsigned b = skey[p] >> 16;
+      const int64_t pos = (int64_t)it_gbase[b] + (p - it_off[b]);
+      r0[pos] = sval[p];
+      rk[pos] = (unsigned short)(skey[p] & 0xFFFF);
+    }
+    __syncthreads();
+  }
+}
+
 // B6: ablation — template-disable phases of the lds3 structure to find the
 // dominant cost.  PH bitmask: 1=ranks, 2=scan, 4=reserve, 8=stage, 16=writeout
 template <int RPT, int RL, int BLK, int PH>
@@ -900,6 +1012,23 @@ int main(int argc, char** argv) {
                            r0, rk);
       }, 26.0 * n);
     };
+    auto scat7 = [&](auto rptTag, auto blkTag) {
+      constexpr int RPT = decltype(rptTag)::value;
+      constexpr int BLK = decltype(blkTag)::value;
+      constexpr int WPB = BLK / 64;
+      snprintf(nm, sizeof nm, "scat_lds7 RPT=%d BLK=%d RL=%d", RPT, BLK, RL);
+      const int64_t tile_sz = (int64_t)BLK * RPT;
+      const int64_t ntiles = (n + tile_sz - 1) / tile_sz;
+      const uint32_t grid = (uint32_t)std::min<int64_t>(ntiles, 2048);
+      const uint32_t lds = tile_sz * 12 + nb * (2 * WPB + 2) * 4 + 16;
+      run(nm, 2, reset_cur, [&] {
+        hipLaunchKernelGGL((k_scat_lds7<RPT, RL, BLK>), dim3(grid), dim3(BLK),
+                           lds, 0, keys, v0, n, n_slots, nb, d_cur, r0, rk);
+      }, 26.0 * n);
+    };
+    scat7(std::integral_constant<int, 24>{}, std::integral_constant<int, 256>{});
+    scat7(std::integral_constant<int, 16>{}, std::integral_constant<int, 512>{});
+    scat_lds3(std::integral_constant<int, 24>{}, std::integral_constant<int, 512>{});
     abl(std::integral_constant<int, 31>{});   // full
     abl(std::integral_constant<int, 15>{});   // no writeout
     abl(std::integral_constant<int, 27>{});   // no reserve
