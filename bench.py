@@ -101,6 +101,8 @@ def main():
     ap.add_argument("--parts", type=int, default=1,
                     help="partitions per rank (gb_accum launches per step)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--no-verify", action="store_true",
+                    help="skip the oracle parity gate before timing")
     args = ap.parse_args()
 
     import modin_amd.distributed as dmod
@@ -128,6 +130,30 @@ def main():
 
     for _ in range(args.warmup):
         step()
+
+    if world == 1 and not args.no_verify:
+        # parity gate outside the timed region: the step's full 1e9-row
+        # result must match the numpy oracle (regenerated from the same
+        # seeds chunk-by-chunk to bound host RAM)
+        import oracle
+        res = step()._modin_frame
+        got_keys = res.index.to_numpy()
+        got_sums = res.to_pandas()["v"].to_numpy()
+        # regenerate with EXACTLY build_frame's draw order (same chunking)
+        vrng = np.random.default_rng([42, 0])
+        acc = np.zeros(args.keys, dtype=np.float64)
+        seen = np.zeros(args.keys, dtype=bool)
+        for cn in oracle.split_row_counts(local_n, max(args.parts, 1), 1):
+            kk = vrng.integers(0, args.keys, cn).astype(np.int64)
+            vv = vrng.random(cn)
+            acc += np.bincount(kk, weights=vv, minlength=args.keys)
+            seen[kk] = True
+        exp_keys = np.nonzero(seen)[0]
+        np.testing.assert_array_equal(got_keys, exp_keys)
+        np.testing.assert_allclose(got_sums, acc[exp_keys], rtol=1e-12,
+                                   atol=1e-9)
+        print(f"# verify ok: {exp_keys.size} groups match the oracle",
+              file=sys.stderr)
 
     lib.kernel_stats_reset()
     if is_dist:
