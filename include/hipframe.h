@@ -97,7 +97,8 @@ enum {
   HF_MAP_FILLNA = 6,/* isnan(x) ? s : x  (f64 only)             */
   HF_MAP_ABS = 7,   /* |x|         */
   HF_MAP_NEG = 8,   /* -x          */
-  HF_MAP_CAST_F64 = 9 /* (double)x : i64 -> f64; scalar ignored */
+  HF_MAP_CAST_F64 = 9, /* (double)x : i64 -> f64; scalar ignored */
+  HF_MAP_CAST_I64 = 10 /* (int64)x : f64 -> i64, C truncation (astype)     */
 };
 int hf_map_scalar(int op, const hf_col* in, double scalar, hf_col** out);
 /* i64 column with an exact int64 scalar (double cannot hold all int64). */
@@ -211,6 +212,9 @@ int hf_gather(const hf_col* col, const hf_col* idx, hf_col** out);
  * multi-partition right frame for the broadcast join, combine() at
  * dataframe.py:2918). */
 int hf_col_concat(const hf_col* const* cols, int ncols, hf_col** out);
+
+/* Row slice [start, start+len) of a column (head/tail/iloc ranges). */
+int hf_col_slice(const hf_col* col, int64_t start, int64_t len, hf_col** out);
 
 /* ---- Compare + Filter (df[df.v > x], dropna) ----
  * Compare is a Map-shaped elementwise kernel producing an int64 0/1 mask

@@ -243,6 +243,43 @@ class HipDataframe:
             return DeviceBlock(out, block.length)
         return self.map(block_fn)
 
+    # ---- row range (head/tail; mask/partition.py:224 row-slice form) ----
+    def take_row_range(self, start: int, stop: int) -> "HipDataframe":
+        start = max(0, min(start, len(self)))
+        stop = max(start, min(stop, len(self)))
+        out_parts, lengths = [], []
+        off = 0
+        for p, ln in zip(self._partitions, self._row_lengths):
+            lo, hi = max(start - off, 0), min(stop - off, ln)
+            if lo < hi:
+                block = p.block()
+                cols = {n: lib.col_slice(c, lo, hi - lo)
+                        for n, c in block.columns.items()}
+                out_parts.append(HipDataframePartition(
+                    DeviceBlock(cols, hi - lo)))
+                lengths.append(hi - lo)
+            off += ln
+        if not out_parts:
+            block = self._partitions[0].block()
+            cols = {n: lib.col_slice(c, 0, 0)
+                    for n, c in block.columns.items()}
+            out_parts, lengths = [HipDataframePartition(DeviceBlock(cols, 0))], [0]
+        idx = self.index[start:stop]
+        return HipDataframe(out_parts, idx, self.columns, lengths, self.dtypes)
+
+    # ---- astype over all columns ----
+    def astype_all(self, dtype) -> "HipDataframe":
+        code = {np.dtype(np.float64): lib.MAP_CAST_F64,
+                np.dtype(np.int64): lib.MAP_CAST_I64}[np.dtype(dtype)]
+
+        def block_fn(block: DeviceBlock) -> DeviceBlock:
+            return DeviceBlock({n: lib.map_scalar(code, c, 0)
+                                if c.dtype_code != (lib.HF_FLOAT64 if code == lib.MAP_CAST_F64 else lib.HF_INT64)
+                                else c
+                                for n, c in block.columns.items()},
+                               block.length)
+        return self.map(block_fn)
+
     # ---- row concat (PartitionManager.concat device form,
     #      partition_manager.py:943: stack the partition lists; no device
     #      copy — partitions are immutable) ----

@@ -45,7 +45,7 @@ HF_FLOAT64 = 1
 
 # map ops
 MAP_ADD, MAP_SUB, MAP_RSUB, MAP_MUL, MAP_DIV, MAP_RDIV, MAP_FILLNA, MAP_ABS, \
-    MAP_NEG, MAP_CAST_F64 = range(10)
+    MAP_NEG, MAP_CAST_F64, MAP_CAST_I64 = range(11)
 # binary ops
 BIN_ADD, BIN_SUB, BIN_MUL, BIN_DIV = range(4)
 
@@ -118,6 +118,8 @@ def load() -> ct.CDLL:
                                               ct.POINTER(ct.c_int64)]),
             "hf_col_concat": (ct.c_int, [ct.POINTER(ct.c_void_p), ct.c_int,
                                          ct.POINTER(ct.c_void_p)]),
+            "hf_col_slice": (ct.c_int, [ct.c_void_p, ct.c_int64, ct.c_int64,
+                                        ct.POINTER(ct.c_void_p)]),
             "hf_join_build": (ct.c_int, [ct.c_void_p, ct.POINTER(ct.c_void_p),
                                          ct.c_int, ct.c_int64, ct.c_int64,
                                          ct.POINTER(ct.c_void_p)]),
@@ -161,7 +163,7 @@ def exported_symbols():
         "hf_memset_raw", "hf_map_scalar", "hf_map_scalar_i64", "hf_binary",
         "hf_reduce", "hf_groupby_accum", "hf_groupby_compact", "hf_fill_f64",
         "hf_fixup_empty", "hf_sort_perm",
-        "hf_col_concat", "hf_join_build", "hf_join_free", "hf_join_probe",
+        "hf_col_concat", "hf_col_slice", "hf_join_build", "hf_join_free", "hf_join_probe",
         "hf_gather", "hf_compare_scalar", "hf_filter_plan", "hf_filter_apply",
         "hf_filter_iota", "hf_filter_plan_free", "hf_profiling",
         "hf_kernel_stats", "hf_kernel_stats_reset",
@@ -280,6 +282,12 @@ def map_scalar(op: int, col: ColumnRef, scalar) -> ColumnRef:
     out = ct.c_void_p()
     if col.dtype_code == HF_INT64 and op in (MAP_DIV, MAP_RDIV, MAP_FILLNA):
         col = cast_f64(col)  # pandas promotes int div to float; fillna no-ops
+    if op == MAP_CAST_I64:
+        if col.dtype_code == HF_INT64:
+            return col
+        _check(load().hf_map_scalar(op, col.handle, 0.0, ct.byref(out)),
+               "hf_map_scalar(cast_i64)")
+        return _wrap(out, col.length, HF_INT64)
     if col.dtype_code == HF_INT64 and op != MAP_CAST_F64:
         _check(load().hf_map_scalar_i64(op, col.handle, int(scalar or 0),
                                         ct.byref(out)), "hf_map_scalar_i64")
@@ -382,6 +390,14 @@ def groupby_compact(sums: int, rowcnt: int, counts: int, nvals: int,
     ccols = ([_wrap(ct.c_void_p(out_counts[c]), n, HF_INT64) for c in range(nvals)]
              if counts else None)
     return kcol, scols, ccols, n
+
+
+def col_slice(col: ColumnRef, start: int, length: int) -> ColumnRef:
+    ensure_ready()
+    out = ct.c_void_p()
+    _check(load().hf_col_slice(col.handle, start, length, ct.byref(out)),
+           "hf_col_slice")
+    return _wrap(out, length, col.dtype_code)
 
 
 def concat(cols: list) -> ColumnRef:

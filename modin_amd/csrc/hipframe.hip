@@ -311,6 +311,14 @@ __global__ void __launch_bounds__(BLOCK) k_cast_i64_f64(const int64_t* __restric
   for (; i < n; i += stride) out[i] = (double)in[i];
 }
 
+__global__ void __launch_bounds__(BLOCK) k_cast_f64_i64(const double* __restrict__ in,
+                                                        int64_t* __restrict__ out,
+                                                        int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = (int64_t)in[i];
+}
+
 // ---------------------------------------------------------------------------
 // kernels: Binary (elementwise column op column) — algebra/binary.py device form
 // ---------------------------------------------------------------------------
@@ -1628,6 +1636,17 @@ int hf_map_scalar(int op, const hf_col* in, double scalar, hf_col** out) {
                          in->len);
     });
   }
+  if (op == HF_MAP_CAST_I64) {
+    if (in->dtype != HF_FLOAT64)
+      return set_err(HF_ERR_ARG, "hf_map_scalar", "cast_i64 needs f64 input");
+    int rc = hf_col_alloc(in->len, HF_INT64, out);
+    if (rc != HF_OK) return rc;
+    return timed_launch("cast_f64_i64", [&] {
+      hipLaunchKernelGGL(k_cast_f64_i64, dim3(grid_for(in->len)), dim3(BLOCK), 0,
+                         g.stream, (const double*)in->dptr,
+                         (int64_t*)(*out)->dptr, in->len);
+    });
+  }
   if (in->dtype == HF_INT64) return hf_map_scalar_i64(op, in, (int64_t)scalar, out);
   int rc = hf_col_alloc(in->len, HF_FLOAT64, out);
   if (rc != HF_OK) return rc;
@@ -2209,6 +2228,20 @@ int hf_col_concat(const hf_col* const* cols, int ncols, hf_col** out) {
                             hipMemcpyDeviceToDevice, g.stream));
     dst += bytes;
   }
+  return HF_OK;
+}
+
+int hf_col_slice(const hf_col* col, int64_t start, int64_t len, hf_col** out) {
+  HF_NEED_INIT("hf_col_slice");
+  if (!col || !out || start < 0 || len < 0 || start + len > col->len)
+    return set_err(HF_ERR_ARG, "hf_col_slice", "bad range");
+  int rc = hf_col_alloc(len, col->dtype, out);
+  if (rc != HF_OK) return rc;
+  const int64_t esz = dtype_size(col->dtype);
+  if (len > 0)
+    HF_HIP("hf_col_slice",
+           hipMemcpyAsync((*out)->dptr, (const char*)col->dptr + start * esz,
+                          len * esz, hipMemcpyDeviceToDevice, g.stream));
   return HF_OK;
 }
 

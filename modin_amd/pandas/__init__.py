@@ -236,6 +236,49 @@ class DataFrame(_HipPandasBase):
             )
         raise lib.HfError("only column selection / boolean masks are supported")
 
+    def head(self, n: int = 5):
+        return DataFrame(
+            query_compiler=self._query_compiler.take_row_range(0, n))
+
+    def tail(self, n: int = 5):
+        total = len(self._query_compiler)
+        return DataFrame(
+            query_compiler=self._query_compiler.take_row_range(total - n,
+                                                               total))
+
+    def astype(self, dtype):
+        return DataFrame(query_compiler=self._query_compiler.astype(dtype))
+
+    def rename(self, columns: dict):
+        qc = self._query_compiler
+        frame = qc._modin_frame
+        new_cols = [columns.get(c, c) for c in frame.columns]
+        # metadata-only: device blocks are re-labelled lazily on access
+        from ..core.dataframe import HipDataframe
+        from ..core.partition import DeviceBlock, HipDataframePartition
+
+        def relabel(block):
+            return DeviceBlock(
+                {columns.get(n, n): c for n, c in block.columns.items()},
+                block.length)
+        parts = [p.add_to_apply_calls(relabel) for p in frame._partitions]
+        import pandas as _pd
+        nf = HipDataframe(parts, frame._index, new_cols, frame._row_lengths,
+                          _pd.Series({columns.get(n, n): d
+                                      for n, d in frame.dtypes.items()}))
+        return DataFrame(query_compiler=type(qc)(nf))
+
+    def reset_index(self, drop: bool = False):
+        if not drop:
+            raise lib.HfError("reset_index(drop=False) keeps the old index "
+                              "as a column — later round")
+        qc = self._query_compiler
+        frame = qc._modin_frame
+        from ..core.dataframe import HipDataframe
+        nf = HipDataframe(frame._partitions, pandas.RangeIndex(len(frame)),
+                          frame.columns, frame._row_lengths, frame.dtypes)
+        return DataFrame(query_compiler=type(qc)(nf))
+
     def sort_values(self, by: str, ascending: bool = True,
                     kind: str = "stable"):
         """Always stable (equals pandas sort_values(kind='stable'), a

@@ -129,6 +129,13 @@ class HipQueryCompiler:
         return self.__constructor__(
             self._modin_frame.filter_rows(mask_qc._modin_frame))
 
+    def take_row_range(self, start: int, stop: int) -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.take_row_range(start, stop))
+
+    def astype(self, dtype) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.astype_all(dtype))
+
     # ---- concat (reference qc.concat -> PartitionManager.concat :943) ----
     def concat(self, others: list) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.concat_rows(

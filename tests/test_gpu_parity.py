@@ -382,6 +382,32 @@ def test_concat_vs_pandas(npartitions):
                                rtol=RTOL)
 
 
+def test_head_tail_astype_rename_reset(npartitions):
+    rng = np.random.default_rng(80)
+    data = {"a": rng.integers(0, 50, 2000).astype(np.int64),
+            "b": rng.random(2000) * 10}
+    df = mpd.DataFrame(data)
+    pdf = pandas.DataFrame(data)
+    pandas.testing.assert_frame_equal(df.head(7).to_pandas(), pdf.head(7))
+    pandas.testing.assert_frame_equal(df.tail(9).to_pandas(), pdf.tail(9))
+    pandas.testing.assert_frame_equal(df.astype("float64").to_pandas(),
+                                      pdf.astype("float64"))
+    pandas.testing.assert_frame_equal(df.astype("int64").to_pandas(),
+                                      pdf.astype("int64"))
+    pandas.testing.assert_frame_equal(
+        df.rename(columns={"a": "x"}).to_pandas(),
+        pdf.rename(columns={"a": "x"}))
+    filt = df[df["b"] > 5.0]
+    pfilt = pdf[pdf["b"] > 5.0]
+    pandas.testing.assert_frame_equal(filt.reset_index(drop=True).to_pandas(),
+                                      pfilt.reset_index(drop=True))
+    # renamed frame still computes
+    out = df.rename(columns={"a": "k"}).groupby("k").sum().to_pandas()
+    expect = pdf.rename(columns={"a": "k"}).groupby("k").sum()
+    np.testing.assert_allclose(out["b"].to_numpy(), expect["b"].to_numpy(),
+                               rtol=RTOL)
+
+
 def test_native_extension_is_loaded():
     """Guard against a silent eager/pandas fallback: the in-tree .so must be
     mapped into this process."""
