@@ -637,6 +637,95 @@ __global__ void __launch_bounds__(BLK) k_scat_lds7(
   }
 }
 
+// B8: wave-autonomous tiles — one WAVE owns each tile; ranks/scan/reserve/
+// stage/writeout all wave-local, zero block barriers (waves self-overlap).
+template <int RPT, int RL, int BLK>
+__global__ void __launch_bounds__(BLK) k_scat_wave(
+    const int64_t* __restrict__ keys, const double* __restrict__ v0, int64_t n,
+    int64_t n_slots, int nb, unsigned* __restrict__ cursors,
+    double* __restrict__ r0, unsigned short* __restrict__ rk) {
+  constexpr int WPB = BLK / 64;
+  constexpr int TILE = 64 * RPT;          // rows per wave-tile
+  constexpr int PAIRS = RPT / 2;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  // per-wave carve: sval[TILE] f64 + skey[TILE] u32 + cnt/off/gbase[nb each]
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  char* my = smem_raw + (size_t)wave * (TILE * 12 + 3 * nb * 4 + 16);
+  double* sval = reinterpret_cast<double*>(my);
+  unsigned* skey = reinterpret_cast<unsigned*>(sval + TILE);
+  unsigned* it_cnt = skey + TILE;
+  unsigned* it_off = it_cnt + nb;
+  unsigned* it_gbase = it_off + nb;
+  const int64_t ntiles = (n + TILE - 1) / TILE;
+  const int64_t npair_total = n >> 1;
+  for (int64_t tile = (int64_t)blockIdx.x * WPB + wave; tile < ntiles;
+       tile += (int64_t)gridDim.x * WPB) {
+    const int64_t t0 = tile * TILE;
+    for (int t = lane; t < nb; t += 64) it_cnt[t] = 0;
+    int lb[RPT];
+    unsigned lk[RPT];
+    unsigned lr[RPT];
+    double lv[RPT];
+#pragma unroll
+    for (int j = 0; j < PAIRS; ++j) {
+      const int64_t pr = (t0 >> 1) + (int64_t)j * 64 + lane;
+      const int a = 2 * j, b = 2 * j + 1;
+      lb[a] = lb[b] = -1;
+      if (pr < npair_total) {
+        const longlong2 kk = reinterpret_cast<const longlong2*>(keys)[pr];
+        const double2 vv = reinterpret_cast<const double2*>(v0)[pr];
+        lb[a] = (int)(kk.x >> RL);
+        lk[a] = (unsigned)(kk.x & ((1 << RL) - 1));
+        lv[a] = vv.x;
+        lb[b] = (int)(kk.y >> RL);
+        lk[b] = (unsigned)(kk.y & ((1 << RL) - 1));
+        lv[b] = vv.y;
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < RPT; ++j)
+      if (lb[j] >= 0) lr[j] = atomicAdd(&it_cnt[lb[j]], 1u);
+    // wave-local exclusive scan of nb counters
+    {
+      unsigned carry = 0;
+      for (int base = 0; base < nb; base += 64) {
+        const int t = base + lane;
+        unsigned v = (t < nb) ? it_cnt[t] : 0;
+        unsigned incl = v;
+#pragma unroll
+        for (int d = 1; d < 64; d <<= 1) {
+          unsigned up = __shfl_up(incl, d);
+          if (lane >= d) incl += up;
+        }
+        if (t < nb) it_off[t] = carry + incl - v;
+        carry += __shfl(incl, 63);
+      }
+    }
+    // wave reserve (lanes stride the buckets)
+    for (int t = lane; t < nb; t += 64) {
+      const unsigned c = it_cnt[t];
+      if (c) it_gbase[t] = atomicAdd(&cursors[t], c);
+    }
+    // stage bucket-sorted
+#pragma unroll
+    for (int j = 0; j < RPT; ++j) {
+      if (lb[j] >= 0) {
+        const unsigned p = it_off[lb[j]] + lr[j];
+        sval[p] = lv[j];
+        skey[p] = ((unsigned)lb[j] << 16) | lk[j];
+      }
+    }
+    // writeout (wave-lockstep; LDS deps handled by lgkmcnt)
+    const int staged = (int)min((int64_t)TILE, n - t0);
+    for (int p = lane; p < staged; p += 64) {
+      const unsigned b = skey[p] >> 16;
+      const int64_t pos = (int64_t)it_gbase[b] + (p - it_off[b]);
+      r0[pos] = sval[p];
+      rk[pos] = (unsigned short)(skey[p] & 0xFFFF);
+    }
+  }
+}
+
 // B6: ablation — template-disable phases of the lds3 structure to find the
 // dominant cost.  PH bitmask: 1=ranks, 2=scan, 4=reserve, 8=stage, 16=writeout
 template <int RPT, int RL, int BLK, int PH>
@@ -1029,6 +1118,27 @@ int main(int argc, char** argv) {
     scat7(std::integral_constant<int, 24>{}, std::integral_constant<int, 256>{});
     scat7(std::integral_constant<int, 16>{}, std::integral_constant<int, 512>{});
     scat_lds3(std::integral_constant<int, 24>{}, std::integral_constant<int, 512>{});
+    auto scatw = [&](auto rptTag, auto blkTag) {
+      constexpr int RPT = decltype(rptTag)::value;
+      constexpr int BLK = decltype(blkTag)::value;
+      constexpr int WPB = BLK / 64;
+      snprintf(nm, sizeof nm, "scat_wave RPT=%d BLK=%d RL=%d", RPT, BLK, RL);
+      const int64_t tile_sz = 64 * RPT;
+      const int64_t ntiles = (n + tile_sz - 1) / tile_sz;
+      const uint32_t grid =
+          (uint32_t)std::min<int64_t>((ntiles + WPB - 1) / WPB, 4096);
+      const uint32_t lds = WPB * (tile_sz * 12 + 3 * nb * 4 + 16);
+      if (lds > 160 * 1024) { printf("%s: LDS too big (%u)\n", nm, lds); }
+      else
+        run(nm, 2, reset_cur, [&] {
+          hipLaunchKernelGGL((k_scat_wave<RPT, RL, BLK>), dim3(grid), dim3(BLK),
+                             lds, 0, keys, v0, n, n_slots, nb, d_cur, r0, rk);
+        }, 26.0 * n);
+    };
+    scatw(std::integral_constant<int, 32>{}, std::integral_constant<int, 256>{});
+    scatw(std::integral_constant<int, 32>{}, std::integral_constant<int, 384>{});
+    scatw(std::integral_constant<int, 48>{}, std::integral_constant<int, 256>{});
+    scatw(std::integral_constant<int, 16>{}, std::integral_constant<int, 512>{});
     abl(std::integral_constant<int, 31>{});   // full
     abl(std::integral_constant<int, 15>{});   // no writeout
     abl(std::integral_constant<int, 27>{});   // no reserve
