@@ -408,6 +408,19 @@ def test_head_tail_astype_rename_reset(npartitions):
                                rtol=RTOL)
 
 
+def test_groupby_keys_only(npartitions):
+    """groupby on a frame whose only column is the key: result is an empty
+    column set with the group index (pandas shape)."""
+    rng = np.random.default_rng(90)
+    k = rng.integers(0, 40, 5000).astype(np.int64)
+    df = mpd.DataFrame({"k": k})
+    out = df.groupby("k").sum().to_pandas()
+    expect = pandas.DataFrame({"k": k}).groupby("k").sum()
+    assert list(out.columns) == list(expect.columns) == []
+    np.testing.assert_array_equal(out.index.to_numpy(),
+                                  expect.index.to_numpy())
+
+
 def test_native_extension_is_loaded():
     """Guard against a silent eager/pandas fallback: the in-tree .so must be
     mapped into this process."""
