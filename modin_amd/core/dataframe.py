@@ -15,7 +15,6 @@ from __future__ import annotations
 import numpy as np
 import pandas
 
-from .. import config
 from . import lib
 from .partition import DeviceBlock, HipDataframePartition
 from .partition_manager import HipDataframePartitionManager

@@ -21,13 +21,12 @@ from __future__ import annotations
 
 import math
 
-import numpy as np
 import pandas
 
 from .. import config
 from ..distributed import maybe_allreduce_table
 from . import lib
-from .partition import DeviceBlock, HipDataframePartition
+from .partition import HipDataframePartition
 
 
 def compute_chunksize(n: int, num_splits: int, min_size: int) -> int:

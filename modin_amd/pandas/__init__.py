@@ -14,7 +14,6 @@ Reductions return pandas.Series (the reference's API layer also lowers
 
 from __future__ import annotations
 
-import numpy as np
 import pandas
 
 from ..core import lib
