@@ -1160,6 +1160,58 @@ int main(int argc, char** argv) {
                            growcnt);
       }, 10.0 * n);
     };
+    // chunked pipeline: scatter+agg per ~CHUNK_ROWS slice, same scratch —
+    // tests Infinity-Cache absorption of the bucket payload (L3-resident
+    // between P1 and P2; HBM should only see the input reads)
+    auto chunked = [&](int64_t chunk_rows) {
+      snprintf(nm, sizeof nm, "chunked %lldM RL=%d", (long long)(chunk_rows / 1000000), RL);
+      const int64_t cap_per_bucket =
+          ((int64_t)(chunk_rows * 1.3 / nb) + 4096 + 63) & ~63LL;
+      std::vector<unsigned> ccur((size_t)nb);
+      std::vector<Work> cwork;
+      for (int b = 0; b < nb; ++b) {
+        ccur[b] = (unsigned)(b * cap_per_bucket);
+        for (int64_t done = 0; done < cap_per_bucket; done += (1 << 21))
+          cwork.push_back(Work{(int64_t)b * cap_per_bucket + done, b,
+                               (int32_t)std::min<int64_t>(1 << 21,
+                                                          cap_per_bucket - done)});
+      }
+      if ((int64_t)nb * cap_per_bucket > alloc_rows) { printf("%s skip (alloc)\n", nm); return; }
+      Work* d_cw;
+      unsigned* d_ccur_init;
+      CHECK(hipMalloc(&d_cw, cwork.size() * sizeof(Work)));
+      CHECK(hipMemcpy(d_cw, cwork.data(), cwork.size() * sizeof(Work),
+                      hipMemcpyHostToDevice));
+      CHECK(hipMalloc(&d_ccur_init, nb * 4));
+      CHECK(hipMemcpy(d_ccur_init, ccur.data(), nb * 4, hipMemcpyHostToDevice));
+      // slack regions beyond each chunk's fill are read by the agg: make
+      // sure they hold in-range lowkeys (first touch would be garbage)
+      CHECK(hipMemset(rk, 0, alloc_rows * 2));
+      auto nop = [] {};
+      run(nm, 2, nop, [&] {
+        for (int64_t c0 = 0; c0 < n; c0 += chunk_rows) {
+          const int64_t cn = std::min(chunk_rows, n - c0);
+          CHECK(hipMemcpyAsync(d_cur, d_ccur_init, nb * 4,
+                               hipMemcpyDeviceToDevice, 0));
+          const int64_t tile_sz = 256 * 24;
+          const uint32_t grid = (uint32_t)std::min<int64_t>(
+              (cn + tile_sz - 1) / tile_sz, 2048);
+          const uint32_t lds = tile_sz * 12 + nb * 16 + 16;
+          hipLaunchKernelGGL((k_scat_lds3<24, RL, 256>), dim3(grid), dim3(256),
+                             lds, 0, keys + c0, v0 + c0, cn, n_slots, nb,
+                             d_cur, r0, rk);
+          hipLaunchKernelGGL((k_agg_u8<RL, 512>), dim3((uint32_t)cwork.size()),
+                             dim3(512), 0, 0, r0, rk, d_cw, n_slots, gsums,
+                             growcnt);
+        }
+      }, 26.0 * n);
+      CHECK(hipFree(d_cw));
+      CHECK(hipFree(d_ccur_init));
+    };
+    chunked(20000000);
+    chunked(10000000);
+    chunked(50000000);
+    chunked(100000000);
     auto agg8 = [&](auto blkTag) {
       constexpr int BLK = decltype(blkTag)::value;
       snprintf(nm, sizeof nm, "agg_u8 BLK=%d RL=%d (%zu wi)", BLK, RL,
