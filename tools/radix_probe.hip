@@ -1166,14 +1166,18 @@ int main(int argc, char** argv) {
     auto chunked = [&](int64_t chunk_rows) {
       snprintf(nm, sizeof nm, "chunked %lldM RL=%d", (long long)(chunk_rows / 1000000), RL);
       const int64_t cap_per_bucket =
-          ((int64_t)(chunk_rows * 1.3 / nb) + 4096 + 63) & ~63LL;
+          ((int64_t)(chunk_rows * 1.15 / nb) + 2048 + 63) & ~63LL;
+      // work granularity sized so each chunk's agg fills the chip (~512+
+      // blocks): items of cap/ceil(cap*nb/что... just target >= 512 items
+      int64_t wchunk = (cap_per_bucket * nb + 511) / 512;
+      wchunk = std::max<int64_t>(64, std::min<int64_t>(wchunk, 1 << 21)) & ~1LL;
       std::vector<unsigned> ccur((size_t)nb);
       std::vector<Work> cwork;
       for (int b = 0; b < nb; ++b) {
         ccur[b] = (unsigned)(b * cap_per_bucket);
-        for (int64_t done = 0; done < cap_per_bucket; done += (1 << 21))
+        for (int64_t done = 0; done < cap_per_bucket; done += wchunk)
           cwork.push_back(Work{(int64_t)b * cap_per_bucket + done, b,
-                               (int32_t)std::min<int64_t>(1 << 21,
+                               (int32_t)std::min<int64_t>(wchunk,
                                                           cap_per_bucket - done)});
       }
       if ((int64_t)nb * cap_per_bucket > alloc_rows) { printf("%s skip (alloc)\n", nm); return; }
