@@ -177,6 +177,59 @@ def alloc_table_torch(nvals: int, n_slots: int, want_counts: bool,
         (counts.data_ptr() if counts is not None else 0)
 
 
+def exchange_splits(t, send_counts):
+    """All-to-all row exchange — the communication step of the
+    range/hash-partitioning shuffle (SURVEY §8e option (ii): the device form
+    of ``PartitionManager.shuffle_partitions``, partition_manager.py:1937,
+    whose p-way split becomes a P-rank exchange over xGMI).
+
+    ``t``: 1-D tensor laid out as P contiguous spans, span i destined for
+    rank i with ``send_counts[i]`` elements.  Returns (recv_tensor,
+    recv_counts).  nccl uses all_to_all_single (pairwise xGMI); gloo (CPU
+    test tier) emulates with ordered pairwise send/recv.
+    """
+    import torch
+    import torch.distributed as dist
+    P = world_size()
+    r = rank()
+    send_counts = [int(c) for c in send_counts]
+    assert len(send_counts) == P and sum(send_counts) == t.numel()
+    cnt = torch.tensor(send_counts, dtype=torch.int64, device=_state["device"])
+    all_cnt = [torch.zeros_like(cnt) for _ in range(P)]
+    dist.all_gather(all_cnt, cnt)
+    recv_counts = [int(all_cnt[src][r].item()) for src in range(P)]
+    out = torch.empty(sum(recv_counts), dtype=t.dtype, device=t.device)
+    if _state["backend"] == "nccl":
+        dist.all_to_all_single(out, t, recv_counts, send_counts)
+        torch.cuda.synchronize()
+        return out, recv_counts
+    # gloo fallback: keep own span, pairwise exchange the rest in ring order
+    send_offs = [0]
+    for c in send_counts:
+        send_offs.append(send_offs[-1] + c)
+    recv_offs = [0]
+    for c in recv_counts:
+        recv_offs.append(recv_offs[-1] + c)
+    out[recv_offs[r]:recv_offs[r + 1]] = t[send_offs[r]:send_offs[r + 1]]
+    for step in range(1, P):
+        to = (r + step) % P
+        frm = (r - step) % P
+        sbuf = t[send_offs[to]:send_offs[to + 1]].contiguous()
+        rbuf = torch.empty(recv_counts[frm], dtype=t.dtype, device=t.device)
+        if r % 2 == 0:  # deadlock-free ordering; both sides know the counts
+            if sbuf.numel():
+                dist.send(sbuf, to)
+            if rbuf.numel():
+                dist.recv(rbuf, frm)
+        else:
+            if rbuf.numel():
+                dist.recv(rbuf, frm)
+            if sbuf.numel():
+                dist.send(sbuf, to)
+        out[recv_offs[frm]:recv_offs[frm + 1]] = rbuf
+    return out, recv_counts
+
+
 def maybe_allreduce_table(table) -> None:
     """RCCL all-reduce of the dense groupby table (reduce phase across GPUs)."""
     if not is_active():

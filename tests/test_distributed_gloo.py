@@ -101,6 +101,59 @@ def _worker(rank, world, port, fail_q):
         fail_q.put(f"rank {rank}: {e}\n{traceback.format_exc()}")
 
 
+def _shuffle_worker(rank, world, port, fail_q):
+    try:
+        os.environ.update({
+            "RANK": str(rank), "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        })
+        import torch
+        import modin_amd.distributed as dist_mod
+        assert dist_mod.init_from_env(backend="gloo", gpu=False)
+        rng = np.random.default_rng(7)  # same stream on all ranks
+        n = 5000
+        keys_all = rng.integers(0, 10**6, (world, n)).astype(np.int64)
+        mine = keys_all[rank]
+        # hash-partition by destination rank, spans ordered by dest — the
+        # host side of the shuffle split (shuffle_partitions device form)
+        dest = mine % world
+        order = np.argsort(dest, kind="stable")
+        counts = np.bincount(dest, minlength=world)
+        t = torch.tensor(mine[order])
+        out, recv_counts = dist_mod.exchange_splits(t, counts.tolist())
+        got = np.sort(out.numpy())
+        # expectation: every key (from any rank) whose hash lands on me
+        expect = np.sort(np.concatenate(
+            [keys_all[src][keys_all[src] % world == rank]
+             for src in range(world)]))
+        np.testing.assert_array_equal(got, expect)
+        assert sum(recv_counts) == expect.size
+        dist_mod.shutdown()
+    except Exception as e:  # pragma: no cover
+        import traceback
+        fail_q.put(f"rank {rank}: {e}\n{traceback.format_exc()}")
+
+
+@pytest.mark.timeout(120)
+@pytest.mark.parametrize("world", [2, 3])
+def test_gloo_shuffle_exchange(world):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29520 + world
+    procs = [ctx.Process(target=_shuffle_worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=110)
+    errs = []
+    while not q.empty():
+        errs.append(q.get())
+    assert not errs, "\n".join(errs)
+    assert all(p.exitcode == 0 for p in procs)
+
+
 @pytest.mark.timeout(120)
 def test_gloo_world2_exchange_logic():
     ctx = mp.get_context("spawn")
