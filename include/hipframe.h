@@ -162,6 +162,26 @@ int hf_groupby_accum(const hf_col* keys,            /* HF_INT64, len n        */
                      uintptr_t sums, uintptr_t rowcnt, uintptr_t counts);
 
 int hf_fill_f64(uintptr_t dptr, double value, int64_t n);
+int hf_fill_i64(uintptr_t dptr, int64_t value, int64_t n);
+
+/* ---- hash-table groupby (unbounded key ranges, single rank) ----
+ * Open-addressing table of H power-of-2 slots (+1 special slot for the
+ * INT64_MIN sentinel key): tkey i64[H+1] initialised to INT64_MIN via
+ * hf_fill_i64; sums f64[nvals][H+1] to the agg identity; rowcnt/counts
+ * i64[...] zeroed.  Insert = splitmix64 probe + atomicCAS claim; a probe
+ * sweep that wraps the table reports "hash table full" (caller grows H and
+ * retries).  Compaction filters present slots, SORTS the surviving keys
+ * (the wide radix sort), and gathers — same output shape as the dense
+ * compact.  Cross-rank hash groupby needs the shuffle exchange (later
+ * round); the dense table remains the multi-GPU path. */
+int hf_groupby_hash_accum(const hf_col* keys, const hf_col* const* vals,
+                          int nvals, int agg_op, int64_t H,
+                          uintptr_t tkey, uintptr_t sums, uintptr_t rowcnt,
+                          uintptr_t counts);
+int hf_groupby_hash_compact(uintptr_t tkey, uintptr_t sums, uintptr_t rowcnt,
+                            uintptr_t counts, int nvals, int64_t H,
+                            hf_col** out_keys, hf_col** out_sums,
+                            hf_col** out_counts, int64_t* n_groups);
 
 /* out[i] = count[i] != 0 ? val[i] : NaN — min/max of an empty (all-NaN)
  * group is NaN in pandas. */

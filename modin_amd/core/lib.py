@@ -106,6 +106,20 @@ def load() -> ct.CDLL:
                                             ct.c_int64, ct.c_size_t,
                                             ct.c_size_t, ct.c_size_t]),
             "hf_fill_f64": (ct.c_int, [ct.c_size_t, ct.c_double, ct.c_int64]),
+            "hf_fill_i64": (ct.c_int, [ct.c_size_t, ct.c_int64, ct.c_int64]),
+            "hf_groupby_hash_accum": (ct.c_int, [ct.c_void_p,
+                                                 ct.POINTER(ct.c_void_p),
+                                                 ct.c_int, ct.c_int,
+                                                 ct.c_int64, ct.c_size_t,
+                                                 ct.c_size_t, ct.c_size_t,
+                                                 ct.c_size_t]),
+            "hf_groupby_hash_compact": (ct.c_int, [ct.c_size_t, ct.c_size_t,
+                                                   ct.c_size_t, ct.c_size_t,
+                                                   ct.c_int, ct.c_int64,
+                                                   ct.POINTER(ct.c_void_p),
+                                                   ct.POINTER(ct.c_void_p),
+                                                   ct.POINTER(ct.c_void_p),
+                                                   ct.POINTER(ct.c_int64)]),
             "hf_sort_perm": (ct.c_int, [ct.c_void_p, ct.c_int,
                                         ct.POINTER(ct.c_void_p)]),
             "hf_fixup_empty": (ct.c_int, [ct.c_void_p, ct.c_void_p,
@@ -162,7 +176,8 @@ def exported_symbols():
         "hf_col_dtype", "hf_col_dptr", "hf_alloc_raw", "hf_free_raw",
         "hf_memset_raw", "hf_map_scalar", "hf_map_scalar_i64", "hf_binary",
         "hf_reduce", "hf_groupby_accum", "hf_groupby_compact", "hf_fill_f64",
-        "hf_fixup_empty", "hf_sort_perm",
+        "hf_fixup_empty", "hf_sort_perm", "hf_fill_i64",
+        "hf_groupby_hash_accum", "hf_groupby_hash_compact",
         "hf_col_concat", "hf_col_slice", "hf_join_build", "hf_join_free", "hf_join_probe",
         "hf_gather", "hf_compare_scalar", "hf_filter_plan", "hf_filter_apply",
         "hf_filter_iota", "hf_filter_plan_free", "hf_profiling",
@@ -362,6 +377,41 @@ def groupby_accum(keys: ColumnRef, vals: list, agg_op: int, key_min: int,
 def fill_f64(dptr: int, value: float, n: int) -> None:
     ensure_ready()
     _check(load().hf_fill_f64(dptr, value, n), "hf_fill_f64")
+
+
+def fill_i64(dptr: int, value: int, n: int) -> None:
+    ensure_ready()
+    _check(load().hf_fill_i64(dptr, value, n), "hf_fill_i64")
+
+
+def groupby_hash_accum(keys: ColumnRef, vals: list, agg_op: int, H: int,
+                       tkey: int, sums: int, rowcnt: int, counts: int) -> None:
+    ensure_ready()
+    arr = (ct.c_void_p * max(len(vals), 1))(*[v.handle for v in vals])
+    _check(load().hf_groupby_hash_accum(keys.handle, arr, len(vals), agg_op,
+                                        H, tkey, sums, rowcnt, counts),
+           "hf_groupby_hash_accum")
+
+
+def groupby_hash_compact(tkey: int, sums: int, rowcnt: int, counts: int,
+                         nvals: int, H: int):
+    ensure_ready()
+    out_keys = ct.c_void_p()
+    out_sums = (ct.c_void_p * max(nvals, 1))()
+    out_counts = (ct.c_void_p * max(nvals, 1))()
+    n_groups = ct.c_int64(0)
+    _check(load().hf_groupby_hash_compact(tkey, sums, rowcnt, counts, nvals,
+                                          H, ct.byref(out_keys), out_sums,
+                                          out_counts if counts else None,
+                                          ct.byref(n_groups)),
+           "hf_groupby_hash_compact")
+    n = n_groups.value
+    kcol = _wrap(out_keys, n, HF_INT64)
+    scols = [_wrap(ct.c_void_p(out_sums[c]), n, HF_FLOAT64)
+             for c in range(nvals)]
+    ccols = ([_wrap(ct.c_void_p(out_counts[c]), n, HF_INT64)
+              for c in range(nvals)] if counts else None)
+    return kcol, scols, ccols, n
 
 
 def fixup_empty(val: ColumnRef, cnt: ColumnRef) -> ColumnRef:

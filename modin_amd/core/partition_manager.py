@@ -222,11 +222,14 @@ class HipDataframePartitionManager:
         if n_slots < 0:
             n_slots = 0
         if n_slots > config.MaxGroupbySlots.get():
-            raise lib.HfError(
-                f"groupby key range {n_slots} exceeds MaxGroupbySlots "
-                f"({config.MaxGroupbySlots.get()}): the dense-key table does "
-                "not apply; hash aggregation is a later round"
-            )
+            from ..distributed import is_active
+            if is_active():
+                raise lib.HfError(
+                    f"groupby key range {n_slots} exceeds MaxGroupbySlots: "
+                    "the multi-GPU hash groupby needs the shuffle exchange — "
+                    "later round (dense ranges all-reduce fine)")
+            return cls._groupby_hash(key_cols, val_cols_per_part, total_rows,
+                                     want_counts, agg_op)
         n_slots = max(n_slots, 1)
         table = GroupbyTable(len(val_names), kmin, n_slots, want_counts,
                              agg_op)
@@ -240,6 +243,50 @@ class HipDataframePartitionManager:
         )
         table.free()
         return keys, sums, counts, n
+
+
+    @classmethod
+    def _groupby_hash(cls, key_cols, val_cols_per_part, total_rows,
+                      want_counts, agg_op):
+        """Open-addressing hash groupby for unbounded key ranges (single
+        rank); grows the table 4x and retries on overflow."""
+        INT64_MIN = -(1 << 63)
+        nv = len(val_cols_per_part[0]) if val_cols_per_part else 0
+        H = 1 << max(12, min(26, (2 * min(total_rows, 1 << 25)).bit_length()))
+        while True:
+            L = H + 1
+            tkey = lib.alloc_raw(8 * L)
+            sums = lib.alloc_raw(8 * nv * L) if nv else lib.alloc_raw(8)
+            rowcnt = lib.alloc_raw(8 * L)
+            counts = lib.alloc_raw(8 * nv * L) if (want_counts and nv) else 0
+            try:
+                lib.fill_i64(tkey, INT64_MIN, L)
+                init = lib.AGG_IDENTITY[agg_op]
+                if nv:
+                    if init == 0.0:
+                        lib.memset_raw(sums, 0, 8 * nv * L)
+                    else:
+                        lib.fill_f64(sums, init, nv * L)
+                lib.memset_raw(rowcnt, 0, 8 * L)
+                if counts:
+                    lib.memset_raw(counts, 0, 8 * nv * L)
+                for kcol, vals in zip(key_cols, val_cols_per_part):
+                    if kcol.length:
+                        lib.groupby_hash_accum(kcol, vals, agg_op, H, tkey,
+                                               sums, rowcnt, counts)
+                return (*lib.groupby_hash_compact(tkey, sums, rowcnt, counts,
+                                                  nv, H),)
+            except lib.HfError as e:
+                if "hash table full" in str(e) and H < (1 << 27):
+                    H <<= 2
+                    continue
+                raise
+            finally:
+                lib.free_raw(tkey)
+                lib.free_raw(sums)
+                lib.free_raw(rowcnt)
+                if counts:
+                    lib.free_raw(counts)
 
 
 def maybe_allreduce_keyrange(kmin, kmax):

@@ -83,6 +83,7 @@ constexpr int64_t SCRATCH_GB_ERR = 128;
 constexpr int64_t SCRATCH_NGROUPS = 136;
 constexpr int64_t SCRATCH_JOIN_HIST_ERR = 144;
 constexpr int64_t SCRATCH_JOIN_FIXUP_ERR = 152;
+constexpr int64_t SCRATCH_HASH_FULL = 160;
 
 int set_err(int code, const char* where, const char* what) {
   g_err = std::string(where) + ": " + what;
@@ -916,6 +917,70 @@ __global__ void __launch_bounds__(512) k_gb_dense(
   }
 }
 
+// ---- hash-table groupby accumulate (unbounded key ranges) ----
+
+__device__ __forceinline__ uint64_t hash_mix64(uint64_t x) {
+  // splitmix64 finalizer
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+
+constexpr long long HASH_EMPTY = 0x8000000000000000LL;  // INT64_MIN sentinel
+
+template <int NVALS, bool COUNTS, int AOP>
+__global__ void __launch_bounds__(BLOCK) k_gb_hash_accum(
+    const int64_t* __restrict__ keys, GbPtrs ptrs, int64_t n, int64_t H,
+    long long* __restrict__ tkey, double* __restrict__ sums,
+    unsigned long long* __restrict__ rowcnt,
+    unsigned long long* __restrict__ counts,
+    unsigned long long* __restrict__ err_full) {
+  const int64_t stride_len = H + 1;  // +1: the INT64_MIN special slot
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const long long k = keys[i];
+    int64_t slot;
+    if (k == HASH_EMPTY) {
+      slot = H;
+    } else {
+      uint64_t h = hash_mix64((uint64_t)k) & (uint64_t)(H - 1);
+      slot = -1;
+      for (int64_t probes = 0; probes < H; ++probes) {
+        const long long cur = (long long)atomicCAS(
+            (unsigned long long*)&tkey[h], (unsigned long long)HASH_EMPTY,
+            (unsigned long long)k);
+        if (cur == HASH_EMPTY || cur == k) {
+          slot = (int64_t)h;
+          break;
+        }
+        h = (h + 1) & (uint64_t)(H - 1);
+      }
+      if (slot < 0) {
+        atomicAdd(err_full, 1ULL);
+        continue;
+      }
+    }
+    atomicAdd(&rowcnt[slot], 1ULL);
+#pragma unroll
+    for (int c = 0; c < NVALS; ++c) {
+      const double v = ptrs.vals[c][i];
+      if (v == v) {
+        glob_slot_agg<AOP>(&sums[(int64_t)c * stride_len + slot], v);
+        if (COUNTS) atomicAdd(&counts[(int64_t)c * stride_len + slot], 1ULL);
+      }
+    }
+  }
+}
+
+__global__ void __launch_bounds__(BLOCK) k_fill_i64(int64_t* __restrict__ p,
+                                                    int64_t v, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) p[i] = v;
+}
+
 // ---- compaction: present slots (rowcnt>0) -> ascending keys + columns ----
 // Fixed tile partitioning so prefix order == slot order.
 constexpr int COMPACT_TILE = 4096;  // slots per tile, one 256-thread block/tile
@@ -1341,6 +1406,94 @@ __global__ void __launch_bounds__(BLOCK) k_sort_scatter(
       __builtin_amdgcn_wave_barrier();
     }
   }
+}
+
+// ---- wide variant: u64 shifted keys + u32 origins as two arrays (for key
+// spans beyond 2^27; up to 8 digit passes) ----
+
+__global__ void __launch_bounds__(BLOCK) k_sort_pack_wide(
+    const int64_t* __restrict__ keys, int64_t n, int64_t key_min,
+    uint64_t key_span, int ascending, unsigned long long* __restrict__ karr,
+    unsigned* __restrict__ iarr) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const uint64_t k = (uint64_t)(keys[i] - key_min);
+    karr[i] = ascending ? k : key_span - k;
+    iarr[i] = (unsigned)i;
+  }
+}
+
+__global__ void __launch_bounds__(BLOCK) k_sort_count_wide(
+    const unsigned long long* __restrict__ karr, int64_t n, int shift,
+    unsigned* __restrict__ C, int64_t ntiles) {
+  __shared__ unsigned hist[SORT_WPB][256];
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  for (int64_t tile = (int64_t)blockIdx.x * SORT_WPB + wave; tile < ntiles;
+       tile += (int64_t)gridDim.x * SORT_WPB) {
+    for (int d = lane; d < 256; d += 64) hist[wave][d] = 0;
+    __builtin_amdgcn_wave_barrier();
+    const int64_t t0 = tile * SORT_TILE;
+#pragma unroll 4
+    for (int j = 0; j < SORT_RPT; ++j) {
+      const int64_t row = t0 + (int64_t)j * 64 + lane;
+      if (row < n) {
+        const unsigned d = (unsigned)(karr[row] >> shift) & 255u;
+        atomicAdd(&hist[wave][d], 1u);
+      }
+    }
+    __builtin_amdgcn_wave_barrier();
+    for (int d = lane; d < 256; d += 64) C[tile * 256 + d] = hist[wave][d];
+  }
+}
+
+__global__ void __launch_bounds__(BLOCK) k_sort_scatter_wide(
+    const unsigned long long* __restrict__ karr,
+    const unsigned* __restrict__ iarr, int64_t n, int shift,
+    const unsigned long long* __restrict__ offs, int64_t ntiles,
+    unsigned long long* __restrict__ karr_out, unsigned* __restrict__ iarr_out) {
+  __shared__ unsigned long long base[SORT_WPB][256];
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  for (int64_t tile = (int64_t)blockIdx.x * SORT_WPB + wave; tile < ntiles;
+       tile += (int64_t)gridDim.x * SORT_WPB) {
+    for (int d = lane; d < 256; d += 64)
+      base[wave][d] = offs[(int64_t)d * ntiles + tile];
+    __builtin_amdgcn_wave_barrier();
+    const int64_t t0 = tile * SORT_TILE;
+    for (int j = 0; j < SORT_RPT; ++j) {
+      const int64_t row = t0 + (int64_t)j * 64 + lane;
+      const bool valid = row < n;
+      const unsigned long long k = valid ? karr[row] : 0;
+      const unsigned idx = valid ? iarr[row] : 0;
+      const unsigned d = (unsigned)(k >> shift) & 255u;
+      unsigned long long exec = __ballot(valid);
+      unsigned long long pos = 0;
+      while (exec) {
+        const int leader = __ffsll((long long)exec) - 1;
+        const unsigned dl = (unsigned)__shfl((int)d, leader);
+        const unsigned long long members = __ballot(valid && d == dl);
+        if (valid && d == dl) {
+          const unsigned rank =
+              (unsigned)__popcll(members & ((1ULL << lane) - 1ULL));
+          pos = base[wave][dl] + rank;
+        }
+        if (lane == leader) base[wave][dl] += __popcll(members);
+        exec &= ~members;
+      }
+      if (valid) {
+        karr_out[pos] = k;
+        iarr_out[pos] = idx;
+      }
+      __builtin_amdgcn_wave_barrier();
+    }
+  }
+}
+
+__global__ void __launch_bounds__(BLOCK) k_sort_unpack_wide(
+    const unsigned* __restrict__ iarr, int64_t n, int64_t* __restrict__ perm) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) perm[i] = (int64_t)iarr[i];
 }
 
 __global__ void __launch_bounds__(BLOCK) k_sort_unpack(
@@ -2231,6 +2384,129 @@ int hf_col_concat(const hf_col* const* cols, int ncols, hf_col** out) {
   return HF_OK;
 }
 
+int hf_fill_i64(uintptr_t dptr, int64_t value, int64_t n) {
+  HF_NEED_INIT("hf_fill_i64");
+  if (n <= 0) return HF_OK;
+  return timed_launch("fill_i64", [&] {
+    hipLaunchKernelGGL(k_fill_i64, dim3((uint32_t)grid_for(n)), dim3(BLOCK), 0,
+                       g.stream, (int64_t*)dptr, value, n);
+  });
+}
+
+int hf_groupby_hash_accum(const hf_col* keys, const hf_col* const* vals,
+                          int nvals, int agg_op, int64_t H,
+                          uintptr_t tkey, uintptr_t sums, uintptr_t rowcnt,
+                          uintptr_t counts) {
+  HF_NEED_INIT("hf_groupby_hash_accum");
+  if (!keys || nvals < 0 || nvals > GB_MAX_VALS || H < 2 || (H & (H - 1)))
+    return set_err(HF_ERR_ARG, "hf_groupby_hash_accum",
+                   "bad args (H power of 2, nvals<=8)");
+  if (keys->dtype != HF_INT64)
+    return set_err(HF_ERR_ARG, "hf_groupby_hash_accum", "keys must be int64");
+  GbPtrs ptrs{};
+  for (int c = 0; c < nvals; ++c) {
+    if (!vals[c] || vals[c]->dtype != HF_FLOAT64 || vals[c]->len != keys->len)
+      return set_err(HF_ERR_ARG, "hf_groupby_hash_accum",
+                     "vals must be float64 columns of keys' length");
+    ptrs.vals[c] = (const double*)vals[c]->dptr;
+  }
+  const int64_t n = keys->len;
+  if (n == 0) return HF_OK;
+  unsigned long long* d_full =
+      (unsigned long long*)((char*)g.d_scratch + SCRATCH_HASH_FULL);
+  const bool cnt = counts != 0;
+  auto launch = [&](auto nvTag, auto cTag, auto aTag) {
+    constexpr int NV = decltype(nvTag)::value;
+    constexpr bool C = decltype(cTag)::value;
+    constexpr int A = decltype(aTag)::value;
+    return timed_launch("gb_hash_accum", [&] {
+      hipLaunchKernelGGL((k_gb_hash_accum<NV, C, A>),
+                         dim3((uint32_t)grid_for(n)), dim3(BLOCK), 0, g.stream,
+                         (const int64_t*)keys->dptr, ptrs, n, H,
+                         (long long*)tkey, (double*)sums,
+                         (unsigned long long*)rowcnt,
+                         (unsigned long long*)counts, d_full);
+    });
+  };
+  auto withA = [&](auto nvTag, auto cTag) {
+    return agg_op == HF_AGG_SUM
+               ? launch(nvTag, cTag, std::integral_constant<int, HF_AGG_SUM>{})
+           : agg_op == HF_AGG_MIN
+               ? launch(nvTag, cTag, std::integral_constant<int, HF_AGG_MIN>{})
+               : launch(nvTag, cTag,
+                        std::integral_constant<int, HF_AGG_MAX>{});
+  };
+  switch (nvals) {
+#define HF_GH_CASE(NV)                                                          case NV:                                                                        return cnt ? withA(std::integral_constant<int, NV>{},                                            std::integral_constant<bool, true>{})                                 : withA(std::integral_constant<int, NV>{},                                            std::integral_constant<bool, false>{});
+    HF_GH_CASE(0) HF_GH_CASE(1) HF_GH_CASE(2) HF_GH_CASE(3) HF_GH_CASE(4)
+    HF_GH_CASE(5) HF_GH_CASE(6) HF_GH_CASE(7) HF_GH_CASE(8)
+#undef HF_GH_CASE
+  }
+  return set_err(HF_ERR_ARG, "hf_groupby_hash_accum", "nvals out of range");
+}
+
+int hf_groupby_hash_compact(uintptr_t tkey, uintptr_t sums, uintptr_t rowcnt,
+                            uintptr_t counts, int nvals, int64_t H,
+                            hf_col** out_keys, hf_col** out_sums,
+                            hf_col** out_counts, int64_t* n_groups) {
+  HF_NEED_INIT("hf_groupby_hash_compact");
+  if (!tkey || !rowcnt || nvals < 0 || nvals > GB_MAX_VALS || !out_keys ||
+      !n_groups)
+    return set_err(HF_ERR_ARG, "hf_groupby_hash_compact", "bad args");
+  // surface "table full" from the accumulate phase
+  unsigned long long h_full = 0;
+  unsigned long long* d_full =
+      (unsigned long long*)((char*)g.d_scratch + SCRATCH_HASH_FULL);
+  HF_HIP("hf_groupby_hash_compact",
+         hipMemcpyAsync(&h_full, d_full, 8, hipMemcpyDeviceToHost, g.stream));
+  HF_HIP("hf_groupby_hash_compact", hipStreamSynchronize(g.stream));
+  if (h_full) {
+    hipMemsetAsync(d_full, 0, 8, g.stream);
+    return set_err(HF_ERR_UNSUPPORTED, "hf_groupby_hash_compact",
+                   "hash table full — retry with a larger H");
+  }
+  const int64_t L = H + 1;
+  const int gpu = g.gpu;
+  // filter present slots (rowcnt doubles as the nonzero mask)
+  hf_col mask_view{(void*)rowcnt, L, HF_INT64, gpu};
+  hf_filterplan* plan = nullptr;
+  int64_t kept = 0;
+  int rc = hf_filter_plan(&mask_view, &plan, &kept);
+  if (rc != HF_OK) return rc;
+  hf_col tkey_view{(void*)tkey, L, HF_INT64, gpu};
+  hf_col* ckeys = nullptr;
+  rc = hf_filter_apply(plan, &tkey_view, &ckeys);
+  std::vector<hf_col*> csums((size_t)(nvals ? nvals : 1), nullptr);
+  std::vector<hf_col*> ccnts((size_t)(nvals ? nvals : 1), nullptr);
+  for (int c = 0; c < nvals && rc == HF_OK; ++c) {
+    hf_col sv{(char*)sums + (int64_t)c * L * 8, L, HF_FLOAT64, gpu};
+    rc = hf_filter_apply(plan, &sv, &csums[c]);
+    if (rc == HF_OK && counts) {
+      hf_col cv{(char*)counts + (int64_t)c * L * 8, L, HF_INT64, gpu};
+      rc = hf_filter_apply(plan, &cv, &ccnts[c]);
+    }
+  }
+  hf_filter_plan_free(plan);
+  // sort surviving keys (wide radix handles any int64 range) and gather
+  hf_col* perm = nullptr;
+  if (rc == HF_OK) rc = hf_sort_perm(ckeys, 1, &perm);
+  if (rc == HF_OK) rc = hf_gather(ckeys, perm, out_keys);
+  for (int c = 0; c < nvals && rc == HF_OK; ++c) {
+    rc = hf_gather(csums[c], perm, &out_sums[c]);
+    if (rc == HF_OK && counts && out_counts)
+      rc = hf_gather(ccnts[c], perm, &out_counts[c]);
+  }
+  hf_col_free(ckeys);
+  hf_col_free(perm);
+  for (int c = 0; c < nvals; ++c) {
+    hf_col_free(csums[c]);
+    hf_col_free(ccnts[c]);
+  }
+  if (rc != HF_OK) return rc;
+  *n_groups = kept;
+  return HF_OK;
+}
+
 int hf_col_slice(const hf_col* col, int64_t start, int64_t len, hf_col** out) {
   HF_NEED_INIT("hf_col_slice");
   if (!col || !out || start < 0 || len < 0 || start + len > col->len)
@@ -2517,11 +2793,77 @@ int hf_sort_perm(const hf_col* keys, int ascending, hf_col** out_perm) {
   hf_reduce_result r;
   rc = hf_reduce(keys, &r);
   if (rc != HF_OK) return rc;
-  const int64_t key_min = r.imn, span = r.imx - r.imn;
-  if (span >= (1LL << 27))
-    return set_err(HF_ERR_UNSUPPORTED, "hf_sort_perm",
-                   "key range too large for the bounded-range radix sort "
-                   "(general 64-bit keys are a later round)");
+  const int64_t key_min = r.imn;
+  const uint64_t span = (uint64_t)r.imx - (uint64_t)r.imn;  // mod-2^64 safe
+  if (span >= (1ULL << 27)) {
+    // wide path: u64 shifted keys + u32 origins, up to 8 digit passes
+    int bits = 0;
+    while (bits < 64 && (span >> bits) != 0) ++bits;
+    const int passes = (bits + 7) / 8;
+    const int64_t ntiles = (n + SORT_TILE - 1) / SORT_TILE;
+    const int64_t L = ntiles * 256;
+    unsigned long long *kA = nullptr, *kB = nullptr, *offs = nullptr;
+    unsigned *iA = nullptr, *iB = nullptr, *C = nullptr, *CT = nullptr;
+    int64_t* scan_tiles = nullptr;
+    const int64_t scan_tiles_n = (L + JOIN_TILE - 1) / JOIN_TILE;
+    HF_HIP("hf_sort_perm", dev_alloc((void**)&kA, n * 8, g.stream));
+    HF_HIP("hf_sort_perm", dev_alloc((void**)&kB, n * 8, g.stream));
+    HF_HIP("hf_sort_perm", dev_alloc((void**)&iA, n * 4, g.stream));
+    HF_HIP("hf_sort_perm", dev_alloc((void**)&iB, n * 4, g.stream));
+    HF_HIP("hf_sort_perm", dev_alloc((void**)&C, L * 4, g.stream));
+    HF_HIP("hf_sort_perm", dev_alloc((void**)&CT, L * 4, g.stream));
+    HF_HIP("hf_sort_perm", dev_alloc((void**)&offs, (L + 1) * 8, g.stream));
+    HF_HIP("hf_sort_perm",
+           dev_alloc((void**)&scan_tiles, scan_tiles_n * 8, g.stream));
+    int64_t* d_total = (int64_t*)((char*)g.d_scratch + SCRATCH_NGROUPS);
+    rc = timed_launch("sort_pack", [&] {
+      hipLaunchKernelGGL(k_sort_pack_wide, dim3((uint32_t)grid_for(n)),
+                         dim3(BLOCK), 0, g.stream, (const int64_t*)keys->dptr,
+                         n, key_min, span, ascending, kA, iA);
+    });
+    unsigned long long *kcur = kA, *kalt = kB;
+    unsigned *icur = iA, *ialt = iB;
+    const uint32_t wgrid =
+        (uint32_t)std::min<int64_t>((ntiles + SORT_WPB - 1) / SORT_WPB, 2048);
+    for (int p = 0; p < passes && rc == HF_OK; ++p) {
+      const int shift = 8 * p;
+      rc = timed_launch("sort_pass", [&] {
+        hipLaunchKernelGGL(k_sort_count_wide, dim3(wgrid), dim3(BLOCK), 0,
+                           g.stream, kcur, n, shift, C, ntiles);
+        hipLaunchKernelGGL(k_transpose256,
+                           dim3((uint32_t)((ntiles + 31) / 32), 8), dim3(1024),
+                           0, g.stream, C, CT, ntiles);
+        hipLaunchKernelGGL(k_tile_sums_u32, dim3((uint32_t)scan_tiles_n),
+                           dim3(BLOCK), 0, g.stream, CT, L, scan_tiles);
+        hipLaunchKernelGGL(k_compact_scan, dim3(1), dim3(1024), 0, g.stream,
+                           scan_tiles, scan_tiles_n, d_total);
+        hipLaunchKernelGGL(k_scan_apply_u32, dim3((uint32_t)scan_tiles_n),
+                           dim3(BLOCK), 0, g.stream, CT, L, scan_tiles,
+                           d_total, offs);
+        hipLaunchKernelGGL(k_sort_scatter_wide, dim3(wgrid), dim3(BLOCK), 0,
+                           g.stream, kcur, icur, n, shift, offs, ntiles, kalt,
+                           ialt);
+      });
+      std::swap(kcur, kalt);
+      std::swap(icur, ialt);
+    }
+    if (rc == HF_OK)
+      rc = timed_launch("sort_unpack", [&] {
+        hipLaunchKernelGGL(k_sort_unpack_wide, dim3((uint32_t)grid_for(n)),
+                           dim3(BLOCK), 0, g.stream, icur, n,
+                           (int64_t*)(*out_perm)->dptr);
+      });
+    dev_free(kA, g.stream);
+    dev_free(kB, g.stream);
+    dev_free(iA, g.stream);
+    dev_free(iB, g.stream);
+    dev_free(C, g.stream);
+    dev_free(CT, g.stream);
+    dev_free(offs, g.stream);
+    dev_free(scan_tiles, g.stream);
+    if (rc != HF_OK) { hf_col_free(*out_perm); *out_perm = nullptr; }
+    return rc;
+  }
   int bits = 0;
   while ((span >> bits) != 0) ++bits;
   const int passes = (bits + 7) / 8;

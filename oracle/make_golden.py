@@ -293,6 +293,35 @@ def gen_sort_cases(mpd, rng):
     return cases
 
 
+def gen_hash_groupby_cases(mpd, rng):
+    """Unbounded key ranges (the hash-table groupby path)."""
+    import pandas
+    cases = {}
+
+    def make(name, k, cols):
+        data = {"k": k.astype(np.int64), **cols}
+        mdf, pdf = mpd.DataFrame(dict(data)), pandas.DataFrame(dict(data))
+        arrays = {"in_k": data["k"]}
+        for cn, cv in cols.items():
+            arrays[f"in_{cn}"] = cv
+        for agg in ("sum", "count", "mean", "min", "max"):
+            mres = getattr(mdf.groupby("k"), agg)()
+            pres = getattr(pdf.groupby("k"), agg)()
+            got = _check_vs_pandas(mres, pres)
+            arrays[f"out_{agg}_keys"] = np.asarray(got.index, dtype=np.int64)
+            for cn in cols:
+                arrays[f"out_{agg}_{cn}"] = got[cn].to_numpy()
+        cases[name] = arrays
+
+    n = 4000
+    make("hh_distinct", rng.integers(-2**60, 2**60, n),
+         {"v": rng.random(n)})
+    hugekeys = rng.integers(-2**60, 2**60, 250)
+    make("hh_dups", rng.choice(hugekeys, n), {"v": rng.random(n),
+                                              "w": rng.standard_normal(n)})
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -304,6 +333,7 @@ def main():
     all_cases.update(gen_merge_cases(mpd, rng))
     all_cases.update(gen_filter_cases(mpd, rng))
     all_cases.update(gen_sort_cases(mpd, rng))
+    all_cases.update(gen_hash_groupby_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)
