@@ -252,7 +252,10 @@ class HipDataframePartitionManager:
         rank); grows the table 4x and retries on overflow."""
         INT64_MIN = -(1 << 63)
         nv = len(val_cols_per_part[0]) if val_cols_per_part else 0
-        H = 1 << max(12, min(26, (2 * min(total_rows, 1 << 25)).bit_length()))
+        # start near 2x the row count (capped): avoids doomed attempts on
+        # high-cardinality inputs
+        H = 1 << max(12, min(27, (2 * total_rows - 1).bit_length()
+                             if total_rows else 12))
         while True:
             L = H + 1
             tkey = lib.alloc_raw(8 * L)
@@ -277,9 +280,14 @@ class HipDataframePartitionManager:
                 return (*lib.groupby_hash_compact(tkey, sums, rowcnt, counts,
                                                   nv, H),)
             except lib.HfError as e:
-                if "hash table full" in str(e) and H < (1 << 27):
-                    H <<= 2
-                    continue
+                if "hash table full" in str(e):
+                    if H < (1 << 27):
+                        H <<= 2
+                        continue
+                    raise lib.HfError(
+                        "groupby cardinality exceeds the hash table cap "
+                        "(~134M groups); the sort-based general groupby is a "
+                        "later round") from e
                 raise
             finally:
                 lib.free_raw(tkey)

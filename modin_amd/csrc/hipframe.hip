@@ -947,7 +947,10 @@ __global__ void __launch_bounds__(BLOCK) k_gb_hash_accum(
     } else {
       uint64_t h = hash_mix64((uint64_t)k) & (uint64_t)(H - 1);
       slot = -1;
-      for (int64_t probes = 0; probes < H; ++probes) {
+      // bounded sweep: a chain this long means the table is effectively
+      // full — fail fast so the host can grow and retry
+      const int64_t maxprobe = H < 4096 ? H : 4096;
+      for (int64_t probes = 0; probes < maxprobe; ++probes) {
         const long long cur = (long long)atomicCAS(
             (unsigned long long*)&tkey[h], (unsigned long long)HASH_EMPTY,
             (unsigned long long)k);
