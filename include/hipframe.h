@@ -183,6 +183,17 @@ int hf_groupby_hash_compact(uintptr_t tkey, uintptr_t sums, uintptr_t rowcnt,
                             hf_col** out_keys, hf_col** out_sums,
                             hf_col** out_counts, int64_t* n_groups);
 
+/* ---- sort-based general groupby (any cardinality, single rank) ----
+ * Input columns must already be KEY-SORTED (hf_sort_perm + hf_gather);
+ * runs of equal keys become groups: run heads are detected in place, a
+ * tile scan assigns run ids, and per-run aggregates accumulate with one
+ * device atomic per row (the extreme-cardinality fallback — runs are short
+ * there, so contention is low). */
+int hf_groupby_sorted(const hf_col* sorted_keys, const hf_col* const* vals,
+                      int nvals, int agg_op, int want_counts,
+                      hf_col** out_keys, hf_col** out_sums,
+                      hf_col** out_counts, int64_t* n_groups);
+
 /* out[i] = count[i] != 0 ? val[i] : NaN — min/max of an empty (all-NaN)
  * group is NaN in pandas. */
 int hf_fixup_empty(const hf_col* val, const hf_col* cnt, hf_col** out);

@@ -120,6 +120,13 @@ def load() -> ct.CDLL:
                                                    ct.POINTER(ct.c_void_p),
                                                    ct.POINTER(ct.c_void_p),
                                                    ct.POINTER(ct.c_int64)]),
+            "hf_groupby_sorted": (ct.c_int, [ct.c_void_p,
+                                             ct.POINTER(ct.c_void_p),
+                                             ct.c_int, ct.c_int, ct.c_int,
+                                             ct.POINTER(ct.c_void_p),
+                                             ct.POINTER(ct.c_void_p),
+                                             ct.POINTER(ct.c_void_p),
+                                             ct.POINTER(ct.c_int64)]),
             "hf_sort_perm": (ct.c_int, [ct.c_void_p, ct.c_int,
                                         ct.POINTER(ct.c_void_p)]),
             "hf_fixup_empty": (ct.c_int, [ct.c_void_p, ct.c_void_p,
@@ -178,6 +185,7 @@ def exported_symbols():
         "hf_reduce", "hf_groupby_accum", "hf_groupby_compact", "hf_fill_f64",
         "hf_fixup_empty", "hf_sort_perm", "hf_fill_i64",
         "hf_groupby_hash_accum", "hf_groupby_hash_compact",
+        "hf_groupby_sorted",
         "hf_col_concat", "hf_col_slice", "hf_join_build", "hf_join_free", "hf_join_probe",
         "hf_gather", "hf_compare_scalar", "hf_filter_plan", "hf_filter_apply",
         "hf_filter_iota", "hf_filter_plan_free", "hf_profiling",
@@ -411,6 +419,32 @@ def groupby_hash_compact(tkey: int, sums: int, rowcnt: int, counts: int,
              for c in range(nvals)]
     ccols = ([_wrap(ct.c_void_p(out_counts[c]), n, HF_INT64)
               for c in range(nvals)] if counts else None)
+    return kcol, scols, ccols, n
+
+
+def groupby_sorted(sorted_keys: ColumnRef, vals: list, agg_op: int,
+                   want_counts: bool):
+    """Segmented aggregation over KEY-SORTED rows — the unbounded-cardinality
+    groupby (runs of equal keys are groups).  Returns (keys, sums, counts, n)
+    like groupby_hash_compact; keys come out ascending because the input is."""
+    ensure_ready()
+    nv = len(vals)
+    arr = (ct.c_void_p * max(nv, 1))(*[v.handle for v in vals])
+    out_keys = ct.c_void_p()
+    out_sums = (ct.c_void_p * max(nv, 1))()
+    out_counts = (ct.c_void_p * max(nv, 1))()
+    n_groups = ct.c_int64(0)
+    _check(load().hf_groupby_sorted(sorted_keys.handle, arr, nv, agg_op,
+                                    1 if want_counts else 0,
+                                    ct.byref(out_keys), out_sums,
+                                    out_counts if want_counts else None,
+                                    ct.byref(n_groups)),
+           "hf_groupby_sorted")
+    n = n_groups.value
+    kcol = _wrap(out_keys, n, HF_INT64)
+    scols = [_wrap(ct.c_void_p(out_sums[c]), n, HF_FLOAT64) for c in range(nv)]
+    ccols = ([_wrap(ct.c_void_p(out_counts[c]), n, HF_INT64)
+              for c in range(nv)] if want_counts else None)
     return kcol, scols, ccols, n
 
 

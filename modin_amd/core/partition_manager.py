@@ -284,10 +284,8 @@ class HipDataframePartitionManager:
                     if H < (1 << 27):
                         H <<= 2
                         continue
-                    raise lib.HfError(
-                        "groupby cardinality exceeds the hash table cap "
-                        "(~134M groups); the sort-based general groupby is a "
-                        "later round") from e
+                    return cls._groupby_sorted(key_cols, val_cols_per_part,
+                                               want_counts, agg_op)
                 raise
             finally:
                 lib.free_raw(tkey)
@@ -295,6 +293,26 @@ class HipDataframePartitionManager:
                 lib.free_raw(rowcnt)
                 if counts:
                     lib.free_raw(counts)
+
+    @classmethod
+    def _groupby_sorted(cls, key_cols, val_cols_per_part, want_counts,
+                        agg_op):
+        """Sort-based groupby: the unbounded-cardinality path (>~134M
+        groups).  Concatenate the partitions' rows, key-sort them (stable LSD
+        radix, hf_sort_perm), then one segmented-aggregation pass over the
+        equal-key runs (hf_groupby_sorted).  Mirrors the reference's
+        fallthrough from the Map-Reduce groupby to the full-column groupby
+        (algebra/groupby.py: GroupByReduce -> GroupByDefault)."""
+        nv = len(val_cols_per_part[0]) if val_cols_per_part else 0
+        keys = (key_cols[0] if len(key_cols) == 1
+                else lib.concat(key_cols))
+        vals = [(val_cols_per_part[0][c] if len(key_cols) == 1
+                 else lib.concat([vp[c] for vp in val_cols_per_part]))
+                for c in range(nv)]
+        perm = lib.sort_perm(keys)
+        skeys = lib.gather(keys, perm)
+        svals = [lib.gather(v, perm) for v in vals]
+        return lib.groupby_sorted(skeys, svals, agg_op, want_counts)
 
 
 def maybe_allreduce_keyrange(kmin, kmax):

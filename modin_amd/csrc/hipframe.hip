@@ -1606,6 +1606,63 @@ __global__ void __launch_bounds__(BLOCK) k_filter_scatter(
   }
 }
 
+// ---- sort-based groupby kernels ----
+
+__global__ void __launch_bounds__(BLOCK) k_head_flags(
+    const int64_t* __restrict__ k, int64_t n, long long* __restrict__ head) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride)
+    head[i] = (i == 0) || (k[i] != k[i - 1]);
+}
+
+// per-row run id = (heads at positions <= i) - 1, via the filter plan's
+// tile bases; one device atomic per row into the per-run aggregate
+template <int AOP, bool CNT, bool ROWCNT, bool HAVE_VAL>
+__global__ void __launch_bounds__(BLOCK) k_segagg(
+    const long long* __restrict__ head, const double* __restrict__ vals,
+    int64_t n, const int64_t* __restrict__ tile_bases,
+    double* __restrict__ gsums, unsigned long long* __restrict__ growcnt,
+    unsigned long long* __restrict__ gcounts) {
+  const int64_t t0 = (int64_t)blockIdx.x * FILT_TILE;
+  const int64_t t1 = min(t0 + (int64_t)FILT_TILE, n);
+  __shared__ int64_t s_base;
+  __shared__ int s_wave_cnt[BLOCK / 64];
+  if (threadIdx.x == 0) s_base = tile_bases[blockIdx.x];
+  __syncthreads();
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  for (int64_t chunk = t0; chunk < t1; chunk += blockDim.x) {
+    const int64_t i = chunk + threadIdx.x;
+    const bool pred = (i < t1) && (head[i] != 0);
+    const uint64_t ballot = __ballot(pred);
+    if (lane == 0) s_wave_cnt[wave] = __popcll(ballot);
+    __syncthreads();
+    int64_t wave_base = 0;
+    for (int w = 0; w < wave; ++w) wave_base += s_wave_cnt[w];
+    if (i < t1) {
+      // inclusive count of heads <= i within the tile, minus 1 for run id
+      const int64_t run =
+          s_base + wave_base +
+          __popcll(ballot & ((lane == 63) ? ~0ULL : ((2ULL << lane) - 1))) - 1;
+      if (ROWCNT) atomicAdd(&growcnt[run], 1ULL);
+      if (HAVE_VAL) {
+        const double v = vals[i];
+        if (v == v) {
+          glob_slot_agg<AOP>(&gsums[run], v);
+          if (CNT) atomicAdd(&gcounts[run], 1ULL);
+        }
+      }
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      int64_t tot = 0;
+      for (int w = 0; w < BLOCK / 64; ++w) tot += s_wave_cnt[w];
+      s_base += tot;
+    }
+    __syncthreads();
+  }
+}
+
 __global__ void __launch_bounds__(BLOCK) k_gather_f64(
     const double* __restrict__ src, const int64_t* __restrict__ idx,
     double* __restrict__ out, int64_t n) {
@@ -2510,6 +2567,150 @@ int hf_groupby_hash_compact(uintptr_t tkey, uintptr_t sums, uintptr_t rowcnt,
   return HF_OK;
 }
 
+static const int64_t* plan_tiles(const hf_filterplan* p);
+
+int hf_groupby_sorted(const hf_col* sorted_keys, const hf_col* const* vals,
+                      int nvals, int agg_op, int want_counts,
+                      hf_col** out_keys, hf_col** out_sums,
+                      hf_col** out_counts, int64_t* n_groups) {
+  HF_NEED_INIT("hf_groupby_sorted");
+  if (!sorted_keys || nvals < 0 || nvals > GB_MAX_VALS || !out_keys ||
+      !n_groups)
+    return set_err(HF_ERR_ARG, "hf_groupby_sorted", "bad args");
+  if (sorted_keys->dtype != HF_INT64)
+    return set_err(HF_ERR_ARG, "hf_groupby_sorted", "keys must be int64");
+  for (int c = 0; c < nvals; ++c)
+    if (!vals[c] || vals[c]->dtype != HF_FLOAT64 ||
+        vals[c]->len != sorted_keys->len)
+      return set_err(HF_ERR_ARG, "hf_groupby_sorted",
+                     "vals must be float64 columns of keys' length");
+  const int64_t n = sorted_keys->len;
+  // head flags -> filter plan (run count + tile bases)
+  hf_col* head = nullptr;
+  int rc = hf_col_alloc(n, HF_INT64, &head);
+  if (rc != HF_OK) return rc;
+  if (n > 0) {
+    rc = timed_launch("gb_sorted_heads", [&] {
+      hipLaunchKernelGGL(k_head_flags, dim3((uint32_t)grid_for(n)),
+                         dim3(BLOCK), 0, g.stream,
+                         (const int64_t*)sorted_keys->dptr, n,
+                         (long long*)head->dptr);
+    });
+    if (rc != HF_OK) { hf_col_free(head); return rc; }
+  }
+  hf_filterplan* plan = nullptr;
+  int64_t C = 0;
+  rc = hf_filter_plan(head, &plan, &C);
+  if (rc != HF_OK) { hf_col_free(head); return rc; }
+  // outputs: unique keys (filter of sorted keys) + per-run aggregates
+  rc = hf_filter_apply(plan, sorted_keys, out_keys);
+  const bool cnt = want_counts != 0;
+  // per-run aggregate buffers
+  const int64_t Ca = C > 0 ? C : 1;
+  double* gsums = nullptr;
+  unsigned long long* growcnt = nullptr;
+  unsigned long long* gcounts = nullptr;
+  if (rc == HF_OK)
+    HF_HIP("hf_groupby_sorted",
+           dev_alloc((void**)&gsums, Ca * 8 * (nvals ? nvals : 1), g.stream));
+  if (rc == HF_OK)
+    HF_HIP("hf_groupby_sorted", dev_alloc((void**)&growcnt, Ca * 8, g.stream));
+  if (rc == HF_OK && cnt)
+    HF_HIP("hf_groupby_sorted",
+           dev_alloc((void**)&gcounts, Ca * 8 * (nvals ? nvals : 1), g.stream));
+  if (rc == HF_OK) {
+    const double init =
+        agg_op == HF_AGG_SUM ? 0.0
+        : agg_op == HF_AGG_MIN ? __builtin_huge_val() : -__builtin_huge_val();
+    if (nvals) {
+      rc = timed_launch("fill_f64", [&] {
+        hipLaunchKernelGGL(k_fill_f64,
+                           dim3((uint32_t)grid_for(Ca * nvals)), dim3(BLOCK),
+                           0, g.stream, gsums, init, Ca * nvals);
+      });
+    }
+    HF_HIP("hf_groupby_sorted", hipMemsetAsync(growcnt, 0, Ca * 8, g.stream));
+    if (cnt && nvals)
+      HF_HIP("hf_groupby_sorted",
+             hipMemsetAsync(gcounts, 0, Ca * 8 * nvals, g.stream));
+  }
+  // aggregate per column (ROWCNT on the first launch only)
+  const int64_t ntiles = n > 0 ? (n + FILT_TILE - 1) / FILT_TILE : 1;
+  auto seg = [&](auto aTag, auto cTag, auto rTag, auto vTag, const double* v,
+                 double* gs, unsigned long long* gc) {
+    constexpr int A = decltype(aTag)::value;
+    constexpr bool CC = decltype(cTag)::value;
+    constexpr bool R = decltype(rTag)::value;
+    constexpr bool V = decltype(vTag)::value;
+    return timed_launch("gb_segagg", [&] {
+      hipLaunchKernelGGL((k_segagg<A, CC, R, V>), dim3((uint32_t)ntiles),
+                         dim3(BLOCK), 0, g.stream, (const long long*)head->dptr,
+                         v, n, plan_tiles(plan), gs, growcnt, gc);
+    });
+  };
+  using T = std::true_type;
+  using F = std::false_type;
+  auto segA = [&](auto cTag, auto rTag, auto vTag, const double* v, double* gs,
+                  unsigned long long* gc) {
+    return agg_op == HF_AGG_SUM
+               ? seg(std::integral_constant<int, HF_AGG_SUM>{}, cTag, rTag,
+                     vTag, v, gs, gc)
+           : agg_op == HF_AGG_MIN
+               ? seg(std::integral_constant<int, HF_AGG_MIN>{}, cTag, rTag,
+                     vTag, v, gs, gc)
+               : seg(std::integral_constant<int, HF_AGG_MAX>{}, cTag, rTag,
+                     vTag, v, gs, gc);
+  };
+  if (rc == HF_OK && n > 0) {
+    if (nvals == 0) {
+      rc = segA(F{}, T{}, F{}, nullptr, nullptr, nullptr);
+    } else {
+      for (int c = 0; c < nvals && rc == HF_OK; ++c) {
+        double* gs = gsums + (int64_t)c * Ca;
+        unsigned long long* gc = cnt ? gcounts + (int64_t)c * Ca : nullptr;
+        const double* v = (const double*)vals[c]->dptr;
+        if (c == 0)
+          rc = cnt ? segA(T{}, T{}, T{}, v, gs, gc)
+                   : segA(F{}, T{}, T{}, v, gs, gc);
+        else
+          rc = cnt ? segA(T{}, F{}, T{}, v, gs, gc)
+                   : segA(F{}, F{}, T{}, v, gs, gc);
+      }
+    }
+  }
+  // wrap aggregate buffers as columns (transfer ownership)
+  for (int c = 0; c < nvals && rc == HF_OK; ++c) {
+    hf_col* sc = new hf_col{};
+    sc->len = C;
+    sc->dtype = HF_FLOAT64;
+    sc->gpu = g.gpu;
+    HF_HIP("hf_groupby_sorted", dev_alloc(&sc->dptr, Ca * 8, g.stream));
+    HF_HIP("hf_groupby_sorted",
+           hipMemcpyAsync(sc->dptr, gsums + (int64_t)c * Ca, Ca * 8,
+                          hipMemcpyDeviceToDevice, g.stream));
+    out_sums[c] = sc;
+    if (cnt && out_counts) {
+      hf_col* cc2 = new hf_col{};
+      cc2->len = C;
+      cc2->dtype = HF_INT64;
+      cc2->gpu = g.gpu;
+      HF_HIP("hf_groupby_sorted", dev_alloc(&cc2->dptr, Ca * 8, g.stream));
+      HF_HIP("hf_groupby_sorted",
+             hipMemcpyAsync(cc2->dptr, gcounts + (int64_t)c * Ca, Ca * 8,
+                            hipMemcpyDeviceToDevice, g.stream));
+      out_counts[c] = cc2;
+    }
+  }
+  hf_filter_plan_free(plan);
+  hf_col_free(head);
+  if (gsums) dev_free(gsums, g.stream);
+  if (growcnt) dev_free(growcnt, g.stream);
+  if (gcounts) dev_free(gcounts, g.stream);
+  if (rc != HF_OK) return rc;
+  *n_groups = C;
+  return HF_OK;
+}
+
 int hf_col_slice(const hf_col* col, int64_t start, int64_t len, hf_col** out) {
   HF_NEED_INIT("hf_col_slice");
   if (!col || !out || start < 0 || len < 0 || start + len > col->len)
@@ -2963,6 +3164,8 @@ struct hf_filterplan {
   int64_t n, total, ntiles;
   int64_t* d_tiles;        // exclusive per-tile kept offsets
 };
+
+static const int64_t* plan_tiles(const hf_filterplan* p) { return p->d_tiles; }
 
 int hf_compare_scalar(int op, const hf_col* col, double scalar, hf_col** out) {
   HF_NEED_INIT("hf_compare_scalar");
