@@ -284,6 +284,21 @@ int hf_filter_plan_free(hf_filterplan* plan);
  * caller gathers payload columns and the index with hf_gather. */
 int hf_sort_perm(const hf_col* keys, int ascending, hf_col** out_perm);
 
+/* ---- multi-GPU range shuffle (device form of the reference's
+ * RangePartitioning shuffle, partition_manager.py:1937 ``shuffle_partitions``
+ * / experimental range-partitioning groupby): per-row destination-rank
+ * binning against host-supplied SORTED splitters.
+ * dest[i] = #{j : splitters[j] <= key[i]}  (int64 compares — exact over the
+ * full key range; groups never straddle a boundary because the rule depends
+ * only on the key value).  nsplit in [0, 63]; keys int64. */
+int hf_shuffle_dest(const hf_col* keys, const int64_t* splitters, int nsplit,
+                    hf_col** dest);
+
+/* Raw device-to-device copy on the hipframe stream — the interop bridge to
+ * RCCL-visible torch buffers (exchange_splits): hf columns are copied into /
+ * out of torch-allocated device tensors by address. */
+int hf_memcpy_dd(uintptr_t dst, uintptr_t src, int64_t bytes);
+
 /* ---- profiling (bench.py roofline leg) ----
  * When enabled, every kernel launch is bracketed by HIP events on the module
  * stream; hf_kernel_stats returns the accumulated count and total ms for the

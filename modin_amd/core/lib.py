@@ -127,6 +127,11 @@ def load() -> ct.CDLL:
                                              ct.POINTER(ct.c_void_p),
                                              ct.POINTER(ct.c_void_p),
                                              ct.POINTER(ct.c_int64)]),
+            "hf_shuffle_dest": (ct.c_int, [ct.c_void_p,
+                                           ct.POINTER(ct.c_int64), ct.c_int,
+                                           ct.POINTER(ct.c_void_p)]),
+            "hf_memcpy_dd": (ct.c_int, [ct.c_size_t, ct.c_size_t,
+                                        ct.c_int64]),
             "hf_sort_perm": (ct.c_int, [ct.c_void_p, ct.c_int,
                                         ct.POINTER(ct.c_void_p)]),
             "hf_fixup_empty": (ct.c_int, [ct.c_void_p, ct.c_void_p,
@@ -185,7 +190,7 @@ def exported_symbols():
         "hf_reduce", "hf_groupby_accum", "hf_groupby_compact", "hf_fill_f64",
         "hf_fixup_empty", "hf_sort_perm", "hf_fill_i64",
         "hf_groupby_hash_accum", "hf_groupby_hash_compact",
-        "hf_groupby_sorted",
+        "hf_groupby_sorted", "hf_shuffle_dest", "hf_memcpy_dd",
         "hf_col_concat", "hf_col_slice", "hf_join_build", "hf_join_free", "hf_join_probe",
         "hf_gather", "hf_compare_scalar", "hf_filter_plan", "hf_filter_apply",
         "hf_filter_iota", "hf_filter_plan_free", "hf_profiling",
@@ -590,6 +595,24 @@ def filter_iota(plan: FilterPlan, base: int) -> ColumnRef:
     _check(load().hf_filter_iota(plan.handle, base, ct.byref(out)),
            "hf_filter_iota")
     return _wrap(out, plan.n_kept, HF_INT64)
+
+
+def shuffle_dest(keys: ColumnRef, splitters) -> ColumnRef:
+    """dest[i] = #{j: splitters[j] <= key[i]} — destination-rank binning for
+    the range shuffle (native int64 compares; exact over the full span)."""
+    ensure_ready()
+    spl = np.asarray(splitters, dtype=np.int64)
+    arr = (ct.c_int64 * max(len(spl), 1))(*spl.tolist())
+    out = ct.c_void_p()
+    _check(load().hf_shuffle_dest(keys.handle, arr, len(spl), ct.byref(out)),
+           "hf_shuffle_dest")
+    return _wrap(out, keys.length, HF_INT64)
+
+
+def memcpy_dd(dst_ptr: int, src_ptr: int, nbytes: int) -> None:
+    """Device-to-device copy on the hipframe stream (torch-buffer interop)."""
+    ensure_ready()
+    _check(load().hf_memcpy_dd(dst_ptr, src_ptr, nbytes), "hf_memcpy_dd")
 
 
 def sort_perm(keys: ColumnRef, ascending: bool = True) -> ColumnRef:

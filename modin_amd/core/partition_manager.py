@@ -224,10 +224,8 @@ class HipDataframePartitionManager:
         if n_slots > config.MaxGroupbySlots.get():
             from ..distributed import is_active
             if is_active():
-                raise lib.HfError(
-                    f"groupby key range {n_slots} exceeds MaxGroupbySlots: "
-                    "the multi-GPU hash groupby needs the shuffle exchange — "
-                    "later round (dense ranges all-reduce fine)")
+                return cls._groupby_shuffle(key_cols, val_cols_per_part,
+                                            want_counts, agg_op)
             return cls._groupby_hash(key_cols, val_cols_per_part, total_rows,
                                      want_counts, agg_op)
         n_slots = max(n_slots, 1)
@@ -313,6 +311,90 @@ class HipDataframePartitionManager:
         skeys = lib.gather(keys, perm)
         svals = [lib.gather(v, perm) for v in vals]
         return lib.groupby_sorted(skeys, svals, agg_op, want_counts)
+
+
+    @classmethod
+    def _groupby_local(cls, key_cols, val_cols_per_part, want_counts, agg_op):
+        """Single-rank groupby over explicit column lists, no collectives:
+        dense table when the local key range is bounded, hash/sorted beyond
+        (used for the post-shuffle per-rank aggregation)."""
+        kmin = kmax = None
+        total_rows = 0
+        for kcol in key_cols:
+            if kcol.length:
+                r = lib.reduce(kcol)
+                kmin = r.imn if kmin is None else min(kmin, r.imn)
+                kmax = r.imx if kmax is None else max(kmax, r.imx)
+            total_rows += kcol.length
+        if kmin is None:
+            kmin, kmax = 0, -1
+        n_slots = max(kmax - kmin + 1, 0)
+        if n_slots > config.MaxGroupbySlots.get():
+            return cls._groupby_hash(key_cols, val_cols_per_part, total_rows,
+                                     want_counts, agg_op)
+        n_slots = max(n_slots, 1)
+        nv = len(val_cols_per_part[0]) if val_cols_per_part else 0
+        table = GroupbyTable(nv, kmin, n_slots, want_counts, agg_op)
+        for kcol, vals in zip(key_cols, val_cols_per_part):
+            if kcol.length:
+                lib.groupby_accum(kcol, vals, agg_op, kmin, n_slots,
+                                  table.sums, table.rowcnt, table.counts)
+        out = lib.groupby_compact(table.sums, table.rowcnt, table.counts,
+                                  nv, kmin, n_slots)
+        table.free()
+        return out
+
+    @classmethod
+    def _groupby_shuffle(cls, key_cols, val_cols_per_part, want_counts,
+                         agg_op):
+        """Multi-GPU unbounded-key groupby: the range-partitioning shuffle
+        (SURVEY §8e option (ii); the device form of the reference's
+        ``shuffle_partitions`` / range-partitioning groupby,
+        partition_manager.py:1937).  Sampled splitters assign each key RANGE
+        to one rank (a group never straddles ranks), rows move once
+        (exchange_splits: RCCL all_to_all over xGMI), every rank aggregates
+        its range locally with the single-rank engine, and the disjoint
+        per-rank results — ascending in rank order by construction — are
+        all-gathered so each rank returns the identical replicated result
+        (the dense-table path's convention)."""
+        import numpy as np
+        from .. import distributed as dist_mod
+        P = dist_mod.world_size()
+        nv = len(val_cols_per_part[0]) if val_cols_per_part else 0
+        keys = key_cols[0] if len(key_cols) == 1 else lib.concat(key_cols)
+        vals = [val_cols_per_part[0][c] if len(key_cols) == 1
+                else lib.concat([vp[c] for vp in val_cols_per_part])
+                for c in range(nv)]
+        # deterministic strided local sample -> identical splitters everywhere
+        n = keys.length
+        S = min(n, 4096)
+        if S:
+            idx = np.linspace(0, n - 1, S).astype(np.int64)
+            sample = lib.get(lib.gather(keys, lib.put(idx)))
+        else:
+            sample = np.empty(0, dtype=np.int64)
+        splitters = dist_mod.sample_splitters(sample)
+        dest = lib.shuffle_dest(keys, splitters)
+        send_counts = []
+        send_k, send_v = [], [[] for _ in range(nv)]
+        for d in range(P):
+            mask = lib.compare_scalar(lib.CMP_EQ, dest, float(d))
+            plan = lib.filter_plan(mask)
+            send_counts.append(plan.n_kept)
+            send_k.append(lib.filter_apply(plan, keys))
+            for c in range(nv):
+                send_v[c].append(lib.filter_apply(plan, vals[c]))
+        rk = dist_mod.exchange_column(lib.concat(send_k), send_counts)
+        rv = [dist_mod.exchange_column(lib.concat(send_v[c]), send_counts)
+              for c in range(nv)]
+        k2, s2, c2, _n2 = cls._groupby_local([rk], [rv], want_counts, agg_op)
+        gk, gs, gc = dist_mod.allgather_groupby(
+            lib.get(k2), [lib.get(c) for c in s2],
+            [lib.get(c) for c in c2] if c2 is not None else None)
+        keys_col = lib.put(gk)
+        sums_cols = [lib.put(a) for a in gs]
+        counts_cols = ([lib.put(a) for a in gc] if gc is not None else None)
+        return keys_col, sums_cols, counts_cols, len(gk)
 
 
 def maybe_allreduce_keyrange(kmin, kmax):

@@ -1663,6 +1663,19 @@ __global__ void __launch_bounds__(BLOCK) k_segagg(
   }
 }
 
+__global__ void __launch_bounds__(BLOCK) k_shuffle_dest(
+    const int64_t* __restrict__ keys, const int64_t* __restrict__ split,
+    int nsplit, long long* __restrict__ dest, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const int64_t k = keys[i];
+    int d = 0;
+    for (int j = 0; j < nsplit; ++j) d += (split[j] <= k);
+    dest[i] = d;
+  }
+}
+
 __global__ void __launch_bounds__(BLOCK) k_gather_f64(
     const double* __restrict__ src, const int64_t* __restrict__ idx,
     double* __restrict__ out, int64_t n) {
@@ -2568,6 +2581,52 @@ int hf_groupby_hash_compact(uintptr_t tkey, uintptr_t sums, uintptr_t rowcnt,
 }
 
 static const int64_t* plan_tiles(const hf_filterplan* p);
+
+int hf_shuffle_dest(const hf_col* keys, const int64_t* splitters, int nsplit,
+                    hf_col** dest) {
+  HF_NEED_INIT("hf_shuffle_dest");
+  if (!keys || !dest || nsplit < 0 || nsplit > 63 ||
+      (nsplit && !splitters))
+    return set_err(HF_ERR_ARG, "hf_shuffle_dest", "bad args");
+  if (keys->dtype != HF_INT64)
+    return set_err(HF_ERR_ARG, "hf_shuffle_dest", "keys must be int64");
+  const int64_t n = keys->len;
+  int rc = hf_col_alloc(n, HF_INT64, dest);
+  if (rc != HF_OK) return rc;
+  int64_t* d_split = nullptr;
+  if (nsplit) {
+    HF_HIP("hf_shuffle_dest", dev_alloc((void**)&d_split, nsplit * 8, g.stream));
+    HF_HIP("hf_shuffle_dest",
+           hipMemcpyAsync(d_split, splitters, nsplit * 8,
+                          hipMemcpyHostToDevice, g.stream));
+  }
+  if (n > 0) {
+    rc = timed_launch("shuffle_dest", [&] {
+      hipLaunchKernelGGL(k_shuffle_dest, dim3((uint32_t)grid_for(n)),
+                         dim3(BLOCK), 0, g.stream,
+                         (const int64_t*)keys->dptr, d_split, nsplit,
+                         (long long*)(*dest)->dptr, n);
+    });
+  }
+  if (d_split) {
+    // splitters are host memory borrowed only until the H2D lands
+    hipStreamSynchronize(g.stream);
+    dev_free(d_split, g.stream);
+  }
+  if (rc != HF_OK) { hf_col_free(*dest); *dest = nullptr; }
+  return rc;
+}
+
+int hf_memcpy_dd(uintptr_t dst, uintptr_t src, int64_t bytes) {
+  HF_NEED_INIT("hf_memcpy_dd");
+  if (bytes < 0 || (bytes && (!dst || !src)))
+    return set_err(HF_ERR_ARG, "hf_memcpy_dd", "bad args");
+  if (bytes)
+    HF_HIP("hf_memcpy_dd",
+           hipMemcpyAsync((void*)dst, (void*)src, (size_t)bytes,
+                          hipMemcpyDeviceToDevice, g.stream));
+  return HF_OK;
+}
 
 int hf_groupby_sorted(const hf_col* sorted_keys, const hf_col* const* vals,
                       int nvals, int agg_op, int want_counts,

@@ -230,6 +230,70 @@ def exchange_splits(t, send_counts):
     return out, recv_counts
 
 
+def sample_splitters(local_sample):
+    """world-1 key splitters from the gathered per-rank samples — the
+    device form of the reference's RangePartitioning sampling
+    (range-partitioning pre-shuffle: sample each partition, combine, take
+    quantiles).  Deterministic and identical on every rank: all_gather is
+    rank-ordered and the quantile rule is pure."""
+    import numpy as np
+    import torch.distributed as dist
+    gathered = [None] * world_size()
+    dist.all_gather_object(gathered, np.asarray(local_sample, dtype=np.int64))
+    alls = np.sort(np.concatenate([g for g in gathered if g is not None and
+                                   len(g)] or
+                                  [np.empty(0, dtype=np.int64)]))
+    P = world_size()
+    if alls.size == 0 or P <= 1:
+        return np.empty(0, dtype=np.int64)
+    qs = [(i * alls.size) // P for i in range(1, P)]
+    return alls[qs]
+
+
+def exchange_column(col, send_counts):
+    """Move one hf column's P destination spans to their ranks
+    (exchange_splits under the hood) and return the received rows as a new
+    hf column.  nccl: zero-host-copy — the column is mirrored into a
+    torch CUDA tensor by device address (hf_memcpy_dd) and travels over
+    xGMI; gloo (CPU test tier / single-GPU multi-rank validation): bounced
+    via host numpy."""
+    import numpy as np
+    import torch
+    from .core import lib
+    tdt = torch.int64 if col.dtype_code == lib.HF_INT64 else torch.float64
+    if _state["backend"] == "nccl":
+        t = torch.empty(col.length, dtype=tdt, device=_state["device"])
+        if col.length:
+            lib.memcpy_dd(t.data_ptr(), col.dptr(), 8 * col.length)
+            lib.sync()  # hf-stream copy must land before the collective
+        out, recv_counts = exchange_splits(t, send_counts)
+        recv = lib.alloc(out.numel(), col.dtype_code)
+        if out.numel():
+            lib.memcpy_dd(recv.dptr(), out.data_ptr(), 8 * out.numel())
+            lib.sync()
+        return recv
+    host = lib.get(col)
+    t = torch.from_numpy(np.ascontiguousarray(host))
+    out, recv_counts = exchange_splits(t, send_counts)
+    return lib.put(out.numpy())
+
+
+def allgather_groupby(keys_np, sums_np, counts_np):
+    """Concatenate the per-rank (disjoint, ascending-range) groupby results
+    in rank order so every rank returns the identical replicated frame —
+    the same output convention as the dense-table all-reduce path."""
+    import numpy as np
+    import torch.distributed as dist
+    gathered = [None] * world_size()
+    dist.all_gather_object(gathered, (keys_np, sums_np, counts_np))
+    gk = np.concatenate([g[0] for g in gathered])
+    nv = len(sums_np)
+    gs = [np.concatenate([g[1][c] for g in gathered]) for c in range(nv)]
+    gc = ([np.concatenate([g[2][c] for g in gathered]) for c in range(nv)]
+          if counts_np is not None else None)
+    return gk, gs, gc
+
+
 def maybe_allreduce_table(table) -> None:
     """RCCL all-reduce of the dense groupby table (reduce phase across GPUs)."""
     if not is_active():

@@ -23,7 +23,9 @@ from .ops import (  # noqa: F401
     inner_join,
     map_op,
     partitioned_groupby_agg,
+    pick_splitters,
     reduce_op,
+    shuffle_dest,
     sort_perm,
     split_row_counts,
 )

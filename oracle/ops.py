@@ -289,3 +289,23 @@ def partitioned_groupby_agg(keys: np.ndarray, vals: dict, agg: str,
                 for k in out_keys
             ])
     return out_keys, out
+
+
+def shuffle_dest(keys, splitters):
+    """Restates csrc k_shuffle_dest (the RangePartitioning bin rule,
+    reference partition_manager.py:1937 shuffle_partitions / range-partition
+    sampling): dest[i] = #{j : splitters[j] <= key[i]}.  Depends only on the
+    key value, so equal keys always land on one destination."""
+    keys = np.asarray(keys, dtype=np.int64)
+    spl = np.asarray(splitters, dtype=np.int64)
+    return np.searchsorted(spl, keys, side="right").astype(np.int64)
+
+
+def pick_splitters(samples, world):
+    """Restates distributed.sample_splitters: sorted global sample ->
+    world-1 quantile splitters (deterministic; identical on every rank)."""
+    s = np.sort(np.asarray(samples, dtype=np.int64))
+    if s.size == 0 or world <= 1:
+        return np.empty(0, dtype=np.int64)
+    qs = [(i * s.size) // world for i in range(1, world)]
+    return s[qs]

@@ -135,6 +135,92 @@ def _shuffle_worker(rank, world, port, fail_q):
         fail_q.put(f"rank {rank}: {e}\n{traceback.format_exc()}")
 
 
+def _range_shuffle_groupby_worker(rank, world, port, fail_q):
+    """The full range-partitioning shuffle-groupby recipe
+    (partition_manager._groupby_shuffle) executed on the oracle backend:
+    real collectives (sample_splitters / exchange_splits /
+    allgather_groupby), numpy in place of the device kernels."""
+    try:
+        os.environ.update({
+            "RANK": str(rank), "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        })
+        import torch
+        import modin_amd.distributed as dist_mod
+        assert dist_mod.init_from_env(backend="gloo", gpu=False)
+        rng = np.random.default_rng(31)  # same stream on all ranks
+        n = 30_000
+        keys_g = rng.integers(-2**62, 2**62, n).astype(np.int64)
+        keys_g[rng.random(n) < 0.2] = 77  # heavy duplicate straddling test
+        vals_g = rng.random(n)
+        vals_g[rng.random(n) < 0.1] = np.nan
+        counts = oracle.split_row_counts(n, world, 32)
+        offs = np.cumsum([0] + counts)
+        sl = slice(offs[rank], offs[rank + 1])
+        k_loc, v_loc = keys_g[sl], vals_g[sl]
+
+        # strided sample -> identical splitters on every rank
+        S = min(k_loc.size, 4096)
+        sample = (k_loc[np.linspace(0, k_loc.size - 1, S).astype(np.int64)]
+                  if S else np.empty(0, dtype=np.int64))
+        splitters = dist_mod.sample_splitters(sample)
+        sp2 = dist_mod.sample_splitters(sample)
+        np.testing.assert_array_equal(splitters, sp2)  # deterministic
+
+        dest = oracle.shuffle_dest(k_loc, splitters)
+        assert dest.min() >= 0 and dest.max() < world if dest.size else True
+        order = np.argsort(dest, kind="stable")
+        send_counts = np.bincount(dest, minlength=world).tolist()
+        rk, _ = dist_mod.exchange_splits(
+            torch.from_numpy(k_loc[order].copy()), send_counts)
+        rv, _ = dist_mod.exchange_splits(
+            torch.from_numpy(v_loc[order].copy()), send_counts)
+        rk, rv = rk.numpy(), rv.numpy()
+
+        def sort_gb(k, v):  # huge-span groupby (bincount oracle would blow up)
+            o = np.argsort(k, kind="stable")
+            ks, vs = k[o], v[o]
+            uk, starts = np.unique(ks, return_index=True)
+            sums = np.add.reduceat(np.nan_to_num(vs), starts) \
+                if ks.size else np.empty(0)
+            cnts = np.add.reduceat((~np.isnan(vs)).astype(np.int64), starts) \
+                if ks.size else np.empty(0, dtype=np.int64)
+            return uk, sums, cnts
+
+        # a key's rows all land on exactly one rank
+        ok_loc, osum_loc, ocnt_loc = sort_gb(rk, rv)
+        gk, gs, gc = dist_mod.allgather_groupby(
+            ok_loc, [osum_loc], [ocnt_loc])
+        ok, osum, ocnt = sort_gb(keys_g, vals_g)
+        np.testing.assert_array_equal(gk, ok)  # ascending, disjoint, complete
+        np.testing.assert_allclose(gs[0], osum, rtol=1e-12, atol=1e-9)
+        np.testing.assert_array_equal(gc[0], ocnt)
+        dist_mod.shutdown()
+    except Exception as e:  # pragma: no cover
+        import traceback
+        fail_q.put(f"rank {rank}: {e}\n{traceback.format_exc()}")
+
+
+@pytest.mark.timeout(120)
+@pytest.mark.parametrize("world", [2, 3])
+def test_gloo_range_shuffle_groupby(world):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29540 + world
+    procs = [ctx.Process(target=_range_shuffle_groupby_worker,
+                         args=(r, world, port, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=110)
+    errs = []
+    while not q.empty():
+        errs.append(q.get())
+    assert not errs, "\n".join(errs)
+    assert all(p.exitcode == 0 for p in procs)
+
+
 @pytest.mark.timeout(120)
 @pytest.mark.parametrize("world", [2, 3])
 def test_gloo_shuffle_exchange(world):

@@ -172,3 +172,30 @@ def test_kernel_stats_profiling():
     n, ms = lib.kernel_stats("map_f64")
     assert n == 3 and ms > 0
     lib.profiling(False)
+
+
+def test_shuffle_dest_kernel_vs_oracle():
+    """Native-i64 splitter binning (k_shuffle_dest) — exact at magnitudes
+    where an f64 compare would mis-bin (keys within 1 of a 2^62-scale
+    splitter)."""
+    rng = np.random.default_rng(17)
+    spl = np.sort(rng.integers(-2**62, 2**62, 7)).astype(np.int64)
+    keys = rng.integers(-2**62, 2**62, 100_000).astype(np.int64)
+    # adversarial: keys exactly at / one-off the splitters
+    edge = np.concatenate([spl, spl - 1, spl + 1]).astype(np.int64)
+    keys = np.concatenate([keys, edge])
+    kcol = lib.put(keys)
+    dest = lib.get(lib.shuffle_dest(kcol, spl))
+    np.testing.assert_array_equal(dest, oracle.shuffle_dest(keys, spl))
+    # empty splitter list -> all zeros
+    np.testing.assert_array_equal(
+        lib.get(lib.shuffle_dest(kcol, np.empty(0, dtype=np.int64))), 0)
+
+
+def test_memcpy_dd_roundtrip():
+    rng = np.random.default_rng(18)
+    a = rng.random(10_000)
+    src = lib.put(a)
+    dst = lib.alloc(10_000, lib.HF_FLOAT64)
+    lib.memcpy_dd(dst.dptr(), src.dptr(), 8 * 10_000)
+    np.testing.assert_array_equal(lib.get(dst), a)

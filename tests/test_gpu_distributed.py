@@ -1,0 +1,88 @@
+"""GPU tier: the multi-rank path end-to-end on ONE GPU (world 2, gloo
+backend with device compute — the single-box stand-in for one-process-per-GPU
+over RCCL; the collectives' device form is covered by the same code paths,
+with gloo's host bounce replacing the xGMI transport).
+
+Covers both distributed groupby routes of partition_manager.groupby_reduce:
+  * dense-range: per-rank table accumulate + table all-reduce + compact,
+  * range shuffle: sampled splitters + all-to-all row exchange + per-rank
+    local aggregation + result all-gather (the unbounded-key route).
+Every rank must return the IDENTICAL replicated result equal to pandas on
+the union of the shards.
+"""
+
+import multiprocessing as mp
+import os
+
+import numpy as np
+import pandas
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+def _worker(rank, world, port, fail_q):
+    try:
+        os.environ.update({
+            "RANK": str(rank), "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        })
+        import modin_amd.distributed as dist_mod
+        assert dist_mod.init_from_env(backend="gloo", gpu=True)
+        import modin_amd.config as config
+        import modin_amd.pandas as mpd
+
+        rng = np.random.default_rng(41)  # same stream on all ranks
+        n = 60_000
+        scenarios = {
+            # dense-range route (key span < MaxGroupbySlots)
+            "dense": rng.integers(-50, 1000, n).astype(np.int64),
+            # shuffle route (full int64 span; heavy dup run straddles shards)
+            "huge": rng.integers(-2**62, 2**62, n).astype(np.int64),
+        }
+        scenarios["huge"][rng.random(n) < 0.2] = 77
+        vals = rng.random(n)
+        vals[rng.random(n) < 0.1] = np.nan
+        ivals = rng.integers(-100, 100, n).astype(np.int64)
+        lo = rank * n // world
+        hi = (rank + 1) * n // world
+        for name, keys in scenarios.items():
+            pdf = pandas.DataFrame({"k": keys, "v": vals, "w": ivals})
+            df = mpd.DataFrame(pdf.iloc[lo:hi])
+            for agg in ("sum", "count", "mean"):
+                out = getattr(df.groupby("k"), agg)().to_pandas()
+                expect = getattr(pdf.groupby("k"), agg)()
+                np.testing.assert_array_equal(
+                    out.index.to_numpy(), expect.index.to_numpy(),
+                    err_msg=f"{name}/{agg} keys")
+                for c in ("v", "w"):
+                    np.testing.assert_allclose(
+                        out[c].to_numpy(), expect[c].to_numpy(),
+                        rtol=1e-12, atol=1e-9, equal_nan=True,
+                        err_msg=f"{name}/{agg}/{c}")
+        dist_mod.shutdown()
+    except Exception as e:  # pragma: no cover
+        import traceback
+        fail_q.put(f"rank {rank}: {e}\n{traceback.format_exc()}")
+
+
+@pytest.mark.timeout(300)
+def test_world2_groupby_dense_and_shuffle_on_gpu(gpu_ready):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29561
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=280)
+    errs = []
+    while not q.empty():
+        errs.append(q.get())
+    for p in procs:
+        if p.is_alive():
+            p.terminate()
+    assert not errs, "\n".join(errs)
+    assert all(p.exitcode == 0 for p in procs)

@@ -82,3 +82,35 @@ def test_partition_call_queue_semantics():
     # p1 still lazily holds only "a"
     p1.drain_call_queue()
     assert p1._block.v == ["a"]
+
+
+def test_shuffle_dest_rule():
+    """oracle.shuffle_dest: searchsorted-right binning — boundary keys go
+    WITH the right bin, equal keys always share a destination, full-span
+    int64 exact."""
+    spl = np.array([-2**62, 0, 2**62], dtype=np.int64)
+    keys = np.array([-2**63, -2**62 - 1, -2**62, -1, 0, 1,
+                     2**62 - 1, 2**62, 2**63 - 1], dtype=np.int64)
+    expect = np.array([0, 0, 1, 1, 2, 2, 2, 3, 3])
+    np.testing.assert_array_equal(oracle.shuffle_dest(keys, spl), expect)
+    # no splitters -> everything to rank 0
+    np.testing.assert_array_equal(
+        oracle.shuffle_dest(keys, np.empty(0, dtype=np.int64)),
+        np.zeros(len(keys), dtype=np.int64))
+    # equal keys -> one destination (groups never straddle ranks)
+    dup = np.full(100, 7, dtype=np.int64)
+    assert len(np.unique(oracle.shuffle_dest(dup, np.array([3, 7, 11])))) == 1
+
+
+def test_pick_splitters_quantiles():
+    s = oracle.pick_splitters(np.arange(100, dtype=np.int64), 4)
+    np.testing.assert_array_equal(s, [25, 50, 75])
+    assert oracle.pick_splitters(np.empty(0, dtype=np.int64), 4).size == 0
+    assert oracle.pick_splitters(np.arange(10), 1).size == 0
+    # splitters are sorted and dests cover [0, P)
+    rng = np.random.default_rng(5)
+    sample = rng.integers(-2**62, 2**62, 1000).astype(np.int64)
+    spl = oracle.pick_splitters(sample, 8)
+    assert np.all(np.diff(spl) >= 0)
+    d = oracle.shuffle_dest(sample, spl)
+    assert d.min() >= 0 and d.max() <= 7
