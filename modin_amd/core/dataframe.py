@@ -157,6 +157,18 @@ class HipDataframe:
             raise lib.HfError("merge: key column must be int64 (dense-range "
                               "CSR join; hashed keys are a later round)")
         rvals = [concat_col(other, n) for n in right_names]
+        from ..distributed import is_active
+        if is_active():
+            # the broadcast join's combine() across ranks: every rank builds
+            # from the FULL right table (all-gather of the right shards),
+            # probes only its own left shard — result stays left-sharded
+            from .. import distributed as dist_mod
+            rdt = [c.dtype_code for c in rvals]
+            gathered = dist_mod.allgather_arrays(
+                [lib.get(rkeys)] + [lib.get(c) for c in rvals])
+            rkeys = lib.put(gathered[0])
+            rvals = [lib.put(a) for a in gathered[1:]]
+            del rdt
         if rkeys.length:
             r = lib.reduce(rkeys)
             kmin, n_slots = r.imn, r.imx - r.imn + 1
@@ -309,6 +321,9 @@ class HipDataframe:
             cols = [p.block().columns[name] for p in self._partitions]
             return cols[0] if len(cols) == 1 else lib.concat(cols)
 
+        from ..distributed import is_active
+        if is_active():
+            return self._sort_rows_distributed(by, ascending, concat_col)
         kcol = concat_col(by)
         perm = lib.sort_perm(kcol, ascending)
         cols = {}
@@ -319,6 +334,56 @@ class HipDataframe:
         part = HipDataframePartition(DeviceBlock(cols, n))
         return HipDataframe([part], DeviceIndex(perm, name=None),
                             self.columns, [n], self.dtypes)
+
+    def _sort_rows_distributed(self, by, ascending, concat_col):
+        """Distributed sort_values: the range-partitioning shuffle + local
+        stable sort (SURVEY §8f.2 "reuses the shuffle"; reference
+        sort_by -> _apply_func_to_range_partitioning, dataframe.py:2742).
+        Sampled splitters put each key RANGE on one rank (rank order =
+        global key order), rows move once (exchange_splits), each rank
+        radix-sorts its range.  Stability: exchange output is source-rank
+        (= global-position) ordered and the per-dest filter preserves
+        order, so the local stable sort reproduces pandas order exactly.
+        Result: the frame stays sharded, rank r holding globally-sorted
+        slice r; the index carries the original global positions."""
+        import numpy as np
+        from .. import distributed as dist_mod
+        P = dist_mod.world_size()
+        kcol = concat_col(by)
+        n = kcol.length
+        S = min(n, 4096)
+        if S:
+            sidx = np.linspace(0, n - 1, S).astype(np.int64)
+            sample = lib.get(lib.gather(kcol, lib.put(sidx)))
+        else:
+            sample = np.empty(0, dtype=np.int64)
+        splitters = dist_mod.sample_splitters(sample)
+        dest = lib.shuffle_dest(kcol, splitters)
+        if not ascending:  # rank 0 takes the LARGEST key range
+            dest = lib.map_scalar(lib.MAP_RSUB, dest, P - 1)
+        base = dist_mod.global_row_base(n)
+        names = list(self.columns)
+        send_cols = {m: [] for m in names}
+        send_pos, send_counts = [], []
+        cols_cat = {m: (kcol if m == by else concat_col(m)) for m in names}
+        for d in range(P):
+            mask = lib.compare_scalar(lib.CMP_EQ, dest, float(d))
+            plan = lib.filter_plan(mask)
+            send_counts.append(plan.n_kept)
+            send_pos.append(lib.filter_iota(plan, base))
+            for m in names:
+                send_cols[m].append(lib.filter_apply(plan, cols_cat[m]))
+        recv = {m: dist_mod.exchange_column(lib.concat(send_cols[m]),
+                                            send_counts)
+                for m in names}
+        rpos = dist_mod.exchange_column(lib.concat(send_pos), send_counts)
+        perm = lib.sort_perm(recv[by], ascending)
+        out_cols = {m: lib.gather(recv[m], perm) for m in names}
+        pos_sorted = lib.gather(rpos, perm)
+        ln = perm.length
+        part = HipDataframePartition(DeviceBlock(out_cols, ln))
+        return HipDataframe([part], DeviceIndex(pos_sorted, name=None),
+                            names, [ln], self.dtypes)
 
     # ---- dropna mask: AND of per-column notna (pandas dropna(how="any")) ----
     def notna_all_mask(self) -> "HipDataframe":

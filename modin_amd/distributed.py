@@ -294,6 +294,27 @@ def allgather_groupby(keys_np, sums_np, counts_np):
     return gk, gs, gc
 
 
+def allgather_arrays(arrays):
+    """All-gather a list of per-rank numpy arrays; returns the rank-order
+    concatenation of each (the host form of ncclAllGatherv for the
+    broadcast-join right side and the sorted-result merge)."""
+    import numpy as np
+    import torch.distributed as dist
+    gathered = [None] * world_size()
+    dist.all_gather_object(gathered, arrays)
+    return [np.concatenate([g[i] for g in gathered])
+            for i in range(len(arrays))]
+
+
+def global_row_base(local_n: int) -> int:
+    """This rank's global row offset: sum of all earlier ranks' shard
+    lengths (the SPMD frame is the rank-order concat of the shards)."""
+    import torch.distributed as dist
+    gathered = [None] * world_size()
+    dist.all_gather_object(gathered, int(local_n))
+    return sum(gathered[: rank()])
+
+
 def maybe_allreduce_table(table) -> None:
     """RCCL all-reduce of the dense groupby table (reduce phase across GPUs)."""
     if not is_active():

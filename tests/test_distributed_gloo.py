@@ -202,6 +202,77 @@ def _range_shuffle_groupby_worker(rank, world, port, fail_q):
         fail_q.put(f"rank {rank}: {e}\n{traceback.format_exc()}")
 
 
+def _dist_sort_worker(rank, world, port, fail_q):
+    """The distributed sort recipe (dataframe._sort_rows_distributed) on the
+    numpy backend with real collectives: splitters -> dest -> stable
+    per-dest split + positions -> exchange -> local stable sort; the
+    rank-order concat must equal the global stable sort, positions
+    included."""
+    try:
+        os.environ.update({
+            "RANK": str(rank), "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        })
+        import torch
+        import modin_amd.distributed as dist_mod
+        assert dist_mod.init_from_env(backend="gloo", gpu=False)
+        rng = np.random.default_rng(57)
+        n = 20_000
+        keys_g = rng.integers(-1000, 1000, n).astype(np.int64)
+        vals_g = rng.random(n)
+        counts = oracle.split_row_counts(n, world, 32)
+        offs = np.cumsum([0] + counts)
+        k_loc = keys_g[offs[rank]:offs[rank + 1]]
+        v_loc = vals_g[offs[rank]:offs[rank + 1]]
+        base = dist_mod.global_row_base(k_loc.size)
+        assert base == offs[rank]
+        for asc in (True, False):
+            S = min(k_loc.size, 4096)
+            sample = k_loc[np.linspace(0, k_loc.size - 1, S).astype(np.int64)]
+            spl = dist_mod.sample_splitters(sample)
+            dest = oracle.shuffle_dest(k_loc, spl)
+            if not asc:
+                dest = (world - 1) - dest
+            order = np.argsort(dest, kind="stable")
+            sc = np.bincount(dest, minlength=world).tolist()
+            pos = base + np.arange(k_loc.size, dtype=np.int64)
+            rk, _ = dist_mod.exchange_splits(
+                torch.from_numpy(k_loc[order].copy()), sc)
+            rp, _ = dist_mod.exchange_splits(
+                torch.from_numpy(pos[order].copy()), sc)
+            rk, rp = rk.numpy(), rp.numpy()
+            o2 = np.argsort(-rk if not asc else rk, kind="stable")
+            gk, gp = dist_mod.allgather_arrays([rk[o2], rp[o2]])
+            eo = np.argsort(-keys_g if not asc else keys_g, kind="stable")
+            np.testing.assert_array_equal(gk, keys_g[eo])
+            np.testing.assert_array_equal(gp, eo)
+            np.testing.assert_array_equal(vals_g[gp], vals_g[eo])
+        dist_mod.shutdown()
+    except Exception as e:  # pragma: no cover
+        import traceback
+        fail_q.put(f"rank {rank}: {e}\n{traceback.format_exc()}")
+
+
+@pytest.mark.timeout(120)
+@pytest.mark.parametrize("world", [2, 3])
+def test_gloo_distributed_sort_recipe(world):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29550 + world
+    procs = [ctx.Process(target=_dist_sort_worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=110)
+    errs = []
+    while not q.empty():
+        errs.append(q.get())
+    assert not errs, "\n".join(errs)
+    assert all(p.exitcode == 0 for p in procs)
+
+
 @pytest.mark.timeout(120)
 @pytest.mark.parametrize("world", [2, 3])
 def test_gloo_range_shuffle_groupby(world):

@@ -7,6 +7,7 @@ through ctypes exactly as modin_amd calls it.
 import numpy as np
 import pytest
 
+import oracle
 from modin_amd.core import lib
 
 pytestmark = pytest.mark.gpu

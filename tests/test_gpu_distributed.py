@@ -61,6 +61,57 @@ def _worker(rank, world, port, fail_q):
                         out[c].to_numpy(), expect[c].to_numpy(),
                         rtol=1e-12, atol=1e-9, equal_nan=True,
                         err_msg=f"{name}/{agg}/{c}")
+        # ---- distributed merge: every rank probes its left shard against
+        # the all-gathered right table; rank outputs are the pandas-merge
+        # slices of the corresponding left shards ----
+        rngm = np.random.default_rng(43)
+        nl, nr = 30_000, 8_000
+        lk = rngm.integers(0, 5_000, nl).astype(np.int64)
+        lv = rngm.random(nl)
+        rk = rngm.integers(0, 5_000, nr).astype(np.int64)
+        rv = rngm.random(nr)
+        lpdf = pandas.DataFrame({"k": lk, "a": lv})
+        rpdf = pandas.DataFrame({"k": rk, "b": rv})
+        llo, lhi = rank * nl // world, (rank + 1) * nl // world
+        rlo, rhi = rank * nr // world, (rank + 1) * nr // world
+        ldf = mpd.DataFrame(lpdf.iloc[llo:lhi].reset_index(drop=True))
+        rdf = mpd.DataFrame(rpdf.iloc[rlo:rhi].reset_index(drop=True))
+        got = ldf.merge(rdf, on="k").to_pandas()
+        expect = lpdf.iloc[llo:lhi].reset_index(drop=True).merge(rpdf, on="k")
+        for c in ("k", "a", "b"):
+            np.testing.assert_allclose(got[c].to_numpy(),
+                                       expect[c].to_numpy(), rtol=0,
+                                       err_msg=f"merge/{c}")
+
+        # ---- distributed sort_values: range shuffle + local radix sort;
+        # the rank-order concat of the shards must equal pandas exactly,
+        # index (original global positions) included ----
+        rngs = np.random.default_rng(44)
+        ns = 40_000
+        sk = rngs.integers(-2**62, 2**62, ns).astype(np.int64)
+        sk[rngs.random(ns) < 0.15] = -5  # dup run straddling shards
+        sv = rngs.random(ns)
+        spdf = pandas.DataFrame({"k": sk, "v": sv})
+        slo, shi = rank * ns // world, (rank + 1) * ns // world
+        sdf = mpd.DataFrame(spdf.iloc[slo:shi].reset_index(drop=True))
+        for asc in (True, False):
+            out = sdf.sort_values("k", ascending=asc).to_pandas()
+            # local index is shard-relative positions offset by the global
+            # base, i.e. already the GLOBAL original positions
+            # the local index already carries GLOBAL original positions
+            gathered = dist_mod.allgather_arrays(
+                [out.index.to_numpy(),
+                 out["k"].to_numpy(), out["v"].to_numpy()])
+            exp = spdf.sort_values("k", ascending=asc, kind="stable")
+            np.testing.assert_array_equal(gathered[0],
+                                          exp.index.to_numpy(),
+                                          err_msg=f"sort asc={asc} idx")
+            np.testing.assert_array_equal(gathered[1],
+                                          exp["k"].to_numpy(),
+                                          err_msg=f"sort asc={asc} k")
+            np.testing.assert_array_equal(gathered[2],
+                                          exp["v"].to_numpy(),
+                                          err_msg=f"sort asc={asc} v")
         dist_mod.shutdown()
     except Exception as e:  # pragma: no cover
         import traceback
