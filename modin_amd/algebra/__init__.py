@@ -38,6 +38,10 @@ def _map_block(op_code, scalar, f64_only=False):
     """Build a DeviceBlock->DeviceBlock elementwise kernel call."""
 
     def block_fn(block: DeviceBlock) -> DeviceBlock:
+        if block.cats:
+            raise lib.HfError(
+                f"arithmetic on string column(s) {sorted(block.cats)}: "
+                "select numeric columns")
         out = {}
         for name, col in block.columns.items():
             src = lib.cast_f64(col) if (f64_only and col.dtype_code == lib.HF_INT64) \
@@ -84,6 +88,10 @@ class Binary(Operator):
             other_frame = other._modin_frame
 
             def zip_fn(lblock: DeviceBlock, rblock: DeviceBlock) -> DeviceBlock:
+                if lblock.cats or rblock.cats:
+                    raise lib.HfError(
+                        "binary ops on string columns: select numeric "
+                        "columns")
                 out = {}
                 if lblock.width == 1 and rblock.width == 1:
                     # Series op Series: positional pairing, left name wins

@@ -21,20 +21,83 @@ from .partition_manager import HipDataframePartitionManager
 
 
 class DeviceIndex:
-    """Lazy index backed by a device int64 column (groupby keys)."""
+    """Lazy index backed by a device int64 column (groupby keys).  With
+    ``cats`` the column holds dictionary codes and materializes to the
+    category values (string groupby keys)."""
 
-    def __init__(self, col: lib.ColumnRef, name=None):
+    def __init__(self, col: lib.ColumnRef, name=None, cats=None):
         self.col = col
         self.name = name
+        self.cats = cats
         self._cache = None
 
     def materialize(self) -> pandas.Index:
         if self._cache is None:
-            self._cache = pandas.Index(lib.get(self.col), name=self.name)
+            arr = lib.get(self.col)
+            if self.cats is not None:
+                from .partition import decode_dict
+                arr = decode_dict(arr, self.cats)
+            self._cache = pandas.Index(arr, name=self.name)
         return self._cache
 
     def __len__(self):
         return self.col.length
+
+
+def recode_dict_col(codes: lib.ColumnRef, old_cats, new_cats):
+    """Re-express dictionary codes in a new dictionary: host LUT
+    (new_cats.get_indexer(old_cats), −1 rows preserved via a +1-shifted
+    sentinel slot) gathered on device — no string ever touches the GPU."""
+    import numpy as np
+    lut = np.empty(len(old_cats) + 1, dtype=np.int64)
+    lut[0] = -1
+    lut[1:] = new_cats.get_indexer(old_cats)
+    lut_col = lib.put(lut)
+    shifted = lib.map_scalar(lib.MAP_ADD, codes, 1)
+    return lib.gather(lut_col, shifted)
+
+
+def _compare_dict_col(op_code, codes, cats, scalar):
+    """Scalar comparison on a dictionary column IN CODE SPACE: sorted
+    categories make order compares a rank threshold (searchsorted), equality
+    a code lookup; NaN (code −1) compares False except ``ne`` (pandas
+    semantics).  Never touches a string on device."""
+    if op_code == lib.CMP_NOTNA:
+        return lib.compare_scalar(lib.CMP_NE, codes, -1.0)
+    if not isinstance(scalar, str):
+        # pandas: str col == non-str -> all False; != -> all True; order
+        # compares raise
+        if op_code == lib.CMP_EQ:
+            return lib.compare_scalar(lib.CMP_EQ, codes, -2.0)
+        if op_code == lib.CMP_NE:
+            return lib.compare_scalar(lib.CMP_NE, codes, -2.0)
+        raise lib.HfError(
+            f"ordering comparison between string column and {type(scalar)}")
+    if op_code in (lib.CMP_EQ, lib.CMP_NE):
+        locs = cats.get_indexer([scalar])
+        loc = float(locs[0]) if locs[0] >= 0 else -2.0
+        return lib.compare_scalar(op_code, codes, loc)
+    left = float(cats.searchsorted(scalar, side="left"))
+    right = float(cats.searchsorted(scalar, side="right"))
+    if op_code == lib.CMP_GT:     # x > s  <=>  code >= rank_right
+        return lib.compare_scalar(lib.CMP_GE, codes, right)
+    if op_code == lib.CMP_GE:
+        return lib.compare_scalar(lib.CMP_GE, codes, left)
+    # LT/LE: exclude NaN's code −1 (pandas: NaN compares False)
+    thr = left if op_code == lib.CMP_LT else right
+    m = lib.compare_scalar(lib.CMP_LT, codes, thr)
+    notna = lib.compare_scalar(lib.CMP_NE, codes, -1.0)
+    return lib.binary(lib.BIN_MUL, m, notna)
+
+
+def union_cats(a, b):
+    if a.equals(b):
+        return a
+    u = a.union(b)
+    try:
+        return u.sort_values()
+    except TypeError:  # mixed types — keep union order
+        return u
 
 
 class HipDataframe:
@@ -60,9 +123,48 @@ class HipDataframe:
     # ---- ingestion (dataframe.py:4592) ----
     @classmethod
     def from_pandas(cls, df: pandas.DataFrame) -> "HipDataframe":
-        parts, row_lengths = cls._partition_mgr_cls.from_pandas(df)
-        return cls(parts, df.index, df.columns, row_lengths,
-                   df.dtypes.copy())
+        """String/object/category columns dictionary-encode at ingestion
+        (ONE frame-wide sorted dictionary — SURVEY §8f.3); numeric columns
+        upload as-is.  At world>1 the dictionaries are unified across ranks
+        here (all-gather + union + host recode of the codes) so every code
+        means the same value on every rank — collectives then operate on
+        codes alone."""
+        from .partition import encode_dict
+        dtypes = df.dtypes.copy()
+        cats_map = {}
+        edf = df
+        for name in df.columns:
+            dt = df.dtypes[name]
+            if dt in (np.dtype(np.int64), np.dtype(np.float64)):
+                continue
+            if (dt == np.dtype(object)
+                    or isinstance(dt, pandas.CategoricalDtype)
+                    or pandas.api.types.is_string_dtype(dt)):
+                if edf is df:
+                    edf = df.copy()
+                codes, cats = encode_dict(df[name])
+                edf[name] = codes
+                cats_map[name] = cats
+            else:
+                raise lib.HfError(
+                    f"column {name!r} has dtype {dt}: the HipNative backend "
+                    "stores int64/float64 device columns (strings are "
+                    "dictionary-encoded)")
+        from ..distributed import is_active
+        if cats_map and is_active():
+            from .. import distributed as dist_mod
+            for name, cats in cats_map.items():
+                gathered = dist_mod.allgather_arrays(
+                    [cats.to_numpy(dtype=object)])[0]
+                union = pandas.Index(pandas.unique(gathered)).sort_values()
+                if not union.equals(cats):
+                    lut = union.get_indexer(cats)
+                    c = edf[name].to_numpy()
+                    edf[name] = np.where(c >= 0, lut[c], -1)
+                cats_map[name] = union
+        parts, row_lengths = cls._partition_mgr_cls.from_pandas(
+            edf, cats=cats_map)
+        return cls(parts, df.index, df.columns, row_lengths, dtypes)
 
     def to_pandas(self) -> pandas.DataFrame:
         out = self._partition_mgr_cls.to_pandas(self._partitions)
@@ -97,6 +199,13 @@ class HipDataframe:
 
     # ---- TreeReduce (dataframe.py:2208) ----
     def tree_reduce(self, col_names):
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        bad = [c for c in col_names if c in blk_cats]
+        if bad:
+            raise lib.HfError(
+                f"reduction over string column(s) {bad}: select numeric "
+                "columns (string agg is a later round)")
         partials = self._partition_mgr_cls.reduce_partitions(
             self._partitions, col_names
         )
@@ -110,8 +219,35 @@ class HipDataframe:
         val_names = [c for c in self.columns if c != by]
         want_counts = agg in ("count", "mean", "min", "max")
         agg_op = lib.AGG_OP_OF[agg]
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        bad = [v for v in val_names if v in blk_cats]
+        if bad:
+            raise lib.HfError(
+                f"groupby aggregation over string column(s) {bad}: select "
+                "numeric columns (string agg is a later round)")
+        key_cats = blk_cats.get(by)
+        parts = self._partitions
+        if key_cats is not None:
+            # pandas drops NaN groups (dropna=True): filter code == -1 rows
+            has_nan = any(
+                p.block().columns[by].length
+                and lib.reduce(p.block().columns[by]).imn < 0
+                for p in parts)
+            if has_nan:
+                fparts = []
+                for p in parts:
+                    block = p.block()
+                    mask = lib.compare_scalar(lib.CMP_GE,
+                                              block.columns[by], 0.0)
+                    plan = lib.filter_plan(mask)
+                    cols = {m: lib.filter_apply(plan, c)
+                            for m, c in block.columns.items()}
+                    fparts.append(HipDataframePartition(
+                        DeviceBlock(cols, plan.n_kept, block.cats)))
+                parts = fparts
         keys, sums, counts, n = self._partition_mgr_cls.groupby_reduce(
-            self._partitions, by, val_names, want_counts, agg_op
+            parts, by, val_names, want_counts, agg_op
         )
         if agg == "sum":
             cols = {name: sums[i] for i, name in enumerate(val_names)}
@@ -132,8 +268,9 @@ class HipDataframe:
             dtypes = pandas.Series({n_: np.dtype(np.float64) for n_ in val_names})
         block = DeviceBlock(cols, n)
         part = HipDataframePartition(block)
-        return HipDataframe([part], DeviceIndex(keys, name=by), val_names,
-                            [n], dtypes)
+        return HipDataframe([part],
+                            DeviceIndex(keys, name=by, cats=key_cats),
+                            val_names, [n], dtypes)
 
     # ---- broadcast inner join (MergeImpl.row_axis_merge device form,
     #      merge.py:104-178: combine() the right frame once, probe per left
@@ -152,7 +289,32 @@ class HipDataframe:
             cols = [p.block().columns[name] for p in frame._partitions]
             return cols[0] if len(cols) == 1 else lib.concat(cols)
 
+        lcats = (self._partitions[0].block().cats
+                 if self._partitions else {})
+        rcats = (other._partitions[0].block().cats
+                 if other._partitions else {})
+        key_cats = None
+        if (on in lcats) != (on in rcats):
+            raise lib.HfError("merge: key column is a string on one side "
+                              "only")
         rkeys = concat_col(other, on)
+        if on in lcats:
+            # dictionary keys: join in the LEFT dictionary's code space —
+            # recode the right key codes (unmatched right values map to a
+            # fresh negative code so they never match; NaN keys are
+            # rejected loudly: pandas matches NaN==NaN in merges, a later
+            # round here)
+            key_cats = lcats[on]
+            for side, kc in (("left", concat_col(self, on)), ("right", rkeys)):
+                if kc.length and lib.reduce(kc).imn < 0:
+                    raise lib.HfError(
+                        f"merge: NaN in the {side} string key (NaN-key "
+                        "matching is a later round)")
+            if not rcats[on].equals(key_cats):
+                rk2 = recode_dict_col(rkeys, rcats[on], key_cats)
+                # unmatched right categories came back −1: remap to −2..
+                # distinct from NaN semantics is unnecessary (no NaNs here)
+                rkeys = rk2
         if rkeys.dtype_code != lib.HF_INT64:
             raise lib.HfError("merge: key column must be int64 (dense-range "
                               "CSR join; hashed keys are a later round)")
@@ -180,7 +342,11 @@ class HipDataframe:
         # build side)
         cache_key = (on, tuple(right_names), kmin, n_slots)
         cached = getattr(other, "_join_build_cache", None)
-        if cached is not None and cached[0] == cache_key:
+        if key_cats is not None:
+            # dictionary keys: the build lives in the LEFT frame's code
+            # space, which varies per left frame — don't cache on the right
+            j = lib.join_build(rkeys, rvals, kmin, n_slots)
+        elif cached is not None and cached[0] == cache_key:
             j = cached[1]
         else:
             j = lib.join_build(rkeys, rvals, kmin, n_slots)
@@ -210,6 +376,18 @@ class HipDataframe:
             dtypes[on if c == on else lout[c]] = self.dtypes[c]
         for c in right_names:
             dtypes[rout[c]] = other.dtypes[c]
+        out_cats = {}
+        if key_cats is not None:
+            out_cats[on] = key_cats
+        for c in self.columns:
+            if c != on and c in lcats:
+                out_cats[lout[c]] = lcats[c]
+        for c in right_names:
+            if c in rcats:
+                out_cats[rout[c]] = rcats[c]
+        if out_cats:
+            for part in out_parts:
+                part._block.cats = dict(out_cats)
         total = sum(lengths)
         return HipDataframe(out_parts, pandas.RangeIndex(total), out_columns,
                             lengths, pandas.Series(dtypes))
@@ -238,8 +416,8 @@ class HipDataframe:
             cols = {name: lib.filter_apply(plan, col)
                     for name, col in block.columns.items()}
             idx_cols.append(lib.filter_iota(plan, base))
-            out_parts.append(HipDataframePartition(DeviceBlock(cols,
-                                                               plan.n_kept)))
+            out_parts.append(HipDataframePartition(
+                DeviceBlock(cols, plan.n_kept, block.cats)))
             lengths.append(plan.n_kept)
             base += length
         idx_col = idx_cols[0] if len(idx_cols) == 1 else lib.concat(idx_cols)
@@ -249,8 +427,18 @@ class HipDataframe:
     # ---- comparison map (mask column) ----
     def compare_scalar(self, op_code: int, scalar) -> "HipDataframe":
         def block_fn(block: DeviceBlock) -> DeviceBlock:
-            out = {name: lib.compare_scalar(op_code, col, scalar)
-                   for name, col in block.columns.items()}
+            out = {}
+            for name, col in block.columns.items():
+                if name in block.cats:
+                    out[name] = _compare_dict_col(op_code, col,
+                                                  block.cats[name], scalar)
+                else:
+                    if isinstance(scalar, str):
+                        raise lib.HfError(
+                            f"comparing numeric column {name!r} to a "
+                            "string scalar")
+                    out[name] = lib.compare_scalar(op_code, col,
+                                                   float(scalar))
             return DeviceBlock(out, block.length)
         return self.map(block_fn)
 
@@ -267,19 +455,23 @@ class HipDataframe:
                 cols = {n: lib.col_slice(c, lo, hi - lo)
                         for n, c in block.columns.items()}
                 out_parts.append(HipDataframePartition(
-                    DeviceBlock(cols, hi - lo)))
+                    DeviceBlock(cols, hi - lo, block.cats)))
                 lengths.append(hi - lo)
             off += ln
         if not out_parts:
             block = self._partitions[0].block()
             cols = {n: lib.col_slice(c, 0, 0)
                     for n, c in block.columns.items()}
-            out_parts, lengths = [HipDataframePartition(DeviceBlock(cols, 0))], [0]
+            out_parts = [HipDataframePartition(
+                DeviceBlock(cols, 0, block.cats))]
+            lengths = [0]
         idx = self.index[start:stop]
         return HipDataframe(out_parts, idx, self.columns, lengths, self.dtypes)
 
     # ---- astype over all columns ----
     def astype_all(self, dtype) -> "HipDataframe":
+        if self._partitions and self._partitions[0].block().cats:
+            raise lib.HfError("astype on string columns is a later round")
         code = {np.dtype(np.float64): lib.MAP_CAST_F64,
                 np.dtype(np.int64): lib.MAP_CAST_I64}[np.dtype(dtype)]
 
@@ -302,7 +494,33 @@ class HipDataframe:
                     "concat: all frames must share the same columns this "
                     "round (NaN-fill alignment is a later round)")
         frames = [self] + list(others)
-        parts = [p for f in frames for p in f._partitions]
+        # unify per-column dictionaries across frames (host union + device
+        # LUT recode where they differ)
+        all_cats = [f._partitions[0].block().cats if f._partitions else {}
+                    for f in frames]
+        dict_names = sorted({n for c in all_cats for n in c})
+        for n in dict_names:
+            if not all(n in c for c in all_cats):
+                raise lib.HfError(
+                    f"concat: column {n!r} is a string in some frames only")
+        parts = []
+        if dict_names:
+            merged = {n: all_cats[0][n] for n in dict_names}
+            for c in all_cats[1:]:
+                for n in dict_names:
+                    merged[n] = union_cats(merged[n], c[n])
+            for f, fcats in zip(frames, all_cats):
+                for p in f._partitions:
+                    block = p.block()
+                    newcols = dict(block.columns)
+                    for n in dict_names:
+                        if not fcats[n].equals(merged[n]):
+                            newcols[n] = recode_dict_col(
+                                block.columns[n], fcats[n], merged[n])
+                    parts.append(HipDataframePartition(
+                        DeviceBlock(newcols, block.length, dict(merged))))
+        else:
+            parts = [p for f in frames for p in f._partitions]
         lengths = [ln for f in frames for ln in f._row_lengths]
         idx = pandas.Index(np.concatenate([np.asarray(f.index) for f in frames]))
         return HipDataframe(parts, idx, cols, lengths, self.dtypes)
@@ -321,6 +539,14 @@ class HipDataframe:
             cols = [p.block().columns[name] for p in self._partitions]
             return cols[0] if len(cols) == 1 else lib.concat(cols)
 
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        if by in blk_cats:
+            kc = concat_col(by)
+            if kc.length and lib.reduce(kc).imn < 0:
+                raise lib.HfError(
+                    "sort_values: NaN in a string sort key (pandas puts "
+                    "NaNs last; the -1 codes sort first — later round)")
         from ..distributed import is_active
         if is_active():
             return self._sort_rows_distributed(by, ascending, concat_col)
@@ -331,7 +557,7 @@ class HipDataframe:
             src = kcol if name == by else concat_col(name)
             cols[name] = lib.gather(src, perm)
         n = perm.length
-        part = HipDataframePartition(DeviceBlock(cols, n))
+        part = HipDataframePartition(DeviceBlock(cols, n, blk_cats))
         return HipDataframe([part], DeviceIndex(perm, name=None),
                             self.columns, [n], self.dtypes)
 
@@ -381,7 +607,9 @@ class HipDataframe:
         out_cols = {m: lib.gather(recv[m], perm) for m in names}
         pos_sorted = lib.gather(rpos, perm)
         ln = perm.length
-        part = HipDataframePartition(DeviceBlock(out_cols, ln))
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        part = HipDataframePartition(DeviceBlock(out_cols, ln, blk_cats))
         return HipDataframe([part], DeviceIndex(pos_sorted, name=None),
                             names, [ln], self.dtypes)
 
@@ -389,8 +617,11 @@ class HipDataframe:
     def notna_all_mask(self) -> "HipDataframe":
         def block_fn(block: DeviceBlock) -> DeviceBlock:
             acc = None
-            for col in block.columns.values():
-                m = lib.compare_scalar(lib.CMP_NOTNA, col, 0.0)
+            for name, col in block.columns.items():
+                if name in block.cats:  # dict-encoded: NaN is code −1
+                    m = lib.compare_scalar(lib.CMP_NE, col, -1.0)
+                else:
+                    m = lib.compare_scalar(lib.CMP_NOTNA, col, 0.0)
                 acc = m if acc is None else lib.binary(lib.BIN_MUL, acc, m)
             return DeviceBlock({"mask": acc}, block.length)
         return self.map(block_fn)

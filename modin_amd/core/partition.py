@@ -22,17 +22,47 @@ import pandas
 from . import lib
 
 
+def decode_dict(codes: np.ndarray, cats: pandas.Index) -> np.ndarray:
+    """codes (int64, −1 = NaN) -> object array of category values/NaN."""
+    out = np.empty(len(codes), dtype=object)
+    valid = codes >= 0
+    out[valid] = cats.to_numpy(dtype=object)[codes[valid]]
+    out[~valid] = np.nan
+    return out
+
+
+def encode_dict(values: pandas.Series):
+    """values -> (int64 codes with −1 = NaN, sorted category Index).
+
+    Sorted categories make code order == value order, so device-side sort
+    and range comparisons run directly on codes."""
+    codes, cats = pandas.factorize(values, sort=True)
+    return codes.astype(np.int64), pandas.Index(cats)
+
+
 class DeviceBlock:
-    """Ordered mapping column-name -> lib.ColumnRef, all of equal length."""
+    """Ordered mapping column-name -> lib.ColumnRef, all of equal length.
 
-    __slots__ = ("columns", "length")
+    String/object columns are DICTIONARY-ENCODED: the device
+    column holds int64 codes (−1 = NaN), ``cats[name]`` holds the host-side
+    sorted category Index (code order == lexicographic order, so sort /
+    range compares work in code space).  The dictionary is built ONCE per
+    frame at ingestion (HipDataframe.from_pandas) and shared by all
+    partitions; blocks only carry the reference.
+    """
 
-    def __init__(self, columns: dict, length: int):
+    __slots__ = ("columns", "length", "cats")
+
+    def __init__(self, columns: dict, length: int, cats: dict = None):
         self.columns = columns  # name -> ColumnRef
         self.length = length
+        self.cats = cats or {}  # name -> pandas.Index (dict-encoded cols)
 
     @classmethod
-    def from_pandas(cls, df: pandas.DataFrame) -> "DeviceBlock":
+    def from_pandas(cls, df: pandas.DataFrame,
+                    cats: dict = None) -> "DeviceBlock":
+        """``df`` must already be device-typed: int64/float64 columns, with
+        string columns pre-encoded to int64 codes (listed in ``cats``)."""
         cols = {}
         for name in df.columns:
             arr = df[name].to_numpy()
@@ -42,16 +72,23 @@ class DeviceBlock:
                     "backend stores int64/float64 device columns only"
                 )
             cols[name] = lib.put(arr)
-        return cls(cols, len(df))
+        mycats = {n: c for n, c in (cats or {}).items() if n in cols}
+        return cls(cols, len(df), mycats)
 
     def to_pandas(self, index=None) -> pandas.DataFrame:
-        data = {name: lib.get(col) for name, col in self.columns.items()}
+        data = {}
+        for name, col in self.columns.items():
+            arr = lib.get(col)
+            if name in self.cats:
+                arr = decode_dict(arr, self.cats[name])
+            data[name] = arr
         if index is None:
             index = pandas.RangeIndex(self.length)
         return pandas.DataFrame(data, index=index)
 
     def select(self, names) -> "DeviceBlock":
-        return DeviceBlock({n: self.columns[n] for n in names}, self.length)
+        return DeviceBlock({n: self.columns[n] for n in names}, self.length,
+                           {n: self.cats[n] for n in names if n in self.cats})
 
     @property
     def width(self) -> int:
@@ -93,9 +130,11 @@ class HipDataframePartition:
         lib.sync()
 
     @classmethod
-    def put(cls, df: pandas.DataFrame) -> "HipDataframePartition":
-        """(partition.py:277) — H2D upload of a pandas block."""
-        return cls(DeviceBlock.from_pandas(df))
+    def put(cls, df: pandas.DataFrame,
+            cats: dict = None) -> "HipDataframePartition":
+        """(partition.py:277) — H2D upload of a pandas block (string
+        columns pre-encoded; ``cats`` holds their dictionaries)."""
+        return cls(DeviceBlock.from_pandas(df, cats))
 
     def get(self) -> pandas.DataFrame:
         self.drain_call_queue()

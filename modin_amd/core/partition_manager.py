@@ -97,7 +97,7 @@ class HipDataframePartitionManager:
 
     # ---- ingestion (partition_manager.py:1070) ----
     @classmethod
-    def from_pandas(cls, df: pandas.DataFrame, num_splits=None):
+    def from_pandas(cls, df: pandas.DataFrame, num_splits=None, cats=None):
         if num_splits is None:
             num_splits = config.NPartitions.get()
         n = len(df)
@@ -107,11 +107,11 @@ class HipDataframePartitionManager:
         start = 0
         while start < n:
             stop = min(start + chunk, n)
-            parts.append(cls._partition_class.put(df.iloc[start:stop]))
+            parts.append(cls._partition_class.put(df.iloc[start:stop], cats))
             row_lengths.append(stop - start)
             start = stop
         if not parts:  # empty frame: keep one empty partition
-            parts = [cls._partition_class.put(df)]
+            parts = [cls._partition_class.put(df, cats)]
             row_lengths = [0]
         return parts, row_lengths
 

@@ -98,7 +98,7 @@ class HipQueryCompiler:
         if not np.isscalar(other):
             raise lib.HfError("comparisons support scalars this round")
         return self.__constructor__(
-            self._modin_frame.compare_scalar(op_code, float(other)))
+            self._modin_frame.compare_scalar(op_code, other))
 
     def gt(self, other):
         return self._compare(lib.CMP_GT, other)

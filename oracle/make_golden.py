@@ -322,6 +322,137 @@ def gen_hash_groupby_cases(mpd, rng):
     return cases
 
 
+NA = "__NA__"  # NaN sentinel inside '<U' string arrays (npz, no pickle)
+
+
+def _enc_str(values):
+    """pandas object/str values -> '<U' array with NA sentinel."""
+    import pandas
+    return np.array([NA if (v is None or (isinstance(v, float) and np.isnan(v)))
+                     else str(v) for v in values])
+
+
+def gen_string_cases(mpd, rng):
+    """Dictionary-encoded string column parity: groupby BY a
+    string key, merge ON a string key, string comparisons/filters, sort by
+    a string column, concat with differing dictionaries — all run through
+    the REAL reference and cross-checked vs pandas."""
+    import pandas
+    cases = {}
+    pool = np.array(["apple", "pear", "zebra", "kiwi", "mango", "fig",
+                     "plum", "apricot", "melon", "lime", "Grape", "banana"])
+
+    # ---- groupby by string key (NaN keys dropped by pandas) ----
+    n = 4000
+    ks = rng.choice(pool, n).astype(object)
+    ks[rng.random(n) < 0.05] = np.nan
+    v = rng.random(n)
+    v[rng.random(n) < 0.1] = np.nan
+    w = rng.standard_normal(n)
+    mdf = mpd.DataFrame({"s": ks, "v": v, "w": w})
+    pdf = pandas.DataFrame({"s": ks, "v": v, "w": w})
+    arrays = {"in_s": _enc_str(ks), "in_v": v, "in_w": w}
+    for agg in ("sum", "count", "mean", "min", "max"):
+        mres = getattr(mdf.groupby("s"), agg)()
+        pres = getattr(pdf.groupby("s"), agg)()
+        got = mres._to_pandas()
+        assert list(got.index) == list(pres.index)
+        np.testing.assert_allclose(got.values.astype(float),
+                                   pres.values.astype(float), rtol=1e-12)
+        arrays[f"out_{agg}_keys"] = _enc_str(got.index)
+        for cn in ("v", "w"):
+            arrays[f"out_{agg}_{cn}"] = got[cn].to_numpy()
+    cases["str_gb"] = arrays
+
+    # ---- merge on string key (differing dictionaries both sides) ----
+    nl, nr = 3000, 800
+    lk = rng.choice(pool[:10], nl).astype(object)
+    rk = rng.choice(pool[4:], nr).astype(object)  # overlap + right-only cats
+    la, rb = rng.random(nl), rng.random(nr)
+    mout = mpd.DataFrame({"s": lk, "a": la}).merge(
+        mpd.DataFrame({"s": rk, "b": rb}), on="s")._to_pandas()
+    pout = pandas.DataFrame({"s": lk, "a": la}).merge(
+        pandas.DataFrame({"s": rk, "b": rb}), on="s")
+    assert list(mout["s"]) == list(pout["s"])
+    np.testing.assert_allclose(mout[["a", "b"]].values,
+                               pout[["a", "b"]].values, rtol=0)
+    cases["str_merge"] = {
+        "in_lk": _enc_str(lk), "in_la": la,
+        "in_rk": _enc_str(rk), "in_rb": rb,
+        "out_s": _enc_str(mout["s"]), "out_a": mout["a"].to_numpy(),
+        "out_b": mout["b"].to_numpy(),
+        "out_idx": mout.index.to_numpy().astype(np.int64),
+    }
+
+    # ---- string comparisons / filter ----
+    nf = 3000
+    fs = rng.choice(pool, nf).astype(object)
+    fs[rng.random(nf) < 0.05] = np.nan
+    fv = rng.random(nf)
+    mdf = mpd.DataFrame({"s": fs, "v": fv})
+    pdf = pandas.DataFrame({"s": fs, "v": fv})
+    farr = {"in_s": _enc_str(fs), "in_v": fv}
+    for tag, fn in [("eq", lambda d: d["s"] == "apple"),
+                    ("ne", lambda d: d["s"] != "apple"),
+                    ("gt", lambda d: d["s"] > "m"),
+                    ("le", lambda d: d["s"] <= "kiwi"),
+                    ("eq_missing", lambda d: d["s"] == "notthere")]:
+        mres = mdf[fn(mdf)]._to_pandas()
+        pres = pdf[fn(pdf)]
+        assert list(mres.index) == list(pres.index)
+        assert list(mres["s"].fillna(NA)) == list(pres["s"].fillna(NA))
+        farr[f"out_{tag}_idx"] = pres.index.to_numpy().astype(np.int64)
+        farr[f"out_{tag}_s"] = _enc_str(pres["s"])
+        farr[f"out_{tag}_v"] = pres["v"].to_numpy()
+    # dropna over the string column
+    mres = mdf.dropna()._to_pandas()
+    pres = pdf.dropna()
+    assert list(mres.index) == list(pres.index)
+    farr["out_dropna_idx"] = pres.index.to_numpy().astype(np.int64)
+    farr["out_dropna_s"] = _enc_str(pres["s"])
+    cases["str_filter"] = farr
+
+    # ---- sort by string column (no NaN: na_position is a later round) ----
+    ns = 2500
+    ss = rng.choice(pool, ns).astype(object)
+    sv = rng.random(ns)
+    mres = mpd.DataFrame({"s": ss, "v": sv}).sort_values("s",
+                                                         kind="stable")
+    pres = pandas.DataFrame({"s": ss, "v": sv}).sort_values("s",
+                                                            kind="stable")
+    mres = mres._to_pandas()
+    assert list(mres["s"]) == list(pres["s"])
+    cases["str_sort"] = {
+        "in_s": _enc_str(ss), "in_v": sv,
+        "out_idx": pres.index.to_numpy().astype(np.int64),
+        "out_s": _enc_str(pres["s"]), "out_v": pres["v"].to_numpy(),
+    }
+
+    # ---- concat with differing dictionaries, then groupby ----
+    n1, n2 = 1500, 1200
+    c1 = rng.choice(pool[:6], n1).astype(object)
+    c2 = rng.choice(pool[3:], n2).astype(object)
+    v1, v2 = rng.random(n1), rng.random(n2)
+    mcat = mpd.concat([mpd.DataFrame({"s": c1, "v": v1}),
+                       mpd.DataFrame({"s": c2, "v": v2})],
+                      ignore_index=True)
+    pcat = pandas.concat([pandas.DataFrame({"s": c1, "v": v1}),
+                          pandas.DataFrame({"s": c2, "v": v2})],
+                         ignore_index=True)
+    mg = mcat.groupby("s").sum()._to_pandas()
+    pg = pcat.groupby("s").sum()
+    assert list(mg.index) == list(pg.index)
+    np.testing.assert_allclose(mg["v"].to_numpy(), pg["v"].to_numpy(),
+                               rtol=1e-12)
+    cases["str_concat"] = {
+        "in_s1": _enc_str(c1), "in_v1": v1,
+        "in_s2": _enc_str(c2), "in_v2": v2,
+        "out_cat_s": _enc_str(pcat["s"]),
+        "out_gb_keys": _enc_str(pg.index), "out_gb_v": pg["v"].to_numpy(),
+    }
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -334,6 +465,7 @@ def main():
     all_cases.update(gen_filter_cases(mpd, rng))
     all_cases.update(gen_sort_cases(mpd, rng))
     all_cases.update(gen_hash_groupby_cases(mpd, rng))
+    all_cases.update(gen_string_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)
