@@ -23,6 +23,15 @@ from ..query_compiler import HipQueryCompiler
 __all__ = ["DataFrame", "Series", "concat", "from_pandas"]
 
 
+def read_parquet(path, columns=None):
+    """Columnar parquet ingestion straight to device:
+    pyarrow -> numpy views / dictionary parts -> hf_put.  Matches
+    pandas.read_parquet output (strings as object, nullable ints as
+    float64, RangeIndex)."""
+    from ..io import read_parquet as _rp
+    return DataFrame(query_compiler=_rp(path, columns=columns))
+
+
 def concat(objs, ignore_index: bool = False):
     """pandas.concat(axis=0) over DataFrames with identical columns."""
     objs = list(objs)

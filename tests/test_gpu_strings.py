@@ -168,3 +168,53 @@ def test_string_category_dtype_roundtrip(npartitions):
     assert list(srt["c"]) == list(exp["c"])
     np.testing.assert_array_equal(srt.index.to_numpy(),
                                   exp.index.to_numpy())
+
+
+def test_read_parquet_vs_pandas(tmp_path, npartitions):
+    """§8f.4 columnar ingestion: pyarrow parquet -> device columns; output
+    matches pandas.read_parquet (strings object + NaN, nullable ints as
+    float64, projection, multi-row-group files)."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    rng = np.random.default_rng(71)
+    n = 20_000
+    s_vals = rng.choice(np.array(["red", "green", "blue", "Amber", "x y"]),
+                        n).astype(object)
+    s_vals[rng.random(n) < 0.07] = None
+    v = rng.random(n)
+    v[rng.random(n) < 0.05] = np.nan
+    k = rng.integers(-1000, 1000, n)
+    ni = k.astype(object)
+    ni[rng.random(n) < 0.03] = None  # nullable int -> float64 in pandas
+    table = pa.table({
+        "s": pa.array(s_vals),
+        "sd": pa.array(s_vals).dictionary_encode(),
+        "v": pa.array(v),
+        "k": pa.array(k, type=pa.int64()),
+        "ni": pa.array(ni.tolist(), type=pa.int64()),
+    })
+    path = str(tmp_path / "t.parquet")
+    pq.write_table(table, path, row_group_size=3000)  # multi-row-group
+    got = mpd.read_parquet(path).to_pandas()
+    exp = pandas.read_parquet(path)
+    assert list(got.columns) == list(exp.columns)
+    for c in ("s", "sd"):
+        assert_str_equal(got[c].to_numpy(),
+                         np.array([NA if x is None else str(x)
+                                   for x in exp[c]]), c)
+    np.testing.assert_array_equal(got["v"].to_numpy(), exp["v"].to_numpy())
+    np.testing.assert_array_equal(got["k"].to_numpy(), exp["k"].to_numpy())
+    assert got["ni"].dtype == np.float64
+    np.testing.assert_array_equal(got["ni"].to_numpy(),
+                                  exp["ni"].to_numpy())
+    # projection
+    got2 = mpd.read_parquet(path, columns=["k", "s"]).to_pandas()
+    assert list(got2.columns) == ["k", "s"]
+    np.testing.assert_array_equal(got2["k"].to_numpy(), exp["k"].to_numpy())
+    # the ingested frame computes: groupby by the string key vs pandas
+    df = mpd.read_parquet(path, columns=["s", "v"])
+    g1 = df.groupby("s").sum().to_pandas()
+    g2 = exp[["s", "v"]].groupby("s").sum()
+    assert list(g1.index) == list(g2.index)
+    np.testing.assert_allclose(g1["v"].to_numpy(), g2["v"].to_numpy(),
+                               rtol=RTOL, atol=1e-9)

@@ -114,3 +114,28 @@ def test_pick_splitters_quantiles():
     assert np.all(np.diff(spl) >= 0)
     d = oracle.shuffle_dest(sample, spl)
     assert d.min() >= 0 and d.max() <= 7
+
+
+def test_parquet_column_conversion():
+    """io._column_to_device_ready: dictionary remap onto the sorted-cats
+    invariant, null semantics (float NaN, nullable-int -> f64, string -1),
+    narrow int widening — all host-side, no GPU."""
+    import pyarrow as pa
+    from modin_amd.io import _column_to_device_ready
+    arr = pa.array(["pear", "apple", None, "pear", "zebra"]).dictionary_encode()
+    codes, cats = _column_to_device_ready(pa.chunked_array([arr]), "s")
+    assert list(cats) == ["apple", "pear", "zebra"]
+    assert codes.tolist() == [1, 0, -1, 1, 2]
+    # plain (non-dictionary) string column dictionary-encodes on the fly
+    codes2, cats2 = _column_to_device_ready(
+        pa.chunked_array([pa.array(["b", "a", "b", None])]), "t")
+    assert list(cats2) == ["a", "b"] and codes2.tolist() == [1, 0, 1, -1]
+    v, c = _column_to_device_ready(
+        pa.chunked_array([pa.array([1.5, None, 2.5])]), "v")
+    assert c is None and np.isnan(v[1]) and v[0] == 1.5
+    v2, _ = _column_to_device_ready(
+        pa.chunked_array([pa.array([1, 2, None], type=pa.int64())]), "i")
+    assert v2.dtype == np.float64 and np.isnan(v2[2])
+    v3, _ = _column_to_device_ready(
+        pa.chunked_array([pa.array([7, 8], type=pa.int32())]), "j")
+    assert v3.dtype == np.int64 and v3.tolist() == [7, 8]
