@@ -249,17 +249,42 @@ class HipDataframe:
         keys, sums, counts, n = self._partition_mgr_cls.groupby_reduce(
             parts, by, val_names, want_counts, agg_op
         )
+        # pandas dtype preservation: sum/min/max of an int64 column stays
+        # int64 (the f64 accumulators are exact — ingestion guards |x|<2^53,
+        # and the per-group sum is checked below before the cast back).
+        int_vals = {n_ for n_ in val_names
+                    if self.dtypes[n_] == np.dtype(np.int64)}
+
+        def back_to_int(name, col):
+            if name not in int_vals:
+                return col, np.dtype(np.float64)
+            if col.length:
+                r = lib.reduce(col)
+                if max(abs(r.mn if r.mn == r.mn else 0.0),
+                       abs(r.mx if r.mx == r.mx else 0.0)) >= 2.0**53:
+                    raise lib.HfError(
+                        f"groupby {agg} of int64 column {name!r} exceeds "
+                        "2^53: exact int accumulation is a later round")
+            return lib.map_scalar(lib.MAP_CAST_I64, col, 0), \
+                np.dtype(np.int64)
+
         if agg == "sum":
-            cols = {name: sums[i] for i, name in enumerate(val_names)}
-            dtypes = pandas.Series({n_: np.dtype(np.float64) for n_ in val_names})
+            cols, dts = {}, {}
+            for i, name in enumerate(val_names):
+                cols[name], dts[name] = back_to_int(name, sums[i])
+            dtypes = pandas.Series(dts)
         elif agg == "count":
             cols = {name: counts[i] for i, name in enumerate(val_names)}
             dtypes = pandas.Series({n_: np.dtype(np.int64) for n_ in val_names})
         elif agg in ("min", "max"):
             # empty (all-NaN) groups hold the agg identity; pandas says NaN
-            cols = {name: lib.fixup_empty(sums[i], counts[i])
-                    for i, name in enumerate(val_names)}
-            dtypes = pandas.Series({n_: np.dtype(np.float64) for n_ in val_names})
+            # (int64 columns can't produce all-NaN groups, so the cast back
+            # is safe)
+            cols, dts = {}, {}
+            for i, name in enumerate(val_names):
+                fixed = lib.fixup_empty(sums[i], counts[i])
+                cols[name], dts[name] = back_to_int(name, fixed)
+            dtypes = pandas.Series(dts)
         else:  # mean = sums / counts (GroupbyReduceImpl mean shape, groupby.py:87)
             cols = {}
             for i, name in enumerate(val_names):

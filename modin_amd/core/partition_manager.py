@@ -207,7 +207,17 @@ class HipDataframePartitionManager:
                     "path); hashed/float keys are a later round"
                 )
             key_cols.append(kcol)
-            vals = [lib.cast_f64(block.columns[v]) for v in val_names]
+            vals = []
+            for v in val_names:
+                c = block.columns[v]
+                if c.dtype_code == lib.HF_INT64 and c.length:
+                    r = lib.reduce(c)
+                    if max(abs(r.imn), abs(r.imx)) >= 1 << 53:
+                        raise lib.HfError(
+                            f"groupby over int64 column {v!r} with values "
+                            "beyond 2^53: f64 accumulation would round — "
+                            "exact int accumulation is a later round")
+                vals.append(lib.cast_f64(c))
             val_cols_per_part.append(vals)
             if kcol.length:
                 r = lib.reduce(kcol)

@@ -200,8 +200,9 @@ def test_read_parquet_vs_pandas(tmp_path, npartitions):
     assert list(got.columns) == list(exp.columns)
     for c in ("s", "sd"):
         assert_str_equal(got[c].to_numpy(),
-                         np.array([NA if x is None else str(x)
-                                   for x in exp[c]]), c)
+                         np.array([NA if (x is None or (isinstance(x, float)
+                                                        and np.isnan(x)))
+                                   else str(x) for x in exp[c]]), c)
     np.testing.assert_array_equal(got["v"].to_numpy(), exp["v"].to_numpy())
     np.testing.assert_array_equal(got["k"].to_numpy(), exp["k"].to_numpy())
     assert got["ni"].dtype == np.float64
