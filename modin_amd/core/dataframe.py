@@ -552,9 +552,51 @@ class HipDataframe:
 
     # ---- sort (PandasDataframe.sort_by device form, dataframe.py:2742;
     #      SURVEY §8f.2): stable radix permutation + column gathers ----
-    def sort_rows(self, by: str, ascending: bool = True) -> "HipDataframe":
-        if by not in self.columns:
-            raise lib.HfError(f"sort_values: column {by!r} missing")
+    @staticmethod
+    def _effective_sort_key(col, is_dict, ascending):
+        """(key col, ascending) -> (int64 key, ascending') whose stable
+        ASCENDING' radix sort realizes pandas order.  Dictionary NaN codes
+        (−1) must sort LAST for both directions (na_position='last'):
+        ascending maps −1 to +2^62, descending negates the codes (order
+        flip) and maps −1 above them — so the pass always sorts
+        ascending."""
+        if not is_dict or not col.length or lib.reduce(col).imn >= 0:
+            return col, ascending
+        BIG = 1 << 62
+        m = lib.compare_scalar(lib.CMP_EQ, col, -1.0)
+        if ascending:
+            t = lib.map_scalar(lib.MAP_MUL, m, BIG + 1)
+            return lib.binary(lib.BIN_ADD, col, t), True
+        neg = lib.map_scalar(lib.MAP_NEG, col, 0)
+        t = lib.map_scalar(lib.MAP_MUL, m, BIG - 1)
+        return lib.binary(lib.BIN_ADD, neg, t), True
+
+    @staticmethod
+    def _compose_sort_perm(eff_keys):
+        """Stable multi-key permutation: LSD over the key list (sort by the
+        LAST key first; stability carries earlier keys' order through) —
+        the device form of pandas lexsort semantics."""
+        perm = None
+        for ekc, ea in reversed(eff_keys):
+            if perm is None:
+                perm = lib.sort_perm(ekc, ea)
+            else:
+                gk = lib.gather(ekc, perm)
+                p2 = lib.sort_perm(gk, ea)
+                perm = lib.gather(perm, p2)
+        return perm
+
+    def sort_rows(self, by, ascending=True) -> "HipDataframe":
+        by_list = [by] if isinstance(by, str) else list(by)
+        if isinstance(ascending, (bool, np.bool_, int)):
+            asc_list = [bool(ascending)] * len(by_list)
+        else:
+            asc_list = [bool(a) for a in ascending]
+        if len(asc_list) != len(by_list) or not by_list:
+            raise lib.HfError("sort_values: by/ascending length mismatch")
+        for b in by_list:
+            if b not in self.columns:
+                raise lib.HfError(f"sort_values: column {b!r} missing")
         if not isinstance(self._index, pandas.RangeIndex) or \
                 self._index.start != 0 or self._index.step != 1:
             raise lib.HfError(
@@ -566,57 +608,71 @@ class HipDataframe:
 
         blk_cats = (self._partitions[0].block().cats
                     if self._partitions else {})
-        if by in blk_cats:
-            kc = concat_col(by)
-            if kc.length and lib.reduce(kc).imn < 0:
-                raise lib.HfError(
-                    "sort_values: NaN in a string sort key (pandas puts "
-                    "NaNs last; the -1 codes sort first — later round)")
         from ..distributed import is_active
         if is_active():
-            return self._sort_rows_distributed(by, ascending, concat_col)
-        kcol = concat_col(by)
-        perm = lib.sort_perm(kcol, ascending)
-        cols = {}
-        for name in self.columns:
-            src = kcol if name == by else concat_col(name)
-            cols[name] = lib.gather(src, perm)
+            return self._sort_rows_distributed(by_list, asc_list, blk_cats,
+                                               concat_col)
+        cache = {}
+
+        def cat_col(name):
+            if name not in cache:
+                cache[name] = concat_col(name)
+            return cache[name]
+
+        eff = []
+        for b, a in zip(by_list, asc_list):
+            kc = cat_col(b)
+            if kc.dtype_code != lib.HF_INT64:
+                raise lib.HfError(
+                    f"sort_values: key column {b!r} must be int64 or "
+                    "string (float sort keys are a later round)")
+            eff.append(self._effective_sort_key(kc, b in blk_cats, a))
+        perm = self._compose_sort_perm(eff)
+        cols = {name: lib.gather(cat_col(name), perm)
+                for name in self.columns}
         n = perm.length
         part = HipDataframePartition(DeviceBlock(cols, n, blk_cats))
         return HipDataframe([part], DeviceIndex(perm, name=None),
                             self.columns, [n], self.dtypes)
 
-    def _sort_rows_distributed(self, by, ascending, concat_col):
+    def _sort_rows_distributed(self, by_list, asc_list, blk_cats,
+                               concat_col):
         """Distributed sort_values: the range-partitioning shuffle + local
         stable sort (SURVEY §8f.2 "reuses the shuffle"; reference
         sort_by -> _apply_func_to_range_partitioning, dataframe.py:2742).
-        Sampled splitters put each key RANGE on one rank (rank order =
-        global key order), rows move once (exchange_splits), each rank
-        radix-sorts its range.  Stability: exchange output is source-rank
-        (= global-position) ordered and the per-dest filter preserves
-        order, so the local stable sort reproduces pandas order exactly.
-        Result: the frame stays sharded, rank r holding globally-sorted
-        slice r; the index carries the original global positions."""
+        Rows shuffle by the PRIMARY effective key (equal primaries share a
+        rank, so secondary keys stay a local matter), each rank then runs
+        the same multi-key stable sort the single-rank path uses.
+        Stability: exchange output is source-rank (= global-position)
+        ordered and the per-dest filter preserves order.  Result: rank r
+        holds globally-sorted slice r; the index carries the original
+        global positions."""
         import numpy as np
         from .. import distributed as dist_mod
         P = dist_mod.world_size()
-        kcol = concat_col(by)
-        n = kcol.length
+        k0 = concat_col(by_list[0])
+        if k0.dtype_code != lib.HF_INT64:
+            raise lib.HfError(
+                f"sort_values: key column {by_list[0]!r} must be int64 or "
+                "string")
+        ek0, ea0 = self._effective_sort_key(k0, by_list[0] in blk_cats,
+                                            asc_list[0])
+        n = ek0.length
         S = min(n, 4096)
         if S:
             sidx = np.linspace(0, n - 1, S).astype(np.int64)
-            sample = lib.get(lib.gather(kcol, lib.put(sidx)))
+            sample = lib.get(lib.gather(ek0, lib.put(sidx)))
         else:
             sample = np.empty(0, dtype=np.int64)
         splitters = dist_mod.sample_splitters(sample)
-        dest = lib.shuffle_dest(kcol, splitters)
-        if not ascending:  # rank 0 takes the LARGEST key range
+        dest = lib.shuffle_dest(ek0, splitters)
+        if not ea0:  # rank 0 takes the LARGEST primary-key range
             dest = lib.map_scalar(lib.MAP_RSUB, dest, P - 1)
         base = dist_mod.global_row_base(n)
         names = list(self.columns)
         send_cols = {m: [] for m in names}
         send_pos, send_counts = [], []
-        cols_cat = {m: (kcol if m == by else concat_col(m)) for m in names}
+        cols_cat = {m: concat_col(m) for m in names}
         for d in range(P):
             mask = lib.compare_scalar(lib.CMP_EQ, dest, float(d))
             plan = lib.filter_plan(mask)
@@ -628,12 +684,12 @@ class HipDataframe:
                                             send_counts)
                 for m in names}
         rpos = dist_mod.exchange_column(lib.concat(send_pos), send_counts)
-        perm = lib.sort_perm(recv[by], ascending)
+        eff = [self._effective_sort_key(recv[b], b in blk_cats, a)
+               for b, a in zip(by_list, asc_list)]
+        perm = self._compose_sort_perm(eff)
         out_cols = {m: lib.gather(recv[m], perm) for m in names}
         pos_sorted = lib.gather(rpos, perm)
         ln = perm.length
-        blk_cats = (self._partitions[0].block().cats
-                    if self._partitions else {})
         part = HipDataframePartition(DeviceBlock(out_cols, ln, blk_cats))
         return HipDataframe([part], DeviceIndex(pos_sorted, name=None),
                             names, [ln], self.dtypes)
