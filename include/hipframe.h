@@ -98,7 +98,8 @@ enum {
   HF_MAP_ABS = 7,   /* |x|         */
   HF_MAP_NEG = 8,   /* -x          */
   HF_MAP_CAST_F64 = 9, /* (double)x : i64 -> f64; scalar ignored */
-  HF_MAP_CAST_I64 = 10 /* (int64)x : f64 -> i64, C truncation (astype)     */
+  HF_MAP_CAST_I64 = 10, /* (int64)x : f64 -> i64, C truncation (astype)    */
+  HF_MAP_SQRT = 11  /* sqrt(x), f64 only (std = sqrt(var))                  */
 };
 int hf_map_scalar(int op, const hf_col* in, double scalar, hf_col** out);
 /* i64 column with an exact int64 scalar (double cannot hold all int64). */

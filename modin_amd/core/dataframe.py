@@ -297,6 +297,141 @@ class HipDataframe:
                             DeviceIndex(keys, name=by, cats=key_cats),
                             val_names, [n], dtypes)
 
+    def groupby_var(self, by: str, ddof: int = 1,
+                    sqrt: bool = False) -> "HipDataframe":
+        """groupby var/std: one extra squared-values pass through the SAME
+        sum machinery (sums of x and x², non-NaN counts), composed on
+        device as (Σx² − (Σx)²/n)/(n−ddof), clamped at 0 (cancellation)
+        and fixed to NaN where n <= ddof (pandas nanvar shape, reference
+        groupby var via GroupbyReduceImpl-style pairs)."""
+        val_names = [c for c in self.columns if c != by]
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        bad = [v for v in val_names if v in blk_cats]
+        if bad:
+            raise lib.HfError(
+                f"groupby var/std over string column(s) {bad}")
+        keys, key_cats, sums, sumsqs, counts, n = self._groupby_moments(
+            by, val_names, blk_cats)
+        cols = {}
+        for i, name in enumerate(val_names):
+            cnt_f = lib.cast_f64(counts[i])
+            t = lib.binary(lib.BIN_MUL, sums[i], sums[i])
+            t = lib.binary(lib.BIN_DIV, t, cnt_f)
+            t = lib.binary(lib.BIN_SUB, sumsqs[i], t)
+            denom_i = lib.map_scalar(lib.MAP_ADD, counts[i], -ddof)
+            var = lib.binary(lib.BIN_DIV, t, lib.cast_f64(denom_i))
+            # clamp tiny negative cancellation: (v+|v|)/2
+            var = lib.map_scalar(
+                lib.MAP_MUL,
+                lib.binary(lib.BIN_ADD, var,
+                           lib.map_scalar(lib.MAP_ABS, var, 0)), 0.5)
+            # n <= ddof (incl. all-NaN groups) -> NaN
+            okm = lib.compare_scalar(lib.CMP_GE, denom_i, 1.0)
+            var = lib.fixup_empty(var,
+                                  lib.binary(lib.BIN_MUL, denom_i, okm))
+            if sqrt:
+                var = lib.map_scalar(lib.MAP_SQRT, var, 0.0)
+            cols[name] = var
+        block = DeviceBlock(cols, n)
+        part = HipDataframePartition(block)
+        dtypes = pandas.Series({v: np.dtype(np.float64) for v in val_names})
+        return HipDataframe([part],
+                            DeviceIndex(keys, name=by, cats=key_cats),
+                            val_names, [n], dtypes)
+
+    def _groupby_moments(self, by: str, val_names, blk_cats):
+        """(keys, key_cats, sums, sumsqs, counts, n) via one SUM pass over
+        values and their squares."""
+        SQ = "\x00sq\x00"
+
+        def add_sq(block: DeviceBlock) -> DeviceBlock:
+            cols = dict(block.columns)
+            for v in val_names:
+                f = lib.cast_f64(block.columns[v])
+                cols[SQ + v] = lib.binary(lib.BIN_MUL, f, f)
+            return DeviceBlock(cols, block.length, block.cats)
+
+        parts = self._partition_mgr_cls.map_partitions(self._partitions,
+                                                       add_sq)
+        key_cats = blk_cats.get(by)
+        if key_cats is not None:
+            fparts = []
+            for p in parts:
+                block = p.block()
+                if block.columns[by].length and \
+                        lib.reduce(block.columns[by]).imn < 0:
+                    mask = lib.compare_scalar(lib.CMP_GE,
+                                              block.columns[by], 0.0)
+                    plan = lib.filter_plan(mask)
+                    cols = {m: lib.filter_apply(plan, c)
+                            for m, c in block.columns.items()}
+                    fparts.append(HipDataframePartition(
+                        DeviceBlock(cols, plan.n_kept, block.cats)))
+                else:
+                    fparts.append(p)
+            parts = fparts
+        ext_names = list(val_names) + [SQ + v for v in val_names]
+        keys, sums, counts, n = self._partition_mgr_cls.groupby_reduce(
+            parts, by, ext_names, True, lib.AGG_SUM)
+        k = len(val_names)
+        return keys, key_cats, sums[:k], sums[k:], counts[:k], n
+
+    def groupby_size(self, by: str) -> "HipDataframe":
+        """groupby().size(): group row counts INCLUDING NaN values (and
+        NaN keys dropped) — a ones column through the sum path."""
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+
+        def ones(block: DeviceBlock) -> DeviceBlock:
+            one = lib.map_scalar(
+                lib.MAP_ADD,
+                lib.map_scalar(lib.MAP_MUL,
+                               lib.cast_f64(block.columns[by])
+                               if block.columns[by].dtype_code
+                               == lib.HF_INT64 else block.columns[by], 0.0),
+                1.0)
+            return DeviceBlock({by: block.columns[by], "\x00size\x00": one},
+                               block.length, block.cats)
+
+        ones_frame = HipDataframe(
+            self._partition_mgr_cls.map_partitions(self._partitions, ones),
+            self._index, [by, "\x00size\x00"], self._row_lengths,
+            pandas.Series({by: self.dtypes[by],
+                           "\x00size\x00": np.dtype(np.float64)}))
+        res = ones_frame.groupby_reduce(by, "sum")
+        # ones are never NaN, so the f64 sums are exact ints: cast
+        block = res._partitions[0].block()
+        col = lib.map_scalar(lib.MAP_CAST_I64,
+                             block.columns["\x00size\x00"], 0)
+        part = HipDataframePartition(DeviceBlock({"size": col}, res._row_lengths[0]))
+        return HipDataframe([part], res._index, ["size"],
+                            res._row_lengths,
+                            pandas.Series({"size": np.dtype(np.int64)}))
+
+    def hconcat(self, others: list) -> "HipDataframe":
+        """Horizontal compose of single-partition frames with identical row
+        count and index (the device form of the reference's axis=1 concat
+        over aligned frames — used by agg-list/dict composition)."""
+        frames = [self] + list(others)
+        n = len(self)
+        cols, dtypes, cats = {}, {}, {}
+        for f in frames:
+            if len(f) != n or len(f._partitions) != 1:
+                raise lib.HfError("hconcat: frames must be aligned "
+                                  "single-partition results")
+            block = f._partitions[0].block()
+            for name, c in block.columns.items():
+                if name in cols:
+                    raise lib.HfError(f"hconcat: duplicate column {name!r}")
+                cols[name] = c
+                dtypes[name] = f.dtypes[name]
+                if name in block.cats:
+                    cats[name] = block.cats[name]
+        part = HipDataframePartition(DeviceBlock(cols, n, cats))
+        return HipDataframe([part], self._index, list(cols), [n],
+                            pandas.Series(dtypes))
+
     # ---- broadcast inner join (MergeImpl.row_axis_merge device form,
     #      merge.py:104-178: combine() the right frame once, probe per left
     #      partition; pandas suffix rules "_x"/"_y" on collisions) ----

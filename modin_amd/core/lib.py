@@ -45,7 +45,7 @@ HF_FLOAT64 = 1
 
 # map ops
 MAP_ADD, MAP_SUB, MAP_RSUB, MAP_MUL, MAP_DIV, MAP_RDIV, MAP_FILLNA, MAP_ABS, \
-    MAP_NEG, MAP_CAST_F64, MAP_CAST_I64 = range(11)
+    MAP_NEG, MAP_CAST_F64, MAP_CAST_I64, MAP_SQRT = range(12)
 # binary ops
 BIN_ADD, BIN_SUB, BIN_MUL, BIN_DIV = range(4)
 

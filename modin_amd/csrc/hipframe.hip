@@ -248,6 +248,7 @@ __device__ __forceinline__ double map1_f64(double x, double s) {
     case HF_MAP_FILLNA: return (x != x) ? s : x;
     case HF_MAP_ABS:    return fabs(x);
     case HF_MAP_NEG:    return -x;
+    case HF_MAP_SQRT:   return sqrt(x);
   }
   return x;
 }
@@ -1886,6 +1887,7 @@ int hf_map_scalar(int op, const hf_col* in, double scalar, hf_col** out) {
     case HF_MAP_FILLNA: rc = launch_map_f64<HF_MAP_FILLNA>(in, scalar, *out); break;
     case HF_MAP_ABS:    rc = launch_map_f64<HF_MAP_ABS>(in, scalar, *out); break;
     case HF_MAP_NEG:    rc = launch_map_f64<HF_MAP_NEG>(in, scalar, *out); break;
+    case HF_MAP_SQRT:   rc = launch_map_f64<HF_MAP_SQRT>(in, scalar, *out); break;
     default:
       hf_col_free(*out);
       *out = nullptr;

@@ -68,6 +68,12 @@ class _HipPandasBase:
     def max(self, **kwargs):
         return self._lower(self._query_compiler.max(**kwargs))
 
+    def var(self, ddof: int = 1):
+        return self._lower(self._query_compiler.var(ddof=ddof))
+
+    def std(self, ddof: int = 1):
+        return self._lower(self._query_compiler.std(ddof=ddof))
+
     # ---- arithmetic ----
     def __add__(self, other):
         return self._rewrap(type(self._query_compiler).add(self._query_compiler,
@@ -268,7 +274,8 @@ class DataFrame(_HipPandasBase):
         def relabel(block):
             return DeviceBlock(
                 {columns.get(n, n): c for n, c in block.columns.items()},
-                block.length)
+                block.length,
+                {columns.get(n, n): c for n, c in block.cats.items()})
         parts = [p.add_to_apply_calls(relabel) for p in frame._partitions]
         import pandas as _pd
         nf = HipDataframe(parts, frame._index, new_cols, frame._row_lengths,
@@ -383,7 +390,51 @@ class DataFrameGroupBy:
     def max(self):
         return self._agg("max")
 
+    def var(self, ddof: int = 1):
+        return DataFrame(query_compiler=self._df._query_compiler.groupby_var(
+            self._by, ddof=ddof))
+
+    def std(self, ddof: int = 1):
+        return DataFrame(query_compiler=self._df._query_compiler.groupby_std(
+            self._by, ddof=ddof))
+
+    def size(self):
+        """pandas DataFrameGroupBy.size(): a Series of group row counts
+        (NaN values included, NaN keys dropped)."""
+        out = DataFrame(
+            query_compiler=self._df._query_compiler.groupby_size(self._by)
+        ).to_pandas()
+        return out["size"].rename(None)
+
+    _AGGS = ("sum", "count", "mean", "min", "max", "var", "std")
+
     def agg(self, how):
+        """str, list-of-str (MultiIndex columns, pandas col-major order) or
+        dict {column: agg} — composed from the single-agg kernels and a
+        device-side horizontal concat (no data copies; columns re-label
+        lazily)."""
         if isinstance(how, str):
             return self._agg(how)
-        raise lib.HfError("groupby.agg accepts a single agg name this round")
+        if isinstance(how, dict):
+            qcs = []
+            for col, a in how.items():
+                if not isinstance(a, str):
+                    raise lib.HfError(
+                        "groupby.agg dict values must be single agg names "
+                        "this round")
+                sub = self._df[[self._by, col]]
+                qcs.append(sub._query_compiler.groupby_agg(self._by, a))
+            return DataFrame(query_compiler=qcs[0].hconcat(qcs[1:]))
+        if isinstance(how, (list, tuple)):
+            val_cols = [c for c in self._df.columns if c != self._by]
+            qcs = []
+            for col in val_cols:  # pandas order: per column, per agg
+                for a in how:
+                    if not isinstance(a, str):
+                        raise lib.HfError("groupby.agg list entries must "
+                                          "be agg names")
+                    sub = self._df[[self._by, col]]
+                    qc = sub._query_compiler.groupby_agg(self._by, a)
+                    qcs.append(qc.rename_columns({col: (col, a)}))
+            return DataFrame(query_compiler=qcs[0].hconcat(qcs[1:]))
+        raise lib.HfError("groupby.agg accepts str / list / dict")

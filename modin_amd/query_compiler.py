@@ -79,6 +79,74 @@ class HipQueryCompiler:
     groupby_min = GroupByReduce.register("min")
     groupby_max = GroupByReduce.register("max")
 
+    def var(self, ddof: int = 1):
+        return self._moment(ddof, sqrt_=False)
+
+    def std(self, ddof: int = 1):
+        return self._moment(ddof, sqrt_=True)
+
+    def _moment(self, ddof, sqrt_):
+        """Column var/std: Σx and Σx² partials from two tree-reduce passes
+        (the squared pass reuses the same single-scan reduce kernel),
+        composed on host — pandas nanvar, ddof default 1."""
+        import math
+        frame = self._modin_frame
+        p1 = frame.tree_reduce(frame.columns)
+
+        def square(block):
+            from modin_amd.core.partition import DeviceBlock
+            out = {}
+            for name, col in block.columns.items():
+                f = lib.cast_f64(col)
+                out[name] = lib.binary(lib.BIN_MUL, f, f)
+            return DeviceBlock(out, block.length)
+
+        sq = frame.map(square)
+        p2 = sq.tree_reduce(frame.columns)
+        vals = []
+        for name in frame.columns:
+            n = p1[name]["count"]
+            if n - ddof <= 0:
+                vals.append(float("nan"))
+                continue
+            v = (p2[name]["sum"] - p1[name]["sum"] ** 2 / n) / (n - ddof)
+            v = max(v, 0.0)
+            vals.append(math.sqrt(v) if sqrt_ else v)
+        return pandas.Series(vals, index=pandas.Index(list(frame.columns)),
+                             dtype=np.float64)
+
+    def groupby_var(self, by: str, ddof: int = 1) -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.groupby_var(by, ddof, sqrt=False))
+
+    def groupby_std(self, by: str, ddof: int = 1) -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.groupby_var(by, ddof, sqrt=True))
+
+    def groupby_size(self, by: str) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.groupby_size(by))
+
+    def hconcat(self, others: list) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.hconcat(
+            [o._modin_frame for o in others]))
+
+    def rename_columns(self, mapping: dict) -> "HipQueryCompiler":
+        from modin_amd.core.dataframe import HipDataframe
+        from modin_amd.core.partition import DeviceBlock
+        frame = self._modin_frame
+        new_cols = [mapping.get(c, c) for c in frame.columns]
+
+        def relabel(block):
+            return DeviceBlock(
+                {mapping.get(n, n): c for n, c in block.columns.items()},
+                block.length,
+                {mapping.get(n, n): c for n, c in block.cats.items()})
+        parts = [p.add_to_apply_calls(relabel) for p in frame._partitions]
+        dts = pandas.Series({mapping.get(n, n): d
+                             for n, d in frame.dtypes.items()})
+        return self.__constructor__(HipDataframe(
+            parts, frame._index, new_cols, frame._row_lengths, dts))
+
     def groupby_agg(self, by: str, agg: str) -> "HipQueryCompiler":
         fn = {
             "sum": type(self).groupby_sum,
@@ -86,6 +154,8 @@ class HipQueryCompiler:
             "mean": type(self).groupby_mean,
             "min": type(self).groupby_min,
             "max": type(self).groupby_max,
+            "var": type(self).groupby_var,
+            "std": type(self).groupby_std,
         }.get(agg)
         if fn is None:
             raise lib.HfError(

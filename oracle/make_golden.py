@@ -531,6 +531,72 @@ def gen_sort2_cases(mpd, rng):
     return cases
 
 
+def gen_var_cases(mpd, rng):
+    """var/std (frame + groupby, ddof 0/1), groupby size, and the agg
+    list/dict forms — all through the REAL reference, checked vs pandas."""
+    import pandas
+    cases = {}
+    n = 4000
+    k = rng.integers(0, 50, n).astype(np.int64)
+    v = rng.random(n) * 10
+    v[rng.random(n) < 0.1] = np.nan
+    v[k == 7] = np.nan          # an all-NaN group -> var NaN
+    k[0] = 999                   # a single-row group -> var NaN (ddof=1)
+    w = rng.standard_normal(n)
+    mdf = mpd.DataFrame({"k": k, "v": v, "w": w})
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w})
+    arrays = {"in_k": k, "in_v": v, "in_w": w}
+    for ddof in (0, 1):
+        for name, mfn, pfn in [
+                (f"var{ddof}", mdf.groupby("k").var, pdf.groupby("k").var),
+                (f"std{ddof}", mdf.groupby("k").std, pdf.groupby("k").std)]:
+            mres = mfn(ddof=ddof)._to_pandas()
+            pres = pfn(ddof=ddof)
+            assert list(mres.index) == list(pres.index)
+            np.testing.assert_allclose(mres.values, pres.values,
+                                       rtol=1e-9, atol=1e-12, equal_nan=True)
+            arrays[f"out_{name}_keys"] = pres.index.to_numpy().astype(np.int64)
+            for cn in ("v", "w"):
+                arrays[f"out_{name}_{cn}"] = pres[cn].to_numpy()
+        # frame-level var/std
+        np.testing.assert_allclose(
+            mdf.var(ddof=ddof)._to_pandas().to_numpy()
+            if hasattr(mdf.var(ddof=ddof), "_to_pandas")
+            else np.asarray(mdf.var(ddof=ddof)),
+            pdf.var(ddof=ddof).to_numpy(), rtol=1e-9, equal_nan=True)
+        arrays[f"out_frame_var{ddof}"] = pdf.var(ddof=ddof).to_numpy()
+        arrays[f"out_frame_std{ddof}"] = pdf.std(ddof=ddof).to_numpy()
+    # size
+    msz = mdf.groupby("k").size()._to_pandas() \
+        if hasattr(mdf.groupby("k").size(), "_to_pandas") \
+        else mdf.groupby("k").size()
+    psz = pdf.groupby("k").size()
+    np.testing.assert_array_equal(np.asarray(msz), psz.to_numpy())
+    arrays["out_size_keys"] = psz.index.to_numpy().astype(np.int64)
+    arrays["out_size"] = psz.to_numpy().astype(np.int64)
+    cases["gbv_moments"] = arrays
+
+    # ---- agg forms ----
+    magg = mdf.groupby("k").agg(["sum", "mean"])._to_pandas()
+    pagg = pdf.groupby("k").agg(["sum", "mean"])
+    assert list(magg.columns) == list(pagg.columns)
+    np.testing.assert_allclose(magg.values, pagg.values, rtol=1e-12,
+                               atol=1e-12, equal_nan=True)
+    mdd = mdf.groupby("k").agg({"v": "sum", "w": "max"})._to_pandas()
+    pdd = pdf.groupby("k").agg({"v": "sum", "w": "max"})
+    assert list(mdd.columns) == list(pdd.columns)
+    cases["gba_forms"] = {
+        "in_k": k, "in_v": v, "in_w": w,
+        "out_list_keys": pagg.index.to_numpy().astype(np.int64),
+        "out_list_cols": np.array([f"{a}|{b}" for a, b in pagg.columns]),
+        "out_list_vals": pagg.to_numpy(),
+        "out_dict_keys": pdd.index.to_numpy().astype(np.int64),
+        "out_dict_cols": np.array(list(pdd.columns)),
+        "out_dict_vals": pdd.to_numpy(),
+    }
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -545,6 +611,7 @@ def main():
     all_cases.update(gen_hash_groupby_cases(mpd, rng))
     all_cases.update(gen_string_cases(mpd, rng))
     all_cases.update(gen_sort2_cases(mpd, rng))
+    all_cases.update(gen_var_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)

@@ -498,6 +498,55 @@ def test_int_groupby_dtype_preserved_vs_golden(npartitions):
         big.groupby("k").sum().to_pandas()
 
 
+def test_var_std_size_vs_golden(npartitions):
+    """frame/groupby var+std (ddof 0 and 1; single-row and all-NaN groups
+    -> NaN) and groupby size, vs the reference."""
+    g = load_golden("gbv_moments")
+    df = mpd.DataFrame({"k": g["in_k"], "v": g["in_v"], "w": g["in_w"]})
+    gb = df.groupby("k")
+    for ddof in (0, 1):
+        for name, res in [(f"var{ddof}", gb.var(ddof=ddof)),
+                          (f"std{ddof}", gb.std(ddof=ddof))]:
+            out = res.to_pandas()
+            np.testing.assert_array_equal(out.index.to_numpy(),
+                                          g[f"out_{name}_keys"])
+            for cn in ("v", "w"):
+                np.testing.assert_allclose(
+                    out[cn].to_numpy(), g[f"out_{name}_{cn}"],
+                    rtol=1e-9, atol=1e-12, equal_nan=True,
+                    err_msg=f"{name}/{cn}")
+        # frame-level (golden order is [k, v, w]; we take v, w)
+        fv = df[["v", "w"]].var(ddof=ddof)
+        fs = df[["v", "w"]].std(ddof=ddof)
+        np.testing.assert_allclose(np.asarray(fv),
+                                   g[f"out_frame_var{ddof}"][1:],
+                                   rtol=1e-9, equal_nan=True)
+        np.testing.assert_allclose(np.asarray(fs),
+                                   g[f"out_frame_std{ddof}"][1:],
+                                   rtol=1e-9, equal_nan=True)
+    sz = df.groupby("k").size()
+    np.testing.assert_array_equal(np.asarray(sz.index), g["out_size_keys"])
+    np.testing.assert_array_equal(np.asarray(sz), g["out_size"])
+
+
+def test_agg_list_dict_forms_vs_golden(npartitions):
+    """groupby.agg(['sum','mean']) -> MultiIndex columns in pandas
+    col-major order; agg({'v':'sum','w':'max'}) -> per-column aggs."""
+    g = load_golden("gba_forms")
+    df = mpd.DataFrame({"k": g["in_k"], "v": g["in_v"], "w": g["in_w"]})
+    out = df.groupby("k").agg(["sum", "mean"]).to_pandas()
+    np.testing.assert_array_equal(out.index.to_numpy(), g["out_list_keys"])
+    got_cols = [f"{c}|{a}" for c, a in out.columns]
+    assert got_cols == list(g["out_list_cols"]), got_cols
+    np.testing.assert_allclose(out.to_numpy(), g["out_list_vals"],
+                               rtol=RTOL, atol=1e-9, equal_nan=True)
+    out2 = df.groupby("k").agg({"v": "sum", "w": "max"}).to_pandas()
+    np.testing.assert_array_equal(out2.index.to_numpy(), g["out_dict_keys"])
+    assert list(out2.columns) == list(g["out_dict_cols"])
+    np.testing.assert_allclose(out2.to_numpy(), g["out_dict_vals"],
+                               rtol=RTOL, atol=1e-9, equal_nan=True)
+
+
 def test_pipeline_filter_merge_groupby_sort(npartitions):
     """Integration chain: filter -> merge -> groupby -> sort, checked against
     the same chain on pandas."""
