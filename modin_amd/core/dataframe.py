@@ -214,8 +214,126 @@ class HipDataframe:
             partials = allreduce_partials(partials, list(col_names))
         return partials
 
+    KEYCOL = "\x00key\x00"  # internal combined multi-key column
+
+    def _combined_key_frame(self, by_list):
+        """Multi-key groupby support: fold the key columns into ONE int64
+        key (k1*span2*span3… + k2*span3… + …, mins subtracted — GLOBAL
+        mins at world>1 so every rank combines identically), giving the
+        existing single-key router (dense/radix/hash/sorted, shuffle) the
+        whole multi-key case.  Rows with a NaN string key are dropped
+        first (pandas dropna=True).  Returns (frame_with_KEYCOL,
+        decode(keys_np) -> pandas.MultiIndex).
+
+        The reference reaches the same point through pandas tuple keys
+        (algebra/groupby.py by-list handling); combined ascending order ==
+        pandas lexicographic order because the fold is monotone."""
+        from ..distributed import is_active
+        from .. import distributed as dist_mod
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        for b in by_list:
+            if b not in self.columns:
+                raise lib.HfError(f"groupby: key column {b!r} missing")
+        parts = self._partitions
+        dict_keys = [b for b in by_list if b in blk_cats]
+        if dict_keys:
+            need = False
+            for p in parts:
+                for b in dict_keys:
+                    c = p.block().columns[b]
+                    if c.length and lib.reduce(c).imn < 0:
+                        need = True
+            if need:
+                fparts = []
+                for p in parts:
+                    block = p.block()
+                    acc = None
+                    for b in dict_keys:
+                        m = lib.compare_scalar(lib.CMP_GE,
+                                               block.columns[b], 0.0)
+                        acc = m if acc is None else lib.binary(
+                            lib.BIN_MUL, acc, m)
+                    plan = lib.filter_plan(acc)
+                    cols = {m2: lib.filter_apply(plan, c)
+                            for m2, c in block.columns.items()}
+                    fparts.append(HipDataframePartition(
+                        DeviceBlock(cols, plan.n_kept, block.cats)))
+                parts = fparts
+        mins, spans = [], []
+        for b in by_list:
+            kmin = kmax = None
+            for p in parts:
+                c = p.block().columns[b]
+                if c.dtype_code != lib.HF_INT64:
+                    raise lib.HfError(
+                        f"groupby: key column {b!r} must be int64 or "
+                        "string")
+                if c.length:
+                    r = lib.reduce(c)
+                    kmin = r.imn if kmin is None else min(kmin, r.imn)
+                    kmax = r.imx if kmax is None else max(kmax, r.imx)
+            if is_active():
+                kmin, kmax = dist_mod.allreduce_minmax(kmin, kmax)
+            if kmin is None:
+                kmin, kmax = 0, 0
+            mins.append(kmin)
+            spans.append(kmax - kmin + 1)
+        total = 1
+        for sp in spans:
+            total *= sp
+            if total > (1 << 62):
+                raise lib.HfError(
+                    "groupby: combined key range of "
+                    f"{by_list} exceeds 2^62 (tuple-hash keys are a "
+                    "later round)")
+        strides = [1] * len(by_list)
+        for i in range(len(by_list) - 2, -1, -1):
+            strides[i] = strides[i + 1] * spans[i + 1]
+        new_parts = []
+        for p in parts:
+            block = p.block()
+            comb = None
+            for b, mn, st in zip(by_list, mins, strides):
+                t = lib.map_scalar(lib.MAP_SUB, block.columns[b], mn)
+                if st != 1:
+                    t = lib.map_scalar(lib.MAP_MUL, t, st)
+                comb = t if comb is None else lib.binary(lib.BIN_ADD,
+                                                         comb, t)
+            cols = dict(block.columns)
+            cols[self.KEYCOL] = comb
+            new_parts.append(HipDataframePartition(
+                DeviceBlock(cols, block.length, block.cats)))
+        columns = list(self.columns) + [self.KEYCOL]
+        dtypes = pandas.concat([self.dtypes, pandas.Series(
+            {self.KEYCOL: np.dtype(np.int64)})])
+        frame = HipDataframe(new_parts, self._index, columns,
+                             [p.block().length for p in new_parts]
+                             if new_parts else [0], dtypes)
+
+        def decode(keys_np):
+            from .partition import decode_dict
+            levels = []
+            for b, mn, st, sp in zip(by_list, mins, strides, spans):
+                lv = (keys_np // st) % sp + mn
+                if b in blk_cats:
+                    lv = decode_dict(lv, blk_cats[b])
+                levels.append(lv)
+            return pandas.MultiIndex.from_arrays(levels, names=by_list)
+
+        return frame, decode
+
     # ---- GroupByReduce (dataframe.py:4530) ----
-    def groupby_reduce(self, by: str, agg: str) -> "HipDataframe":
+    def groupby_reduce(self, by, agg: str) -> "HipDataframe":
+        if isinstance(by, (list, tuple)):
+            if len(by) == 1:
+                by = by[0]
+            else:
+                cf, decode = self._combined_key_frame(list(by))
+                keep = [c for c in cf.columns if c not in by]
+                res = cf.take_columns(keep).groupby_reduce(self.KEYCOL, agg)
+                res._index = decode(lib.get(res._index.col))
+                return res
         val_names = [c for c in self.columns if c != by]
         want_counts = agg in ("count", "mean", "min", "max")
         agg_op = lib.AGG_OP_OF[agg]
@@ -304,6 +422,16 @@ class HipDataframe:
         device as (Σx² − (Σx)²/n)/(n−ddof), clamped at 0 (cancellation)
         and fixed to NaN where n <= ddof (pandas nanvar shape, reference
         groupby var via GroupbyReduceImpl-style pairs)."""
+        if isinstance(by, (list, tuple)):
+            if len(by) == 1:
+                by = by[0]
+            else:
+                cf, decode = self._combined_key_frame(list(by))
+                keep = [c for c in cf.columns if c not in by]
+                res = cf.take_columns(keep).groupby_var(self.KEYCOL, ddof,
+                                                        sqrt)
+                res._index = decode(lib.get(res._index.col))
+                return res
         val_names = [c for c in self.columns if c != by]
         blk_cats = (self._partitions[0].block().cats
                     if self._partitions else {})
@@ -377,9 +505,18 @@ class HipDataframe:
         k = len(val_names)
         return keys, key_cats, sums[:k], sums[k:], counts[:k], n
 
-    def groupby_size(self, by: str) -> "HipDataframe":
+    def groupby_size(self, by) -> "HipDataframe":
         """groupby().size(): group row counts INCLUDING NaN values (and
         NaN keys dropped) — a ones column through the sum path."""
+        if isinstance(by, (list, tuple)):
+            if len(by) == 1:
+                by = by[0]
+            else:
+                cf, decode = self._combined_key_frame(list(by))
+                res = cf.take_columns([self.KEYCOL]).groupby_size(
+                    self.KEYCOL)
+                res._index = decode(lib.get(res._index.col))
+                return res
         blk_cats = (self._partitions[0].block().cats
                     if self._partitions else {})
 

@@ -308,10 +308,14 @@ class DataFrame(_HipPandasBase):
         return DataFrame(query_compiler=self._query_compiler.merge(
             other._query_compiler, on=on, how=how))
 
-    def groupby(self, by: str) -> "DataFrameGroupBy":
-        if not isinstance(by, str) or by not in list(self.columns):
-            raise lib.HfError("groupby(by=<column name>) only")
-        return DataFrameGroupBy(self, by)
+    def groupby(self, by) -> "DataFrameGroupBy":
+        bys = list(by) if isinstance(by, (list, tuple)) else [by]
+        for b in bys:
+            if not isinstance(b, str) or b not in list(self.columns):
+                raise lib.HfError(
+                    "groupby(by=<column name> | [column names]) only")
+        return DataFrameGroupBy(self, by if isinstance(by, (list, tuple))
+                                else by)
 
     def to_pandas(self) -> pandas.DataFrame:
         return self._query_compiler.to_pandas()
@@ -422,18 +426,22 @@ class DataFrameGroupBy:
                     raise lib.HfError(
                         "groupby.agg dict values must be single agg names "
                         "this round")
-                sub = self._df[[self._by, col]]
+                bys = (list(self._by) if isinstance(self._by, (list, tuple))
+                       else [self._by])
+                sub = self._df[[*bys, col]]
                 qcs.append(sub._query_compiler.groupby_agg(self._by, a))
             return DataFrame(query_compiler=qcs[0].hconcat(qcs[1:]))
         if isinstance(how, (list, tuple)):
-            val_cols = [c for c in self._df.columns if c != self._by]
+            bys = (list(self._by) if isinstance(self._by, (list, tuple))
+                   else [self._by])
+            val_cols = [c for c in self._df.columns if c not in bys]
             qcs = []
             for col in val_cols:  # pandas order: per column, per agg
                 for a in how:
                     if not isinstance(a, str):
                         raise lib.HfError("groupby.agg list entries must "
                                           "be agg names")
-                    sub = self._df[[self._by, col]]
+                    sub = self._df[[*bys, col]]
                     qc = sub._query_compiler.groupby_agg(self._by, a)
                     qcs.append(qc.rename_columns({col: (col, a)}))
             return DataFrame(query_compiler=qcs[0].hconcat(qcs[1:]))

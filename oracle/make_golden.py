@@ -597,6 +597,61 @@ def gen_var_cases(mpd, rng):
     return cases
 
 
+def gen_multikey_cases(mpd, rng):
+    """groupby by a KEY LIST (int+int and string+int), all reduce aggs +
+    size — the reference's by-list path vs pandas; our backend folds the
+    keys into one int64 (monotone, so index order matches pandas
+    lexicographic MultiIndex order)."""
+    import pandas
+    cases = {}
+    n = 4000
+    a = rng.integers(-20, 20, n).astype(np.int64)
+    b = rng.integers(0, 15, n).astype(np.int64)
+    v = rng.random(n)
+    v[rng.random(n) < 0.1] = np.nan
+    w = rng.standard_normal(n)
+    mdf = mpd.DataFrame({"a": a, "b": b, "v": v, "w": w})
+    pdf = pandas.DataFrame({"a": a, "b": b, "v": v, "w": w})
+    arrays = {"in_a": a, "in_b": b, "in_v": v, "in_w": w}
+    for agg in ("sum", "count", "mean", "min", "max"):
+        mres = getattr(mdf.groupby(["a", "b"]), agg)()._to_pandas()
+        pres = getattr(pdf.groupby(["a", "b"]), agg)()
+        assert list(mres.index) == list(pres.index)
+        np.testing.assert_allclose(mres.values.astype(float),
+                                   pres.values.astype(float), rtol=1e-12,
+                                   atol=1e-12, equal_nan=True)
+        arrays[f"out_{agg}_ka"] = pres.index.get_level_values(0).to_numpy()
+        arrays[f"out_{agg}_kb"] = pres.index.get_level_values(1).to_numpy()
+        for cn in ("v", "w"):
+            arrays[f"out_{agg}_{cn}"] = pres[cn].to_numpy()
+    psz = pdf.groupby(["a", "b"]).size()
+    msz = mdf.groupby(["a", "b"]).size()
+    np.testing.assert_array_equal(np.asarray(msz._to_pandas()
+                                             if hasattr(msz, "_to_pandas")
+                                             else msz), psz.to_numpy())
+    arrays["out_size"] = psz.to_numpy().astype(np.int64)
+    cases["gbm_ints"] = arrays
+
+    ns = 3000
+    pool = np.array(["north", "south", "east", "west"])
+    sarr = rng.choice(pool, ns).astype(object)
+    sarr[rng.random(ns) < 0.05] = np.nan  # NaN key rows drop entirely
+    g2 = rng.integers(0, 8, ns).astype(np.int64)
+    v2 = rng.random(ns)
+    mdf = mpd.DataFrame({"s": sarr, "g": g2, "v": v2})
+    pdf = pandas.DataFrame({"s": sarr, "g": g2, "v": v2})
+    arr2 = {"in_s": _enc_str(sarr), "in_g": g2, "in_v": v2}
+    for agg in ("sum", "mean"):
+        mres = getattr(mdf.groupby(["s", "g"]), agg)()._to_pandas()
+        pres = getattr(pdf.groupby(["s", "g"]), agg)()
+        assert list(mres.index) == list(pres.index)
+        arr2[f"out_{agg}_ks"] = _enc_str(pres.index.get_level_values(0))
+        arr2[f"out_{agg}_kg"] = pres.index.get_level_values(1).to_numpy()
+        arr2[f"out_{agg}_v"] = pres["v"].to_numpy()
+    cases["gbm_strint"] = arr2
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -612,6 +667,7 @@ def main():
     all_cases.update(gen_string_cases(mpd, rng))
     all_cases.update(gen_sort2_cases(mpd, rng))
     all_cases.update(gen_var_cases(mpd, rng))
+    all_cases.update(gen_multikey_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)

@@ -547,6 +547,50 @@ def test_agg_list_dict_forms_vs_golden(npartitions):
                                rtol=RTOL, atol=1e-9, equal_nan=True)
 
 
+def test_multikey_groupby_vs_golden(npartitions):
+    """groupby(by=[a, b]): the combined-key fold through the single-key
+    router; MultiIndex result in pandas lexicographic order, size
+    included."""
+    g = load_golden("gbm_ints")
+    df = mpd.DataFrame({"a": g["in_a"], "b": g["in_b"], "v": g["in_v"],
+                        "w": g["in_w"]})
+    for agg in ("sum", "count", "mean", "min", "max"):
+        out = getattr(df.groupby(["a", "b"]), agg)().to_pandas()
+        np.testing.assert_array_equal(
+            out.index.get_level_values(0).to_numpy(), g[f"out_{agg}_ka"],
+            err_msg=f"{agg} ka")
+        np.testing.assert_array_equal(
+            out.index.get_level_values(1).to_numpy(), g[f"out_{agg}_kb"],
+            err_msg=f"{agg} kb")
+        assert out.index.names == ["a", "b"]
+        for cn in ("v", "w"):
+            expect = g[f"out_{agg}_{cn}"]
+            np.testing.assert_allclose(out[cn].to_numpy(), expect,
+                                       rtol=RTOL, atol=1e-9,
+                                       equal_nan=True,
+                                       err_msg=f"{agg}/{cn}")
+    sz = df.groupby(["a", "b"]).size()
+    np.testing.assert_array_equal(np.asarray(sz), g["out_size"])
+
+
+def test_multikey_groupby_string_int_vs_golden(npartitions):
+    """groupby(by=[string, int]): dictionary codes fold with the int key;
+    NaN string keys drop the whole row (pandas dropna)."""
+    from tests.test_gpu_strings import assert_str_equal, dec
+    g = load_golden("gbm_strint")
+    df = mpd.DataFrame({"s": dec(g["in_s"]), "g": g["in_g"],
+                        "v": g["in_v"]})
+    for agg in ("sum", "mean"):
+        out = getattr(df.groupby(["s", "g"]), agg)().to_pandas()
+        assert_str_equal(out.index.get_level_values(0).to_numpy(),
+                         g[f"out_{agg}_ks"], f"{agg} ks")
+        np.testing.assert_array_equal(
+            out.index.get_level_values(1).to_numpy(), g[f"out_{agg}_kg"])
+        np.testing.assert_allclose(out["v"].to_numpy(),
+                                   g[f"out_{agg}_v"], rtol=RTOL,
+                                   atol=1e-9, equal_nan=True)
+
+
 def test_pipeline_filter_merge_groupby_sort(npartitions):
     """Integration chain: filter -> merge -> groupby -> sort, checked against
     the same chain on pandas."""
