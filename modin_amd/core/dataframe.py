@@ -546,6 +546,69 @@ class HipDataframe:
                             res._row_lengths,
                             pandas.Series({"size": np.dtype(np.int64)}))
 
+    def distinct_stats(self, name: str):
+        """Distinct values of one int64/dict column with first-appearance
+        positions and group sizes — the engine under Series.unique /
+        value_counts / nunique (reference: qc unique/value_counts).  One
+        groupby-min pass over the global row position plus one size pass;
+        NaN (dict code −1) tracked separately (pandas: unique keeps NaN,
+        value_counts/nunique drop it)."""
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        POS = "\x00pos\x00"
+        parts2 = []
+        base = 0
+        nan_first = None
+        nan_count = 0
+        for p, ln in zip(self._partitions, self._row_lengths):
+            block = p.block()
+            col = block.columns[name]
+            if col.dtype_code != lib.HF_INT64:
+                raise lib.HfError(
+                    f"unique/value_counts: column {name!r} must be int64 "
+                    "or string")
+            ones = lib.compare_scalar(lib.CMP_NOTNA, col, 0.0)
+            plan = lib.filter_plan(ones)
+            pos = lib.filter_iota(plan, base)
+            cols = {name: col, POS: pos}
+            if name in blk_cats and col.length and \
+                    lib.reduce(col).imn < 0:
+                nm = lib.compare_scalar(lib.CMP_EQ, col, -1.0)
+                nplan = lib.filter_plan(nm)
+                nan_count += nplan.n_kept
+                if nplan.n_kept:
+                    npos = lib.filter_iota(nplan, base)
+                    first = lib.reduce(npos).imn
+                    nan_first = first if nan_first is None \
+                        else min(nan_first, first)
+                # drop the NaN rows from the grouped pass
+                keepm = lib.compare_scalar(lib.CMP_GE, col, 0.0)
+                kplan = lib.filter_plan(keepm)
+                cols = {name: lib.filter_apply(kplan, col),
+                        POS: lib.filter_apply(kplan, pos)}
+            parts2.append(HipDataframePartition(
+                DeviceBlock(cols, cols[name].length, block.cats)))
+            base += ln
+        frame2 = HipDataframe(
+            parts2, pandas.RangeIndex(base), [name, POS],
+            [pp.block().length for pp in parts2] if parts2 else [0],
+            pandas.Series({name: self.dtypes[name],
+                           POS: np.dtype(np.int64)}))
+        fmin = frame2.groupby_reduce(name, "min")
+        fsize = frame2.take_columns([name]).groupby_size(name)
+        keys = lib.get(fmin._index.col)
+        firstpos = lib.get(
+            fmin._partitions[0].block().columns[POS]).astype(np.int64)
+        counts = lib.get(fsize._partitions[0].block().columns["size"])
+        if name in blk_cats:
+            from .partition import decode_dict
+            values = decode_dict(keys, blk_cats[name])
+        else:
+            values = keys
+        return {"values": values, "counts": counts, "firstpos": firstpos,
+                "nan_count": int(nan_count),
+                "nan_firstpos": nan_first}
+
     def hconcat(self, others: list) -> "HipDataframe":
         """Horizontal compose of single-partition frames with identical row
         count and index (the device form of the reference's axis=1 concat

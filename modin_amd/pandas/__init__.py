@@ -352,6 +352,62 @@ class Series(_HipPandasBase):
     def __len__(self):
         return len(self._query_compiler)
 
+    def unique(self):
+        """pandas Series.unique: distinct values in FIRST-APPEARANCE order,
+        NaN included at its appearance position (device groupby-min over
+        global row positions)."""
+        st = self._query_compiler.distinct_stats()
+        order = np.argsort(st["firstpos"], kind="stable")
+        vals = list(np.asarray(st["values"], dtype=object)[order])
+        pos = list(st["firstpos"][order])
+        if st["nan_count"]:
+            i = int(np.searchsorted(np.asarray(pos), st["nan_firstpos"]))
+            vals.insert(i, np.nan)
+        if all(isinstance(v, (int, np.integer)) for v in vals):
+            return np.array(vals, dtype=np.int64)
+        return np.array(vals, dtype=object)
+
+    def value_counts(self):
+        """pandas Series.value_counts: counts desc, ties in appearance
+        order, NaN dropped (default dropna=True)."""
+        st = self._query_compiler.distinct_stats()
+        order = np.lexsort((st["firstpos"], -st["counts"]))
+        idx = pandas.Index(np.asarray(st["values"], dtype=object)[order],
+                           name=self.name)
+        return pandas.Series(st["counts"][order].astype(np.int64),
+                             index=idx, name="count")
+
+    def nunique(self) -> int:
+        return int(len(self._query_compiler.distinct_stats()["values"]))
+
+    def isin(self, values) -> "Series":
+        """Membership mask composed from EQ compares, OR-folded as
+        a+b-a*b over the 0/1 masks (<= 64 values; string Series translate
+        through the dictionary; NaN values rejected loudly)."""
+        values = list(values)
+        if len(values) > 64:
+            raise lib.HfError("isin supports up to 64 values this round")
+        for v in values:
+            if isinstance(v, float) and v != v:
+                raise lib.HfError(
+                    "isin with NaN in the value list is a later round")
+        qc = self._query_compiler
+        if not values:
+            acc = qc.eq(float("inf"))  # all False (NaN == inf is False too)
+        else:
+            acc = None
+            for v in values:
+                m = qc.eq(v)
+                if acc is None:
+                    acc = m
+                else:  # OR of 0/1 masks: a + b - a*b
+                    t = type(acc).add(acc, m)
+                    u = type(acc).mul(acc, m)
+                    acc = type(t).sub(t, u)
+        out = Series(query_compiler=acc, name=self.name)
+        out._bool_mask = True
+        return out
+
     def to_pandas(self) -> pandas.Series:
         df = self._query_compiler.to_pandas()
         s = df[df.columns[0]]

@@ -147,6 +147,9 @@ class HipQueryCompiler:
         return self.__constructor__(HipDataframe(
             parts, frame._index, new_cols, frame._row_lengths, dts))
 
+    def distinct_stats(self):
+        return self._modin_frame.distinct_stats(self._modin_frame.columns[0])
+
     def groupby_agg(self, by: str, agg: str) -> "HipQueryCompiler":
         fn = {
             "sum": type(self).groupby_sum,

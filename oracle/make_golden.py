@@ -652,6 +652,55 @@ def gen_multikey_cases(mpd, rng):
     return cases
 
 
+def gen_series_cases(mpd, rng):
+    """Series.unique / value_counts / nunique / isin vs the reference:
+    appearance order, count-desc ties, NaN handling, string dictionaries."""
+    import pandas
+    cases = {}
+    n = 5000
+    k = rng.integers(-30, 30, n).astype(np.int64)
+    pool = np.array(["ash", "oak", "elm", "fir", "Yew"])
+    sv = rng.choice(pool, n).astype(object)
+    sv[rng.random(n) < 0.06] = np.nan
+    mdf = mpd.DataFrame({"k": k, "s": sv})
+    pdf = pandas.DataFrame({"k": k, "s": sv})
+    arrays = {"in_k": k, "in_s": _enc_str(sv)}
+    # int column
+    mu = mdf["k"].unique()
+    pu = pdf["k"].unique()
+    np.testing.assert_array_equal(np.asarray(mu), pu)
+    arrays["out_k_unique"] = pu.astype(np.int64)
+    mvc = mdf["k"].value_counts()
+    pvc = pdf["k"].value_counts()
+    np.testing.assert_array_equal(np.asarray(mvc._to_pandas()
+                                             if hasattr(mvc, "_to_pandas")
+                                             else mvc), pvc.to_numpy())
+    arrays["out_k_vc_idx"] = pvc.index.to_numpy().astype(np.int64)
+    arrays["out_k_vc"] = pvc.to_numpy().astype(np.int64)
+    assert int(mdf["k"].nunique()) == int(pdf["k"].nunique())
+    arrays["out_k_nunique"] = np.array([pdf["k"].nunique()], dtype=np.int64)
+    # string column (unique keeps NaN at appearance position)
+    pu_s = pdf["s"].unique()
+    arrays["out_s_unique"] = _enc_str(pu_s)
+    pvc_s = pdf["s"].value_counts()
+    arrays["out_s_vc_idx"] = _enc_str(pvc_s.index)
+    arrays["out_s_vc"] = pvc_s.to_numpy().astype(np.int64)
+    arrays["out_s_nunique"] = np.array([pdf["s"].nunique()],
+                                       dtype=np.int64)
+    mu_s = mdf["s"].unique()
+    assert len(mu_s) == len(pu_s)
+    # isin masks
+    pk = pdf["k"].isin([3, -7, 999]).to_numpy()
+    ps = pdf["s"].isin(["oak", "Yew", "missing"]).to_numpy()
+    np.testing.assert_array_equal(
+        np.asarray(mdf["k"].isin([3, -7, 999])._to_pandas()), pk)
+    arrays["out_k_isin"] = pk.astype(np.int64)
+    arrays["out_s_isin"] = ps.astype(np.int64)
+    arrays["out_empty_isin"] = pdf["k"].isin([]).to_numpy().astype(np.int64)
+    cases["ser_utils"] = arrays
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -668,6 +717,7 @@ def main():
     all_cases.update(gen_sort2_cases(mpd, rng))
     all_cases.update(gen_var_cases(mpd, rng))
     all_cases.update(gen_multikey_cases(mpd, rng))
+    all_cases.update(gen_series_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)

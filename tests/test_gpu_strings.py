@@ -237,3 +237,31 @@ def test_read_parquet_vs_pandas(tmp_path, npartitions):
     assert list(g1.index) == list(g2.index)
     np.testing.assert_allclose(g1["v"].to_numpy(), g2["v"].to_numpy(),
                                rtol=RTOL, atol=1e-9)
+
+
+def test_series_utils_vs_golden(npartitions):
+    """Series.unique (appearance order, NaN kept), value_counts (count
+    desc, ties by appearance, NaN dropped), nunique, isin — int and string
+    columns vs the reference."""
+    g = load_golden("ser_utils")
+    df = mpd.DataFrame({"k": g["in_k"], "s": dec(g["in_s"])})
+    np.testing.assert_array_equal(df["k"].unique(), g["out_k_unique"])
+    vc = df["k"].value_counts()
+    np.testing.assert_array_equal(vc.index.to_numpy(), g["out_k_vc_idx"])
+    np.testing.assert_array_equal(vc.to_numpy(), g["out_k_vc"])
+    assert df["k"].nunique() == int(g["out_k_nunique"][0])
+    su = df["s"].unique()
+    assert_str_equal(su, g["out_s_unique"], "s unique")
+    svc = df["s"].value_counts()
+    assert_str_equal(svc.index.to_numpy(), g["out_s_vc_idx"], "s vc idx")
+    np.testing.assert_array_equal(svc.to_numpy(), g["out_s_vc"])
+    assert df["s"].nunique() == int(g["out_s_nunique"][0])
+    np.testing.assert_array_equal(
+        df["k"].isin([3, -7, 999]).to_pandas().to_numpy().astype(np.int64),
+        g["out_k_isin"])
+    np.testing.assert_array_equal(
+        df["s"].isin(["oak", "Yew", "missing"]).to_pandas().to_numpy()
+        .astype(np.int64), g["out_s_isin"])
+    np.testing.assert_array_equal(
+        df["k"].isin([]).to_pandas().to_numpy().astype(np.int64),
+        g["out_empty_isin"])
