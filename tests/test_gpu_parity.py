@@ -687,3 +687,38 @@ def test_native_extension_is_loaded():
     import modin_amd.core.lib as l
     maps = open("/proc/self/maps").read()
     assert "libhipframe.so" in maps
+
+
+def test_new_paths_empty_and_edge(npartitions):
+    """Empty/degenerate frames through the round-1 late additions:
+    multi-key groupby, var/std, size, unique/value_counts, agg forms."""
+    empty = mpd.DataFrame(pandas.DataFrame({
+        "a": np.array([], dtype=np.int64),
+        "b": np.array([], dtype=np.int64),
+        "v": np.array([], dtype=np.float64)}))
+    out = empty.groupby(["a", "b"]).sum().to_pandas()
+    assert len(out) == 0
+    assert len(empty.groupby("a").var().to_pandas()) == 0
+    assert empty.groupby("a").size().empty
+    assert len(mpd.DataFrame(pandas.DataFrame(
+        {"k": np.array([], dtype=np.int64)}))["k"].unique()) == 0
+    # single row: var ddof=1 -> NaN, ddof=0 -> 0
+    one = mpd.DataFrame(pandas.DataFrame({"k": [1], "v": [2.5]}))
+    v1 = one.groupby("k").var().to_pandas()
+    assert np.isnan(v1["v"].iloc[0])
+    v0 = one.groupby("k").var(ddof=0).to_pandas()
+    assert v0["v"].iloc[0] == 0.0
+    # multi-key with one key column degenerates to single-key
+    pdf = pandas.DataFrame({"a": [1, 1, 2], "v": [1.0, 2.0, 3.0]})
+    df = mpd.DataFrame(pdf)
+    out = df.groupby(["a"]).sum().to_pandas()
+    exp = pdf.groupby(["a"]).sum()
+    np.testing.assert_array_equal(out["v"].to_numpy(),
+                                  exp["v"].to_numpy())
+    # combined-range overflow is loud
+    big = mpd.DataFrame(pandas.DataFrame({
+        "a": np.array([0, 2**40], dtype=np.int64),
+        "b": np.array([0, 2**40], dtype=np.int64),
+        "v": np.array([1.0, 2.0])}))
+    with pytest.raises(lib.HfError, match="2\\^62"):
+        big.groupby(["a", "b"]).sum().to_pandas()
