@@ -14,6 +14,7 @@ Reductions return pandas.Series (the reference's API layer also lowers
 
 from __future__ import annotations
 
+import numpy as np
 import pandas
 
 from ..core import lib
@@ -308,14 +309,13 @@ class DataFrame(_HipPandasBase):
         return DataFrame(query_compiler=self._query_compiler.merge(
             other._query_compiler, on=on, how=how))
 
-    def groupby(self, by) -> "DataFrameGroupBy":
+    def groupby(self, by, as_index: bool = True) -> "DataFrameGroupBy":
         bys = list(by) if isinstance(by, (list, tuple)) else [by]
         for b in bys:
             if not isinstance(b, str) or b not in list(self.columns):
                 raise lib.HfError(
                     "groupby(by=<column name> | [column names]) only")
-        return DataFrameGroupBy(self, by if isinstance(by, (list, tuple))
-                                else by)
+        return DataFrameGroupBy(self, by, as_index=as_index)
 
     def to_pandas(self) -> pandas.DataFrame:
         return self._query_compiler.to_pandas()
@@ -425,15 +425,42 @@ class Series(_HipPandasBase):
 
 class DataFrameGroupBy:
     """Mirrors modin/pandas/groupby.py DataFrameGroupBy for the reduce aggs
-    (sum :1330, count, mean -> _wrap_aggregation -> qc.groupby_<agg>)."""
+    (sum :1330, count, mean -> _wrap_aggregation -> qc.groupby_<agg>);
+    ``as_index=False`` applies the reference's reduce fix-up
+    (algebra/groupby.py:278 — keys become leading columns over a fresh
+    RangeIndex); ``gb[col]`` / ``gb[[cols]]`` select aggregation columns
+    (SeriesGroupBy shape for a single name)."""
 
-    def __init__(self, df: DataFrame, by: str):
+    def __init__(self, df: DataFrame, by, as_index: bool = True,
+                 series_out: bool = False):
         self._df = df
         self._by = by
+        self._as_index = as_index
+        self._series_out = series_out
+
+    def __getitem__(self, key):
+        bys = list(self._by) if isinstance(self._by, (list, tuple)) \
+            else [self._by]
+        names = [key] if isinstance(key, str) else list(key)
+        for n2 in names:
+            if n2 not in list(self._df.columns):
+                raise lib.HfError(f"groupby selection: column {n2!r} "
+                                  "missing")
+        sub = self._df[[*bys, *names]]
+        return DataFrameGroupBy(sub, self._by, as_index=self._as_index,
+                                series_out=isinstance(key, str))
 
     def _agg(self, how: str) -> DataFrame:
         qc = self._df._query_compiler.groupby_agg(self._by, how)
-        return DataFrame(query_compiler=qc)
+        out = DataFrame(query_compiler=qc)
+        if self._series_out and self._as_index:
+            name = list(qc._modin_frame.columns)[0]
+            return Series(query_compiler=qc, name=name)
+        if not self._as_index:
+            # reference reduce fix-up (algebra/groupby.py:278): keys become
+            # leading columns over a fresh RangeIndex
+            return from_pandas(out.to_pandas().reset_index())
+        return out
 
     def sum(self):
         return self._agg("sum")
