@@ -295,6 +295,15 @@ int hf_sort_perm(const hf_col* keys, int ascending, hf_col** out_perm);
 int hf_shuffle_dest(const hf_col* keys, const int64_t* splitters, int nsplit,
                     hf_col** dest);
 
+/* Exact-match binary search: out[i] = j with sorted[j] == keys[i], else -1.
+ * Densifies unbounded int64 join keys through the sorted distinct right
+ * keys (lower_bound per row), so the dense-range CSR join
+ * (hf_join_build's 2^27 slot cap) covers ANY key span with <= 2^27
+ * DISTINCT build keys — the device form of the reference's hash-join key
+ * lookup (storage_formats/pandas/merge.py row_axis_merge). */
+int hf_search_sorted(const hf_col* keys, const hf_col* sorted_uniq,
+                     hf_col** out);
+
 /* Raw device-to-device copy on the hipframe stream — the interop bridge to
  * RCCL-visible torch buffers (exchange_splits): hf columns are copied into /
  * out of torch-allocated device tensors by address. */

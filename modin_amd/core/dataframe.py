@@ -696,21 +696,38 @@ class HipDataframe:
             kmin, n_slots = r.imn, r.imx - r.imn + 1
         else:
             kmin, n_slots = 0, 1
+        uniq = None  # set on the unbounded-span path (codes join)
+        if n_slots > (1 << 27):
+            # unbounded key span: densify through the sorted DISTINCT right
+            # keys (hf_search_sorted binary search per row; the CSR join
+            # then runs in code space [0, n_uniq)).  Covers any int64 span
+            # with <= 2^27 distinct right keys.
+            perm = lib.sort_perm(rkeys)
+            skeys = lib.gather(rkeys, perm)
+            uniq, _su, _cu, n_uniq = lib.groupby_sorted(
+                skeys, [], lib.AGG_SUM, False)
+            if n_uniq > (1 << 27):
+                raise lib.HfError(
+                    "merge: more than 2^27 distinct right keys (the "
+                    "co-shuffled giant-right merge is a later round)")
+            rkeys = lib.search_sorted(rkeys, uniq)
+            kmin, n_slots = 0, max(n_uniq, 1)
         # cache the build side on the right frame: its columns are immutable,
         # so repeated merges with the same right frame skip hist/scan/fill
         # (the lazy-metadata pattern again; a real broadcast join caches its
         # build side)
-        cache_key = (on, tuple(right_names), kmin, n_slots)
+        cache_key = (on, tuple(right_names), kmin, n_slots,
+                     uniq is not None)
         cached = getattr(other, "_join_build_cache", None)
         if key_cats is not None:
             # dictionary keys: the build lives in the LEFT frame's code
             # space, which varies per left frame — don't cache on the right
             j = lib.join_build(rkeys, rvals, kmin, n_slots)
         elif cached is not None and cached[0] == cache_key:
-            j = cached[1]
+            j, uniq = cached[1], cached[2]
         else:
             j = lib.join_build(rkeys, rvals, kmin, n_slots)
-            other._join_build_cache = (cache_key, j)
+            other._join_build_cache = (cache_key, j, uniq)
 
         out_parts, lengths = [], []
         for p in self._partitions:
@@ -718,7 +735,11 @@ class HipDataframe:
             lkeys = block.columns[on]
             if lkeys.dtype_code != lib.HF_INT64:
                 raise lib.HfError("merge: key column must be int64")
+            if uniq is not None:  # code space: unmatched lefts become -1
+                lkeys = lib.search_sorted(lkeys, uniq)
             keys_c, lidx, rcols, nout = lib.join_probe(j, lkeys)
+            if uniq is not None:  # decode output codes back to key values
+                keys_c = lib.gather(uniq, keys_c)
             cols = {}
             for name in self.columns:  # left column order, key in place
                 if name == on:

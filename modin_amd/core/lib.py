@@ -127,6 +127,8 @@ def load() -> ct.CDLL:
                                              ct.POINTER(ct.c_void_p),
                                              ct.POINTER(ct.c_void_p),
                                              ct.POINTER(ct.c_int64)]),
+            "hf_search_sorted": (ct.c_int, [ct.c_void_p, ct.c_void_p,
+                                            ct.POINTER(ct.c_void_p)]),
             "hf_shuffle_dest": (ct.c_int, [ct.c_void_p,
                                            ct.POINTER(ct.c_int64), ct.c_int,
                                            ct.POINTER(ct.c_void_p)]),
@@ -191,6 +193,7 @@ def exported_symbols():
         "hf_fixup_empty", "hf_sort_perm", "hf_fill_i64",
         "hf_groupby_hash_accum", "hf_groupby_hash_compact",
         "hf_groupby_sorted", "hf_shuffle_dest", "hf_memcpy_dd",
+        "hf_search_sorted",
         "hf_col_concat", "hf_col_slice", "hf_join_build", "hf_join_free", "hf_join_probe",
         "hf_gather", "hf_compare_scalar", "hf_filter_plan", "hf_filter_apply",
         "hf_filter_iota", "hf_filter_plan_free", "hf_profiling",
@@ -595,6 +598,15 @@ def filter_iota(plan: FilterPlan, base: int) -> ColumnRef:
     _check(load().hf_filter_iota(plan.handle, base, ct.byref(out)),
            "hf_filter_iota")
     return _wrap(out, plan.n_kept, HF_INT64)
+
+
+def search_sorted(keys: ColumnRef, sorted_uniq: ColumnRef) -> ColumnRef:
+    """out[i] = exact-match index of keys[i] in sorted_uniq, else -1."""
+    ensure_ready()
+    out = ct.c_void_p()
+    _check(load().hf_search_sorted(keys.handle, sorted_uniq.handle,
+                                   ct.byref(out)), "hf_search_sorted")
+    return _wrap(out, keys.length, HF_INT64)
 
 
 def shuffle_dest(keys: ColumnRef, splitters) -> ColumnRef:

@@ -1664,6 +1664,23 @@ __global__ void __launch_bounds__(BLOCK) k_segagg(
   }
 }
 
+__global__ void __launch_bounds__(BLOCK) k_search_sorted(
+    const int64_t* __restrict__ keys, int64_t n,
+    const int64_t* __restrict__ sorted, int64_t m,
+    int64_t* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const int64_t k = keys[i];
+    int64_t lo = 0, hi = m;
+    while (lo < hi) {  // lower_bound; first levels stay L2-resident
+      const int64_t mid = (lo + hi) >> 1;
+      if (sorted[mid] < k) lo = mid + 1; else hi = mid;
+    }
+    out[i] = (lo < m && sorted[lo] == k) ? lo : -1;
+  }
+}
+
 __global__ void __launch_bounds__(BLOCK) k_shuffle_dest(
     const int64_t* __restrict__ keys, const int64_t* __restrict__ split,
     int nsplit, long long* __restrict__ dest, int64_t n) {
@@ -2583,6 +2600,29 @@ int hf_groupby_hash_compact(uintptr_t tkey, uintptr_t sums, uintptr_t rowcnt,
 }
 
 static const int64_t* plan_tiles(const hf_filterplan* p);
+
+int hf_search_sorted(const hf_col* keys, const hf_col* sorted_uniq,
+                     hf_col** out) {
+  HF_NEED_INIT("hf_search_sorted");
+  if (!keys || !sorted_uniq || !out)
+    return set_err(HF_ERR_ARG, "hf_search_sorted", "null");
+  if (keys->dtype != HF_INT64 || sorted_uniq->dtype != HF_INT64)
+    return set_err(HF_ERR_ARG, "hf_search_sorted", "int64 columns required");
+  const int64_t n = keys->len;
+  int rc = hf_col_alloc(n, HF_INT64, out);
+  if (rc != HF_OK) return rc;
+  if (n > 0) {
+    rc = timed_launch("search_sorted", [&] {
+      hipLaunchKernelGGL(k_search_sorted, dim3((uint32_t)grid_for(n)),
+                         dim3(BLOCK), 0, g.stream,
+                         (const int64_t*)keys->dptr, n,
+                         (const int64_t*)sorted_uniq->dptr,
+                         sorted_uniq->len, (int64_t*)(*out)->dptr);
+    });
+    if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  }
+  return rc;
+}
 
 int hf_shuffle_dest(const hf_col* keys, const int64_t* splitters, int nsplit,
                     hf_col** dest) {

@@ -701,6 +701,35 @@ def gen_series_cases(mpd, rng):
     return cases
 
 
+def gen_merge2_cases(mpd, rng):
+    """Unbounded-span merge keys (full int64 range; the densify-via-
+    distinct-keys join path) vs the reference."""
+    import pandas
+    cases = {}
+    nl, nr = 4000, 900
+    base = rng.integers(-2**60, 2**60, 700)
+    lk = rng.choice(base, nl)                       # some match
+    miss = rng.random(nl) < 0.2                      # some never match
+    lk[miss] = rng.integers(-2**60, 2**60, int(miss.sum()))
+    rk = rng.choice(base, nr)
+    la, rb = rng.random(nl), rng.random(nr)
+    mout = mpd.DataFrame({"k": lk, "a": la}).merge(
+        mpd.DataFrame({"k": rk, "b": rb}), on="k")._to_pandas()
+    pout = pandas.DataFrame({"k": lk, "a": la}).merge(
+        pandas.DataFrame({"k": rk, "b": rb}), on="k")
+    np.testing.assert_array_equal(mout["k"].to_numpy(), pout["k"].to_numpy())
+    np.testing.assert_allclose(mout[["a", "b"]].values,
+                               pout[["a", "b"]].values, rtol=0)
+    cases["mg2_hugespan"] = {
+        "in_lk": lk.astype(np.int64), "in_la": la,
+        "in_rk": rk.astype(np.int64), "in_rb": rb,
+        "out_k": pout["k"].to_numpy().astype(np.int64),
+        "out_a": pout["a"].to_numpy(), "out_b": pout["b"].to_numpy(),
+        "out_idx": pout.index.to_numpy().astype(np.int64),
+    }
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -718,6 +747,7 @@ def main():
     all_cases.update(gen_var_cases(mpd, rng))
     all_cases.update(gen_multikey_cases(mpd, rng))
     all_cases.update(gen_series_cases(mpd, rng))
+    all_cases.update(gen_merge2_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)

@@ -722,3 +722,35 @@ def test_new_paths_empty_and_edge(npartitions):
         "v": np.array([1.0, 2.0])}))
     with pytest.raises(lib.HfError, match="2\\^62"):
         big.groupby(["a", "b"]).sum().to_pandas()
+
+
+def test_merge_unbounded_span_vs_golden(npartitions):
+    """Merge keys over the full int64 span: the densify path (sorted
+    distinct right keys + device binary search -> CSR join in code
+    space) vs the reference."""
+    g = load_golden("mg2_hugespan")
+    left = mpd.DataFrame({"k": g["in_lk"], "a": g["in_la"]})
+    right = mpd.DataFrame({"k": g["in_rk"], "b": g["in_rb"]})
+    out = left.merge(right, on="k").to_pandas()
+    np.testing.assert_array_equal(out["k"].to_numpy(), g["out_k"])
+    np.testing.assert_array_equal(out["a"].to_numpy(), g["out_a"])
+    np.testing.assert_array_equal(out["b"].to_numpy(), g["out_b"])
+    # repeat merges hit the cached densified build
+    out2 = left.merge(right, on="k").to_pandas()
+    np.testing.assert_array_equal(out2["k"].to_numpy(), g["out_k"])
+
+
+def test_search_sorted_kernel():
+    rng = np.random.default_rng(23)
+    uniq = np.unique(rng.integers(-2**62, 2**62, 5000)).astype(np.int64)
+    keys = np.concatenate([rng.choice(uniq, 20_000),
+                           rng.integers(-2**62, 2**62, 5000)]).astype(np.int64)
+    out = lib.get(lib.search_sorted(lib.put(keys), lib.put(uniq)))
+    expect = np.searchsorted(uniq, keys)
+    expect = np.where((expect < len(uniq)) & (uniq[np.minimum(expect,
+                                                              len(uniq) - 1)]
+                                              == keys), expect, -1)
+    np.testing.assert_array_equal(out, expect)
+    # empty inputs
+    assert lib.get(lib.search_sorted(lib.put(np.empty(0, dtype=np.int64)),
+                                     lib.put(uniq))).size == 0
