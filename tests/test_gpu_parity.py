@@ -754,3 +754,35 @@ def test_search_sorted_kernel():
     # empty inputs
     assert lib.get(lib.search_sorted(lib.put(np.empty(0, dtype=np.int64)),
                                      lib.put(uniq))).size == 0
+
+
+def test_groupby_selection_and_as_index(npartitions):
+    """gb[col] / gb[[cols]] selection and as_index=False (reference
+    reduce fix-up algebra/groupby.py:278) vs pandas inline."""
+    rng = np.random.default_rng(81)
+    n = 50_000
+    pdf = pandas.DataFrame({"k": rng.integers(0, 100, n),
+                            "v": rng.random(n),
+                            "w": rng.random(n)})
+    df = mpd.DataFrame(pdf)
+    s1 = df.groupby("k")["v"].sum()
+    p1 = pdf.groupby("k")["v"].sum()
+    assert s1.name == "v"
+    out = s1.to_pandas()
+    np.testing.assert_array_equal(out.index.to_numpy(), p1.index.to_numpy())
+    np.testing.assert_allclose(out.to_numpy(), p1.to_numpy(), rtol=RTOL)
+    d2 = df.groupby("k")[["w", "v"]].mean().to_pandas()
+    p2 = pdf.groupby("k")[["w", "v"]].mean()
+    assert list(d2.columns) == list(p2.columns)
+    np.testing.assert_allclose(d2.to_numpy(), p2.to_numpy(), rtol=RTOL)
+    d3 = df.groupby("k", as_index=False).sum().to_pandas()
+    p3 = pdf.groupby("k", as_index=False).sum()
+    assert list(d3.columns) == list(p3.columns)
+    np.testing.assert_array_equal(d3["k"].to_numpy(), p3["k"].to_numpy())
+    np.testing.assert_allclose(d3["v"].to_numpy(), p3["v"].to_numpy(),
+                               rtol=RTOL)
+    d4 = df.groupby(["k"], as_index=False)["v"].sum()
+    p4 = pdf.groupby(["k"], as_index=False)["v"].sum()
+    got4 = d4.to_pandas() if hasattr(d4, "to_pandas") else d4
+    np.testing.assert_allclose(got4["v"].to_numpy(), p4["v"].to_numpy(),
+                               rtol=RTOL)
