@@ -92,3 +92,31 @@ def read_parquet(path, columns=None):
     frame = HipDataframe(parts, pandas.RangeIndex(n), names, row_lengths,
                          pandas.Series(dtypes))
     return HipQueryCompiler(frame)
+
+
+def write_parquet(qc, path):
+    """Device columns -> pyarrow table -> parquet.  Dictionary columns
+    rebuild as pyarrow DictionaryArrays straight from the codes (−1 ->
+    null); numeric columns transfer as numpy."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    frame = qc._modin_frame
+    arrays, names = [], []
+    blocks = [p.block() for p in frame._partitions]
+    for name in frame.columns:
+        parts = [b.columns[name] for b in blocks]
+        cats = blocks[0].cats.get(name) if blocks else None
+        nps = [lib.get(c) for c in parts]
+        merged = np.concatenate(nps) if len(nps) != 1 else nps[0]
+        if cats is not None:
+            mask = merged < 0
+            idx = pa.array(np.where(mask, 0, merged).astype(np.int32),
+                           mask=mask)
+            arr = pa.DictionaryArray.from_arrays(
+                idx, pa.array(cats.to_numpy(dtype=object).tolist()))
+        else:
+            arr = pa.array(merged)
+        arrays.append(arr)
+        names.append(name)
+    pq.write_table(pa.table(dict(zip(names, arrays))), path)

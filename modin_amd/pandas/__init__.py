@@ -264,6 +264,34 @@ class DataFrame(_HipPandasBase):
     def astype(self, dtype):
         return DataFrame(query_compiler=self._query_compiler.astype(dtype))
 
+    def drop(self, columns=None):
+        """pandas DataFrame.drop(columns=...): metadata-only column
+        removal (device blocks re-select lazily)."""
+        if columns is None:
+            raise lib.HfError("drop: only drop(columns=...) is supported")
+        drop_set = {columns} if isinstance(columns, str) else set(columns)
+        missing = drop_set - set(self.columns)
+        if missing:
+            raise lib.HfError(f"drop: columns not found: {sorted(missing)}")
+        keep = [c for c in self.columns if c not in drop_set]
+        return self[keep]
+
+    def nunique(self):
+        """Per-column distinct count (NaN excluded), as a pandas Series."""
+        vals = {}
+        for c in self.columns:
+            vals[c] = Series(
+                query_compiler=self._query_compiler.getitem_column_array(
+                    [c]), name=c).nunique()
+        return pandas.Series(vals, dtype=np.int64)
+
+    def to_parquet(self, path):
+        """Round-trip columnar write: device columns -> pyarrow -> parquet
+        (strings rebuilt as dictionary arrays from the codes, so no per-row
+        Python objects either direction)."""
+        from ..io import write_parquet
+        write_parquet(self._query_compiler, path)
+
     def rename(self, columns: dict):
         qc = self._query_compiler
         frame = qc._modin_frame
@@ -351,6 +379,12 @@ class Series(_HipPandasBase):
 
     def __len__(self):
         return len(self._query_compiler)
+
+    def sort_values(self, ascending: bool = True, kind: str = "stable"):
+        """pandas Series.sort_values (always stable)."""
+        name = list(self._query_compiler._modin_frame.columns)[0]
+        qc = self._query_compiler.sort_rows_by_column_values(name, ascending)
+        return Series(query_compiler=qc, name=self.name)
 
     def unique(self):
         """pandas Series.unique: distinct values in FIRST-APPEARANCE order,

@@ -265,3 +265,50 @@ def test_series_utils_vs_golden(npartitions):
     np.testing.assert_array_equal(
         df["k"].isin([]).to_pandas().to_numpy().astype(np.int64),
         g["out_empty_isin"])
+
+
+def test_parquet_write_roundtrip(tmp_path, npartitions):
+    """to_parquet -> read_parquet round trip: dictionary columns travel as
+    parquet dictionary pages, numerics as plain columns."""
+    rng = np.random.default_rng(83)
+    n = 15_000
+    sv = rng.choice(np.array(["aa", "bb", "cc"]), n).astype(object)
+    sv[rng.random(n) < 0.05] = None
+    pdf = pandas.DataFrame({"s": sv, "v": rng.random(n),
+                            "k": rng.integers(-5, 5, n)})
+    df = mpd.DataFrame(pdf)
+    path = str(tmp_path / "w.parquet")
+    df.to_parquet(path)
+    back = mpd.read_parquet(path).to_pandas()
+    exp = pandas.read_parquet(path)
+    assert list(back.columns) == ["s", "v", "k"]
+    np.testing.assert_array_equal(back["v"].to_numpy(),
+                                  pdf["v"].to_numpy())
+    np.testing.assert_array_equal(back["k"].to_numpy(),
+                                  pdf["k"].to_numpy())
+    for g, e in zip(back["s"], pdf["s"]):
+        if e is None:
+            assert isinstance(g, float) and np.isnan(g)
+        else:
+            assert g == e
+    # pandas reads our file identically
+    np.testing.assert_array_equal(exp["v"].to_numpy(), pdf["v"].to_numpy())
+
+
+def test_drop_nunique_series_sort(npartitions):
+    rng = np.random.default_rng(84)
+    n = 20_000
+    pdf = pandas.DataFrame({"k": rng.integers(0, 7, n),
+                            "v": rng.random(n),
+                            "w": rng.integers(0, 1000, n)})
+    df = mpd.DataFrame(pdf)
+    d = df.drop(columns=["v"]).to_pandas()
+    assert list(d.columns) == ["k", "w"]
+    nu = df.nunique()
+    pnu = pdf.nunique()
+    np.testing.assert_array_equal(np.asarray(nu[["k", "w"]]),
+                                  pnu[["k", "w"]].to_numpy())
+    ss = df["w"].sort_values().to_pandas()
+    ps = pdf["w"].sort_values(kind="stable")
+    np.testing.assert_array_equal(ss.to_numpy(), ps.to_numpy())
+    np.testing.assert_array_equal(ss.index.to_numpy(), ps.index.to_numpy())
