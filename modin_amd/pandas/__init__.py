@@ -402,14 +402,19 @@ class Series(_HipPandasBase):
         return np.array(vals, dtype=object)
 
     def value_counts(self):
-        """pandas Series.value_counts: counts desc, ties in appearance
-        order, NaN dropped (default dropna=True)."""
+        """pandas Series.value_counts: the device groupby supplies distinct
+        values + counts + first-appearance positions; the final count-desc
+        sort over the (small) distinct set is delegated to pandas
+        sort_values on the appearance-ordered counts — bit-identical tie
+        order with pandas (which starts from its hashtable's appearance
+        order and quicksorts)."""
         st = self._query_compiler.distinct_stats()
-        order = np.lexsort((st["firstpos"], -st["counts"]))
+        order = np.argsort(st["firstpos"], kind="stable")
         idx = pandas.Index(np.asarray(st["values"], dtype=object)[order],
                            name=self.name)
-        return pandas.Series(st["counts"][order].astype(np.int64),
-                             index=idx, name="count")
+        pre = pandas.Series(st["counts"][order].astype(np.int64),
+                            index=idx, name="count")
+        return pre.sort_values(ascending=False)
 
     def nunique(self) -> int:
         return int(len(self._query_compiler.distinct_stats()["values"]))
