@@ -295,6 +295,14 @@ int hf_sort_perm(const hf_col* keys, int ascending, hf_col** out_perm);
 int hf_shuffle_dest(const hf_col* keys, const int64_t* splitters, int nsplit,
                     hf_col** dest);
 
+/* Order-preserving f64 <-> i64 bit transform (total order trick: flip all
+ * bits of negatives, flip only the sign of non-negatives), so float sort /
+ * groupby / merge keys ride the int64 radix machinery.  -0.0 normalizes to
+ * +0.0 first (pandas groups them together); NaN maps above +inf (pandas
+ * na_position='last').  direction=0: f64 col -> ordered i64 col;
+ * direction=1: ordered i64 col -> f64 col (inverse). */
+int hf_ordered_i64(const hf_col* col, int direction, hf_col** out);
+
 /* Exact-match binary search: out[i] = j with sorted[j] == keys[i], else -1.
  * Densifies unbounded int64 join keys through the sorted distinct right
  * keys (lower_bound per row), so the dense-range CSR join

@@ -346,6 +346,27 @@ class HipDataframe:
                 "numeric columns (string agg is a later round)")
         key_cats = blk_cats.get(by)
         parts = self._partitions
+        float_key = (parts and key_cats is None
+                     and parts[0].block().columns[by].dtype_code
+                     == lib.HF_FLOAT64)
+        if float_key:
+            # pandas drops NaN keys; valid float keys ride the ordered
+            # f64->i64 bit transform (monotone, so output order == pandas)
+            fparts = []
+            for p in parts:
+                block = p.block()
+                kc = block.columns[by]
+                cols = dict(block.columns)
+                if kc.length and lib.reduce(kc).count < kc.length:
+                    mask = lib.compare_scalar(lib.CMP_NOTNA, kc, 0.0)
+                    plan = lib.filter_plan(mask)
+                    cols = {m: lib.filter_apply(plan, c)
+                            for m, c in block.columns.items()}
+                    kc = cols[by]
+                cols[by] = lib.ordered_i64(kc)
+                fparts.append(HipDataframePartition(
+                    DeviceBlock(cols, cols[by].length, block.cats)))
+            parts = fparts
         if key_cats is not None:
             # pandas drops NaN groups (dropna=True): filter code == -1 rows
             has_nan = any(
@@ -411,9 +432,12 @@ class HipDataframe:
             dtypes = pandas.Series({n_: np.dtype(np.float64) for n_ in val_names})
         block = DeviceBlock(cols, n)
         part = HipDataframePartition(block)
-        return HipDataframe([part],
-                            DeviceIndex(keys, name=by, cats=key_cats),
-                            val_names, [n], dtypes)
+        if float_key:
+            idx = pandas.Index(lib.ordered_to_f64_np(lib.get(keys)),
+                               name=by)
+        else:
+            idx = DeviceIndex(keys, name=by, cats=key_cats)
+        return HipDataframe([part], idx, val_names, [n], dtypes)
 
     def groupby_var(self, by: str, ddof: int = 1,
                     sqrt: bool = False) -> "HipDataframe":
@@ -560,13 +584,13 @@ class HipDataframe:
         base = 0
         nan_first = None
         nan_count = 0
+        float_col = (self._partitions
+                     and name not in blk_cats
+                     and self._partitions[0].block().columns[name]
+                     .dtype_code == lib.HF_FLOAT64)
         for p, ln in zip(self._partitions, self._row_lengths):
             block = p.block()
             col = block.columns[name]
-            if col.dtype_code != lib.HF_INT64:
-                raise lib.HfError(
-                    f"unique/value_counts: column {name!r} must be int64 "
-                    "or string")
             ones = lib.compare_scalar(lib.CMP_NOTNA, col, 0.0)
             plan = lib.filter_plan(ones)
             pos = lib.filter_iota(plan, base)
@@ -586,6 +610,25 @@ class HipDataframe:
                 kplan = lib.filter_plan(keepm)
                 cols = {name: lib.filter_apply(kplan, col),
                         POS: lib.filter_apply(kplan, pos)}
+            elif float_col:
+                # the NOTNA plan above already excluded NaN rows from
+                # ``pos``; track the NaNs separately and transform the
+                # kept keys to ordered-i64
+                kc = lib.filter_apply(plan, col)
+                dropped = col.length - plan.n_kept
+                if dropped:
+                    nan_count += dropped
+                    nm_inv = lib.map_scalar(lib.MAP_RSUB, ones, 1)
+                    nplan = lib.filter_plan(nm_inv)
+                    npos = lib.filter_iota(nplan, base)
+                    first = lib.reduce(npos).imn
+                    nan_first = first if nan_first is None \
+                        else min(nan_first, first)
+                cols = {name: lib.ordered_i64(kc), POS: pos}
+            elif col.dtype_code != lib.HF_INT64:
+                raise lib.HfError(
+                    f"unique/value_counts: column {name!r} has an "
+                    "unsupported dtype")
             parts2.append(HipDataframePartition(
                 DeviceBlock(cols, cols[name].length, block.cats)))
             base += ln
@@ -603,6 +646,8 @@ class HipDataframe:
         if name in blk_cats:
             from .partition import decode_dict
             values = decode_dict(keys, blk_cats[name])
+        elif float_col:
+            values = lib.ordered_to_f64_np(keys)
         else:
             values = keys
         return {"values": values, "counts": counts, "firstpos": firstpos,
@@ -911,14 +956,28 @@ class HipDataframe:
     @staticmethod
     def _effective_sort_key(col, is_dict, ascending):
         """(key col, ascending) -> (int64 key, ascending') whose stable
-        ASCENDING' radix sort realizes pandas order.  Dictionary NaN codes
-        (−1) must sort LAST for both directions (na_position='last'):
-        ascending maps −1 to +2^62, descending negates the codes (order
-        flip) and maps −1 above them — so the pass always sorts
-        ascending."""
+        ASCENDING' radix sort realizes pandas order.  NaNs — dictionary
+        code −1 or float NaN — sort LAST for both directions
+        (na_position='last'): the NaN rows map to +2^62 and valid keys
+        keep (asc) or negate (desc) their order, so the pass always runs
+        ascending when an adjustment is needed.  Float keys first ride the
+        order-preserving f64->i64 bit transform (hf_ordered_i64)."""
+        BIG = 1 << 62
+        if col.dtype_code == lib.HF_FLOAT64:
+            r = lib.reduce(col) if col.length else None
+            okey = lib.ordered_i64(col)
+            if r is None or r.count == col.length:  # no NaN
+                return okey, ascending
+            notna = lib.compare_scalar(lib.CMP_NOTNA, col, 0.0)
+            isna_big = lib.map_scalar(
+                lib.MAP_MUL, lib.map_scalar(lib.MAP_RSUB, notna, 1), BIG)
+            base = okey if ascending else lib.map_scalar(lib.MAP_NEG,
+                                                         okey, 0)
+            return lib.binary(lib.BIN_ADD,
+                              lib.binary(lib.BIN_MUL, base, notna),
+                              isna_big), True
         if not is_dict or not col.length or lib.reduce(col).imn >= 0:
             return col, ascending
-        BIG = 1 << 62
         m = lib.compare_scalar(lib.CMP_EQ, col, -1.0)
         if ascending:
             t = lib.map_scalar(lib.MAP_MUL, m, BIG + 1)
@@ -975,14 +1034,8 @@ class HipDataframe:
                 cache[name] = concat_col(name)
             return cache[name]
 
-        eff = []
-        for b, a in zip(by_list, asc_list):
-            kc = cat_col(b)
-            if kc.dtype_code != lib.HF_INT64:
-                raise lib.HfError(
-                    f"sort_values: key column {b!r} must be int64 or "
-                    "string (float sort keys are a later round)")
-            eff.append(self._effective_sort_key(kc, b in blk_cats, a))
+        eff = [self._effective_sort_key(cat_col(b), b in blk_cats, a)
+               for b, a in zip(by_list, asc_list)]
         perm = self._compose_sort_perm(eff)
         cols = {name: lib.gather(cat_col(name), perm)
                 for name in self.columns}
@@ -1007,10 +1060,6 @@ class HipDataframe:
         from .. import distributed as dist_mod
         P = dist_mod.world_size()
         k0 = concat_col(by_list[0])
-        if k0.dtype_code != lib.HF_INT64:
-            raise lib.HfError(
-                f"sort_values: key column {by_list[0]!r} must be int64 or "
-                "string")
         ek0, ea0 = self._effective_sort_key(k0, by_list[0] in blk_cats,
                                             asc_list[0])
         n = ek0.length

@@ -129,6 +129,8 @@ def load() -> ct.CDLL:
                                              ct.POINTER(ct.c_int64)]),
             "hf_search_sorted": (ct.c_int, [ct.c_void_p, ct.c_void_p,
                                             ct.POINTER(ct.c_void_p)]),
+            "hf_ordered_i64": (ct.c_int, [ct.c_void_p, ct.c_int,
+                                          ct.POINTER(ct.c_void_p)]),
             "hf_shuffle_dest": (ct.c_int, [ct.c_void_p,
                                            ct.POINTER(ct.c_int64), ct.c_int,
                                            ct.POINTER(ct.c_void_p)]),
@@ -193,7 +195,7 @@ def exported_symbols():
         "hf_fixup_empty", "hf_sort_perm", "hf_fill_i64",
         "hf_groupby_hash_accum", "hf_groupby_hash_compact",
         "hf_groupby_sorted", "hf_shuffle_dest", "hf_memcpy_dd",
-        "hf_search_sorted",
+        "hf_search_sorted", "hf_ordered_i64",
         "hf_col_concat", "hf_col_slice", "hf_join_build", "hf_join_free", "hf_join_probe",
         "hf_gather", "hf_compare_scalar", "hf_filter_plan", "hf_filter_apply",
         "hf_filter_iota", "hf_filter_plan_free", "hf_profiling",
@@ -598,6 +600,24 @@ def filter_iota(plan: FilterPlan, base: int) -> ColumnRef:
     _check(load().hf_filter_iota(plan.handle, base, ct.byref(out)),
            "hf_filter_iota")
     return _wrap(out, plan.n_kept, HF_INT64)
+
+
+def ordered_i64(col: ColumnRef, inverse: bool = False) -> ColumnRef:
+    """Order-preserving f64 <-> i64 bit transform (float keys on the int64
+    radix machinery; -0.0 == +0.0; NaN above +inf)."""
+    ensure_ready()
+    out = ct.c_void_p()
+    _check(load().hf_ordered_i64(col.handle, 1 if inverse else 0,
+                                 ct.byref(out)), "hf_ordered_i64")
+    return _wrap(out, col.length, HF_FLOAT64 if inverse else HF_INT64)
+
+
+def ordered_to_f64_np(keys_np):
+    """Host inverse of the ordered transform (decode compacted group
+    keys)."""
+    s_ = np.asarray(keys_np, dtype=np.int64)
+    bits = np.where(s_ < 0, ~(s_ ^ np.int64(-2**63)), s_)
+    return bits.view(np.float64)
 
 
 def search_sorted(keys: ColumnRef, sorted_uniq: ColumnRef) -> ColumnRef:

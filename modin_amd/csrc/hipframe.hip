@@ -1664,6 +1664,30 @@ __global__ void __launch_bounds__(BLOCK) k_segagg(
   }
 }
 
+__global__ void __launch_bounds__(BLOCK) k_f64_ordered(
+    const double* __restrict__ in, int64_t* __restrict__ out, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    double x = in[i] + 0.0;  // -0.0 -> +0.0 (pandas groups them together)
+    long long v = __double_as_longlong(x);
+    // signed total order: non-negative floats keep their bits (already
+    // increasing as signed i64); negative floats reverse below zero
+    out[i] = (v < 0) ? (~v ^ 0x8000000000000000LL) : v;
+  }
+}
+
+__global__ void __launch_bounds__(BLOCK) k_ordered_f64(
+    const int64_t* __restrict__ in, double* __restrict__ out, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const long long v = in[i];
+    out[i] = __longlong_as_double(
+        (v < 0) ? ~(v ^ 0x8000000000000000LL) : v);
+  }
+}
+
 __global__ void __launch_bounds__(BLOCK) k_search_sorted(
     const int64_t* __restrict__ keys, int64_t n,
     const int64_t* __restrict__ sorted, int64_t m,
@@ -2600,6 +2624,35 @@ int hf_groupby_hash_compact(uintptr_t tkey, uintptr_t sums, uintptr_t rowcnt,
 }
 
 static const int64_t* plan_tiles(const hf_filterplan* p);
+
+int hf_ordered_i64(const hf_col* col, int direction, hf_col** out) {
+  HF_NEED_INIT("hf_ordered_i64");
+  if (!col || !out) return set_err(HF_ERR_ARG, "hf_ordered_i64", "null");
+  const int want = direction ? HF_INT64 : HF_FLOAT64;
+  if (col->dtype != want)
+    return set_err(HF_ERR_ARG, "hf_ordered_i64",
+                   direction ? "inverse needs an int64 column"
+                             : "forward needs a float64 column");
+  const int64_t n = col->len;
+  int rc = hf_col_alloc(n, direction ? HF_FLOAT64 : HF_INT64, out);
+  if (rc != HF_OK) return rc;
+  if (n > 0) {
+    rc = timed_launch("f64_ordered", [&] {
+      if (direction)
+        hipLaunchKernelGGL(k_ordered_f64, dim3((uint32_t)grid_for(n)),
+                           dim3(BLOCK), 0, g.stream,
+                           (const int64_t*)col->dptr, (double*)(*out)->dptr,
+                           n);
+      else
+        hipLaunchKernelGGL(k_f64_ordered, dim3((uint32_t)grid_for(n)),
+                           dim3(BLOCK), 0, g.stream,
+                           (const double*)col->dptr,
+                           (int64_t*)(*out)->dptr, n);
+    });
+    if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  }
+  return rc;
+}
 
 int hf_search_sorted(const hf_col* keys, const hf_col* sorted_uniq,
                      hf_col** out) {

@@ -730,6 +730,48 @@ def gen_merge2_cases(mpd, rng):
     return cases
 
 
+def gen_float_key_cases(mpd, rng):
+    """Float sort keys (NaN last both directions), float groupby keys
+    (NaN dropped), float unique/value_counts — the ordered f64<->i64
+    transform path vs the reference."""
+    import pandas
+    cases = {}
+    n = 4000
+    f = np.round(rng.standard_normal(n) * 100, 2)
+    f[rng.random(n) < 0.07] = np.nan
+    f[rng.random(n) < 0.02] = -0.0  # groups with +0.0
+    f[rng.random(n) < 0.02] = 0.0
+    w = rng.integers(0, 12, n).astype(np.int64)
+    v = rng.random(n)
+    mdf = mpd.DataFrame({"f": f, "w": w, "v": v})
+    pdf = pandas.DataFrame({"f": f, "w": w, "v": v})
+    arrays = {"in_f": f, "in_w": w, "in_v": v}
+    for tag, by, asc in [("f_asc", "f", True), ("f_desc", "f", False),
+                         ("fw", ["f", "w"], True),
+                         ("wf_mixed", ["w", "f"], [True, False])]:
+        mres = mdf.sort_values(by, ascending=asc, kind="stable")._to_pandas()
+        pres = pdf.sort_values(by, ascending=asc, kind="stable")
+        assert list(mres.index) == list(pres.index), tag
+        arrays[f"out_{tag}_idx"] = pres.index.to_numpy().astype(np.int64)
+        arrays[f"out_{tag}_f"] = pres["f"].to_numpy()
+    for agg in ("sum", "mean", "count"):
+        mres = getattr(mdf.groupby("f"), agg)()._to_pandas()
+        pres = getattr(pdf.groupby("f"), agg)()
+        assert list(mres.index) == list(pres.index), agg
+        arrays[f"out_gb_{agg}_keys"] = pres.index.to_numpy()
+        for cn in ("w", "v"):
+            arrays[f"out_gb_{agg}_{cn}"] = pres[cn].to_numpy()
+    pu = pdf["f"].unique()
+    arrays["out_unique"] = pu
+    pvc = pdf["f"].value_counts()
+    arrays["out_vc_idx"] = pvc.index.to_numpy()
+    arrays["out_vc"] = pvc.to_numpy().astype(np.int64)
+    arrays["out_nunique"] = np.array([pdf["f"].nunique()], dtype=np.int64)
+    np.testing.assert_array_equal(np.asarray(mdf["f"].unique()), pu)
+    cases["flt_keys"] = arrays
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -748,6 +790,7 @@ def main():
     all_cases.update(gen_multikey_cases(mpd, rng))
     all_cases.update(gen_series_cases(mpd, rng))
     all_cases.update(gen_merge2_cases(mpd, rng))
+    all_cases.update(gen_float_key_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)

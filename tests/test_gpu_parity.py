@@ -786,3 +786,40 @@ def test_groupby_selection_and_as_index(npartitions):
     got4 = d4.to_pandas() if hasattr(d4, "to_pandas") else d4
     np.testing.assert_allclose(got4["v"].to_numpy(), p4["v"].to_numpy(),
                                rtol=RTOL)
+
+
+def test_float_keys_vs_golden(npartitions):
+    """Float sort/groupby/unique keys via the ordered f64<->i64 bit
+    transform: NaN-last sorts both directions, NaN-dropped groups,
+    -0.0 == +0.0, exact key decode."""
+    g = load_golden("flt_keys")
+    df = mpd.DataFrame({"f": g["in_f"], "w": g["in_w"], "v": g["in_v"]})
+    for tag, by, asc in [("f_asc", "f", True), ("f_desc", "f", False),
+                         ("fw", ["f", "w"], True),
+                         ("wf_mixed", ["w", "f"], [True, False])]:
+        out = df.sort_values(by, ascending=asc).to_pandas()
+        np.testing.assert_array_equal(out.index.to_numpy(),
+                                      g[f"out_{tag}_idx"], err_msg=tag)
+        np.testing.assert_array_equal(out["f"].to_numpy(),
+                                      g[f"out_{tag}_f"], err_msg=tag)
+    for agg in ("sum", "mean", "count"):
+        out = getattr(df.groupby("f"), agg)().to_pandas()
+        np.testing.assert_array_equal(out.index.to_numpy(),
+                                      g[f"out_gb_{agg}_keys"],
+                                      err_msg=f"gb {agg} keys")
+        for cn in ("w", "v"):
+            expect = g[f"out_gb_{agg}_{cn}"]
+            np.testing.assert_allclose(out[cn].to_numpy(), expect,
+                                       rtol=RTOL, atol=1e-9,
+                                       equal_nan=True,
+                                       err_msg=f"gb {agg}/{cn}")
+    u = df["f"].unique()
+    expu = g["out_unique"]
+    assert len(u) == len(expu)
+    np.testing.assert_array_equal(np.asarray(u, dtype=np.float64),
+                                  expu.astype(np.float64))
+    vc = df["f"].value_counts()
+    np.testing.assert_array_equal(vc.index.to_numpy().astype(np.float64),
+                                  g["out_vc_idx"].astype(np.float64))
+    np.testing.assert_array_equal(vc.to_numpy(), g["out_vc"])
+    assert df["f"].nunique() == int(g["out_nunique"][0])
