@@ -654,6 +654,106 @@ class HipDataframe:
                 "nan_count": int(nan_count),
                 "nan_firstpos": nan_first}
 
+    def median_columns(self):
+        """Per-column median (NaN skipped): NOTNA filter -> ordered radix
+        sort -> middle element(s) sliced on device; pandas nanmedian."""
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        out = {}
+        for name in self.columns:
+            if name in blk_cats:
+                raise lib.HfError(f"median over string column {name!r}")
+            cols = [p.block().columns[name] for p in self._partitions]
+            col = cols[0] if len(cols) == 1 else lib.concat(cols)
+            if col.dtype_code == lib.HF_FLOAT64 and col.length:
+                mask = lib.compare_scalar(lib.CMP_NOTNA, col, 0.0)
+                plan = lib.filter_plan(mask)
+                if plan.n_kept < col.length:
+                    col = lib.filter_apply(plan, col)
+            n = col.length
+            if n == 0:
+                out[name] = float("nan")
+                continue
+            key = (lib.ordered_i64(col)
+                   if col.dtype_code == lib.HF_FLOAT64 else col)
+            perm = lib.sort_perm(key)
+            sv = lib.gather(lib.cast_f64(col)
+                            if col.dtype_code == lib.HF_INT64 else col,
+                            perm)
+            lo, hi = (n - 1) // 2, n // 2
+            mid = lib.get(lib.col_slice(sv, lo, hi - lo + 1))
+            out[name] = float(mid.mean())
+        return out
+
+    def groupby_median(self, by) -> "HipDataframe":
+        """groupby().median(): sort by (key, value na-last) once per value
+        column, then gather the per-group middle elements — offsets from
+        groupby size, non-NaN counts from groupby count (pandas
+        nanmedian per group)."""
+        from ..distributed import is_active
+        if is_active():
+            raise lib.HfError(
+                "distributed groupby.median is a later round")
+        if isinstance(by, (list, tuple)):
+            if len(by) == 1:
+                by = by[0]
+            else:
+                cf, decode = self._combined_key_frame(list(by))
+                keep = [c for c in cf.columns if c not in by]
+                res = cf.take_columns(keep).groupby_median(self.KEYCOL)
+                res._index = decode(lib.get(res._index.col))
+                return res
+        val_names = [c for c in self.columns if c != by]
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        bad = [v for v in val_names if v in blk_cats]
+        if bad:
+            raise lib.HfError(f"groupby median over string column(s) {bad}")
+        sizes_f = self.groupby_size(by)
+        sizes = lib.get(sizes_f._partitions[0].block().columns["size"])
+        offs = np.concatenate([[0], np.cumsum(sizes)[:-1]])
+        cnt_res = self.groupby_reduce(by, "count")
+        cnt_block = cnt_res._partitions[0].block()
+        ng = len(sizes)
+
+        def concat_col(m):
+            cols = [p.block().columns[m] for p in self._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        key_col = concat_col(by)
+        key_cats = blk_cats.get(by)
+        # NaN keys (dict -1 / float NaN) sort LAST via the effective key,
+        # so the leading group runs line up with the (NaN-key-dropped)
+        # size/count tables
+        ekey, _ = self._effective_sort_key(key_col, key_cats is not None,
+                                           True)
+        out_cols = {}
+        for v in val_names:
+            vcol = concat_col(v)
+            eval_, _ = self._effective_sort_key(
+                lib.cast_f64(vcol) if vcol.dtype_code == lib.HF_INT64
+                else vcol, False, True)
+            perm = self._compose_sort_perm([(ekey, True), (eval_, True)])
+            sv = lib.gather(lib.cast_f64(vcol)
+                            if vcol.dtype_code == lib.HF_INT64 else vcol,
+                            perm)
+            cnt = lib.get(cnt_block.columns[v]) if ng else np.empty(0)
+            cnt = np.asarray(cnt, dtype=np.int64)
+            lo = offs + np.maximum((cnt - 1) // 2, 0)
+            hi = offs + np.maximum(cnt // 2, 0)
+            lo_c = lib.put(lo.astype(np.int64))
+            hi_c = lib.put(hi.astype(np.int64))
+            a = lib.get(lib.gather(sv, lo_c)) if ng else np.empty(0)
+            b = lib.get(lib.gather(sv, hi_c)) if ng else np.empty(0)
+            med = (a + b) / 2.0
+            med[cnt == 0] = np.nan
+            out_cols[v] = lib.put(med)
+        part = HipDataframePartition(DeviceBlock(out_cols, ng))
+        dtypes = pandas.Series({v: np.dtype(np.float64)
+                                for v in val_names})
+        return HipDataframe([part], cnt_res._index, val_names, [ng],
+                            dtypes)
+
     def hconcat(self, others: list) -> "HipDataframe":
         """Horizontal compose of single-partition frames with identical row
         count and index (the device form of the reference's axis=1 concat
@@ -968,9 +1068,13 @@ class HipDataframe:
             okey = lib.ordered_i64(col)
             if r is None or r.count == col.length:  # no NaN
                 return okey, ascending
+            # NaN sentinel must exceed ordered(+inf) ~ 0x7FF0... ~ 9.22e18
+            # (2^62 would land BELOW the ordered bits of floats >= 2.0)
+            NANKEY = (1 << 63) - 1
             notna = lib.compare_scalar(lib.CMP_NOTNA, col, 0.0)
             isna_big = lib.map_scalar(
-                lib.MAP_MUL, lib.map_scalar(lib.MAP_RSUB, notna, 1), BIG)
+                lib.MAP_MUL, lib.map_scalar(lib.MAP_RSUB, notna, 1),
+                NANKEY)
             base = okey if ascending else lib.map_scalar(lib.MAP_NEG,
                                                          okey, 0)
             return lib.binary(lib.BIN_ADD,

@@ -69,6 +69,9 @@ class _HipPandasBase:
     def max(self, **kwargs):
         return self._lower(self._query_compiler.max(**kwargs))
 
+    def median(self, **kwargs):
+        return self._lower(self._query_compiler.median())
+
     def var(self, ddof: int = 1):
         return self._lower(self._query_compiler.var(ddof=ddof))
 
@@ -524,6 +527,9 @@ class DataFrameGroupBy:
         return DataFrame(query_compiler=self._df._query_compiler.groupby_std(
             self._by, ddof=ddof))
 
+    def median(self):
+        return self._agg("median")
+
     def size(self):
         """pandas DataFrameGroupBy.size(): a Series of group row counts
         (NaN values included, NaN keys dropped)."""
@@ -532,7 +538,8 @@ class DataFrameGroupBy:
         ).to_pandas()
         return out["size"].rename(None)
 
-    _AGGS = ("sum", "count", "mean", "min", "max", "var", "std")
+    _AGGS = ("sum", "count", "mean", "min", "max", "var", "std",
+             "median")
 
     def agg(self, how):
         """str, list-of-str (MultiIndex columns, pandas col-major order) or

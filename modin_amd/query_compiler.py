@@ -123,6 +123,13 @@ class HipQueryCompiler:
         return self.__constructor__(
             self._modin_frame.groupby_var(by, ddof, sqrt=True))
 
+    def median(self):
+        vals = self._modin_frame.median_columns()
+        return pandas.Series(vals, dtype=np.float64)
+
+    def groupby_median(self, by) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.groupby_median(by))
+
     def groupby_size(self, by: str) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.groupby_size(by))
 
@@ -159,6 +166,7 @@ class HipQueryCompiler:
             "max": type(self).groupby_max,
             "var": type(self).groupby_var,
             "std": type(self).groupby_std,
+            "median": type(self).groupby_median,
         }.get(agg)
         if fn is None:
             raise lib.HfError(

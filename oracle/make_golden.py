@@ -772,6 +772,57 @@ def gen_float_key_cases(mpd, rng):
     return cases
 
 
+def gen_median_cases(mpd, rng):
+    """df.median and groupby.median (NaN values skipped, even/odd group
+    sizes, string and multi keys) vs the reference."""
+    import pandas
+    cases = {}
+    n = 4000
+    k = rng.integers(0, 40, n).astype(np.int64)
+    pool = np.array(["red", "blue", "green"])
+    sk = rng.choice(pool, n).astype(object)
+    sk[rng.random(n) < 0.05] = np.nan
+    v = rng.random(n) * 100
+    v[rng.random(n) < 0.12] = np.nan
+    w = rng.integers(-50, 50, n).astype(np.int64)
+    mdf = mpd.DataFrame({"k": k, "s": sk, "v": v, "w": w})
+    pdf = pandas.DataFrame({"k": k, "s": sk, "v": v, "w": w})
+    arrays = {"in_k": k, "in_s": _enc_str(sk), "in_v": v, "in_w": w}
+    mm = mdf[["k", "v", "w"]].median()
+    pm = pdf[["k", "v", "w"]].median()
+    np.testing.assert_allclose(np.asarray(mm), pm.to_numpy(), rtol=1e-12)
+    arrays["out_frame_median"] = pm.to_numpy()
+    for tag, by in [("k", "k"), ("s", "s"), ("ks", ["k", "s"])]:
+        sub = [c for c in ("k", "s", "v", "w")
+               if c not in (by if isinstance(by, list) else [by])]
+        msel = mdf[([by] if isinstance(by, str) else by) + sub]
+        psel = pdf[([by] if isinstance(by, str) else by) + sub]
+        # drop the string column from the aggregation side
+        numeric = [c for c in sub if c != "s"]
+        mres = msel[([by] if isinstance(by, str) else by) + numeric] \
+            .groupby(by).median()._to_pandas()
+        pres = psel[([by] if isinstance(by, str) else by) + numeric] \
+            .groupby(by).median()
+        assert list(mres.index) == list(pres.index), tag
+        np.testing.assert_allclose(mres.values, pres.values, rtol=1e-12,
+                                   atol=1e-12, equal_nan=True)
+        if isinstance(by, str) and by == "k":
+            arrays["out_gbk_keys"] = pres.index.to_numpy().astype(np.int64)
+        elif isinstance(by, str):
+            arrays["out_gbs_keys"] = _enc_str(pres.index)
+        else:
+            arrays["out_gbks_ka"] = pres.index.get_level_values(0) \
+                .to_numpy().astype(np.int64)
+            arrays["out_gbks_kb"] = _enc_str(
+                pres.index.get_level_values(1))
+        sfx = ("gbk" if by == "k" else
+               "gbs" if isinstance(by, str) else "gbks")
+        for cn in numeric:
+            arrays[f"out_{sfx}_{cn}"] = pres[cn].to_numpy()
+    cases["med_cases"] = arrays
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -791,6 +842,7 @@ def main():
     all_cases.update(gen_series_cases(mpd, rng))
     all_cases.update(gen_merge2_cases(mpd, rng))
     all_cases.update(gen_float_key_cases(mpd, rng))
+    all_cases.update(gen_median_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)

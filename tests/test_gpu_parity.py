@@ -159,13 +159,21 @@ def test_property_map_roundtrip_idempotence():
 
 def test_error_surfaces():
     rng = np.random.default_rng(16)
-    df = mpd.DataFrame({"k": rng.random(100), "v": rng.random(100)})
-    with pytest.raises(lib.HfError, match="int64"):
-        df.groupby("k").sum()  # float keys -> loud, not silent fallback
+    # float groupby keys ride the ordered transform (was a loud error in
+    # early round 1; now parity-checked here against pandas)
+    kf = rng.random(100)
+    vf = rng.random(100)
+    pdf = pandas.DataFrame({"k": kf, "v": vf})
+    out = mpd.DataFrame(pdf).groupby("k").sum().to_pandas()
+    exp = pdf.groupby("k").sum()
+    np.testing.assert_array_equal(out.index.to_numpy(),
+                                  exp.index.to_numpy())
+    np.testing.assert_allclose(out["v"].to_numpy(), exp["v"].to_numpy(),
+                               rtol=RTOL)
     dfi = mpd.DataFrame({"k": rng.integers(0, 5, 100).astype(np.int64),
                          "v": rng.random(100)})
     with pytest.raises(lib.HfError, match="not implemented"):
-        dfi.groupby("k").agg("median")
+        dfi.groupby("k").agg("prod")
     # key range beyond the dense-table cap routes to the hash path
     old = config.MaxGroupbySlots.get()
     config.MaxGroupbySlots.put(10)
@@ -823,3 +831,37 @@ def test_float_keys_vs_golden(npartitions):
                                   g["out_vc_idx"].astype(np.float64))
     np.testing.assert_array_equal(vc.to_numpy(), g["out_vc"])
     assert df["f"].nunique() == int(g["out_nunique"][0])
+
+
+def test_median_vs_golden(npartitions):
+    """df.median and groupby.median (int/string/multi keys, NaN values,
+    even/odd group sizes) vs the reference."""
+    from tests.test_gpu_strings import assert_str_equal, dec
+    g = load_golden("med_cases")
+    df = mpd.DataFrame({"k": g["in_k"], "s": dec(g["in_s"]),
+                        "v": g["in_v"], "w": g["in_w"]})
+    fm = df[["k", "v", "w"]].median()
+    np.testing.assert_allclose(np.asarray(fm), g["out_frame_median"],
+                               rtol=RTOL)
+    out = df[["k", "v", "w"]].groupby("k").median().to_pandas()
+    np.testing.assert_array_equal(out.index.to_numpy(), g["out_gbk_keys"])
+    for cn in ("v", "w"):
+        np.testing.assert_allclose(out[cn].to_numpy(),
+                                   g[f"out_gbk_{cn}"], rtol=RTOL,
+                                   atol=1e-12, equal_nan=True,
+                                   err_msg=f"gbk/{cn}")
+    out = df[["s", "v", "w"]].groupby("s").median().to_pandas()
+    assert_str_equal(out.index.to_numpy(), g["out_gbs_keys"], "gbs keys")
+    for cn in ("v", "w"):
+        np.testing.assert_allclose(out[cn].to_numpy(),
+                                   g[f"out_gbs_{cn}"], rtol=RTOL,
+                                   atol=1e-12, equal_nan=True)
+    out = df.groupby(["k", "s"]).median().to_pandas()
+    np.testing.assert_array_equal(
+        out.index.get_level_values(0).to_numpy(), g["out_gbks_ka"])
+    assert_str_equal(out.index.get_level_values(1).to_numpy(),
+                     g["out_gbks_kb"], "gbks kb")
+    for cn in ("v", "w"):
+        np.testing.assert_allclose(out[cn].to_numpy(),
+                                   g[f"out_gbks_{cn}"], rtol=RTOL,
+                                   atol=1e-12, equal_nan=True)
