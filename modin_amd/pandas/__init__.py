@@ -383,6 +383,19 @@ class Series(_HipPandasBase):
     def __len__(self):
         return len(self._query_compiler)
 
+    def __getitem__(self, key):
+        """Boolean-mask selection: s[s > 0] (pandas Series mask form)."""
+        if isinstance(key, Series):
+            return Series(
+                query_compiler=self._query_compiler.getitem_array(
+                    key._query_compiler), name=self.name)
+        raise lib.HfError("Series supports boolean-mask selection only")
+
+    def dropna(self) -> "Series":
+        qc = self._query_compiler
+        return Series(query_compiler=qc.getitem_array(qc.notna()),
+                      name=self.name)
+
     def sort_values(self, ascending: bool = True, kind: str = "stable"):
         """pandas Series.sort_values (always stable)."""
         name = list(self._query_compiler._modin_frame.columns)[0]

@@ -865,3 +865,21 @@ def test_median_vs_golden(npartitions):
         np.testing.assert_allclose(out[cn].to_numpy(),
                                    g[f"out_gbks_{cn}"], rtol=RTOL,
                                    atol=1e-12, equal_nan=True)
+
+
+def test_series_mask_and_dropna(npartitions):
+    rng = np.random.default_rng(86)
+    v = rng.standard_normal(10_000)
+    v[rng.random(10_000) < 0.1] = np.nan
+    ps = pandas.Series(v, name="v")
+    df = mpd.DataFrame({"v": v})
+    s_ = df["v"]
+    got = s_[s_ > 0.5].to_pandas()
+    exp = ps[ps > 0.5]
+    np.testing.assert_array_equal(got.index.to_numpy(),
+                                  exp.index.to_numpy())
+    np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy())
+    gd = s_.dropna().to_pandas()
+    ed = ps.dropna()
+    np.testing.assert_array_equal(gd.index.to_numpy(), ed.index.to_numpy())
+    np.testing.assert_array_equal(gd.to_numpy(), ed.to_numpy())
