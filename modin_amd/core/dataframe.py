@@ -780,9 +780,14 @@ class HipDataframe:
     # ---- broadcast inner join (MergeImpl.row_axis_merge device form,
     #      merge.py:104-178: combine() the right frame once, probe per left
     #      partition; pandas suffix rules "_x"/"_y" on collisions) ----
-    def broadcast_join(self, other: "HipDataframe", on: str) -> "HipDataframe":
+    def broadcast_join(self, other: "HipDataframe", on: str,
+                       how: str = "inner") -> "HipDataframe":
         if on not in self.columns or on not in other.columns:
             raise lib.HfError(f"merge: key column {on!r} missing")
+        if how not in ("inner", "left"):
+            raise lib.HfError(
+                f"merge how={how!r} not implemented (inner/left broadcast "
+                "join this round)")
         left_names = [c for c in self.columns if c != on]
         right_names = [c for c in other.columns if c != on]
         common = set(left_names) & set(right_names)
@@ -874,17 +879,82 @@ class HipDataframe:
             j = lib.join_build(rkeys, rvals, kmin, n_slots)
             other._join_build_cache = (cache_key, j, uniq)
 
+        runiq = uniq  # probe-space distinct right keys (left-join test)
+        if how == "left" and runiq is None:
+            if rkeys.length:
+                rperm = lib.sort_perm(rkeys)
+                runiq, _ru, _rc2, _rn = lib.groupby_sorted(
+                    lib.gather(rkeys, rperm), [], lib.AGG_SUM, False)
+            else:
+                runiq = lib.put(np.empty(0, dtype=np.int64))
+        # pandas dtype rule: a left join that introduces NaN rights turns
+        # int64 right columns into float64 — decided GLOBALLY first
+        total_unmatched = 0
+        un_plans = []
+        if how == "left":
+            for p in self._partitions:
+                block = p.block()
+                lk = block.columns[on]
+                if uniq is not None:
+                    lk = lib.search_sorted(lk, uniq)
+                    m = lib.compare_scalar(lib.CMP_EQ, lk, -1.0)
+                else:
+                    code = lib.search_sorted(lk, runiq)
+                    m = lib.compare_scalar(lib.CMP_EQ, code, -1.0)
+                plan = lib.filter_plan(m)
+                un_plans.append((plan, lk))
+                total_unmatched += plan.n_kept
+
         out_parts, lengths = [], []
-        for p in self._partitions:
+        for pi, p in enumerate(self._partitions):
             block = p.block()
             lkeys = block.columns[on]
             if lkeys.dtype_code != lib.HF_INT64:
                 raise lib.HfError("merge: key column must be int64")
             if uniq is not None:  # code space: unmatched lefts become -1
-                lkeys = lib.search_sorted(lkeys, uniq)
+                lkeys = un_plans[pi][1] if how == "left" \
+                    else lib.search_sorted(lkeys, uniq)
             keys_c, lidx, rcols, nout = lib.join_probe(j, lkeys)
             if uniq is not None:  # decode output codes back to key values
                 keys_c = lib.gather(uniq, keys_c)
+            if how == "left":
+                plan = un_plans[pi][0]
+                n_un = plan.n_kept
+                if n_un:
+                    ulidx = lib.filter_iota(plan, 0)
+                    lidx_all = lib.concat([lidx, ulidx])
+                    order = lib.sort_perm(lidx_all)
+                    lidx_s = lib.gather(lidx_all, order)
+                    cols = {}
+                    for name in self.columns:
+                        key = on if name == on else lout[name]
+                        cols[key] = lib.gather(block.columns[name], lidx_s)
+                    for i, rn in enumerate(right_names):
+                        rc = rcols[i]
+                        if rn in rcats:  # dict payload: NaN is code −1
+                            nanfill = lib.alloc(n_un, lib.HF_INT64)
+                            lib.fill_i64(nanfill.dptr(), -1, n_un)
+                        else:
+                            if rc.dtype_code == lib.HF_INT64:
+                                rc = lib.cast_f64(rc)
+                            nanfill = lib.alloc(n_un, lib.HF_FLOAT64)
+                            lib.fill_f64(nanfill.dptr(), float("nan"),
+                                         n_un)
+                        cols[rout[rn]] = lib.gather(
+                            lib.concat([rc, nanfill]), order)
+                    nout2 = nout + n_un
+                    out_parts.append(HipDataframePartition(
+                        DeviceBlock(cols, nout2)))
+                    lengths.append(nout2)
+                    continue
+                # no unmatched rows in this partition: fall through, but
+                # keep the global dtype decision consistent
+                if total_unmatched:
+                    rcols = [lib.cast_f64(rc)
+                             if (rn not in rcats
+                                 and rc.dtype_code == lib.HF_INT64)
+                             else rc
+                             for rn, rc in zip(right_names, rcols)]
             cols = {}
             for name in self.columns:  # left column order, key in place
                 if name == on:
@@ -901,7 +971,11 @@ class HipDataframe:
         for c in self.columns:
             dtypes[on if c == on else lout[c]] = self.dtypes[c]
         for c in right_names:
-            dtypes[rout[c]] = other.dtypes[c]
+            d = other.dtypes[c]
+            if (how == "left" and total_unmatched and c not in rcats
+                    and d == np.dtype(np.int64)):
+                d = np.dtype(np.float64)
+            dtypes[rout[c]] = d
         out_cats = {}
         if key_cats is not None:
             out_cats[on] = key_cats

@@ -233,12 +233,8 @@ class HipQueryCompiler:
     #      storage_formats/pandas/merge.py:104) ----
     def merge(self, right: "HipQueryCompiler", on: str,
               how: str = "inner") -> "HipQueryCompiler":
-        if how != "inner":
-            raise lib.HfError(
-                f"merge how={how!r} not implemented (inner broadcast join "
-                "this round)")
         return self.__constructor__(
-            self._modin_frame.broadcast_join(right._modin_frame, on)
+            self._modin_frame.broadcast_join(right._modin_frame, on, how)
         )
 
     # ---- projection ----

@@ -823,6 +823,88 @@ def gen_median_cases(mpd, rng):
     return cases
 
 
+def gen_left_merge_cases(mpd, rng):
+    """merge(how='left'): unmatched lefts keep NaN rights (int right
+    columns become float64), pandas left row order, string keys and
+    huge-span keys included — vs the reference."""
+    import pandas
+    cases = {}
+    nl, nr = 3000, 700
+    lk = rng.integers(0, 2000, nl).astype(np.int64)   # many unmatched
+    rk = rng.integers(0, 900, nr).astype(np.int64)
+    la = rng.random(nl)
+    rb = rng.random(nr)
+    ri = rng.integers(-100, 100, nr).astype(np.int64)
+    mout = mpd.DataFrame({"k": lk, "a": la}).merge(
+        mpd.DataFrame({"k": rk, "b": rb, "i": ri}), on="k",
+        how="left")._to_pandas()
+    pout = pandas.DataFrame({"k": lk, "a": la}).merge(
+        pandas.DataFrame({"k": rk, "b": rb, "i": ri}), on="k", how="left")
+    assert list(mout.dtypes) == list(pout.dtypes)
+    np.testing.assert_array_equal(mout["k"].to_numpy(), pout["k"].to_numpy())
+    np.testing.assert_allclose(mout[["a", "b", "i"]].values,
+                               pout[["a", "b", "i"]].values, rtol=0,
+                               equal_nan=True)
+    cases["mgl_basic"] = {
+        "in_lk": lk, "in_la": la, "in_rk": rk, "in_rb": rb, "in_ri": ri,
+        "out_k": pout["k"].to_numpy().astype(np.int64),
+        "out_a": pout["a"].to_numpy(), "out_b": pout["b"].to_numpy(),
+        "out_i": pout["i"].to_numpy(),
+    }
+
+    # all-matched: int right col stays int64
+    lk2 = rng.choice(rk, 1500)
+    m2 = mpd.DataFrame({"k": lk2, "a": rng.random(1500)}).merge(
+        mpd.DataFrame({"k": rk, "i": ri}), on="k", how="left")._to_pandas()
+    p2 = pandas.DataFrame({"k": lk2,
+                           "a": np.zeros(1500)}).merge(
+        pandas.DataFrame({"k": rk, "i": ri}), on="k", how="left")
+    assert m2["i"].dtype == p2["i"].dtype == np.dtype(np.int64)
+    cases["mgl_allmatch"] = {
+        "in_lk": lk2.astype(np.int64), "in_rk": rk, "in_ri": ri,
+        "out_k": p2["k"].to_numpy().astype(np.int64),
+        "out_i": p2["i"].to_numpy().astype(np.int64),
+    }
+
+    # string keys + string payload
+    pool = np.array(["ant", "bee", "cat", "dog", "eel", "fox"])
+    lks = rng.choice(pool, 2000).astype(object)
+    rks = rng.choice(pool[:4], 500).astype(object)
+    rs = rng.choice(np.array(["x", "y"]), 500).astype(object)
+    mout = mpd.DataFrame({"s": lks, "a": rng.random(2000)}).merge(
+        mpd.DataFrame({"s": rks, "t": rs}), on="s", how="left")._to_pandas()
+    pout = pandas.DataFrame({"s": lks, "a": np.zeros(2000)}).merge(
+        pandas.DataFrame({"s": rks, "t": rs}), on="s", how="left")
+    assert list(mout["s"]) == list(pout["s"])
+    assert list(mout["t"].fillna(NA)) == list(pout["t"].fillna(NA))
+    cases["mgl_str"] = {
+        "in_ls": _enc_str(lks), "in_rs": _enc_str(rks),
+        "in_rt": _enc_str(rs),
+        "out_s": _enc_str(pout["s"]), "out_t": _enc_str(pout["t"]),
+        "out_idx": pout.index.to_numpy().astype(np.int64),
+    }
+
+    # huge-span keys (densify path) left join
+    base = rng.integers(-2**60, 2**60, 400)
+    lkh = rng.choice(base, 2000)
+    miss = rng.random(2000) < 0.3
+    lkh[miss] = rng.integers(-2**60, 2**60, int(miss.sum()))
+    rkh = rng.choice(base, 300)
+    rbv = rng.random(300)
+    mout = mpd.DataFrame({"k": lkh, "a": rng.random(2000)}).merge(
+        mpd.DataFrame({"k": rkh, "b": rbv}), on="k", how="left")._to_pandas()
+    pout = pandas.DataFrame({"k": lkh, "a": np.zeros(2000)}).merge(
+        pandas.DataFrame({"k": rkh, "b": rbv}), on="k", how="left")
+    np.testing.assert_array_equal(mout["k"].to_numpy(), pout["k"].to_numpy())
+    cases["mgl_huge"] = {
+        "in_lk": lkh.astype(np.int64), "in_rk": rkh.astype(np.int64),
+        "in_rb": rbv,
+        "out_k": pout["k"].to_numpy().astype(np.int64),
+        "out_b": pout["b"].to_numpy(),
+    }
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -843,6 +925,7 @@ def main():
     all_cases.update(gen_merge2_cases(mpd, rng))
     all_cases.update(gen_float_key_cases(mpd, rng))
     all_cases.update(gen_median_cases(mpd, rng))
+    all_cases.update(gen_left_merge_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)

@@ -883,3 +883,39 @@ def test_series_mask_and_dropna(npartitions):
     ed = ps.dropna()
     np.testing.assert_array_equal(gd.index.to_numpy(), ed.index.to_numpy())
     np.testing.assert_array_equal(gd.to_numpy(), ed.to_numpy())
+
+
+def test_left_merge_vs_golden(npartitions):
+    """merge(how='left'): unmatched lefts keep NaN rights, int right
+    columns become float64 iff NaNs were introduced, pandas left row
+    order; huge-span (densify) keys too."""
+    g = load_golden("mgl_basic")
+    left = mpd.DataFrame({"k": g["in_lk"], "a": g["in_la"]})
+    right = mpd.DataFrame({"k": g["in_rk"], "b": g["in_rb"],
+                           "i": g["in_ri"]})
+    out = left.merge(right, on="k", how="left").to_pandas()
+    np.testing.assert_array_equal(out["k"].to_numpy(), g["out_k"])
+    np.testing.assert_allclose(out["a"].to_numpy(), g["out_a"], rtol=0)
+    np.testing.assert_allclose(out["b"].to_numpy(), g["out_b"], rtol=0,
+                               equal_nan=True)
+    assert out["i"].dtype == np.float64  # NaNs introduced
+    np.testing.assert_allclose(out["i"].to_numpy(), g["out_i"], rtol=0,
+                               equal_nan=True)
+
+    g2 = load_golden("mgl_allmatch")
+    l2 = mpd.DataFrame({"k": g2["in_lk"],
+                        "a": np.zeros(len(g2["in_lk"]))})
+    r2 = mpd.DataFrame({"k": g2["in_rk"], "i": g2["in_ri"]})
+    o2 = l2.merge(r2, on="k", how="left").to_pandas()
+    assert o2["i"].dtype == np.int64  # all matched: int stays int
+    np.testing.assert_array_equal(o2["k"].to_numpy(), g2["out_k"])
+    np.testing.assert_array_equal(o2["i"].to_numpy(), g2["out_i"])
+
+    g4 = load_golden("mgl_huge")
+    l4 = mpd.DataFrame({"k": g4["in_lk"],
+                        "a": np.zeros(len(g4["in_lk"]))})
+    r4 = mpd.DataFrame({"k": g4["in_rk"], "b": g4["in_rb"]})
+    o4 = l4.merge(r4, on="k", how="left").to_pandas()
+    np.testing.assert_array_equal(o4["k"].to_numpy(), g4["out_k"])
+    np.testing.assert_allclose(o4["b"].to_numpy(), g4["out_b"], rtol=0,
+                               equal_nan=True)

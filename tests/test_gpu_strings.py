@@ -312,3 +312,16 @@ def test_drop_nunique_series_sort(npartitions):
     ps = pdf["w"].sort_values(kind="stable")
     np.testing.assert_array_equal(ss.to_numpy(), ps.to_numpy())
     np.testing.assert_array_equal(ss.index.to_numpy(), ps.index.to_numpy())
+
+
+def test_left_merge_string_vs_golden(npartitions):
+    """Left merge on STRING keys with a string payload column: unmatched
+    lefts get NaN payload (dict code −1)."""
+    g = load_golden("mgl_str")
+    left = mpd.DataFrame({"s": dec(g["in_ls"]),
+                          "a": np.zeros(len(g["in_ls"]))})
+    right = mpd.DataFrame({"s": dec(g["in_rs"]), "t": dec(g["in_rt"])})
+    out = left.merge(right, on="s", how="left").to_pandas()
+    assert_str_equal(out["s"].to_numpy(), g["out_s"], "left key")
+    assert_str_equal(out["t"].to_numpy(), g["out_t"], "left payload")
+    np.testing.assert_array_equal(out.index.to_numpy(), g["out_idx"])
