@@ -754,6 +754,94 @@ class HipDataframe:
         return HipDataframe([part], cnt_res._index, val_names, [ng],
                             dtypes)
 
+    def groupby_firstlast(self, by, last: bool = False) -> "HipDataframe":
+        """groupby().first()/last(): per value column, the value at the
+        min/max ORIGINAL ROW POSITION among that column's non-NaN rows of
+        each group (pandas skips NaN) — one groupby-min/max of a position
+        column per value column, then a device gather; groups with no
+        non-NaN value fill NaN (float) / code −1 (dict)."""
+        from ..distributed import is_active
+        if is_active():
+            raise lib.HfError(
+                "distributed groupby.first/last is a later round")
+        if isinstance(by, (list, tuple)):
+            if len(by) == 1:
+                by = by[0]
+            else:
+                cf, decode = self._combined_key_frame(list(by))
+                keep = [c for c in cf.columns if c not in by]
+                res = cf.take_columns(keep).groupby_firstlast(self.KEYCOL,
+                                                              last)
+                res._index = decode(lib.get(res._index.col))
+                return res
+        val_names = [c for c in self.columns if c != by]
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        POS = "\x00pos\x00"
+        res0 = self.groupby_size(by)
+        gkeys = res0.index
+        ng = len(gkeys)
+        out_cols, dtypes, out_cats = {}, {}, {}
+        for v in val_names:
+            parts2 = []
+            base = 0
+            vcols = []
+            for p, ln in zip(self._partitions, self._row_lengths):
+                block = p.block()
+                kcol = block.columns[by]
+                vcol = block.columns[v]
+                vcols.append(vcol)
+                # keep rows whose key AND value are usable: float-NaN
+                # keys (dropped groups) and NaN values (pandas first/last
+                # skip NaN) both filter out; positions stay global
+                ones = lib.compare_scalar(lib.CMP_NOTNA, kcol, 0.0)
+                if v in blk_cats:
+                    keep = lib.compare_scalar(lib.CMP_NE, vcol, -1.0)
+                elif vcol.dtype_code == lib.HF_FLOAT64:
+                    keep = lib.compare_scalar(lib.CMP_NOTNA, vcol, 0.0)
+                else:
+                    keep = None
+                mask = (ones if keep is None
+                        else lib.binary(lib.BIN_MUL, ones, keep))
+                plan0 = lib.filter_plan(mask)
+                pos = lib.filter_iota(plan0, base)
+                cols = {by: lib.filter_apply(plan0, kcol), POS: pos}
+                parts2.append(HipDataframePartition(
+                    DeviceBlock(cols, cols[by].length, block.cats)))
+                base += ln
+            sub = HipDataframe(
+                parts2, pandas.RangeIndex(base), [by, POS],
+                [pp.block().length for pp in parts2] if parts2 else [0],
+                pandas.Series({by: self.dtypes[by],
+                               POS: np.dtype(np.int64)}))
+            posres = sub.groupby_reduce(by, "max" if last else "min")
+            pos_np = lib.get(posres._partitions[0].block().columns[POS])
+            skeys = posres.index
+            vcat = vcols[0] if len(vcols) == 1 else lib.concat(vcols)
+            vals_at = lib.get(lib.gather(vcat,
+                                         lib.put(pos_np.astype(np.int64)))) \
+                if len(pos_np) else np.empty(0, dtype=vcat.np_dtype)
+            idx = gkeys.get_indexer(skeys)
+            if v in blk_cats:
+                out = np.full(ng, -1, dtype=np.int64)
+                out[idx] = vals_at
+                dtypes[v] = np.dtype(object)
+                out_cats[v] = blk_cats[v]
+            elif vcat.dtype_code == lib.HF_INT64:
+                out = np.empty(ng, dtype=np.int64)
+                out[:] = 0
+                out[idx] = vals_at  # int columns: every group has a value
+                dtypes[v] = np.dtype(np.int64)
+            else:
+                out = np.full(ng, np.nan)
+                out[idx] = vals_at
+                dtypes[v] = np.dtype(np.float64)
+            out_cols[v] = lib.put(out)
+        part = HipDataframePartition(DeviceBlock(out_cols, ng,
+                                                 out_cats))
+        return HipDataframe([part], res0._index, val_names, [ng],
+                            pandas.Series(dtypes))
+
     def hconcat(self, others: list) -> "HipDataframe":
         """Horizontal compose of single-partition frames with identical row
         count and index (the device form of the reference's axis=1 concat

@@ -543,6 +543,12 @@ class DataFrameGroupBy:
     def median(self):
         return self._agg("median")
 
+    def first(self):
+        return self._agg("first")
+
+    def last(self):
+        return self._agg("last")
+
     def size(self):
         """pandas DataFrameGroupBy.size(): a Series of group row counts
         (NaN values included, NaN keys dropped)."""
@@ -552,7 +558,7 @@ class DataFrameGroupBy:
         return out["size"].rename(None)
 
     _AGGS = ("sum", "count", "mean", "min", "max", "var", "std",
-             "median")
+             "median", "first", "last")
 
     def agg(self, how):
         """str, list-of-str (MultiIndex columns, pandas col-major order) or

@@ -130,6 +130,14 @@ class HipQueryCompiler:
     def groupby_median(self, by) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.groupby_median(by))
 
+    def groupby_first(self, by) -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.groupby_firstlast(by, last=False))
+
+    def groupby_last(self, by) -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.groupby_firstlast(by, last=True))
+
     def groupby_size(self, by: str) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.groupby_size(by))
 
@@ -167,6 +175,8 @@ class HipQueryCompiler:
             "var": type(self).groupby_var,
             "std": type(self).groupby_std,
             "median": type(self).groupby_median,
+            "first": type(self).groupby_first,
+            "last": type(self).groupby_last,
         }.get(agg)
         if fn is None:
             raise lib.HfError(

@@ -214,7 +214,7 @@ def test_merge_error_surfaces():
     ok = mpd.DataFrame({"k": rng.integers(0, 5, 10).astype(np.int64),
                         "v": rng.random(10)})
     with pytest.raises(lib.HfError, match="not implemented"):
-        ok.merge(right, on="k", how="left")
+        ok.merge(right, on="k", how="outer")
 
 
 def test_filter_vs_golden(npartitions):
