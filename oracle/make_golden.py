@@ -905,6 +905,40 @@ def gen_left_merge_cases(mpd, rng):
     return cases
 
 
+def gen_firstlast_cases(mpd, rng):
+    """groupby.first/last (NaN skipped; all-NaN groups -> NaN; string
+    values and keys; int dtype preserved) vs the reference."""
+    import pandas
+    cases = {}
+    n = 3000
+    k = rng.integers(0, 30, n).astype(np.int64)
+    v = rng.random(n)
+    v[rng.random(n) < 0.15] = np.nan
+    v[k == 5] = np.nan  # all-NaN group
+    w = rng.integers(-99, 99, n).astype(np.int64)
+    pool = np.array(["aa", "bb", "cc", "dd"])
+    sv = rng.choice(pool, n).astype(object)
+    sv[rng.random(n) < 0.1] = np.nan
+    mdf = mpd.DataFrame({"k": k, "v": v, "w": w, "s": sv})
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w, "s": sv})
+    arrays = {"in_k": k, "in_v": v, "in_w": w, "in_s": _enc_str(sv)}
+    for agg in ("first", "last"):
+        mres = getattr(mdf.groupby("k"), agg)()._to_pandas()
+        pres = getattr(pdf.groupby("k"), agg)()
+        assert list(mres.index) == list(pres.index)
+        assert list(mres["w"]) == list(pres["w"])
+        assert list(mres["s"].fillna(NA)) == list(pres["s"].fillna(NA))
+        np.testing.assert_allclose(mres["v"].to_numpy(),
+                                   pres["v"].to_numpy(), rtol=0,
+                                   equal_nan=True)
+        arrays[f"out_{agg}_keys"] = pres.index.to_numpy().astype(np.int64)
+        arrays[f"out_{agg}_v"] = pres["v"].to_numpy()
+        arrays[f"out_{agg}_w"] = pres["w"].to_numpy().astype(np.int64)
+        arrays[f"out_{agg}_s"] = _enc_str(pres["s"])
+    cases["gbfl_cases"] = arrays
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -926,6 +960,7 @@ def main():
     all_cases.update(gen_float_key_cases(mpd, rng))
     all_cases.update(gen_median_cases(mpd, rng))
     all_cases.update(gen_left_merge_cases(mpd, rng))
+    all_cases.update(gen_firstlast_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)

@@ -325,3 +325,23 @@ def test_left_merge_string_vs_golden(npartitions):
     assert_str_equal(out["s"].to_numpy(), g["out_s"], "left key")
     assert_str_equal(out["t"].to_numpy(), g["out_t"], "left payload")
     np.testing.assert_array_equal(out.index.to_numpy(), g["out_idx"])
+
+
+def test_groupby_first_last_vs_golden(npartitions):
+    """groupby.first/last: value at min/max original position among each
+    group's non-NaN rows — float (all-NaN group -> NaN), int (dtype
+    preserved) and STRING value columns, vs the reference."""
+    g = load_golden("gbfl_cases")
+    df = mpd.DataFrame({"k": g["in_k"], "v": g["in_v"], "w": g["in_w"],
+                        "s": dec(g["in_s"])})
+    for agg in ("first", "last"):
+        out = getattr(df.groupby("k"), agg)().to_pandas()
+        np.testing.assert_array_equal(out.index.to_numpy(),
+                                      g[f"out_{agg}_keys"])
+        np.testing.assert_allclose(out["v"].to_numpy(), g[f"out_{agg}_v"],
+                                   rtol=0, equal_nan=True, err_msg=agg)
+        assert out["w"].dtype == np.int64
+        np.testing.assert_array_equal(out["w"].to_numpy(),
+                                      g[f"out_{agg}_w"])
+        assert_str_equal(out["s"].to_numpy(), g[f"out_{agg}_s"],
+                         f"{agg} s")
