@@ -655,14 +655,21 @@ class HipDataframe:
                 "nan_firstpos": nan_first}
 
     def median_columns(self):
-        """Per-column median (NaN skipped): NOTNA filter -> ordered radix
-        sort -> middle element(s) sliced on device; pandas nanmedian."""
+        """Per-column median (NaN skipped): quantile_columns([0.5])."""
+        return {name: v[0]
+                for name, v in self.quantile_columns([0.5]).items()}
+
+    def quantile_columns(self, qs):
+        """Per-column quantiles (linear interpolation, NaN skipped):
+        NOTNA filter -> ordered radix sort -> the two bracketing elements
+        per q sliced on device (pandas nanquantile)."""
         blk_cats = (self._partitions[0].block().cats
                     if self._partitions else {})
         out = {}
         for name in self.columns:
             if name in blk_cats:
-                raise lib.HfError(f"median over string column {name!r}")
+                raise lib.HfError(
+                    f"median/quantile over string column {name!r}")
             cols = [p.block().columns[name] for p in self._partitions]
             col = cols[0] if len(cols) == 1 else lib.concat(cols)
             if col.dtype_code == lib.HF_FLOAT64 and col.length:
@@ -672,7 +679,7 @@ class HipDataframe:
                     col = lib.filter_apply(plan, col)
             n = col.length
             if n == 0:
-                out[name] = float("nan")
+                out[name] = [float("nan")] * len(qs)
                 continue
             key = (lib.ordered_i64(col)
                    if col.dtype_code == lib.HF_FLOAT64 else col)
@@ -680,9 +687,15 @@ class HipDataframe:
             sv = lib.gather(lib.cast_f64(col)
                             if col.dtype_code == lib.HF_INT64 else col,
                             perm)
-            lo, hi = (n - 1) // 2, n // 2
-            mid = lib.get(lib.col_slice(sv, lo, hi - lo + 1))
-            out[name] = float(mid.mean())
+            vals = []
+            for q in qs:
+                pos = (n - 1) * float(q)
+                lo = int(np.floor(pos))
+                hi = int(np.ceil(pos))
+                pair = lib.get(lib.col_slice(sv, lo, hi - lo + 1))
+                vals.append(float(pair[0] + (pair[-1] - pair[0])
+                                  * (pos - lo)))
+            out[name] = vals
         return out
 
     def groupby_median(self, by) -> "HipDataframe":

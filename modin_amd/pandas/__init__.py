@@ -72,6 +72,21 @@ class _HipPandasBase:
     def median(self, **kwargs):
         return self._lower(self._query_compiler.median())
 
+    def isna(self):
+        """Boolean NaN mask (inverse of notna)."""
+        qc = self._query_compiler
+        notna = qc.notna()
+        one = type(notna).mul(notna, -1)
+        inv = type(one).add(one, 1)  # 1 - notna
+        out = self._rewrap(inv)
+        out._bool_mask = True
+        return out
+
+    def notna(self):
+        out = self._rewrap(self._query_compiler.notna())
+        out._bool_mask = True
+        return out
+
     def var(self, ddof: int = 1):
         return self._lower(self._query_compiler.var(ddof=ddof))
 
@@ -266,6 +281,30 @@ class DataFrame(_HipPandasBase):
 
     def astype(self, dtype):
         return DataFrame(query_compiler=self._query_compiler.astype(dtype))
+
+    def quantile(self, q=0.5):
+        """Per-column quantiles (linear interpolation, NaN skipped)."""
+        qs = [q] if np.isscalar(q) else list(q)
+        out = self._query_compiler.quantile(qs)
+        if np.isscalar(q):
+            return out.iloc[0]
+        return out
+
+    def describe(self):
+        """count/mean/std/min/25%/50%/75%/max per numeric column (pandas
+        describe), composed from the one-pass reduce and the sorted
+        quantile machinery."""
+        qc = self._query_compiler
+        cnt = qc.count()
+        mean = qc.mean()
+        std = qc.std()
+        mn = qc.min()
+        mx = qc.max()
+        qs = qc.quantile([0.25, 0.5, 0.75])
+        rows = {"count": cnt.astype(float), "mean": mean, "std": std,
+                "min": mn, "25%": qs.iloc[0], "50%": qs.iloc[1],
+                "75%": qs.iloc[2], "max": mx}
+        return pandas.DataFrame(rows).T[list(self.columns)]
 
     def drop(self, columns=None):
         """pandas DataFrame.drop(columns=...): metadata-only column

@@ -127,6 +127,10 @@ class HipQueryCompiler:
         vals = self._modin_frame.median_columns()
         return pandas.Series(vals, dtype=np.float64)
 
+    def quantile(self, qs):
+        vals = self._modin_frame.quantile_columns(list(qs))
+        return pandas.DataFrame(vals, index=pandas.Index(list(qs)))
+
     def groupby_median(self, by) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.groupby_median(by))
 

@@ -919,3 +919,29 @@ def test_left_merge_vs_golden(npartitions):
     np.testing.assert_array_equal(o4["k"].to_numpy(), g4["out_k"])
     np.testing.assert_allclose(o4["b"].to_numpy(), g4["out_b"], rtol=0,
                                equal_nan=True)
+
+
+def test_quantile_describe_vs_pandas(npartitions):
+    """df.quantile (scalar and list q, linear interpolation, NaN skipped)
+    and df.describe vs pandas inline (pandas is the reference's own
+    arbiter — SURVEY §8c)."""
+    rng = np.random.default_rng(87)
+    n = 30_000
+    v = rng.standard_normal(n) * 7
+    v[rng.random(n) < 0.1] = np.nan
+    w = rng.integers(-1000, 1000, n)
+    pdf = pandas.DataFrame({"v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    for q in (0.5, 0.25, 0.99, 0.0, 1.0):
+        got = df.quantile(q)
+        exp = pdf.quantile(q)
+        np.testing.assert_allclose(np.asarray(got), exp.to_numpy(),
+                                   rtol=RTOL, err_msg=str(q))
+    got = df.quantile([0.1, 0.5, 0.9])
+    exp = pdf.quantile([0.1, 0.5, 0.9])
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(), rtol=RTOL)
+    gd = df.describe()
+    ed = pdf.describe()
+    assert list(gd.index) == list(ed.index)
+    assert list(gd.columns) == list(ed.columns)
+    np.testing.assert_allclose(gd.to_numpy(), ed.to_numpy(), rtol=1e-9)
