@@ -885,10 +885,10 @@ class HipDataframe:
                        how: str = "inner") -> "HipDataframe":
         if on not in self.columns or on not in other.columns:
             raise lib.HfError(f"merge: key column {on!r} missing")
-        if how not in ("inner", "left"):
+        if how not in ("inner", "left", "outer"):
             raise lib.HfError(
-                f"merge how={how!r} not implemented (inner/left broadcast "
-                "join this round)")
+                f"merge how={how!r} not implemented (inner/left/outer/"
+                "right this round)")
         left_names = [c for c in self.columns if c != on]
         right_names = [c for c in other.columns if c != on]
         common = set(left_names) & set(right_names)
@@ -981,7 +981,7 @@ class HipDataframe:
             other._join_build_cache = (cache_key, j, uniq)
 
         runiq = uniq  # probe-space distinct right keys (left-join test)
-        if how == "left" and runiq is None:
+        if how in ("left", "outer") and runiq is None:
             if rkeys.length:
                 rperm = lib.sort_perm(rkeys)
                 runiq, _ru, _rc2, _rn = lib.groupby_sorted(
@@ -992,7 +992,7 @@ class HipDataframe:
         # int64 right columns into float64 — decided GLOBALLY first
         total_unmatched = 0
         un_plans = []
-        if how == "left":
+        if how in ("left", "outer"):
             for p in self._partitions:
                 block = p.block()
                 lk = block.columns[on]
@@ -1013,12 +1013,12 @@ class HipDataframe:
             if lkeys.dtype_code != lib.HF_INT64:
                 raise lib.HfError("merge: key column must be int64")
             if uniq is not None:  # code space: unmatched lefts become -1
-                lkeys = un_plans[pi][1] if how == "left" \
+                lkeys = un_plans[pi][1] if how in ("left", "outer") \
                     else lib.search_sorted(lkeys, uniq)
             keys_c, lidx, rcols, nout = lib.join_probe(j, lkeys)
             if uniq is not None:  # decode output codes back to key values
                 keys_c = lib.gather(uniq, keys_c)
-            if how == "left":
+            if how in ("left", "outer"):
                 plan = un_plans[pi][0]
                 n_un = plan.n_kept
                 if n_un:
@@ -1066,15 +1066,64 @@ class HipDataframe:
                 cols[rout[rn]] = rcols[i]
             out_parts.append(HipDataframePartition(DeviceBlock(cols, nout)))
             lengths.append(nout)
+        n_run = 0
+        if how == "outer":
+            # unmatched RIGHT rows: keys absent from the left key set
+            luniq = None
+            lk_all = concat_col(self, on)
+            if uniq is not None:
+                lk_all = lib.search_sorted(lk_all, uniq)
+            if lk_all.length:
+                lperm = lib.sort_perm(lk_all)
+                luniq, _l1, _l2, _l3 = lib.groupby_sorted(
+                    lib.gather(lk_all, lperm), [], lib.AGG_SUM, False)
+            else:
+                luniq = lib.put(np.empty(0, dtype=np.int64))
+            rk_probe = rkeys  # possibly code space already
+            rm = lib.compare_scalar(lib.CMP_EQ,
+                                    lib.search_sorted(rk_probe, luniq),
+                                    -1.0)
+            rplan = lib.filter_plan(rm)
+            n_run = rplan.n_kept
+            if n_run:
+                cols = {}
+                rk_out = lib.filter_apply(rplan, rkeys)
+                if uniq is not None:
+                    rk_out = lib.gather(uniq, rk_out)
+                for name in self.columns:
+                    key = on if name == on else lout[name]
+                    if name == on:
+                        cols[key] = rk_out
+                    elif name in lcats:
+                        c2 = lib.alloc(n_run, lib.HF_INT64)
+                        lib.fill_i64(c2.dptr(), -1, n_run)
+                        cols[key] = c2
+                    else:
+                        c2 = lib.alloc(n_run, lib.HF_FLOAT64)
+                        lib.fill_f64(c2.dptr(), float("nan"), n_run)
+                        cols[key] = c2
+                for i, rn in enumerate(right_names):
+                    rc = lib.filter_apply(rplan, rvals[i])
+                    if rn not in rcats and rc.dtype_code == lib.HF_INT64 \
+                            and total_unmatched:
+                        rc = lib.cast_f64(rc)
+                    cols[rout[rn]] = rc
+                out_parts.append(HipDataframePartition(
+                    DeviceBlock(cols, n_run)))
+                lengths.append(n_run)
         out_columns = ([on if c == on else lout[c] for c in self.columns]
                        + [rout[c] for c in right_names])
         dtypes = {}
         for c in self.columns:
-            dtypes[on if c == on else lout[c]] = self.dtypes[c]
+            d = self.dtypes[c]
+            if (how == "outer" and n_run and c != on and c not in lcats
+                    and d == np.dtype(np.int64)):
+                d = np.dtype(np.float64)
+            dtypes[on if c == on else lout[c]] = d
         for c in right_names:
             d = other.dtypes[c]
-            if (how == "left" and total_unmatched and c not in rcats
-                    and d == np.dtype(np.int64)):
+            if (how in ("left", "outer") and total_unmatched
+                    and c not in rcats and d == np.dtype(np.int64)):
                 d = np.dtype(np.float64)
             dtypes[rout[c]] = d
         out_cats = {}
@@ -1090,8 +1139,14 @@ class HipDataframe:
             for part in out_parts:
                 part._block.cats = dict(out_cats)
         total = sum(lengths)
-        return HipDataframe(out_parts, pandas.RangeIndex(total), out_columns,
-                            lengths, pandas.Series(dtypes))
+        res = HipDataframe(out_parts, pandas.RangeIndex(total), out_columns,
+                           lengths, pandas.Series(dtypes))
+        if how == "outer":
+            # pandas sorts outer-join keys; stable sort keeps the
+            # within-key left-order/right-match expansion order
+            res = res.sort_rows(on, True)
+            res._index = pandas.RangeIndex(total)
+        return res
 
     # ---- row filter (PandasDataframe.filter / mask device form,
     #      partition.py:224) ----

@@ -247,6 +247,36 @@ class HipQueryCompiler:
     #      storage_formats/pandas/merge.py:104) ----
     def merge(self, right: "HipQueryCompiler", on: str,
               how: str = "inner") -> "HipQueryCompiler":
+        if how == "right":
+            # pandas right join == swapped left join with the suffix roles
+            # flipped back and columns restored to left-then-right order
+            swapped = right.merge(self, on=on, how="left")
+            frame = swapped._modin_frame
+            lcols = [c for c in self.columns if c != on]
+            rcols = [c for c in right.columns if c != on]
+            common = set(lcols) & set(rcols)
+            # in the swapped join, OUR columns got "_y" and right's "_x"
+            ren = {}
+            for c in lcols:
+                if c in common:
+                    ren[c + "_y"] = c + "_x"
+            for c in rcols:
+                if c in common:
+                    ren[c + "_x"] = c + "_y"
+            # two-step rename through temporaries to avoid collisions
+            tmp = {k: ("\x00tmp\x00" + k) for k in ren}
+            out = swapped.rename_columns(tmp) if ren else swapped
+            if ren:
+                out = out.rename_columns(
+                    {("\x00tmp\x00" + k): v for k, v in ren.items()})
+            # pandas puts the key at its left-frame position; ours keeps
+            # the left column order with the key in place
+            left_order = [on if c == on else
+                          (c + "_x" if c in common else c)
+                          for c in self.columns]
+            order = left_order + [(c + "_y" if c in common else c)
+                                  for c in rcols]
+            return out.getitem_column_array(order)
         return self.__constructor__(
             self._modin_frame.broadcast_join(right._modin_frame, on, how)
         )

@@ -939,6 +939,43 @@ def gen_firstlast_cases(mpd, rng):
     return cases
 
 
+def gen_outer_right_merge_cases(mpd, rng):
+    """merge how='outer' (left rows then unmatched rights) and how='right'
+    vs the reference, including common-column suffixing."""
+    import pandas
+    cases = {}
+    nl, nr = 2500, 900
+    lk = rng.integers(0, 1500, nl).astype(np.int64)
+    rk = rng.integers(500, 2500, nr).astype(np.int64)  # both-side unmatched
+    la = rng.random(nl)
+    rb = rng.random(nr)
+    lw = rng.integers(-9, 9, nl).astype(np.int64)
+    rw = rng.integers(100, 200, nr).astype(np.int64)  # suffixed common col
+    for how in ("outer", "right"):
+        mout = mpd.DataFrame({"k": lk, "a": la, "w": lw}).merge(
+            mpd.DataFrame({"k": rk, "b": rb, "w": rw}), on="k",
+            how=how)._to_pandas()
+        pout = pandas.DataFrame({"k": lk, "a": la, "w": lw}).merge(
+            pandas.DataFrame({"k": rk, "b": rb, "w": rw}), on="k", how=how)
+        assert list(mout.columns) == list(pout.columns)
+        np.testing.assert_array_equal(mout["k"].to_numpy(),
+                                      pout["k"].to_numpy())
+        np.testing.assert_allclose(
+            mout[["a", "w_x", "b", "w_y"]].values,
+            pout[["a", "w_x", "b", "w_y"]].values, rtol=0, equal_nan=True)
+        cases[f"mg_{how}"] = {
+            "in_lk": lk, "in_la": la, "in_lw": lw,
+            "in_rk": rk, "in_rb": rb, "in_rw": rw,
+            "out_k": pout["k"].to_numpy().astype(np.int64),
+            "out_a": pout["a"].to_numpy(),
+            "out_wx": pout["w_x"].to_numpy(),
+            "out_b": pout["b"].to_numpy(),
+            "out_wy": pout["w_y"].to_numpy(),
+            "out_cols": np.array(list(pout.columns)),
+        }
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -961,6 +998,7 @@ def main():
     all_cases.update(gen_median_cases(mpd, rng))
     all_cases.update(gen_left_merge_cases(mpd, rng))
     all_cases.update(gen_firstlast_cases(mpd, rng))
+    all_cases.update(gen_outer_right_merge_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)

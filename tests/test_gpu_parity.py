@@ -945,3 +945,24 @@ def test_quantile_describe_vs_pandas(npartitions):
     assert list(gd.index) == list(ed.index)
     assert list(gd.columns) == list(ed.columns)
     np.testing.assert_allclose(gd.to_numpy(), ed.to_numpy(), rtol=1e-9)
+
+
+def test_outer_right_merge_vs_golden(npartitions):
+    """merge how='outer' (key-sorted union, NaN fills both directions)
+    and how='right' (right row order, suffix roles preserved) vs the
+    reference."""
+    for how in ("outer", "right"):
+        g = load_golden(f"mg_{how}")
+        left = mpd.DataFrame({"k": g["in_lk"], "a": g["in_la"],
+                              "w": g["in_lw"]})
+        right = mpd.DataFrame({"k": g["in_rk"], "b": g["in_rb"],
+                               "w": g["in_rw"]})
+        out = left.merge(right, on="k", how=how).to_pandas()
+        assert list(out.columns) == list(g["out_cols"]), how
+        np.testing.assert_array_equal(out["k"].to_numpy(), g["out_k"],
+                                      err_msg=how)
+        for cn, gk in [("a", "out_a"), ("w_x", "out_wx"),
+                       ("b", "out_b"), ("w_y", "out_wy")]:
+            np.testing.assert_allclose(out[cn].to_numpy(), g[gk],
+                                       rtol=0, equal_nan=True,
+                                       err_msg=f"{how}/{cn}")
