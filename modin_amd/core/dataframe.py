@@ -855,6 +855,95 @@ class HipDataframe:
         return HipDataframe([part], res0._index, val_names, [ng],
                             pandas.Series(dtypes))
 
+    def shift_rows(self, periods: int) -> "HipDataframe":
+        """pandas shift(axis=0): device slice + NaN block concat per
+        column (int columns become float64, the pandas rule).  Operates on
+        the frame as one sequence (partitions concatenated)."""
+        if periods == 0:
+            return self
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        if blk_cats:
+            raise lib.HfError("shift over string columns is a later round")
+        n = len(self)
+        k = min(abs(periods), n)
+
+        def concat_col(name):
+            cols = [p.block().columns[name] for p in self._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        out_cols, dtypes = {}, {}
+        for name in self.columns:
+            col = concat_col(name)
+            if col.dtype_code == lib.HF_INT64:
+                col = lib.cast_f64(col)
+            nanb = lib.alloc(k, lib.HF_FLOAT64)
+            if k:
+                lib.fill_f64(nanb.dptr(), float("nan"), k)
+            if periods > 0:
+                kept = lib.col_slice(col, 0, n - k)
+                out_cols[name] = lib.concat([nanb, kept])
+            else:
+                kept = lib.col_slice(col, k, n - k)
+                out_cols[name] = lib.concat([kept, nanb])
+            dtypes[name] = np.dtype(np.float64)
+        part = HipDataframePartition(DeviceBlock(out_cols, n))
+        return HipDataframe([part], self._index, self.columns, [n],
+                            pandas.Series(dtypes))
+
+    def diff_rows(self, periods: int) -> "HipDataframe":
+        """pandas diff(axis=0): x − x.shift(periods), composed on the
+        concatenated columns (both operands single-partition aligned)."""
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        if blk_cats:
+            raise lib.HfError("diff over string columns is a later round")
+        shifted = self.shift_rows(periods)
+        n = len(self)
+        out_cols = {}
+        for name in self.columns:
+            cols = [p.block().columns[name] for p in self._partitions]
+            col = cols[0] if len(cols) == 1 else lib.concat(cols)
+            if col.dtype_code == lib.HF_INT64:
+                col = lib.cast_f64(col)
+            sh = shifted._partitions[0].block().columns[name]
+            out_cols[name] = lib.binary(lib.BIN_SUB, col, sh)
+        part = HipDataframePartition(DeviceBlock(out_cols, n))
+        dtypes = pandas.Series({c: np.dtype(np.float64)
+                                for c in self.columns})
+        return HipDataframe([part], self._index, self.columns, [n],
+                            dtypes)
+
+    def idx_extreme(self, maximum: bool) -> dict:
+        """Per-column idxmax/idxmin: the FIRST original position holding
+        the column's max/min (NaN skipped) — one cached reduce + an EQ
+        filter + a position min."""
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        bad = [c for c in self.columns if c in blk_cats]
+        if bad:
+            raise lib.HfError(f"idxmax/idxmin over string column(s) {bad}")
+        out = {}
+        for name in self.columns:
+            cols = [p.block().columns[name] for p in self._partitions]
+            col = cols[0] if len(cols) == 1 else lib.concat(cols)
+            if not col.length:
+                out[name] = float("nan")
+                continue
+            r = lib.reduce(col)
+            if r.count == 0:
+                out[name] = float("nan")
+                continue
+            if col.dtype_code == lib.HF_INT64:
+                target = float(r.imx if maximum else r.imn)
+            else:
+                target = r.mx if maximum else r.mn
+            m = lib.compare_scalar(lib.CMP_EQ, col, target)
+            plan = lib.filter_plan(m)
+            pos = lib.filter_iota(plan, 0)
+            out[name] = int(lib.reduce(pos).imn)
+        return out
+
     def hconcat(self, others: list) -> "HipDataframe":
         """Horizontal compose of single-partition frames with identical row
         count and index (the device form of the reference's axis=1 concat

@@ -90,6 +90,19 @@ class _HipPandasBase:
     def var(self, ddof: int = 1):
         return self._lower(self._query_compiler.var(ddof=ddof))
 
+    def shift(self, periods: int = 1):
+        return self._rewrap(self._query_compiler.shift(int(periods)))
+
+    def diff(self, periods: int = 1):
+        """x - x.shift(periods) (pandas diff)."""
+        return self._rewrap(self._query_compiler.diff(int(periods)))
+
+    def idxmax(self):
+        return self._lower(self._query_compiler.idxmax())
+
+    def idxmin(self):
+        return self._lower(self._query_compiler.idxmin())
+
     def std(self, ddof: int = 1):
         return self._lower(self._query_compiler.std(ddof=ddof))
 

@@ -127,6 +127,20 @@ class HipQueryCompiler:
         vals = self._modin_frame.median_columns()
         return pandas.Series(vals, dtype=np.float64)
 
+    def shift(self, periods: int) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.shift_rows(periods))
+
+    def diff(self, periods: int) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.diff_rows(periods))
+
+    def idxmax(self):
+        vals = self._modin_frame.idx_extreme(True)
+        return pandas.Series(vals)
+
+    def idxmin(self):
+        vals = self._modin_frame.idx_extreme(False)
+        return pandas.Series(vals)
+
     def quantile(self, qs):
         vals = self._modin_frame.quantile_columns(list(qs))
         return pandas.DataFrame(vals, index=pandas.Index(list(qs)))

@@ -966,3 +966,33 @@ def test_outer_right_merge_vs_golden(npartitions):
             np.testing.assert_allclose(out[cn].to_numpy(), g[gk],
                                        rtol=0, equal_nan=True,
                                        err_msg=f"{how}/{cn}")
+
+
+def test_shift_diff_idx_vs_pandas(npartitions):
+    rng = np.random.default_rng(88)
+    n = 20_000
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.05] = np.nan
+    w = rng.integers(-100, 100, n)
+    pdf = pandas.DataFrame({"v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    for p_ in (1, 3, -2, 0, n + 5):
+        got = df.shift(p_).to_pandas()
+        exp = pdf.shift(p_)
+        np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(),
+                                   rtol=0, equal_nan=True,
+                                   err_msg=f"shift {p_}")
+        gd = df.diff(p_).to_pandas()
+        ed = pdf.diff(p_)
+        np.testing.assert_allclose(gd.to_numpy(), ed.to_numpy(),
+                                   rtol=0, equal_nan=True,
+                                   err_msg=f"diff {p_}")
+    im = df.idxmax()
+    em = pdf.idxmax()
+    np.testing.assert_array_equal(np.asarray(im), em.to_numpy())
+    ii = df.idxmin()
+    ei = pdf.idxmin()
+    np.testing.assert_array_equal(np.asarray(ii), ei.to_numpy())
+    sv = df["v"]
+    assert sv.idxmax() == pdf["v"].idxmax()
+    assert sv.idxmin() == pdf["v"].idxmin()
