@@ -1227,6 +1227,21 @@ class HipDataframe:
         if out_cats:
             for part in out_parts:
                 part._block.cats = dict(out_cats)
+        # align device dtypes to the declared result dtypes: partitions
+        # with no NaN fills still hold int64 where the frame-wide rule
+        # promoted to float64 (left/outer joins) — cast so device-side
+        # concat/sort stay type-uniform
+        for part in out_parts:
+            block = part._block
+            changed = False
+            cols2 = dict(block.columns)
+            for cn, c in cols2.items():
+                if (dtypes.get(cn) == np.dtype(np.float64)
+                        and c.dtype_code == lib.HF_INT64):
+                    cols2[cn] = lib.cast_f64(c)
+                    changed = True
+            if changed:
+                part._block = DeviceBlock(cols2, block.length, block.cats)
         total = sum(lengths)
         res = HipDataframe(out_parts, pandas.RangeIndex(total), out_columns,
                            lengths, pandas.Series(dtypes))

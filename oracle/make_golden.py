@@ -963,7 +963,7 @@ def gen_outer_right_merge_cases(mpd, rng):
         np.testing.assert_allclose(
             mout[["a", "w_x", "b", "w_y"]].values,
             pout[["a", "w_x", "b", "w_y"]].values, rtol=0, equal_nan=True)
-        cases[f"mg_{how}"] = {
+        cases[f"mgo_{how}"] = {
             "in_lk": lk, "in_la": la, "in_lw": lw,
             "in_rk": rk, "in_rb": rb, "in_rw": rw,
             "out_k": pout["k"].to_numpy().astype(np.int64),

@@ -214,7 +214,7 @@ def test_merge_error_surfaces():
     ok = mpd.DataFrame({"k": rng.integers(0, 5, 10).astype(np.int64),
                         "v": rng.random(10)})
     with pytest.raises(lib.HfError, match="not implemented"):
-        ok.merge(right, on="k", how="outer")
+        ok.merge(right, on="k", how="cross")
 
 
 def test_filter_vs_golden(npartitions):
@@ -952,7 +952,7 @@ def test_outer_right_merge_vs_golden(npartitions):
     and how='right' (right row order, suffix roles preserved) vs the
     reference."""
     for how in ("outer", "right"):
-        g = load_golden(f"mg_{how}")
+        g = load_golden(f"mgo_{how}")
         left = mpd.DataFrame({"k": g["in_lk"], "a": g["in_la"],
                               "w": g["in_lw"]})
         right = mpd.DataFrame({"k": g["in_rk"], "b": g["in_rb"],
