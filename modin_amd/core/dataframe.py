@@ -898,7 +898,7 @@ class HipDataframe:
                     if self._partitions else {})
         if blk_cats:
             raise lib.HfError("diff over string columns is a later round")
-        shifted = self.shift_rows(periods)
+        shifted = self.shift_rows(periods) if periods else None
         n = len(self)
         out_cols = {}
         for name in self.columns:
@@ -906,13 +906,32 @@ class HipDataframe:
             col = cols[0] if len(cols) == 1 else lib.concat(cols)
             if col.dtype_code == lib.HF_INT64:
                 col = lib.cast_f64(col)
-            sh = shifted._partitions[0].block().columns[name]
+            sh = (shifted._partitions[0].block().columns[name]
+                  if shifted is not None else col)
             out_cols[name] = lib.binary(lib.BIN_SUB, col, sh)
         part = HipDataframePartition(DeviceBlock(out_cols, n))
         dtypes = pandas.Series({c: np.dtype(np.float64)
                                 for c in self.columns})
         return HipDataframe([part], self._index, self.columns, [n],
                             dtypes)
+
+    def cumsum_rows(self) -> "HipDataframe":
+        """pandas cumsum(axis=0): the device three-phase scan per column
+        (int64 exact, float64 NaN-skipping)."""
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        if blk_cats:
+            raise lib.HfError("cumsum over string columns")
+        n = len(self)
+        out_cols, dtypes = {}, {}
+        for name in self.columns:
+            cols = [p.block().columns[name] for p in self._partitions]
+            col = cols[0] if len(cols) == 1 else lib.concat(cols)
+            out_cols[name] = lib.cumsum(col)
+            dtypes[name] = self.dtypes[name]
+        part = HipDataframePartition(DeviceBlock(out_cols, n))
+        return HipDataframe([part], self._index, self.columns, [n],
+                            pandas.Series(dtypes))
 
     def idx_extreme(self, maximum: bool) -> dict:
         """Per-column idxmax/idxmin: the FIRST original position holding

@@ -131,6 +131,8 @@ def load() -> ct.CDLL:
                                             ct.POINTER(ct.c_void_p)]),
             "hf_ordered_i64": (ct.c_int, [ct.c_void_p, ct.c_int,
                                           ct.POINTER(ct.c_void_p)]),
+            "hf_cumsum": (ct.c_int, [ct.c_void_p,
+                                     ct.POINTER(ct.c_void_p)]),
             "hf_shuffle_dest": (ct.c_int, [ct.c_void_p,
                                            ct.POINTER(ct.c_int64), ct.c_int,
                                            ct.POINTER(ct.c_void_p)]),
@@ -195,7 +197,7 @@ def exported_symbols():
         "hf_fixup_empty", "hf_sort_perm", "hf_fill_i64",
         "hf_groupby_hash_accum", "hf_groupby_hash_compact",
         "hf_groupby_sorted", "hf_shuffle_dest", "hf_memcpy_dd",
-        "hf_search_sorted", "hf_ordered_i64",
+        "hf_search_sorted", "hf_ordered_i64", "hf_cumsum",
         "hf_col_concat", "hf_col_slice", "hf_join_build", "hf_join_free", "hf_join_probe",
         "hf_gather", "hf_compare_scalar", "hf_filter_plan", "hf_filter_apply",
         "hf_filter_iota", "hf_filter_plan_free", "hf_profiling",
@@ -600,6 +602,15 @@ def filter_iota(plan: FilterPlan, base: int) -> ColumnRef:
     _check(load().hf_filter_iota(plan.handle, base, ct.byref(out)),
            "hf_filter_iota")
     return _wrap(out, plan.n_kept, HF_INT64)
+
+
+def cumsum(col: ColumnRef) -> ColumnRef:
+    """Inclusive prefix sum (pandas cumsum axis=0): f64 skips NaN, i64
+    exact."""
+    ensure_ready()
+    out = ct.c_void_p()
+    _check(load().hf_cumsum(col.handle, ct.byref(out)), "hf_cumsum")
+    return _wrap(out, col.length, col.dtype_code)
 
 
 def ordered_i64(col: ColumnRef, inverse: bool = False) -> ColumnRef:

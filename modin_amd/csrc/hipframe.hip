@@ -1664,6 +1664,101 @@ __global__ void __launch_bounds__(BLOCK) k_segagg(
   }
 }
 
+template <typename T>
+__device__ __forceinline__ T cs_load(const T* p_, int64_t i);
+template <>
+__device__ __forceinline__ double cs_load<double>(const double* p_, int64_t i) {
+  const double v = p_[i];
+  return (v != v) ? 0.0 : v;  // pandas cumsum skips NaN
+}
+template <>
+__device__ __forceinline__ int64_t cs_load<int64_t>(const int64_t* p_,
+                                                    int64_t i) {
+  return p_[i];
+}
+
+template <typename T>
+__global__ void __launch_bounds__(BLOCK) k_cumsum_tiles(
+    const T* __restrict__ in, int64_t n, T* __restrict__ tile_sums) {
+  const int64_t t0 = (int64_t)blockIdx.x * FILT_TILE;
+  const int64_t t1 = min(t0 + (int64_t)FILT_TILE, n);
+  T local = T(0);
+  for (int64_t i = t0 + threadIdx.x; i < t1; i += blockDim.x)
+    local += cs_load<T>(in, i);
+  __shared__ T sc[BLOCK / 64];
+  for (int off = 32; off > 0; off >>= 1)
+    local += __shfl_down(local, off);
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) sc[wave] = local;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    T tot = T(0);
+    for (int w = 0; w < BLOCK / 64; ++w) tot += sc[w];
+    tile_sums[blockIdx.x] = tot;
+  }
+}
+
+template <typename T>
+__global__ void __launch_bounds__(1024) k_cumsum_scan_tiles(
+    T* __restrict__ tile_sums, int64_t ntiles) {
+  __shared__ T carry;
+  if (threadIdx.x == 0) carry = T(0);
+  __syncthreads();
+  __shared__ T buf[1024];
+  for (int64_t base = 0; base < ntiles; base += 1024) {
+    const int64_t i = base + threadIdx.x;
+    T v = (i < ntiles) ? tile_sums[i] : T(0);
+    buf[threadIdx.x] = v;
+    __syncthreads();
+    for (int off = 1; off < 1024; off <<= 1) {
+      T add = (threadIdx.x >= off) ? buf[threadIdx.x - off] : T(0);
+      __syncthreads();
+      buf[threadIdx.x] += add;
+      __syncthreads();
+    }
+    const T incl = buf[threadIdx.x];
+    if (i < ntiles) tile_sums[i] = carry + incl - v;  // exclusive
+    __syncthreads();
+    if (threadIdx.x == 1023) carry += incl;
+    __syncthreads();
+  }
+}
+
+template <typename T>
+__global__ void __launch_bounds__(BLOCK) k_cumsum_apply(
+    const T* __restrict__ in, int64_t n, const T* __restrict__ tile_base,
+    T* __restrict__ out) {
+  const int64_t t0 = (int64_t)blockIdx.x * FILT_TILE;
+  const int64_t t1 = min(t0 + (int64_t)FILT_TILE, n);
+  constexpr int PER = FILT_TILE / BLOCK;  // consecutive elems per thread
+  const int64_t s0 = t0 + (int64_t)threadIdx.x * PER;
+  T loc[PER];
+  T run = T(0);
+  for (int j = 0; j < PER; ++j) {
+    const int64_t i = s0 + j;
+    if (i < t1) run += cs_load<T>(in, i);
+    loc[j] = run;  // inclusive within the thread's segment
+  }
+  // exclusive base across threads (Hillis-Steele over thread totals)
+  __shared__ T buf[BLOCK];
+  buf[threadIdx.x] = run;
+  __syncthreads();
+  for (int off = 1; off < BLOCK; off <<= 1) {
+    T add = (threadIdx.x >= off) ? buf[threadIdx.x - off] : T(0);
+    __syncthreads();
+    buf[threadIdx.x] += add;
+    __syncthreads();
+  }
+  const T tbase = tile_base[blockIdx.x] + buf[threadIdx.x] - run;
+  for (int j = 0; j < PER; ++j) {
+    const int64_t i = s0 + j;
+    if (i < t1) {
+      const T v0 = in[i];
+      out[i] = (v0 != v0) ? v0 : tbase + loc[j];  // NaN stays NaN (f64)
+    }
+  }
+}
+
 __global__ void __launch_bounds__(BLOCK) k_f64_ordered(
     const double* __restrict__ in, int64_t* __restrict__ out, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -2624,6 +2719,41 @@ int hf_groupby_hash_compact(uintptr_t tkey, uintptr_t sums, uintptr_t rowcnt,
 }
 
 static const int64_t* plan_tiles(const hf_filterplan* p);
+
+int hf_cumsum(const hf_col* col, hf_col** out) {
+  HF_NEED_INIT("hf_cumsum");
+  if (!col || !out) return set_err(HF_ERR_ARG, "hf_cumsum", "null");
+  const int64_t n = col->len;
+  int rc = hf_col_alloc(n, col->dtype, out);
+  if (rc != HF_OK) return rc;
+  if (n == 0) return HF_OK;
+  const int64_t ntiles = (n + FILT_TILE - 1) / FILT_TILE;
+  void* d_ts = nullptr;
+  HF_HIP("hf_cumsum", dev_alloc(&d_ts, ntiles * 8, g.stream));
+  auto run = [&](auto tTag) -> int {
+    using T = decltype(tTag);
+    int r2 = timed_launch("cumsum_tiles", [&] {
+      hipLaunchKernelGGL((k_cumsum_tiles<T>), dim3((uint32_t)ntiles),
+                         dim3(BLOCK), 0, g.stream, (const T*)col->dptr, n,
+                         (T*)d_ts);
+    });
+    if (r2 != HF_OK) return r2;
+    r2 = timed_launch("cumsum_scan", [&] {
+      hipLaunchKernelGGL((k_cumsum_scan_tiles<T>), dim3(1), dim3(1024), 0,
+                         g.stream, (T*)d_ts, ntiles);
+    });
+    if (r2 != HF_OK) return r2;
+    return timed_launch("cumsum_apply", [&] {
+      hipLaunchKernelGGL((k_cumsum_apply<T>), dim3((uint32_t)ntiles),
+                         dim3(BLOCK), 0, g.stream, (const T*)col->dptr, n,
+                         (const T*)d_ts, (T*)(*out)->dptr);
+    });
+  };
+  rc = (col->dtype == HF_FLOAT64) ? run(double{}) : run(int64_t{});
+  dev_free(d_ts, g.stream);
+  if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  return rc;
+}
 
 int hf_ordered_i64(const hf_col* col, int direction, hf_col** out) {
   HF_NEED_INIT("hf_ordered_i64");

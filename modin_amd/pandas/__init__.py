@@ -90,6 +90,9 @@ class _HipPandasBase:
     def var(self, ddof: int = 1):
         return self._lower(self._query_compiler.var(ddof=ddof))
 
+    def cumsum(self):
+        return self._rewrap(self._query_compiler.cumsum())
+
     def shift(self, periods: int = 1):
         return self._rewrap(self._query_compiler.shift(int(periods)))
 

@@ -127,6 +127,9 @@ class HipQueryCompiler:
         vals = self._modin_frame.median_columns()
         return pandas.Series(vals, dtype=np.float64)
 
+    def cumsum(self) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.cumsum_rows())
+
     def shift(self, periods: int) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.shift_rows(periods))
 

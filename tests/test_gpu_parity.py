@@ -996,3 +996,24 @@ def test_shift_diff_idx_vs_pandas(npartitions):
     sv = df["v"]
     assert sv.idxmax() == pdf["v"].idxmax()
     assert sv.idxmin() == pdf["v"].idxmin()
+
+
+def test_cumsum_vs_pandas(npartitions):
+    """Device three-phase prefix scan: i64 exact, f64 NaN-skipping, at a
+    multi-tile size."""
+    rng = np.random.default_rng(89)
+    n = 1_000_000  # ~245 tiles
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.05] = np.nan
+    w = rng.integers(-100, 100, n)
+    pdf = pandas.DataFrame({"v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    got = df.cumsum().to_pandas()
+    exp = pdf.cumsum()
+    assert got["w"].dtype == np.int64
+    np.testing.assert_array_equal(got["w"].to_numpy(), exp["w"].to_numpy())
+    np.testing.assert_allclose(got["v"].to_numpy(), exp["v"].to_numpy(),
+                               rtol=1e-12, atol=1e-9, equal_nan=True)
+    s_ = df["v"].cumsum().to_pandas()
+    np.testing.assert_allclose(s_.to_numpy(), pdf["v"].cumsum().to_numpy(),
+                               rtol=1e-12, atol=1e-9, equal_nan=True)
