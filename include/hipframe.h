@@ -303,11 +303,12 @@ int hf_shuffle_dest(const hf_col* keys, const int64_t* splitters, int nsplit,
  * direction=1: ordered i64 col -> f64 col (inverse). */
 int hf_ordered_i64(const hf_col* col, int direction, hf_col** out);
 
-/* Inclusive prefix sum down the column (pandas cumsum, axis=0): f64 skips
- * NaN (NaN rows stay NaN, later sums unaffected), int64 is exact.  Device
- * three-phase scan: per-tile sums -> single-workgroup exclusive tile scan
- * -> per-tile apply. */
-int hf_cumsum(const hf_col* col, hf_col** out);
+/* Inclusive prefix scan down the column (pandas cumsum/cummin/cummax,
+ * axis=0; agg_op = HF_AGG_SUM/MIN/MAX): f64 skips NaN (NaN rows stay NaN,
+ * later results unaffected), int64 is exact.  Device three-phase scan:
+ * per-tile combine -> single-workgroup exclusive tile scan -> per-tile
+ * apply. */
+int hf_cumsum(const hf_col* col, int agg_op, hf_col** out);
 
 /* Exact-match binary search: out[i] = j with sorted[j] == keys[i], else -1.
  * Densifies unbounded int64 join keys through the sorted distinct right

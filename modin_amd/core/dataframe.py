@@ -915,19 +915,19 @@ class HipDataframe:
         return HipDataframe([part], self._index, self.columns, [n],
                             dtypes)
 
-    def cumsum_rows(self) -> "HipDataframe":
-        """pandas cumsum(axis=0): the device three-phase scan per column
-        (int64 exact, float64 NaN-skipping)."""
+    def cumsum_rows(self, agg_op: int = 0) -> "HipDataframe":
+        """pandas cumsum/cummin/cummax (axis=0): the device three-phase
+        scan per column (int64 exact, float64 NaN-skipping)."""
         blk_cats = (self._partitions[0].block().cats
                     if self._partitions else {})
         if blk_cats:
-            raise lib.HfError("cumsum over string columns")
+            raise lib.HfError("cumulative ops over string columns")
         n = len(self)
         out_cols, dtypes = {}, {}
         for name in self.columns:
             cols = [p.block().columns[name] for p in self._partitions]
             col = cols[0] if len(cols) == 1 else lib.concat(cols)
-            out_cols[name] = lib.cumsum(col)
+            out_cols[name] = lib.cumsum(col, agg_op)
             dtypes[name] = self.dtypes[name]
         part = HipDataframePartition(DeviceBlock(out_cols, n))
         return HipDataframe([part], self._index, self.columns, [n],

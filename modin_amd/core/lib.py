@@ -131,7 +131,7 @@ def load() -> ct.CDLL:
                                             ct.POINTER(ct.c_void_p)]),
             "hf_ordered_i64": (ct.c_int, [ct.c_void_p, ct.c_int,
                                           ct.POINTER(ct.c_void_p)]),
-            "hf_cumsum": (ct.c_int, [ct.c_void_p,
+            "hf_cumsum": (ct.c_int, [ct.c_void_p, ct.c_int,
                                      ct.POINTER(ct.c_void_p)]),
             "hf_shuffle_dest": (ct.c_int, [ct.c_void_p,
                                            ct.POINTER(ct.c_int64), ct.c_int,
@@ -604,12 +604,13 @@ def filter_iota(plan: FilterPlan, base: int) -> ColumnRef:
     return _wrap(out, plan.n_kept, HF_INT64)
 
 
-def cumsum(col: ColumnRef) -> ColumnRef:
-    """Inclusive prefix sum (pandas cumsum axis=0): f64 skips NaN, i64
-    exact."""
+def cumsum(col: ColumnRef, agg_op: int = 0) -> ColumnRef:
+    """Inclusive prefix scan (pandas cumsum/cummin/cummax axis=0,
+    agg_op AGG_SUM/MIN/MAX): f64 skips NaN, i64 exact."""
     ensure_ready()
     out = ct.c_void_p()
-    _check(load().hf_cumsum(col.handle, ct.byref(out)), "hf_cumsum")
+    _check(load().hf_cumsum(col.handle, agg_op, ct.byref(out)),
+           "hf_cumsum")
     return _wrap(out, col.length, col.dtype_code)
 
 

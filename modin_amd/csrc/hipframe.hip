@@ -1664,67 +1664,86 @@ __global__ void __launch_bounds__(BLOCK) k_segagg(
   }
 }
 
-template <typename T>
-__device__ __forceinline__ T cs_load(const T* p_, int64_t i);
-template <>
-__device__ __forceinline__ double cs_load<double>(const double* p_, int64_t i) {
-  const double v = p_[i];
-  return (v != v) ? 0.0 : v;  // pandas cumsum skips NaN
+template <typename T, int OP>
+__device__ __forceinline__ T cs_ident() {
+  if (OP == HF_AGG_MIN)
+    return (sizeof(T) == 8 && (T)0.5 == 0) ? (T)INT64_MAX
+                                           : (T)__builtin_huge_val();
+  if (OP == HF_AGG_MAX)
+    return (sizeof(T) == 8 && (T)0.5 == 0) ? (T)INT64_MIN
+                                           : (T)-__builtin_huge_val();
+  return T(0);
 }
-template <>
-__device__ __forceinline__ int64_t cs_load<int64_t>(const int64_t* p_,
-                                                    int64_t i) {
-  return p_[i];
+template <typename T, int OP>
+__device__ __forceinline__ T cs_comb(T a, T b) {
+  if (OP == HF_AGG_MIN) return a < b ? a : b;
+  if (OP == HF_AGG_MAX) return a > b ? a : b;
+  return a + b;
+}
+template <typename T, int OP>
+__device__ __forceinline__ T cs_load(const T* p_, int64_t i) {
+  const T v = p_[i];
+  return (v != v) ? cs_ident<T, OP>() : v;  // NaN skipped (f64)
 }
 
-template <typename T>
+template <typename T, int OP>
 __global__ void __launch_bounds__(BLOCK) k_cumsum_tiles(
     const T* __restrict__ in, int64_t n, T* __restrict__ tile_sums) {
   const int64_t t0 = (int64_t)blockIdx.x * FILT_TILE;
   const int64_t t1 = min(t0 + (int64_t)FILT_TILE, n);
-  T local = T(0);
+  T local = cs_ident<T, OP>();
   for (int64_t i = t0 + threadIdx.x; i < t1; i += blockDim.x)
-    local += cs_load<T>(in, i);
+    local = cs_comb<T, OP>(local, cs_load<T, OP>(in, i));
   __shared__ T sc[BLOCK / 64];
   for (int off = 32; off > 0; off >>= 1)
-    local += __shfl_down(local, off);
+    local = cs_comb<T, OP>(local, (T)__shfl_down(local, off));
   const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
   if (lane == 0) sc[wave] = local;
   __syncthreads();
   if (threadIdx.x == 0) {
-    T tot = T(0);
-    for (int w = 0; w < BLOCK / 64; ++w) tot += sc[w];
+    T tot = cs_ident<T, OP>();
+    for (int w = 0; w < BLOCK / 64; ++w) tot = cs_comb<T, OP>(tot, sc[w]);
     tile_sums[blockIdx.x] = tot;
   }
 }
 
-template <typename T>
+template <typename T, int OP>
 __global__ void __launch_bounds__(1024) k_cumsum_scan_tiles(
     T* __restrict__ tile_sums, int64_t ntiles) {
+  // exclusive scan under cs_comb: track each element's PREFIX (exclusive)
+  // explicitly since min/max have no inverse
   __shared__ T carry;
-  if (threadIdx.x == 0) carry = T(0);
+  if (threadIdx.x == 0) carry = cs_ident<T, OP>();
   __syncthreads();
   __shared__ T buf[1024];
   for (int64_t base = 0; base < ntiles; base += 1024) {
     const int64_t i = base + threadIdx.x;
-    T v = (i < ntiles) ? tile_sums[i] : T(0);
+    T v = (i < ntiles) ? tile_sums[i] : cs_ident<T, OP>();
     buf[threadIdx.x] = v;
     __syncthreads();
+    T excl = cs_ident<T, OP>();
     for (int off = 1; off < 1024; off <<= 1) {
-      T add = (threadIdx.x >= off) ? buf[threadIdx.x - off] : T(0);
+      T add = (threadIdx.x >= off) ? buf[threadIdx.x - off]
+                                   : cs_ident<T, OP>();
       __syncthreads();
-      buf[threadIdx.x] += add;
+      buf[threadIdx.x] = cs_comb<T, OP>(buf[threadIdx.x], add);
       __syncthreads();
     }
     const T incl = buf[threadIdx.x];
-    if (i < ntiles) tile_sums[i] = carry + incl - v;  // exclusive
+    // exclusive = inclusive of the previous lane (or identity at lane 0)
+    excl = (threadIdx.x == 0) ? cs_ident<T, OP>() : T();
     __syncthreads();
-    if (threadIdx.x == 1023) carry += incl;
+    buf[threadIdx.x] = incl;  // reuse buf to read neighbor inclusives
+    __syncthreads();
+    excl = (threadIdx.x == 0) ? cs_ident<T, OP>() : buf[threadIdx.x - 1];
+    if (i < ntiles) tile_sums[i] = cs_comb<T, OP>(carry, excl);
+    __syncthreads();
+    if (threadIdx.x == 1023) carry = cs_comb<T, OP>(carry, incl);
     __syncthreads();
   }
 }
 
-template <typename T>
+template <typename T, int OP>
 __global__ void __launch_bounds__(BLOCK) k_cumsum_apply(
     const T* __restrict__ in, int64_t n, const T* __restrict__ tile_base,
     T* __restrict__ out) {
@@ -1733,28 +1752,35 @@ __global__ void __launch_bounds__(BLOCK) k_cumsum_apply(
   constexpr int PER = FILT_TILE / BLOCK;  // consecutive elems per thread
   const int64_t s0 = t0 + (int64_t)threadIdx.x * PER;
   T loc[PER];
-  T run = T(0);
+  T run = cs_ident<T, OP>();
   for (int j = 0; j < PER; ++j) {
     const int64_t i = s0 + j;
-    if (i < t1) run += cs_load<T>(in, i);
+    if (i < t1) run = cs_comb<T, OP>(run, cs_load<T, OP>(in, i));
     loc[j] = run;  // inclusive within the thread's segment
   }
-  // exclusive base across threads (Hillis-Steele over thread totals)
+  // exclusive base across threads: scan thread totals, read neighbor
   __shared__ T buf[BLOCK];
   buf[threadIdx.x] = run;
   __syncthreads();
   for (int off = 1; off < BLOCK; off <<= 1) {
-    T add = (threadIdx.x >= off) ? buf[threadIdx.x - off] : T(0);
+    T add = (threadIdx.x >= off) ? buf[threadIdx.x - off]
+                                 : cs_ident<T, OP>();
     __syncthreads();
-    buf[threadIdx.x] += add;
+    buf[threadIdx.x] = cs_comb<T, OP>(buf[threadIdx.x], add);
     __syncthreads();
   }
-  const T tbase = tile_base[blockIdx.x] + buf[threadIdx.x] - run;
+  const T incl = buf[threadIdx.x];
+  __syncthreads();
+  buf[threadIdx.x] = incl;
+  __syncthreads();
+  const T texcl = (threadIdx.x == 0) ? cs_ident<T, OP>()
+                                     : buf[threadIdx.x - 1];
+  const T tbase = cs_comb<T, OP>(tile_base[blockIdx.x], texcl);
   for (int j = 0; j < PER; ++j) {
     const int64_t i = s0 + j;
     if (i < t1) {
       const T v0 = in[i];
-      out[i] = (v0 != v0) ? v0 : tbase + loc[j];  // NaN stays NaN (f64)
+      out[i] = (v0 != v0) ? v0 : cs_comb<T, OP>(tbase, loc[j]);
     }
   }
 }
@@ -2720,9 +2746,11 @@ int hf_groupby_hash_compact(uintptr_t tkey, uintptr_t sums, uintptr_t rowcnt,
 
 static const int64_t* plan_tiles(const hf_filterplan* p);
 
-int hf_cumsum(const hf_col* col, hf_col** out) {
+int hf_cumsum(const hf_col* col, int agg_op, hf_col** out) {
   HF_NEED_INIT("hf_cumsum");
   if (!col || !out) return set_err(HF_ERR_ARG, "hf_cumsum", "null");
+  if (agg_op != HF_AGG_SUM && agg_op != HF_AGG_MIN && agg_op != HF_AGG_MAX)
+    return set_err(HF_ERR_ARG, "hf_cumsum", "bad agg_op");
   const int64_t n = col->len;
   int rc = hf_col_alloc(n, col->dtype, out);
   if (rc != HF_OK) return rc;
@@ -2730,26 +2758,35 @@ int hf_cumsum(const hf_col* col, hf_col** out) {
   const int64_t ntiles = (n + FILT_TILE - 1) / FILT_TILE;
   void* d_ts = nullptr;
   HF_HIP("hf_cumsum", dev_alloc(&d_ts, ntiles * 8, g.stream));
-  auto run = [&](auto tTag) -> int {
+  auto run = [&](auto tTag, auto opTag) -> int {
     using T = decltype(tTag);
+    constexpr int OP = decltype(opTag)::value;
     int r2 = timed_launch("cumsum_tiles", [&] {
-      hipLaunchKernelGGL((k_cumsum_tiles<T>), dim3((uint32_t)ntiles),
+      hipLaunchKernelGGL((k_cumsum_tiles<T, OP>), dim3((uint32_t)ntiles),
                          dim3(BLOCK), 0, g.stream, (const T*)col->dptr, n,
                          (T*)d_ts);
     });
     if (r2 != HF_OK) return r2;
     r2 = timed_launch("cumsum_scan", [&] {
-      hipLaunchKernelGGL((k_cumsum_scan_tiles<T>), dim3(1), dim3(1024), 0,
-                         g.stream, (T*)d_ts, ntiles);
+      hipLaunchKernelGGL((k_cumsum_scan_tiles<T, OP>), dim3(1), dim3(1024),
+                         0, g.stream, (T*)d_ts, ntiles);
     });
     if (r2 != HF_OK) return r2;
     return timed_launch("cumsum_apply", [&] {
-      hipLaunchKernelGGL((k_cumsum_apply<T>), dim3((uint32_t)ntiles),
+      hipLaunchKernelGGL((k_cumsum_apply<T, OP>), dim3((uint32_t)ntiles),
                          dim3(BLOCK), 0, g.stream, (const T*)col->dptr, n,
                          (const T*)d_ts, (T*)(*out)->dptr);
     });
   };
-  rc = (col->dtype == HF_FLOAT64) ? run(double{}) : run(int64_t{});
+  auto runT = [&](auto opTag) -> int {
+    return (col->dtype == HF_FLOAT64) ? run(double{}, opTag)
+                                      : run(int64_t{}, opTag);
+  };
+  rc = agg_op == HF_AGG_MIN
+           ? runT(std::integral_constant<int, HF_AGG_MIN>{})
+       : agg_op == HF_AGG_MAX
+           ? runT(std::integral_constant<int, HF_AGG_MAX>{})
+           : runT(std::integral_constant<int, HF_AGG_SUM>{});
   dev_free(d_ts, g.stream);
   if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
   return rc;

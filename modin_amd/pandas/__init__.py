@@ -93,6 +93,12 @@ class _HipPandasBase:
     def cumsum(self):
         return self._rewrap(self._query_compiler.cumsum())
 
+    def cummin(self):
+        return self._rewrap(self._query_compiler.cummin())
+
+    def cummax(self):
+        return self._rewrap(self._query_compiler.cummax())
+
     def shift(self, periods: int = 1):
         return self._rewrap(self._query_compiler.shift(int(periods)))
 

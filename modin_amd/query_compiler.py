@@ -128,7 +128,13 @@ class HipQueryCompiler:
         return pandas.Series(vals, dtype=np.float64)
 
     def cumsum(self) -> "HipQueryCompiler":
-        return self.__constructor__(self._modin_frame.cumsum_rows())
+        return self.__constructor__(self._modin_frame.cumsum_rows(lib.AGG_SUM))
+
+    def cummin(self) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.cumsum_rows(lib.AGG_MIN))
+
+    def cummax(self) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.cumsum_rows(lib.AGG_MAX))
 
     def shift(self, periods: int) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.shift_rows(periods))

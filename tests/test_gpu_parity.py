@@ -1017,3 +1017,12 @@ def test_cumsum_vs_pandas(npartitions):
     s_ = df["v"].cumsum().to_pandas()
     np.testing.assert_allclose(s_.to_numpy(), pdf["v"].cumsum().to_numpy(),
                                rtol=1e-12, atol=1e-9, equal_nan=True)
+    for name, fn in (("cummax", "cummax"), ("cummin", "cummin")):
+        got2 = getattr(df, fn)().to_pandas()
+        exp2 = getattr(pdf, fn)()
+        assert got2["w"].dtype == np.int64
+        np.testing.assert_array_equal(got2["w"].to_numpy(),
+                                      exp2["w"].to_numpy(), err_msg=name)
+        np.testing.assert_allclose(got2["v"].to_numpy(),
+                                   exp2["v"].to_numpy(), rtol=0,
+                                   equal_nan=True, err_msg=name)
