@@ -699,10 +699,14 @@ class HipDataframe:
         return out
 
     def groupby_median(self, by) -> "HipDataframe":
-        """groupby().median(): sort by (key, value na-last) once per value
-        column, then gather the per-group middle elements — offsets from
+        return self.groupby_quantile(by, 0.5)
+
+    def groupby_quantile(self, by, q: float = 0.5) -> "HipDataframe":
+        """groupby().quantile(q) (median is q=0.5): sort by (key, value
+        na-last) once per value column, then gather the per-group
+        bracketing elements and interpolate linearly — offsets from
         groupby size, non-NaN counts from groupby count (pandas
-        nanmedian per group)."""
+        nanquantile per group)."""
         from ..distributed import is_active
         if is_active():
             raise lib.HfError(
@@ -713,7 +717,7 @@ class HipDataframe:
             else:
                 cf, decode = self._combined_key_frame(list(by))
                 keep = [c for c in cf.columns if c not in by]
-                res = cf.take_columns(keep).groupby_median(self.KEYCOL)
+                res = cf.take_columns(keep).groupby_quantile(self.KEYCOL, q)
                 res._index = decode(lib.get(res._index.col))
                 return res
         val_names = [c for c in self.columns if c != by]
@@ -752,13 +756,15 @@ class HipDataframe:
                             perm)
             cnt = lib.get(cnt_block.columns[v]) if ng else np.empty(0)
             cnt = np.asarray(cnt, dtype=np.int64)
-            lo = offs + np.maximum((cnt - 1) // 2, 0)
-            hi = offs + np.maximum(cnt // 2, 0)
+            pos = np.maximum(cnt - 1, 0) * float(q)
+            lo = offs + np.floor(pos).astype(np.int64)
+            hi = offs + np.ceil(pos).astype(np.int64)
             lo_c = lib.put(lo.astype(np.int64))
             hi_c = lib.put(hi.astype(np.int64))
             a = lib.get(lib.gather(sv, lo_c)) if ng else np.empty(0)
             b = lib.get(lib.gather(sv, hi_c)) if ng else np.empty(0)
-            med = (a + b) / 2.0
+            frac = pos - np.floor(pos)
+            med = a + (b - a) * frac
             med[cnt == 0] = np.nan
             out_cols[v] = lib.put(med)
         part = HipDataframePartition(DeviceBlock(out_cols, ng))

@@ -604,6 +604,17 @@ class DataFrameGroupBy:
     def median(self):
         return self._agg("median")
 
+    def quantile(self, q: float = 0.5):
+        out = DataFrame(
+            query_compiler=self._df._query_compiler.groupby_quantile(
+                self._by, float(q)))
+        if self._series_out and self._as_index:
+            name = list(out._query_compiler._modin_frame.columns)[0]
+            return Series(query_compiler=out._query_compiler, name=name)
+        if not self._as_index:
+            return from_pandas(out.to_pandas().reset_index())
+        return out
+
     def first(self):
         return self._agg("first")
 

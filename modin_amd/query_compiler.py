@@ -157,6 +157,10 @@ class HipQueryCompiler:
     def groupby_median(self, by) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.groupby_median(by))
 
+    def groupby_quantile(self, by, q: float) -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.groupby_quantile(by, q))
+
     def groupby_first(self, by) -> "HipQueryCompiler":
         return self.__constructor__(
             self._modin_frame.groupby_firstlast(by, last=False))

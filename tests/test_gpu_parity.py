@@ -1026,3 +1026,22 @@ def test_cumsum_vs_pandas(npartitions):
         np.testing.assert_allclose(got2["v"].to_numpy(),
                                    exp2["v"].to_numpy(), rtol=0,
                                    equal_nan=True, err_msg=name)
+
+
+def test_groupby_quantile_vs_pandas(npartitions):
+    rng = np.random.default_rng(91)
+    n = 40_000
+    k = rng.integers(0, 60, n)
+    v = rng.standard_normal(n) * 5
+    v[rng.random(n) < 0.1] = np.nan
+    pdf = pandas.DataFrame({"k": k, "v": v})
+    df = mpd.DataFrame(pdf)
+    for q in (0.5, 0.25, 0.9, 0.0, 1.0):
+        got = df.groupby("k").quantile(q).to_pandas()
+        exp = pdf.groupby("k").quantile(q)
+        np.testing.assert_array_equal(got.index.to_numpy(),
+                                      exp.index.to_numpy())
+        np.testing.assert_allclose(got["v"].to_numpy(),
+                                   exp["v"].to_numpy(), rtol=1e-12,
+                                   atol=1e-12, equal_nan=True,
+                                   err_msg=str(q))
