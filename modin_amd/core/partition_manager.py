@@ -257,7 +257,35 @@ class HipDataframePartitionManager:
     def _groupby_hash(cls, key_cols, val_cols_per_part, total_rows,
                       want_counts, agg_op):
         """Open-addressing hash groupby for unbounded key ranges (single
-        rank); grows the table 4x and retries on overflow."""
+        rank); grows the table 4x and retries on overflow.  Near-unique
+        keys route straight to the sort-based path: a strided 64K-key
+        sample estimating the distinct ratio avoids a doomed (and
+        expensive) full-table attempt when the cardinality clearly
+        exceeds the 2^27-slot cap."""
+        if total_rows > (1 << 27):
+            S = 65536
+            import numpy as np
+            samples = []
+            need = S
+            for kcol in key_cols:
+                if need <= 0 or not kcol.length:
+                    continue
+                take = min(need, S * kcol.length // max(total_rows, 1) + 1)
+                idx = np.linspace(0, kcol.length - 1, take).astype(np.int64)
+                samples.append(lib.get(lib.gather(kcol, lib.put(idx))))
+                need -= take
+            if samples:
+                # birthday estimator: s draws from D uniform distinct give
+                # ~s^2/2D collisions, so D^ = s^2 / (2 * collisions); only
+                # clearly-over-cap cardinalities skip the hash attempt
+                sarr = np.concatenate(samples)
+                ssz = len(sarr)
+                coll = ssz - len(np.unique(sarr))
+                d_est = float("inf") if coll == 0 \
+                    else ssz * ssz / (2.0 * coll)
+                if d_est > 1.5 * (1 << 27):
+                    return cls._groupby_sorted(key_cols, val_cols_per_part,
+                                               want_counts, agg_op)
         INT64_MIN = -(1 << 63)
         nv = len(val_cols_per_part[0]) if val_cols_per_part else 0
         # start near 2x the row count (capped): avoids doomed attempts on
