@@ -1749,13 +1749,24 @@ __global__ void __launch_bounds__(BLOCK) k_cumsum_apply(
     T* __restrict__ out) {
   const int64_t t0 = (int64_t)blockIdx.x * FILT_TILE;
   const int64_t t1 = min(t0 + (int64_t)FILT_TILE, n);
-  constexpr int PER = FILT_TILE / BLOCK;  // consecutive elems per thread
-  const int64_t s0 = t0 + (int64_t)threadIdx.x * PER;
+  constexpr int PER = FILT_TILE / BLOCK;  // elems per thread
+  // stage the tile through LDS so HBM access stays COALESCED (striped)
+  // while each thread scans PER CONSECUTIVE elements from LDS
+  __shared__ T stage[FILT_TILE];
+  for (int j = 0; j < PER; ++j) {
+    const int64_t i = t0 + (int64_t)j * BLOCK + threadIdx.x;
+    if (i < t1) stage[(int)(i - t0)] = in[i];
+  }
+  __syncthreads();
+  const int s0 = threadIdx.x * PER;
+  const int lim = (int)(t1 - t0);
   T loc[PER];
   T run = cs_ident<T, OP>();
   for (int j = 0; j < PER; ++j) {
-    const int64_t i = s0 + j;
-    if (i < t1) run = cs_comb<T, OP>(run, cs_load<T, OP>(in, i));
+    if (s0 + j < lim) {
+      const T v = stage[s0 + j];
+      run = cs_comb<T, OP>(run, (v != v) ? cs_ident<T, OP>() : v);
+    }
     loc[j] = run;  // inclusive within the thread's segment
   }
   // exclusive base across threads: scan thread totals, read neighbor
@@ -1776,12 +1787,17 @@ __global__ void __launch_bounds__(BLOCK) k_cumsum_apply(
   const T texcl = (threadIdx.x == 0) ? cs_ident<T, OP>()
                                      : buf[threadIdx.x - 1];
   const T tbase = cs_comb<T, OP>(tile_base[blockIdx.x], texcl);
+  __syncthreads();
   for (int j = 0; j < PER; ++j) {
-    const int64_t i = s0 + j;
-    if (i < t1) {
-      const T v0 = in[i];
-      out[i] = (v0 != v0) ? v0 : cs_comb<T, OP>(tbase, loc[j]);
+    if (s0 + j < lim) {
+      const T v0 = stage[s0 + j];
+      stage[s0 + j] = (v0 != v0) ? v0 : cs_comb<T, OP>(tbase, loc[j]);
     }
+  }
+  __syncthreads();
+  for (int j = 0; j < PER; ++j) {
+    const int64_t i = t0 + (int64_t)j * BLOCK + threadIdx.x;
+    if (i < t1) out[i] = stage[(int)(i - t0)];
   }
 }
 
