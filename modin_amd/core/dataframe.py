@@ -1050,12 +1050,10 @@ class HipDataframe:
             # from the FULL right table (all-gather of the right shards),
             # probes only its own left shard — result stays left-sharded
             from .. import distributed as dist_mod
-            rdt = [c.dtype_code for c in rvals]
             gathered = dist_mod.allgather_arrays(
                 [lib.get(rkeys)] + [lib.get(c) for c in rvals])
             rkeys = lib.put(gathered[0])
             rvals = [lib.put(a) for a in gathered[1:]]
-            del rdt
         if rkeys.length:
             r = lib.reduce(rkeys)
             kmin, n_slots = r.imn, r.imx - r.imn + 1

@@ -1730,11 +1730,10 @@ __global__ void __launch_bounds__(1024) k_cumsum_scan_tiles(
       __syncthreads();
     }
     const T incl = buf[threadIdx.x];
-    // exclusive = inclusive of the previous lane (or identity at lane 0)
-    excl = (threadIdx.x == 0) ? cs_ident<T, OP>() : T();
     __syncthreads();
     buf[threadIdx.x] = incl;  // reuse buf to read neighbor inclusives
     __syncthreads();
+    // exclusive = inclusive of the previous lane (or identity at lane 0)
     excl = (threadIdx.x == 0) ? cs_ident<T, OP>() : buf[threadIdx.x - 1];
     if (i < ntiles) tile_sums[i] = cs_comb<T, OP>(carry, excl);
     __syncthreads();
