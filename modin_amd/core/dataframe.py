@@ -969,6 +969,106 @@ class HipDataframe:
             out[name] = int(lib.reduce(pos).imn)
         return out
 
+    def groupby_nunique(self, by) -> "HipDataframe":
+        """groupby().nunique(): per value column, the count of DISTINCT
+        non-NaN values in each group — sort by (key, value), mark run
+        boundaries with exact device compares (int64 subtract for keys,
+        f64 subtract for NaN-free values), filter to the distinct pairs
+        and size-count them per key; groups whose values are all NaN
+        report 0 (pandas dropna=True)."""
+        from ..distributed import is_active
+        if is_active():
+            raise lib.HfError("distributed groupby.nunique is a later "
+                              "round")
+        if isinstance(by, (list, tuple)):
+            if len(by) == 1:
+                by = by[0]
+            else:
+                cf, decode = self._combined_key_frame(list(by))
+                keep = [c for c in cf.columns if c not in by]
+                res = cf.take_columns(keep).groupby_nunique(self.KEYCOL)
+                res._index = decode(lib.get(res._index.col))
+                return res
+        val_names = [c for c in self.columns if c != by]
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        res0 = self.groupby_size(by)
+        gkeys = res0.index
+        ng = len(gkeys)
+        out_cols = {}
+        for v in val_names:
+            sub = self.take_columns([by, v])
+            fparts = []
+            for p in sub._partitions:
+                block = p.block()
+                vcol = block.columns[v]
+                if v in blk_cats:
+                    keep = lib.compare_scalar(lib.CMP_NE, vcol, -1.0)
+                elif vcol.dtype_code == lib.HF_FLOAT64:
+                    keep = lib.compare_scalar(lib.CMP_NOTNA, vcol, 0.0)
+                else:
+                    keep = None
+                if keep is not None:
+                    plan = lib.filter_plan(keep)
+                    cols = {m: lib.filter_apply(plan, c)
+                            for m, c in block.columns.items()}
+                    fparts.append(HipDataframePartition(
+                        DeviceBlock(cols, plan.n_kept, block.cats)))
+                else:
+                    fparts.append(p)
+            sub = HipDataframe(fparts, pandas.RangeIndex(
+                sum(pp.block().length for pp in fparts)),
+                [by, v],
+                [pp.block().length for pp in fparts] if fparts else [0],
+                self.dtypes[[by, v]])
+            srt = sub.sort_rows([by, v], True)
+            sblock = srt._partitions[0].block()
+            n2 = sblock.length
+            if n2 == 0:
+                counts_sub = np.empty(0, dtype=np.int64)
+                skeys_idx = pandas.Index(np.empty(0, dtype=np.int64),
+                                         name=by)
+            else:
+                def run_head(col):
+                    # head[i] = (col[i] != col[i-1]); row 0 is ALWAYS a
+                    # head: the shifted-in sentinel is col[0]^1 (i64) /
+                    # NaN (f64), guaranteed != col[0]
+                    prev_body = lib.col_slice(col, 0, n2 - 1)
+                    first = lib.alloc(1, col.dtype_code)
+                    if col.dtype_code == lib.HF_FLOAT64:
+                        lib.fill_f64(first.dptr(), float("nan"), 1)
+                    else:
+                        v0 = int(lib.get(lib.col_slice(col, 0, 1))[0])
+                        lib.fill_i64(first.dptr(), v0 ^ 1, 1)
+                    prev = lib.concat([first, prev_body])
+                    d = lib.binary(lib.BIN_SUB, col, prev)
+                    return lib.compare_scalar(lib.CMP_NE, d, 0.0)
+
+                hk = run_head(sblock.columns[by])
+                hv = run_head(sblock.columns[v])
+                head = lib.binary(lib.BIN_ADD, hk, hv)  # >0 == new pair
+                head = lib.compare_scalar(lib.CMP_GE, head, 1.0)
+                plan = lib.filter_plan(head)
+                dkeys = lib.filter_apply(plan, sblock.columns[by])
+                dpart = HipDataframePartition(DeviceBlock(
+                    {by: dkeys}, plan.n_kept,
+                    {by: blk_cats[by]} if by in blk_cats else {}))
+                dframe = HipDataframe([dpart],
+                                      pandas.RangeIndex(plan.n_kept),
+                                      [by], [plan.n_kept],
+                                      self.dtypes[[by]])
+                dsz = dframe.groupby_size(by)
+                counts_sub = lib.get(
+                    dsz._partitions[0].block().columns["size"])
+                skeys_idx = dsz.index
+            out = np.zeros(ng, dtype=np.int64)
+            idx = gkeys.get_indexer(skeys_idx)
+            out[idx] = counts_sub
+            out_cols[v] = lib.put(out)
+        part = HipDataframePartition(DeviceBlock(out_cols, ng))
+        dtypes = pandas.Series({v: np.dtype(np.int64) for v in val_names})
+        return HipDataframe([part], res0._index, val_names, [ng], dtypes)
+
     def hconcat(self, others: list) -> "HipDataframe":
         """Horizontal compose of single-partition frames with identical row
         count and index (the device form of the reference's axis=1 concat

@@ -615,6 +615,16 @@ class DataFrameGroupBy:
             return from_pandas(out.to_pandas().reset_index())
         return out
 
+    def nunique(self):
+        qc = self._df._query_compiler.groupby_nunique(self._by)
+        out = DataFrame(query_compiler=qc)
+        if self._series_out and self._as_index:
+            name = list(qc._modin_frame.columns)[0]
+            return Series(query_compiler=qc, name=name)
+        if not self._as_index:
+            return from_pandas(out.to_pandas().reset_index())
+        return out
+
     def first(self):
         return self._agg("first")
 

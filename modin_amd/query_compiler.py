@@ -169,6 +169,9 @@ class HipQueryCompiler:
         return self.__constructor__(
             self._modin_frame.groupby_firstlast(by, last=True))
 
+    def groupby_nunique(self, by) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.groupby_nunique(by))
+
     def groupby_size(self, by: str) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.groupby_size(by))
 

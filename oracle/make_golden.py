@@ -976,6 +976,39 @@ def gen_outer_right_merge_cases(mpd, rng):
     return cases
 
 
+def gen_nunique_cases(mpd, rng):
+    """groupby.nunique per value column (distinct non-NaN; all-NaN groups
+    0; int/float/string values; string and multi keys) vs the
+    reference."""
+    import pandas
+    cases = {}
+    n = 3000
+    k = rng.integers(0, 25, n).astype(np.int64)
+    v = np.round(rng.random(n) * 20, 1)
+    v[rng.random(n) < 0.15] = np.nan
+    v[k == 3] = np.nan  # all-NaN group -> 0
+    w = rng.integers(0, 8, n).astype(np.int64)
+    pool = np.array(["p", "q", "r", "s"])
+    sv = rng.choice(pool, n).astype(object)
+    sv[rng.random(n) < 0.1] = np.nan
+    mdf = mpd.DataFrame({"k": k, "v": v, "w": w, "s": sv})
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w, "s": sv})
+    mres = mdf.groupby("k").nunique()._to_pandas()
+    pres = pdf.groupby("k").nunique()
+    assert list(mres.index) == list(pres.index)
+    np.testing.assert_array_equal(mres.values, pres.values)
+    arrays = {"in_k": k, "in_v": v, "in_w": w, "in_s": _enc_str(sv),
+              "out_keys": pres.index.to_numpy().astype(np.int64)}
+    for cn in ("v", "w", "s"):
+        arrays[f"out_{cn}"] = pres[cn].to_numpy().astype(np.int64)
+    # string key
+    pres2 = pdf[["s", "v"]].groupby("s").nunique()
+    arrays["out_sk_keys"] = _enc_str(pres2.index)
+    arrays["out_sk_v"] = pres2["v"].to_numpy().astype(np.int64)
+    cases["gbnu_cases"] = arrays
+    return cases
+
+
 def main():
     os.makedirs(GOLDEN_DIR, exist_ok=True)
     mpd = _setup_reference()
@@ -999,6 +1032,7 @@ def main():
     all_cases.update(gen_left_merge_cases(mpd, rng))
     all_cases.update(gen_firstlast_cases(mpd, rng))
     all_cases.update(gen_outer_right_merge_cases(mpd, rng))
+    all_cases.update(gen_nunique_cases(mpd, rng))
     for name, arrays in all_cases.items():
         path = os.path.join(GOLDEN_DIR, f"{name}.npz")
         np.savez_compressed(path, **arrays)

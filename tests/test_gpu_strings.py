@@ -345,3 +345,20 @@ def test_groupby_first_last_vs_golden(npartitions):
                                       g[f"out_{agg}_w"])
         assert_str_equal(out["s"].to_numpy(), g[f"out_{agg}_s"],
                          f"{agg} s")
+
+
+def test_groupby_nunique_vs_golden(npartitions):
+    """groupby.nunique: distinct non-NaN values per group via sorted
+    run-boundary masks; int/float/string values; all-NaN groups 0."""
+    g = load_golden("gbnu_cases")
+    df = mpd.DataFrame({"k": g["in_k"], "v": g["in_v"], "w": g["in_w"],
+                        "s": dec(g["in_s"])})
+    out = df.groupby("k").nunique().to_pandas()
+    np.testing.assert_array_equal(out.index.to_numpy(), g["out_keys"])
+    for cn in ("v", "w", "s"):
+        assert out[cn].dtype == np.int64
+        np.testing.assert_array_equal(out[cn].to_numpy(), g[f"out_{cn}"],
+                                      err_msg=cn)
+    out2 = df[["s", "v"]].groupby("s").nunique().to_pandas()
+    assert_str_equal(out2.index.to_numpy(), g["out_sk_keys"], "sk keys")
+    np.testing.assert_array_equal(out2["v"].to_numpy(), g["out_sk_v"])
