@@ -1,0 +1,121 @@
+"""GPU tier: randomized cross-operation pipelines vs pandas.
+
+Each case builds a random frame (int64/float64/string columns, NaNs
+planted) and runs a random composition of the backend's operations,
+comparing every intermediate that materializes against pandas computing
+the same chain.  Seeds are FIXED — failures reproduce exactly.  pandas is
+the arbiter the reference's own df_equals tests use.
+"""
+
+import numpy as np
+import pandas
+import pytest
+
+import modin_amd.pandas as mpd
+
+pytestmark = pytest.mark.gpu
+
+POOL = np.array(["ash", "birch", "cedar", "dub", "elm", "Fir", "ginkgo"])
+
+
+@pytest.fixture(autouse=True)
+def _ready(gpu_ready):
+    yield
+
+
+def make_frame(rng, n):
+    k = rng.integers(0, rng.integers(5, 500), n).astype(np.int64)
+    v = rng.standard_normal(n) * rng.integers(1, 50)
+    v[rng.random(n) < rng.random() * 0.2] = np.nan
+    w = rng.integers(-1000, 1000, n).astype(np.int64)
+    s = rng.choice(POOL[: rng.integers(2, len(POOL))], n).astype(object)
+    s[rng.random(n) < rng.random() * 0.1] = np.nan
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w, "s": s})
+    return pdf
+
+
+def check(df, pdf, msg):
+    got = df.to_pandas()
+    assert list(got.columns) == list(pdf.columns), msg
+    np.testing.assert_array_equal(got.index.to_numpy(),
+                                  pdf.index.to_numpy(), err_msg=msg)
+    for c in pdf.columns:
+        g, e = got[c].to_numpy(), pdf[c].to_numpy()
+        if e.dtype == object:
+            for i, (a, b) in enumerate(zip(g, e)):
+                bn = isinstance(b, float) and np.isnan(b)
+                gn = isinstance(a, float) and np.isnan(a)
+                assert gn == bn and (gn or a == b), f"{msg}/{c}[{i}]"
+        else:
+            np.testing.assert_allclose(g, e, rtol=1e-12, atol=1e-9,
+                                       equal_nan=True,
+                                       err_msg=f"{msg}/{c}")
+
+
+@pytest.mark.parametrize("seed", range(20))
+def test_fuzz_pipeline(seed):
+    rng = np.random.default_rng(1000 + seed)
+    n = int(rng.integers(500, 40_000))
+    pdf = make_frame(rng, n)
+    df = mpd.DataFrame(pdf)
+
+    steps = rng.integers(2, 5)
+    for si in range(steps):
+        op = rng.choice(["filter", "sort", "head", "dropna", "arith"])
+        msg = f"seed {seed} step {si} op {op}"
+        if op == "filter":
+            thr = float(np.round(rng.standard_normal() * 10, 2))
+            df = df[df["v"] > thr]
+            pdf = pdf[pdf["v"] > thr]
+            df = mpd.DataFrame(df.to_pandas().reset_index(drop=True))
+            pdf = pdf.reset_index(drop=True)
+        elif op == "sort":
+            by = [["k"], ["s"], ["k", "w"], ["v"], ["s", "k"]][
+                rng.integers(0, 5)]
+            if "s" in by and pdf["s"].isna().any() and len(by) > 1 \
+                    and by[0] != "s":
+                by = ["k", "w"]
+            asc = bool(rng.integers(0, 2))
+            df = df.sort_values(by, ascending=asc)
+            pdf = pdf.sort_values(by, ascending=asc, kind="stable")
+            check(df, pdf, msg)
+            df = mpd.DataFrame(df.to_pandas().reset_index(drop=True))
+            pdf = pdf.reset_index(drop=True)
+        elif op == "head":
+            m = int(rng.integers(1, max(len(pdf), 2)))
+            df, pdf = df.head(m), pdf.head(m)
+        elif op == "dropna":
+            df = df.dropna()
+            pdf = pdf.dropna()
+            df = mpd.DataFrame(df.to_pandas().reset_index(drop=True))
+            pdf = pdf.reset_index(drop=True)
+        else:
+            c = float(np.round(rng.standard_normal(), 3))
+            sub = [col for col in ("v", "w") if col in pdf.columns]
+            for col in sub:
+                got = (df[col] + c).to_pandas()
+                exp = pdf[col] + c
+                np.testing.assert_allclose(got.to_numpy(),
+                                           exp.to_numpy(), rtol=1e-12,
+                                           atol=1e-12, equal_nan=True,
+                                           err_msg=msg)
+        check(df, pdf, msg)
+        if len(pdf) == 0:
+            break
+
+    # closing aggregation on whatever survived
+    if len(pdf):
+        by = ["k", "s"][rng.integers(0, 2)]
+        agg = ["sum", "mean", "count", "min", "max", "var", "median",
+               "first", "last"][rng.integers(0, 9)]
+        sub = [by, "v", "w"]  # numeric values only (string agg is loud)
+        gout = getattr(df[sub].groupby(by), agg)().to_pandas()
+        pout = getattr(pdf[sub].groupby(by), agg)()
+        assert len(gout) == len(pout), f"seed {seed} closing {by}/{agg}"
+        for c in pout.columns:
+            g, e = gout[c].to_numpy(), pout[c].to_numpy()
+            if e.dtype == object:
+                continue
+            np.testing.assert_allclose(
+                g.astype(float), e.astype(float), rtol=1e-9, atol=1e-9,
+                equal_nan=True, err_msg=f"seed {seed} {by}/{agg}/{c}")
