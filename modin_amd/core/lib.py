@@ -317,6 +317,10 @@ def map_scalar(op: int, col: ColumnRef, scalar) -> ColumnRef:
     out = ct.c_void_p()
     if col.dtype_code == HF_INT64 and op in (MAP_DIV, MAP_RDIV, MAP_FILLNA):
         col = cast_f64(col)  # pandas promotes int div to float; fillna no-ops
+    elif (col.dtype_code == HF_INT64
+          and op in (MAP_ADD, MAP_SUB, MAP_RSUB, MAP_MUL)
+          and isinstance(scalar, float) and not scalar.is_integer()):
+        col = cast_f64(col)  # pandas: int64 op non-integral float -> float64
     if op == MAP_CAST_I64:
         if col.dtype_code == HF_INT64:
             return col
