@@ -99,7 +99,9 @@ enum {
   HF_MAP_NEG = 8,   /* -x          */
   HF_MAP_CAST_F64 = 9, /* (double)x : i64 -> f64; scalar ignored */
   HF_MAP_CAST_I64 = 10, /* (int64)x : f64 -> i64, C truncation (astype)    */
-  HF_MAP_SQRT = 11  /* sqrt(x), f64 only (std = sqrt(var))                  */
+  HF_MAP_SQRT = 11, /* sqrt(x), f64 only (std = sqrt(var))                  */
+  HF_MAP_MIN = 12,  /* fmin(x, s) — clip upper; NaN passes through (f64)    */
+  HF_MAP_MAX = 13   /* fmax(x, s) — clip lower; NaN passes through (f64)    */
 };
 int hf_map_scalar(int op, const hf_col* in, double scalar, hf_col** out);
 /* i64 column with an exact int64 scalar (double cannot hold all int64). */
@@ -109,7 +111,9 @@ int hf_map_scalar_i64(int op, const hf_col* in, int64_t scalar, hf_col** out);
  * Device form of algebra/binary.py:293-459 frame branch (n_ary_op zip-apply,
  * dataframe.py:3851) for two row-aligned columns of equal length. */
 enum {
-  HF_BIN_ADD = 0, HF_BIN_SUB = 1, HF_BIN_MUL = 2, HF_BIN_DIV = 3
+  HF_BIN_ADD = 0, HF_BIN_SUB = 1, HF_BIN_MUL = 2, HF_BIN_DIV = 3,
+  HF_BIN_MIN = 4, /* fmin / integer min — NaN-skipping (axis=1 folds)   */
+  HF_BIN_MAX = 5  /* fmax / integer max                                  */
 };
 int hf_binary(int op, const hf_col* a, const hf_col* b, hf_col** out);
 

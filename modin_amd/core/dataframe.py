@@ -939,6 +939,77 @@ class HipDataframe:
         return HipDataframe([part], self._index, self.columns, [n],
                             pandas.Series(dtypes))
 
+    def clip_columns(self, lower, upper) -> "HipDataframe":
+        """pandas clip(axis=None): per-column fmax(lower) then
+        fmin(upper); int columns stay int64 for integral bounds (the
+        pandas dtype rule rides the map wrapper's promotion)."""
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        if blk_cats:
+            raise lib.HfError("clip over string columns")
+
+        def block_fn(block: DeviceBlock) -> DeviceBlock:
+            out = {}
+            for name, col in block.columns.items():
+                c = col
+                if lower is not None:
+                    c = lib.map_scalar(lib.MAP_MAX, c, lower)
+                if upper is not None:
+                    c = lib.map_scalar(lib.MAP_MIN, c, upper)
+                out[name] = c
+            return DeviceBlock(out, block.length)
+        return self.map(block_fn)
+
+    def reduce_axis1(self, op: str) -> "HipDataframe":
+        """Row-wise reductions (pandas axis=1): fold the columns with the
+        elementwise kernels — sum/count fold ADD over NaN-zeroed values /
+        NOTNA masks, min/max fold fmin/fmax (NaN-skipping), mean =
+        sum/count.  Returns a 1-column frame aligned with the rows."""
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        if blk_cats:
+            raise lib.HfError(f"{op}(axis=1) over string columns")
+        all_int = all(self.dtypes[c] == np.dtype(np.int64)
+                      for c in self.columns)
+
+        def block_fn(block: DeviceBlock) -> DeviceBlock:
+            cols = list(block.columns.values())
+            if op in ("min", "max"):
+                bop = lib.BIN_MIN if op == "min" else lib.BIN_MAX
+                acc = cols[0]
+                for c in cols[1:]:
+                    acc = lib.binary(bop, acc, c)
+                return DeviceBlock({"\x00r\x00": acc}, block.length)
+            sums = None
+            cnts = None
+            for c in cols:
+                f = lib.cast_f64(c) if c.dtype_code == lib.HF_INT64 else c
+                fz = lib.map_scalar(lib.MAP_FILLNA, f, 0.0)
+                sums = fz if sums is None else lib.binary(lib.BIN_ADD,
+                                                          sums, fz)
+                if op in ("mean", "count"):
+                    m = lib.compare_scalar(lib.CMP_NOTNA, c, 0.0)
+                    cnts = m if cnts is None else lib.binary(lib.BIN_ADD,
+                                                             cnts, m)
+            if op == "count":
+                return DeviceBlock({"\x00r\x00": cnts}, block.length)
+            if op == "mean":
+                return DeviceBlock(
+                    {"\x00r\x00": lib.binary(lib.BIN_DIV, sums,
+                                              lib.cast_f64(cnts))},
+                    block.length)
+            if all_int:  # exact: int sums (no NaN possible)
+                sums = lib.map_scalar(lib.MAP_CAST_I64, sums, 0)
+            return DeviceBlock({"\x00r\x00": sums}, block.length)
+
+        res = self.map(block_fn)
+        dt = (np.dtype(np.int64)
+              if (op == "count" or (all_int and op in ("sum", "min",
+                                                       "max")))
+              else np.dtype(np.float64))
+        res.dtypes = pandas.Series({"\x00r\x00": dt})
+        return res
+
     def idx_extreme(self, maximum: bool) -> dict:
         """Per-column idxmax/idxmin: the FIRST original position holding
         the column's max/min (NaN skipped) — one cached reduce + an EQ

@@ -45,9 +45,9 @@ HF_FLOAT64 = 1
 
 # map ops
 MAP_ADD, MAP_SUB, MAP_RSUB, MAP_MUL, MAP_DIV, MAP_RDIV, MAP_FILLNA, MAP_ABS, \
-    MAP_NEG, MAP_CAST_F64, MAP_CAST_I64, MAP_SQRT = range(12)
+    MAP_NEG, MAP_CAST_F64, MAP_CAST_I64, MAP_SQRT, MAP_MIN, MAP_MAX = range(14)
 # binary ops
-BIN_ADD, BIN_SUB, BIN_MUL, BIN_DIV = range(4)
+BIN_ADD, BIN_SUB, BIN_MUL, BIN_DIV, BIN_MIN, BIN_MAX = range(6)
 
 _NP_TO_HF = {np.dtype(np.int64): HF_INT64, np.dtype(np.float64): HF_FLOAT64}
 _HF_TO_NP = {HF_INT64: np.dtype(np.int64), HF_FLOAT64: np.dtype(np.float64)}
@@ -318,7 +318,7 @@ def map_scalar(op: int, col: ColumnRef, scalar) -> ColumnRef:
     if col.dtype_code == HF_INT64 and op in (MAP_DIV, MAP_RDIV, MAP_FILLNA):
         col = cast_f64(col)  # pandas promotes int div to float; fillna no-ops
     elif (col.dtype_code == HF_INT64
-          and op in (MAP_ADD, MAP_SUB, MAP_RSUB, MAP_MUL)
+          and op in (MAP_ADD, MAP_SUB, MAP_RSUB, MAP_MUL, MAP_MIN, MAP_MAX)
           and isinstance(scalar, float) and not scalar.is_integer()):
         col = cast_f64(col)  # pandas: int64 op non-integral float -> float64
     if op == MAP_CAST_I64:

@@ -249,6 +249,8 @@ __device__ __forceinline__ double map1_f64(double x, double s) {
     case HF_MAP_ABS:    return fabs(x);
     case HF_MAP_NEG:    return -x;
     case HF_MAP_SQRT:   return sqrt(x);
+    case HF_MAP_MIN:    return fmin(x, s);
+    case HF_MAP_MAX:    return fmax(x, s);
   }
   return x;
 }
@@ -281,6 +283,8 @@ __device__ __forceinline__ int64_t map1_i64(int64_t x, int64_t s) {
     case HF_MAP_RSUB: return s - x;
     case HF_MAP_MUL:  return x * s;
     case HF_MAP_ABS:  return x < 0 ? -x : x;
+    case HF_MAP_MIN:  return x < s ? x : s;
+    case HF_MAP_MAX:  return x > s ? x : s;
     case HF_MAP_NEG:  return -x;
   }
   return x;
@@ -325,6 +329,19 @@ __global__ void __launch_bounds__(BLOCK) k_cast_f64_i64(const double* __restrict
 // kernels: Binary (elementwise column op column) — algebra/binary.py device form
 // ---------------------------------------------------------------------------
 
+__device__ __forceinline__ double hf_minv(double a, double b) {
+  return fmin(a, b);  // NaN-skipping: pandas axis=1 min/max fold
+}
+__device__ __forceinline__ int64_t hf_minv(int64_t a, int64_t b) {
+  return a < b ? a : b;
+}
+__device__ __forceinline__ double hf_maxv(double a, double b) {
+  return fmax(a, b);
+}
+__device__ __forceinline__ int64_t hf_maxv(int64_t a, int64_t b) {
+  return a > b ? a : b;
+}
+
 template <int OP, typename T>
 __device__ __forceinline__ T bin1(T a, T b) {
   switch (OP) {
@@ -332,6 +349,8 @@ __device__ __forceinline__ T bin1(T a, T b) {
     case HF_BIN_SUB: return a - b;
     case HF_BIN_MUL: return a * b;
     case HF_BIN_DIV: return a / b;
+    case HF_BIN_MIN: return hf_minv(a, b);
+    case HF_BIN_MAX: return hf_maxv(a, b);
   }
   return a;
 }
@@ -2065,6 +2084,8 @@ int hf_map_scalar(int op, const hf_col* in, double scalar, hf_col** out) {
     case HF_MAP_ABS:    rc = launch_map_f64<HF_MAP_ABS>(in, scalar, *out); break;
     case HF_MAP_NEG:    rc = launch_map_f64<HF_MAP_NEG>(in, scalar, *out); break;
     case HF_MAP_SQRT:   rc = launch_map_f64<HF_MAP_SQRT>(in, scalar, *out); break;
+    case HF_MAP_MIN:    rc = launch_map_f64<HF_MAP_MIN>(in, scalar, *out); break;
+    case HF_MAP_MAX:    rc = launch_map_f64<HF_MAP_MAX>(in, scalar, *out); break;
     default:
       hf_col_free(*out);
       *out = nullptr;
@@ -2088,6 +2109,8 @@ int hf_map_scalar_i64(int op, const hf_col* in, int64_t scalar, hf_col** out) {
     case HF_MAP_MUL:  rc = launch_map_i64<HF_MAP_MUL>(in, scalar, *out); break;
     case HF_MAP_ABS:  rc = launch_map_i64<HF_MAP_ABS>(in, scalar, *out); break;
     case HF_MAP_NEG:  rc = launch_map_i64<HF_MAP_NEG>(in, scalar, *out); break;
+    case HF_MAP_MIN:  rc = launch_map_i64<HF_MAP_MIN>(in, scalar, *out); break;
+    case HF_MAP_MAX:  rc = launch_map_i64<HF_MAP_MAX>(in, scalar, *out); break;
     default:
       hf_col_free(*out);
       *out = nullptr;
@@ -2138,6 +2161,10 @@ int hf_binary(int op, const hf_col* a, const hf_col* b, hf_col** out) {
     case HF_BIN_MUL: rc = f64 ? Lf(std::integral_constant<int, HF_BIN_MUL>{})
                               : Li(std::integral_constant<int, HF_BIN_MUL>{}); break;
     case HF_BIN_DIV: rc = Lf(std::integral_constant<int, HF_BIN_DIV>{}); break;
+    case HF_BIN_MIN: rc = f64 ? Lf(std::integral_constant<int, HF_BIN_MIN>{})
+                              : Li(std::integral_constant<int, HF_BIN_MIN>{}); break;
+    case HF_BIN_MAX: rc = f64 ? Lf(std::integral_constant<int, HF_BIN_MAX>{})
+                              : Li(std::integral_constant<int, HF_BIN_MAX>{}); break;
     default:
       hf_col_free(*out);
       *out = nullptr;

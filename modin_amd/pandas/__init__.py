@@ -54,20 +54,31 @@ class _HipPandasBase:
     _query_compiler: HipQueryCompiler
 
     # ---- reductions ----
+    def _reduce(self, name, **kwargs):
+        axis = kwargs.pop("axis", 0)
+        if axis in (1, "columns"):
+            qc = self._query_compiler.reduce_axis1(name)
+            out = Series(query_compiler=qc, name=None)
+            return out
+        return self._lower(getattr(self._query_compiler, name)(**kwargs))
+
     def sum(self, **kwargs):
-        return self._lower(self._query_compiler.sum(**kwargs))
+        return self._reduce("sum", **kwargs)
 
     def mean(self, **kwargs):
-        return self._lower(self._query_compiler.mean(**kwargs))
+        return self._reduce("mean", **kwargs)
 
     def count(self, **kwargs):
-        return self._lower(self._query_compiler.count(**kwargs))
+        return self._reduce("count", **kwargs)
 
     def min(self, **kwargs):
-        return self._lower(self._query_compiler.min(**kwargs))
+        return self._reduce("min", **kwargs)
 
     def max(self, **kwargs):
-        return self._lower(self._query_compiler.max(**kwargs))
+        return self._reduce("max", **kwargs)
+
+    def clip(self, lower=None, upper=None):
+        return self._rewrap(self._query_compiler.clip(lower, upper))
 
     def median(self, **kwargs):
         return self._lower(self._query_compiler.median())

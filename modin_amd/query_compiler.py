@@ -127,6 +127,13 @@ class HipQueryCompiler:
         vals = self._modin_frame.median_columns()
         return pandas.Series(vals, dtype=np.float64)
 
+    def clip(self, lower=None, upper=None) -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.clip_columns(lower, upper))
+
+    def reduce_axis1(self, op: str) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.reduce_axis1(op))
+
     def cumsum(self) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.cumsum_rows(lib.AGG_SUM))
 

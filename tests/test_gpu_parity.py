@@ -1045,3 +1045,40 @@ def test_groupby_quantile_vs_pandas(npartitions):
                                    exp["v"].to_numpy(), rtol=1e-12,
                                    atol=1e-12, equal_nan=True,
                                    err_msg=str(q))
+
+
+def test_clip_axis1_vs_pandas(npartitions):
+    """clip bounds and axis=1 reductions (sum/mean/min/max/count row
+    folds, NaN-skipping) vs pandas."""
+    rng = np.random.default_rng(92)
+    n = 30_000
+    a = rng.standard_normal(n) * 10
+    a[rng.random(n) < 0.1] = np.nan
+    b = rng.standard_normal(n) * 10
+    b[rng.random(n) < 0.1] = np.nan
+    w = rng.integers(-50, 50, n)
+    pdf = pandas.DataFrame({"a": a, "b": b, "w": w})
+    df = mpd.DataFrame(pdf)
+    got = df.clip(-5, 5).to_pandas()
+    exp = pdf.clip(-5, 5)
+    assert list(got.dtypes) == list(exp.dtypes)
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(), rtol=0,
+                               equal_nan=True)
+    got = df.clip(lower=0).to_pandas()
+    exp = pdf.clip(lower=0)
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(), rtol=0,
+                               equal_nan=True)
+    for op in ("sum", "mean", "min", "max", "count"):
+        g = getattr(df, op)(axis=1).to_pandas()
+        e = getattr(pdf, op)(axis=1)
+        np.testing.assert_allclose(g.to_numpy().astype(float),
+                                   e.to_numpy().astype(float), rtol=1e-12,
+                                   atol=1e-12, equal_nan=True,
+                                   err_msg=op)
+    # all-int frame keeps int64 sums/extremes
+    pint = pandas.DataFrame({"x": w, "y": rng.integers(0, 9, n)})
+    dint = mpd.DataFrame(pint)
+    gs = dint.sum(axis=1).to_pandas()
+    assert gs.dtype == np.int64
+    np.testing.assert_array_equal(gs.to_numpy(),
+                                  pint.sum(axis=1).to_numpy())
