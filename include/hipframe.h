@@ -314,6 +314,23 @@ int hf_ordered_i64(const hf_col* col, int direction, hf_col** out);
  * apply. */
 int hf_cumsum(const hf_col* col, int agg_op, hf_col** out);
 
+/* SEGMENTED inclusive prefix scan: restart at every row whose `heads` entry
+ * is nonzero (int64 0/1 column, same length).  The device form of the
+ * pandas groupby transform family (DataFrameGroupBy.cumsum/cummin/cummax,
+ * modin/pandas/groupby.py) applied after a stable sort by key: key-run
+ * head flags delimit the segments.  NaN rows stay NaN and contribute the
+ * identity (pandas skipna); int64 is exact.  The combine is the classic
+ * segmented-scan pair operator ((fa,va),(fb,vb)) -> (fa|fb, fb ? vb :
+ * comb(va,vb)) — associative, not commutative, so every fold is ordered. */
+int hf_seg_cumsum(const hf_col* col, const hf_col* heads, int agg_op,
+                  hf_col** out);
+
+/* Inverse-permutation scatter: out[idx[i]] = col[i]; idx must be a
+ * permutation of [0, len) (a hf_sort_perm result).  Restores original row
+ * order after a sort-then-transform composition — the device analog of the
+ * reference reindexing a transform result back to the caller's index. */
+int hf_scatter(const hf_col* col, const hf_col* idx, hf_col** out);
+
 /* Exact-match binary search: out[i] = j with sorted[j] == keys[i], else -1.
  * Densifies unbounded int64 join keys through the sorted distinct right
  * keys (lower_bound per row), so the dense-range CSR join

@@ -1140,6 +1140,221 @@ class HipDataframe:
         dtypes = pandas.Series({v: np.dtype(np.int64) for v in val_names})
         return HipDataframe([part], res0._index, val_names, [ng], dtypes)
 
+    # ---- groupby transforms (same-length results in ORIGINAL row order:
+    # pandas DataFrameGroupBy.cumsum/cummin/cummax/cumcount/rank;
+    # reference: modin/pandas/groupby.py fallback through
+    # _wrap_aggregation / default2pandas transform) ----
+    @staticmethod
+    def _run_head_col(col, n):
+        """head[i] = (col[i] != col[i-1]) over an EXACT-comparable (i64 or
+        NaN-free f64) column; row 0 is always a head (the shifted-in
+        sentinel differs by construction)."""
+        prev_body = lib.col_slice(col, 0, n - 1)
+        first = lib.alloc(1, col.dtype_code)
+        if col.dtype_code == lib.HF_FLOAT64:
+            lib.fill_f64(first.dptr(), float("nan"), 1)
+        else:
+            v0 = int(lib.get(lib.col_slice(col, 0, 1))[0])
+            lib.fill_i64(first.dptr(), v0 ^ 1, 1)
+        prev = lib.concat([first, prev_body])
+        d = lib.binary(lib.BIN_SUB, col, prev)
+        return lib.compare_scalar(lib.CMP_NE, d, 0.0)
+
+    @staticmethod
+    def _iota(n):
+        """Device [0, n) int64 column (an all-ones filter plan's kept
+        positions)."""
+        ones = lib.alloc(n, lib.HF_INT64)
+        lib.fill_i64(ones.dptr(), 1, n)
+        plan = lib.filter_plan(ones)
+        return lib.filter_iota(plan, 0)
+
+    @staticmethod
+    def _const_i64(value):
+        c = lib.alloc(1, lib.HF_INT64)
+        lib.fill_i64(c.dptr(), int(value), 1)
+        return c
+
+    def groupby_transform(self, by, how: str, ascending: bool = True,
+                          method: str = "average") -> "HipDataframe":
+        """Same-length groupby transforms in original row order.
+
+        how: 'cumsum' | 'cummin' | 'cummax' (segmented scan), 'cumcount'
+        (position within group), 'rank' (method 'average'|'min'|'first',
+        ascending per pandas; na_option='keep').
+
+        Device composition: stable sort by key (LSD radix keeps original
+        order within groups), key-run head flags delimit the segments, the
+        segmented scan / position arithmetic runs in sorted order, and the
+        inverse-permutation scatter restores original order.  Rows whose
+        key is NaN belong to no group (pandas dropna=True) and come back
+        NaN — which also forces float64 results, exactly pandas' dtype
+        rule (int64 stays int64 only when every key is valid).
+
+        Reference semantics pinned against pandas 2.3.3 in-container
+        (NaN keys -> NaN in every transform incl. cumcount; NaN values
+        stay NaN and don't advance cum* state; rank na_option='keep')."""
+        from ..distributed import is_active
+        if is_active():
+            raise lib.HfError("distributed groupby transforms are a later "
+                              "round")
+        by_list = [by] if isinstance(by, str) else list(by)
+        for b in by_list:
+            if b not in self.columns:
+                raise lib.HfError(f"groupby: key column {b!r} missing")
+        if how not in ("cumsum", "cummin", "cummax", "cumcount", "rank"):
+            raise lib.HfError(f"groupby transform {how!r} not supported")
+        if how == "rank" and method not in ("average", "min", "first"):
+            raise lib.HfError(f"rank method {method!r} not supported")
+        if not isinstance(self._index, pandas.RangeIndex) or \
+                self._index.start != 0 or self._index.step != 1:
+            raise lib.HfError("groupby transforms: only RangeIndex frames "
+                              "this round")
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        val_names = [c for c in self.columns if c not in by_list]
+        for v in val_names:
+            if v in blk_cats:
+                raise lib.HfError(
+                    f"groupby {how}: string column {v!r} unsupported "
+                    "(pandas raises on non-numeric transforms)")
+        n = len(self)
+        if n == 0 or not val_names:
+            dts = pandas.Series({v: np.dtype(np.float64)
+                                 for v in val_names})
+            part = HipDataframePartition(DeviceBlock(
+                {v: lib.alloc(0, lib.HF_FLOAT64) for v in val_names}, 0))
+            return HipDataframe([part], pandas.RangeIndex(0), val_names,
+                                [0], dts)
+
+        def concat_col(name):
+            cols = [p.block().columns[name] for p in self._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        # per-row key validity in ORIGINAL order (pandas dropna=True)
+        valid = None
+        for b in by_list:
+            c = concat_col(b)
+            if b in blk_cats:
+                m = lib.compare_scalar(lib.CMP_GE, c, 0.0)
+            elif c.dtype_code == lib.HF_FLOAT64:
+                m = lib.compare_scalar(lib.CMP_NOTNA, c, 0.0)
+            else:
+                continue
+            valid = m if valid is None else lib.binary(lib.BIN_MUL, valid, m)
+        if valid is not None and lib.reduce(valid).isum == n:
+            valid = None  # every key valid — int dtypes survive
+        eff_keys = [self._effective_sort_key(concat_col(b), b in blk_cats,
+                                             True)
+                    for b in by_list]
+        if how == "rank":
+            return self._groupby_rank(by_list, val_names, eff_keys, valid,
+                                      n, concat_col, ascending, method)
+        perm = self._compose_sort_perm(eff_keys)
+        head = None
+        for ekc, _ in eff_keys:
+            h = self._run_head_col(lib.gather(ekc, perm), n)
+            head = h if head is None else lib.binary(lib.BIN_ADD, head, h)
+        if len(eff_keys) > 1:
+            head = lib.compare_scalar(lib.CMP_GE, head, 1.0)
+        out_cols, dts = {}, {}
+        if how == "cumcount":
+            plan = lib.filter_plan(head)
+            hp = lib.filter_iota(plan, 0)           # run start positions
+            rid = lib.map_scalar(lib.MAP_SUB, lib.cumsum(head), 1)
+            start = lib.gather(hp, rid)
+            cc = lib.binary(lib.BIN_SUB, self._iota(n), start)
+            res = lib.scatter(cc, perm)
+            if valid is not None:
+                res = lib.fixup_empty(lib.cast_f64(res), valid)
+            name = "cumcount"
+            part = HipDataframePartition(DeviceBlock({name: res}, n))
+            dt = np.dtype(np.int64 if valid is None else np.float64)
+            return HipDataframe([part], pandas.RangeIndex(n), [name], [n],
+                                pandas.Series({name: dt}))
+        agg_op = {"cumsum": lib.AGG_SUM, "cummin": lib.AGG_MIN,
+                  "cummax": lib.AGG_MAX}[how]
+        for v in val_names:
+            vc = concat_col(v)
+            src_int = vc.dtype_code == lib.HF_INT64
+            if src_int and valid is not None:
+                vc = lib.cast_f64(vc)
+            sv = lib.gather(vc, perm)
+            seg = lib.seg_cumsum(sv, head, agg_op)
+            res = lib.scatter(seg, perm)
+            if valid is not None:
+                res = lib.fixup_empty(res, valid)
+            out_cols[v] = res
+            dts[v] = np.dtype(np.int64 if (src_int and valid is None)
+                              else np.float64)
+        part = HipDataframePartition(DeviceBlock(out_cols, n))
+        return HipDataframe([part], pandas.RangeIndex(n), val_names, [n],
+                            pandas.Series(dts))
+
+    def _groupby_rank(self, by_list, val_names, eff_keys, valid, n,
+                      concat_col, ascending, method):
+        """rank within groups (pandas DataFrameGroupBy.rank,
+        na_option='keep'): per value column, sort by (keys…, value) with
+        the effective-key transform (NaN value sorts last), 1-based
+        position within the key run, tie runs collapsed per `method`
+        ('average' -> first + (len-1)/2, 'min' -> first, 'first' -> the
+        position itself)."""
+        out_cols = {}
+        for v in val_names:
+            vc = concat_col(v)
+            eff_v = self._effective_sort_key(vc, False, ascending)
+            perm = self._compose_sort_perm(eff_keys + [eff_v])
+            khead = None
+            for ekc, _ in eff_keys:
+                h = self._run_head_col(lib.gather(ekc, perm), n)
+                khead = h if khead is None else lib.binary(lib.BIN_ADD,
+                                                           khead, h)
+            if len(eff_keys) > 1:
+                khead = lib.compare_scalar(lib.CMP_GE, khead, 1.0)
+            kplan = lib.filter_plan(khead)
+            hp = lib.filter_iota(kplan, 0)
+            rid = lib.map_scalar(lib.MAP_SUB, lib.cumsum(khead), 1)
+            start = lib.gather(hp, rid)
+            pos1 = lib.map_scalar(
+                lib.MAP_ADD,
+                lib.binary(lib.BIN_SUB, self._iota(n), start), 1)
+            if method == "first":
+                rank_s = lib.cast_f64(pos1)
+            else:
+                vhead = self._run_head_col(lib.gather(eff_v[0], perm), n)
+                thead = lib.compare_scalar(
+                    lib.CMP_GE, lib.binary(lib.BIN_ADD, khead, vhead), 1.0)
+                tplan = lib.filter_plan(thead)
+                nt = tplan.n_kept
+                tstarts = lib.filter_iota(tplan, 0)
+                first_pos = lib.filter_apply(tplan, pos1)
+                if method == "average":
+                    nxt = lib.concat([lib.col_slice(tstarts, 1, nt - 1),
+                                      self._const_i64(n)]) if nt > 1 \
+                        else self._const_i64(n)
+                    tlen = lib.binary(lib.BIN_SUB, nxt, tstarts)
+                    avg = lib.binary(
+                        lib.BIN_ADD, lib.cast_f64(first_pos),
+                        lib.map_scalar(
+                            lib.MAP_DIV,
+                            lib.cast_f64(lib.map_scalar(lib.MAP_SUB,
+                                                        tlen, 1)), 2.0))
+                else:
+                    avg = lib.cast_f64(first_pos)
+                trid = lib.map_scalar(lib.MAP_SUB, lib.cumsum(thead), 1)
+                rank_s = lib.gather(avg, trid)
+            res = lib.scatter(rank_s, perm)
+            if vc.dtype_code == lib.HF_FLOAT64:
+                notna_v = lib.compare_scalar(lib.CMP_NOTNA, vc, 0.0)
+                res = lib.fixup_empty(res, notna_v)
+            if valid is not None:
+                res = lib.fixup_empty(res, valid)
+            out_cols[v] = res
+        part = HipDataframePartition(DeviceBlock(out_cols, n))
+        dts = pandas.Series({v: np.dtype(np.float64) for v in val_names})
+        return HipDataframe([part], pandas.RangeIndex(n), val_names, [n],
+                            dts)
+
     def hconcat(self, others: list) -> "HipDataframe":
         """Horizontal compose of single-partition frames with identical row
         count and index (the device form of the reference's axis=1 concat
@@ -1582,13 +1797,15 @@ class HipDataframe:
     # ---- sort (PandasDataframe.sort_by device form, dataframe.py:2742;
     #      SURVEY §8f.2): stable radix permutation + column gathers ----
     @staticmethod
-    def _effective_sort_key(col, is_dict, ascending):
+    def _effective_sort_key(col, is_dict, ascending, na_first=False):
         """(key col, ascending) -> (int64 key, ascending') whose stable
         ASCENDING' radix sort realizes pandas order.  NaNs — dictionary
-        code −1 or float NaN — sort LAST for both directions
-        (na_position='last'): the NaN rows map to +2^62 and valid keys
-        keep (asc) or negate (desc) their order, so the pass always runs
-        ascending when an adjustment is needed.  Float keys first ride the
+        code −1 or float NaN — sort LAST for both directions by default
+        (na_position='last'): the NaN rows map to a sentinel above every
+        valid key and valid keys keep (asc) or negate (desc) their order,
+        so the pass always runs ascending when an adjustment is needed.
+        na_first=True (na_position='first') mirrors the sentinel BELOW
+        every valid key instead.  Float keys first ride the
         order-preserving f64->i64 bit transform (hf_ordered_i64)."""
         BIG = 1 << 62
         if col.dtype_code == lib.HF_FLOAT64:
@@ -1596,9 +1813,9 @@ class HipDataframe:
             okey = lib.ordered_i64(col)
             if r is None or r.count == col.length:  # no NaN
                 return okey, ascending
-            # NaN sentinel must exceed ordered(+inf) ~ 0x7FF0... ~ 9.22e18
-            # (2^62 would land BELOW the ordered bits of floats >= 2.0)
-            NANKEY = (1 << 63) - 1
+            # NaN sentinel must clear ordered(±inf) ~ ±0x7FF0... ~ 9.22e18
+            # (2^62 would land INSIDE the ordered bits of floats >= 2.0)
+            NANKEY = ((1 << 63) - 1) * (-1 if na_first else 1)
             notna = lib.compare_scalar(lib.CMP_NOTNA, col, 0.0)
             isna_big = lib.map_scalar(
                 lib.MAP_MUL, lib.map_scalar(lib.MAP_RSUB, notna, 1),
@@ -1610,12 +1827,16 @@ class HipDataframe:
                               isna_big), True
         if not is_dict or not col.length or lib.reduce(col).imn >= 0:
             return col, ascending
+        if ascending and na_first:
+            return col, ascending  # code −1 already sorts first
         m = lib.compare_scalar(lib.CMP_EQ, col, -1.0)
         if ascending:
             t = lib.map_scalar(lib.MAP_MUL, m, BIG + 1)
             return lib.binary(lib.BIN_ADD, col, t), True
         neg = lib.map_scalar(lib.MAP_NEG, col, 0)
-        t = lib.map_scalar(lib.MAP_MUL, m, BIG - 1)
+        # descending: NaN code −1 negates to +1; push it far above (last)
+        # or far below (first) the negated valid codes
+        t = lib.map_scalar(lib.MAP_MUL, m, -BIG if na_first else BIG - 1)
         return lib.binary(lib.BIN_ADD, neg, t), True
 
     @staticmethod
@@ -1633,7 +1854,12 @@ class HipDataframe:
                 perm = lib.gather(perm, p2)
         return perm
 
-    def sort_rows(self, by, ascending=True) -> "HipDataframe":
+    def sort_rows(self, by, ascending=True,
+                  na_position: str = "last") -> "HipDataframe":
+        if na_position not in ("last", "first"):
+            raise lib.HfError("sort_values: na_position must be 'last' or "
+                              "'first'")
+        na_first = na_position == "first"
         by_list = [by] if isinstance(by, str) else list(by)
         if isinstance(ascending, (bool, np.bool_, int)):
             asc_list = [bool(ascending)] * len(by_list)
@@ -1658,7 +1884,7 @@ class HipDataframe:
         from ..distributed import is_active
         if is_active():
             return self._sort_rows_distributed(by_list, asc_list, blk_cats,
-                                               concat_col)
+                                               concat_col, na_first)
         cache = {}
 
         def cat_col(name):
@@ -1666,7 +1892,8 @@ class HipDataframe:
                 cache[name] = concat_col(name)
             return cache[name]
 
-        eff = [self._effective_sort_key(cat_col(b), b in blk_cats, a)
+        eff = [self._effective_sort_key(cat_col(b), b in blk_cats, a,
+                                        na_first)
                for b, a in zip(by_list, asc_list)]
         perm = self._compose_sort_perm(eff)
         cols = {name: lib.gather(cat_col(name), perm)
@@ -1677,7 +1904,7 @@ class HipDataframe:
                             self.columns, [n], self.dtypes)
 
     def _sort_rows_distributed(self, by_list, asc_list, blk_cats,
-                               concat_col):
+                               concat_col, na_first=False):
         """Distributed sort_values: the range-partitioning shuffle + local
         stable sort (SURVEY §8f.2 "reuses the shuffle"; reference
         sort_by -> _apply_func_to_range_partitioning, dataframe.py:2742).
@@ -1693,7 +1920,7 @@ class HipDataframe:
         P = dist_mod.world_size()
         k0 = concat_col(by_list[0])
         ek0, ea0 = self._effective_sort_key(k0, by_list[0] in blk_cats,
-                                            asc_list[0])
+                                            asc_list[0], na_first)
         n = ek0.length
         S = min(n, 4096)
         if S:
@@ -1721,7 +1948,8 @@ class HipDataframe:
                                             send_counts)
                 for m in names}
         rpos = dist_mod.exchange_column(lib.concat(send_pos), send_counts)
-        eff = [self._effective_sort_key(recv[b], b in blk_cats, a)
+        eff = [self._effective_sort_key(recv[b], b in blk_cats, a,
+                                        na_first)
                for b, a in zip(by_list, asc_list)]
         perm = self._compose_sort_perm(eff)
         out_cols = {m: lib.gather(recv[m], perm) for m in names}

@@ -133,6 +133,10 @@ def load() -> ct.CDLL:
                                           ct.POINTER(ct.c_void_p)]),
             "hf_cumsum": (ct.c_int, [ct.c_void_p, ct.c_int,
                                      ct.POINTER(ct.c_void_p)]),
+            "hf_seg_cumsum": (ct.c_int, [ct.c_void_p, ct.c_void_p, ct.c_int,
+                                         ct.POINTER(ct.c_void_p)]),
+            "hf_scatter": (ct.c_int, [ct.c_void_p, ct.c_void_p,
+                                      ct.POINTER(ct.c_void_p)]),
             "hf_shuffle_dest": (ct.c_int, [ct.c_void_p,
                                            ct.POINTER(ct.c_int64), ct.c_int,
                                            ct.POINTER(ct.c_void_p)]),
@@ -197,7 +201,8 @@ def exported_symbols():
         "hf_fixup_empty", "hf_sort_perm", "hf_fill_i64",
         "hf_groupby_hash_accum", "hf_groupby_hash_compact",
         "hf_groupby_sorted", "hf_shuffle_dest", "hf_memcpy_dd",
-        "hf_search_sorted", "hf_ordered_i64", "hf_cumsum",
+        "hf_search_sorted", "hf_ordered_i64", "hf_cumsum", "hf_seg_cumsum",
+        "hf_scatter",
         "hf_col_concat", "hf_col_slice", "hf_join_build", "hf_join_free", "hf_join_probe",
         "hf_gather", "hf_compare_scalar", "hf_filter_plan", "hf_filter_apply",
         "hf_filter_iota", "hf_filter_plan_free", "hf_profiling",
@@ -615,6 +620,28 @@ def cumsum(col: ColumnRef, agg_op: int = 0) -> ColumnRef:
     out = ct.c_void_p()
     _check(load().hf_cumsum(col.handle, agg_op, ct.byref(out)),
            "hf_cumsum")
+    return _wrap(out, col.length, col.dtype_code)
+
+
+def seg_cumsum(col: ColumnRef, heads: ColumnRef,
+               agg_op: int = 0) -> ColumnRef:
+    """Segmented inclusive prefix scan: restart at rows whose `heads` entry
+    is nonzero (groupby cumsum/cummin/cummax after a stable sort by key).
+    f64 skips NaN, i64 exact."""
+    ensure_ready()
+    out = ct.c_void_p()
+    _check(load().hf_seg_cumsum(col.handle, heads.handle, agg_op,
+                                ct.byref(out)), "hf_seg_cumsum")
+    return _wrap(out, col.length, col.dtype_code)
+
+
+def scatter(col: ColumnRef, idx: ColumnRef) -> ColumnRef:
+    """out[idx[i]] = col[i]; idx must be a permutation of [0, len) — the
+    inverse of gather(col, idx), restoring pre-sort row order."""
+    ensure_ready()
+    out = ct.c_void_p()
+    _check(load().hf_scatter(col.handle, idx.handle, ct.byref(out)),
+           "hf_scatter")
     return _wrap(out, col.length, col.dtype_code)
 
 

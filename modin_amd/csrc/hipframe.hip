@@ -1888,6 +1888,209 @@ __global__ void __launch_bounds__(BLOCK) k_gather_i64(
   for (; i < n; i += stride) out[i] = src[idx[i]];
 }
 
+// Inverse of gather: out[idx[i]] = src[i].  idx must be a permutation of
+// [0, n) (the caller holds a sort_perm) so writes never collide; reads are
+// coalesced and the 8 B scattered writes coalesce in L2 like the radix
+// scatter's do.
+template <typename T>
+__global__ void __launch_bounds__(BLOCK) k_scatter(
+    const T* __restrict__ src, const int64_t* __restrict__ idx,
+    T* __restrict__ out, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[idx[i]] = src[i];
+}
+
+// ---------------------------------------------------------------------------
+// Segmented scan (groupby transforms: cumsum/cummin/cummax WITHIN key runs
+// of a key-sorted column — the device form of pandas
+// DataFrameGroupBy.cumsum, groupby.py "transform" family).
+// State is the pair (f, v) — "range contains a segment head", "aggregate of
+// the range's tail segment" — under
+//   seg_comb((fa,va), (fb,vb)) = (fa|fb, fb ? vb : cs_comb(va,vb))
+// which is associative but NOT commutative: every fold/scan below is an
+// ordered left-to-right pass (the plain cumsum kernels' strided folds and
+// shfl tree reductions are commutative-only and cannot be reused).
+// NaN elements (f64) emit NaN at their position and contribute the identity
+// to the running state (pandas cum* skipna); min/max propagate through the
+// same cs_* helpers as hf_cumsum.  Three phases like hf_cumsum: per-tile
+// summaries, one-block exclusive tile scan, per-element apply.
+// ---------------------------------------------------------------------------
+template <typename T, int OP>
+__global__ void __launch_bounds__(BLOCK) k_seg_tiles(
+    const T* __restrict__ in, const int64_t* __restrict__ heads, int64_t n,
+    T* __restrict__ tile_v, unsigned* __restrict__ tile_f) {
+  const int64_t t0 = (int64_t)blockIdx.x * FILT_TILE;
+  const int64_t t1 = min(t0 + (int64_t)FILT_TILE, n);
+  constexpr int PER = FILT_TILE / BLOCK;
+  __shared__ T sv[FILT_TILE];
+  __shared__ unsigned char sf[FILT_TILE];
+  for (int j = 0; j < PER; ++j) {
+    const int64_t i = t0 + (int64_t)j * BLOCK + threadIdx.x;
+    if (i < t1) {
+      sv[(int)(i - t0)] = in[i];
+      sf[(int)(i - t0)] = heads[i] != 0;
+    }
+  }
+  __syncthreads();
+  const int s0 = threadIdx.x * PER;
+  const int lim = (int)(t1 - t0);
+  unsigned char f = 0;
+  T v = cs_ident<T, OP>();
+  for (int j = 0; j < PER; ++j) {
+    if (s0 + j < lim) {
+      const T x = sv[s0 + j];
+      const T xv = (x != x) ? cs_ident<T, OP>() : x;
+      if (sf[s0 + j]) { f = 1; v = xv; }
+      else v = cs_comb<T, OP>(v, xv);
+    }
+  }
+  __shared__ T bv[BLOCK];
+  __shared__ unsigned char bf[BLOCK];
+  bv[threadIdx.x] = v;
+  bf[threadIdx.x] = f;
+  __syncthreads();
+  for (int off = 1; off < BLOCK; off <<= 1) {
+    T pv = cs_ident<T, OP>();
+    unsigned char pf = 0;
+    if (threadIdx.x >= off) { pv = bv[threadIdx.x - off];
+                              pf = bf[threadIdx.x - off]; }
+    __syncthreads();
+    const unsigned char nf = pf | bf[threadIdx.x];
+    const T nv = bf[threadIdx.x] ? bv[threadIdx.x]
+                                 : cs_comb<T, OP>(pv, bv[threadIdx.x]);
+    __syncthreads();
+    bv[threadIdx.x] = nv;
+    bf[threadIdx.x] = nf;
+    __syncthreads();
+  }
+  if (threadIdx.x == BLOCK - 1) {
+    tile_v[blockIdx.x] = bv[threadIdx.x];
+    tile_f[blockIdx.x] = bf[threadIdx.x];
+  }
+}
+
+template <typename T, int OP>
+__global__ void __launch_bounds__(1024) k_seg_scan_tiles(
+    T* __restrict__ tile_v, unsigned* __restrict__ tile_f, int64_t ntiles) {
+  __shared__ T carry_v;
+  __shared__ unsigned carry_f;
+  if (threadIdx.x == 0) { carry_v = cs_ident<T, OP>(); carry_f = 0; }
+  __shared__ T bv[1024];
+  __shared__ unsigned char bf[1024];
+  __syncthreads();
+  for (int64_t base = 0; base < ntiles; base += 1024) {
+    const int64_t i = base + threadIdx.x;
+    T v = cs_ident<T, OP>();
+    unsigned char f = 0;
+    if (i < ntiles) { v = tile_v[i]; f = (unsigned char)tile_f[i]; }
+    bv[threadIdx.x] = v;
+    bf[threadIdx.x] = f;
+    __syncthreads();
+    for (int off = 1; off < 1024; off <<= 1) {
+      T pv = cs_ident<T, OP>();
+      unsigned char pf = 0;
+      if (threadIdx.x >= off) { pv = bv[threadIdx.x - off];
+                                pf = bf[threadIdx.x - off]; }
+      __syncthreads();
+      const unsigned char nf = pf | bf[threadIdx.x];
+      const T nv = bf[threadIdx.x] ? bv[threadIdx.x]
+                                   : cs_comb<T, OP>(pv, bv[threadIdx.x]);
+      __syncthreads();
+      bv[threadIdx.x] = nv;
+      bf[threadIdx.x] = nf;
+      __syncthreads();
+    }
+    const T incl_v = bv[threadIdx.x];
+    const unsigned char incl_f = bf[threadIdx.x];
+    // exclusive = carry ⊕ previous lane's inclusive (identity at lane 0)
+    const T pv = (threadIdx.x == 0) ? cs_ident<T, OP>()
+                                    : bv[threadIdx.x - 1];
+    const unsigned char pf = (threadIdx.x == 0) ? 0 : bf[threadIdx.x - 1];
+    const unsigned ef = carry_f | pf;
+    const T ev = pf ? pv : cs_comb<T, OP>(carry_v, pv);
+    if (i < ntiles) { tile_v[i] = ev; tile_f[i] = ef; }
+    __syncthreads();  // all lanes read carry before 1023 advances it
+    if (threadIdx.x == 1023) {
+      carry_v = incl_f ? incl_v : cs_comb<T, OP>(carry_v, incl_v);
+      carry_f = carry_f | incl_f;
+    }
+    __syncthreads();
+  }
+}
+
+template <typename T, int OP>
+__global__ void __launch_bounds__(BLOCK) k_seg_apply(
+    const T* __restrict__ in, const int64_t* __restrict__ heads, int64_t n,
+    const T* __restrict__ tile_v, const unsigned* __restrict__ tile_f,
+    T* __restrict__ out) {
+  const int64_t t0 = (int64_t)blockIdx.x * FILT_TILE;
+  const int64_t t1 = min(t0 + (int64_t)FILT_TILE, n);
+  constexpr int PER = FILT_TILE / BLOCK;
+  __shared__ T sv[FILT_TILE];
+  __shared__ unsigned char sf[FILT_TILE];
+  for (int j = 0; j < PER; ++j) {
+    const int64_t i = t0 + (int64_t)j * BLOCK + threadIdx.x;
+    if (i < t1) {
+      sv[(int)(i - t0)] = in[i];
+      sf[(int)(i - t0)] = heads[i] != 0;
+    }
+  }
+  __syncthreads();
+  const int s0 = threadIdx.x * PER;
+  const int lim = (int)(t1 - t0);
+  T lv[PER];
+  unsigned char lf[PER];
+  unsigned char f = 0;
+  T v = cs_ident<T, OP>();
+  for (int j = 0; j < PER; ++j) {
+    if (s0 + j < lim) {
+      const T x = sv[s0 + j];
+      const T xv = (x != x) ? cs_ident<T, OP>() : x;
+      if (sf[s0 + j]) { f = 1; v = xv; }
+      else v = cs_comb<T, OP>(v, xv);
+    }
+    lv[j] = v;  // inclusive pair within the thread's run, per element
+    lf[j] = f;
+  }
+  __shared__ T bv[BLOCK];
+  __shared__ unsigned char bf[BLOCK];
+  bv[threadIdx.x] = v;
+  bf[threadIdx.x] = f;
+  __syncthreads();
+  for (int off = 1; off < BLOCK; off <<= 1) {
+    T pv = cs_ident<T, OP>();
+    unsigned char pf = 0;
+    if (threadIdx.x >= off) { pv = bv[threadIdx.x - off];
+                              pf = bf[threadIdx.x - off]; }
+    __syncthreads();
+    const unsigned char nf = pf | bf[threadIdx.x];
+    const T nv = bf[threadIdx.x] ? bv[threadIdx.x]
+                                 : cs_comb<T, OP>(pv, bv[threadIdx.x]);
+    __syncthreads();
+    bv[threadIdx.x] = nv;
+    bf[threadIdx.x] = nf;
+    __syncthreads();
+  }
+  // thread base = tile exclusive pair ⊕ preceding threads' inclusive pair
+  const T pv = (threadIdx.x == 0) ? cs_ident<T, OP>() : bv[threadIdx.x - 1];
+  const unsigned char pf = (threadIdx.x == 0) ? 0 : bf[threadIdx.x - 1];
+  const T base_v = pf ? pv : cs_comb<T, OP>(tile_v[blockIdx.x], pv);
+  __syncthreads();
+  for (int j = 0; j < PER; ++j) {
+    if (s0 + j < lim) {
+      const T x = sv[s0 + j];
+      const T r = lf[j] ? lv[j] : cs_comb<T, OP>(base_v, lv[j]);
+      sv[s0 + j] = (x != x) ? x : r;
+    }
+  }
+  __syncthreads();
+  for (int j = 0; j < PER; ++j) {
+    const int64_t i = t0 + (int64_t)j * BLOCK + threadIdx.x;
+    if (i < t1) out[i] = sv[(int)(i - t0)];
+  }
+}
+
 }  // namespace
 
 struct hf_join {
@@ -2830,6 +3033,89 @@ int hf_cumsum(const hf_col* col, int agg_op, hf_col** out) {
            ? runT(std::integral_constant<int, HF_AGG_MAX>{})
            : runT(std::integral_constant<int, HF_AGG_SUM>{});
   dev_free(d_ts, g.stream);
+  if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  return rc;
+}
+
+int hf_seg_cumsum(const hf_col* col, const hf_col* heads, int agg_op,
+                  hf_col** out) {
+  HF_NEED_INIT("hf_seg_cumsum");
+  if (!col || !heads || !out)
+    return set_err(HF_ERR_ARG, "hf_seg_cumsum", "null");
+  if (heads->dtype != HF_INT64 || heads->len != col->len)
+    return set_err(HF_ERR_ARG, "hf_seg_cumsum",
+                   "heads must be an int64 0/1 column of the same length");
+  if (agg_op != HF_AGG_SUM && agg_op != HF_AGG_MIN && agg_op != HF_AGG_MAX)
+    return set_err(HF_ERR_ARG, "hf_seg_cumsum", "bad agg_op");
+  const int64_t n = col->len;
+  int rc = hf_col_alloc(n, col->dtype, out);
+  if (rc != HF_OK) return rc;
+  if (n == 0) return HF_OK;
+  const int64_t ntiles = (n + FILT_TILE - 1) / FILT_TILE;
+  void* d_tv = nullptr;
+  void* d_tf = nullptr;
+  HF_HIP("hf_seg_cumsum", dev_alloc(&d_tv, ntiles * 8, g.stream));
+  rc = [&]() -> int {
+    HF_HIP("hf_seg_cumsum", dev_alloc(&d_tf, ntiles * 4, g.stream));
+    auto run = [&](auto tTag, auto opTag) -> int {
+      using T = decltype(tTag);
+      constexpr int OP = decltype(opTag)::value;
+      int r2 = timed_launch("seg_tiles", [&] {
+        hipLaunchKernelGGL((k_seg_tiles<T, OP>), dim3((uint32_t)ntiles),
+                           dim3(BLOCK), 0, g.stream, (const T*)col->dptr,
+                           (const int64_t*)heads->dptr, n, (T*)d_tv,
+                           (unsigned*)d_tf);
+      });
+      if (r2 != HF_OK) return r2;
+      r2 = timed_launch("seg_scan", [&] {
+        hipLaunchKernelGGL((k_seg_scan_tiles<T, OP>), dim3(1), dim3(1024),
+                           0, g.stream, (T*)d_tv, (unsigned*)d_tf, ntiles);
+      });
+      if (r2 != HF_OK) return r2;
+      return timed_launch("seg_apply", [&] {
+        hipLaunchKernelGGL((k_seg_apply<T, OP>), dim3((uint32_t)ntiles),
+                           dim3(BLOCK), 0, g.stream, (const T*)col->dptr,
+                           (const int64_t*)heads->dptr, n, (const T*)d_tv,
+                           (const unsigned*)d_tf, (T*)(*out)->dptr);
+      });
+    };
+    auto runT = [&](auto opTag) -> int {
+      return (col->dtype == HF_FLOAT64) ? run(double{}, opTag)
+                                        : run(int64_t{}, opTag);
+    };
+    return agg_op == HF_AGG_MIN
+               ? runT(std::integral_constant<int, HF_AGG_MIN>{})
+           : agg_op == HF_AGG_MAX
+               ? runT(std::integral_constant<int, HF_AGG_MAX>{})
+               : runT(std::integral_constant<int, HF_AGG_SUM>{});
+  }();
+  if (d_tf) dev_free(d_tf, g.stream);
+  dev_free(d_tv, g.stream);
+  if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  return rc;
+}
+
+int hf_scatter(const hf_col* col, const hf_col* idx, hf_col** out) {
+  HF_NEED_INIT("hf_scatter");
+  if (!col || !idx || !out) return set_err(HF_ERR_ARG, "hf_scatter", "null");
+  if (idx->dtype != HF_INT64 || idx->len != col->len)
+    return set_err(HF_ERR_ARG, "hf_scatter",
+                   "index must be an int64 permutation of the column");
+  int rc = hf_col_alloc(col->len, col->dtype, out);
+  if (rc != HF_OK) return rc;
+  const int64_t n = col->len;
+  if (n == 0) return HF_OK;
+  rc = timed_launch("scatter", [&] {
+    if (col->dtype == HF_FLOAT64)
+      hipLaunchKernelGGL((k_scatter<double>), dim3((uint32_t)grid_for(n)),
+                         dim3(BLOCK), 0, g.stream, (const double*)col->dptr,
+                         (const int64_t*)idx->dptr, (double*)(*out)->dptr, n);
+    else
+      hipLaunchKernelGGL((k_scatter<int64_t>), dim3((uint32_t)grid_for(n)),
+                         dim3(BLOCK), 0, g.stream, (const int64_t*)col->dptr,
+                         (const int64_t*)idx->dptr, (int64_t*)(*out)->dptr,
+                         n);
+  });
   if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
   return rc;
 }

@@ -399,12 +399,13 @@ class DataFrame(_HipPandasBase):
         return DataFrame(query_compiler=type(qc)(nf))
 
     def sort_values(self, by: str, ascending: bool = True,
-                    kind: str = "stable"):
+                    kind: str = "stable", na_position: str = "last"):
         """Always stable (equals pandas sort_values(kind='stable'), a
-        stronger guarantee than the default quicksort)."""
+        stronger guarantee than the default quicksort).  na_position
+        'last'/'first' as pandas."""
         return DataFrame(
             query_compiler=self._query_compiler.sort_rows_by_column_values(
-                by, ascending))
+                by, ascending, na_position=na_position))
 
     def merge(self, other: "DataFrame", on: str, how: str = "inner"):
         """Inner merge on an int64 key column (modin/pandas API ->
@@ -468,10 +469,12 @@ class Series(_HipPandasBase):
         return Series(query_compiler=qc.getitem_array(qc.notna()),
                       name=self.name)
 
-    def sort_values(self, ascending: bool = True, kind: str = "stable"):
+    def sort_values(self, ascending: bool = True, kind: str = "stable",
+                    na_position: str = "last"):
         """pandas Series.sort_values (always stable)."""
         name = list(self._query_compiler._modin_frame.columns)[0]
-        qc = self._query_compiler.sort_rows_by_column_values(name, ascending)
+        qc = self._query_compiler.sort_rows_by_column_values(
+            name, ascending, na_position=na_position)
         return Series(query_compiler=qc, name=self.name)
 
     def unique(self):
@@ -641,6 +644,36 @@ class DataFrameGroupBy:
 
     def last(self):
         return self._agg("last")
+
+    def _transform(self, how: str, **kw):
+        """Same-length transforms in original row order (pandas
+        DataFrameGroupBy.cumsum/cummin/cummax/cumcount/rank).  as_index is
+        irrelevant (pandas keeps the caller's index for transforms)."""
+        qc = self._df._query_compiler.groupby_transform(self._by, how, **kw)
+        if self._series_out or how == "cumcount":
+            name = list(qc._modin_frame.columns)[0]
+            return Series(query_compiler=qc,
+                          name=None if how == "cumcount" else name)
+        return DataFrame(query_compiler=qc)
+
+    def cumsum(self):
+        return self._transform("cumsum")
+
+    def cummin(self):
+        return self._transform("cummin")
+
+    def cummax(self):
+        return self._transform("cummax")
+
+    def cumcount(self):
+        return self._transform("cumcount")
+
+    def rank(self, method: str = "average", ascending: bool = True,
+             na_option: str = "keep"):
+        if na_option != "keep":
+            raise lib.HfError("rank: only na_option='keep' this round")
+        return self._transform("rank", ascending=bool(ascending),
+                               method=method)
 
     def size(self):
         """pandas DataFrameGroupBy.size(): a Series of group row counts

@@ -179,6 +179,15 @@ class HipQueryCompiler:
     def groupby_nunique(self, by) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.groupby_nunique(by))
 
+    def groupby_transform(self, by, how: str, ascending: bool = True,
+                          method: str = "average") -> "HipQueryCompiler":
+        """Same-length transforms in original row order (pandas
+        DataFrameGroupBy.cumsum/cummin/cummax/cumcount/rank; reference
+        routes these through modin/pandas/groupby.py -> qc groupby
+        methods)."""
+        return self.__constructor__(self._modin_frame.groupby_transform(
+            by, how, ascending=ascending, method=method))
+
     def groupby_size(self, by: str) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.groupby_size(by))
 
@@ -275,10 +284,12 @@ class HipQueryCompiler:
 
     # ---- sort (reference qc.sort_rows_by_column_values) ----
     def sort_rows_by_column_values(self, by: str,
-                                   ascending: bool = True
+                                   ascending: bool = True,
+                                   na_position: str = "last"
                                    ) -> "HipQueryCompiler":
         return self.__constructor__(
-            self._modin_frame.sort_rows(by, ascending))
+            self._modin_frame.sort_rows(by, ascending,
+                                        na_position=na_position))
 
     # ---- merge (query_compiler merge -> MergeImpl.row_axis_merge,
     #      storage_formats/pandas/merge.py:104) ----

@@ -1082,3 +1082,184 @@ def test_clip_axis1_vs_pandas(npartitions):
     assert gs.dtype == np.int64
     np.testing.assert_array_equal(gs.to_numpy(),
                                   pint.sum(axis=1).to_numpy())
+
+
+def test_groupby_transforms_vs_pandas(npartitions):
+    """groupby cumsum/cummin/cummax/cumcount in ORIGINAL row order
+    (segmented scan after a stable key sort + inverse-permutation
+    scatter), incl. NaN keys (no group -> NaN, pandas dropna=True), NaN
+    values (stay NaN, don't advance the state) and the dtype rule (int64
+    survives only when every key is valid)."""
+    rng = np.random.default_rng(93)
+    n = 200_000
+    k = rng.integers(0, 500, n).astype(np.float64)
+    k[rng.random(n) < 0.02] = np.nan
+    v = rng.standard_normal(n) * 3
+    v[rng.random(n) < 0.05] = np.nan
+    w = rng.integers(-50, 50, n)
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    for how in ("cumsum", "cummin", "cummax"):
+        got = getattr(df.groupby("k"), how)().to_pandas()
+        exp = getattr(pdf.groupby("k"), how)()
+        assert list(got.columns) == list(exp.columns)
+        for c in exp.columns:
+            np.testing.assert_allclose(
+                got[c].to_numpy(), exp[c].to_numpy(),
+                rtol=1e-12 if how == "cumsum" else 0, atol=1e-9,
+                equal_nan=True, err_msg=f"{how}/{c}")
+    got_cc = df.groupby("k").cumcount().to_pandas()
+    exp_cc = pdf.groupby("k").cumcount()
+    np.testing.assert_allclose(got_cc.to_numpy(), exp_cc.to_numpy(),
+                               rtol=0, equal_nan=True)
+    # int keys, no NaN anywhere: int64 results, exact
+    pdf2 = pandas.DataFrame({"k": rng.integers(0, 50, 5000),
+                             "w": rng.integers(-9, 9, 5000)})
+    df2 = mpd.DataFrame(pdf2)
+    for how in ("cumsum", "cummin", "cummax"):
+        got2 = getattr(df2.groupby("k"), how)().to_pandas()
+        exp2 = getattr(pdf2.groupby("k"), how)()
+        assert got2["w"].dtype == np.int64, how
+        np.testing.assert_array_equal(got2["w"].to_numpy(),
+                                      exp2["w"].to_numpy(), err_msg=how)
+    got2 = df2.groupby("k").cumcount().to_pandas()
+    assert got2.dtype == np.int64
+    np.testing.assert_array_equal(got2.to_numpy(),
+                                  pdf2.groupby("k").cumcount().to_numpy())
+    # Series selection form
+    s_ = df2.groupby("k")["w"].cumsum().to_pandas()
+    pd_s = pdf2.groupby("k")["w"].cumsum()
+    assert s_.name == "w"
+    np.testing.assert_array_equal(s_.to_numpy(), pd_s.to_numpy())
+
+
+def test_groupby_rank_vs_pandas(npartitions):
+    """groupby.rank: methods average/min/first, ascending both ways,
+    na_option='keep' (NaN values rank NaN; ties averaged within the
+    group)."""
+    rng = np.random.default_rng(94)
+    n = 60_000
+    k = rng.integers(0, 200, n)
+    v = rng.integers(-20, 20, n).astype(np.float64)  # many exact ties
+    v[rng.random(n) < 0.05] = np.nan
+    w = rng.integers(-5, 5, n)  # int64 values with heavy ties
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    for method in ("average", "min", "first"):
+        for asc in (True, False):
+            got = df.groupby("k").rank(method=method,
+                                       ascending=asc).to_pandas()
+            exp = pdf.groupby("k").rank(method=method, ascending=asc)
+            for c in exp.columns:
+                np.testing.assert_allclose(
+                    got[c].to_numpy(), exp[c].to_numpy(), rtol=0,
+                    equal_nan=True, err_msg=f"{method}/asc={asc}/{c}")
+    s_ = df.groupby("k")["v"].rank().to_pandas()
+    np.testing.assert_allclose(s_.to_numpy(),
+                               pdf.groupby("k")["v"].rank().to_numpy(),
+                               rtol=0, equal_nan=True)
+
+
+def test_groupby_transform_multikey_vs_pandas(npartitions):
+    """Multi-key transforms (string + int keys): group identity via the
+    per-column run-head OR — no combined-key fold, so any span works."""
+    rng = np.random.default_rng(95)
+    n = 30_000
+    a = rng.choice(["x", "y", "zz", "w"], n)
+    b = rng.integers(-3, 4, n)
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.05] = np.nan
+    pdf = pandas.DataFrame({"a": a, "b": b, "v": v})
+    df = mpd.DataFrame(pdf)
+    got = df.groupby(["a", "b"]).cumsum().to_pandas()
+    exp = pdf.groupby(["a", "b"]).cumsum()
+    np.testing.assert_allclose(got["v"].to_numpy(), exp["v"].to_numpy(),
+                               rtol=1e-12, atol=1e-9, equal_nan=True)
+    got_cc = df.groupby(["a", "b"]).cumcount().to_pandas()
+    np.testing.assert_array_equal(
+        got_cc.to_numpy(), pdf.groupby(["a", "b"]).cumcount().to_numpy())
+    got_r = df.groupby(["a", "b"]).rank().to_pandas()
+    exp_r = pdf.groupby(["a", "b"]).rank()
+    np.testing.assert_allclose(got_r["v"].to_numpy(),
+                               exp_r["v"].to_numpy(), rtol=0,
+                               equal_nan=True)
+
+
+def test_groupby_transform_edge_cases(npartitions):
+    """Single group, all-NaN keys, one row, empty value set."""
+    pdf = pandas.DataFrame({"k": [7, 7, 7], "v": [1.0, np.nan, 2.0]})
+    df = mpd.DataFrame(pdf)
+    np.testing.assert_allclose(
+        df.groupby("k").cumsum().to_pandas()["v"].to_numpy(),
+        pdf.groupby("k").cumsum()["v"].to_numpy(), rtol=0, equal_nan=True)
+    pdf2 = pandas.DataFrame({"k": [np.nan, np.nan], "v": [1.0, 2.0]})
+    df2 = mpd.DataFrame(pdf2)
+    np.testing.assert_allclose(
+        df2.groupby("k").cumsum().to_pandas()["v"].to_numpy(),
+        pdf2.groupby("k").cumsum()["v"].to_numpy(), rtol=0, equal_nan=True)
+    pdf3 = pandas.DataFrame({"k": [1], "v": [5.0]})
+    df3 = mpd.DataFrame(pdf3)
+    np.testing.assert_allclose(
+        df3.groupby("k").rank().to_pandas()["v"].to_numpy(),
+        pdf3.groupby("k").rank()["v"].to_numpy(), rtol=0)
+
+
+def test_scatter_seg_cumsum_kernels():
+    """Kernel-level checks: hf_scatter inverts hf_gather; hf_seg_cumsum
+    equals a per-segment host scan (sum/min/max, i64 + f64 with NaN)."""
+    rng = np.random.default_rng(96)
+    n = 50_000
+    perm_np = rng.permutation(n).astype(np.int64)
+    x = rng.standard_normal(n)
+    cx = lib.put(x)
+    cp = lib.put(perm_np)
+    gathered = lib.gather(cx, cp)
+    back = lib.scatter(gathered, cp)
+    np.testing.assert_array_equal(lib.get(back), x)
+    # segmented scan vs host reference
+    heads = (rng.random(n) < 0.001).astype(np.int64)
+    heads[0] = 1
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.05] = np.nan
+    seg_id = np.cumsum(heads) - 1
+    for op, name in ((lib.AGG_SUM, "sum"), (lib.AGG_MIN, "min"),
+                     (lib.AGG_MAX, "max")):
+        got = lib.get(lib.seg_cumsum(lib.put(v), lib.put(heads), op))
+        exp = pandas.Series(v).groupby(seg_id).transform(
+            {"sum": "cumsum", "min": "cummin", "max": "cummax"}[name]
+        ).to_numpy()
+        np.testing.assert_allclose(got, exp, rtol=1e-12, atol=1e-9,
+                                   equal_nan=True, err_msg=name)
+    w = rng.integers(-1000, 1000, n)
+    got = lib.get(lib.seg_cumsum(lib.put(w), lib.put(heads), lib.AGG_SUM))
+    exp = pandas.Series(w).groupby(seg_id).cumsum().to_numpy()
+    np.testing.assert_array_equal(got, exp)
+
+
+def test_sort_na_position_first(npartitions):
+    """sort_values(na_position='first'): NaN float keys and NaN string
+    (dict code −1) keys lead the result for both directions; stability
+    preserved."""
+    rng = np.random.default_rng(97)
+    n = 30_000
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.1] = np.nan
+    s = rng.choice(["b", "a", "cc"], n).astype(object)
+    s[rng.random(n) < 0.1] = None
+    w = rng.integers(0, 100, n)
+    pdf = pandas.DataFrame({"v": v, "s": s, "w": w})
+    df = mpd.DataFrame(pdf)
+    for by in ("v", "s"):
+        for asc in (True, False):
+            got = df.sort_values(by, ascending=asc,
+                                 na_position="first").to_pandas()
+            exp = pdf.sort_values(by, ascending=asc, kind="stable",
+                                  na_position="first")
+            np.testing.assert_array_equal(got.index.to_numpy(),
+                                          exp.index.to_numpy(),
+                                          err_msg=f"{by}/asc={asc}")
+            np.testing.assert_array_equal(got["w"].to_numpy(),
+                                          exp["w"].to_numpy())
+    s2 = df["v"].sort_values(na_position="first").to_pandas()
+    e2 = pdf["v"].sort_values(na_position="first", kind="stable")
+    np.testing.assert_array_equal(s2.index.to_numpy(), e2.index.to_numpy())
