@@ -331,6 +331,10 @@ int hf_seg_cumsum(const hf_col* col, const hf_col* heads, int agg_op,
  * reference reindexing a transform result back to the caller's index. */
 int hf_scatter(const hf_col* col, const hf_col* idx, hf_col** out);
 
+/* Cartesian-product gather indices (pandas merge how='cross'):
+ * lidx[i] = i / nr, ridx[i] = i % nr over n = nl*nr rows. */
+int hf_cross_idx(int64_t nl, int64_t nr, hf_col** lidx, hf_col** ridx);
+
 /* Exact-match binary search: out[i] = j with sorted[j] == keys[i], else -1.
  * Densifies unbounded int64 join keys through the sorted distinct right
  * keys (lower_bound per row), so the dense-range CSR join

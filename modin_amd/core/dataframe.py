@@ -44,14 +44,20 @@ class DeviceIndex:
         return self.col.length
 
 
-def recode_dict_col(codes: lib.ColumnRef, old_cats, new_cats):
+def recode_dict_col(codes: lib.ColumnRef, old_cats, new_cats,
+                    missing: int = -1):
     """Re-express dictionary codes in a new dictionary: host LUT
     (new_cats.get_indexer(old_cats), −1 rows preserved via a +1-shifted
-    sentinel slot) gathered on device — no string ever touches the GPU."""
+    sentinel slot) gathered on device — no string ever touches the GPU.
+    `missing` is the code for old categories absent from new_cats (merge
+    passes −2 so unmatched categories never collide with NaN's −1)."""
     import numpy as np
     lut = np.empty(len(old_cats) + 1, dtype=np.int64)
     lut[0] = -1
-    lut[1:] = new_cats.get_indexer(old_cats)
+    idxr = new_cats.get_indexer(old_cats)
+    if missing != -1:
+        idxr = np.where(idxr < 0, missing, idxr)
+    lut[1:] = idxr
     lut_col = lib.put(lut)
     shifted = lib.map_scalar(lib.MAP_ADD, codes, 1)
     return lib.gather(lut_col, shifted)
@@ -1176,7 +1182,8 @@ class HipDataframe:
         return c
 
     def groupby_transform(self, by, how: str, ascending: bool = True,
-                          method: str = "average") -> "HipDataframe":
+                          method: str = "average",
+                          periods: int = 1) -> "HipDataframe":
         """Same-length groupby transforms in original row order.
 
         how: 'cumsum' | 'cummin' | 'cummax' (segmented scan), 'cumcount'
@@ -1202,7 +1209,8 @@ class HipDataframe:
         for b in by_list:
             if b not in self.columns:
                 raise lib.HfError(f"groupby: key column {b!r} missing")
-        if how not in ("cumsum", "cummin", "cummax", "cumcount", "rank"):
+        if how not in ("cumsum", "cummin", "cummax", "cumcount", "rank",
+                       "ngroup", "shift", "diff"):
             raise lib.HfError(f"groupby transform {how!r} not supported")
         if how == "rank" and method not in ("average", "min", "first"):
             raise lib.HfError(f"rank method {method!r} not supported")
@@ -1219,12 +1227,13 @@ class HipDataframe:
                     f"groupby {how}: string column {v!r} unsupported "
                     "(pandas raises on non-numeric transforms)")
         n = len(self)
-        if n == 0 or not val_names:
-            dts = pandas.Series({v: np.dtype(np.float64)
-                                 for v in val_names})
+        needs_vals = how not in ("cumcount", "ngroup")
+        if n == 0 or (needs_vals and not val_names):
+            names = val_names if needs_vals else [how]
+            dts = pandas.Series({v: np.dtype(np.float64) for v in names})
             part = HipDataframePartition(DeviceBlock(
-                {v: lib.alloc(0, lib.HF_FLOAT64) for v in val_names}, 0))
-            return HipDataframe([part], pandas.RangeIndex(0), val_names,
+                {v: lib.alloc(0, lib.HF_FLOAT64) for v in names}, 0))
+            return HipDataframe([part], pandas.RangeIndex(0), names,
                                 [0], dts)
 
         def concat_col(name):
@@ -1258,20 +1267,65 @@ class HipDataframe:
         if len(eff_keys) > 1:
             head = lib.compare_scalar(lib.CMP_GE, head, 1.0)
         out_cols, dts = {}, {}
-        if how == "cumcount":
+        if how in ("cumcount", "ngroup"):
             plan = lib.filter_plan(head)
             hp = lib.filter_iota(plan, 0)           # run start positions
             rid = lib.map_scalar(lib.MAP_SUB, lib.cumsum(head), 1)
-            start = lib.gather(hp, rid)
-            cc = lib.binary(lib.BIN_SUB, self._iota(n), start)
+            if how == "cumcount":
+                start = lib.gather(hp, rid)
+                cc = lib.binary(lib.BIN_SUB, self._iota(n), start)
+            else:
+                # ngroup: 0-based group id in SORTED key order (pandas
+                # sort=True default) == the run id; NaN-key runs sort last
+                # so valid group numbering is unaffected before the fixup
+                cc = rid
             res = lib.scatter(cc, perm)
             if valid is not None:
                 res = lib.fixup_empty(lib.cast_f64(res), valid)
-            name = "cumcount"
+            name = how
             part = HipDataframePartition(DeviceBlock({name: res}, n))
             dt = np.dtype(np.int64 if valid is None else np.float64)
             return HipDataframe([part], pandas.RangeIndex(n), [name], [n],
                                 pandas.Series({name: dt}))
+        if how in ("shift", "diff"):
+            periods = int(periods)
+            plan = lib.filter_plan(head)
+            ng = plan.n_kept
+            hp = lib.filter_iota(plan, 0)
+            rid = lib.map_scalar(lib.MAP_SUB, lib.cumsum(head), 1)
+            start = lib.gather(hp, rid)
+            nxt = lib.concat([lib.col_slice(hp, 1, ng - 1),
+                              self._const_i64(n)]) if ng > 1 \
+                else self._const_i64(n)
+            end = lib.gather(nxt, rid)              # run end (exclusive)
+            pos = self._iota(n)
+            idx = lib.map_scalar(lib.MAP_SUB, pos, periods)
+            # source row must stay inside the run: start <= idx < end
+            ok = lib.binary(
+                lib.BIN_MUL,
+                lib.compare_scalar(
+                    lib.CMP_GE, lib.binary(lib.BIN_SUB, idx, start), 0.0),
+                lib.compare_scalar(
+                    lib.CMP_GE,
+                    lib.binary(lib.BIN_SUB,
+                               lib.map_scalar(lib.MAP_SUB, end, 1), idx),
+                    0.0))
+            cidx = lib.map_scalar(
+                lib.MAP_MAX,
+                lib.map_scalar(lib.MAP_MIN, idx, max(n - 1, 0)), 0)
+            for v in val_names:
+                sv = lib.gather(lib.cast_f64(concat_col(v)), perm)
+                sh = lib.fixup_empty(lib.gather(sv, cidx), ok)
+                if how == "diff":
+                    sh = lib.binary(lib.BIN_SUB, sv, sh)
+                res = lib.scatter(sh, perm)
+                if valid is not None:
+                    res = lib.fixup_empty(res, valid)
+                out_cols[v] = res
+                dts[v] = np.dtype(np.float64)
+            part = HipDataframePartition(DeviceBlock(out_cols, n))
+            return HipDataframe([part], pandas.RangeIndex(n), val_names,
+                                [n], pandas.Series(dts))
         agg_op = {"cumsum": lib.AGG_SUM, "cummin": lib.AGG_MIN,
                   "cummax": lib.AGG_MAX}[how]
         for v in val_names:
@@ -1381,6 +1435,63 @@ class HipDataframe:
     # ---- broadcast inner join (MergeImpl.row_axis_merge device form,
     #      merge.py:104-178: combine() the right frame once, probe per left
     #      partition; pandas suffix rules "_x"/"_y" on collisions) ----
+    def cross_join(self, other: "HipDataframe") -> "HipDataframe":
+        """pandas merge(how='cross'): the nl x nr cartesian product —
+        device gather through hf_cross_idx row indices; '_x'/'_y' suffixes
+        on every colliding column name (no key column to exempt).
+        Reference: MergeImpl/pandas cross merge (merge.py)."""
+        from ..distributed import is_active
+        if is_active():
+            raise lib.HfError("distributed cross merge is a later round")
+        nl, nr = len(self), len(other)
+        if nl * max(nr, 1) > (1 << 31):
+            raise lib.HfError("cross merge result exceeds 2^31 rows")
+
+        def concat_col(frame, name):
+            cols = [p.block().columns[name] for p in frame._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        lcats = (self._partitions[0].block().cats
+                 if self._partitions else {})
+        rcats = (other._partitions[0].block().cats
+                 if other._partitions else {})
+        common = set(self.columns) & set(other.columns)
+        lout = {n2: (n2 + "_x" if n2 in common else n2)
+                for n2 in self.columns}
+        rout = {n2: (n2 + "_y" if n2 in common else n2)
+                for n2 in other.columns}
+        n = nl * nr
+        if nr == 0 or nl == 0:
+            cols = {}
+            dts = {}
+            for c in self.columns:
+                cols[lout[c]] = lib.alloc(0, lib.HF_INT64)
+                dts[lout[c]] = self.dtypes[c]
+            for c in other.columns:
+                cols[rout[c]] = lib.alloc(0, lib.HF_INT64)
+                dts[rout[c]] = other.dtypes[c]
+            part = HipDataframePartition(DeviceBlock(cols, 0))
+            names = list(cols)
+            return HipDataframe([part], pandas.RangeIndex(0), names, [0],
+                                pandas.Series(dts))
+        lidx, ridx = lib.cross_idx(nl, nr)
+        cols, cats = {}, {}
+        for c in self.columns:
+            cols[lout[c]] = lib.gather(concat_col(self, c), lidx)
+            if c in lcats:
+                cats[lout[c]] = lcats[c]
+        for c in other.columns:
+            cols[rout[c]] = lib.gather(concat_col(other, c), ridx)
+            if c in rcats:
+                cats[rout[c]] = rcats[c]
+        names = [lout[c] for c in self.columns] + \
+            [rout[c] for c in other.columns]
+        dts = pandas.Series(
+            {**{lout[c]: self.dtypes[c] for c in self.columns},
+             **{rout[c]: other.dtypes[c] for c in other.columns}})
+        part = HipDataframePartition(DeviceBlock(cols, n, cats))
+        return HipDataframe([part], pandas.RangeIndex(n), names, [n], dts)
+
     def broadcast_join(self, other: "HipDataframe", on: str,
                        how: str = "inner") -> "HipDataframe":
         if on not in self.columns or on not in other.columns:
@@ -1409,23 +1520,28 @@ class HipDataframe:
             raise lib.HfError("merge: key column is a string on one side "
                               "only")
         rkeys = concat_col(other, on)
+        # float keys ride the order-preserving bit transform: all NaNs
+        # canonicalize to ONE ordered value, so NaN==NaN matches — exactly
+        # pandas merge semantics for NaN keys.  An int64/float64 mixed key
+        # promotes to float64 first (pandas rule).
+        key_f64 = on not in lcats and (
+            self.dtypes[on] == np.dtype(np.float64)
+            or other.dtypes[on] == np.dtype(np.float64))
+
+        def enc_key(col):
+            return lib.ordered_i64(lib.cast_f64(col)) if key_f64 else col
+
+        rkeys = enc_key(rkeys)
         if on in lcats:
             # dictionary keys: join in the LEFT dictionary's code space —
-            # recode the right key codes (unmatched right values map to a
-            # fresh negative code so they never match; NaN keys are
-            # rejected loudly: pandas matches NaN==NaN in merges, a later
-            # round here)
+            # recode the right key codes; NaN keys are code −1 on BOTH
+            # sides, so they match each other (pandas NaN==NaN in merges);
+            # right categories absent from the left dictionary map to −2
+            # (recode_dict_col) so they never collide with NaN
             key_cats = lcats[on]
-            for side, kc in (("left", concat_col(self, on)), ("right", rkeys)):
-                if kc.length and lib.reduce(kc).imn < 0:
-                    raise lib.HfError(
-                        f"merge: NaN in the {side} string key (NaN-key "
-                        "matching is a later round)")
             if not rcats[on].equals(key_cats):
-                rk2 = recode_dict_col(rkeys, rcats[on], key_cats)
-                # unmatched right categories came back −1: remap to −2..
-                # distinct from NaN semantics is unnecessary (no NaNs here)
-                rkeys = rk2
+                rkeys = recode_dict_col(rkeys, rcats[on], key_cats,
+                                        missing=-2)
         if rkeys.dtype_code != lib.HF_INT64:
             raise lib.HfError("merge: key column must be int64 (dense-range "
                               "CSR join; hashed keys are a later round)")
@@ -1493,7 +1609,7 @@ class HipDataframe:
         if how in ("left", "outer"):
             for p in self._partitions:
                 block = p.block()
-                lk = block.columns[on]
+                lk = enc_key(block.columns[on])
                 if uniq is not None:
                     lk = lib.search_sorted(lk, uniq)
                     m = lib.compare_scalar(lib.CMP_EQ, lk, -1.0)
@@ -1507,7 +1623,7 @@ class HipDataframe:
         out_parts, lengths = [], []
         for pi, p in enumerate(self._partitions):
             block = p.block()
-            lkeys = block.columns[on]
+            lkeys = enc_key(block.columns[on])
             if lkeys.dtype_code != lib.HF_INT64:
                 raise lib.HfError("merge: key column must be int64")
             if uniq is not None:  # code space: unmatched lefts become -1
@@ -1557,7 +1673,8 @@ class HipDataframe:
             cols = {}
             for name in self.columns:  # left column order, key in place
                 if name == on:
-                    cols[on] = keys_c
+                    cols[on] = lib.ordered_i64(keys_c, inverse=True) \
+                        if key_f64 else keys_c
                 else:
                     cols[lout[name]] = lib.gather(block.columns[name], lidx)
             for i, rn in enumerate(right_names):
@@ -1568,7 +1685,7 @@ class HipDataframe:
         if how == "outer":
             # unmatched RIGHT rows: keys absent from the left key set
             luniq = None
-            lk_all = concat_col(self, on)
+            lk_all = enc_key(concat_col(self, on))
             if uniq is not None:
                 lk_all = lib.search_sorted(lk_all, uniq)
             if lk_all.length:
@@ -1588,6 +1705,8 @@ class HipDataframe:
                 rk_out = lib.filter_apply(rplan, rkeys)
                 if uniq is not None:
                     rk_out = lib.gather(uniq, rk_out)
+                if key_f64:
+                    rk_out = lib.ordered_i64(rk_out, inverse=True)
                 for name in self.columns:
                     key = on if name == on else lout[name]
                     if name == on:
@@ -1614,6 +1733,8 @@ class HipDataframe:
         dtypes = {}
         for c in self.columns:
             d = self.dtypes[c]
+            if c == on and key_f64:
+                d = np.dtype(np.float64)
             if (how == "outer" and n_run and c != on and c not in lcats
                     and d == np.dtype(np.int64)):
                 d = np.dtype(np.float64)

@@ -137,6 +137,9 @@ def load() -> ct.CDLL:
                                          ct.POINTER(ct.c_void_p)]),
             "hf_scatter": (ct.c_int, [ct.c_void_p, ct.c_void_p,
                                       ct.POINTER(ct.c_void_p)]),
+            "hf_cross_idx": (ct.c_int, [ct.c_int64, ct.c_int64,
+                                        ct.POINTER(ct.c_void_p),
+                                        ct.POINTER(ct.c_void_p)]),
             "hf_shuffle_dest": (ct.c_int, [ct.c_void_p,
                                            ct.POINTER(ct.c_int64), ct.c_int,
                                            ct.POINTER(ct.c_void_p)]),
@@ -202,7 +205,7 @@ def exported_symbols():
         "hf_groupby_hash_accum", "hf_groupby_hash_compact",
         "hf_groupby_sorted", "hf_shuffle_dest", "hf_memcpy_dd",
         "hf_search_sorted", "hf_ordered_i64", "hf_cumsum", "hf_seg_cumsum",
-        "hf_scatter",
+        "hf_scatter", "hf_cross_idx",
         "hf_col_concat", "hf_col_slice", "hf_join_build", "hf_join_free", "hf_join_probe",
         "hf_gather", "hf_compare_scalar", "hf_filter_plan", "hf_filter_apply",
         "hf_filter_iota", "hf_filter_plan_free", "hf_profiling",
@@ -633,6 +636,16 @@ def seg_cumsum(col: ColumnRef, heads: ColumnRef,
     _check(load().hf_seg_cumsum(col.handle, heads.handle, agg_op,
                                 ct.byref(out)), "hf_seg_cumsum")
     return _wrap(out, col.length, col.dtype_code)
+
+
+def cross_idx(nl: int, nr: int):
+    """(lidx, ridx) gather indices of the nl x nr cartesian product
+    (merge how='cross')."""
+    ensure_ready()
+    li, ri = ct.c_void_p(), ct.c_void_p()
+    _check(load().hf_cross_idx(nl, nr, ct.byref(li), ct.byref(ri)),
+           "hf_cross_idx")
+    return _wrap(li, nl * nr, HF_INT64), _wrap(ri, nl * nr, HF_INT64)
 
 
 def scatter(col: ColumnRef, idx: ColumnRef) -> ColumnRef:

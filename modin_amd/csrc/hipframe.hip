@@ -1825,7 +1825,11 @@ __global__ void __launch_bounds__(BLOCK) k_f64_ordered(
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
     double x = in[i] + 0.0;  // -0.0 -> +0.0 (pandas groups them together)
-    long long v = __double_as_longlong(x);
+    // canonicalize NaN (any sign/payload) to one ordered value: pandas
+    // treats every NaN as equal — groupby keys, sort ties AND merge keys
+    // (pandas matches NaN==NaN in merges)
+    long long v = (x != x) ? 0x7FF8000000000000LL
+                           : __double_as_longlong(x);
     // signed total order: non-negative floats keep their bits (already
     // increasing as signed i64); negative floats reverse below zero
     out[i] = (v < 0) ? (~v ^ 0x8000000000000000LL) : v;
@@ -1886,6 +1890,19 @@ __global__ void __launch_bounds__(BLOCK) k_gather_i64(
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) out[i] = src[idx[i]];
+}
+
+// Cross-join row indices: lidx[i] = i / nr, ridx[i] = i % nr — the gather
+// indices materializing a cartesian product (pandas merge how='cross').
+__global__ void __launch_bounds__(BLOCK) k_cross_idx(
+    int64_t n, int64_t nr, int64_t* __restrict__ lidx,
+    int64_t* __restrict__ ridx) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    lidx[i] = i / nr;
+    ridx[i] = i - (i / nr) * nr;
+  }
 }
 
 // Inverse of gather: out[idx[i]] = src[i].  idx must be a permutation of
@@ -3092,6 +3109,28 @@ int hf_seg_cumsum(const hf_col* col, const hf_col* heads, int agg_op,
   if (d_tf) dev_free(d_tf, g.stream);
   dev_free(d_tv, g.stream);
   if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
+  return rc;
+}
+
+int hf_cross_idx(int64_t nl, int64_t nr, hf_col** lidx, hf_col** ridx) {
+  HF_NEED_INIT("hf_cross_idx");
+  if (!lidx || !ridx || nl < 0 || nr <= 0)
+    return set_err(HF_ERR_ARG, "hf_cross_idx", "bad args");
+  const int64_t n = nl * nr;
+  int rc = hf_col_alloc(n, HF_INT64, lidx);
+  if (rc != HF_OK) return rc;
+  rc = hf_col_alloc(n, HF_INT64, ridx);
+  if (rc != HF_OK) { hf_col_free(*lidx); *lidx = nullptr; return rc; }
+  if (n == 0) return HF_OK;
+  rc = timed_launch("cross_idx", [&] {
+    hipLaunchKernelGGL(k_cross_idx, dim3((uint32_t)grid_for(n)), dim3(BLOCK),
+                       0, g.stream, n, nr, (int64_t*)(*lidx)->dptr,
+                       (int64_t*)(*ridx)->dptr);
+  });
+  if (rc != HF_OK) {
+    hf_col_free(*lidx); hf_col_free(*ridx);
+    *lidx = *ridx = nullptr;
+  }
   return rc;
 }
 

@@ -407,9 +407,11 @@ class DataFrame(_HipPandasBase):
             query_compiler=self._query_compiler.sort_rows_by_column_values(
                 by, ascending, na_position=na_position))
 
-    def merge(self, other: "DataFrame", on: str, how: str = "inner"):
-        """Inner merge on an int64 key column (modin/pandas API ->
-        qc.merge -> broadcast-right device join)."""
+    def merge(self, other: "DataFrame", on: str = None,
+              how: str = "inner"):
+        """Merge on an int64/float64/string key column (modin/pandas API
+        -> qc.merge -> broadcast-right device join); how='cross' takes no
+        key (cartesian product)."""
         return DataFrame(query_compiler=self._query_compiler.merge(
             other._query_compiler, on=on, how=how))
 
@@ -650,10 +652,11 @@ class DataFrameGroupBy:
         DataFrameGroupBy.cumsum/cummin/cummax/cumcount/rank).  as_index is
         irrelevant (pandas keeps the caller's index for transforms)."""
         qc = self._df._query_compiler.groupby_transform(self._by, how, **kw)
-        if self._series_out or how == "cumcount":
+        if self._series_out or how in ("cumcount", "ngroup"):
             name = list(qc._modin_frame.columns)[0]
             return Series(query_compiler=qc,
-                          name=None if how == "cumcount" else name)
+                          name=None if how in ("cumcount", "ngroup")
+                          else name)
         return DataFrame(query_compiler=qc)
 
     def cumsum(self):
@@ -667,6 +670,15 @@ class DataFrameGroupBy:
 
     def cumcount(self):
         return self._transform("cumcount")
+
+    def ngroup(self):
+        return self._transform("ngroup")
+
+    def shift(self, periods: int = 1):
+        return self._transform("shift", periods=int(periods))
+
+    def diff(self, periods: int = 1):
+        return self._transform("diff", periods=int(periods))
 
     def rank(self, method: str = "average", ascending: bool = True,
              na_option: str = "keep"):

@@ -180,13 +180,14 @@ class HipQueryCompiler:
         return self.__constructor__(self._modin_frame.groupby_nunique(by))
 
     def groupby_transform(self, by, how: str, ascending: bool = True,
-                          method: str = "average") -> "HipQueryCompiler":
+                          method: str = "average",
+                          periods: int = 1) -> "HipQueryCompiler":
         """Same-length transforms in original row order (pandas
-        DataFrameGroupBy.cumsum/cummin/cummax/cumcount/rank; reference
-        routes these through modin/pandas/groupby.py -> qc groupby
-        methods)."""
+        DataFrameGroupBy.cumsum/cummin/cummax/cumcount/rank/ngroup/
+        shift/diff; reference routes these through
+        modin/pandas/groupby.py -> qc groupby methods)."""
         return self.__constructor__(self._modin_frame.groupby_transform(
-            by, how, ascending=ascending, method=method))
+            by, how, ascending=ascending, method=method, periods=periods))
 
     def groupby_size(self, by: str) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.groupby_size(by))
@@ -293,8 +294,15 @@ class HipQueryCompiler:
 
     # ---- merge (query_compiler merge -> MergeImpl.row_axis_merge,
     #      storage_formats/pandas/merge.py:104) ----
-    def merge(self, right: "HipQueryCompiler", on: str,
+    def merge(self, right: "HipQueryCompiler", on: str = None,
               how: str = "inner") -> "HipQueryCompiler":
+        if how == "cross":
+            if on is not None:
+                raise lib.HfError("merge: how='cross' forbids 'on'")
+            return self.__constructor__(
+                self._modin_frame.cross_join(right._modin_frame))
+        if on is None:
+            raise lib.HfError("merge: 'on' required (except how='cross')")
         if how == "right":
             # pandas right join == swapped left join with the suffix roles
             # flipped back and columns restored to left-then-right order
