@@ -249,8 +249,10 @@ __device__ __forceinline__ double map1_f64(double x, double s) {
     case HF_MAP_ABS:    return fabs(x);
     case HF_MAP_NEG:    return -x;
     case HF_MAP_SQRT:   return sqrt(x);
-    case HF_MAP_MIN:    return fmin(x, s);
-    case HF_MAP_MAX:    return fmax(x, s);
+    // NaN-PROPAGATING (pandas clip leaves NaN alone; fmin/fmax would
+    // replace NaN with the bound)
+    case HF_MAP_MIN:    return (x != x) ? x : fmin(x, s);
+    case HF_MAP_MAX:    return (x != x) ? x : fmax(x, s);
   }
   return x;
 }
