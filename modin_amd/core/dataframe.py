@@ -1008,13 +1008,15 @@ class HipDataframe:
                 sums = lib.map_scalar(lib.MAP_CAST_I64, sums, 0)
             return DeviceBlock({"\x00r\x00": sums}, block.length)
 
-        res = self.map(block_fn)
+        mgr = self._partition_mgr_cls
+        parts = mgr.map_partitions(self._partitions, block_fn)
         dt = (np.dtype(np.int64)
               if (op == "count" or (all_int and op in ("sum", "min",
                                                        "max")))
               else np.dtype(np.float64))
-        res.dtypes = pandas.Series({"\x00r\x00": dt})
-        return res
+        return HipDataframe(parts, self._index, ["\x00r\x00"],
+                            self._row_lengths,
+                            pandas.Series({"\x00r\x00": dt}))
 
     def idx_extreme(self, maximum: bool) -> dict:
         """Per-column idxmax/idxmin: the FIRST original position holding
