@@ -1212,7 +1212,8 @@ class HipDataframe:
             if b not in self.columns:
                 raise lib.HfError(f"groupby: key column {b!r} missing")
         if how not in ("cumsum", "cummin", "cummax", "cumcount", "rank",
-                       "ngroup", "shift", "diff"):
+                       "ngroup", "shift", "diff",
+                       "bsum", "bmin", "bmax", "bcount", "bmean"):
             raise lib.HfError(f"groupby transform {how!r} not supported")
         if how == "rank" and method not in ("average", "min", "first"):
             raise lib.HfError(f"rank method {method!r} not supported")
@@ -1325,6 +1326,80 @@ class HipDataframe:
                     res = lib.fixup_empty(res, valid)
                 out_cols[v] = res
                 dts[v] = np.dtype(np.float64)
+            part = HipDataframePartition(DeviceBlock(out_cols, n))
+            return HipDataframe([part], pandas.RangeIndex(n), val_names,
+                                [n], pandas.Series(dts))
+        if how.startswith("b"):
+            # broadcast aggregate (pandas gb.transform('sum'|'mean'|...)):
+            # per-run aggregate = segmented-scan value at the run's LAST
+            # row (NaN pre-filled with the identity so the end value is
+            # the true aggregate), gathered back per row id
+            agg = how[1:]
+            op = {"sum": lib.AGG_SUM, "mean": lib.AGG_SUM,
+                  "count": lib.AGG_SUM, "min": lib.AGG_MIN,
+                  "max": lib.AGG_MAX}[agg]
+            ident = {"min": float("inf"), "max": float("-inf")}.get(agg,
+                                                                    0.0)
+            plan = lib.filter_plan(head)
+            ng = plan.n_kept
+            hp = lib.filter_iota(plan, 0)
+            rid = lib.map_scalar(lib.MAP_SUB, lib.cumsum(head), 1)
+            ends = lib.concat([lib.col_slice(hp, 1, ng - 1),
+                               self._const_i64(n)]) if ng > 1 \
+                else self._const_i64(n)
+            ends_m1 = lib.map_scalar(lib.MAP_SUB, ends, 1)  # run last row
+            for v in val_names:
+                vc = concat_col(v)
+                src_int = vc.dtype_code == lib.HF_INT64
+                sv = lib.gather(vc, perm)
+                if src_int:
+                    m = None
+                    dense = sv
+                else:
+                    m = lib.compare_scalar(lib.CMP_NOTNA, sv, 0.0)
+                    dense = lib.map_scalar(lib.MAP_FILLNA, sv, ident)
+                if agg == "count":
+                    if m is None:
+                        m = lib.alloc(n, lib.HF_INT64)
+                        lib.fill_i64(m.dptr(), 1, n)
+                    seg_a = lib.seg_cumsum(m, head, lib.AGG_SUM)
+                    br = lib.gather(lib.gather(seg_a, ends_m1), rid)
+                else:
+                    seg_a = lib.seg_cumsum(dense, head, op)
+                    br = lib.gather(lib.gather(seg_a, ends_m1), rid)
+                    if agg == "mean" or (agg in ("min", "max")
+                                         and m is not None):
+                        segc = lib.seg_cumsum(
+                            m if m is not None else None, head,
+                            lib.AGG_SUM) if m is not None else None
+                        if agg == "mean":
+                            if segc is None:
+                                ones = lib.alloc(n, lib.HF_INT64)
+                                lib.fill_i64(ones.dptr(), 1, n)
+                                segc = lib.seg_cumsum(ones, head,
+                                                      lib.AGG_SUM)
+                            brc = lib.gather(lib.gather(segc, ends_m1),
+                                             rid)
+                            br = lib.binary(lib.BIN_DIV, lib.cast_f64(br),
+                                            lib.cast_f64(brc))
+                        else:
+                            # all-NaN group: min/max is NaN, not ±inf
+                            brc = lib.gather(lib.gather(segc, ends_m1),
+                                             rid)
+                            br = lib.fixup_empty(
+                                lib.cast_f64(br),
+                                lib.compare_scalar(lib.CMP_GE,
+                                                   lib.cast_f64(brc),
+                                                   1.0))
+                res = lib.scatter(br, perm)
+                if valid is not None:
+                    res = lib.fixup_empty(lib.cast_f64(res), valid)
+                out_cols[v] = res
+                is_int = (valid is None
+                          and (agg == "count"
+                               or (src_int
+                                   and agg in ("sum", "min", "max"))))
+                dts[v] = np.dtype(np.int64 if is_int else np.float64)
             part = HipDataframePartition(DeviceBlock(out_cols, n))
             return HipDataframe([part], pandas.RangeIndex(n), val_names,
                                 [n], pandas.Series(dts))

@@ -687,6 +687,18 @@ class DataFrameGroupBy:
         return self._transform("rank", ascending=bool(ascending),
                                method=method)
 
+    def transform(self, func):
+        """pandas DataFrameGroupBy.transform: broadcast aggregates
+        ('sum'/'mean'/'count'/'min'/'max') back to every row, or the
+        same-length transforms ('cumsum'/'cummin'/'cummax'/'rank') by
+        name."""
+        if func in ("sum", "mean", "count", "min", "max"):
+            return self._transform("b" + func)
+        if func in ("cumsum", "cummin", "cummax", "rank"):
+            return self._transform(func)
+        raise lib.HfError(f"groupby.transform({func!r}) not supported "
+                          "(named aggs/transforms only)")
+
     def size(self):
         """pandas DataFrameGroupBy.size(): a Series of group row counts
         (NaN values included, NaN keys dropped)."""

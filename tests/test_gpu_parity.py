@@ -1395,3 +1395,46 @@ def test_merge_cross_vs_pandas(npartitions):
             np.testing.assert_allclose(got[c].to_numpy(),
                                        exp[c].to_numpy(), rtol=0,
                                        equal_nan=True, err_msg=c)
+
+
+def test_groupby_transform_broadcast_vs_pandas(npartitions):
+    """gb.transform('sum'/'mean'/'count'/'min'/'max'): per-group aggregate
+    broadcast to every row (NaN-value rows receive the group result; NaN
+    keys get NaN; all-NaN groups: sum 0.0, min/max/mean NaN; int64
+    survives only with valid keys)."""
+    rng = np.random.default_rng(102)
+    n = 100_000
+    k = rng.integers(0, 400, n).astype(np.float64)
+    k[rng.random(n) < 0.02] = np.nan
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.05] = np.nan
+    w = rng.integers(-50, 50, n)
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    for agg in ("sum", "mean", "count", "min", "max"):
+        got = df.groupby("k").transform(agg).to_pandas()
+        exp = pdf.groupby("k").transform(agg)
+        for c in exp.columns:
+            np.testing.assert_allclose(
+                got[c].to_numpy().astype(float),
+                exp[c].to_numpy().astype(float),
+                rtol=1e-12 if agg in ("sum", "mean") else 0, atol=1e-9,
+                equal_nan=True, err_msg=f"{agg}/{c}")
+    # all-NaN group + int dtype rule with valid keys
+    pdf2 = pandas.DataFrame({"k": [1, 1, 2, 2, 3],
+                             "v": [np.nan, np.nan, 1.0, 2.0, 5.0],
+                             "w": [1, 2, 3, 4, 5]})
+    df2 = mpd.DataFrame(pdf2)
+    for agg in ("sum", "mean", "count", "min", "max"):
+        got2 = df2.groupby("k").transform(agg).to_pandas()
+        exp2 = pdf2.groupby("k").transform(agg)
+        assert list(got2.dtypes) == list(exp2.dtypes), agg
+        for c in exp2.columns:
+            np.testing.assert_allclose(
+                got2[c].to_numpy().astype(float),
+                exp2[c].to_numpy().astype(float), rtol=0,
+                equal_nan=True, err_msg=f"{agg}/{c}")
+    s_ = df2.groupby("k")["v"].transform("mean").to_pandas()
+    np.testing.assert_allclose(
+        s_.to_numpy(), pdf2.groupby("k")["v"].transform("mean").to_numpy(),
+        rtol=0, equal_nan=True)
