@@ -1422,6 +1422,143 @@ class HipDataframe:
         return HipDataframe([part], pandas.RangeIndex(n), val_names, [n],
                             pandas.Series(dts))
 
+    def rank_rows(self, ascending: bool = True,
+                  method: str = "average") -> "HipDataframe":
+        """Frame-level pandas rank(axis=0): the groupby rank machinery
+        over ONE synthetic constant-key group (a zeros key column)."""
+        parts = []
+        for p in self._partitions:
+            b = p.block()
+            z = lib.alloc(b.length, lib.HF_INT64)
+            lib.fill_i64(z.dptr(), 0, b.length)
+            cols = dict(b.columns)
+            cols[self.KEYCOL] = z
+            parts.append(HipDataframePartition(
+                DeviceBlock(cols, b.length, b.cats)))
+        dtypes = pandas.concat([self.dtypes, pandas.Series(
+            {self.KEYCOL: np.dtype(np.int64)})])
+        tmp = HipDataframe(parts, self._index,
+                           list(self.columns) + [self.KEYCOL],
+                           self._row_lengths, dtypes)
+        return tmp.groupby_transform(self.KEYCOL, "rank",
+                                     ascending=ascending, method=method)
+
+    def groupby_idxminmax(self, by, maximum: bool) -> "HipDataframe":
+        """groupby.idxmax/idxmin: per group and value column, the ORIGINAL
+        row position (RangeIndex label) of the first occurrence of the
+        extreme; all-NaN groups report NaN (float64 column, pandas 2.3).
+        Composition: stable sort by (key, value) — the extreme's tie block
+        start IS the first original occurrence; non-NaN counts locate the
+        last non-NaN row (values sort NaN-last within each key run)."""
+        from ..distributed import is_active
+        if is_active():
+            raise lib.HfError("distributed groupby idxmax/idxmin is a "
+                              "later round")
+        by_list = [by] if isinstance(by, str) else list(by)
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        val_names = [c for c in self.columns if c not in by_list]
+        for v in val_names:
+            if v in blk_cats:
+                raise lib.HfError("idxmax/idxmin over string columns")
+        if not isinstance(self._index, pandas.RangeIndex) or \
+                self._index.start != 0 or self._index.step != 1:
+            raise lib.HfError("groupby idxmax/idxmin: only RangeIndex "
+                              "frames this round")
+
+        def concat_col(name):
+            cols = [p.block().columns[name] for p in self._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        n = len(self)
+        res0 = self.groupby_size(by)      # group-key index, NaN dropped
+        ngv = len(res0)
+        out_cols, dts = {}, {}
+        if n == 0 or ngv == 0:
+            part = HipDataframePartition(DeviceBlock(
+                {v: lib.alloc(0, lib.HF_INT64) for v in val_names}, 0))
+            return HipDataframe([part], res0._index, val_names, [0],
+                                pandas.Series({v: np.dtype(np.int64)
+                                               for v in val_names}))
+        if len(by_list) > 1:
+            # trailing-sentinel slicing below assumes NaN-key rows sort
+            # last, which only holds for a single key column
+            for b in by_list:
+                c = concat_col(b)
+                bad = (lib.reduce(c).imn < 0) if b in blk_cats else (
+                    c.dtype_code == lib.HF_FLOAT64
+                    and lib.reduce(c).count != c.length)
+                if c.length and bad:
+                    raise lib.HfError("idxmax/idxmin: NaN in a multi-key "
+                                      "groupby key is a later round")
+        eff_keys = [self._effective_sort_key(concat_col(b), b in blk_cats,
+                                             True)
+                    for b in by_list]
+        for v in val_names:
+            vc = concat_col(v)
+            eff_v = self._effective_sort_key(vc, False, True)
+            perm = self._compose_sort_perm(eff_keys + [eff_v])
+            khead = None
+            for ekc, _ in eff_keys:
+                h = self._run_head_col(lib.gather(ekc, perm), n)
+                khead = h if khead is None else lib.binary(lib.BIN_ADD,
+                                                           khead, h)
+            if len(eff_keys) > 1:
+                khead = lib.compare_scalar(lib.CMP_GE, khead, 1.0)
+            kplan = lib.filter_plan(khead)
+            ng = kplan.n_kept
+            hp = lib.filter_iota(kplan, 0)
+            # non-NaN count per run (values sort NaN-last within the run)
+            sv = lib.gather(vc, perm)
+            if vc.dtype_code == lib.HF_FLOAT64:
+                m = lib.compare_scalar(lib.CMP_NOTNA, sv, 0.0)
+            else:
+                m = lib.alloc(n, lib.HF_INT64)
+                lib.fill_i64(m.dptr(), 1, n)
+            segc = lib.seg_cumsum(m, khead, lib.AGG_SUM)
+            ends = lib.concat([lib.col_slice(hp, 1, ng - 1),
+                               self._const_i64(n)]) if ng > 1 \
+                else self._const_i64(n)
+            cnt = lib.gather(segc, lib.map_scalar(lib.MAP_SUB, ends, 1))
+            if maximum:
+                # last non-NaN row of the run, then its tie-block start
+                pos = lib.binary(
+                    lib.BIN_ADD, hp,
+                    lib.map_scalar(lib.MAP_MAX,
+                                   lib.map_scalar(lib.MAP_SUB, cnt, 1),
+                                   0))
+                vhead = self._run_head_col(lib.gather(eff_v[0], perm), n)
+                thead = lib.compare_scalar(
+                    lib.CMP_GE, lib.binary(lib.BIN_ADD, khead, vhead),
+                    1.0)
+                trid = lib.map_scalar(lib.MAP_SUB, lib.cumsum(thead), 1)
+                tplan = lib.filter_plan(thead)
+                t_hp = lib.filter_iota(tplan, 0)
+                tstart_row = lib.gather(t_hp, trid)   # [n]
+                ts = lib.gather(tstart_row, pos)      # [ng]
+            else:
+                ts = hp  # ascending sort: the min's first occurrence
+            orig = lib.gather(perm, ts)               # [ng]
+            # keep only the valid (non-NaN-key) groups: the sentinel runs
+            # sort last, so they are the trailing ng - ngv runs
+            orig = lib.col_slice(orig, 0, ngv)
+            cntv = lib.col_slice(cnt, 0, ngv)
+            n_empty = ngv - int(lib.reduce(
+                lib.compare_scalar(lib.CMP_GE, lib.cast_f64(cntv),
+                                   1.0)).isum)
+            if n_empty:
+                orig = lib.fixup_empty(
+                    lib.cast_f64(orig),
+                    lib.compare_scalar(lib.CMP_GE, lib.cast_f64(cntv),
+                                       1.0))
+                dts[v] = np.dtype(np.float64)
+            else:
+                dts[v] = np.dtype(np.int64)
+            out_cols[v] = orig
+        part = HipDataframePartition(DeviceBlock(out_cols, ngv))
+        return HipDataframe([part], res0._index, val_names, [ngv],
+                            pandas.Series(dts))
+
     def _groupby_rank(self, by_list, val_names, eff_keys, valid, n,
                       concat_col, ascending, method):
         """rank within groups (pandas DataFrameGroupBy.rank,

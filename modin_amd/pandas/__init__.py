@@ -101,6 +101,14 @@ class _HipPandasBase:
     def var(self, ddof: int = 1):
         return self._lower(self._query_compiler.var(ddof=ddof))
 
+    def rank(self, method: str = "average", ascending: bool = True,
+             na_option: str = "keep"):
+        """pandas rank(axis=0): always float64; na_option='keep'."""
+        if na_option != "keep":
+            raise lib.HfError("rank: only na_option='keep' this round")
+        return self._rewrap(self._query_compiler.rank(
+            method=method, ascending=bool(ascending)))
+
     def cumsum(self):
         return self._rewrap(self._query_compiler.cumsum())
 
@@ -686,6 +694,24 @@ class DataFrameGroupBy:
             raise lib.HfError("rank: only na_option='keep' this round")
         return self._transform("rank", ascending=bool(ascending),
                                method=method)
+
+    def idxmax(self):
+        """Original row label of each group's first max per column
+        (all-NaN groups: NaN)."""
+        qc = self._df._query_compiler.groupby_idxmax(self._by)
+        out = DataFrame(query_compiler=qc)
+        if self._series_out and self._as_index:
+            name = list(qc._modin_frame.columns)[0]
+            return Series(query_compiler=qc, name=name)
+        return out
+
+    def idxmin(self):
+        qc = self._df._query_compiler.groupby_idxmin(self._by)
+        out = DataFrame(query_compiler=qc)
+        if self._series_out and self._as_index:
+            name = list(qc._modin_frame.columns)[0]
+            return Series(query_compiler=qc, name=name)
+        return out
 
     def transform(self, func):
         """pandas DataFrameGroupBy.transform: broadcast aggregates

@@ -189,6 +189,19 @@ class HipQueryCompiler:
         return self.__constructor__(self._modin_frame.groupby_transform(
             by, how, ascending=ascending, method=method, periods=periods))
 
+    def groupby_idxmax(self, by) -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.groupby_idxminmax(by, maximum=True))
+
+    def groupby_idxmin(self, by) -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.groupby_idxminmax(by, maximum=False))
+
+    def rank(self, method: str = "average",
+             ascending: bool = True) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.rank_rows(
+            ascending=ascending, method=method))
+
     def groupby_size(self, by: str) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.groupby_size(by))
 

@@ -1438,3 +1438,77 @@ def test_groupby_transform_broadcast_vs_pandas(npartitions):
     np.testing.assert_allclose(
         s_.to_numpy(), pdf2.groupby("k")["v"].transform("mean").to_numpy(),
         rtol=0, equal_nan=True)
+
+
+def test_groupby_idxmax_idxmin_vs_pandas(npartitions):
+    """gb.idxmax/idxmin: first original row label of the group extreme;
+    all-NaN groups -> NaN (float64 column); ties resolve to first
+    occurrence; NaN keys dropped."""
+    rng = np.random.default_rng(103)
+    n = 50_000
+    k = rng.integers(0, 150, n).astype(np.float64)
+    k[rng.random(n) < 0.02] = np.nan
+    v = rng.integers(-6, 6, n).astype(np.float64)  # heavy ties
+    v[rng.random(n) < 0.05] = np.nan
+    w = rng.integers(-100, 100, n)
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    for fn in ("idxmax", "idxmin"):
+        got = getattr(df.groupby("k"), fn)().to_pandas()
+        exp = getattr(pdf.groupby("k"), fn)()
+        np.testing.assert_array_equal(got.index.to_numpy(),
+                                      exp.index.to_numpy())
+        for c in exp.columns:
+            np.testing.assert_allclose(
+                got[c].to_numpy().astype(float),
+                exp[c].to_numpy().astype(float), rtol=0, equal_nan=True,
+                err_msg=f"{fn}/{c}")
+    # all-NaN group + int columns + selection form
+    pdf2 = pandas.DataFrame({"k": [1, 1, 2, 2, 3],
+                             "v": [3.0, 3.0, np.nan, np.nan, 1.0],
+                             "w": [9, 2, 3, 4, 5]})
+    df2 = mpd.DataFrame(pdf2)
+    for fn in ("idxmax", "idxmin"):
+        got2 = getattr(df2.groupby("k"), fn)().to_pandas()
+        exp2 = getattr(pdf2.groupby("k"), fn)()
+        assert list(got2.dtypes) == list(exp2.dtypes), fn
+        for c in exp2.columns:
+            np.testing.assert_allclose(
+                got2[c].to_numpy().astype(float),
+                exp2[c].to_numpy().astype(float), rtol=0, equal_nan=True,
+                err_msg=f"{fn}/{c}")
+    s_ = df2.groupby("k")["w"].idxmax().to_pandas()
+    np.testing.assert_array_equal(
+        s_.to_numpy(), pdf2.groupby("k")["w"].idxmax().to_numpy())
+    # multi-key (no NaN keys)
+    pdf3 = pandas.DataFrame({"a": rng.integers(0, 5, 3000),
+                             "b": rng.choice(["x", "y"], 3000),
+                             "v": rng.standard_normal(3000)})
+    df3 = mpd.DataFrame(pdf3)
+    got3 = df3.groupby(["a", "b"]).idxmax().to_pandas()
+    exp3 = pdf3.groupby(["a", "b"]).idxmax()
+    np.testing.assert_array_equal(got3["v"].to_numpy(),
+                                  exp3["v"].to_numpy())
+
+
+def test_frame_rank_vs_pandas(npartitions):
+    """DataFrame.rank / Series.rank (axis=0): the one-group rank
+    composition."""
+    rng = np.random.default_rng(104)
+    n = 40_000
+    v = rng.integers(-30, 30, n).astype(np.float64)
+    v[rng.random(n) < 0.05] = np.nan
+    w = rng.integers(-1000, 1000, n)
+    pdf = pandas.DataFrame({"v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    for method in ("average", "min", "first"):
+        for asc in (True, False):
+            got = df.rank(method=method, ascending=asc).to_pandas()
+            exp = pdf.rank(method=method, ascending=asc)
+            for c in exp.columns:
+                np.testing.assert_allclose(
+                    got[c].to_numpy(), exp[c].to_numpy(), rtol=0,
+                    equal_nan=True, err_msg=f"{method}/{asc}/{c}")
+    s_ = df["v"].rank().to_pandas()
+    np.testing.assert_allclose(s_.to_numpy(), pdf["v"].rank().to_numpy(),
+                               rtol=0, equal_nan=True)
