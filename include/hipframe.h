@@ -101,7 +101,9 @@ enum {
   HF_MAP_CAST_I64 = 10, /* (int64)x : f64 -> i64, C truncation (astype)    */
   HF_MAP_SQRT = 11, /* sqrt(x), f64 only (std = sqrt(var))                  */
   HF_MAP_MIN = 12,  /* fmin(x, s) — clip upper; NaN passes through (f64)    */
-  HF_MAP_MAX = 13   /* fmax(x, s) — clip lower; NaN passes through (f64)    */
+  HF_MAP_MAX = 13,  /* fmax(x, s) — clip lower; NaN passes through (f64)    */
+  HF_MAP_ROUND = 14 /* rint(x*s)/s, f64 only (pandas round(d): s = 10^d;
+                       half-even like numpy)                               */
 };
 int hf_map_scalar(int op, const hf_col* in, double scalar, hf_col** out);
 /* i64 column with an exact int64 scalar (double cannot hold all int64). */

@@ -1184,8 +1184,8 @@ class HipDataframe:
         return c
 
     def groupby_transform(self, by, how: str, ascending: bool = True,
-                          method: str = "average",
-                          periods: int = 1) -> "HipDataframe":
+                          method: str = "average", periods: int = 1,
+                          dropna: bool = True) -> "HipDataframe":
         """Same-length groupby transforms in original row order.
 
         how: 'cumsum' | 'cummin' | 'cummax' (segmented scan), 'cumcount'
@@ -1243,9 +1243,11 @@ class HipDataframe:
             cols = [p.block().columns[name] for p in self._partitions]
             return cols[0] if len(cols) == 1 else lib.concat(cols)
 
-        # per-row key validity in ORIGINAL order (pandas dropna=True)
+        # per-row key validity in ORIGINAL order (pandas dropna=True);
+        # dropna=False keeps NaN keys as a real group (the canonical-NaN
+        # effective key makes all NaNs one run) — duplicated() rides this
         valid = None
-        for b in by_list:
+        for b in by_list if dropna else []:
             c = concat_col(b)
             if b in blk_cats:
                 m = lib.compare_scalar(lib.CMP_GE, c, 0.0)
@@ -1622,6 +1624,83 @@ class HipDataframe:
         dts = pandas.Series({v: np.dtype(np.float64) for v in val_names})
         return HipDataframe([part], pandas.RangeIndex(n), val_names, [n],
                             dts)
+
+    def where_rows(self, mask_frame: "HipDataframe",
+                   other=None) -> "HipDataframe":
+        """pandas where(cond, other): keep values where the row mask is
+        true, else `other` (NaN default).  Exact NaN bookkeeping: an
+        original NaN under a TRUE cond stays NaN even when `other` fills
+        the false rows."""
+        mcols = []
+        for p in mask_frame._partitions:
+            b = p.block()
+            mcols.append(b.columns[list(b.columns)[0]])
+        m = mcols[0] if len(mcols) == 1 else lib.concat(mcols)
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+
+        def concat_col(name):
+            cols = [p.block().columns[name] for p in self._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        n = len(self)
+        if m.length != n:
+            raise lib.HfError("where/mask: condition length mismatch")
+        out_cols, dts, cats = {}, {}, {}
+        for c in self.columns:
+            col = concat_col(c)
+            if c in blk_cats:
+                if other is not None:
+                    raise lib.HfError("where/mask with a fill value over "
+                                      "string columns is a later round")
+                # codes: (code+1)*m - 1 -> untouched codes / −1 (NaN)
+                t = lib.map_scalar(
+                    lib.MAP_SUB,
+                    lib.binary(lib.BIN_MUL,
+                               lib.map_scalar(lib.MAP_ADD, col, 1), m), 1)
+                out_cols[c] = t
+                dts[c] = self.dtypes[c]
+                cats[c] = blk_cats[c]
+            elif (col.dtype_code == lib.HF_INT64 and other is not None
+                    and isinstance(other, (int, np.integer))):
+                t1 = lib.binary(lib.BIN_MUL, col, m)
+                t2 = lib.map_scalar(
+                    lib.MAP_MUL, lib.map_scalar(lib.MAP_RSUB, m, 1),
+                    int(other))
+                out_cols[c] = lib.binary(lib.BIN_ADD, t1, t2)
+                dts[c] = np.dtype(np.int64)
+            else:
+                cf = lib.cast_f64(col)
+                t = lib.fixup_empty(cf, m)
+                if other is not None:
+                    isna = lib.map_scalar(
+                        lib.MAP_RSUB,
+                        lib.compare_scalar(lib.CMP_NOTNA, cf, 0.0), 1)
+                    nan_keep = lib.binary(lib.BIN_MUL, isna, m)
+                    t = lib.map_scalar(lib.MAP_FILLNA, t, float(other))
+                    t = lib.fixup_empty(
+                        t, lib.map_scalar(lib.MAP_RSUB, nan_keep, 1))
+                out_cols[c] = t
+                dts[c] = np.dtype(np.float64)
+        part = HipDataframePartition(DeviceBlock(out_cols, n, cats))
+        return HipDataframe([part], self._index, list(self.columns), [n],
+                            pandas.Series(dts))
+
+    def round_cols(self, decimals: int = 0) -> "HipDataframe":
+        """pandas round(decimals): half-even via rint(x*10^d)/10^d (the
+        numpy scaling rule); int columns unchanged."""
+        scale = float(10.0 ** int(decimals))
+
+        def block_fn(block: DeviceBlock) -> DeviceBlock:
+            out = {}
+            for name, col in block.columns.items():
+                if (col.dtype_code == lib.HF_FLOAT64
+                        and name not in block.cats):
+                    out[name] = lib.map_scalar(lib.MAP_ROUND, col, scale)
+                else:
+                    out[name] = col
+            return DeviceBlock(out, block.length, block.cats)
+        return self.map(block_fn)
 
     def hconcat(self, others: list) -> "HipDataframe":
         """Horizontal compose of single-partition frames with identical row

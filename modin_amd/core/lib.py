@@ -45,7 +45,8 @@ HF_FLOAT64 = 1
 
 # map ops
 MAP_ADD, MAP_SUB, MAP_RSUB, MAP_MUL, MAP_DIV, MAP_RDIV, MAP_FILLNA, MAP_ABS, \
-    MAP_NEG, MAP_CAST_F64, MAP_CAST_I64, MAP_SQRT, MAP_MIN, MAP_MAX = range(14)
+    MAP_NEG, MAP_CAST_F64, MAP_CAST_I64, MAP_SQRT, MAP_MIN, MAP_MAX, \
+    MAP_ROUND = range(15)
 # binary ops
 BIN_ADD, BIN_SUB, BIN_MUL, BIN_DIV, BIN_MIN, BIN_MAX = range(6)
 

@@ -251,6 +251,7 @@ __device__ __forceinline__ double map1_f64(double x, double s) {
     case HF_MAP_SQRT:   return sqrt(x);
     // NaN-PROPAGATING (pandas clip leaves NaN alone; fmin/fmax would
     // replace NaN with the bound)
+    case HF_MAP_ROUND:  return rint(x * s) / s;
     case HF_MAP_MIN:    return (x != x) ? x : fmin(x, s);
     case HF_MAP_MAX:    return (x != x) ? x : fmax(x, s);
   }
@@ -2308,6 +2309,7 @@ int hf_map_scalar(int op, const hf_col* in, double scalar, hf_col** out) {
     case HF_MAP_SQRT:   rc = launch_map_f64<HF_MAP_SQRT>(in, scalar, *out); break;
     case HF_MAP_MIN:    rc = launch_map_f64<HF_MAP_MIN>(in, scalar, *out); break;
     case HF_MAP_MAX:    rc = launch_map_f64<HF_MAP_MAX>(in, scalar, *out); break;
+    case HF_MAP_ROUND:  rc = launch_map_f64<HF_MAP_ROUND>(in, scalar, *out); break;
     default:
       hf_col_free(*out);
       *out = nullptr;

@@ -109,6 +109,37 @@ class _HipPandasBase:
         return self._rewrap(self._query_compiler.rank(
             method=method, ascending=bool(ascending)))
 
+    def round(self, decimals: int = 0):  # noqa: A003
+        """pandas round: half-even on float columns, ints unchanged."""
+        return self._rewrap(self._query_compiler.round(int(decimals)))
+
+    def where(self, cond, other=None):
+        """pandas where(cond, other): cond is a boolean Series row mask;
+        false rows become `other` (NaN default)."""
+        if not isinstance(cond, Series):
+            raise lib.HfError("where/mask take a boolean Series condition")
+        return self._rewrap(self._query_compiler.where_mask(
+            cond._query_compiler, other))
+
+    def mask(self, cond, other=None):  # noqa: A003
+        """pandas mask = where(~cond)."""
+        if not isinstance(cond, Series):
+            raise lib.HfError("where/mask take a boolean Series condition")
+        return self.where(~cond, other)
+
+    def duplicated(self, subset=None):
+        """pandas duplicated(keep='first'): boolean Series; NaN keys
+        compare equal (pandas semantics)."""
+        qc = self._query_compiler.duplicated(subset)
+        out = Series(query_compiler=qc, name=None)
+        out._bool_mask = True
+        return out
+
+    def drop_duplicates(self, subset=None):
+        """pandas drop_duplicates(keep='first'): original index labels of
+        the kept rows."""
+        return self._rewrap(self._query_compiler.drop_duplicates(subset))
+
     def cumsum(self):
         return self._rewrap(self._query_compiler.cumsum())
 
@@ -431,6 +462,16 @@ class DataFrame(_HipPandasBase):
                     "groupby(by=<column name> | [column names]) only")
         return DataFrameGroupBy(self, by, as_index=as_index)
 
+    def nlargest(self, n: int, columns: str):
+        """pandas nlargest(keep='first'): stable descending sort, NaN
+        excluded (head clipped to the non-NaN count)."""
+        k = min(int(n), int(self[columns].count()))
+        return self.sort_values(columns, ascending=False).head(k)
+
+    def nsmallest(self, n: int, columns: str):
+        k = min(int(n), int(self[columns].count()))
+        return self.sort_values(columns, ascending=True).head(k)
+
     def to_pandas(self) -> pandas.DataFrame:
         return self._query_compiler.to_pandas()
 
@@ -486,6 +527,21 @@ class Series(_HipPandasBase):
         qc = self._query_compiler.sort_rows_by_column_values(
             name, ascending, na_position=na_position)
         return Series(query_compiler=qc, name=self.name)
+
+    def head(self, n: int = 5):
+        return Series(
+            query_compiler=self._query_compiler.take_row_range(0, n),
+            name=self.name)
+
+    def nlargest(self, n: int = 5):
+        """pandas Series.nlargest(keep='first'): stable descending sort,
+        NaN excluded."""
+        k = min(int(n), int(self.count()))
+        return self.sort_values(ascending=False).head(k)
+
+    def nsmallest(self, n: int = 5):
+        k = min(int(n), int(self.count()))
+        return self.sort_values(ascending=True).head(k)
 
     def unique(self):
         """pandas Series.unique: distinct values in FIRST-APPEARANCE order,

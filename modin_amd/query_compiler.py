@@ -202,6 +202,30 @@ class HipQueryCompiler:
         return self.__constructor__(self._modin_frame.rank_rows(
             ascending=ascending, method=method))
 
+    def duplicated(self, subset=None) -> "HipQueryCompiler":
+        """Row-duplicate mask, keep='first' (pandas duplicated): cumcount
+        over ALL subset columns with dropna=False (NaN==NaN, the
+        canonical-NaN effective key) > 0."""
+        by = (list(self._modin_frame.columns) if subset is None
+              else ([subset] if isinstance(subset, str) else list(subset)))
+        cc = self._modin_frame.groupby_transform(by, "cumcount",
+                                                 dropna=False)
+        return self.__constructor__(cc.compare_scalar(lib.CMP_GE, 1.0))
+
+    def drop_duplicates(self, subset=None) -> "HipQueryCompiler":
+        dup = self.duplicated(subset)
+        inv = dup._modin_frame.compare_scalar(lib.CMP_EQ, 0.0)
+        return self.__constructor__(self._modin_frame.filter_rows(inv))
+
+    def where_mask(self, cond: "HipQueryCompiler",
+                   other=None) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.where_rows(
+            cond._modin_frame, other))
+
+    def round(self, decimals: int = 0) -> "HipQueryCompiler":  # noqa: A003
+        return self.__constructor__(
+            self._modin_frame.round_cols(decimals))
+
     def groupby_size(self, by: str) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.groupby_size(by))
 

@@ -1512,3 +1512,93 @@ def test_frame_rank_vs_pandas(npartitions):
     s_ = df["v"].rank().to_pandas()
     np.testing.assert_allclose(s_.to_numpy(), pdf["v"].rank().to_numpy(),
                                rtol=0, equal_nan=True)
+
+
+def test_duplicated_drop_duplicates_vs_pandas(npartitions):
+    """duplicated/drop_duplicates keep='first': cumcount>0 over ALL
+    subset columns with dropna=False (NaN==NaN, pandas semantics);
+    drop_duplicates keeps original index labels."""
+    rng = np.random.default_rng(105)
+    n = 40_000
+    a = rng.integers(0, 40, n).astype(np.float64)
+    a[rng.random(n) < 0.05] = np.nan
+    b = rng.choice(["x", "y", "zz"], n)
+    w = rng.integers(0, 6, n)
+    pdf = pandas.DataFrame({"a": a, "b": b, "w": w})
+    df = mpd.DataFrame(pdf)
+    for subset in (None, ["a"], ["a", "b"], "w"):
+        got = df.duplicated(subset).to_pandas()
+        exp = pdf.duplicated(subset=subset)
+        np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy(),
+                                      err_msg=str(subset))
+    got_d = df.drop_duplicates(["a", "b"]).to_pandas()
+    exp_d = pdf.drop_duplicates(subset=["a", "b"])
+    np.testing.assert_array_equal(got_d.index.to_numpy(),
+                                  exp_d.index.to_numpy())
+    np.testing.assert_array_equal(got_d["w"].to_numpy(),
+                                  exp_d["w"].to_numpy())
+    s_ = df["w"].drop_duplicates().to_pandas()
+    e_ = pdf["w"].drop_duplicates()
+    np.testing.assert_array_equal(s_.index.to_numpy(), e_.index.to_numpy())
+    np.testing.assert_array_equal(s_.to_numpy(), e_.to_numpy())
+
+
+def test_where_mask_round_vs_pandas(npartitions):
+    rng = np.random.default_rng(106)
+    n = 30_000
+    v = rng.standard_normal(n) * 10
+    v[rng.random(n) < 0.1] = np.nan
+    w = rng.integers(-50, 50, n)
+    pdf = pandas.DataFrame({"v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    cond_p = pdf["v"] > 0
+    cond_m = df["v"] > 0
+    got = df.where(cond_m).to_pandas()
+    exp = pdf.where(cond_p)
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(), rtol=0,
+                               equal_nan=True)
+    got = df.where(cond_m, -1).to_pandas()
+    exp = pdf.where(cond_p, -1)
+    assert list(got.dtypes) == list(exp.dtypes)
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(), rtol=0,
+                               equal_nan=True)
+    got = df.mask(cond_m).to_pandas()
+    exp = pdf.mask(cond_p)
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(), rtol=0,
+                               equal_nan=True)
+    s_ = df["v"].where(cond_m, 0.5).to_pandas()
+    np.testing.assert_allclose(s_.to_numpy(),
+                               pdf["v"].where(cond_p, 0.5).to_numpy(),
+                               rtol=0, equal_nan=True)
+    for d in (0, 1, 2, -1):
+        got = df.round(d).to_pandas()
+        exp = pdf.round(d)
+        assert list(got.dtypes) == list(exp.dtypes)
+        np.testing.assert_allclose(got["v"].to_numpy(),
+                                   exp["v"].to_numpy(), rtol=0,
+                                   equal_nan=True, err_msg=f"round({d})")
+        np.testing.assert_array_equal(got["w"].to_numpy(),
+                                      exp["w"].to_numpy())
+
+
+def test_nlargest_nsmallest_vs_pandas(npartitions):
+    rng = np.random.default_rng(107)
+    n = 20_000
+    v = rng.integers(-100, 100, n).astype(np.float64)  # ties
+    v[rng.random(n) < 0.05] = np.nan
+    w = rng.integers(0, 9, n)
+    pdf = pandas.DataFrame({"v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    for k in (5, 100, n + 50):
+        got = df.nlargest(k, "v").to_pandas()
+        exp = pdf.nlargest(k, "v")
+        np.testing.assert_array_equal(got.index.to_numpy(),
+                                      exp.index.to_numpy(), err_msg=str(k))
+        got = df.nsmallest(k, "v").to_pandas()
+        exp = pdf.nsmallest(k, "v")
+        np.testing.assert_array_equal(got.index.to_numpy(),
+                                      exp.index.to_numpy())
+    s_ = df["v"].nlargest(17).to_pandas()
+    e_ = pdf["v"].nlargest(17)
+    np.testing.assert_array_equal(s_.index.to_numpy(), e_.index.to_numpy())
+    np.testing.assert_allclose(s_.to_numpy(), e_.to_numpy(), rtol=0)
