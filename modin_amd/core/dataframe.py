@@ -1224,13 +1224,14 @@ class HipDataframe:
         blk_cats = (self._partitions[0].block().cats
                     if self._partitions else {})
         val_names = [c for c in self.columns if c not in by_list]
-        for v in val_names:
-            if v in blk_cats:
-                raise lib.HfError(
-                    f"groupby {how}: string column {v!r} unsupported "
-                    "(pandas raises on non-numeric transforms)")
         n = len(self)
         needs_vals = how not in ("cumcount", "ngroup")
+        if needs_vals:
+            for v in val_names:
+                if v in blk_cats:
+                    raise lib.HfError(
+                        f"groupby {how}: string column {v!r} unsupported "
+                        "(pandas raises on non-numeric transforms)")
         if n == 0 or (needs_vals and not val_names):
             names = val_names if needs_vals else [how]
             dts = pandas.Series({v: np.dtype(np.float64) for v in names})
