@@ -1695,9 +1695,17 @@ class HipDataframe:
         def block_fn(block: DeviceBlock) -> DeviceBlock:
             out = {}
             for name, col in block.columns.items():
-                if (col.dtype_code == lib.HF_FLOAT64
-                        and name not in block.cats):
+                if name in block.cats:
+                    out[name] = col
+                elif col.dtype_code == lib.HF_FLOAT64:
                     out[name] = lib.map_scalar(lib.MAP_ROUND, col, scale)
+                elif decimals < 0:
+                    # pandas rounds int columns too for negative decimals
+                    # (to tens/hundreds/…), keeping int64
+                    out[name] = lib.map_scalar(
+                        lib.MAP_CAST_I64,
+                        lib.map_scalar(lib.MAP_ROUND, lib.cast_f64(col),
+                                       scale), 0)
                 else:
                     out[name] = col
             return DeviceBlock(out, block.length, block.cats)
