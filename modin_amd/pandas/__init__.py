@@ -463,14 +463,13 @@ class DataFrame(_HipPandasBase):
         return DataFrameGroupBy(self, by, as_index=as_index)
 
     def nlargest(self, n: int, columns: str):
-        """pandas nlargest(keep='first'): stable descending sort, NaN
-        excluded (head clipped to the non-NaN count)."""
-        k = min(int(n), int(self[columns].count()))
-        return self.sort_values(columns, ascending=False).head(k)
+        """pandas nlargest(keep='first'): stable descending NaN-last sort
+        + head(n) (NaN rows only appear once n exceeds the non-NaN
+        count — pandas behavior)."""
+        return self.sort_values(columns, ascending=False).head(int(n))
 
     def nsmallest(self, n: int, columns: str):
-        k = min(int(n), int(self[columns].count()))
-        return self.sort_values(columns, ascending=True).head(k)
+        return self.sort_values(columns, ascending=True).head(int(n))
 
     def to_pandas(self) -> pandas.DataFrame:
         return self._query_compiler.to_pandas()
@@ -534,14 +533,12 @@ class Series(_HipPandasBase):
             name=self.name)
 
     def nlargest(self, n: int = 5):
-        """pandas Series.nlargest(keep='first'): stable descending sort,
-        NaN excluded."""
-        k = min(int(n), int(self.count()))
-        return self.sort_values(ascending=False).head(k)
+        """pandas Series.nlargest(keep='first'): stable descending
+        NaN-last sort + head(n)."""
+        return self.sort_values(ascending=False).head(int(n))
 
     def nsmallest(self, n: int = 5):
-        k = min(int(n), int(self.count()))
-        return self.sort_values(ascending=True).head(k)
+        return self.sort_values(ascending=True).head(int(n))
 
     def unique(self):
         """pandas Series.unique: distinct values in FIRST-APPEARANCE order,
