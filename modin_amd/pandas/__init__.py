@@ -156,6 +156,10 @@ class _HipPandasBase:
         """x - x.shift(periods) (pandas diff)."""
         return self._rewrap(self._query_compiler.diff(int(periods)))
 
+    def pct_change(self, periods: int = 1):
+        """pandas pct_change(fill_method=None): x / x.shift(p) - 1."""
+        return self / self.shift(periods) - 1
+
     def idxmax(self):
         return self._lower(self._query_compiler.idxmax())
 
@@ -518,6 +522,21 @@ class Series(_HipPandasBase):
         qc = self._query_compiler
         return Series(query_compiler=qc.getitem_array(qc.notna()),
                       name=self.name)
+
+    def between(self, left, right, inclusive: str = "both") -> "Series":
+        """pandas Series.between: boolean mask (NaN -> False)."""
+        if inclusive == "both":
+            out = (self >= left) & (self <= right)
+        elif inclusive == "neither":
+            out = (self > left) & (self < right)
+        elif inclusive == "left":
+            out = (self >= left) & (self < right)
+        elif inclusive == "right":
+            out = (self > left) & (self <= right)
+        else:
+            raise lib.HfError("between: bad 'inclusive'")
+        out._bool_mask = True
+        return out
 
     def sort_values(self, ascending: bool = True, kind: str = "stable",
                     na_position: str = "last"):

@@ -1615,3 +1615,25 @@ def test_nlargest_nsmallest_vs_pandas(npartitions):
     e_ = pdf["v"].nlargest(17)
     np.testing.assert_array_equal(s_.index.to_numpy(), e_.index.to_numpy())
     np.testing.assert_allclose(s_.to_numpy(), e_.to_numpy(), rtol=0)
+
+
+def test_pct_change_between_vs_pandas(npartitions):
+    rng = np.random.default_rng(108)
+    n = 20_000
+    v = rng.standard_normal(n) + 5
+    v[rng.random(n) < 0.05] = np.nan
+    pdf = pandas.DataFrame({"v": v, "w": rng.integers(1, 9, n)})
+    df = mpd.DataFrame(pdf)
+    for p in (1, 3):
+        got = df.pct_change(p).to_pandas()
+        exp = pdf.pct_change(p, fill_method=None)
+        for c in exp.columns:
+            np.testing.assert_allclose(got[c].to_numpy(),
+                                       exp[c].to_numpy(), rtol=1e-12,
+                                       atol=1e-12, equal_nan=True,
+                                       err_msg=f"pct({p})/{c}")
+    for inc in ("both", "neither", "left", "right"):
+        got = df["v"].between(4.0, 6.0, inclusive=inc).to_pandas()
+        exp = pdf["v"].between(4.0, 6.0, inclusive=inc)
+        np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy(),
+                                      err_msg=inc)
