@@ -61,7 +61,8 @@ def test_fuzz_pipeline(seed):
 
     steps = rng.integers(2, 5)
     for si in range(steps):
-        op = rng.choice(["filter", "sort", "head", "dropna", "arith"])
+        op = rng.choice(["filter", "sort", "head", "dropna", "arith",
+                         "round", "where", "dedup"])
         msg = f"seed {seed} step {si} op {op}"
         if op == "filter":
             thr = float(np.round(rng.standard_normal() * 10, 2))
@@ -89,6 +90,21 @@ def test_fuzz_pipeline(seed):
             pdf = pdf.dropna()
             df = mpd.DataFrame(df.to_pandas().reset_index(drop=True))
             pdf = pdf.reset_index(drop=True)
+        elif op == "round":
+            d = int(rng.integers(-1, 3))
+            df = df.round(d)
+            pdf = pdf.round(d)
+        elif op == "where":
+            thr = float(np.round(rng.standard_normal() * 5, 2))
+            df = df.where(df["v"] > thr)
+            pdf = pdf.where(pdf["v"] > thr)
+        elif op == "dedup":
+            subs = [["k"], ["k", "s"], ["s", "w"], None][rng.integers(0, 4)]
+            df = df.drop_duplicates(subs)
+            pdf = pdf.drop_duplicates(subset=subs)
+            check(df, pdf, msg)
+            df = mpd.DataFrame(df.to_pandas().reset_index(drop=True))
+            pdf = pdf.reset_index(drop=True)
         else:
             c = float(np.round(rng.standard_normal(), 3))
             sub = [col for col in ("v", "w") if col in pdf.columns]
@@ -102,6 +118,37 @@ def test_fuzz_pipeline(seed):
         check(df, pdf, msg)
         if len(pdf) == 0:
             break
+
+    # closing transform on whatever survived (original-row-order family)
+    if len(pdf):
+        by = ["k", "s"][rng.integers(0, 2)]
+        tr = ["cumsum", "cumcount", "shift", "rank", "tsum",
+              "tmean"][rng.integers(0, 6)]
+        sub = [by, "v", "w"]
+        gb_g = df[sub].groupby(by)
+        gb_p = pdf[sub].groupby(by)
+        if tr == "cumcount":
+            g = gb_g.cumcount().to_pandas().to_numpy().astype(float)
+            e = gb_p.cumcount().to_numpy().astype(float)
+            np.testing.assert_allclose(g, e, rtol=0, equal_nan=True,
+                                       err_msg=f"seed {seed} cumcount")
+        else:
+            if tr in ("tsum", "tmean"):
+                gout = gb_g.transform(tr[1:]).to_pandas()
+                pout = gb_p.transform(tr[1:])
+            elif tr == "shift":
+                pr = int(rng.integers(-2, 3)) or 1
+                gout = gb_g.shift(pr).to_pandas()
+                pout = gb_p.shift(pr)
+            else:
+                gout = getattr(gb_g, tr)().to_pandas()
+                pout = getattr(gb_p, tr)()
+            for c in pout.columns:
+                np.testing.assert_allclose(
+                    gout[c].to_numpy().astype(float),
+                    pout[c].to_numpy().astype(float), rtol=1e-12,
+                    atol=1e-9, equal_nan=True,
+                    err_msg=f"seed {seed} {by}/{tr}/{c}")
 
     # closing aggregation on whatever survived
     if len(pdf):
