@@ -187,16 +187,38 @@ class HipDataframe:
         return HipDataframe(parts, self._index, self.columns,
                             self._row_lengths, new_dtypes)
 
+    def repartition_like(self, row_lengths) -> "HipDataframe":
+        """Device re-slice of this frame's columns to the given row
+        splits (same total length) — the device form of the reference's
+        `_copartition` re-split (dataframe.py:3709), no host round trip."""
+        if list(self._row_lengths) == list(row_lengths):
+            return self
+
+        def concat_col(name):
+            cols = [p.block().columns[name] for p in self._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        cats = (self._partitions[0].block().cats
+                if self._partitions else {})
+        full = {c: concat_col(c) for c in self.columns}
+        parts, off = [], 0
+        for ln in row_lengths:
+            cols = {c: lib.col_slice(col, off, ln)
+                    for c, col in full.items()}
+            parts.append(HipDataframePartition(
+                DeviceBlock(cols, ln, cats)))
+            off += ln
+        return HipDataframe(parts, self._index, self.columns,
+                            list(row_lengths), self.dtypes)
+
     # ---- Binary zip (dataframe.py:3851) ----
     def n_ary_op(self, zip_fn, other: "HipDataframe") -> "HipDataframe":
         if self._row_lengths != other._row_lengths:
             if len(self) != len(other):
                 raise lib.HfError("n_ary_op: length mismatch")
-            # co-partition (dataframe.py:3709 _copartition): re-split the rhs
-            # to the lhs row splits via host round trip is NOT offered — the
-            # deterministic from_pandas chunking makes equal-length frames
-            # align; anything else is a later round.
-            raise lib.HfError("n_ary_op: frames are not co-partitioned")
+            # co-partition (dataframe.py:3709 _copartition): re-slice the
+            # rhs to the lhs row splits, device-side
+            other = other.repartition_like(self._row_lengths)
         parts = self._partition_mgr_cls.binary_partitions(
             self._partitions, other._partitions, zip_fn
         )
@@ -2405,8 +2427,12 @@ class HipDataframe:
 
 
 def _peek_dtypes(parts, frame) -> pandas.Series:
-    """Derive result dtypes from the first partition's (drained) block."""
+    """Derive result dtypes from the first partition's (drained) block.
+    Dictionary-encoded (string) columns hold int64 codes on device but
+    are object frames API-side."""
     block = parts[0].block()
-    return pandas.Series({name: col.np_dtype for name, col in block.columns.items()})
+    return pandas.Series({
+        name: (np.dtype(object) if name in block.cats else col.np_dtype)
+        for name, col in block.columns.items()})
 
 

@@ -213,25 +213,12 @@ class HipQueryCompiler:
         return self.__constructor__(cc.compare_scalar(lib.CMP_GE, 1.0))
 
     def drop_duplicates(self, subset=None) -> "HipQueryCompiler":
-        from modin_amd.core.dataframe import HipDataframe
-        from modin_amd.core.partition import (DeviceBlock,
-                                              HipDataframePartition)
         dup = self.duplicated(subset)
         inv = dup._modin_frame.compare_scalar(lib.CMP_EQ, 0.0)
         # the transform result is one partition; re-slice the mask to the
         # frame's partition lengths so filter_rows stays co-partitioned
         frame = self._modin_frame
-        (mcol,) = inv._partitions[0].block().columns.values()
-        parts, off = [], 0
-        for ln in frame._row_lengths:
-            parts.append(HipDataframePartition(DeviceBlock(
-                {"mask": lib.col_slice(mcol, off, ln)}, ln)))
-            off += ln
-        import pandas as _pd
-        import numpy as _np
-        mask2 = HipDataframe(parts, frame._index, ["mask"],
-                             list(frame._row_lengths),
-                             _pd.Series({"mask": _np.dtype(_np.int64)}))
+        mask2 = inv.repartition_like(frame._row_lengths)
         return self.__constructor__(frame.filter_rows(mask2))
 
     def where_mask(self, cond: "HipQueryCompiler",
