@@ -179,9 +179,10 @@ class GroupByReduce(Operator):
                 "sum/count/mean/min/max)"
             )
 
-        def caller(query_compiler, by: str, **kwargs):
+        def caller(query_compiler, by: str, dropna: bool = True,
+                   **kwargs):
             frame = query_compiler._modin_frame
-            result = frame.groupby_reduce(by, agg)
+            result = frame.groupby_reduce(by, agg, dropna=dropna)
             return query_compiler.__constructor__(result)
 
         return caller

@@ -352,11 +352,15 @@ class HipDataframe:
         return frame, decode
 
     # ---- GroupByReduce (dataframe.py:4530) ----
-    def groupby_reduce(self, by, agg: str) -> "HipDataframe":
+    def groupby_reduce(self, by, agg: str,
+                       dropna: bool = True) -> "HipDataframe":
         if isinstance(by, (list, tuple)):
             if len(by) == 1:
                 by = by[0]
             else:
+                if not dropna:
+                    raise lib.HfError("multi-key groupby(dropna=False) is "
+                                      "a later round")
                 cf, decode = self._combined_key_frame(list(by))
                 keep = [c for c in cf.columns if c not in by]
                 res = cf.take_columns(keep).groupby_reduce(self.KEYCOL, agg)
@@ -377,15 +381,21 @@ class HipDataframe:
         float_key = (parts and key_cats is None
                      and parts[0].block().columns[by].dtype_code
                      == lib.HF_FLOAT64)
+        NANCODE = 1 << 62  # dict NaN group key under dropna=False
         if float_key:
-            # pandas drops NaN keys; valid float keys ride the ordered
-            # f64->i64 bit transform (monotone, so output order == pandas)
+            # pandas drops NaN keys (dropna=True); valid float keys ride
+            # the ordered f64->i64 bit transform (monotone, so output
+            # order == pandas).  dropna=False keeps them: every NaN
+            # canonicalizes to ONE ordered key above ordered(+inf), so
+            # the NaN group lands LAST — pandas' dropna=False index order
+            # — and the inverse transform decodes it back to NaN.
             fparts = []
             for p in parts:
                 block = p.block()
                 kc = block.columns[by]
                 cols = dict(block.columns)
-                if kc.length and lib.reduce(kc).count < kc.length:
+                if dropna and kc.length and \
+                        lib.reduce(kc).count < kc.length:
                     mask = lib.compare_scalar(lib.CMP_NOTNA, kc, 0.0)
                     plan = lib.filter_plan(mask)
                     cols = {m: lib.filter_apply(plan, c)
@@ -396,22 +406,36 @@ class HipDataframe:
                     DeviceBlock(cols, cols[by].length, block.cats)))
             parts = fparts
         if key_cats is not None:
-            # pandas drops NaN groups (dropna=True): filter code == -1 rows
             has_nan = any(
                 p.block().columns[by].length
                 and lib.reduce(p.block().columns[by]).imn < 0
                 for p in parts)
             if has_nan:
+                # dropna=True: filter code == -1 rows (pandas drops NaN
+                # groups).  dropna=False: remap -1 to a key ABOVE every
+                # code so the NaN group sorts last like pandas; decoded
+                # back to -1 (= NaN) on the result keys below.
                 fparts = []
                 for p in parts:
                     block = p.block()
                     mask = lib.compare_scalar(lib.CMP_GE,
                                               block.columns[by], 0.0)
-                    plan = lib.filter_plan(mask)
-                    cols = {m: lib.filter_apply(plan, c)
-                            for m, c in block.columns.items()}
-                    fparts.append(HipDataframePartition(
-                        DeviceBlock(cols, plan.n_kept, block.cats)))
+                    if dropna:
+                        plan = lib.filter_plan(mask)
+                        cols = {m: lib.filter_apply(plan, c)
+                                for m, c in block.columns.items()}
+                        fparts.append(HipDataframePartition(
+                            DeviceBlock(cols, plan.n_kept, block.cats)))
+                    else:
+                        isna = lib.map_scalar(lib.MAP_RSUB, mask, 1)
+                        k2 = lib.binary(
+                            lib.BIN_ADD, block.columns[by],
+                            lib.map_scalar(lib.MAP_MUL, isna,
+                                           NANCODE + 1))
+                        cols = dict(block.columns)
+                        cols[by] = k2
+                        fparts.append(HipDataframePartition(
+                            DeviceBlock(cols, block.length, block.cats)))
                 parts = fparts
         keys, sums, counts, n = self._partition_mgr_cls.groupby_reduce(
             parts, by, val_names, want_counts, agg_op
@@ -464,6 +488,12 @@ class HipDataframe:
             idx = pandas.Index(lib.ordered_to_f64_np(lib.get(keys)),
                                name=by)
         else:
+            if key_cats is not None and not dropna and keys.length and \
+                    lib.reduce(keys).imx >= NANCODE:
+                m2 = lib.compare_scalar(lib.CMP_GE, keys, float(NANCODE))
+                keys = lib.binary(
+                    lib.BIN_SUB, keys,
+                    lib.map_scalar(lib.MAP_MUL, m2, NANCODE + 1))
             idx = DeviceIndex(keys, name=by, cats=key_cats)
         return HipDataframe([part], idx, val_names, [n], dtypes)
 

@@ -458,13 +458,14 @@ class DataFrame(_HipPandasBase):
         return DataFrame(query_compiler=self._query_compiler.merge(
             other._query_compiler, on=on, how=how))
 
-    def groupby(self, by, as_index: bool = True) -> "DataFrameGroupBy":
+    def groupby(self, by, as_index: bool = True,
+                dropna: bool = True) -> "DataFrameGroupBy":
         bys = list(by) if isinstance(by, (list, tuple)) else [by]
         for b in bys:
             if not isinstance(b, str) or b not in list(self.columns):
                 raise lib.HfError(
                     "groupby(by=<column name> | [column names]) only")
-        return DataFrameGroupBy(self, by, as_index=as_index)
+        return DataFrameGroupBy(self, by, as_index=as_index, dropna=dropna)
 
     def nlargest(self, n: int, columns: str):
         """pandas nlargest(keep='first'): stable descending NaN-last sort
@@ -644,11 +645,12 @@ class DataFrameGroupBy:
     (SeriesGroupBy shape for a single name)."""
 
     def __init__(self, df: DataFrame, by, as_index: bool = True,
-                 series_out: bool = False):
+                 series_out: bool = False, dropna: bool = True):
         self._df = df
         self._by = by
         self._as_index = as_index
         self._series_out = series_out
+        self._dropna = dropna
 
     def __getitem__(self, key):
         bys = list(self._by) if isinstance(self._by, (list, tuple)) \
@@ -660,10 +662,12 @@ class DataFrameGroupBy:
                                   "missing")
         sub = self._df[[*bys, *names]]
         return DataFrameGroupBy(sub, self._by, as_index=self._as_index,
-                                series_out=isinstance(key, str))
+                                series_out=isinstance(key, str),
+                                dropna=self._dropna)
 
     def _agg(self, how: str) -> DataFrame:
-        qc = self._df._query_compiler.groupby_agg(self._by, how)
+        qc = self._df._query_compiler.groupby_agg(self._by, how,
+                                                  dropna=self._dropna)
         out = DataFrame(query_compiler=qc)
         if self._series_out and self._as_index:
             name = list(qc._modin_frame.columns)[0]
@@ -689,11 +693,20 @@ class DataFrameGroupBy:
     def max(self):
         return self._agg("max")
 
+    def _need_dropna(self, what: str):
+        if not self._dropna:
+            raise lib.HfError(
+                f"groupby(dropna=False).{what} is a later round "
+                "(aggs sum/count/mean/min/max and the transform family "
+                "support dropna=False)")
+
     def var(self, ddof: int = 1):
+        self._need_dropna("var")
         return DataFrame(query_compiler=self._df._query_compiler.groupby_var(
             self._by, ddof=ddof))
 
     def std(self, ddof: int = 1):
+        self._need_dropna("std")
         return DataFrame(query_compiler=self._df._query_compiler.groupby_std(
             self._by, ddof=ddof))
 
@@ -701,6 +714,7 @@ class DataFrameGroupBy:
         return self._agg("median")
 
     def quantile(self, q: float = 0.5):
+        self._need_dropna("quantile")
         out = DataFrame(
             query_compiler=self._df._query_compiler.groupby_quantile(
                 self._by, float(q)))
@@ -712,6 +726,7 @@ class DataFrameGroupBy:
         return out
 
     def nunique(self):
+        self._need_dropna("nunique")
         qc = self._df._query_compiler.groupby_nunique(self._by)
         out = DataFrame(query_compiler=qc)
         if self._series_out and self._as_index:
@@ -731,7 +746,8 @@ class DataFrameGroupBy:
         """Same-length transforms in original row order (pandas
         DataFrameGroupBy.cumsum/cummin/cummax/cumcount/rank).  as_index is
         irrelevant (pandas keeps the caller's index for transforms)."""
-        qc = self._df._query_compiler.groupby_transform(self._by, how, **kw)
+        qc = self._df._query_compiler.groupby_transform(
+            self._by, how, dropna=self._dropna, **kw)
         if self._series_out or how in ("cumcount", "ngroup"):
             name = list(qc._modin_frame.columns)[0]
             return Series(query_compiler=qc,
@@ -770,6 +786,7 @@ class DataFrameGroupBy:
     def idxmax(self):
         """Original row label of each group's first max per column
         (all-NaN groups: NaN)."""
+        self._need_dropna("idxmax")
         qc = self._df._query_compiler.groupby_idxmax(self._by)
         out = DataFrame(query_compiler=qc)
         if self._series_out and self._as_index:
@@ -778,6 +795,7 @@ class DataFrameGroupBy:
         return out
 
     def idxmin(self):
+        self._need_dropna("idxmin")
         qc = self._df._query_compiler.groupby_idxmin(self._by)
         out = DataFrame(query_compiler=qc)
         if self._series_out and self._as_index:
@@ -800,6 +818,7 @@ class DataFrameGroupBy:
     def size(self):
         """pandas DataFrameGroupBy.size(): a Series of group row counts
         (NaN values included, NaN keys dropped)."""
+        self._need_dropna("size")
         out = DataFrame(
             query_compiler=self._df._query_compiler.groupby_size(self._by)
         ).to_pandas()

@@ -180,14 +180,15 @@ class HipQueryCompiler:
         return self.__constructor__(self._modin_frame.groupby_nunique(by))
 
     def groupby_transform(self, by, how: str, ascending: bool = True,
-                          method: str = "average",
-                          periods: int = 1) -> "HipQueryCompiler":
+                          method: str = "average", periods: int = 1,
+                          dropna: bool = True) -> "HipQueryCompiler":
         """Same-length transforms in original row order (pandas
         DataFrameGroupBy.cumsum/cummin/cummax/cumcount/rank/ngroup/
         shift/diff; reference routes these through
         modin/pandas/groupby.py -> qc groupby methods)."""
         return self.__constructor__(self._modin_frame.groupby_transform(
-            by, how, ascending=ascending, method=method, periods=periods))
+            by, how, ascending=ascending, method=method, periods=periods,
+            dropna=dropna))
 
     def groupby_idxmax(self, by) -> "HipQueryCompiler":
         return self.__constructor__(
@@ -257,7 +258,8 @@ class HipQueryCompiler:
     def distinct_stats(self):
         return self._modin_frame.distinct_stats(self._modin_frame.columns[0])
 
-    def groupby_agg(self, by: str, agg: str) -> "HipQueryCompiler":
+    def groupby_agg(self, by: str, agg: str,
+                    dropna: bool = True) -> "HipQueryCompiler":
         fn = {
             "sum": type(self).groupby_sum,
             "count": type(self).groupby_count,
@@ -274,6 +276,11 @@ class HipQueryCompiler:
             raise lib.HfError(
                 f"groupby agg {agg!r} not implemented on the HipNative backend"
             )
+        if agg in ("sum", "count", "mean", "min", "max"):
+            return fn(self, by, dropna=dropna)
+        if not dropna:
+            raise lib.HfError(f"groupby(dropna=False).{agg} is a later "
+                              "round (sum/count/mean/min/max only)")
         return fn(self, by)
 
     # ---- comparisons (query_compiler gt/lt/eq bindings) -> int64 0/1 mask

@@ -1637,3 +1637,60 @@ def test_pct_change_between_vs_pandas(npartitions):
         exp = pdf["v"].between(4.0, 6.0, inclusive=inc)
         np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy(),
                                       err_msg=inc)
+
+
+def test_groupby_dropna_false_vs_pandas(npartitions):
+    """groupby(dropna=False): NaN keys form a real group, sorted LAST
+    (pandas index order), for the reduce aggs and the transform family;
+    int value dtypes survive (NaN KEYS don't force float values)."""
+    rng = np.random.default_rng(109)
+    n = 60_000
+    kf = rng.integers(0, 50, n).astype(np.float64)
+    kf[rng.random(n) < 0.05] = np.nan
+    ks = rng.choice(["a", "b", "c"], n).astype(object)
+    ks[rng.random(n) < 0.05] = None
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.05] = np.nan
+    w = rng.integers(-40, 40, n)
+    for key in ("kf", "ks"):
+        pdf = pandas.DataFrame({"k": kf if key == "kf" else ks,
+                                "v": v, "w": w})
+        df = mpd.DataFrame(pdf)
+        for agg in ("sum", "count", "mean", "min", "max"):
+            got = getattr(df.groupby("k", dropna=False), agg)().to_pandas()
+            exp = getattr(pdf.groupby("k", dropna=False), agg)()
+            assert len(got) == len(exp), f"{key}/{agg}"
+            if key == "kf":
+                np.testing.assert_allclose(
+                    got.index.to_numpy().astype(float),
+                    exp.index.to_numpy().astype(float), rtol=0,
+                    equal_nan=True, err_msg=f"{key}/{agg}")
+            else:
+                np.testing.assert_array_equal(
+                    pandas.Series(got.index).fillna("<NA>").to_numpy(),
+                    pandas.Series(exp.index).fillna("<NA>").to_numpy(),
+                    err_msg=f"{key}/{agg}")
+            assert list(got.dtypes) == list(exp.dtypes), f"{key}/{agg}"
+            for c in exp.columns:
+                np.testing.assert_allclose(
+                    got[c].to_numpy().astype(float),
+                    exp[c].to_numpy().astype(float), rtol=1e-12,
+                    atol=1e-9, equal_nan=True, err_msg=f"{key}/{agg}/{c}")
+        # transform family under dropna=False: NaN-key rows get REAL
+        # results (they belong to the NaN group)
+        got_t = df.groupby("k", dropna=False).cumsum().to_pandas()
+        exp_t = pdf.groupby("k", dropna=False).cumsum()
+        for c in exp_t.columns:
+            np.testing.assert_allclose(
+                got_t[c].to_numpy(), exp_t[c].to_numpy(), rtol=1e-12,
+                atol=1e-9, equal_nan=True, err_msg=f"{key}/cumsum/{c}")
+        got_n = df.groupby("k", dropna=False).ngroup().to_pandas()
+        exp_n = pdf.groupby("k", dropna=False).ngroup()
+        np.testing.assert_array_equal(got_n.to_numpy(), exp_n.to_numpy(),
+                                      err_msg=f"{key}/ngroup")
+    # unsupported dropna=False paths raise loudly
+    df2 = mpd.DataFrame(pandas.DataFrame({"k": [1, 2], "v": [1.0, 2.0]}))
+    with pytest.raises(lib.HfError, match="dropna=False"):
+        df2.groupby("k", dropna=False).var()
+    with pytest.raises(lib.HfError, match="dropna=False"):
+        df2.groupby("k", dropna=False).median()
