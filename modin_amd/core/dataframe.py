@@ -1413,39 +1413,28 @@ class HipDataframe:
                 else:
                     m = lib.compare_scalar(lib.CMP_NOTNA, sv, 0.0)
                     dense = lib.map_scalar(lib.MAP_FILLNA, sv, ident)
+                def per_run_count():
+                    mm = m
+                    if mm is None:  # int column: every row counts
+                        mm = lib.alloc(n, lib.HF_INT64)
+                        lib.fill_i64(mm.dptr(), 1, n)
+                    segc = lib.seg_cumsum(mm, head, lib.AGG_SUM)
+                    return lib.gather(lib.gather(segc, ends_m1), rid)
+
                 if agg == "count":
-                    if m is None:
-                        m = lib.alloc(n, lib.HF_INT64)
-                        lib.fill_i64(m.dptr(), 1, n)
-                    seg_a = lib.seg_cumsum(m, head, lib.AGG_SUM)
-                    br = lib.gather(lib.gather(seg_a, ends_m1), rid)
+                    br = per_run_count()
                 else:
                     seg_a = lib.seg_cumsum(dense, head, op)
                     br = lib.gather(lib.gather(seg_a, ends_m1), rid)
-                    if agg == "mean" or (agg in ("min", "max")
-                                         and m is not None):
-                        segc = lib.seg_cumsum(
-                            m if m is not None else None, head,
-                            lib.AGG_SUM) if m is not None else None
-                        if agg == "mean":
-                            if segc is None:
-                                ones = lib.alloc(n, lib.HF_INT64)
-                                lib.fill_i64(ones.dptr(), 1, n)
-                                segc = lib.seg_cumsum(ones, head,
-                                                      lib.AGG_SUM)
-                            brc = lib.gather(lib.gather(segc, ends_m1),
-                                             rid)
-                            br = lib.binary(lib.BIN_DIV, lib.cast_f64(br),
-                                            lib.cast_f64(brc))
-                        else:
-                            # all-NaN group: min/max is NaN, not ±inf
-                            brc = lib.gather(lib.gather(segc, ends_m1),
-                                             rid)
-                            br = lib.fixup_empty(
-                                lib.cast_f64(br),
-                                lib.compare_scalar(lib.CMP_GE,
-                                                   lib.cast_f64(brc),
-                                                   1.0))
+                    if agg == "mean":
+                        br = lib.binary(lib.BIN_DIV, lib.cast_f64(br),
+                                        lib.cast_f64(per_run_count()))
+                    elif agg in ("min", "max") and m is not None:
+                        # all-NaN group: min/max is NaN, not ±inf
+                        br = lib.fixup_empty(
+                            lib.cast_f64(br),
+                            lib.compare_scalar(lib.CMP_GE,
+                                               per_run_count(), 1.0))
                 res = lib.scatter(br, perm)
                 if valid is not None:
                     res = lib.fixup_empty(lib.cast_f64(res), valid)
