@@ -1925,7 +1925,7 @@ class HipDataframe:
         # (the lazy-metadata pattern again; a real broadcast join caches its
         # build side)
         cache_key = (on, tuple(right_names), kmin, n_slots,
-                     uniq is not None)
+                     uniq is not None, key_f64)
         cached = getattr(other, "_join_build_cache", None)
         if key_cats is not None:
             # dictionary keys: the build lives in the LEFT frame's code
