@@ -1,12 +1,20 @@
 """modin_amd.pandas — the drop-in user API (reference L1) for the hot path.
 
 Mirrors ``modin/pandas``'s DataFrame/Series/GroupBy surface for the operators
-the HipNative backend accelerates: construction/from_pandas,
-elementwise arithmetic (+ - * /), fillna/abs, sum/mean/count/min/max, column
-projection and ``groupby(key).sum()/count()/mean()/agg(...)``
+the HipNative backend accelerates (SURVEY.md §8a + §8f, see README/DESIGN
+for the full coverage list): construction/from_pandas/read_parquet,
+elementwise arithmetic and comparisons, map ops (fillna/abs/round/clip/
+where/mask), reductions (sum/mean/count/min/max/var/std/median/quantile/
+idxmax/idxmin, axis 0 and 1), sort_values (multi-key, na_position),
+filter/dropna/duplicated/drop_duplicates/nlargest, merge
+(inner/left/right/outer/cross, int64/float64/string keys incl. NaN
+matching), concat, and the groupby family: the reduce aggs + agg forms
 (modin/pandas/dataframe.py:2188 sum; modin/pandas/groupby.py:1330
-DataFrameGroupBy.sum -> _wrap_aggregation).  Everything else the reference's
-API offers is out of scope per SURVEY.md §8 and raises loudly.
+DataFrameGroupBy.sum -> _wrap_aggregation), size/nunique/first/last/
+median/quantile/idxmax/idxmin, dropna=False, and the same-length
+transforms (cum*/cumcount/ngroup/rank/shift/diff/transform(agg)).
+Anything beyond this surface raises loudly (no silent pandas fallback —
+DESIGN.md "No CPU fallback").
 
 Reductions return pandas.Series (the reference's API layer also lowers
 1×N reduce results into Series via ``_reduce_dimension``).
