@@ -139,3 +139,54 @@ def test_parquet_column_conversion():
     v3, _ = _column_to_device_ready(
         pa.chunked_array([pa.array([7, 8], type=pa.int32())]), "j")
     assert v3.dtype == np.int64 and v3.tolist() == [7, 8]
+
+
+def test_ordered_inverse_host_roundtrip():
+    """ordered_to_f64_np (the host inverse of the device f64->i64 ordered
+    bit transform) inverts a numpy restatement of the forward rule on
+    every special value — the same rule k_f64_ordered implements
+    (negatives bit-reversed below zero, -0.0 -> +0.0, every NaN
+    canonicalized to one key)."""
+    import numpy as np
+    from modin_amd.core.lib import ordered_to_f64_np
+
+    def forward(x):
+        x = np.asarray(x, dtype=np.float64) + 0.0  # -0.0 -> +0.0
+        v = x.view(np.int64).copy()
+        v[np.isnan(x)] = 0x7FF8000000000000
+        neg = v < 0
+        v[neg] = ~v[neg] ^ np.int64(-2**63)
+        return v
+
+    xs = np.array([0.0, -0.0, 1.5, -1.5, np.inf, -np.inf, np.nan,
+                   5e-324, -5e-324, 1e308, -1e308, 2.0, -2.0])
+    back = ordered_to_f64_np(forward(xs))
+    np.testing.assert_array_equal(back, np.where(xs == -0.0, 0.0, xs))
+    # the forward rule is order-preserving on ordinary values
+    rng = np.random.default_rng(7)
+    r = rng.standard_normal(1000) * 1e6
+    f = forward(np.sort(r))
+    # element compare, not np.diff: the key span is nearly the full int64
+    # range, so a sign-crossing difference overflows int64
+    assert (f[1:] > f[:-1]).all()
+    # all NaNs (any payload/sign) map to ONE key above ordered(+inf)
+    nans = np.array([np.nan, -np.nan, np.float64("nan")])
+    fk = forward(nans)
+    assert len(set(fk.tolist())) == 1 and fk[0] > forward([np.inf])[0]
+
+
+def test_decode_encode_dict_roundtrip():
+    """Host dictionary encode/decode round-trips values incl. NaN."""
+    import numpy as np
+    import pandas
+    from modin_amd.core.partition import decode_dict, encode_dict
+    s = pandas.Series(["b", None, "a", "b", np.nan, "cc"])
+    codes, cats = encode_dict(s)
+    assert list(cats) == ["a", "b", "cc"]  # sorted-cats invariant
+    assert codes.dtype == np.int64 and codes[1] == -1 and codes[4] == -1
+    back = decode_dict(np.asarray(codes), cats)
+    for got, exp in zip(back, s):
+        if isinstance(exp, str):
+            assert got == exp
+        else:
+            assert isinstance(got, float) and np.isnan(got)
