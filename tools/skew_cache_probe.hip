@@ -13,10 +13,11 @@
 // Variants (all produce the dense (sums, rowcnt) tables; checked against
 // a host reference):
 //   v1  global-atomic accumulate (production k_gb_accum shape; baseline)
-//   v2  LDS direct-mapped cache (key & (C-1)), evict-on-conflict to the
-//       global table, flush at block end
-//   v3  v2 with 2-way associativity (evict the slot with the smaller
-//       rowcnt — cheap LFU approximation)
+//   v2  LDS direct-mapped cache (key & (C-1)); occupied-slot conflicts
+//       BYPASS to the global table (no eviction — tags are write-once, so
+//       the kernel stays race-free without slot locks); flush at the end.
+// If v2 pays, follow-ups to measure live: 2-way associativity with an
+// LFU-ish evict, and wiring the cache into gb_scatter's front end.
 // Run on 1 GPU:  ./skew_cache_probe [rows] [keys]
 // Build: hipcc --offload-arch=gfx950 -O3 tools/skew_cache_probe.hip \
 //        -o tools/skew_cache_probe
