@@ -190,3 +190,35 @@ def test_oracle_vs_pandas_property(seed):
         exp = getattr(pandas.Series(vals), op)()
         np.testing.assert_allclose(float(got), float(exp), rtol=1e-12,
                                    err_msg=op)
+
+
+@pytest.mark.parametrize("seed", range(4))
+def test_oracle_join_filter_vs_pandas_property(seed):
+    """Oracle inner join + filter vs pandas on randomized inputs."""
+    import pandas
+    rng = np.random.default_rng(700 + seed)
+    nl, nr = int(rng.integers(500, 8000)), int(rng.integers(200, 3000))
+    lk = rng.integers(0, 400, nl).astype(np.int64)
+    rk = rng.integers(0, 400, nr).astype(np.int64)
+    lv = rng.standard_normal(nl)
+    rv = rng.standard_normal(nr)
+    ok, _lidx, olv, orv = oracle.inner_join(lk, {"a": lv}, rk, {"b": rv})
+    pl = pandas.DataFrame({"k": lk, "a": lv})
+    pr = pandas.DataFrame({"k": rk, "b": rv})
+    exp = pl.merge(pr, on="k")
+    assert len(ok) == len(exp)
+    np.testing.assert_allclose(np.sort(np.asarray(olv["a"])),
+                               np.sort(exp["a"].to_numpy()), rtol=1e-15)
+    np.testing.assert_allclose(np.sort(np.asarray(orv["b"])),
+                               np.sort(exp["b"].to_numpy()), rtol=1e-15)
+    # filter mask semantics (NaN compares False except NE)
+    v = rng.standard_normal(2000)
+    v[rng.random(2000) < 0.2] = np.nan
+    thr = float(rng.standard_normal())
+    mask = oracle.compare_op("gt", v, thr)
+    np.testing.assert_array_equal(
+        mask.astype(bool), pandas.Series(v).gt(thr).to_numpy())
+    kept = oracle.filter_rows(mask, {"v": v})["v"]
+    np.testing.assert_allclose(
+        np.asarray(kept), v[pandas.Series(v).gt(thr).to_numpy()],
+        rtol=0, equal_nan=True)
