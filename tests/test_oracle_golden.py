@@ -218,7 +218,8 @@ def test_oracle_join_filter_vs_pandas_property(seed):
     mask = oracle.compare_op("gt", v, thr)
     np.testing.assert_array_equal(
         mask.astype(bool), pandas.Series(v).gt(thr).to_numpy())
-    kept = oracle.filter_rows(mask, {"v": v})["v"]
+    _pos, fcols = oracle.filter_rows(mask, {"v": v})
+    kept = fcols["v"]
     np.testing.assert_allclose(
         np.asarray(kept), v[pandas.Series(v).gt(thr).to_numpy()],
         rtol=0, equal_nan=True)
