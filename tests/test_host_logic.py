@@ -190,3 +190,59 @@ def test_decode_encode_dict_roundtrip():
             assert got == exp
         else:
             assert isinstance(got, float) and np.isnan(got)
+
+
+def test_ffill_bfill_composition_prototype():
+    """Round-2 de-risk (DESIGN roadmap 6): the groupby ffill/bfill device
+    composition — stable key sort, run heads, segmented MAX over
+    (valid ? position : −1), gather, inverse scatter, NaN-key fixup —
+    restated in numpy and pinned against pandas groupby.ffill/bfill.
+    Every step below maps 1:1 onto an existing hf_* primitive."""
+    import numpy as np
+    import pandas
+
+    def seg_ffill(keys, vals, reverse=False):
+        n = len(keys)
+        valid_key = ~np.isnan(keys)
+        # effective key: canonical-NaN sentinel sorts last (ordered_i64)
+        eff = np.where(valid_key, keys, np.inf)
+        if reverse:
+            # bfill = ffill over reversed rows (hf_gather by reversed iota)
+            rev = np.arange(n - 1, 0 - 1, -1)
+            out = seg_ffill(keys[rev], vals[rev], False)
+            return out[rev]
+        perm = np.argsort(eff, kind="stable")      # hf_sort_perm
+        sk, sv = eff[perm], vals[perm]
+        head = np.empty(n, dtype=np.int64)         # run_head compose
+        head[0] = 1
+        head[1:] = (sk[1:] != sk[:-1]).astype(np.int64)
+        rid = np.cumsum(head) - 1                  # hf_cumsum (i64)
+        posv = np.where(~np.isnan(sv), np.arange(n), -1)  # compare+mul
+        # segmented max scan == hf_seg_cumsum(posv, head, AGG_MAX)
+        segmax = np.empty(n, dtype=np.int64)
+        cur = -1
+        for i in range(n):                         # (device: 3-phase scan)
+            if head[i]:
+                cur = -1
+            cur = max(cur, posv[i])
+            segmax[i] = cur
+        filled = np.where(segmax >= 0, sv[np.clip(segmax, 0, n - 1)],
+                          np.nan)                  # hf_gather + fixup
+        out = np.empty(n)
+        out[perm] = filled                         # hf_scatter
+        out[~valid_key] = np.nan                   # valid-key fixup
+        return out
+
+    rng = np.random.default_rng(11)
+    n = 4000
+    keys = rng.integers(0, 40, n).astype(np.float64)
+    keys[rng.random(n) < 0.05] = np.nan
+    vals = rng.standard_normal(n)
+    vals[rng.random(n) < 0.3] = np.nan
+    pdf = pandas.DataFrame({"k": keys, "v": vals})
+    exp_f = pdf.groupby("k")["v"].ffill().to_numpy()
+    got_f = seg_ffill(keys, vals)
+    np.testing.assert_allclose(got_f, exp_f, rtol=0, equal_nan=True)
+    exp_b = pdf.groupby("k")["v"].bfill().to_numpy()
+    got_b = seg_ffill(keys, vals, reverse=True)
+    np.testing.assert_allclose(got_b, exp_b, rtol=0, equal_nan=True)
