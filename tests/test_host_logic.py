@@ -246,3 +246,50 @@ def test_ffill_bfill_composition_prototype():
     exp_b = pdf.groupby("k")["v"].bfill().to_numpy()
     got_b = seg_ffill(keys, vals, reverse=True)
     np.testing.assert_allclose(got_b, exp_b, rtol=0, equal_nan=True)
+
+
+def test_coshuffled_merge_composition_prototype():
+    """Round-2 de-risk (DESIGN roadmap 2): the co-shuffled merge for
+    giant right tables — both sides range-binned by the same splitters
+    (hf_shuffle_dest), per-bin dense CSR join, and pandas' left-major
+    match order restored by a stable sort on the global left row id.
+    Restated in numpy and pinned against pandas.merge; every step maps
+    onto an existing primitive (shuffle_dest / filter / join_probe /
+    sort_perm / gather)."""
+    import numpy as np
+    import pandas
+
+    rng = np.random.default_rng(13)
+    nl, nr = 30_000, 8_000
+    lk = rng.integers(-10**12, 10**12, nl) // 10**7  # clustered keys
+    rk = rng.integers(-10**12, 10**12, nr) // 10**7
+    lv = rng.standard_normal(nl)
+    rv = rng.standard_normal(nr)
+    P = 8
+    spl = oracle.pick_splitters(np.sort(rk)[:: max(nr // 512, 1)], P)
+    ld = oracle.shuffle_dest(lk, spl)
+    rd = oracle.shuffle_dest(rk, spl)
+    out_keys, out_lidx, out_a, out_b = [], [], [], []
+    for p in range(P):                       # per-bin local CSR join
+        lsel = np.nonzero(ld == p)[0]
+        rsel = np.nonzero(rd == p)[0]
+        k, lidx, la, rb = oracle.inner_join(
+            lk[lsel], {"a": lv[lsel]}, rk[rsel], {"b": rv[rsel]})
+        out_keys.append(k)
+        out_lidx.append(lsel[lidx])          # back to GLOBAL left rows
+        out_a.append(la["a"])
+        out_b.append(rb["b"])
+    keys = np.concatenate(out_keys)
+    glidx = np.concatenate(out_lidx)
+    a = np.concatenate(out_a)
+    b = np.concatenate(out_b)
+    # pandas order: left-row-major, right matches in right-row order —
+    # a STABLE sort by global left id restores it because each bin's
+    # per-left matches are already right-row ordered
+    order = np.argsort(glidx, kind="stable")
+    keys, a, b = keys[order], a[order], b[order]
+    exp = pandas.DataFrame({"k": lk, "a": lv}).merge(
+        pandas.DataFrame({"k": rk, "b": rv}), on="k")
+    np.testing.assert_array_equal(keys, exp["k"].to_numpy())
+    np.testing.assert_allclose(a, exp["a"].to_numpy(), rtol=0)
+    np.testing.assert_allclose(b, exp["b"].to_numpy(), rtol=0)
