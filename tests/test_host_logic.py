@@ -293,3 +293,72 @@ def test_coshuffled_merge_composition_prototype():
     np.testing.assert_array_equal(keys, exp["k"].to_numpy())
     np.testing.assert_allclose(a, exp["a"].to_numpy(), rtol=0)
     np.testing.assert_allclose(b, exp["b"].to_numpy(), rtol=0)
+
+
+def test_distributed_median_composition_prototype():
+    """Round-2 de-risk: distributed median/quantile rides the existing
+    range shuffle — after shuffling VALUES (ordered-transformed) by
+    sampled splitters, rank r holds a contiguous value range; the global
+    middle positions land on one rank, found by an exclusive prefix of
+    per-rank non-NaN counts (one tiny allgather).  numpy restatement vs
+    numpy median."""
+    import numpy as np
+
+    rng = np.random.default_rng(17)
+    n = 50_001
+    v = rng.standard_normal(n) * 100
+    v[rng.random(n) < 0.1] = np.nan
+    P = 4
+    vv = v[~np.isnan(v)]                    # NOTNA filter per rank
+    spl = oracle.pick_splitters(
+        np.sort(vv)[:: max(len(vv) // 512, 1)].astype(np.float64), P)
+    # the real path bins the ordered-i64 form via hf_shuffle_dest; float
+    # compares are equivalent for finite values here
+    dest = np.searchsorted(spl, vv, side="right")
+    shards = [np.sort(vv[dest == p], kind="stable") for p in range(P)]
+    counts = np.array([len(s) for s in shards])
+    total = counts.sum()
+    mids = [(total - 1) // 2, total // 2]   # lower/upper middle
+    got = []
+    base = np.concatenate([[0], np.cumsum(counts)])
+    for m in mids:
+        r = int(np.searchsorted(base, m, side="right")) - 1
+        got.append(shards[r][m - base[r]])
+    np.testing.assert_allclose(np.mean(got), np.nanmedian(v), rtol=1e-15)
+
+
+def test_multikey_idx_nan_keys_prototype():
+    """Round-2 de-risk: NaN keys in MULTI-key idxmax — filter invalid
+    rows first (keeping original positions via filter_iota), run the
+    existing composition on the filtered frame, and report the kept
+    positions; pinned vs pandas."""
+    import numpy as np
+    import pandas
+
+    rng = np.random.default_rng(19)
+    n = 5000
+    a = rng.integers(0, 6, n).astype(np.float64)
+    a[rng.random(n) < 0.05] = np.nan
+    b = rng.integers(0, 4, n).astype(np.float64)
+    b[rng.random(n) < 0.05] = np.nan
+    v = rng.integers(-5, 5, n).astype(np.float64)  # heavy ties
+    pdf = pandas.DataFrame({"a": a, "b": b, "v": v})
+    keep = ~(np.isnan(a) | np.isnan(b))            # filter_plan
+    pos = np.nonzero(keep)[0]                      # filter_iota
+    fa, fb, fv = a[keep], b[keep], v[keep]
+    # existing composition on the filtered rows: stable sort by
+    # (a, b, v), key-run ends via counts, tie-block start = first max
+    perm = np.lexsort((fv, fb, fa))                # device: 3 stable passes
+    sa, sb, sv = fa[perm], fb[perm], fv[perm]
+    head = np.ones(len(sa), dtype=bool)
+    head[1:] = (sa[1:] != sa[:-1]) | (sb[1:] != sb[:-1])
+    starts = np.nonzero(head)[0]
+    ends = np.append(starts[1:], len(sa))
+    got = {}
+    for s, e in zip(starts, ends):
+        mx = sv[e - 1]                             # NaN-free values here
+        ts = s + int(np.searchsorted(sv[s:e], mx, side="left"))
+        got[(sa[s], sb[s])] = pos[perm[ts]]
+    exp = pdf.groupby(["a", "b"]).idxmax()
+    for (ka, kb), row in zip(exp.index, exp["v"]):
+        assert got[(ka, kb)] == row, (ka, kb)
