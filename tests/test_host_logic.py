@@ -390,3 +390,36 @@ def test_string_minmax_first_composition_prototype():
         exp = getattr(pdf.groupby("k")["s"], agg)()
         np.testing.assert_array_equal(gk, exp.index.to_numpy())
         np.testing.assert_array_equal(got, exp.to_numpy())
+
+
+def test_unbounded_multikey_groupby_prototype():
+    """Round-2 de-risk: multi-key groupby BEYOND the 2^62 combined-span
+    fold needs no tuple hash — per-column stable LSD sort (any spans),
+    OR'd per-column run heads delimit the tuple groups exactly, and the
+    sorted/segmented aggregation path consumes the head flags.  numpy
+    restatement vs pandas; collision-free by construction."""
+    import numpy as np
+    import pandas
+
+    rng = np.random.default_rng(29)
+    n = 20_000
+    a = rng.integers(-2**62, 2**62, n)          # full-span keys
+    a = a - (a % 10**15)                        # ~9e3 distinct
+    b = rng.integers(-2**62, 2**62, n)
+    b = b - (b % (3 * 10**17))
+    v = rng.standard_normal(n)
+    perm = np.lexsort((b, a))                   # device: stable LSD passes
+    sa, sb, sv = a[perm], b[perm], v[perm]
+    head = np.ones(n, dtype=bool)
+    head[1:] = (sa[1:] != sa[:-1]) | (sb[1:] != sb[:-1])
+    starts = np.nonzero(head)[0]
+    ends = np.append(starts[1:], n)
+    sums = np.add.reduceat(sv, starts)
+    exp = pandas.DataFrame({"a": a, "b": b, "v": v}).groupby(
+        ["a", "b"])["v"].sum()
+    np.testing.assert_array_equal(sa[starts],
+                                  exp.index.get_level_values(0).to_numpy())
+    np.testing.assert_array_equal(sb[starts],
+                                  exp.index.get_level_values(1).to_numpy())
+    np.testing.assert_allclose(sums, exp.to_numpy(), rtol=1e-12)
+    assert len(starts) == len(exp) and ends[-1] == n
