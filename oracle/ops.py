@@ -263,17 +263,32 @@ def partitioned_groupby_agg(keys: np.ndarray, vals: dict, agg: str,
     change the result beyond fp reassociation."""
     counts = split_row_counts(len(keys), num_splits, min_size)
     offs = np.cumsum([0] + counts)
-    # map phase: per-partition partials (sum & count cover all three aggs)
+    # map phase: per-partition partials.  sum & count cover sum/count/mean
+    # (GroupbyReduceImpl pairs, storage_formats/pandas/groupby.py:237-248:
+    # mean = map concat(sum,count) / reduce divide); min/max reduce with
+    # themselves ({min: ("min","min")}) over NaN-skipping partials.
     partial = {}
     for i in range(len(counts)):
         sl = slice(offs[i], offs[i + 1])
-        uk, psums = groupby_agg(keys[sl], {n: v[sl] for n, v in vals.items()}, "sum")
-        _, pcnts = groupby_agg(keys[sl], {n: v[sl] for n, v in vals.items()}, "count")
+        sub = {n: v[sl] for n, v in vals.items()}
+        uk, psums = groupby_agg(keys[sl], sub, "sum")
+        _, pcnts = groupby_agg(keys[sl], sub, "count")
+        if agg in ("min", "max"):
+            _, pext = groupby_agg(keys[sl], sub, agg)
         for j, k in enumerate(uk):
-            acc = partial.setdefault(int(k), {n: [0.0, 0] for n in vals})
+            acc = partial.setdefault(
+                int(k), {n: [0.0, 0, float("nan")] for n in vals})
             for n in vals:
                 acc[n][0] += psums[n][j]
                 acc[n][1] += pcnts[n][j]
+                if agg in ("min", "max"):
+                    e = pext[n][j]
+                    cur = acc[n][2]
+                    if np.isnan(cur):
+                        acc[n][2] = e
+                    elif not np.isnan(e):
+                        acc[n][2] = min(cur, e) if agg == "min" \
+                            else max(cur, e)
     out_keys = np.array(sorted(partial.keys()), dtype=np.int64)
     out = {}
     for n in vals:
@@ -282,6 +297,8 @@ def partitioned_groupby_agg(keys: np.ndarray, vals: dict, agg: str,
         elif agg == "count":
             out[n] = np.array([partial[int(k)][n][1] for k in out_keys],
                               dtype=np.int64)
+        elif agg in ("min", "max"):
+            out[n] = np.array([partial[int(k)][n][2] for k in out_keys])
         else:
             out[n] = np.array([
                 partial[int(k)][n][0] / partial[int(k)][n][1]

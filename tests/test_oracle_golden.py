@@ -223,3 +223,50 @@ def test_oracle_join_filter_vs_pandas_property(seed):
     np.testing.assert_allclose(
         np.asarray(kept), v[pandas.Series(v).gt(thr).to_numpy()],
         rtol=0, equal_nan=True)
+
+
+@pytest.mark.parametrize("seed", range(10))
+def test_oracle_fuzz_vs_pandas(seed):
+    """Randomized oracle fuzz: groupby (all aggs, random key spans and
+    NaN densities, incl. single-group and all-NaN-column cases),
+    partitioned==global, sort asc/desc ties, joins with empty sides.
+    The bench's pre-timing parity gate trusts these functions."""
+    import pandas
+    rng = np.random.default_rng(900 + seed)
+    n = int(rng.integers(100, 30_000))
+    span = int(rng.integers(1, 1000))
+    keys = rng.integers(-span, span + 1, n).astype(np.int64)
+    vals = rng.standard_normal(n)
+    nan_frac = float(rng.random()) * 0.9
+    vals[rng.random(n) < nan_frac] = np.nan
+    pdf = pandas.DataFrame({"k": keys, "v": vals})
+    for agg in ("sum", "count", "mean", "min", "max"):
+        gk, gv = oracle.groupby_agg(keys, {"v": vals}, agg)
+        exp = getattr(pdf.groupby("k")["v"], agg)()
+        np.testing.assert_array_equal(gk, exp.index.to_numpy(),
+                                      err_msg=f"{seed}/{agg}")
+        np.testing.assert_allclose(
+            np.asarray(gv["v"], dtype=np.float64),
+            exp.to_numpy().astype(np.float64), rtol=1e-12, atol=1e-12,
+            equal_nan=True, err_msg=f"{seed}/{agg}")
+        # partitioned map-reduce == global (the multi-partition bench path)
+        splits = int(rng.integers(1, 9))
+        pk, pv = oracle.partitioned_groupby_agg(keys, {"v": vals}, agg,
+                                                splits)
+        np.testing.assert_array_equal(pk, gk)
+        np.testing.assert_allclose(np.asarray(pv["v"], dtype=np.float64),
+                                   np.asarray(gv["v"], dtype=np.float64),
+                                   rtol=1e-12, atol=1e-12, equal_nan=True,
+                                   err_msg=f"{seed}/{agg}/part{splits}")
+    for asc in (True, False):
+        perm = oracle.sort_perm(keys, ascending=asc)
+        exp_idx = pdf.sort_values("k", ascending=asc,
+                                  kind="stable").index.to_numpy()
+        np.testing.assert_array_equal(perm, exp_idx, err_msg=str(asc))
+    # join with a possibly-empty right
+    nr = int(rng.integers(0, 500))
+    rk = rng.integers(-span, span + 1, nr).astype(np.int64)
+    rv = rng.standard_normal(nr)
+    jk, _jl, _ja, _jb = oracle.inner_join(keys, {"a": vals}, rk, {"b": rv})
+    exp_len = len(pdf.merge(pandas.DataFrame({"k": rk, "b": rv}), on="k"))
+    assert len(jk) == exp_len
