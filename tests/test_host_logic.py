@@ -423,3 +423,62 @@ def test_unbounded_multikey_groupby_prototype():
                                   exp.index.get_level_values(1).to_numpy())
     np.testing.assert_allclose(sums, exp.to_numpy(), rtol=1e-12)
     assert len(starts) == len(exp) and ends[-1] == n
+
+
+def test_rolling_composition_prototype():
+    """Round-2 design pin: pandas rolling(w) without new kernel families.
+    sum/mean: windowed difference of NaN-zero-filled prefix sums + a
+    windowed non-NaN count against min_periods (pandas default
+    min_periods=w -> any NaN in the window yields NaN).  min/max: the
+    van Herk/Gil-Werman two-scan trick — tile the rows at width w, take
+    suffix-max within each tile (a segmented scan with heads at tile
+    starts, exactly hf_seg_cumsum AGG_MAX) and prefix-max (the existing
+    hf_cumsum-per-tile shape); window max at i = comb(suffix[i-w+1],
+    prefix[i]).  Pinned vs pandas."""
+    import numpy as np
+    import pandas
+
+    rng = np.random.default_rng(31)
+    n = 5000
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.1] = np.nan
+    s = pandas.Series(v)
+    for w in (1, 3, 16, 100):
+        zf = np.where(np.isnan(v), 0.0, v)
+        cs = np.concatenate([[0.0], np.cumsum(zf)])
+        cc = np.concatenate([[0], np.cumsum(~np.isnan(v))])
+        wsum = cs[w:] - cs[:-w]                # rows w-1..n-1
+        wcnt = cc[w:] - cc[:-w]
+        out = np.full(n, np.nan)
+        ok = wcnt == w                         # min_periods = w
+        out[w - 1:][ok] = wsum[ok]
+        np.testing.assert_allclose(out, s.rolling(w).sum().to_numpy(),
+                                   rtol=1e-12, atol=1e-12, equal_nan=True,
+                                   err_msg=f"sum/w={w}")
+        outm = np.full(n, np.nan)
+        outm[w - 1:][ok] = wsum[ok] / w
+        np.testing.assert_allclose(outm, s.rolling(w).mean().to_numpy(),
+                                   rtol=1e-12, atol=1e-12, equal_nan=True,
+                                   err_msg=f"mean/w={w}")
+        # van Herk max: NaN rows poison their windows under min_periods=w,
+        # so compute over zero... no — compute over v with NaN->-inf and
+        # mask windows containing NaN afterwards
+        vv = np.where(np.isnan(v), -np.inf, v)
+        ntiles = -(-n // w)
+        pad = ntiles * w
+        vp = np.full(pad, -np.inf)
+        vp[:n] = vv
+        tiles = vp.reshape(ntiles, w)
+        pref = np.maximum.accumulate(tiles, axis=1).reshape(pad)
+        suff = np.maximum.accumulate(tiles[:, ::-1], axis=1)[:, ::-1] \
+            .reshape(pad)
+        wmax = np.full(n, np.nan)
+        for i in range(w - 1, n):
+            lo = i - w + 1
+            wmax[i] = max(suff[lo], pref[i]) if (lo // w) != (i // w) \
+                else pref[i]
+        res = np.full(n, np.nan)
+        res[w - 1:][ok] = wmax[w - 1:][ok]
+        np.testing.assert_allclose(res, s.rolling(w).max().to_numpy(),
+                                   rtol=0, equal_nan=True,
+                                   err_msg=f"max/w={w}")
