@@ -852,7 +852,8 @@ class DataFrameGroupBy:
                 bys = (list(self._by) if isinstance(self._by, (list, tuple))
                        else [self._by])
                 sub = self._df[[*bys, col]]
-                qcs.append(sub._query_compiler.groupby_agg(self._by, a))
+                qcs.append(sub._query_compiler.groupby_agg(
+                    self._by, a, dropna=self._dropna))
             return DataFrame(query_compiler=qcs[0].hconcat(qcs[1:]))
         if isinstance(how, (list, tuple)):
             bys = (list(self._by) if isinstance(self._by, (list, tuple))
@@ -865,7 +866,8 @@ class DataFrameGroupBy:
                         raise lib.HfError("groupby.agg list entries must "
                                           "be agg names")
                     sub = self._df[[*bys, col]]
-                    qc = sub._query_compiler.groupby_agg(self._by, a)
+                    qc = sub._query_compiler.groupby_agg(
+                        self._by, a, dropna=self._dropna)
                     qcs.append(qc.rename_columns({col: (col, a)}))
             return DataFrame(query_compiler=qcs[0].hconcat(qcs[1:]))
         raise lib.HfError("groupby.agg accepts str / list / dict")
