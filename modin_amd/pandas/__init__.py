@@ -354,13 +354,16 @@ class DataFrame(_HipPandasBase):
         raise lib.HfError("only column selection / boolean masks are supported")
 
     def head(self, n: int = 5):
+        total = len(self._query_compiler)
+        stop = max(0, total + n) if n < 0 else n  # pandas head(-n)
         return DataFrame(
-            query_compiler=self._query_compiler.take_row_range(0, n))
+            query_compiler=self._query_compiler.take_row_range(0, stop))
 
     def tail(self, n: int = 5):
         total = len(self._query_compiler)
+        start = min(total, -n) if n < 0 else total - n  # pandas tail(-n)
         return DataFrame(
-            query_compiler=self._query_compiler.take_row_range(total - n,
+            query_compiler=self._query_compiler.take_row_range(start,
                                                                total))
 
     def astype(self, dtype):
@@ -556,8 +559,10 @@ class Series(_HipPandasBase):
         return Series(query_compiler=qc, name=self.name)
 
     def head(self, n: int = 5):
+        total = len(self._query_compiler)
+        stop = max(0, total + n) if n < 0 else n  # pandas head(-n)
         return Series(
-            query_compiler=self._query_compiler.take_row_range(0, n),
+            query_compiler=self._query_compiler.take_row_range(0, stop),
             name=self.name)
 
     def nlargest(self, n: int = 5):

@@ -657,6 +657,10 @@ def test_head_tail_astype_rename_reset(npartitions):
     pdf = pandas.DataFrame(data)
     pandas.testing.assert_frame_equal(df.head(7).to_pandas(), pdf.head(7))
     pandas.testing.assert_frame_equal(df.tail(9).to_pandas(), pdf.tail(9))
+    pandas.testing.assert_frame_equal(df.head(-3).to_pandas(),
+                                      pdf.head(-3))
+    pandas.testing.assert_frame_equal(df.tail(-3).to_pandas(),
+                                      pdf.tail(-3))
     pandas.testing.assert_frame_equal(df.astype("float64").to_pandas(),
                                       pdf.astype("float64"))
     pandas.testing.assert_frame_equal(df.astype("int64").to_pandas(),
