@@ -1,0 +1,131 @@
+"""CPU tier: the Python COMPOSITION layer over the numpy lib mock.
+
+These run the REAL modin_amd/core/dataframe.py composition code (the
+sequences of gather/scatter/scan/filter calls behind groupby transforms,
+rank, where/mask/round, sort and dedup) against pandas, with
+tests/mocklib.py standing in for the HIP kernels.  Kernel parity stays on
+the GPU tier; this tier catches composition regressions (metadata,
+ordering, NaN bookkeeping) before any gpurun call is spent.  Paths whose
+kernels are NOT mocked (groupby aggregations, merge) are exercised on the
+GPU tier only.
+"""
+
+import numpy as np
+import pandas
+import pytest
+
+from tests import mocklib
+
+
+@pytest.fixture()
+def mlib(monkeypatch):
+    mocklib.install(monkeypatch)
+    import modin_amd.pandas as mpd
+    yield mpd
+
+
+def _frames(rng, n=4000, nan_keys=True):
+    k = rng.integers(0, 40, n).astype(np.float64)
+    if nan_keys:
+        k[rng.random(n) < 0.05] = np.nan
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.1] = np.nan
+    w = rng.integers(-30, 30, n)
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w})
+    return pdf
+
+
+def test_mock_groupby_transforms(mlib):
+    rng = np.random.default_rng(41)
+    pdf = _frames(rng)
+    df = mlib.DataFrame(pdf)
+    for how in ("cumsum", "cummin", "cummax"):
+        got = getattr(df.groupby("k"), how)().to_pandas()
+        exp = getattr(pdf.groupby("k"), how)()
+        for c in exp.columns:
+            np.testing.assert_allclose(got[c].to_numpy(),
+                                       exp[c].to_numpy(), rtol=1e-12,
+                                       atol=1e-9, equal_nan=True,
+                                       err_msg=f"{how}/{c}")
+    np.testing.assert_allclose(
+        df.groupby("k").cumcount().to_pandas().to_numpy(),
+        pdf.groupby("k").cumcount().to_numpy(), rtol=0, equal_nan=True)
+    np.testing.assert_allclose(
+        df.groupby("k").ngroup().to_pandas().to_numpy(),
+        pdf.groupby("k").ngroup().to_numpy().astype(float), rtol=0,
+        equal_nan=True)
+    for p in (1, -2):
+        got = df.groupby("k").shift(p).to_pandas()
+        exp = pdf.groupby("k").shift(p)
+        for c in exp.columns:
+            np.testing.assert_allclose(got[c].to_numpy(),
+                                       exp[c].to_numpy(), rtol=0,
+                                       equal_nan=True,
+                                       err_msg=f"shift({p})/{c}")
+    for agg in ("sum", "mean", "count", "min", "max"):
+        got = df.groupby("k").transform(agg).to_pandas()
+        exp = pdf.groupby("k").transform(agg)
+        for c in exp.columns:
+            np.testing.assert_allclose(
+                got[c].to_numpy().astype(float),
+                exp[c].to_numpy().astype(float), rtol=1e-12, atol=1e-9,
+                equal_nan=True, err_msg=f"t-{agg}/{c}")
+
+
+def test_mock_rank_and_frame_rank(mlib):
+    rng = np.random.default_rng(43)
+    pdf = _frames(rng, 3000)
+    df = mlib.DataFrame(pdf)
+    for method in ("average", "min", "first"):
+        for asc in (True, False):
+            got = df.groupby("k").rank(method=method,
+                                       ascending=asc).to_pandas()
+            exp = pdf.groupby("k").rank(method=method, ascending=asc)
+            for c in exp.columns:
+                np.testing.assert_allclose(
+                    got[c].to_numpy(), exp[c].to_numpy(), rtol=0,
+                    equal_nan=True, err_msg=f"{method}/{asc}/{c}")
+    got = df.rank().to_pandas()
+    exp = pdf.rank()
+    for c in ("v", "w"):
+        np.testing.assert_allclose(got[c].to_numpy(), exp[c].to_numpy(),
+                                   rtol=0, equal_nan=True, err_msg=c)
+
+
+def test_mock_where_round_sort_dedup(mlib):
+    rng = np.random.default_rng(47)
+    pdf = _frames(rng, 3000, nan_keys=False)
+    df = mlib.DataFrame(pdf)
+    cond_p = pdf["v"] > 0
+    got = df.where(df["v"] > 0, -2.5).to_pandas()
+    exp = pdf.where(cond_p, -2.5)
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(), rtol=0,
+                               equal_nan=True)
+    got = df.mask(df["v"] > 0).to_pandas()
+    exp = pdf.mask(cond_p)
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(), rtol=0,
+                               equal_nan=True)
+    for d in (0, 2, -1):
+        got = df.round(d).to_pandas()
+        exp = pdf.round(d)
+        np.testing.assert_allclose(got["v"].to_numpy(),
+                                   exp["v"].to_numpy(), rtol=0,
+                                   equal_nan=True, err_msg=f"round{d}")
+        np.testing.assert_array_equal(got["w"].to_numpy(),
+                                      exp["w"].to_numpy())
+    for by, asc, napos in (("w", True, "last"), ("v", False, "last"),
+                           ("v", True, "first")):
+        got = df.sort_values(by, ascending=asc,
+                             na_position=napos).to_pandas()
+        exp = pdf.sort_values(by, ascending=asc, kind="stable",
+                              na_position=napos)
+        np.testing.assert_array_equal(got.index.to_numpy(),
+                                      exp.index.to_numpy(),
+                                      err_msg=f"{by}/{asc}/{napos}")
+    got = df.duplicated(["w"]).to_pandas()
+    exp = pdf.duplicated(subset=["w"])
+    np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy())
+    got = df.drop_duplicates(["w"]).to_pandas()
+    exp = pdf.drop_duplicates(subset=["w"])
+    np.testing.assert_array_equal(got.index.to_numpy(),
+                                  exp.index.to_numpy())
