@@ -1264,7 +1264,7 @@ class HipDataframe:
             if b not in self.columns:
                 raise lib.HfError(f"groupby: key column {b!r} missing")
         if how not in ("cumsum", "cummin", "cummax", "cumcount", "rank",
-                       "ngroup", "shift", "diff",
+                       "ngroup", "shift", "diff", "ffill", "bfill",
                        "bsum", "bmin", "bmax", "bcount", "bmean"):
             raise lib.HfError(f"groupby transform {how!r} not supported")
         if how == "rank" and method not in ("average", "min", "first"):
@@ -1384,6 +1384,56 @@ class HipDataframe:
             part = HipDataframePartition(DeviceBlock(out_cols, n))
             return HipDataframe([part], pandas.RangeIndex(n), val_names,
                                 [n], pandas.Series(dts))
+        if how in ("ffill", "bfill"):
+            # pinned prototype (test_host_logic.py
+            # test_ffill_bfill_composition_prototype): segmented MAX over
+            # (valid ? sorted position : −1) finds each row's last valid
+            # source within its key run; bfill = ffill over reversed rows
+            # (the reversal permutation is its own inverse).
+            rstate = None
+            if how == "bfill":
+                # reversal permutation (its own inverse) + reversed sort
+                # state, built once; the index upload is 8n B H2D —
+                # a device iota-reversal is a later micro-optimization
+                ridx = lib.put(np.arange(n - 1, -1, -1, dtype=np.int64))
+                reff = [(lib.gather(ekc, ridx), ea)
+                        for ekc, ea in eff_keys]
+                rperm = self._compose_sort_perm(reff)
+                rhead = None
+                for ekc, _ in reff:
+                    h = self._run_head_col(lib.gather(ekc, rperm), n)
+                    rhead = h if rhead is None else lib.binary(
+                        lib.BIN_ADD, rhead, h)
+                if len(reff) > 1:
+                    rhead = lib.compare_scalar(lib.CMP_GE, rhead, 1.0)
+                rstate = (ridx, rperm, rhead)
+            for v in val_names:
+                vc = concat_col(v)
+                src_int = vc.dtype_code == lib.HF_INT64
+                if src_int:
+                    # int columns hold no NaN: fill is the identity
+                    if valid is None:
+                        out_cols[v] = vc
+                        dts[v] = np.dtype(np.int64)
+                    else:
+                        out_cols[v] = lib.fixup_empty(lib.cast_f64(vc),
+                                                      valid)
+                        dts[v] = np.dtype(np.float64)
+                    continue
+                if how == "bfill":
+                    ridx, rperm, rhead = rstate
+                    filled = self._seg_ffill_col(lib.gather(vc, ridx),
+                                                 rperm, rhead, n)
+                    res = lib.gather(filled, ridx)
+                else:
+                    res = self._seg_ffill_col(vc, perm, head, n)
+                if valid is not None:
+                    res = lib.fixup_empty(res, valid)
+                out_cols[v] = res
+                dts[v] = np.dtype(np.float64)
+            part = HipDataframePartition(DeviceBlock(out_cols, n))
+            return HipDataframe([part], pandas.RangeIndex(n), val_names,
+                                [n], pandas.Series(dts))
         if how.startswith("b"):
             # broadcast aggregate (pandas gb.transform('sum'|'mean'|...)):
             # per-run aggregate = segmented-scan value at the run's LAST
@@ -1466,10 +1516,27 @@ class HipDataframe:
         return HipDataframe([part], pandas.RangeIndex(n), val_names, [n],
                             pandas.Series(dts))
 
-    def rank_rows(self, ascending: bool = True,
-                  method: str = "average") -> "HipDataframe":
-        """Frame-level pandas rank(axis=0): the groupby rank machinery
-        over ONE synthetic constant-key group (a zeros key column)."""
+    @staticmethod
+    def _seg_ffill_col(vc, perm, head, n):
+        """ffill one f64 column within key runs: segmented MAX over
+        (valid ? sorted position : −1), gather the source rows, NaN
+        where no prior valid exists, inverse-scatter to original order
+        (the pinned prototype, test_host_logic.py)."""
+        sv = lib.gather(vc, perm)
+        m = lib.compare_scalar(lib.CMP_NOTNA, sv, 0.0)
+        pos = HipDataframe._iota(n)
+        posv = lib.binary(lib.BIN_ADD,
+                          lib.binary(lib.BIN_MUL, pos, m),
+                          lib.map_scalar(lib.MAP_SUB, m, 1))
+        segmax = lib.seg_cumsum(posv, head, lib.AGG_MAX)
+        ok = lib.compare_scalar(lib.CMP_GE, segmax, 0.0)
+        cidx = lib.map_scalar(lib.MAP_MAX, segmax, 0)
+        filled = lib.fixup_empty(lib.gather(sv, cidx), ok)
+        return lib.scatter(filled, perm)
+
+    def _with_const_key(self) -> "HipDataframe":
+        """Copy of this frame with a zeros KEYCOL — frame-level transforms
+        (rank, ffill/bfill) ride the groupby machinery over ONE group."""
         parts = []
         for p in self._partitions:
             b = p.block()
@@ -1481,11 +1548,19 @@ class HipDataframe:
                 DeviceBlock(cols, b.length, b.cats)))
         dtypes = pandas.concat([self.dtypes, pandas.Series(
             {self.KEYCOL: np.dtype(np.int64)})])
-        tmp = HipDataframe(parts, self._index,
-                           list(self.columns) + [self.KEYCOL],
-                           self._row_lengths, dtypes)
-        return tmp.groupby_transform(self.KEYCOL, "rank",
-                                     ascending=ascending, method=method)
+        return HipDataframe(parts, self._index,
+                            list(self.columns) + [self.KEYCOL],
+                            self._row_lengths, dtypes)
+
+    def rank_rows(self, ascending: bool = True,
+                  method: str = "average") -> "HipDataframe":
+        """Frame-level pandas rank(axis=0) over one constant-key group."""
+        return self._with_const_key().groupby_transform(
+            self.KEYCOL, "rank", ascending=ascending, method=method)
+
+    def fill_rows(self, how: str) -> "HipDataframe":
+        """Frame-level pandas ffill/bfill over one constant-key group."""
+        return self._with_const_key().groupby_transform(self.KEYCOL, how)
 
     def groupby_idxminmax(self, by, maximum: bool) -> "HipDataframe":
         """groupby.idxmax/idxmin: per group and value column, the ORIGINAL

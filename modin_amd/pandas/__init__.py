@@ -168,6 +168,16 @@ class _HipPandasBase:
         """pandas pct_change(fill_method=None): x / x.shift(p) - 1."""
         return self / self.shift(periods) - 1
 
+    def ffill(self):
+        """pandas ffill (forward fill down the rows)."""
+        return self._rewrap(self._query_compiler.fillna_directional(
+            "ffill"))
+
+    def bfill(self):
+        """pandas bfill (backward fill down the rows)."""
+        return self._rewrap(self._query_compiler.fillna_directional(
+            "bfill"))
+
     def idxmax(self):
         return self._lower(self._query_compiler.idxmax())
 
@@ -788,6 +798,14 @@ class DataFrameGroupBy:
 
     def diff(self, periods: int = 1):
         return self._transform("diff", periods=int(periods))
+
+    def ffill(self):
+        """pandas DataFrameGroupBy.ffill: forward fill within groups."""
+        return self._transform("ffill")
+
+    def bfill(self):
+        """pandas DataFrameGroupBy.bfill: backward fill within groups."""
+        return self._transform("bfill")
 
     def rank(self, method: str = "average", ascending: bool = True,
              na_option: str = "keep"):

@@ -203,6 +203,10 @@ class HipQueryCompiler:
         return self.__constructor__(self._modin_frame.rank_rows(
             ascending=ascending, method=method))
 
+    def fillna_directional(self, how: str) -> "HipQueryCompiler":
+        """pandas ffill/bfill (frame-level, one constant-key group)."""
+        return self.__constructor__(self._modin_frame.fill_rows(how))
+
     def duplicated(self, subset=None) -> "HipQueryCompiler":
         """Row-duplicate mask, keep='first' (pandas duplicated): cumcount
         over ALL subset columns with dropna=False (NaN==NaN, the

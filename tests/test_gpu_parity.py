@@ -1698,3 +1698,51 @@ def test_groupby_dropna_false_vs_pandas(npartitions):
         df2.groupby("k", dropna=False).var()
     with pytest.raises(lib.HfError, match="dropna=False"):
         df2.groupby("k", dropna=False).median()
+
+
+def test_groupby_ffill_bfill_vs_pandas(npartitions):
+    """groupby/frame ffill & bfill: segmented MAX over valid positions
+    (composition CPU-validated on the numpy lib mock,
+    tests/test_mock_compositions.py; here the same assertions run on the
+    real kernels)."""
+    rng = np.random.default_rng(110)
+    n = 60_000
+    k = rng.integers(0, 200, n).astype(np.float64)
+    k[rng.random(n) < 0.03] = np.nan
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.3] = np.nan
+    w = rng.integers(-40, 40, n)
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    for how in ("ffill", "bfill"):
+        got = getattr(df.groupby("k"), how)().to_pandas()
+        exp = getattr(pdf.groupby("k"), how)()
+        assert list(got.dtypes) == list(exp.dtypes), how
+        for c in exp.columns:
+            np.testing.assert_allclose(got[c].to_numpy(),
+                                       exp[c].to_numpy(), rtol=0,
+                                       equal_nan=True,
+                                       err_msg=f"{how}/{c}")
+        got2 = getattr(df, how)().to_pandas()
+        exp2 = getattr(pdf, how)()
+        assert list(got2.dtypes) == list(exp2.dtypes), how
+        for c in exp2.columns:
+            np.testing.assert_allclose(got2[c].to_numpy(),
+                                       exp2[c].to_numpy(), rtol=0,
+                                       equal_nan=True,
+                                       err_msg=f"frame-{how}/{c}")
+    # multi-key + selection forms
+    pdf2 = pandas.DataFrame({"a": rng.integers(0, 5, 5000),
+                             "b": rng.choice(["x", "y"], 5000),
+                             "v": np.where(rng.random(5000) < 0.4, np.nan,
+                                           rng.standard_normal(5000))})
+    df2 = mpd.DataFrame(pdf2)
+    got3 = df2.groupby(["a", "b"]).ffill().to_pandas()
+    exp3 = pdf2.groupby(["a", "b"]).ffill()
+    np.testing.assert_allclose(got3["v"].to_numpy(),
+                               exp3["v"].to_numpy(), rtol=0,
+                               equal_nan=True)
+    s_ = df2.groupby(["a", "b"])["v"].bfill().to_pandas()
+    np.testing.assert_allclose(
+        s_.to_numpy(), pdf2.groupby(["a", "b"])["v"].bfill().to_numpy(),
+        rtol=0, equal_nan=True)

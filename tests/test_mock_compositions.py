@@ -129,3 +129,30 @@ def test_mock_where_round_sort_dedup(mlib):
     exp = pdf.drop_duplicates(subset=["w"])
     np.testing.assert_array_equal(got.index.to_numpy(),
                                   exp.index.to_numpy())
+
+
+def test_mock_ffill_bfill(mlib):
+    rng = np.random.default_rng(53)
+    pdf = _frames(rng, 3000)
+    df = mlib.DataFrame(pdf)
+    for how in ("ffill", "bfill"):
+        got = getattr(df.groupby("k"), how)().to_pandas()
+        exp = getattr(pdf.groupby("k"), how)()
+        assert list(got.dtypes) == list(exp.dtypes), how
+        for c in exp.columns:
+            np.testing.assert_allclose(got[c].to_numpy(),
+                                       exp[c].to_numpy(), rtol=0,
+                                       equal_nan=True,
+                                       err_msg=f"{how}/{c}")
+        got2 = getattr(df, how)().to_pandas()
+        exp2 = getattr(pdf, how)()
+        assert list(got2.dtypes) == list(exp2.dtypes), how
+        for c in exp2.columns:
+            np.testing.assert_allclose(got2[c].to_numpy(),
+                                       exp2[c].to_numpy(), rtol=0,
+                                       equal_nan=True,
+                                       err_msg=f"frame-{how}/{c}")
+    s_ = df.groupby("k")["v"].ffill().to_pandas()
+    np.testing.assert_allclose(s_.to_numpy(),
+                               pdf.groupby("k")["v"].ffill().to_numpy(),
+                               rtol=0, equal_nan=True)
