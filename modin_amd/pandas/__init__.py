@@ -799,6 +799,16 @@ class DataFrameGroupBy:
     def diff(self, periods: int = 1):
         return self._transform("diff", periods=int(periods))
 
+    def pct_change(self, periods: int = 1):
+        """pandas DataFrameGroupBy.pct_change(fill_method=None):
+        x / x.shift(p within group) - 1."""
+        bys = (list(self._by) if isinstance(self._by, (list, tuple))
+               else [self._by])
+        vals = [c for c in self._df.columns if c not in bys]
+        shifted = self._transform("shift", periods=int(periods))
+        base = self._df[vals[0]] if self._series_out else self._df[vals]
+        return base / shifted - 1
+
     def ffill(self):
         """pandas DataFrameGroupBy.ffill: forward fill within groups."""
         return self._transform("ffill")

@@ -1746,3 +1746,28 @@ def test_groupby_ffill_bfill_vs_pandas(npartitions):
     np.testing.assert_allclose(
         s_.to_numpy(), pdf2.groupby(["a", "b"])["v"].bfill().to_numpy(),
         rtol=0, equal_nan=True)
+
+
+def test_groupby_pct_change_vs_pandas(npartitions):
+    rng = np.random.default_rng(111)
+    n = 40_000
+    k = rng.integers(0, 150, n).astype(np.float64)
+    k[rng.random(n) < 0.02] = np.nan
+    v = rng.standard_normal(n) + 4
+    v[rng.random(n) < 0.1] = np.nan
+    w = rng.integers(1, 50, n)
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    for p in (1, 3):
+        got = df.groupby("k").pct_change(p).to_pandas()
+        exp = pdf.groupby("k").pct_change(p, fill_method=None)
+        for c in exp.columns:
+            np.testing.assert_allclose(got[c].to_numpy(),
+                                       exp[c].to_numpy(), rtol=1e-12,
+                                       atol=1e-12, equal_nan=True,
+                                       err_msg=f"pct({p})/{c}")
+    s_ = df.groupby("k")["v"].pct_change().to_pandas()
+    np.testing.assert_allclose(
+        s_.to_numpy(),
+        pdf.groupby("k")["v"].pct_change(fill_method=None).to_numpy(),
+        rtol=1e-12, atol=1e-12, equal_nan=True)

@@ -156,3 +156,22 @@ def test_mock_ffill_bfill(mlib):
     np.testing.assert_allclose(s_.to_numpy(),
                                pdf.groupby("k")["v"].ffill().to_numpy(),
                                rtol=0, equal_nan=True)
+
+
+def test_mock_groupby_pct_change(mlib):
+    rng = np.random.default_rng(59)
+    pdf = _frames(rng, 2500)
+    df = mlib.DataFrame(pdf)
+    for p in (1, 2):
+        got = df.groupby("k").pct_change(p).to_pandas()
+        exp = pdf.groupby("k").pct_change(p, fill_method=None)
+        for c in exp.columns:
+            np.testing.assert_allclose(got[c].to_numpy(),
+                                       exp[c].to_numpy(), rtol=1e-12,
+                                       atol=1e-12, equal_nan=True,
+                                       err_msg=f"pct({p})/{c}")
+    s_ = df.groupby("k")["v"].pct_change().to_pandas()
+    np.testing.assert_allclose(
+        s_.to_numpy(),
+        pdf.groupby("k")["v"].pct_change(fill_method=None).to_numpy(),
+        rtol=1e-12, atol=1e-12, equal_nan=True)
