@@ -231,6 +231,11 @@ class _HipPandasBase:
         return self.__truediv__(other)
 
     def fillna(self, value):
+        """pandas fillna: scalar, or {column: scalar} dict."""
+        if isinstance(value, dict):
+            return self._rewrap(
+                self._query_compiler.fillna_dict(
+                    {k: float(v) for k, v in value.items()}))
         return self._rewrap(type(self._query_compiler).fillna(self._query_compiler,
                                                               float(value)))
 
@@ -544,6 +549,16 @@ class Series(_HipPandasBase):
         qc = self._query_compiler
         return Series(query_compiler=qc.getitem_array(qc.notna()),
                       name=self.name)
+
+    def replace(self, to_replace, value) -> "Series":
+        """pandas Series.replace(scalar, scalar): where(self != a, b) —
+        NaN rows survive untouched (NaN != a is True)."""
+        if not (np.isscalar(to_replace) and np.isscalar(value)):
+            raise lib.HfError("replace: scalar to_replace/value only "
+                              "this round")
+        if isinstance(to_replace, float) and np.isnan(to_replace):
+            return self.fillna(value)  # pandas replace(nan, x) == fillna
+        return self.where(self != to_replace, value)
 
     def between(self, left, right, inclusive: str = "both") -> "Series":
         """pandas Series.between: boolean mask (NaN -> False)."""

@@ -207,6 +207,26 @@ class HipQueryCompiler:
         """pandas ffill/bfill (frame-level, one constant-key group)."""
         return self.__constructor__(self._modin_frame.fill_rows(how))
 
+    def fillna_dict(self, values: dict) -> "HipQueryCompiler":
+        """pandas fillna({column: scalar}): per-column fill values;
+        unlisted columns pass through."""
+        from modin_amd.core.partition import DeviceBlock
+        frame = self._modin_frame
+        for c in values:
+            if c not in frame.columns:
+                raise lib.HfError(f"fillna: column {c!r} missing")
+
+        def block_fn(block):
+            out = {}
+            for name, col in block.columns.items():
+                if name in values and name not in block.cats:
+                    out[name] = lib.map_scalar(lib.MAP_FILLNA, col,
+                                               values[name])
+                else:
+                    out[name] = col
+            return DeviceBlock(out, block.length, block.cats)
+        return self.__constructor__(frame.map(block_fn))
+
     def duplicated(self, subset=None) -> "HipQueryCompiler":
         """Row-duplicate mask, keep='first' (pandas duplicated): cumcount
         over ALL subset columns with dropna=False (NaN==NaN, the

@@ -1771,3 +1771,28 @@ def test_groupby_pct_change_vs_pandas(npartitions):
         s_.to_numpy(),
         pdf.groupby("k")["v"].pct_change(fill_method=None).to_numpy(),
         rtol=1e-12, atol=1e-12, equal_nan=True)
+
+
+def test_fillna_dict_replace_vs_pandas(npartitions):
+    rng = np.random.default_rng(112)
+    n = 20_000
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.15] = np.nan
+    u = rng.standard_normal(n)
+    u[rng.random(n) < 0.15] = np.nan
+    w = rng.integers(0, 9, n)
+    pdf = pandas.DataFrame({"v": v, "u": u, "w": w})
+    df = mpd.DataFrame(pdf)
+    got = df.fillna({"v": -1.0, "u": 2.5}).to_pandas()
+    exp = pdf.fillna({"v": -1.0, "u": 2.5})
+    for c in exp.columns:
+        np.testing.assert_allclose(got[c].to_numpy(), exp[c].to_numpy(),
+                                   rtol=0, equal_nan=True, err_msg=c)
+    s_ = df["v"].replace(float(v[0]), 99.0).to_pandas()
+    e_ = pdf["v"].replace(float(v[0]), 99.0)
+    np.testing.assert_allclose(s_.to_numpy(), e_.to_numpy(), rtol=0,
+                               equal_nan=True)
+    s2 = df["w"].replace(3, -7).to_pandas()
+    e2 = pdf["w"].replace(3, -7)
+    assert s2.dtype == e2.dtype
+    np.testing.assert_array_equal(s2.to_numpy(), e2.to_numpy())

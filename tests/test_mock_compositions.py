@@ -175,3 +175,25 @@ def test_mock_groupby_pct_change(mlib):
         s_.to_numpy(),
         pdf.groupby("k")["v"].pct_change(fill_method=None).to_numpy(),
         rtol=1e-12, atol=1e-12, equal_nan=True)
+
+
+def test_mock_fillna_dict_replace(mlib):
+    rng = np.random.default_rng(61)
+    pdf = _frames(rng, 2000, nan_keys=False)
+    df = mlib.DataFrame(pdf)
+    got = df.fillna({"v": -1.5}).to_pandas()
+    exp = pdf.fillna({"v": -1.5})
+    for c in exp.columns:
+        np.testing.assert_allclose(got[c].to_numpy(), exp[c].to_numpy(),
+                                   rtol=0, equal_nan=True, err_msg=c)
+    s = pdf["w"].astype(np.float64)
+    s[rng.random(2000) < 0.1] = np.nan
+    pdf2 = pandas.DataFrame({"x": s})
+    df2 = mlib.DataFrame(pdf2)
+    got = df2["x"].replace(7.0, -99.0).to_pandas()
+    exp = pdf2["x"].replace(7.0, -99.0)
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(), rtol=0,
+                               equal_nan=True)
+    got = df2["x"].replace(np.nan, 0.5).to_pandas()
+    exp = pdf2["x"].replace(np.nan, 0.5)
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(), rtol=0)
