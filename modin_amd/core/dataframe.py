@@ -1552,6 +1552,120 @@ class HipDataframe:
                             list(self.columns) + [self.KEYCOL],
                             self._row_lengths, dtypes)
 
+    def rolling_agg(self, window: int, min_periods, op: str
+                    ) -> "HipDataframe":
+        """pandas rolling(window, min_periods).sum/mean/count/min/max —
+        no new kernels (the pinned prototype, test_host_logic.py):
+        sum/mean/count ride windowed differences of NaN-zero-filled
+        prefix sums; min/max ride the van Herk/Gil-Werman two-scan (per
+        w-aligned tile, prefix extreme = segmented scan with tile-start
+        heads; suffix extreme = the same scan over reversed rows) with
+        window extreme = comb(suffix[lo], prefix[i]).  Gates: count
+        emits when the window holds >= min_periods ROWS; the other aggs
+        when it holds >= min_periods OBSERVATIONS (non-NaN) — measured
+        pandas 2.3.3 behavior.  Results float64 (pandas)."""
+        w = int(window)
+        minp = w if min_periods is None else int(min_periods)
+        if w < 1 or minp < 0 or minp > w:
+            raise lib.HfError("rolling: need 1 <= min_periods <= window")
+        if op not in ("sum", "mean", "count", "min", "max"):
+            raise lib.HfError(f"rolling.{op} not supported")
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        if blk_cats:
+            raise lib.HfError("rolling over string columns")
+        if not isinstance(self._index, pandas.RangeIndex) or \
+                self._index.start != 0 or self._index.step != 1:
+            raise lib.HfError("rolling: only RangeIndex frames this round")
+        n = len(self)
+        names = list(self.columns)
+        if n == 0:
+            part = HipDataframePartition(DeviceBlock(
+                {c: lib.alloc(0, lib.HF_FLOAT64) for c in names}, 0))
+            return HipDataframe([part], pandas.RangeIndex(0), names, [0],
+                                pandas.Series({c: np.dtype(np.float64)
+                                               for c in names}))
+
+        def concat_col(name):
+            cols = [p.block().columns[name] for p in self._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        pos = self._iota(n)
+        lo = lib.map_scalar(lib.MAP_MAX,
+                            lib.map_scalar(lib.MAP_SUB, pos, w - 1), 0)
+        pidx = lib.map_scalar(lib.MAP_MAX,
+                              lib.map_scalar(lib.MAP_SUB, lo, 1), 0)
+        has_prev = lib.compare_scalar(
+            lib.CMP_GE, lib.map_scalar(lib.MAP_SUB, pos, w - 1), 1.0)
+        avail_ok = lib.compare_scalar(
+            lib.CMP_GE,
+            lib.map_scalar(lib.MAP_ADD,
+                           lib.binary(lib.BIN_SUB, pos, lo), 1),
+            float(minp))
+        full = lib.compare_scalar(lib.CMP_GE, pos, float(w - 1))
+        pad = -(-n // w) * w
+        rev = tile_head = None
+        if op in ("min", "max"):
+            rev = lib.put(np.arange(pad - 1, -1, -1, dtype=np.int64))
+            ones = lib.alloc(pad, lib.HF_INT64)
+            lib.fill_i64(ones.dptr(), 1, pad)
+            ipad = lib.filter_iota(lib.filter_plan(ones), 0)
+            tid = lib.map_scalar(
+                lib.MAP_CAST_I64,
+                lib.map_scalar(lib.MAP_DIV, ipad, float(w)), 0)
+            tile_head = lib.compare_scalar(
+                lib.CMP_EQ,
+                lib.binary(lib.BIN_SUB, ipad,
+                           lib.map_scalar(lib.MAP_MUL, tid, w)), 0.0)
+        out_cols = {}
+        for c in names:
+            vc = lib.cast_f64(concat_col(c))
+            m = lib.compare_scalar(lib.CMP_NOTNA, vc, 0.0)
+            ccnt = lib.cumsum(m)
+            wcnt = lib.binary(
+                lib.BIN_SUB, ccnt,
+                lib.binary(lib.BIN_MUL, lib.gather(ccnt, pidx), has_prev))
+            obs_ok = lib.compare_scalar(lib.CMP_GE, wcnt, float(minp))
+            if op == "count":
+                res = lib.fixup_empty(lib.cast_f64(wcnt), avail_ok)
+            elif op in ("sum", "mean"):
+                zf = lib.map_scalar(lib.MAP_FILLNA, vc, 0.0)
+                cz = lib.cumsum(zf)
+                wsum = lib.binary(
+                    lib.BIN_SUB, cz,
+                    lib.binary(lib.BIN_MUL, lib.gather(cz, pidx),
+                               lib.cast_f64(has_prev)))
+                if op == "mean":
+                    wsum = lib.binary(lib.BIN_DIV, wsum,
+                                      lib.cast_f64(wcnt))
+                res = lib.fixup_empty(wsum, obs_ok)
+            else:
+                ident = float("inf") if op == "min" else float("-inf")
+                agg_op = lib.AGG_MIN if op == "min" else lib.AGG_MAX
+                bop = lib.BIN_MIN if op == "min" else lib.BIN_MAX
+                vfill = lib.map_scalar(lib.MAP_FILLNA, vc, ident)
+                if pad > n:
+                    tailc = lib.alloc(pad - n, lib.HF_FLOAT64)
+                    lib.fill_f64(tailc.dptr(), ident, pad - n)
+                    vpad = lib.concat([vfill, tailc])
+                else:
+                    vpad = vfill
+                pref = lib.seg_cumsum(vpad, tile_head, agg_op)
+                suff = lib.gather(
+                    lib.seg_cumsum(lib.gather(vpad, rev), tile_head,
+                                   agg_op), rev)
+                # head windows (< w rows) must use the prefix alone:
+                # NaN-select the suffix leg there (fmin/fmax skip NaN)
+                suff_sel = lib.fixup_empty(lib.gather(suff, lo), full)
+                ans = lib.binary(bop, suff_sel,
+                                 lib.col_slice(pref, 0, n))
+                res = lib.fixup_empty(ans, obs_ok)
+            out_cols[c] = res
+        part = HipDataframePartition(DeviceBlock(out_cols, n))
+        return HipDataframe([part], pandas.RangeIndex(n), names, [n],
+                            pandas.Series({c: np.dtype(np.float64)
+                                           for c in names}))
+
     def rank_rows(self, ascending: bool = True,
                   method: str = "average") -> "HipDataframe":
         """Frame-level pandas rank(axis=0) over one constant-key group."""

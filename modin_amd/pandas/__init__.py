@@ -168,6 +168,11 @@ class _HipPandasBase:
         """pandas pct_change(fill_method=None): x / x.shift(p) - 1."""
         return self / self.shift(periods) - 1
 
+    def rolling(self, window: int, min_periods=None):
+        """pandas rolling(window, min_periods): fixed forward-closed
+        windows; .sum/.mean/.count/.min/.max."""
+        return Rolling(self, int(window), min_periods)
+
     def ffill(self):
         """pandas ffill (forward fill down the rows)."""
         return self._rewrap(self._query_compiler.fillna_directional(
@@ -919,3 +924,35 @@ class DataFrameGroupBy:
                     qcs.append(qc.rename_columns({col: (col, a)}))
             return DataFrame(query_compiler=qcs[0].hconcat(qcs[1:]))
         raise lib.HfError("groupby.agg accepts str / list / dict")
+
+
+class Rolling:
+    """pandas Rolling over fixed windows (mirrors pandas.core.window
+    Rolling for the sum/mean/count/min/max aggs; min_periods rules as
+    measured on pandas 2.3.3 — count gates on window rows, the rest on
+    non-NaN observations)."""
+
+    def __init__(self, obj, window: int, min_periods=None):
+        self._obj = obj
+        self._window = window
+        self._min_periods = min_periods
+
+    def _agg(self, op: str):
+        qc = self._obj._query_compiler.rolling_agg(
+            self._window, self._min_periods, op)
+        return self._obj._rewrap(qc)
+
+    def sum(self):
+        return self._agg("sum")
+
+    def mean(self):
+        return self._agg("mean")
+
+    def count(self):
+        return self._agg("count")
+
+    def min(self):
+        return self._agg("min")
+
+    def max(self):
+        return self._agg("max")

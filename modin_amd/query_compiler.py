@@ -207,6 +207,11 @@ class HipQueryCompiler:
         """pandas ffill/bfill (frame-level, one constant-key group)."""
         return self.__constructor__(self._modin_frame.fill_rows(how))
 
+    def rolling_agg(self, window: int, min_periods,
+                    op: str) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.rolling_agg(
+            window, min_periods, op))
+
     def fillna_dict(self, values: dict) -> "HipQueryCompiler":
         """pandas fillna({column: scalar}): per-column fill values;
         unlisted columns pass through."""

@@ -1796,3 +1796,30 @@ def test_fillna_dict_replace_vs_pandas(npartitions):
     e2 = pdf["w"].replace(3, -7)
     assert s2.dtype == e2.dtype
     np.testing.assert_array_equal(s2.to_numpy(), e2.to_numpy())
+
+
+def test_rolling_vs_pandas(npartitions):
+    """rolling(w, min_periods).sum/mean/count/min/max: prefix-sum windows
+    + van Herk two-scan extremes (composition CPU-validated on the mock
+    tier; here on the real kernels)."""
+    rng = np.random.default_rng(113)
+    n = 50_000
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.15] = np.nan
+    w = rng.integers(-30, 30, n)
+    pdf = pandas.DataFrame({"v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    for win, mp in ((3, None), (16, 4), (250, 1), (7, 7)):
+        for op in ("sum", "mean", "count", "min", "max"):
+            got = getattr(df.rolling(win, min_periods=mp), op)() \
+                .to_pandas()
+            exp = getattr(pdf.rolling(win, min_periods=mp), op)()
+            for c in exp.columns:
+                np.testing.assert_allclose(
+                    got[c].to_numpy(), exp[c].to_numpy(), rtol=1e-12,
+                    atol=1e-9, equal_nan=True,
+                    err_msg=f"{op}/w={win}/mp={mp}/{c}")
+    s_ = df["v"].rolling(5).mean().to_pandas()
+    np.testing.assert_allclose(s_.to_numpy(),
+                               pdf["v"].rolling(5).mean().to_numpy(),
+                               rtol=1e-12, atol=1e-12, equal_nan=True)

@@ -197,3 +197,28 @@ def test_mock_fillna_dict_replace(mlib):
     got = df2["x"].replace(np.nan, 0.5).to_pandas()
     exp = pdf2["x"].replace(np.nan, 0.5)
     np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(), rtol=0)
+
+
+def test_mock_rolling(mlib):
+    rng = np.random.default_rng(67)
+    n = 3000
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.15] = np.nan
+    w = rng.integers(-20, 20, n)
+    pdf = pandas.DataFrame({"v": v, "w": w})
+    df = mlib.DataFrame(pdf)
+    for win, mp in ((1, None), (3, None), (3, 1), (16, 4), (100, None),
+                    (7, 7)):
+        for op in ("sum", "mean", "count", "min", "max"):
+            got = getattr(df.rolling(win, min_periods=mp), op)() \
+                .to_pandas()
+            exp = getattr(pdf.rolling(win, min_periods=mp), op)()
+            for c in exp.columns:
+                np.testing.assert_allclose(
+                    got[c].to_numpy(), exp[c].to_numpy(), rtol=1e-12,
+                    atol=1e-12, equal_nan=True,
+                    err_msg=f"{op}/w={win}/mp={mp}/{c}")
+    s_ = df["v"].rolling(5).mean().to_pandas()
+    np.testing.assert_allclose(s_.to_numpy(),
+                               pdf["v"].rolling(5).mean().to_numpy(),
+                               rtol=1e-12, atol=1e-12, equal_nan=True)
