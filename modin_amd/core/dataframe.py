@@ -1666,6 +1666,59 @@ class HipDataframe:
                             pandas.Series({c: np.dtype(np.float64)
                                            for c in names}))
 
+    def expanding_agg(self, min_periods, op: str) -> "HipDataframe":
+        """pandas expanding(min_periods).sum/mean/count/min/max — the
+        prefix scans directly (identity-filled so running extremes pass
+        NaN rows through), gated like rolling: count on available rows,
+        the rest on non-NaN observations."""
+        minp = 1 if min_periods is None else int(min_periods)
+        if minp < 0:
+            raise lib.HfError("expanding: min_periods >= 0")
+        if op not in ("sum", "mean", "count", "min", "max"):
+            raise lib.HfError(f"expanding.{op} not supported")
+        if (self._partitions and self._partitions[0].block().cats):
+            raise lib.HfError("expanding over string columns")
+        n = len(self)
+        names = list(self.columns)
+
+        def concat_col(name):
+            cols = [p.block().columns[name] for p in self._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        if n == 0:
+            part = HipDataframePartition(DeviceBlock(
+                {c: lib.alloc(0, lib.HF_FLOAT64) for c in names}, 0))
+            return HipDataframe([part], pandas.RangeIndex(0), names, [0],
+                                pandas.Series({c: np.dtype(np.float64)
+                                               for c in names}))
+        pos = self._iota(n)
+        avail_ok = lib.compare_scalar(
+            lib.CMP_GE, lib.map_scalar(lib.MAP_ADD, pos, 1), float(minp))
+        out_cols = {}
+        for c in names:
+            vc = lib.cast_f64(concat_col(c))
+            m = lib.compare_scalar(lib.CMP_NOTNA, vc, 0.0)
+            ccnt = lib.cumsum(m)
+            obs_ok = lib.compare_scalar(lib.CMP_GE, ccnt, float(minp))
+            if op == "count":
+                res = lib.fixup_empty(lib.cast_f64(ccnt), avail_ok)
+            elif op in ("sum", "mean"):
+                cz = lib.cumsum(lib.map_scalar(lib.MAP_FILLNA, vc, 0.0))
+                if op == "mean":
+                    cz = lib.binary(lib.BIN_DIV, cz, lib.cast_f64(ccnt))
+                res = lib.fixup_empty(cz, obs_ok)
+            else:
+                ident = float("inf") if op == "min" else float("-inf")
+                agg_op = lib.AGG_MIN if op == "min" else lib.AGG_MAX
+                run = lib.cumsum(
+                    lib.map_scalar(lib.MAP_FILLNA, vc, ident), agg_op)
+                res = lib.fixup_empty(run, obs_ok)
+            out_cols[c] = res
+        part = HipDataframePartition(DeviceBlock(out_cols, n))
+        return HipDataframe([part], self._index, names, [n],
+                            pandas.Series({c: np.dtype(np.float64)
+                                           for c in names}))
+
     def rank_rows(self, ascending: bool = True,
                   method: str = "average") -> "HipDataframe":
         """Frame-level pandas rank(axis=0) over one constant-key group."""

@@ -173,6 +173,10 @@ class _HipPandasBase:
         windows; .sum/.mean/.count/.min/.max."""
         return Rolling(self, int(window), min_periods)
 
+    def expanding(self, min_periods: int = 1):
+        """pandas expanding(min_periods); .sum/.mean/.count/.min/.max."""
+        return Expanding(self, min_periods)
+
     def ffill(self):
         """pandas ffill (forward fill down the rows)."""
         return self._rewrap(self._query_compiler.fillna_directional(
@@ -940,6 +944,34 @@ class Rolling:
     def _agg(self, op: str):
         qc = self._obj._query_compiler.rolling_agg(
             self._window, self._min_periods, op)
+        return self._obj._rewrap(qc)
+
+    def sum(self):
+        return self._agg("sum")
+
+    def mean(self):
+        return self._agg("mean")
+
+    def count(self):
+        return self._agg("count")
+
+    def min(self):
+        return self._agg("min")
+
+    def max(self):
+        return self._agg("max")
+
+
+class Expanding:
+    """pandas Expanding (growing windows) for sum/mean/count/min/max."""
+
+    def __init__(self, obj, min_periods: int = 1):
+        self._obj = obj
+        self._min_periods = min_periods
+
+    def _agg(self, op: str):
+        qc = self._obj._query_compiler.expanding_agg(self._min_periods,
+                                                     op)
         return self._obj._rewrap(qc)
 
     def sum(self):

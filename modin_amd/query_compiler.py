@@ -212,6 +212,10 @@ class HipQueryCompiler:
         return self.__constructor__(self._modin_frame.rolling_agg(
             window, min_periods, op))
 
+    def expanding_agg(self, min_periods, op: str) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.expanding_agg(
+            min_periods, op))
+
     def fillna_dict(self, values: dict) -> "HipQueryCompiler":
         """pandas fillna({column: scalar}): per-column fill values;
         unlisted columns pass through."""

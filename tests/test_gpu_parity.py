@@ -1823,3 +1823,22 @@ def test_rolling_vs_pandas(npartitions):
     np.testing.assert_allclose(s_.to_numpy(),
                                pdf["v"].rolling(5).mean().to_numpy(),
                                rtol=1e-12, atol=1e-12, equal_nan=True)
+
+
+def test_expanding_vs_pandas(npartitions):
+    rng = np.random.default_rng(114)
+    n = 30_000
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.2] = np.nan
+    w = rng.integers(-9, 9, n)
+    pdf = pandas.DataFrame({"v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    for mp in (1, 5):
+        for op in ("sum", "mean", "count", "min", "max"):
+            got = getattr(df.expanding(mp), op)().to_pandas()
+            exp = getattr(pdf.expanding(mp), op)()
+            for c in exp.columns:
+                np.testing.assert_allclose(
+                    got[c].to_numpy(), exp[c].to_numpy(), rtol=1e-12,
+                    atol=1e-9, equal_nan=True,
+                    err_msg=f"{op}/mp={mp}/{c}")

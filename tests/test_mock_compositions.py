@@ -222,3 +222,22 @@ def test_mock_rolling(mlib):
     np.testing.assert_allclose(s_.to_numpy(),
                                pdf["v"].rolling(5).mean().to_numpy(),
                                rtol=1e-12, atol=1e-12, equal_nan=True)
+
+
+def test_mock_expanding(mlib):
+    rng = np.random.default_rng(71)
+    n = 2000
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.2] = np.nan
+    w = rng.integers(-9, 9, n)
+    pdf = pandas.DataFrame({"v": v, "w": w})
+    df = mlib.DataFrame(pdf)
+    for mp in (1, 3, 50):
+        for op in ("sum", "mean", "count", "min", "max"):
+            got = getattr(df.expanding(mp), op)().to_pandas()
+            exp = getattr(pdf.expanding(mp), op)()
+            for c in exp.columns:
+                np.testing.assert_allclose(
+                    got[c].to_numpy(), exp[c].to_numpy(), rtol=1e-12,
+                    atol=1e-12, equal_nan=True,
+                    err_msg=f"{op}/mp={mp}/{c}")
