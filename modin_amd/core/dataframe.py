@@ -1659,7 +1659,11 @@ class HipDataframe:
                 suff_sel = lib.fixup_empty(lib.gather(suff, lo), full)
                 ans = lib.binary(bop, suff_sel,
                                  lib.col_slice(pref, 0, n))
-                res = lib.fixup_empty(ans, obs_ok)
+                # min/max need >= 1 observation even at min_periods=0,
+                # else the +/-inf scan identity leaks (pandas: NaN)
+                mm_ok = obs_ok if minp >= 1 else lib.compare_scalar(
+                    lib.CMP_GE, wcnt, 1.0)
+                res = lib.fixup_empty(ans, mm_ok)
             out_cols[c] = res
         part = HipDataframePartition(DeviceBlock(out_cols, n))
         return HipDataframe([part], pandas.RangeIndex(n), names, [n],
@@ -1712,7 +1716,10 @@ class HipDataframe:
                 agg_op = lib.AGG_MIN if op == "min" else lib.AGG_MAX
                 run = lib.cumsum(
                     lib.map_scalar(lib.MAP_FILLNA, vc, ident), agg_op)
-                res = lib.fixup_empty(run, obs_ok)
+                # min/max need >= 1 observation even at min_periods=0
+                mm_ok = obs_ok if minp >= 1 else lib.compare_scalar(
+                    lib.CMP_GE, ccnt, 1.0)
+                res = lib.fixup_empty(run, mm_ok)
             out_cols[c] = res
         part = HipDataframePartition(DeviceBlock(out_cols, n))
         return HipDataframe([part], self._index, names, [n],
@@ -2049,13 +2056,22 @@ class HipDataframe:
         if nr == 0 or nl == 0:
             cols = {}
             dts = {}
+            cats = {}
             for c in self.columns:
-                cols[lout[c]] = lib.alloc(0, lib.HF_INT64)
+                code = (lib.HF_FLOAT64 if self.dtypes[c] == np.float64
+                        else lib.HF_INT64)
+                cols[lout[c]] = lib.alloc(0, code)
                 dts[lout[c]] = self.dtypes[c]
+                if c in lcats:
+                    cats[lout[c]] = lcats[c]
             for c in other.columns:
-                cols[rout[c]] = lib.alloc(0, lib.HF_INT64)
+                code = (lib.HF_FLOAT64 if other.dtypes[c] == np.float64
+                        else lib.HF_INT64)
+                cols[rout[c]] = lib.alloc(0, code)
                 dts[rout[c]] = other.dtypes[c]
-            part = HipDataframePartition(DeviceBlock(cols, 0))
+                if c in rcats:
+                    cats[rout[c]] = rcats[c]
+            part = HipDataframePartition(DeviceBlock(cols, 0, cats))
             names = list(cols)
             return HipDataframe([part], pandas.RangeIndex(0), names, [0],
                                 pandas.Series(dts))

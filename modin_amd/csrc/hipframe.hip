@@ -3075,8 +3075,10 @@ int hf_seg_cumsum(const hf_col* col, const hf_col* heads, int agg_op,
   const int64_t ntiles = (n + FILT_TILE - 1) / FILT_TILE;
   void* d_tv = nullptr;
   void* d_tf = nullptr;
-  HF_HIP("hf_seg_cumsum", dev_alloc(&d_tv, ntiles * 8, g.stream));
   rc = [&]() -> int {
+    // both tile allocs inside the lambda so the hf_col_free(*out) error
+    // path below covers an alloc failure (no *out leak)
+    HF_HIP("hf_seg_cumsum", dev_alloc(&d_tv, ntiles * 8, g.stream));
     HF_HIP("hf_seg_cumsum", dev_alloc(&d_tf, ntiles * 4, g.stream));
     auto run = [&](auto tTag, auto opTag) -> int {
       using T = decltype(tTag);
@@ -3111,7 +3113,7 @@ int hf_seg_cumsum(const hf_col* col, const hf_col* heads, int agg_op,
                : runT(std::integral_constant<int, HF_AGG_SUM>{});
   }();
   if (d_tf) dev_free(d_tf, g.stream);
-  dev_free(d_tv, g.stream);
+  if (d_tv) dev_free(d_tv, g.stream);
   if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
   return rc;
 }

@@ -228,7 +228,10 @@ class HipQueryCompiler:
         def block_fn(block):
             out = {}
             for name, col in block.columns.items():
-                if name in values and name not in block.cats:
+                # int64 columns are NaN-free: pandas leaves them (and
+                # their dtype) untouched — skip MAP_FILLNA's f64 promotion
+                if (name in values and name not in block.cats
+                        and col.dtype_code != lib.HF_INT64):
                     out[name] = lib.map_scalar(lib.MAP_FILLNA, col,
                                                values[name])
                 else:
