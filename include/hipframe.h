@@ -171,6 +171,19 @@ int hf_groupby_accum(const hf_col* keys,            /* HF_INT64, len n        */
 int hf_fill_f64(uintptr_t dptr, double value, int64_t n);
 int hf_fill_i64(uintptr_t dptr, int64_t value, int64_t n);
 
+/* ---- deterministic device-side random fills (synthetic data generation:
+ * the bench's frames are born in HBM; oracle.rand_* mirrors the splitmix64
+ * formulas bit-exactly so the verify gate can regenerate expectations on the
+ * host — the device analog of the reference benchmarks' np.random frames,
+ * e.g. modin/tests/pandas/test_groupby.py fixtures).
+ *   randint : col[i] = lo + splitmix64(seed+i) % (hi-lo)     (HF_INT64)
+ *   randf64 : col[i] = (splitmix64(seed+i) >> 11) * 2^-53    (HF_FLOAT64)
+ *   randcdf : u as randf64; col[i] = searchsorted(cdf, u, side='right')
+ *             over a sorted f64 cdf column (zipf & friends)  (HF_INT64) */
+int hf_fill_randint(hf_col* col, uint64_t seed, int64_t lo, int64_t hi);
+int hf_fill_randf64(hf_col* col, uint64_t seed);
+int hf_fill_randcdf(hf_col* col, uint64_t seed, const hf_col* cdf);
+
 /* ---- hash-table groupby (unbounded key ranges, single rank) ----
  * Open-addressing table of H power-of-2 slots (+1 special slot for the
  * INT64_MIN sentinel key): tkey i64[H+1] initialised to INT64_MIN via

@@ -108,6 +108,11 @@ def load() -> ct.CDLL:
                                             ct.c_size_t, ct.c_size_t]),
             "hf_fill_f64": (ct.c_int, [ct.c_size_t, ct.c_double, ct.c_int64]),
             "hf_fill_i64": (ct.c_int, [ct.c_size_t, ct.c_int64, ct.c_int64]),
+            "hf_fill_randint": (ct.c_int, [ct.c_void_p, ct.c_uint64,
+                                           ct.c_int64, ct.c_int64]),
+            "hf_fill_randf64": (ct.c_int, [ct.c_void_p, ct.c_uint64]),
+            "hf_fill_randcdf": (ct.c_int, [ct.c_void_p, ct.c_uint64,
+                                           ct.c_void_p]),
             "hf_groupby_hash_accum": (ct.c_int, [ct.c_void_p,
                                                  ct.POINTER(ct.c_void_p),
                                                  ct.c_int, ct.c_int,
@@ -203,6 +208,7 @@ def exported_symbols():
         "hf_memset_raw", "hf_map_scalar", "hf_map_scalar_i64", "hf_binary",
         "hf_reduce", "hf_groupby_accum", "hf_groupby_compact", "hf_fill_f64",
         "hf_fixup_empty", "hf_sort_perm", "hf_fill_i64",
+        "hf_fill_randint", "hf_fill_randf64", "hf_fill_randcdf",
         "hf_groupby_hash_accum", "hf_groupby_hash_compact",
         "hf_groupby_sorted", "hf_shuffle_dest", "hf_memcpy_dd",
         "hf_search_sorted", "hf_ordered_i64", "hf_cumsum", "hf_seg_cumsum",
@@ -415,6 +421,31 @@ def fill_f64(dptr: int, value: float, n: int) -> None:
 def fill_i64(dptr: int, value: int, n: int) -> None:
     ensure_ready()
     _check(load().hf_fill_i64(dptr, value, n), "hf_fill_i64")
+
+
+def fill_randint(n: int, seed: int, lo: int, hi: int) -> ColumnRef:
+    """Device-generated int64 column: lo + splitmix64(seed+i) % (hi-lo)
+    (oracle.rand_int mirrors this bit-exactly)."""
+    col = alloc(n, HF_INT64)
+    _check(load().hf_fill_randint(col.handle, seed, lo, hi),
+           "hf_fill_randint")
+    return col
+
+
+def fill_randf64(n: int, seed: int) -> ColumnRef:
+    """Device-generated float64 column uniform in [0,1) (oracle.rand_f64)."""
+    col = alloc(n, HF_FLOAT64)
+    _check(load().hf_fill_randf64(col.handle, seed), "hf_fill_randf64")
+    return col
+
+
+def fill_randcdf(n: int, seed: int, cdf: ColumnRef) -> ColumnRef:
+    """Device inverse-CDF draw: searchsorted(cdf, U[0,1), side='right')
+    (oracle.rand_cdf)."""
+    col = alloc(n, HF_INT64)
+    _check(load().hf_fill_randcdf(col.handle, seed, cdf.handle),
+           "hf_fill_randcdf")
+    return col
 
 
 def groupby_hash_accum(keys: ColumnRef, vals: list, agg_op: int, H: int,

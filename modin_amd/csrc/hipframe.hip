@@ -697,6 +697,12 @@ __global__ void __launch_bounds__(BLOCK) k_conv_keys32(
   }
 }
 
+// Software-pipelined tile loop (round-2 probe winner, tools/ring_probe.hip:
+// P1 6.69 -> 5.52 ms on the 1e9-row north star).  Raw global loads for tile
+// t+1 are issued right after tile t's LDS staging — the registers are dead at
+// that point — so their ~700-cycle latency hides under t's writeout + barrier
+// instead of serializing at the top of the next iteration (the 81%
+// wave-parked stall of the round-1 kernel, profiles/r01b/sq_scatter.txt).
 template <int NV, int RPT, int BLK, int RL, bool K32>
 __global__ void __launch_bounds__(BLK) k_gb_scatter(
     const int64_t* __restrict__ keys, const unsigned* __restrict__ keys32,
@@ -734,46 +740,54 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
                                       // region (a static __shared__ would
                                       // shift the base off 16B — G17)
   const int64_t ntiles = (neven + TILE - 1) / TILE;
-  for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
-    const int64_t t0 = tile * TILE;
-    for (int t = threadIdx.x; t < nb; t += blockDim.x) it_cnt[t] = 0;
+
+  // only the K32-selected key array and the first NV value arrays are ever
+  // touched; the others constant-fold away (template params) and cost no
+  // registers
+  uint2 kraw[PAIRS];
+  longlong2 kraw64[PAIRS];
+  double2 vraw0[PAIRS];
+  double2 vraw1[PAIRS];
+  auto issue_loads = [&](int64_t tile) {
+#pragma unroll
+    for (int j = 0; j < PAIRS; ++j) {
+      const int64_t pr = tile * (TILE / 2) + (int64_t)j * BLK + threadIdx.x;
+      if (pr < npair_total) {
+        if (K32)
+          kraw[j] = reinterpret_cast<const uint2*>(keys32)[pr];
+        else
+          kraw64[j] = reinterpret_cast<const longlong2*>(keys)[pr];
+        if (NV > 0) vraw0[j] = reinterpret_cast<const double2*>(v0)[pr];
+        if (NV > 1) vraw1[j] = reinterpret_cast<const double2*>(v1)[pr];
+      }
+    }
+  };
+
+  int64_t tile = blockIdx.x;
+  if (tile < ntiles) issue_loads(tile);
+  for (int t = threadIdx.x; t < nb; t += blockDim.x) it_cnt[t] = 0;
+  for (; tile < ntiles; tile += gridDim.x) {
     int lb[RPT];
     unsigned lk[RPT];
     unsigned lr[RPT];
-    double lv0[RPT], lv1[RPT];
-    __syncthreads();
+    __syncthreads();  // it_cnt cleared (prologue or previous iteration)
 #pragma unroll
     for (int j = 0; j < PAIRS; ++j) {
-      const int64_t pr = (t0 >> 1) + (int64_t)j * BLK + threadIdx.x;
+      const int64_t pr = tile * (TILE / 2) + (int64_t)j * BLK + threadIdx.x;
       const int a = 2 * j, bslot = 2 * j + 1;
       lb[a] = lb[bslot] = -1;
       if (pr < npair_total) {
-        int64_t ka, kb;
-        if (K32) {
-          const uint2 kk32 = reinterpret_cast<const uint2*>(keys32)[pr];
-          ka = (int64_t)kk32.x;
-          kb = (int64_t)kk32.y;
-        } else {
-          const longlong2 kk = reinterpret_cast<const longlong2*>(keys)[pr];
-          ka = kk.x - key_min;
-          kb = kk.y - key_min;
-        }
-        double2 vv0{}, vv1{};
-        if (NV > 0) vv0 = reinterpret_cast<const double2*>(v0)[pr];
-        if (NV > 1) vv1 = reinterpret_cast<const double2*>(v1)[pr];
+        const int64_t ka = K32 ? (int64_t)kraw[j].x : kraw64[j].x - key_min;
+        const int64_t kb = K32 ? (int64_t)kraw[j].y : kraw64[j].y - key_min;
         if ((uint64_t)ka < (uint64_t)n_slots) {
           lb[a] = (int)(ka >> RL);
           lk[a] = (unsigned)(ka & ((1 << RL) - 1));
-          if (NV > 0) lv0[a] = vv0.x;
-          if (NV > 1) lv1[a] = vv1.x;
         } else {
           atomicAdd(err, 1ULL);
         }
         if ((uint64_t)kb < (uint64_t)n_slots) {
           lb[bslot] = (int)(kb >> RL);
           lk[bslot] = (unsigned)(kb & ((1 << RL) - 1));
-          if (NV > 0) lv0[bslot] = vv0.y;
-          if (NV > 1) lv1[bslot] = vv1.y;
         } else {
           atomicAdd(err, 1ULL);
         }
@@ -811,11 +825,16 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
       if (lb[j] >= 0) {
         const unsigned p = it_off[lb[j]] + lr[j];
         skey[p] = ((unsigned)lb[j] << 16) | lk[j];
-        if (NV > 0) sval0[p] = lv0[j];
-        if (NV > 1) sval1[p] = lv1[j];
+        if (NV > 0) sval0[p] = (j & 1) ? vraw0[j >> 1].y : vraw0[j >> 1].x;
+        if (NV > 1) sval1[p] = (j & 1) ? vraw1[j >> 1].y : vraw1[j >> 1].x;
       }
     }
-    __syncthreads();
+    __syncthreads();  // staging visible; it_cnt reads (scan) long done
+    // registers are dead — issue the NEXT tile's loads so they fly during
+    // this tile's writeout, and clear it_cnt for the next rank phase
+    const int64_t nxt = tile + gridDim.x;
+    if (nxt < ntiles) issue_loads(nxt);
+    for (int t = threadIdx.x; t < nb; t += blockDim.x) it_cnt[t] = 0;
     const int staged = (int)*s_total;
     for (int p = threadIdx.x; p < staged; p += blockDim.x) {
       const unsigned b = skey[p] >> 16;
@@ -824,7 +843,8 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
       if (NV > 0) r0[pos] = sval0[p];
       if (NV > 1) r1[pos] = sval1[p];
     }
-    __syncthreads();
+    // loop-top barrier orders the writeout's LDS reads against the next
+    // tile's staging writes
   }
 }
 
@@ -997,6 +1017,54 @@ __global__ void __launch_bounds__(BLOCK) k_gb_hash_accum(
         if (COUNTS) atomicAdd(&counts[(int64_t)c * stride_len + slot], 1ULL);
       }
     }
+  }
+}
+
+// Deterministic counter-based RNG fills (bench/test data generation on
+// device — bench.py's synthetic frames are born in HBM so the bench is GPU
+// work, not host numpy; oracle.rand_* mirrors these formulas bit-exactly
+// for the verify gate).  splitmix64 finalizer over (seed + index).
+__device__ __forceinline__ uint64_t rng_mix64(uint64_t x) {
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+
+__global__ void __launch_bounds__(BLOCK) k_fill_randint(
+    int64_t* __restrict__ p, int64_t n, uint64_t seed, int64_t lo,
+    uint64_t span) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride)
+    p[i] = lo + (int64_t)(rng_mix64(seed + (uint64_t)i) % span);
+}
+
+__global__ void __launch_bounds__(BLOCK) k_fill_randf64(
+    double* __restrict__ p, int64_t n, uint64_t seed) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride)
+    p[i] = (double)(rng_mix64(seed + (uint64_t)i) >> 11) *
+           (1.0 / 9007199254740992.0);
+}
+
+// inverse-CDF draw (zipf and friends): u uniform in [0,1), output =
+// np.searchsorted(cdf, u, side='right') on a sorted f64 cdf column
+__global__ void __launch_bounds__(BLOCK) k_fill_randcdf(
+    int64_t* __restrict__ p, int64_t n, uint64_t seed,
+    const double* __restrict__ cdf, int64_t m) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const double u = (double)(rng_mix64(seed + (uint64_t)i) >> 11) *
+                     (1.0 / 9007199254740992.0);
+    int64_t a = 0, b = m;          // first idx with cdf[idx] > u
+    while (a < b) {
+      const int64_t mid = (a + b) >> 1;
+      if (cdf[mid] <= u) a = mid + 1; else b = mid;
+    }
+    p[i] = a;
   }
 }
 
@@ -2588,12 +2656,14 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
                         hipMemcpyHostToDevice, g.stream));
   // host vectors must outlive the async H2D of pageable memory
   HF_HIP("gb_radix", hipStreamSynchronize(g.stream));
-  // P1 scatter: 12288-row tiles (512x24) for <=1 value column — 147 KB LDS,
-  // 1 block/CU of 8 waves (probe winner); 6144-row tiles for 2 columns
-  auto scat = [&](auto nvTag, auto rptTag) {
+  // P1 scatter (pipelined): 12288-row tiles as 1024x12 for <=1 value column
+  // — 147 KB LDS, 1 block/CU of 16 waves (round-2 probe: 1024x12 edges out
+  // 512x24 and both beat the round-1 kernel by ~17%); 6144-row tiles as
+  // 512x12 for 2 columns
+  auto scat = [&](auto nvTag, auto rptTag, auto blkTag) {
     constexpr int NVv = decltype(nvTag)::value;
     constexpr int RPTv = decltype(rptTag)::value;
-    constexpr int BLKv = 512;
+    constexpr int BLKv = decltype(blkTag)::value;
     const int64_t tile_sz = (int64_t)BLKv * RPTv;
     const int64_t ntiles = ((n & ~1LL) + tile_sz - 1) / tile_sz;
     const uint32_t sgrid =
@@ -2610,11 +2680,14 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
     });
   };
   rc = nvals == 0 ? scat(std::integral_constant<int, 0>{},
-                         std::integral_constant<int, 24>{})
+                         std::integral_constant<int, 12>{},
+                         std::integral_constant<int, 1024>{})
        : nvals == 1 ? scat(std::integral_constant<int, 1>{},
-                           std::integral_constant<int, 24>{})
+                           std::integral_constant<int, 12>{},
+                           std::integral_constant<int, 1024>{})
                     : scat(std::integral_constant<int, 2>{},
-                           std::integral_constant<int, 12>{});
+                           std::integral_constant<int, 12>{},
+                           std::integral_constant<int, 512>{});
   if (rc != HF_OK) return rc;
   // P2 aggregate, one column per launch
   const bool cnt = counts != 0;
@@ -2893,6 +2966,44 @@ int hf_fill_i64(uintptr_t dptr, int64_t value, int64_t n) {
   return timed_launch("fill_i64", [&] {
     hipLaunchKernelGGL(k_fill_i64, dim3((uint32_t)grid_for(n)), dim3(BLOCK), 0,
                        g.stream, (int64_t*)dptr, value, n);
+  });
+}
+
+int hf_fill_randint(hf_col* col, uint64_t seed, int64_t lo, int64_t hi) {
+  HF_NEED_INIT("hf_fill_randint");
+  if (!col || col->dtype != HF_INT64 || hi <= lo)
+    return set_err(HF_ERR_ARG, "hf_fill_randint", "int64 col, hi > lo");
+  if (col->len == 0) return HF_OK;
+  return timed_launch("fill_randint", [&] {
+    hipLaunchKernelGGL(k_fill_randint, dim3((uint32_t)grid_for(col->len)),
+                       dim3(BLOCK), 0, g.stream, (int64_t*)col->dptr,
+                       col->len, seed, lo, (uint64_t)(hi - lo));
+  });
+}
+
+int hf_fill_randf64(hf_col* col, uint64_t seed) {
+  HF_NEED_INIT("hf_fill_randf64");
+  if (!col || col->dtype != HF_FLOAT64)
+    return set_err(HF_ERR_ARG, "hf_fill_randf64", "float64 col");
+  if (col->len == 0) return HF_OK;
+  return timed_launch("fill_randf64", [&] {
+    hipLaunchKernelGGL(k_fill_randf64, dim3((uint32_t)grid_for(col->len)),
+                       dim3(BLOCK), 0, g.stream, (double*)col->dptr,
+                       col->len, seed);
+  });
+}
+
+int hf_fill_randcdf(hf_col* col, uint64_t seed, const hf_col* cdf) {
+  HF_NEED_INIT("hf_fill_randcdf");
+  if (!col || col->dtype != HF_INT64 || !cdf || cdf->dtype != HF_FLOAT64 ||
+      cdf->len <= 0)
+    return set_err(HF_ERR_ARG, "hf_fill_randcdf",
+                   "int64 col, non-empty float64 cdf");
+  if (col->len == 0) return HF_OK;
+  return timed_launch("fill_randcdf", [&] {
+    hipLaunchKernelGGL(k_fill_randcdf, dim3((uint32_t)grid_for(col->len)),
+                       dim3(BLOCK), 0, g.stream, (int64_t*)col->dptr,
+                       col->len, seed, (const double*)cdf->dptr, cdf->len);
   });
 }
 

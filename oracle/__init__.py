@@ -24,8 +24,12 @@ from .ops import (  # noqa: F401
     map_op,
     partitioned_groupby_agg,
     pick_splitters,
+    rand_cdf,
+    rand_f64,
+    rand_int,
     reduce_op,
     shuffle_dest,
     sort_perm,
     split_row_counts,
+    zipf_cdf,
 )

@@ -326,3 +326,48 @@ def pick_splitters(samples, world):
         return np.empty(0, dtype=np.int64)
     qs = [(i * s.size) // world for i in range(1, world)]
     return s[qs]
+
+
+# ---- deterministic RNG mirrors of the device fills (hf_fill_rand*) --------
+# splitmix64 finalizer over (seed + index), bit-exact vs hipframe.hip's
+# rng_mix64 — the verify gate regenerates device-born bench frames here.
+
+def _mix64(x: np.ndarray) -> np.ndarray:
+    with np.errstate(over="ignore"):
+        x = (x + np.uint64(0x9E3779B97F4A7C15))
+        x = (x ^ (x >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+        x = (x ^ (x >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+        return x ^ (x >> np.uint64(31))
+
+
+def _rand_bits(seed: int, n: int, offset: int = 0) -> np.ndarray:
+    idx = np.arange(offset, offset + n, dtype=np.uint64)
+    with np.errstate(over="ignore"):
+        return _mix64(np.uint64(seed % (1 << 64)) + idx)
+
+
+def rand_int(seed: int, n: int, lo: int, hi: int,
+             offset: int = 0) -> np.ndarray:
+    """Mirror of hf_fill_randint: lo + splitmix64(seed+i) % (hi-lo)."""
+    return (lo + (_rand_bits(seed, n, offset) %
+                  np.uint64(hi - lo)).astype(np.int64))
+
+
+def rand_f64(seed: int, n: int, offset: int = 0) -> np.ndarray:
+    """Mirror of hf_fill_randf64: (bits >> 11) * 2^-53 in [0,1)."""
+    return ((_rand_bits(seed, n, offset) >> np.uint64(11)).astype(np.float64)
+            * (1.0 / 9007199254740992.0))
+
+
+def rand_cdf(seed: int, n: int, cdf: np.ndarray,
+             offset: int = 0) -> np.ndarray:
+    """Mirror of hf_fill_randcdf: searchsorted(cdf, U[0,1), side='right')."""
+    u = rand_f64(seed, n, offset)
+    return np.searchsorted(cdf, u, side="right").astype(np.int64)
+
+
+def zipf_cdf(n_keys: int, s: float) -> np.ndarray:
+    """CDF of zipf(s) over ranks 1..n_keys (key k has p ~ 1/(k+1)^s) —
+    BASELINE §8d's skew variant; shared by bench.py and the device draw."""
+    w = 1.0 / np.power(np.arange(1, n_keys + 1, dtype=np.float64), s)
+    return np.cumsum(w) / w.sum()

@@ -1842,3 +1842,61 @@ def test_expanding_vs_pandas(npartitions):
                     got[c].to_numpy(), exp[c].to_numpy(), rtol=1e-12,
                     atol=1e-9, equal_nan=True,
                     err_msg=f"{op}/mp={mp}/{c}")
+
+
+def test_device_rng_matches_oracle_mirror(npartitions):
+    """hf_fill_randint/randf64/randcdf must be bit-exact vs the oracle's
+    splitmix64 mirrors (bench.py's verify gate depends on this)."""
+    n = 1_000_003
+    k = lib.get(lib.fill_randint(n, 42, 0, 1_000_000))
+    np.testing.assert_array_equal(k, oracle.rand_int(42, n, 0, 1_000_000))
+    assert k.min() >= 0 and k.max() < 1_000_000
+    v = lib.get(lib.fill_randf64(n, 777))
+    np.testing.assert_array_equal(v, oracle.rand_f64(777, n))
+    assert v.min() >= 0.0 and v.max() < 1.0
+    cdf = oracle.zipf_cdf(10_000, 1.2)
+    z = lib.get(lib.fill_randcdf(n, 5, lib.put(cdf)))
+    np.testing.assert_array_equal(z, oracle.rand_cdf(5, n, cdf))
+    assert z.min() >= 0 and z.max() < 10_000
+    # offset streams (bench shards chunks along one stream)
+    k2 = lib.get(lib.fill_randint(500, 42 + 250, 0, 1_000_000))
+    np.testing.assert_array_equal(
+        k2, oracle.rand_int(42 + 250, 500, 0, 1_000_000))
+
+
+def test_groupby_sum_on_device_generated_frame(npartitions):
+    """End-to-end bench shape at small n: device-generated frame through the
+    L1 groupby, checked against the oracle regeneration."""
+    from modin_amd.core.dataframe import HipDataframe
+    from modin_amd.core.partition import DeviceBlock, HipDataframePartition
+    from modin_amd.query_compiler import HipQueryCompiler
+
+    n, K = 2_000_000, 5000
+    chunks = oracle.split_row_counts(n, 3, 1)
+    parts = []
+    off = 0
+    for cn in chunks:
+        kc = lib.fill_randint(cn, 1000 + off, 0, K)
+        vc = lib.fill_randf64(cn, 2000 + off)
+        parts.append(HipDataframePartition(DeviceBlock({"k": kc, "v": vc},
+                                                       cn)))
+        off += cn
+    frame = HipDataframe(parts, pandas.RangeIndex(n), ["k", "v"], chunks,
+                         pandas.Series({"k": np.dtype(np.int64),
+                                        "v": np.dtype(np.float64)}))
+    df = mpd.DataFrame(query_compiler=HipQueryCompiler(frame))
+    out = df.groupby("k").sum().to_pandas()
+
+    acc = np.zeros(K)
+    seen = np.zeros(K, dtype=bool)
+    off = 0
+    for cn in chunks:
+        kk = oracle.rand_int(1000 + off, cn, 0, K)
+        vv = oracle.rand_f64(2000 + off, cn)
+        acc += np.bincount(kk, weights=vv, minlength=K)
+        seen[kk] = True
+        off += cn
+    exp_keys = np.nonzero(seen)[0]
+    np.testing.assert_array_equal(out.index.to_numpy(), exp_keys)
+    np.testing.assert_allclose(out["v"].to_numpy(), acc[exp_keys],
+                               rtol=RTOL, atol=1e-9)
