@@ -230,6 +230,37 @@ def exchange_splits(t, send_counts):
     return out, recv_counts
 
 
+def _allgather_varlen(t):
+    """All-gather a 1-D tensor of per-rank-varying length: size exchange +
+    pad-to-max all_gather (NCCL has no allgatherv) + trim.  Device tensor
+    collectives throughout — no host pickling.
+    Returns the list of per-rank tensors on the collective device."""
+    import torch
+    import torch.distributed as dist
+    P = world_size()
+    dev = _state["device"]
+    t = t.to(dev)
+    n = torch.tensor([t.numel()], dtype=torch.int64, device=dev)
+    sizes = [torch.zeros_like(n) for _ in range(P)]
+    dist.all_gather(sizes, n)
+    sizes = [int(s.item()) for s in sizes]
+    m = max(sizes + [1])
+    pad = torch.zeros(m, dtype=t.dtype, device=dev)
+    if t.numel():
+        pad[: t.numel()] = t
+    outs = [torch.empty(m, dtype=t.dtype, device=dev) for _ in range(P)]
+    dist.all_gather(outs, pad)
+    return [outs[i][: sizes[i]] for i in range(P)]
+
+
+def _gather_np_varlen(arr):
+    """numpy in, rank-order list of numpy out, via _allgather_varlen."""
+    import numpy as np
+    import torch
+    t = torch.from_numpy(np.ascontiguousarray(arr))
+    return [g.cpu().numpy() for g in _allgather_varlen(t)]
+
+
 def sample_splitters(local_sample):
     """world-1 key splitters from the gathered per-rank samples — the
     device form of the reference's RangePartitioning sampling
@@ -237,11 +268,8 @@ def sample_splitters(local_sample):
     quantiles).  Deterministic and identical on every rank: all_gather is
     rank-ordered and the quantile rule is pure."""
     import numpy as np
-    import torch.distributed as dist
-    gathered = [None] * world_size()
-    dist.all_gather_object(gathered, np.asarray(local_sample, dtype=np.int64))
-    alls = np.sort(np.concatenate([g for g in gathered if g is not None and
-                                   len(g)] or
+    gathered = _gather_np_varlen(np.asarray(local_sample, dtype=np.int64))
+    alls = np.sort(np.concatenate([g for g in gathered if len(g)] or
                                   [np.empty(0, dtype=np.int64)]))
     P = world_size()
     if alls.size == 0 or P <= 1:
@@ -250,22 +278,55 @@ def sample_splitters(local_sample):
     return alls[qs]
 
 
+class _CAIView:
+    """Zero-copy view of an hf device column for torch (the send side of the
+    RCCL exchange reads hipframe memory directly over xGMI — no staging
+    mirror).  The ColumnRef must outlive the view (caller scope does)."""
+
+    def __init__(self, ptr: int, length: int, typestr: str):
+        self.__cuda_array_interface__ = {
+            "shape": (length,),
+            "typestr": typestr,
+            "data": (ptr, False),
+            "strides": None,
+            "version": 2,
+        }
+
+
+def _as_torch_view(col):
+    """torch CUDA tensor aliasing an hf column (no copy), or None if this
+    torch build rejects the __cuda_array_interface__ handoff."""
+    import torch
+    from .core import lib
+    ts = "<i8" if col.dtype_code == lib.HF_INT64 else "<f8"
+    try:
+        t = torch.as_tensor(_CAIView(col.dptr(), col.length, ts),
+                            device=_state["device"])
+        if t.data_ptr() != col.dptr():
+            return None  # a copy was made onto the torch allocator — fine too
+        return t
+    except Exception:
+        return None
+
+
 def exchange_column(col, send_counts):
     """Move one hf column's P destination spans to their ranks
     (exchange_splits under the hood) and return the received rows as a new
-    hf column.  nccl: zero-host-copy — the column is mirrored into a
-    torch CUDA tensor by device address (hf_memcpy_dd) and travels over
-    xGMI; gloo (CPU test tier / single-GPU multi-rank validation): bounced
-    via host numpy."""
+    hf column.  nccl: the send buffer is a zero-copy torch view of the hf
+    column (RCCL reads it in place over xGMI) and the received rows land in
+    one device copy into hf-owned memory; gloo (CPU test tier / single-GPU
+    multi-rank validation): bounced via host numpy."""
     import numpy as np
     import torch
     from .core import lib
     tdt = torch.int64 if col.dtype_code == lib.HF_INT64 else torch.float64
     if _state["backend"] == "nccl":
-        t = torch.empty(col.length, dtype=tdt, device=_state["device"])
-        if col.length:
-            lib.memcpy_dd(t.data_ptr(), col.dptr(), 8 * col.length)
-            lib.sync()  # hf-stream copy must land before the collective
+        t = _as_torch_view(col) if col.length else None
+        if t is None:
+            t = torch.empty(col.length, dtype=tdt, device=_state["device"])
+            if col.length:
+                lib.memcpy_dd(t.data_ptr(), col.dptr(), 8 * col.length)
+        lib.sync()  # hf-stream writes must land before the collective
         out, recv_counts = exchange_splits(t, send_counts)
         recv = lib.alloc(out.numel(), col.dtype_code)
         if out.numel():
@@ -283,14 +344,13 @@ def allgather_groupby(keys_np, sums_np, counts_np):
     in rank order so every rank returns the identical replicated frame —
     the same output convention as the dense-table all-reduce path."""
     import numpy as np
-    import torch.distributed as dist
-    gathered = [None] * world_size()
-    dist.all_gather_object(gathered, (keys_np, sums_np, counts_np))
-    gk = np.concatenate([g[0] for g in gathered])
+    gk = np.concatenate(_gather_np_varlen(np.asarray(keys_np, np.int64)))
     nv = len(sums_np)
-    gs = [np.concatenate([g[1][c] for g in gathered]) for c in range(nv)]
-    gc = ([np.concatenate([g[2][c] for g in gathered]) for c in range(nv)]
-          if counts_np is not None else None)
+    gs = [np.concatenate(_gather_np_varlen(
+        np.asarray(sums_np[c], np.float64))) for c in range(nv)]
+    gc = ([np.concatenate(_gather_np_varlen(
+        np.asarray(counts_np[c], np.int64))) for c in range(nv)]
+        if counts_np is not None else None)
     return gk, gs, gc
 
 
@@ -299,20 +359,25 @@ def allgather_arrays(arrays):
     concatenation of each (the host form of ncclAllGatherv for the
     broadcast-join right side and the sorted-result merge)."""
     import numpy as np
+    return [np.concatenate(_gather_np_varlen(np.ascontiguousarray(a)))
+            for a in arrays]
+
+
+def allgather_lengths(local_n: int) -> list:
+    """Every rank's shard length, in rank order (one tiny i64 all_gather)."""
+    import torch
     import torch.distributed as dist
-    gathered = [None] * world_size()
-    dist.all_gather_object(gathered, arrays)
-    return [np.concatenate([g[i] for g in gathered])
-            for i in range(len(arrays))]
+    dev = _state["device"]
+    t = torch.tensor([int(local_n)], dtype=torch.int64, device=dev)
+    outs = [torch.zeros_like(t) for _ in range(world_size())]
+    dist.all_gather(outs, t)
+    return [int(o.item()) for o in outs]
 
 
 def global_row_base(local_n: int) -> int:
     """This rank's global row offset: sum of all earlier ranks' shard
     lengths (the SPMD frame is the rank-order concat of the shards)."""
-    import torch.distributed as dist
-    gathered = [None] * world_size()
-    dist.all_gather_object(gathered, int(local_n))
-    return sum(gathered[: rank()])
+    return sum(allgather_lengths(local_n)[: rank()])
 
 
 def maybe_allreduce_table(table) -> None:
