@@ -2512,9 +2512,78 @@ class HipDataframe:
                         DeviceBlock(newcols, block.length, dict(merged))))
         else:
             parts = [p for f in frames for p in f._partitions]
+        from .. import distributed as dist_mod
+        if dist_mod.is_active() and dist_mod.world_size() > 1:
+            blk_cats = (parts[0].block().cats if parts else {})
+            return self._concat_rows_exchange(frames, parts, cols, blk_cats)
         lengths = [ln for f in frames for ln in f._row_lengths]
         idx = pandas.Index(np.concatenate([np.asarray(f.index) for f in frames]))
         return HipDataframe(parts, idx, cols, lengths, self.dtypes)
+
+    def _concat_rows_exchange(self, frames, parts, cols, blk_cats):
+        """World>1 concat: re-shard so the GLOBAL row order equals pandas
+        concat order — all of frame 0 (ranks in order), then frame 1, … —
+        instead of the rank-local interleave (the round-1 known deviation).
+        Local rows are already ascending in global concat position, so the
+        per-destination spans are contiguous interval overlaps (host
+        arithmetic, no per-row planning) and ride exchange_column; the
+        received source-major pieces are restored to global order with one
+        device gather per column."""
+        from .. import distributed as dist_mod
+        P = dist_mod.world_size()
+        r = dist_mod.rank()
+        F = len(frames)
+        lens = [dist_mod.allgather_lengths(len(f)) for f in frames]
+        FB = np.cumsum([0] + [sum(L) for L in lens])
+        N = int(FB[-1])
+        runs = [(int(FB[fi]) + sum(lens[fi][:r]), lens[fi][r])
+                for fi in range(F)]
+        per = -(-N // P) if N else 0
+        T = [min(i * per, N) for i in range(P + 1)]
+        send_counts = [0] * P
+        for g0, L in runs:
+            for d in range(P):
+                lo, hi = max(g0, T[d]), min(g0 + L, T[d + 1])
+                if hi > lo:
+                    send_counts[d] += hi - lo
+        local_cols = {}
+        for m in cols:
+            cs = [p.block().columns[m] for p in parts]
+            local_cols[m] = cs[0] if len(cs) == 1 else lib.concat(cs)
+        recv = {m: dist_mod.exchange_column(local_cols[m], send_counts)
+                for m in cols}
+        idx_arrays = [np.asarray(f.index) for f in frames]
+        if any(a.dtype == object for a in idx_arrays):
+            raise lib.HfError("distributed concat: non-numeric index is a "
+                              "later round")
+        as_f64 = any(a.dtype.kind == "f" for a in idx_arrays)
+        idx_cat = np.concatenate(
+            [a.astype(np.float64 if as_f64 else np.int64)
+             for a in idx_arrays]) if idx_arrays else np.empty(0, np.int64)
+        ridx = dist_mod.exchange_column(lib.put(idx_cat), send_counts)
+        # received layout: src-major, frames in order within src; each piece
+        # is the (src, frame) shard ∩ my target interval
+        pieces = []
+        off = 0
+        for src in range(P):
+            for fi in range(F):
+                g0 = int(FB[fi]) + sum(lens[fi][:src])
+                L = lens[fi][src]
+                lo, hi = max(g0, T[r]), min(g0 + L, T[r + 1])
+                if hi > lo:
+                    pieces.append((lo, off, hi - lo))
+                    off += hi - lo
+        my_n = off
+        pieces.sort()
+        gather_idx = (np.concatenate(
+            [np.arange(o, o + L, dtype=np.int64) for (_, o, L) in pieces])
+            if pieces else np.empty(0, dtype=np.int64))
+        gidx = lib.put(gather_idx)
+        out_cols = {m: lib.gather(recv[m], gidx) for m in cols}
+        idx_vals = lib.get(lib.gather(ridx, gidx))
+        part = HipDataframePartition(DeviceBlock(out_cols, my_n, blk_cats))
+        return HipDataframe([part], pandas.Index(idx_vals), cols, [my_n],
+                            self.dtypes)
 
     # ---- sort (PandasDataframe.sort_by device form, dataframe.py:2742;
     #      SURVEY §8f.2): stable radix permutation + column gathers ----

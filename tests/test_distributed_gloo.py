@@ -326,3 +326,98 @@ def test_gloo_world2_exchange_logic():
         errs.append(q.get())
     assert not errs, "\n".join(errs)
     assert all(p.exitcode == 0 for p in procs)
+
+
+def _concat_worker(rank, world, port, fail_q):
+    """World>1 concat must produce the GLOBAL pandas concat order (frame 0
+    across ranks, then frame 1, ...) — the exchange-based concat, exercised
+    through the REAL dataframe.py composition code on the numpy lib mock."""
+    try:
+        os.environ.update({
+            "RANK": str(rank), "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        })
+        import pandas
+        import modin_amd.distributed as dist_mod
+        from tests import mocklib
+
+        class _RawPatch:
+            def setattr(self, obj, name, fn):
+                setattr(obj, name, fn)
+
+        mocklib.install(_RawPatch())
+        assert dist_mod.init_from_env(backend="gloo", gpu=False)
+        from modin_amd.core import lib
+        from modin_amd.core.dataframe import HipDataframe
+        from modin_amd.core.partition import (DeviceBlock,
+                                              HipDataframePartition)
+
+        rng = np.random.default_rng(31)  # same stream on all ranks
+        nA, nB = 1000, 700
+        A_k = rng.integers(0, 50, nA).astype(np.int64)
+        A_v = rng.random(nA)
+        B_k = rng.integers(0, 50, nB).astype(np.int64)
+        B_v = rng.random(nB)
+
+        def shard(arr):
+            counts = oracle.split_row_counts(len(arr), world, 1)
+            while len(counts) < world:
+                counts.append(0)
+            offs = np.cumsum([0] + counts)
+            return arr[offs[rank]:offs[rank + 1]], offs[rank]
+
+        def frame(k, v, idx0):
+            n = len(k)
+            block = DeviceBlock({"k": lib.put(k), "v": lib.put(v)}, n)
+            return HipDataframe(
+                [HipDataframePartition(block)],
+                pandas.RangeIndex(idx0, idx0 + n), ["k", "v"], [n],
+                pandas.Series({"k": np.dtype(np.int64),
+                               "v": np.dtype(np.float64)}))
+
+        (ak, aoff) = shard(A_k)
+        (av, _) = shard(A_v)
+        (bk, boff) = shard(B_k)
+        (bv, _) = shard(B_v)
+        fa = frame(ak, av, aoff)
+        fb = frame(bk, bv, boff)
+        out = fa.concat_rows([fb])
+
+        # gather every rank's shard in rank order == global pandas concat
+        got_k = np.concatenate(dist_mod._gather_np_varlen(
+            lib.get(out._partitions[0].block().columns["k"])))
+        got_v = np.concatenate(dist_mod._gather_np_varlen(
+            lib.get(out._partitions[0].block().columns["v"])))
+        got_idx = np.concatenate(dist_mod._gather_np_varlen(
+            np.asarray(out.index).astype(np.int64)))
+        exp = pandas.concat([
+            pandas.DataFrame({"k": A_k, "v": A_v}),
+            pandas.DataFrame({"k": B_k, "v": B_v}),
+        ])
+        np.testing.assert_array_equal(got_k, exp["k"].to_numpy())
+        np.testing.assert_allclose(got_v, exp["v"].to_numpy(), rtol=0)
+        np.testing.assert_array_equal(got_idx, exp.index.to_numpy())
+        dist_mod.shutdown()
+    except Exception as e:  # pragma: no cover
+        import traceback
+        fail_q.put(f"rank {rank}: {e}\n{traceback.format_exc()}")
+
+
+@pytest.mark.timeout(120)
+@pytest.mark.parametrize("world", [2, 3])
+def test_gloo_concat_global_order(world):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29530 + world
+    procs = [ctx.Process(target=_concat_worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=110)
+    errs = []
+    while not q.empty():
+        errs.append(q.get())
+    assert not errs, "\n".join(errs)
+    assert all(p.exitcode == 0 for p in procs)
