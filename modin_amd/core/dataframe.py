@@ -12,6 +12,8 @@ performs in ``PartitionManager.get_indices`` (partition_manager.py:1220).
 
 from __future__ import annotations
 
+import os
+
 import numpy as np
 import pandas
 
@@ -2093,8 +2095,145 @@ class HipDataframe:
         part = HipDataframePartition(DeviceBlock(cols, n, cats))
         return HipDataframe([part], pandas.RangeIndex(n), names, [n], dts)
 
+    def _binned_merge(self, other: "HipDataframe", on: str, how: str,
+                      key_f64: bool, n_bins: int = 0) -> "HipDataframe":
+        """Range-binned (co-shuffled) merge for giant right tables — the
+        device form of the reference's range_partitioning_merge
+        (storage_formats/pandas/merge.py:39 -> dataframe.py:4087): both
+        sides are binned by splitters drawn from the right side's sorted
+        DISTINCT keys (hf_shuffle_dest), each bin runs the existing dense
+        CSR broadcast join, and pandas' left-major match order is restored
+        by one stable sort over a hidden global-left-row column (the
+        composition pinned in tests/test_host_logic.py:251).  Removes the
+        2^27-distinct-right-keys cap; inner/left on non-dictionary keys
+        (right joins arrive here pre-swapped as left).  At world>1 the
+        splitters are made identical on every rank (sample all-gather), so
+        each bin's broadcast join gathers exactly the global right rows of
+        its key range."""
+        from .. import distributed as dist_mod
+
+        def concat_col(frame, name):
+            cs = [p.block().columns[name] for p in frame._partitions]
+            return cs[0] if len(cs) == 1 else lib.concat(cs)
+
+        def enc_key(col):
+            return lib.ordered_i64(lib.cast_f64(col)) if key_f64 else col
+
+        lk = enc_key(concat_col(self, on))
+        rk = enc_key(concat_col(other, on))
+        # splitters: quantiles of the right side's sorted distinct keys
+        if rk.length:
+            perm = lib.sort_perm(rk)
+            uniq, _u1, _u2, n_uniq = lib.groupby_sorted(
+                lib.gather(rk, perm), [], lib.AGG_SUM, False)
+        else:
+            uniq, n_uniq = lib.put(np.empty(0, dtype=np.int64)), 0
+        n_uniq_glob = int(n_uniq)
+        if dist_mod.is_active():
+            n_uniq_glob = sum(dist_mod.allgather_lengths(int(n_uniq)))
+        B = n_bins or max(2, -(-n_uniq_glob // (1 << 26)))
+        S = min(int(n_uniq), 4096)
+        if S:
+            sidx = np.linspace(0, int(n_uniq) - 1, S).astype(np.int64)
+            sample = lib.get(lib.gather(uniq, lib.put(sidx)))
+        else:
+            sample = np.empty(0, dtype=np.int64)
+        if dist_mod.is_active():
+            # identical splitters on every rank (each bin's broadcast join
+            # is a collective)
+            sample = np.sort(dist_mod.allgather_arrays([sample])[0])
+        if sample.size == 0:
+            splitters = np.empty(0, dtype=np.int64)
+            B = 1
+        else:
+            qs = [(i * sample.size) // B for i in range(1, B)]
+            splitters = np.unique(sample[qs])
+            B = splitters.size + 1
+        ld = lib.shuffle_dest(lk, splitters)
+        rd = lib.shuffle_dest(rk, splitters)
+
+        lcats = (self._partitions[0].block().cats
+                 if self._partitions else {})
+        rcats = (other._partitions[0].block().cats
+                 if other._partitions else {})
+        LROW = "__hf_lrow__"
+        lcols = {c: concat_col(self, c) for c in self.columns}
+        rcols = {c: concat_col(other, c) for c in other.columns}
+        nl = lk.length
+        lrow = lib.filter_iota(lib.filter_plan(
+            lib.compare_scalar(lib.CMP_GE, ld, 0.0)), 0) if nl else \
+            lib.put(np.empty(0, dtype=np.int64))
+
+        bin_frames = []
+        for b in range(B):
+            lplan = lib.filter_plan(
+                lib.compare_scalar(lib.CMP_EQ, ld, float(b)))
+            rplan = lib.filter_plan(
+                lib.compare_scalar(lib.CMP_EQ, rd, float(b)))
+            any_rows = lplan.n_kept or (dist_mod.is_active()
+                                        and rplan.n_kept)
+            if not any_rows and not dist_mod.is_active():
+                continue
+            lbin = {c: lib.filter_apply(lplan, lcols[c])
+                    for c in self.columns}
+            lbin[LROW] = lib.filter_apply(lplan, lrow)
+            rbin = {c: lib.filter_apply(rplan, rcols[c])
+                    for c in other.columns}
+            Lb = HipDataframe(
+                [HipDataframePartition(
+                    DeviceBlock(lbin, lplan.n_kept, dict(lcats)))],
+                pandas.RangeIndex(lplan.n_kept),
+                list(self.columns) + [LROW], [lplan.n_kept],
+                pandas.Series({**{c: self.dtypes[c] for c in self.columns},
+                               LROW: np.dtype(np.int64)}))
+            Rb = HipDataframe(
+                [HipDataframePartition(
+                    DeviceBlock(rbin, rplan.n_kept, dict(rcats)))],
+                pandas.RangeIndex(rplan.n_kept),
+                list(other.columns), [rplan.n_kept],
+                pandas.Series({c: other.dtypes[c] for c in other.columns}))
+            bin_frames.append(Lb.broadcast_join(Rb, on, how, _no_bin=True))
+        if not bin_frames:
+            empty = self.take_row_range(0, 0)
+            return empty.broadcast_join(other.take_row_range(0, 0), on, how)
+        # unify per-column dtypes across bins (a bin with NaN fills promoted
+        # int64 right columns to float64 — the pandas rule is global)
+        out_columns = list(bin_frames[0].columns)
+        final_dt = {}
+        for c in out_columns:
+            dts = [f.dtypes[c] for f in bin_frames]
+            final_dt[c] = (np.dtype(np.float64)
+                           if any(d == np.dtype(np.float64) for d in dts)
+                           else dts[0])
+        cat_map = {}
+        for f in bin_frames:
+            cat_map.update(f._partitions[0].block().cats)
+        merged = {}
+        for c in out_columns:
+            pieces = []
+            for f in bin_frames:
+                col = f._partitions[0].block().columns[c]
+                if (final_dt[c] == np.dtype(np.float64)
+                        and col.dtype_code == lib.HF_INT64
+                        and c not in cat_map):
+                    col = lib.cast_f64(col)
+                pieces.append(col)
+            merged[c] = pieces[0] if len(pieces) == 1 else lib.concat(pieces)
+        order = lib.sort_perm(merged[LROW])
+        n_out = merged[LROW].length
+        out_cols = {c: lib.gather(merged[c], order) for c in out_columns
+                    if c != LROW}
+        out_columns = [c for c in out_columns if c != LROW]
+        part = HipDataframePartition(
+            DeviceBlock(out_cols, n_out,
+                        {c: v for c, v in cat_map.items() if c != LROW}))
+        return HipDataframe(
+            [part], pandas.RangeIndex(n_out), out_columns, [n_out],
+            pandas.Series({c: final_dt[c] for c in out_columns}))
+
     def broadcast_join(self, other: "HipDataframe", on: str,
-                       how: str = "inner") -> "HipDataframe":
+                       how: str = "inner",
+                       _no_bin: bool = False) -> "HipDataframe":
         if on not in self.columns or on not in other.columns:
             raise lib.HfError(f"merge: key column {on!r} missing")
         if how not in ("inner", "left", "outer"):
@@ -2131,6 +2270,15 @@ class HipDataframe:
 
         def enc_key(col):
             return lib.ordered_i64(lib.cast_f64(col)) if key_f64 else col
+
+        # test hook: MODIN_AMD_MERGE_BINS=B forces the co-shuffled binned
+        # merge at any size (parity tests exercise the giant-right path on
+        # small data)
+        force_bins = int(os.environ.get("MODIN_AMD_MERGE_BINS", "0"))
+        if (force_bins > 0 and not _no_bin and on not in lcats
+                and how in ("inner", "left")):
+            return self._binned_merge(other, on, how, key_f64,
+                                      n_bins=force_bins)
 
         rkeys = enc_key(rkeys)
         if on in lcats:
@@ -2173,9 +2321,13 @@ class HipDataframe:
             uniq, _su, _cu, n_uniq = lib.groupby_sorted(
                 skeys, [], lib.AGG_SUM, False)
             if n_uniq > (1 << 27):
+                if key_cats is None and how in ("inner", "left"):
+                    # giant right: range-binned (co-shuffled) merge
+                    return self._binned_merge(other, on, how, key_f64)
                 raise lib.HfError(
-                    "merge: more than 2^27 distinct right keys (the "
-                    "co-shuffled giant-right merge is a later round)")
+                    "merge: more than 2^27 distinct right keys with "
+                    f"how={how!r}/string keys (binned merge covers "
+                    "inner/left on non-dictionary keys)")
             rkeys = lib.search_sorted(rkeys, uniq)
             kmin, n_slots = 0, max(n_uniq, 1)
         # cache the build side on the right frame: its columns are immutable,

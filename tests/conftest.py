@@ -16,6 +16,9 @@ def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: test requires a visible MI355X (run via gpurun)"
     )
+    config.addinivalue_line(
+        "markers", "slow: multi-GB / multi-minute case (run explicitly)"
+    )
 
 
 @pytest.fixture(scope="session")
