@@ -758,6 +758,122 @@ class HipDataframe:
             out[name] = vals
         return out
 
+    def _shuffle_frame_by_key(self, by, with_pos: bool = False):
+        """Re-shard the whole frame by the groupby key's sampled ranges
+        (hf_shuffle_dest + exchange_column): afterwards every key range —
+        hence every GROUP — lives on exactly one rank, so the single-rank
+        engines run unchanged on the local shard (the frame-level form of
+        partition_manager._groupby_shuffle; reference shuffle_partitions,
+        partition_manager.py:1937).  with_pos adds the original global row
+        position as a hidden column (transforms route results back)."""
+        from .. import distributed as dist_mod
+        P = dist_mod.world_size()
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+
+        def concat_col(m):
+            cs = [p.block().columns[m] for p in self._partitions]
+            return cs[0] if len(cs) == 1 else lib.concat(cs)
+
+        kcol = concat_col(by)
+        ekey, _ = self._effective_sort_key(kcol, by in blk_cats, True)
+        n = ekey.length
+        S = min(n, 4096)
+        if S:
+            sidx = np.linspace(0, n - 1, S).astype(np.int64)
+            sample = lib.get(lib.gather(ekey, lib.put(sidx)))
+        else:
+            sample = np.empty(0, dtype=np.int64)
+        splitters = dist_mod.sample_splitters(sample)
+        dest = lib.shuffle_dest(ekey, splitters)
+        names = list(self.columns)
+        cols_cat = {m: concat_col(m) for m in names}
+        POS = "\x00gpos\x00"
+        if with_pos:
+            base = dist_mod.global_row_base(n)
+            cols_cat[POS] = None  # filled per plan below
+        send_cols = {m: [] for m in cols_cat}
+        send_counts = []
+        for d in range(P):
+            plan = lib.filter_plan(
+                lib.compare_scalar(lib.CMP_EQ, dest, float(d)))
+            send_counts.append(plan.n_kept)
+            for m in names:
+                send_cols[m].append(lib.filter_apply(plan, cols_cat[m]))
+            if with_pos:
+                send_cols[POS].append(lib.filter_iota(plan, base))
+        recv = {m: dist_mod.exchange_column(lib.concat(send_cols[m]),
+                                            send_counts)
+                for m in send_cols}
+        pos_col = recv.pop(POS, None)
+        ln = recv[names[0]].length if names else 0
+        part = HipDataframePartition(DeviceBlock(recv, ln, dict(blk_cats)))
+        shuf = HipDataframe([part], pandas.RangeIndex(ln), names, [ln],
+                            self.dtypes)
+        return (shuf, pos_col) if with_pos else shuf
+
+    def _route_back_by_pos(self, res: "HipDataframe",
+                           pos_col) -> "HipDataframe":
+        """Return shuffled same-length results to their origin ranks: each
+        result row travels to the rank owning its original global position
+        (hf_shuffle_dest over the rank boundaries), then an inverse-
+        permutation scatter restores the original local order.  `self` is
+        the pre-shuffle frame (its shard lengths define the boundaries)."""
+        from .. import distributed as dist_mod
+        P = dist_mod.world_size()
+        r = dist_mod.rank()
+        lens = dist_mod.allgather_lengths(len(self))
+        bounds = np.cumsum(lens)[:-1].astype(np.int64)  # P-1 splitters
+        dest = lib.shuffle_dest(pos_col, bounds)
+        names = list(res.columns)
+        blk = res._partitions[0].block()
+        send_cols = {m: [] for m in names}
+        send_pos, send_counts = [], []
+        for d in range(P):
+            plan = lib.filter_plan(
+                lib.compare_scalar(lib.CMP_EQ, dest, float(d)))
+            send_counts.append(plan.n_kept)
+            send_pos.append(lib.filter_apply(plan, pos_col))
+            for m in names:
+                send_cols[m].append(lib.filter_apply(plan, blk.columns[m]))
+        recv = {m: dist_mod.exchange_column(lib.concat(send_cols[m]),
+                                            send_counts)
+                for m in names}
+        rpos = dist_mod.exchange_column(lib.concat(send_pos), send_counts)
+        my_base = sum(lens[:r])
+        lpos = lib.map_scalar(lib.MAP_SUB, rpos, my_base)
+        out_cols = {m: lib.scatter(recv[m], lpos) for m in names}
+        ln = lens[r]
+        part = HipDataframePartition(
+            DeviceBlock(out_cols, ln, dict(blk.cats)))
+        return HipDataframe([part], pandas.RangeIndex(ln), names, [ln],
+                            res.dtypes)
+
+    @staticmethod
+    def _replicate_result_frame(res: "HipDataframe") -> "HipDataframe":
+        """All-gather the per-rank disjoint (ascending-range) result shards
+        in rank order so every rank returns the identical replicated frame
+        — the output convention of the dense-table path."""
+        from .. import distributed as dist_mod
+        names = list(res.columns)
+        blk = res._partitions[0].block() if res._partitions else None
+        idx_vals = np.asarray(res.index)
+        if idx_vals.dtype == object:
+            raise lib.HfError("distributed groupby result with a "
+                              "non-numeric index is a later round")
+        datas = [lib.get(blk.columns[m]) for m in names] if blk else \
+            [np.empty(0) for _ in names]
+        as_f = idx_vals.dtype.kind == "f"
+        g = dist_mod.allgather_arrays(
+            datas + [idx_vals.astype(np.float64 if as_f else np.int64)])
+        cols = {m: lib.put(g[i]) for i, m in enumerate(names)}
+        idx = pandas.Index(g[-1])
+        ln = len(idx)
+        cats = dict(blk.cats) if blk else {}
+        return HipDataframe(
+            [HipDataframePartition(DeviceBlock(cols, ln, cats))], idx,
+            names, [ln], res.dtypes)
+
     def groupby_median(self, by) -> "HipDataframe":
         return self.groupby_quantile(by, 0.5)
 
@@ -767,10 +883,22 @@ class HipDataframe:
         bracketing elements and interpolate linearly — offsets from
         groupby size, non-NaN counts from groupby count (pandas
         nanquantile per group)."""
-        from ..distributed import is_active
-        if is_active():
-            raise lib.HfError(
-                "distributed groupby.median is a later round")
+        from .. import distributed as dist_mod
+        if dist_mod.is_active():
+            key = by if isinstance(by, str) else \
+                (by[0] if len(by) == 1 else None)
+            if key is None:
+                # multi-key: combine first (global span), then shuffle on
+                # the combined key through the single-key path below
+                cf, decode = self._combined_key_frame(list(by))
+                keep = [c for c in cf.columns if c not in by]
+                res = cf.take_columns(keep).groupby_quantile(self.KEYCOL, q)
+                res._index = decode(np.asarray(res.index).astype(np.int64))
+                return res
+            shuf = self._shuffle_frame_by_key(key)
+            with dist_mod.local_mode():
+                res = shuf.groupby_quantile(key, q)
+            return self._replicate_result_frame(res)
         if isinstance(by, (list, tuple)):
             if len(by) == 1:
                 by = by[0]
@@ -839,10 +967,21 @@ class HipDataframe:
         each group (pandas skips NaN) — one groupby-min/max of a position
         column per value column, then a device gather; groups with no
         non-NaN value fill NaN (float) / code −1 (dict)."""
-        from ..distributed import is_active
-        if is_active():
-            raise lib.HfError(
-                "distributed groupby.first/last is a later round")
+        from .. import distributed as dist_mod
+        if dist_mod.is_active():
+            key = by if isinstance(by, str) else \
+                (by[0] if len(by) == 1 else None)
+            if key is None:
+                cf, decode = self._combined_key_frame(list(by))
+                keep = [c for c in cf.columns if c not in by]
+                res = cf.take_columns(keep).groupby_firstlast(self.KEYCOL,
+                                                              last)
+                res._index = decode(np.asarray(res.index).astype(np.int64))
+                return res
+            shuf = self._shuffle_frame_by_key(key)
+            with dist_mod.local_mode():
+                res = shuf.groupby_firstlast(key, last)
+            return self._replicate_result_frame(res)
         if isinstance(by, (list, tuple)):
             if len(by) == 1:
                 by = by[0]
@@ -1257,10 +1396,39 @@ class HipDataframe:
         Reference semantics pinned against pandas 2.3.3 in-container
         (NaN keys -> NaN in every transform incl. cumcount; NaN values
         stay NaN and don't advance cum* state; rank na_option='keep')."""
-        from ..distributed import is_active
-        if is_active():
-            raise lib.HfError("distributed groupby transforms are a later "
-                              "round")
+        from .. import distributed as dist_mod
+        if dist_mod.is_active():
+            # shuffle by the FIRST key's ranges (every multi-key group
+            # shares its first key, so groups stay rank-local), run the
+            # single-rank transform locally, route rows back to their
+            # origin rank by global position (reference: the same
+            # shuffle_partitions recipe, applied row-preserving)
+            by_l = [by] if isinstance(by, str) else list(by)
+            shuf, pos_col = self._shuffle_frame_by_key(by_l[0],
+                                                       with_pos=True)
+            with dist_mod.local_mode():
+                res = shuf.groupby_transform(by, how, ascending=ascending,
+                                             method=method, periods=periods,
+                                             dropna=dropna)
+            if how == "ngroup":
+                # global group ordinal = local ordinal + #groups on lower
+                # ranks (rank ranges ascend, pandas numbers sorted groups)
+                col = res._partitions[0].block().columns[how]
+                r = lib.reduce(lib.cast_f64(col)) if col.length else None
+                n_local = int(r.mx) + 1 if r is not None and r.count else 0
+                offs = dist_mod.allgather_lengths(n_local)
+                my_off = sum(offs[: dist_mod.rank()])
+                if my_off:
+                    newc = lib.map_scalar(lib.MAP_ADD, col, float(my_off))
+                    blk = res._partitions[0].block()
+                    cols2 = dict(blk.columns)
+                    cols2[how] = newc
+                    res = HipDataframe(
+                        [HipDataframePartition(
+                            DeviceBlock(cols2, blk.length, blk.cats))],
+                        res._index, list(res.columns), [blk.length],
+                        res.dtypes)
+            return self._route_back_by_pos(res, pos_col)
         by_list = [by] if isinstance(by, str) else list(by)
         for b in by_list:
             if b not in self.columns:

@@ -69,6 +69,23 @@ def init_from_env(backend=None, gpu: bool = True):
     return True
 
 
+class local_mode:
+    """Temporarily mark the distributed state inactive: after a key-range
+    shuffle each rank owns disjoint groups, so the LOCAL single-rank
+    engine must run without the dense-table all-reduce re-merging them
+    (the shuffle-then-local recipe of _groupby_shuffle, applied at the
+    frame level for median/quantile/first/last/transforms)."""
+
+    def __enter__(self):
+        self._prev = _state["active"]
+        _state["active"] = False
+        return self
+
+    def __exit__(self, *exc):
+        _state["active"] = self._prev
+        return False
+
+
 def shutdown():
     import torch.distributed as dist
     if dist.is_initialized():

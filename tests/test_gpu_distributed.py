@@ -112,6 +112,92 @@ def _worker(rank, world, port, fail_q):
             np.testing.assert_array_equal(gathered[2],
                                           exp["v"].to_numpy(),
                                           err_msg=f"sort asc={asc} v")
+        # ---- distributed median/quantile/first/last (shuffle-then-local,
+        # round-2 lift): every rank returns the identical replicated
+        # result equal to pandas on the union ----
+        rngq = np.random.default_rng(45)
+        nq = 30_000
+        qk = rngq.integers(0, 500, nq).astype(np.int64)
+        qv = rngq.random(nq)
+        qv[rngq.random(nq) < 0.1] = np.nan
+        qw = rngq.integers(-50, 50, nq).astype(np.int64)
+        qpdf = pandas.DataFrame({"k": qk, "v": qv, "w": qw})
+        qlo, qhi = rank * nq // world, (rank + 1) * nq // world
+        qdf = mpd.DataFrame(qpdf.iloc[qlo:qhi].reset_index(drop=True))
+        for op in ("median", "first", "last"):
+            out = getattr(qdf.groupby("k"), op)().to_pandas()
+            expect = getattr(qpdf.groupby("k"), op)()
+            np.testing.assert_array_equal(out.index.to_numpy(),
+                                          expect.index.to_numpy(),
+                                          err_msg=f"{op} keys")
+            for c in ("v", "w"):
+                np.testing.assert_allclose(
+                    out[c].to_numpy(dtype=np.float64),
+                    expect[c].to_numpy(dtype=np.float64), rtol=1e-12,
+                    atol=1e-12, equal_nan=True, err_msg=f"{op}/{c}")
+
+        # ---- distributed transforms (shuffle + local + route-back):
+        # row-aligned results on each rank's own shard ----
+        tdf_p = qpdf.iloc[qlo:qhi].reset_index(drop=True)
+        tdf = mpd.DataFrame(tdf_p)
+        for how in ("cumsum", "cumcount", "ngroup", "rank", "shift"):
+            gb = tdf.groupby("k")
+            egb = qpdf.groupby("k")
+            if how == "cumsum":
+                got = gb.cumsum().to_pandas()
+                exp = qpdf.groupby("k").cumsum().iloc[qlo:qhi]
+                for c in ("v", "w"):
+                    np.testing.assert_allclose(
+                        got[c].to_numpy(dtype=np.float64),
+                        exp[c].to_numpy(dtype=np.float64), rtol=1e-12,
+                        atol=1e-9, equal_nan=True,
+                        err_msg=f"transform cumsum/{c}")
+            elif how == "cumcount":
+                got = gb.cumcount().to_pandas()
+                exp = egb.cumcount().iloc[qlo:qhi]
+                np.testing.assert_array_equal(
+                    np.asarray(got).reshape(-1),
+                    exp.to_numpy(), err_msg="transform cumcount")
+            elif how == "ngroup":
+                got = gb.ngroup().to_pandas()
+                exp = egb.ngroup().iloc[qlo:qhi]
+                np.testing.assert_array_equal(
+                    np.asarray(got).reshape(-1), exp.to_numpy(),
+                    err_msg="transform ngroup")
+            elif how == "rank":
+                got = gb.rank().to_pandas()
+                exp = egb.rank().iloc[qlo:qhi]
+                for c in ("v", "w"):
+                    np.testing.assert_allclose(
+                        got[c].to_numpy(), exp[c].to_numpy(), rtol=1e-12,
+                        equal_nan=True, err_msg=f"transform rank/{c}")
+            else:
+                got = gb.shift(1).to_pandas()
+                exp = egb.shift(1).iloc[qlo:qhi]
+                for c in ("v", "w"):
+                    np.testing.assert_allclose(
+                        got[c].to_numpy(dtype=np.float64),
+                        exp[c].to_numpy(dtype=np.float64), rtol=0,
+                        equal_nan=True, err_msg=f"transform shift/{c}")
+
+        # ---- distributed concat: global pandas row order ----
+        cpdf_a = qpdf.iloc[qlo:qhi].reset_index(drop=True)
+        cpdf_b = spdf.iloc[slo:shi].reset_index(drop=True)[["k", "v"]]
+        ca = mpd.DataFrame(cpdf_a[["k", "v"]])
+        cb = mpd.DataFrame(cpdf_b)
+        cc = mpd.concat([ca, cb])
+        blk = cc._query_compiler._modin_frame._partitions[0].block()
+        from modin_amd.core import lib as hl
+        gathered = dist_mod.allgather_arrays(
+            [hl.get(blk.columns["k"]), hl.get(blk.columns["v"])])
+        exp_cat = pandas.concat([qpdf[["k", "v"]],
+                                 spdf[["k", "v"]]])
+        np.testing.assert_array_equal(gathered[0],
+                                      exp_cat["k"].to_numpy(),
+                                      err_msg="concat k order")
+        np.testing.assert_allclose(gathered[1], exp_cat["v"].to_numpy(),
+                                   rtol=0, equal_nan=True,
+                                   err_msg="concat v order")
         dist_mod.shutdown()
     except Exception as e:  # pragma: no cover
         import traceback
