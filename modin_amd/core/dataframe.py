@@ -121,9 +121,12 @@ class HipDataframe:
     # ---- metadata ----
     @property
     def index(self) -> pandas.Index:
-        if isinstance(self._index, DeviceIndex):
-            return self._index.materialize()
-        return self._index
+        idx = (self._index.materialize()
+               if isinstance(self._index, DeviceIndex) else self._index)
+        dt = getattr(self, "_index_dtype", None)
+        if dt is not None and idx.dtype != dt:
+            idx = idx.astype(dt)
+        return idx
 
     def __len__(self):
         return sum(self._row_lengths)
@@ -144,6 +147,23 @@ class HipDataframe:
         for name in df.columns:
             dt = df.dtypes[name]
             if dt in (np.dtype(np.int64), np.dtype(np.float64)):
+                continue
+            if isinstance(dt, pandas.DatetimeTZDtype):
+                raise lib.HfError(
+                    f"column {name!r}: timezone-aware datetimes are a "
+                    "later round (convert with tz_localize(None))")
+            if np.issubdtype(dt, np.datetime64):
+                # typed-column tag (SURVEY §8f.3 gateway): the device
+                # stores the int64 ns view; every sort/groupby/merge/
+                # compare path runs on int64, to_pandas restores the tag
+                if df[name].isna().any():
+                    raise lib.HfError(
+                        f"column {name!r}: NaT values are a later round")
+                if edf is df:
+                    edf = df.copy()
+                edf[name] = df[name].to_numpy().astype(
+                    "datetime64[ns]").view(np.int64)
+                dtypes[name] = np.dtype("datetime64[ns]")
                 continue
             if (dt == np.dtype(object)
                     or isinstance(dt, pandas.CategoricalDtype)
@@ -439,27 +459,32 @@ class HipDataframe:
                         fparts.append(HipDataframePartition(
                             DeviceBlock(cols, block.length, block.cats)))
                 parts = fparts
-        keys, sums, counts, n = self._partition_mgr_cls.groupby_reduce(
-            parts, by, val_names, want_counts, agg_op
-        )
-        # pandas dtype preservation: sum/min/max of an int64 column stays
-        # int64 (the f64 accumulators are exact — ingestion guards |x|<2^53,
-        # and the per-group sum is checked below before the cast back).
+        keys, sums, counts, n, biases = \
+            self._partition_mgr_cls.groupby_reduce(
+                parts, by, val_names, want_counts, agg_op)
+        # pandas dtype preservation: sum/min/max of an int64 (or the int64
+        # view of a datetime64) column stays typed — the f64 accumulators
+        # are exact below 2^53, huge min/max ride the per-column bias the
+        # partition manager applied (added back INT-side here).
         int_vals = {n_ for n_ in val_names
-                    if self.dtypes[n_] == np.dtype(np.int64)}
+                    if (self.dtypes[n_] == np.dtype(np.int64)
+                        or np.issubdtype(self.dtypes[n_], np.datetime64))}
 
         def back_to_int(name, col):
             if name not in int_vals:
                 return col, np.dtype(np.float64)
-            if col.length:
+            bias = biases[val_names.index(name)]
+            if col.length and not bias:
                 r = lib.reduce(col)
                 if max(abs(r.mn if r.mn == r.mn else 0.0),
                        abs(r.mx if r.mx == r.mx else 0.0)) >= 2.0**53:
                     raise lib.HfError(
                         f"groupby {agg} of int64 column {name!r} exceeds "
                         "2^53: exact int accumulation is a later round")
-            return lib.map_scalar(lib.MAP_CAST_I64, col, 0), \
-                np.dtype(np.int64)
+            out = lib.map_scalar(lib.MAP_CAST_I64, col, 0)
+            if bias:
+                out = lib.map_scalar(lib.MAP_ADD, out, bias)
+            return out, self.dtypes[name]
 
         if agg == "sum":
             cols, dts = {}, {}
@@ -584,8 +609,9 @@ class HipDataframe:
                     fparts.append(p)
             parts = fparts
         ext_names = list(val_names) + [SQ + v for v in val_names]
-        keys, sums, counts, n = self._partition_mgr_cls.groupby_reduce(
-            parts, by, ext_names, True, lib.AGG_SUM)
+        keys, sums, counts, n, _biases = \
+            self._partition_mgr_cls.groupby_reduce(
+                parts, by, ext_names, True, lib.AGG_SUM)
         k = len(val_names)
         return keys, key_cats, sums[:k], sums[k:], counts[:k], n
 

@@ -197,7 +197,7 @@ class HipDataframePartitionManager:
         # key range via the i64 reduce kernel
         kmin, kmax, total_rows = None, None, 0
         key_cols = []
-        val_cols_per_part = []
+        raw_cols_per_part = []
         for p in parts:
             block = p.block()
             kcol = block.columns[by_name]
@@ -207,23 +207,54 @@ class HipDataframePartitionManager:
                     "path); hashed/float keys are a later round"
                 )
             key_cols.append(kcol)
-            vals = []
-            for v in val_names:
-                c = block.columns[v]
-                if c.dtype_code == lib.HF_INT64 and c.length:
-                    r = lib.reduce(c)
-                    if max(abs(r.imn), abs(r.imx)) >= 1 << 53:
-                        raise lib.HfError(
-                            f"groupby over int64 column {v!r} with values "
-                            "beyond 2^53: f64 accumulation would round — "
-                            "exact int accumulation is a later round")
-                vals.append(lib.cast_f64(c))
-            val_cols_per_part.append(vals)
+            raw_cols_per_part.append([block.columns[v] for v in val_names])
             if kcol.length:
                 r = lib.reduce(kcol)
                 kmin = r.imn if kmin is None else min(kmin, r.imn)
                 kmax = r.imx if kmax is None else max(kmax, r.imx)
             total_rows += kcol.length
+        # int64 value columns: the f64 accumulators are exact below 2^53.
+        # For MIN/MAX a global per-column BIAS (subtract the column min)
+        # keeps huge-magnitude ints (datetime64 ns views) exact as long as
+        # the RANGE fits 2^53 — the caller adds the bias back int-side.
+        biases = [0] * len(val_names)
+        from ..distributed import is_active
+        for ci in range(len(val_names)):
+            imn = imx = None
+            is_int = False
+            for raw in raw_cols_per_part:
+                c = raw[ci]
+                if c.dtype_code != lib.HF_INT64:
+                    continue
+                is_int = True
+                if c.length:
+                    r = lib.reduce(c)
+                    imn = r.imn if imn is None else min(imn, r.imn)
+                    imx = r.imx if imx is None else max(imx, r.imx)
+            if not is_int:
+                continue
+            if is_active():
+                imn, imx = maybe_allreduce_keyrange(imn, imx)
+            if imn is None:
+                continue
+            if max(abs(imn), abs(imx)) < 1 << 53:
+                continue
+            if agg_op in (lib.AGG_MIN, lib.AGG_MAX) and \
+                    imx - imn < 1 << 53:
+                biases[ci] = imn
+                continue
+            raise lib.HfError(
+                f"groupby over int64 column {val_names[ci]!r} with values "
+                "beyond 2^53: f64 accumulation would round — exact int "
+                "accumulation covers min/max ranges below 2^53 only")
+        val_cols_per_part = []
+        for raw in raw_cols_per_part:
+            vals = []
+            for ci, c in enumerate(raw):
+                if biases[ci] and c.length:
+                    c = lib.map_scalar(lib.MAP_SUB, c, biases[ci])
+                vals.append(lib.cast_f64(c))
+            val_cols_per_part.append(vals)
 
         kmin, kmax = maybe_allreduce_keyrange(kmin, kmax)
         if kmin is None:  # globally empty
@@ -232,12 +263,14 @@ class HipDataframePartitionManager:
         if n_slots < 0:
             n_slots = 0
         if n_slots > config.MaxGroupbySlots.get():
-            from ..distributed import is_active
             if is_active():
-                return cls._groupby_shuffle(key_cols, val_cols_per_part,
-                                            want_counts, agg_op)
-            return cls._groupby_hash(key_cols, val_cols_per_part, total_rows,
-                                     want_counts, agg_op)
+                k4, s4, c4, n4 = cls._groupby_shuffle(
+                    key_cols, val_cols_per_part, want_counts, agg_op)
+            else:
+                k4, s4, c4, n4 = cls._groupby_hash(
+                    key_cols, val_cols_per_part, total_rows, want_counts,
+                    agg_op)
+            return k4, s4, c4, n4, biases
         n_slots = max(n_slots, 1)
         table = GroupbyTable(len(val_names), kmin, n_slots, want_counts,
                              agg_op)
@@ -250,7 +283,7 @@ class HipDataframePartitionManager:
             table.sums, table.rowcnt, table.counts, len(val_names), kmin, n_slots
         )
         table.free()
-        return keys, sums, counts, n
+        return keys, sums, counts, n, biases
 
 
     @classmethod

@@ -115,13 +115,24 @@ class HipQueryCompiler:
         return pandas.Series(vals, index=pandas.Index(list(frame.columns)),
                              dtype=np.float64)
 
+    def _tag_key_index(self, res: "HipQueryCompiler", by):
+        """Groupby by a datetime64 key: the result index materializes back
+        to the tagged dtype (the int64-ns typed-column layer)."""
+        key = by if isinstance(by, str) else (
+            by[0] if isinstance(by, (list, tuple)) and len(by) == 1 else None)
+        if key is not None:
+            dt = dict(self._modin_frame.dtypes).get(key)
+            if dt is not None and np.issubdtype(dt, np.datetime64):
+                res._modin_frame._index_dtype = dt
+        return res
+
     def groupby_var(self, by: str, ddof: int = 1) -> "HipQueryCompiler":
-        return self.__constructor__(
-            self._modin_frame.groupby_var(by, ddof, sqrt=False))
+        return self._tag_key_index(self.__constructor__(
+            self._modin_frame.groupby_var(by, ddof, sqrt=False)), by)
 
     def groupby_std(self, by: str, ddof: int = 1) -> "HipQueryCompiler":
-        return self.__constructor__(
-            self._modin_frame.groupby_var(by, ddof, sqrt=True))
+        return self._tag_key_index(self.__constructor__(
+            self._modin_frame.groupby_var(by, ddof, sqrt=True)), by)
 
     def median(self):
         vals = self._modin_frame.median_columns()
@@ -313,14 +324,19 @@ class HipQueryCompiler:
                 f"groupby agg {agg!r} not implemented on the HipNative backend"
             )
         if agg in ("sum", "count", "mean", "min", "max"):
-            return fn(self, by, dropna=dropna)
+            return self._tag_key_index(fn(self, by, dropna=dropna), by)
         if not dropna:
             raise lib.HfError(f"groupby(dropna=False).{agg} is a later "
                               "round (sum/count/mean/min/max only)")
-        return fn(self, by)
+        return self._tag_key_index(fn(self, by), by)
 
     # ---- comparisons (query_compiler gt/lt/eq bindings) -> int64 0/1 mask
     def _compare(self, op_code, other):
+        import datetime
+        if isinstance(other, (pandas.Timestamp, np.datetime64,
+                              datetime.datetime)):
+            # datetime operand vs the int64-ns typed column
+            other = int(pandas.Timestamp(other).value)
         if not np.isscalar(other):
             raise lib.HfError("comparisons support scalars this round")
         return self.__constructor__(

@@ -1958,3 +1958,74 @@ def test_binned_merge_beyond_csr_cap(npartitions):
     np.testing.assert_allclose(out["a"].to_numpy(), lv[mask], rtol=0)
     np.testing.assert_allclose(out["b"].to_numpy(),
                                lk[mask].astype(np.float64) * 2.5, rtol=0)
+
+
+def test_datetime64_typed_columns(npartitions):
+    """datetime64[ns] rides the int64 typed-column layer (DESIGN §Round-2
+    roadmap 4): round trip, sort, filter vs Timestamp scalars, groupby
+    (datetime values and datetime KEY with DatetimeIndex result), merge on
+    datetime keys — all vs pandas (NaT-free; NaT is a loud later-round)."""
+    rng = np.random.default_rng(88)
+    n = 50_000
+    base = pandas.Timestamp("2021-03-01").value
+    tvals = base + rng.integers(0, 10**15, n)
+    t = tvals.astype("datetime64[ns]")
+    k = rng.integers(0, 300, n).astype(np.int64)
+    v = rng.random(n)
+    pdf = pandas.DataFrame({"t": t, "k": k, "v": v})
+    df = mpd.DataFrame(pdf)
+
+    # round trip preserves dtype + values
+    back = df.to_pandas()
+    assert back["t"].dtype == np.dtype("datetime64[ns]")
+    np.testing.assert_array_equal(back["t"].to_numpy(), t)
+
+    # sort by the datetime column
+    got = df.sort_values("t").to_pandas()
+    exp = pdf.sort_values("t", kind="stable")
+    np.testing.assert_array_equal(got["t"].to_numpy(), exp["t"].to_numpy())
+    np.testing.assert_array_equal(got.index.to_numpy(),
+                                  exp.index.to_numpy())
+
+    # filter vs a Timestamp scalar
+    cut = pandas.Timestamp(base + 5 * 10**14)
+    got = df[df["t"] > cut].to_pandas()
+    exp = pdf[pdf["t"] > cut]
+    np.testing.assert_array_equal(got["t"].to_numpy(), exp["t"].to_numpy())
+    np.testing.assert_allclose(got["v"].to_numpy(), exp["v"].to_numpy(),
+                               rtol=0)
+
+    # groupby with datetime VALUES (min/max of timestamps)
+    got = df.groupby("k").max().to_pandas()
+    exp = pdf.groupby("k").max()
+    assert got["t"].dtype == np.dtype("datetime64[ns]")
+    np.testing.assert_array_equal(got["t"].to_numpy(),
+                                  exp["t"].to_numpy())
+
+    # groupby BY a datetime key -> DatetimeIndex result
+    day = pdf["t"].dt.floor("D")
+    pdf2 = pandas.DataFrame({"d": day, "v": v})
+    df2 = mpd.DataFrame(pdf2)
+    got = df2.groupby("d").sum().to_pandas()
+    exp = pdf2.groupby("d").sum()
+    assert got.index.dtype == np.dtype("datetime64[ns]")
+    np.testing.assert_array_equal(got.index.to_numpy(),
+                                  exp.index.to_numpy())
+    np.testing.assert_allclose(got["v"].to_numpy(), exp["v"].to_numpy(),
+                               rtol=1e-12)
+
+    # merge on a datetime key
+    keys = pandas.Series(t).iloc[:500].reset_index(drop=True)
+    rpdf = pandas.DataFrame({"t": keys, "b": rng.random(500)})
+    got = df.merge(mpd.DataFrame(rpdf), on="t").to_pandas()
+    exp = pdf.merge(rpdf, on="t")
+    assert got["t"].dtype == np.dtype("datetime64[ns]")
+    np.testing.assert_array_equal(got["t"].to_numpy(), exp["t"].to_numpy())
+    np.testing.assert_allclose(got["b"].to_numpy(), exp["b"].to_numpy(),
+                               rtol=0)
+
+    # NaT and tz-aware are loud errors
+    bad = pandas.DataFrame({"t": pandas.to_datetime(
+        ["2020-01-01", None])})
+    with pytest.raises(lib.HfError, match="NaT"):
+        mpd.DataFrame(bad)
