@@ -52,6 +52,14 @@ struct hf_col {
   // Same immutable-column caching rationale as d_hist.
   void*   d_k32 = nullptr;
   int64_t k32_min = 0;
+  // cached radix-scatter layout derived from d_hist: device cursor-init
+  // array (region starts) + device work items — steady-state groupby
+  // launches copy cursors D2D and never touch the host (no H2D, no sync)
+  void*   d_curinit = nullptr;   // u32[nb]
+  void*   d_work = nullptr;      // GbWorkItem[work_n]
+  int64_t work_n = 0;
+  int64_t radix_rows = 0;        // payload rows (64-aligned region total)
+  int     radix_rl = 0;          // validity: matches (hist_kmin, hist_nb)
 };
 
 namespace {
@@ -64,6 +72,12 @@ struct State {
   bool        inited = false;
   int         gpu = -1;
   hipStream_t stream = nullptr;
+  // consumer stream for the overlapped radix-groupby P2 (all OTHER compute
+  // stays on `stream`; stream2 work is always fenced back onto `stream`
+  // with an event before any buffer it read is freed, so the allocator's
+  // single-stream reuse argument still holds)
+  hipStream_t stream2 = nullptr;
+  hipEvent_t  ov_ev[10] = {};
   // small persistent device scratch: reduce accumulators + error word +
   // compact bookkeeping
   void*       d_scratch = nullptr;   // see layout below
@@ -128,9 +142,10 @@ int resolve_stats(const char* where) {
   return HF_OK;
 }
 
-// launch helper with optional event bracketing
+// launch helper with optional event bracketing; pass the stream the launch
+// targets (default: the module stream)
 template <typename F>
-int timed_launch(const char* name, F&& launch) {
+int timed_launch_on(const char* name, hipStream_t s, F&& launch) {
   if (!g.profiling) {
     launch();
     hipError_t e = hipGetLastError();
@@ -141,14 +156,19 @@ int timed_launch(const char* name, F&& launch) {
   p.name = name;
   HF_HIP(name, hipEventCreate(&p.a));
   HF_HIP(name, hipEventCreate(&p.b));
-  HF_HIP(name, hipEventRecord(p.a, g.stream));
+  HF_HIP(name, hipEventRecord(p.a, s));
   launch();
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) return set_hip_err(name, e);
-  HF_HIP(name, hipEventRecord(p.b, g.stream));
+  HF_HIP(name, hipEventRecord(p.b, s));
   g.pending.push_back(p);
   if (g.pending.size() > 4096) return resolve_stats(name);
   return HF_OK;
+}
+
+template <typename F>
+int timed_launch(const char* name, F&& launch) {
+  return timed_launch_on(name, g.stream, std::forward<F>(launch));
 }
 
 constexpr int BLOCK = 256;
@@ -703,31 +723,35 @@ __global__ void __launch_bounds__(BLOCK) k_conv_keys32(
 // that point — so their ~700-cycle latency hides under t's writeout + barrier
 // instead of serializing at the top of the next iteration (the 81%
 // wave-parked stall of the round-1 kernel, profiles/r01b/sq_scatter.txt).
+// row0/row1 bound the input rows (the overlapped-P2 path scatters in
+// chunks; row0 is always even, only the LAST chunk may end odd)
 template <int NV, int RPT, int BLK, int RL, bool K32>
 __global__ void __launch_bounds__(BLK) k_gb_scatter(
     const int64_t* __restrict__ keys, const unsigned* __restrict__ keys32,
     const double* __restrict__ v0,
-    const double* __restrict__ v1, int64_t n, int64_t key_min, int64_t n_slots,
+    const double* __restrict__ v1, int64_t row0, int64_t row1,
+    int64_t key_min, int64_t n_slots,
     int nb, unsigned* __restrict__ cursors,
     double* __restrict__ r0, double* __restrict__ r1,
     unsigned short* __restrict__ rk, unsigned long long* __restrict__ err) {
   constexpr int TILE = BLK * RPT;
   constexpr int PAIRS = RPT / 2;
-  if ((n & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
+  if ((row1 & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
     // odd tail row: direct single-row reservation + write
-    const int64_t k = K32 ? (int64_t)keys32[n - 1] : keys[n - 1] - key_min;
+    const int64_t k =
+        K32 ? (int64_t)keys32[row1 - 1] : keys[row1 - 1] - key_min;
     if ((uint64_t)k < (uint64_t)n_slots) {
       const int b = (int)(k >> RL);
       const int64_t pos = (int64_t)atomicAdd(&cursors[b], 1u);
       rk[pos] = (unsigned short)(k & ((1 << RL) - 1));
-      if (NV > 0) r0[pos] = v0[n - 1];
-      if (NV > 1) r1[pos] = v1[n - 1];
+      if (NV > 0) r0[pos] = v0[row1 - 1];
+      if (NV > 1) r1[pos] = v1[row1 - 1];
     } else {
       atomicAdd(err, 1ULL);
     }
   }
-  const int64_t neven = n & ~1LL;
-  const int64_t npair_total = neven >> 1;
+  const int64_t pr0 = row0 >> 1;
+  const int64_t npair_total = ((row1 & ~1LL) - row0) >> 1;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   double* sval0 = reinterpret_cast<double*>(smem_raw);            // [TILE]
   double* sval1 = sval0 + (NV > 1 ? TILE : 0);
@@ -739,7 +763,7 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
   unsigned* s_total = it_gbase + nb;  // scalar; keep ALL LDS in the dynamic
                                       // region (a static __shared__ would
                                       // shift the base off 16B — G17)
-  const int64_t ntiles = (neven + TILE - 1) / TILE;
+  const int64_t ntiles = (npair_total * 2 + TILE - 1) / TILE;
 
   // only the K32-selected key array and the first NV value arrays are ever
   // touched; the others constant-fold away (template params) and cost no
@@ -751,8 +775,9 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
   auto issue_loads = [&](int64_t tile) {
 #pragma unroll
     for (int j = 0; j < PAIRS; ++j) {
-      const int64_t pr = tile * (TILE / 2) + (int64_t)j * BLK + threadIdx.x;
-      if (pr < npair_total) {
+      const int64_t pj = tile * (TILE / 2) + (int64_t)j * BLK + threadIdx.x;
+      if (pj < npair_total) {
+        const int64_t pr = pr0 + pj;
         if (K32)
           kraw[j] = reinterpret_cast<const uint2*>(keys32)[pr];
         else
@@ -773,10 +798,10 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
     __syncthreads();  // it_cnt cleared (prologue or previous iteration)
 #pragma unroll
     for (int j = 0; j < PAIRS; ++j) {
-      const int64_t pr = tile * (TILE / 2) + (int64_t)j * BLK + threadIdx.x;
+      const int64_t pj = tile * (TILE / 2) + (int64_t)j * BLK + threadIdx.x;
       const int a = 2 * j, bslot = 2 * j + 1;
       lb[a] = lb[bslot] = -1;
-      if (pr < npair_total) {
+      if (pj < npair_total) {
         const int64_t ka = K32 ? (int64_t)kraw[j].x : kraw64[j].x - key_min;
         const int64_t kb = K32 ? (int64_t)kraw[j].y : kraw64[j].y - key_min;
         if ((uint64_t)ka < (uint64_t)n_slots) {
@@ -911,6 +936,85 @@ __global__ void __launch_bounds__(512) k_gb_bucket_agg(
     if (HAVE_VAL) glob_slot_agg<AOP>(&gsums[gbase + s], lsums[s]);
     if (ROWCNT) atomicAdd(&growcnt[gbase + s], 1ULL);
     if (CNT) atomicAdd(&gcounts[gbase + s], (unsigned long long)lcnt[s]);
+  }
+}
+
+// Overlapped-P2 owner form (round 2): block b owns bucket b for the whole
+// step and consumes the payload range [lo[b], hi[b]) — the region slice a
+// P1 scatter chunk just completed (lo/hi are device cursor snapshots, so
+// the host never syncs mid-step).  The merge into the global table is a
+// PLAIN read-modify-write: the sole writer of slice b is this block, and
+// all chunk launches serialize on the consumer stream.  Runs concurrently
+// with the next P1 chunk on the main stream (disjoint payload ranges).
+template <bool ROWCNT, bool CNT, bool HAVE_VAL, int RL, int AOP>
+__global__ void __launch_bounds__(512) k_gb_agg_range(
+    const double* __restrict__ vals, const unsigned short* __restrict__ lowkeys,
+    const unsigned* __restrict__ lo, const unsigned* __restrict__ hi,
+    int64_t n_slots, double* __restrict__ gsums,
+    unsigned long long* __restrict__ growcnt,
+    unsigned long long* __restrict__ gcounts) {
+  constexpr int RANGE = 1 << RL;
+  __shared__ double lsums[HAVE_VAL ? RANGE : 1];
+  __shared__ unsigned lcnt[CNT ? RANGE : 1];
+  __shared__ unsigned char ltouch[RANGE];
+  const int b = blockIdx.x;
+  const int64_t s0 = lo[b];
+  const int64_t s1 = hi[b];
+  if (s1 <= s0) return;
+  for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
+    if (HAVE_VAL) lsums[s] = agg_identity<AOP>();
+    ltouch[s] = 0;
+    if (CNT) lcnt[s] = 0;
+  }
+  __syncthreads();
+  // cursor snapshots have arbitrary parity: scalar head/tail rows keep the
+  // vectorized middle 2-row aligned
+  auto one_row = [&](int64_t i) {
+    const int slot = lowkeys[i];
+    ltouch[slot] = 1;
+    if (HAVE_VAL) {
+      const double v = vals[i];
+      if (v == v) {
+        lds_slot_agg<AOP>(&lsums[slot], v);
+        if (CNT) atomicAdd(&lcnt[slot], 1u);
+      }
+    }
+  };
+  int64_t a0 = s0;
+  if ((a0 & 1) && threadIdx.x == 0) one_row(a0);
+  a0 += (a0 & 1);
+  const int64_t npair = (s1 - a0) >> 1;
+  const ushort2* k2 = reinterpret_cast<const ushort2*>(lowkeys + a0);
+  const double2* v2 = reinterpret_cast<const double2*>(vals + a0);
+  for (int64_t i = threadIdx.x; i < npair; i += blockDim.x) {
+    const ushort2 kk = k2[i];
+    ltouch[kk.x] = 1;
+    ltouch[kk.y] = 1;
+    if (HAVE_VAL) {
+      const double2 vv = v2[i];
+      if (vv.x == vv.x) {
+        lds_slot_agg<AOP>(&lsums[kk.x], vv.x);
+        if (CNT) atomicAdd(&lcnt[kk.x], 1u);
+      }
+      if (vv.y == vv.y) {
+        lds_slot_agg<AOP>(&lsums[kk.y], vv.y);
+        if (CNT) atomicAdd(&lcnt[kk.y], 1u);
+      }
+    }
+  }
+  if (((s1 - a0) & 1) && threadIdx.x == 0) one_row(s1 - 1);
+  __syncthreads();
+  const int64_t gbase = (int64_t)b << RL;
+  for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
+    if (!ltouch[s] || gbase + s >= n_slots) continue;
+    if (HAVE_VAL) {
+      double* a = &gsums[gbase + s];
+      if (AOP == HF_AGG_SUM) *a += lsums[s];
+      else if (AOP == HF_AGG_MIN) *a = fmin(*a, lsums[s]);
+      else *a = fmax(*a, lsums[s]);
+    }
+    if (ROWCNT) growcnt[gbase + s] += 1ULL;
+    if (CNT) gcounts[gbase + s] += (unsigned long long)lcnt[s];
   }
 }
 
@@ -2232,6 +2336,9 @@ int hf_init(int gpu) {
     return set_err(HF_ERR_ARG, "hf_init", "gpu index out of range");
   HF_HIP("hf_init", hipSetDevice(gpu));
   HF_HIP("hf_init", hipStreamCreate(&g.stream));
+  HF_HIP("hf_init", hipStreamCreate(&g.stream2));
+  for (auto& ev : g.ov_ev)
+    HF_HIP("hf_init", hipEventCreateWithFlags(&ev, hipEventDisableTiming));
   HF_HIP("hf_init", hipMalloc(&g.d_scratch, SCRATCH_BYTES));
   HF_HIP("hf_init", hipMemset(g.d_scratch, 0, SCRATCH_BYTES));
   g.gpu = gpu;
@@ -2242,12 +2349,16 @@ int hf_init(int gpu) {
 int hf_shutdown(void) {
   if (!g.inited) return HF_OK;
   hipStreamSynchronize(g.stream);
+  if (g.stream2) hipStreamSynchronize(g.stream2);
   g_cache.trim();
   for (auto& p : g.pending) { hipEventDestroy(p.a); hipEventDestroy(p.b); }
   g.pending.clear();
   g.stats.clear();
   if (g.d_scratch) hipFree(g.d_scratch);
+  for (auto& ev : g.ov_ev)
+    if (ev) hipEventDestroy(ev);
   hipStreamDestroy(g.stream);
+  if (g.stream2) hipStreamDestroy(g.stream2);
   g = State{};
   return HF_OK;
 }
@@ -2308,6 +2419,8 @@ int hf_col_free(hf_col* col) {
   if (g.inited && col->dptr) dev_free(col->dptr, g.stream);
   if (col->d_hist) free(col->d_hist);  // host-side cached histogram
   if (g.inited && col->d_k32) dev_free(col->d_k32, g.stream);
+  if (g.inited && col->d_curinit) dev_free(col->d_curinit, g.stream);
+  if (g.inited && col->d_work) dev_free(col->d_work, g.stream);
   delete col;
   return HF_OK;
 }
@@ -2526,6 +2639,16 @@ int ensure_host_hist(hf_col* keys, int64_t key_min, int64_t n_slots,
   if (keys->d_hist && keys->hist_kmin == key_min && keys->hist_nb == nb)
     return HF_OK;
   if (keys->d_hist) { free(keys->d_hist); keys->d_hist = nullptr; }
+  // the radix layout cache derives from this histogram: invalidate with it
+  if (keys->d_curinit) {
+    dev_free(keys->d_curinit, g.stream);
+    keys->d_curinit = nullptr;
+  }
+  if (keys->d_work) {
+    dev_free(keys->d_work, g.stream);
+    keys->d_work = nullptr;
+  }
+  keys->radix_rl = 0;
   unsigned long long* d_h = nullptr;
   HF_HIP("gb_hist", dev_alloc((void**)&d_h, nb * 8, g.stream));
   HF_HIP("gb_hist", hipMemsetAsync(d_h, 0, nb * 8, g.stream));
@@ -2616,29 +2739,61 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
   if (rc != HF_OK) return rc;
   const int64_t* h = keys->d_hist ? (const int64_t*)keys->d_hist : nullptr;
   // exact per-bucket regions, 64-row aligned; u32 cursors cap one partition
-  // at ~4.29e9 rows (shard larger frames)
-  std::vector<unsigned> cur_init((size_t)nb);
-  std::vector<GbWorkItem> work;
-  int64_t off = 0;
-  for (int64_t b = 0; b < nb; ++b) {
-    cur_init[b] = (unsigned)off;
-    const int64_t rows_b = h[b];
-    int64_t done = 0;
-    while (done < rows_b) {
-      const int64_t len = std::min(AGG_CHUNK, rows_b - done);
-      work.push_back(GbWorkItem{off + done, (int32_t)b, (int32_t)len});
-      done += len;
+  // at ~4.29e9 rows (shard larger frames).  The cursor-init array and the
+  // work items derive only from the cached histogram, so they cache on the
+  // immutable key column too: steady-state groupby calls copy cursors D2D
+  // and never touch the host (no pageable H2D, no mid-op sync).
+  if (!keys->d_curinit || keys->radix_rl != RL) {
+    if (keys->d_curinit) {
+      dev_free(keys->d_curinit, g.stream);
+      keys->d_curinit = nullptr;
     }
-    off += (rows_b + 63) & ~63LL;
+    if (keys->d_work) {
+      dev_free(keys->d_work, g.stream);
+      keys->d_work = nullptr;
+    }
+    std::vector<unsigned> cur_init((size_t)nb);
+    std::vector<GbWorkItem> work;
+    int64_t off = 0;
+    for (int64_t b = 0; b < nb; ++b) {
+      cur_init[b] = (unsigned)off;
+      const int64_t rows_b = h[b];
+      int64_t done = 0;
+      while (done < rows_b) {
+        const int64_t len = std::min(AGG_CHUNK, rows_b - done);
+        work.push_back(GbWorkItem{off + done, (int32_t)b, (int32_t)len});
+        done += len;
+      }
+      off += (rows_b + 63) & ~63LL;
+    }
+    if (off > 0xFFFFFFF0LL)
+      return set_err(HF_ERR_UNSUPPORTED, "hf_groupby_accum",
+                     "partition too large for radix scatter (shard it)");
+    HF_HIP("gb_radix", dev_alloc(&keys->d_curinit, nb * 4, g.stream));
+    HF_HIP("gb_radix",
+           dev_alloc(&keys->d_work,
+                     std::max<size_t>(work.size(), 1) * sizeof(GbWorkItem),
+                     g.stream));
+    HF_HIP("gb_radix", hipMemcpyAsync(keys->d_curinit, cur_init.data(),
+                                      nb * 4, hipMemcpyHostToDevice,
+                                      g.stream));
+    if (!work.empty())
+      HF_HIP("gb_radix",
+             hipMemcpyAsync(keys->d_work, work.data(),
+                            work.size() * sizeof(GbWorkItem),
+                            hipMemcpyHostToDevice, g.stream));
+    // host vectors must outlive the async H2D of pageable memory
+    HF_HIP("gb_radix", hipStreamSynchronize(g.stream));
+    keys->work_n = (int64_t)work.size();
+    keys->radix_rows = off;
+    keys->radix_rl = RL;
   }
-  if (off > 0xFFFFFFF0LL)
-    return set_err(HF_ERR_UNSUPPORTED, "hf_groupby_accum",
-                   "partition too large for radix scatter (shard it)");
+  const int64_t off = keys->radix_rows;
+  GbWorkItem* d_work = (GbWorkItem*)keys->d_work;
   double* r0 = nullptr;
   double* r1 = nullptr;
   unsigned short* rk = nullptr;
   unsigned* d_cur = nullptr;
-  GbWorkItem* d_work = nullptr;
   const int64_t alloc_rows = off > 0 ? off : 64;
   if (nvals > 0)
     HF_HIP("gb_radix", dev_alloc((void**)&r0, alloc_rows * 8, g.stream));
@@ -2646,16 +2801,113 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
     HF_HIP("gb_radix", dev_alloc((void**)&r1, alloc_rows * 8, g.stream));
   HF_HIP("gb_radix", dev_alloc((void**)&rk, alloc_rows * 2, g.stream));
   HF_HIP("gb_radix", dev_alloc((void**)&d_cur, nb * 4, g.stream));
-  HF_HIP("gb_radix",
-         dev_alloc((void**)&d_work, work.size() * sizeof(GbWorkItem),
-                        g.stream));
-  HF_HIP("gb_radix", hipMemcpyAsync(d_cur, cur_init.data(), nb * 4,
-                                    hipMemcpyHostToDevice, g.stream));
-  HF_HIP("gb_radix",
-         hipMemcpyAsync(d_work, work.data(), work.size() * sizeof(GbWorkItem),
-                        hipMemcpyHostToDevice, g.stream));
-  // host vectors must outlive the async H2D of pageable memory
-  HF_HIP("gb_radix", hipStreamSynchronize(g.stream));
+  HF_HIP("gb_radix", hipMemcpyAsync(d_cur, keys->d_curinit, nb * 4,
+                                    hipMemcpyDeviceToDevice, g.stream));
+
+  // ---- overlapped path (round 2): scatter in 4 chunks on the main
+  // stream; after each chunk the consumer stream aggregates the region
+  // slices that chunk completed (owner-block k_gb_agg_range, plain RMW —
+  // cursor snapshots travel device-to-device).  The last chunk is small
+  // (10%) so the only non-overlapped P2 work is cheap; the rest of P2
+  // rides bandwidth the latency-bound scatter leaves unused.
+  const bool overlap = n >= (64LL << 20) && nvals <= 2 && nb <= 1024;
+  if (overlap) {
+    const int NCH = 4;
+    const double fracs[NCH] = {0.30, 0.30, 0.30, 0.10};
+    unsigned* d_snap[NCH];
+    for (int c = 0; c < NCH; ++c)
+      HF_HIP("gb_radix", dev_alloc((void**)&d_snap[c], nb * 4, g.stream));
+    int64_t bounds[NCH + 1];
+    bounds[0] = 0;
+    double accf = 0;
+    for (int c = 0; c < NCH - 1; ++c) {
+      accf += fracs[c];
+      bounds[c + 1] = ((int64_t)(n * accf)) & ~1LL;
+    }
+    bounds[NCH] = n;
+    auto scat_chunk = [&](auto nvTag, auto rptTag, auto blkTag, int c) {
+      constexpr int NVv = decltype(nvTag)::value;
+      constexpr int RPTv = decltype(rptTag)::value;
+      constexpr int BLKv = decltype(blkTag)::value;
+      const int64_t tile_sz = (int64_t)BLKv * RPTv;
+      const int64_t rows_c = bounds[c + 1] - bounds[c];
+      const int64_t ntiles = (rows_c + tile_sz - 1) / tile_sz;
+      const uint32_t sgrid =
+          (uint32_t)std::min<int64_t>(std::max<int64_t>(ntiles, 1), 2048);
+      const uint32_t lds =
+          (uint32_t)(tile_sz * (8 * NVv + 4) + nb * 12 + 16);
+      return timed_launch("gb_scatter", [&] {
+        hipLaunchKernelGGL((k_gb_scatter<NVv, RPTv, BLKv, RL, true>),
+                           dim3(sgrid), dim3(BLKv), lds, g.stream,
+                           (const int64_t*)keys->dptr,
+                           (const unsigned*)keys->d_k32, ptrs.vals[0],
+                           ptrs.vals[1], bounds[c], bounds[c + 1], key_min,
+                           n_slots, (int)nb, d_cur, r0, r1, rk, d_err);
+      });
+    };
+    auto agg_chunk = [&](auto rTag, auto cTag, auto vTag, const double* v,
+                         double* gs, unsigned long long* gc,
+                         const unsigned* clo, const unsigned* chi) {
+      constexpr bool R = decltype(rTag)::value, C = decltype(cTag)::value,
+                     V = decltype(vTag)::value;
+      return timed_launch_on("gb_bucket_agg", g.stream2, [&] {
+        hipLaunchKernelGGL((k_gb_agg_range<R, C, V, RL, AOP>),
+                           dim3((uint32_t)nb), dim3(512), 0, g.stream2, v,
+                           rk, clo, chi, n_slots, gs,
+                           (unsigned long long*)rowcnt, gc);
+      });
+    };
+    using T = std::true_type;
+    using F = std::false_type;
+    const bool cnt = counts != 0;
+    for (int c = 0; c < NCH && rc == HF_OK; ++c) {
+      rc = nvals == 0 ? scat_chunk(std::integral_constant<int, 0>{},
+                                   std::integral_constant<int, 12>{},
+                                   std::integral_constant<int, 1024>{}, c)
+           : nvals == 1 ? scat_chunk(std::integral_constant<int, 1>{},
+                                     std::integral_constant<int, 12>{},
+                                     std::integral_constant<int, 1024>{}, c)
+                        : scat_chunk(std::integral_constant<int, 2>{},
+                                     std::integral_constant<int, 12>{},
+                                     std::integral_constant<int, 512>{}, c);
+      if (rc != HF_OK) break;
+      HF_HIP("gb_radix", hipMemcpyAsync(d_snap[c], d_cur, nb * 4,
+                                        hipMemcpyDeviceToDevice, g.stream));
+      HF_HIP("gb_radix", hipEventRecord(g.ov_ev[c], g.stream));
+      HF_HIP("gb_radix", hipStreamWaitEvent(g.stream2, g.ov_ev[c], 0));
+      const unsigned* clo =
+          c == 0 ? (const unsigned*)keys->d_curinit : d_snap[c - 1];
+      const unsigned* chi = d_snap[c];
+      if (nvals == 0) {
+        rc = agg_chunk(T{}, F{}, F{}, nullptr, nullptr, nullptr, clo, chi);
+      } else {
+        for (int col = 0; col < nvals && rc == HF_OK; ++col) {
+          const double* v = col == 0 ? r0 : r1;
+          double* gs = (double*)sums + (int64_t)col * n_slots;
+          unsigned long long* gc =
+              cnt ? (unsigned long long*)counts + (int64_t)col * n_slots
+                  : nullptr;
+          if (col == 0 && with_rowcnt)
+            rc = cnt ? agg_chunk(T{}, T{}, T{}, v, gs, gc, clo, chi)
+                     : agg_chunk(T{}, F{}, T{}, v, gs, gc, clo, chi);
+          else
+            rc = cnt ? agg_chunk(F{}, T{}, T{}, v, gs, gc, clo, chi)
+                     : agg_chunk(F{}, F{}, T{}, v, gs, gc, clo, chi);
+        }
+      }
+    }
+    // fence the consumer stream back onto the main stream BEFORE the
+    // payload buffers are released (keeps the allocator's single-stream
+    // reuse argument intact)
+    HF_HIP("gb_radix", hipEventRecord(g.ov_ev[NCH], g.stream2));
+    HF_HIP("gb_radix", hipStreamWaitEvent(g.stream, g.ov_ev[NCH], 0));
+    for (int c = 0; c < NCH; ++c) dev_free(d_snap[c], g.stream);
+    if (r0) dev_free(r0, g.stream);
+    if (r1) dev_free(r1, g.stream);
+    dev_free(rk, g.stream);
+    dev_free(d_cur, g.stream);
+    return rc;
+  }
   // P1 scatter (pipelined): 12288-row tiles as 1024x12 for <=1 value column
   // — 147 KB LDS, 1 block/CU of 16 waves (round-2 probe: 1024x12 edges out
   // 512x24 and both beat the round-1 kernel by ~17%); 6144-row tiles as
@@ -2675,8 +2927,8 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
                          dim3(sgrid), dim3(BLKv), lds, g.stream,
                          (const int64_t*)keys->dptr,
                          (const unsigned*)keys->d_k32, ptrs.vals[0],
-                         ptrs.vals[1], n, key_min, n_slots, (int)nb, d_cur, r0,
-                         r1, rk, d_err);
+                         ptrs.vals[1], (int64_t)0, n, key_min, n_slots,
+                         (int)nb, d_cur, r0, r1, rk, d_err);
     });
   };
   rc = nvals == 0 ? scat(std::integral_constant<int, 0>{},
@@ -2691,7 +2943,7 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
   if (rc != HF_OK) return rc;
   // P2 aggregate, one column per launch
   const bool cnt = counts != 0;
-  const uint32_t agrid = (uint32_t)work.size();
+  const uint32_t agrid = (uint32_t)keys->work_n;
   auto agg = [&](auto rTag, auto cTag, auto vTag, const double* v, double* gs,
                  unsigned long long* gc) {
     constexpr bool R = decltype(rTag)::value, C = decltype(cTag)::value,
@@ -2726,7 +2978,6 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
   if (r1) dev_free(r1, g.stream);
   dev_free(rk, g.stream);
   dev_free(d_cur, g.stream);
-  dev_free(d_work, g.stream);
   return rc;
 }
 
