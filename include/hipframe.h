@@ -157,7 +157,8 @@ int hf_reduce(const hf_col* in, hf_reduce_result* out); /* syncs the stream */
  *                                     (optional: pass 0 to skip; needed for
  *                                     count/mean)
  * All buffers must be zeroed by the caller before the first accumulate. */
-enum { HF_AGG_SUM = 0, HF_AGG_MIN = 1, HF_AGG_MAX = 2 };
+enum { HF_AGG_SUM = 0, HF_AGG_MIN = 1, HF_AGG_MAX = 2,
+       HF_AGG_PROD = 3 /* scans only (hf_cumsum / hf_seg_cumsum) */ };
 /* agg_op picks the per-slot combine for the value table (the "sums" buffer
  * doubles as the min/max table; initialize it to 0 / +inf / -inf with
  * hf_fill_f64 before the first accumulate — GroupbyReduceImpl's other

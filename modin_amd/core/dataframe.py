@@ -1459,8 +1459,8 @@ class HipDataframe:
         for b in by_list:
             if b not in self.columns:
                 raise lib.HfError(f"groupby: key column {b!r} missing")
-        if how not in ("cumsum", "cummin", "cummax", "cumcount", "rank",
-                       "ngroup", "shift", "diff", "ffill", "bfill",
+        if how not in ("cumsum", "cummin", "cummax", "cumprod", "cumcount",
+                       "rank", "ngroup", "shift", "diff", "ffill", "bfill",
                        "bsum", "bmin", "bmax", "bcount", "bmean"):
             raise lib.HfError(f"groupby transform {how!r} not supported")
         if how == "rank" and method not in ("average", "min", "first"):
@@ -1694,7 +1694,7 @@ class HipDataframe:
             return HipDataframe([part], pandas.RangeIndex(n), val_names,
                                 [n], pandas.Series(dts))
         agg_op = {"cumsum": lib.AGG_SUM, "cummin": lib.AGG_MIN,
-                  "cummax": lib.AGG_MAX}[how]
+                  "cummax": lib.AGG_MAX, "cumprod": lib.AGG_PROD}[how]
         for v in val_names:
             vc = concat_col(v)
             src_int = vc.dtype_code == lib.HF_INT64

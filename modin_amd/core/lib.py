@@ -398,7 +398,7 @@ def memset_raw(dptr: int, value: int, nbytes: int) -> None:
     _check(load().hf_memset_raw(dptr, value, nbytes), "hf_memset_raw")
 
 
-AGG_SUM, AGG_MIN, AGG_MAX = range(3)
+AGG_SUM, AGG_MIN, AGG_MAX, AGG_PROD = range(4)
 AGG_OP_OF = {"sum": AGG_SUM, "count": AGG_SUM, "mean": AGG_SUM,
              "min": AGG_MIN, "max": AGG_MAX}
 AGG_IDENTITY = {AGG_SUM: 0.0, AGG_MIN: float("inf"), AGG_MAX: float("-inf")}

@@ -1866,12 +1866,14 @@ __device__ __forceinline__ T cs_ident() {
   if (OP == HF_AGG_MAX)
     return (sizeof(T) == 8 && (T)0.5 == 0) ? (T)INT64_MIN
                                            : (T)-__builtin_huge_val();
+  if (OP == HF_AGG_PROD) return T(1);
   return T(0);
 }
 template <typename T, int OP>
 __device__ __forceinline__ T cs_comb(T a, T b) {
   if (OP == HF_AGG_MIN) return a < b ? a : b;
   if (OP == HF_AGG_MAX) return a > b ? a : b;
+  if (OP == HF_AGG_PROD) return a * b;
   return a + b;
 }
 template <typename T, int OP>
@@ -3377,7 +3379,8 @@ static const int64_t* plan_tiles(const hf_filterplan* p);
 int hf_cumsum(const hf_col* col, int agg_op, hf_col** out) {
   HF_NEED_INIT("hf_cumsum");
   if (!col || !out) return set_err(HF_ERR_ARG, "hf_cumsum", "null");
-  if (agg_op != HF_AGG_SUM && agg_op != HF_AGG_MIN && agg_op != HF_AGG_MAX)
+  if (agg_op != HF_AGG_SUM && agg_op != HF_AGG_MIN &&
+      agg_op != HF_AGG_MAX && agg_op != HF_AGG_PROD)
     return set_err(HF_ERR_ARG, "hf_cumsum", "bad agg_op");
   const int64_t n = col->len;
   int rc = hf_col_alloc(n, col->dtype, out);
@@ -3414,6 +3417,8 @@ int hf_cumsum(const hf_col* col, int agg_op, hf_col** out) {
            ? runT(std::integral_constant<int, HF_AGG_MIN>{})
        : agg_op == HF_AGG_MAX
            ? runT(std::integral_constant<int, HF_AGG_MAX>{})
+       : agg_op == HF_AGG_PROD
+           ? runT(std::integral_constant<int, HF_AGG_PROD>{})
            : runT(std::integral_constant<int, HF_AGG_SUM>{});
   dev_free(d_ts, g.stream);
   if (rc != HF_OK) { hf_col_free(*out); *out = nullptr; }
@@ -3428,7 +3433,8 @@ int hf_seg_cumsum(const hf_col* col, const hf_col* heads, int agg_op,
   if (heads->dtype != HF_INT64 || heads->len != col->len)
     return set_err(HF_ERR_ARG, "hf_seg_cumsum",
                    "heads must be an int64 0/1 column of the same length");
-  if (agg_op != HF_AGG_SUM && agg_op != HF_AGG_MIN && agg_op != HF_AGG_MAX)
+  if (agg_op != HF_AGG_SUM && agg_op != HF_AGG_MIN &&
+      agg_op != HF_AGG_MAX && agg_op != HF_AGG_PROD)
     return set_err(HF_ERR_ARG, "hf_seg_cumsum", "bad agg_op");
   const int64_t n = col->len;
   int rc = hf_col_alloc(n, col->dtype, out);
@@ -3472,6 +3478,8 @@ int hf_seg_cumsum(const hf_col* col, const hf_col* heads, int agg_op,
                ? runT(std::integral_constant<int, HF_AGG_MIN>{})
            : agg_op == HF_AGG_MAX
                ? runT(std::integral_constant<int, HF_AGG_MAX>{})
+           : agg_op == HF_AGG_PROD
+               ? runT(std::integral_constant<int, HF_AGG_PROD>{})
                : runT(std::integral_constant<int, HF_AGG_SUM>{});
   }();
   if (d_tf) dev_free(d_tf, g.stream);

@@ -157,6 +157,9 @@ class _HipPandasBase:
     def cummax(self):
         return self._rewrap(self._query_compiler.cummax())
 
+    def cumprod(self):
+        return self._rewrap(self._query_compiler.cumprod())
+
     def shift(self, periods: int = 1):
         return self._rewrap(self._query_compiler.shift(int(periods)))
 
@@ -810,6 +813,9 @@ class DataFrameGroupBy:
 
     def cummax(self):
         return self._transform("cummax")
+
+    def cumprod(self):
+        return self._transform("cumprod")
 
     def cumcount(self):
         return self._transform("cumcount")

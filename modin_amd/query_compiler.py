@@ -154,6 +154,10 @@ class HipQueryCompiler:
     def cummax(self) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.cumsum_rows(lib.AGG_MAX))
 
+    def cumprod(self) -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.cumsum_rows(lib.AGG_PROD))
+
     def shift(self, periods: int) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.shift_rows(periods))
 

@@ -112,23 +112,24 @@ def install(monkeypatch):
     def cumsum(col, agg_op=0):
         x = col.arr
         if col.dtype_code == HF_INT64:
-            fns = {0: np.add, 1: np.minimum, 2: np.maximum}
+            fns = {0: np.add, 1: np.minimum, 2: np.maximum, 3: np.multiply}
             return _reg(MockCol(fns[agg_op].accumulate(x)))
         nan = np.isnan(x)
-        ident = {0: 0.0, 1: np.inf, 2: -np.inf}[agg_op]
+        ident = {0: 0.0, 1: np.inf, 2: -np.inf, 3: 1.0}[agg_op]
         z = np.where(nan, ident, x)
-        fns = {0: np.add, 1: np.minimum, 2: np.maximum}
+        fns = {0: np.add, 1: np.minimum, 2: np.maximum, 3: np.multiply}
         acc = fns[agg_op].accumulate(z)
         return _reg(MockCol(np.where(nan, np.nan, acc)))
 
     def seg_cumsum(col, heads, agg_op=0):
         x = col.arr
         h = heads.arr != 0
-        ident = {0: 0.0, 1: np.inf, 2: -np.inf}[agg_op] \
+        ident = {0: 0.0, 1: np.inf, 2: -np.inf, 3: 1.0}[agg_op] \
             if col.dtype_code == HF_FLOAT64 else \
             {0: 0, 1: np.iinfo(np.int64).max,
-             2: np.iinfo(np.int64).min}[agg_op]
-        comb = {0: lambda a, b: a + b, 1: min, 2: max}[agg_op]
+             2: np.iinfo(np.int64).min, 3: 1}[agg_op]
+        comb = {0: lambda a, b: a + b, 1: min, 2: max,
+                3: lambda a, b: a * b}[agg_op]
         out = np.empty_like(x)
         run = ident
         for i in range(x.size):

@@ -2029,3 +2029,30 @@ def test_datetime64_typed_columns(npartitions):
         ["2020-01-01", None])})
     with pytest.raises(lib.HfError, match="NaT"):
         mpd.DataFrame(bad)
+
+
+def test_cumprod_vs_pandas(npartitions):
+    """Product scan (AGG_PROD through the segmented/linear scan kernels):
+    groupby.cumprod + frame cumprod vs pandas, NaN values skip-but-stay."""
+    rng = np.random.default_rng(123)
+    n = 60_000
+    k = rng.integers(0, 500, n).astype(np.int64)
+    v = np.clip(rng.standard_normal(n), -1.5, 1.5)
+    v[rng.random(n) < 0.1] = np.nan
+    w = (rng.integers(0, 3, n) - 1).astype(np.int64)
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    got = df.groupby("k").cumprod().to_pandas()
+    exp = pdf.groupby("k").cumprod()
+    for c in exp.columns:
+        np.testing.assert_allclose(got[c].to_numpy(dtype=float),
+                                   exp[c].to_numpy(dtype=float),
+                                   rtol=1e-12, atol=1e-300, equal_nan=True,
+                                   err_msg=f"gb-cumprod/{c}")
+    got = df[["v", "w"]].cumprod().to_pandas()
+    exp = pdf[["v", "w"]].cumprod()
+    for c in exp.columns:
+        np.testing.assert_allclose(got[c].to_numpy(dtype=float),
+                                   exp[c].to_numpy(dtype=float),
+                                   rtol=1e-12, atol=1e-300, equal_nan=True,
+                                   err_msg=f"cumprod/{c}")

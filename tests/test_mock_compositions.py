@@ -241,3 +241,28 @@ def test_mock_expanding(mlib):
                     got[c].to_numpy(), exp[c].to_numpy(), rtol=1e-12,
                     atol=1e-12, equal_nan=True,
                     err_msg=f"{op}/mp={mp}/{c}")
+
+
+def test_mock_cumprod(mlib):
+    """groupby.cumprod + frame cumprod (product scan, AGG_PROD): NaN values
+    stay NaN and do not advance the running product (pandas semantics)."""
+    rng = np.random.default_rng(99)
+    pdf = _frames(rng, n=3000)
+    # keep magnitudes tame so f64 products stay finite
+    pdf["v"] = np.clip(pdf["v"], -1.5, 1.5)
+    pdf["w"] = (pdf["w"] % 3) - 1
+    df = mlib.DataFrame(pdf)
+    got = df.groupby("k").cumprod().to_pandas()
+    exp = pdf.groupby("k").cumprod()
+    for c in exp.columns:
+        np.testing.assert_allclose(got[c].to_numpy(dtype=float),
+                                   exp[c].to_numpy(dtype=float),
+                                   rtol=1e-12, atol=1e-300, equal_nan=True,
+                                   err_msg=f"gb-cumprod/{c}")
+    got = df[["v", "w"]].cumprod().to_pandas()
+    exp = pdf[["v", "w"]].cumprod()
+    for c in exp.columns:
+        np.testing.assert_allclose(got[c].to_numpy(dtype=float),
+                                   exp[c].to_numpy(dtype=float),
+                                   rtol=1e-12, atol=1e-300, equal_nan=True,
+                                   err_msg=f"cumprod/{c}")
