@@ -1404,7 +1404,8 @@ class HipDataframe:
 
     def groupby_transform(self, by, how: str, ascending: bool = True,
                           method: str = "average", periods: int = 1,
-                          dropna: bool = True) -> "HipDataframe":
+                          dropna: bool = True,
+                          na_option: str = "keep") -> "HipDataframe":
         """Same-length groupby transforms in original row order.
 
         how: 'cumsum' | 'cummin' | 'cummax' (segmented scan), 'cumcount'
@@ -1435,7 +1436,8 @@ class HipDataframe:
             with dist_mod.local_mode():
                 res = shuf.groupby_transform(by, how, ascending=ascending,
                                              method=method, periods=periods,
-                                             dropna=dropna)
+                                             dropna=dropna,
+                                             na_option=na_option)
             if how == "ngroup":
                 # global group ordinal = local ordinal + #groups on lower
                 # ranks (rank ranges ascend, pandas numbers sorted groups)
@@ -1512,7 +1514,8 @@ class HipDataframe:
                     for b in by_list]
         if how == "rank":
             return self._groupby_rank(by_list, val_names, eff_keys, valid,
-                                      n, concat_col, ascending, method)
+                                      n, concat_col, ascending, method,
+                                      na_option)
         perm = self._compose_sort_perm(eff_keys)
         head = None
         for ekc, _ in eff_keys:
@@ -1922,11 +1925,12 @@ class HipDataframe:
                             pandas.Series({c: np.dtype(np.float64)
                                            for c in names}))
 
-    def rank_rows(self, ascending: bool = True,
-                  method: str = "average") -> "HipDataframe":
+    def rank_rows(self, ascending: bool = True, method: str = "average",
+                  na_option: str = "keep") -> "HipDataframe":
         """Frame-level pandas rank(axis=0) over one constant-key group."""
         return self._with_const_key().groupby_transform(
-            self.KEYCOL, "rank", ascending=ascending, method=method)
+            self.KEYCOL, "rank", ascending=ascending, method=method,
+            na_option=na_option)
 
     def fill_rows(self, how: str) -> "HipDataframe":
         """Frame-level pandas ffill/bfill over one constant-key group."""
@@ -2049,17 +2053,20 @@ class HipDataframe:
                             pandas.Series(dts))
 
     def _groupby_rank(self, by_list, val_names, eff_keys, valid, n,
-                      concat_col, ascending, method):
-        """rank within groups (pandas DataFrameGroupBy.rank,
-        na_option='keep'): per value column, sort by (keys…, value) with
-        the effective-key transform (NaN value sorts last), 1-based
-        position within the key run, tie runs collapsed per `method`
-        ('average' -> first + (len-1)/2, 'min' -> first, 'first' -> the
-        position itself)."""
+                      concat_col, ascending, method, na_option="keep"):
+        """rank within groups (pandas DataFrameGroupBy.rank): per value
+        column, sort by (keys…, value) with the effective-key transform,
+        1-based position within the key run, tie runs collapsed per
+        `method` ('average' -> first + (len-1)/2, 'min' -> first,
+        'first' -> the position itself).  na_option: 'keep' (NaN -> NaN),
+        'top'/'bottom' (NaN values rank lowest/highest — the NaN sentinel
+        sorts first/last and forms ONE tie run, so the normal tie
+        arithmetic produces pandas' shared NaN ranks)."""
         out_cols = {}
         for v in val_names:
             vc = concat_col(v)
-            eff_v = self._effective_sort_key(vc, False, ascending)
+            eff_v = self._effective_sort_key(vc, False, ascending,
+                                             na_first=(na_option == "top"))
             perm = self._compose_sort_perm(eff_keys + [eff_v])
             khead = None
             for ekc, _ in eff_keys:
@@ -2101,7 +2108,7 @@ class HipDataframe:
                 trid = lib.map_scalar(lib.MAP_SUB, lib.cumsum(thead), 1)
                 rank_s = lib.gather(avg, trid)
             res = lib.scatter(rank_s, perm)
-            if vc.dtype_code == lib.HF_FLOAT64:
+            if na_option == "keep" and vc.dtype_code == lib.HF_FLOAT64:
                 notna_v = lib.compare_scalar(lib.CMP_NOTNA, vc, 0.0)
                 res = lib.fixup_empty(res, notna_v)
             if valid is not None:

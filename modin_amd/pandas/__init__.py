@@ -111,11 +111,12 @@ class _HipPandasBase:
 
     def rank(self, method: str = "average", ascending: bool = True,
              na_option: str = "keep"):
-        """pandas rank(axis=0): always float64; na_option='keep'."""
-        if na_option != "keep":
-            raise lib.HfError("rank: only na_option='keep' this round")
+        """pandas rank(axis=0): always float64."""
+        if na_option not in ("keep", "top", "bottom"):
+            raise lib.HfError(f"rank: bad na_option {na_option!r}")
         return self._rewrap(self._query_compiler.rank(
-            method=method, ascending=bool(ascending)))
+            method=method, ascending=bool(ascending),
+            na_option=na_option))
 
     def round(self, decimals: int = 0):  # noqa: A003
         """pandas round: half-even on float columns, ints unchanged."""
@@ -849,10 +850,10 @@ class DataFrameGroupBy:
 
     def rank(self, method: str = "average", ascending: bool = True,
              na_option: str = "keep"):
-        if na_option != "keep":
-            raise lib.HfError("rank: only na_option='keep' this round")
+        if na_option not in ("keep", "top", "bottom"):
+            raise lib.HfError(f"rank: bad na_option {na_option!r}")
         return self._transform("rank", ascending=bool(ascending),
-                               method=method)
+                               method=method, na_option=na_option)
 
     def idxmax(self):
         """Original row label of each group's first max per column

@@ -196,14 +196,15 @@ class HipQueryCompiler:
 
     def groupby_transform(self, by, how: str, ascending: bool = True,
                           method: str = "average", periods: int = 1,
-                          dropna: bool = True) -> "HipQueryCompiler":
+                          dropna: bool = True,
+                          na_option: str = "keep") -> "HipQueryCompiler":
         """Same-length transforms in original row order (pandas
-        DataFrameGroupBy.cumsum/cummin/cummax/cumcount/rank/ngroup/
-        shift/diff; reference routes these through
+        DataFrameGroupBy.cumsum/cummin/cummax/cumprod/cumcount/rank/
+        ngroup/shift/diff; reference routes these through
         modin/pandas/groupby.py -> qc groupby methods)."""
         return self.__constructor__(self._modin_frame.groupby_transform(
             by, how, ascending=ascending, method=method, periods=periods,
-            dropna=dropna))
+            dropna=dropna, na_option=na_option))
 
     def groupby_idxmax(self, by) -> "HipQueryCompiler":
         return self.__constructor__(
@@ -213,10 +214,10 @@ class HipQueryCompiler:
         return self.__constructor__(
             self._modin_frame.groupby_idxminmax(by, maximum=False))
 
-    def rank(self, method: str = "average",
-             ascending: bool = True) -> "HipQueryCompiler":
+    def rank(self, method: str = "average", ascending: bool = True,
+             na_option: str = "keep") -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.rank_rows(
-            ascending=ascending, method=method))
+            ascending=ascending, method=method, na_option=na_option))
 
     def fillna_directional(self, how: str) -> "HipQueryCompiler":
         """pandas ffill/bfill (frame-level, one constant-key group)."""

@@ -2056,3 +2056,33 @@ def test_cumprod_vs_pandas(npartitions):
                                    exp[c].to_numpy(dtype=float),
                                    rtol=1e-12, atol=1e-300, equal_nan=True,
                                    err_msg=f"cumprod/{c}")
+
+
+def test_rank_na_option_vs_pandas(npartitions):
+    rng = np.random.default_rng(321)
+    n = 40_000
+    k = rng.integers(0, 300, n).astype(np.int64)
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.15] = np.nan
+    w = rng.integers(-9, 9, n).astype(np.int64)
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    for na in ("top", "bottom", "keep"):
+        for method in ("average", "min", "first"):
+            for asc in (True, False):
+                got = df.groupby("k").rank(method=method, ascending=asc,
+                                           na_option=na).to_pandas()
+                exp = pdf.groupby("k").rank(method=method, ascending=asc,
+                                            na_option=na)
+                for c in exp.columns:
+                    np.testing.assert_allclose(
+                        got[c].to_numpy(), exp[c].to_numpy(), rtol=0,
+                        equal_nan=True,
+                        err_msg=f"gb/{na}/{method}/asc={asc}/{c}")
+        got = df[["v", "w"]].rank(na_option=na).to_pandas()
+        exp = pdf[["v", "w"]].rank(na_option=na)
+        for c in exp.columns:
+            np.testing.assert_allclose(got[c].to_numpy(),
+                                       exp[c].to_numpy(), rtol=0,
+                                       equal_nan=True,
+                                       err_msg=f"frame/{na}/{c}")

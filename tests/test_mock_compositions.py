@@ -266,3 +266,32 @@ def test_mock_cumprod(mlib):
                                    exp[c].to_numpy(dtype=float),
                                    rtol=1e-12, atol=1e-300, equal_nan=True,
                                    err_msg=f"cumprod/{c}")
+
+
+def test_mock_rank_na_option(mlib):
+    """rank na_option='top'/'bottom': NaN values take the lowest/highest
+    shared ranks (one tie run at the NaN sentinel) — vs pandas for every
+    (method, ascending) combination, frame and groupby forms."""
+    rng = np.random.default_rng(7)
+    pdf = _frames(rng, n=2500)
+    df = mlib.DataFrame(pdf)
+    for na in ("top", "bottom", "keep"):
+        for method in ("average", "min", "first"):
+            for asc in (True, False):
+                got = df.groupby("k").rank(
+                    method=method, ascending=asc,
+                    na_option=na).to_pandas()
+                exp = pdf.groupby("k").rank(method=method, ascending=asc,
+                                            na_option=na)
+                for c in exp.columns:
+                    np.testing.assert_allclose(
+                        got[c].to_numpy(), exp[c].to_numpy(), rtol=0,
+                        equal_nan=True,
+                        err_msg=f"gb/{na}/{method}/asc={asc}/{c}")
+        got = df[["v", "w"]].rank(na_option=na).to_pandas()
+        exp = pdf[["v", "w"]].rank(na_option=na)
+        for c in exp.columns:
+            np.testing.assert_allclose(got[c].to_numpy(),
+                                       exp[c].to_numpy(), rtol=0,
+                                       equal_nan=True,
+                                       err_msg=f"frame/{na}/{c}")
