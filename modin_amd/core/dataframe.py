@@ -784,6 +784,25 @@ class HipDataframe:
             out[name] = vals
         return out
 
+    def encode_keys_ordered(self, names) -> "HipDataframe":
+        """Copy with the named float64 key columns re-expressed as the
+        order-preserving int64 transform (hf_ordered_i64): all NaNs become
+        ONE canonical sentinel above every finite key — the device form of
+        pandas groupby(dropna=False)'s NaN group, which sorts last."""
+        parts = []
+        for p in self._partitions:
+            b = p.block()
+            cols = dict(b.columns)
+            for nm in names:
+                cols[nm] = lib.ordered_i64(lib.cast_f64(cols[nm]))
+            parts.append(HipDataframePartition(
+                DeviceBlock(cols, b.length, dict(b.cats))))
+        dts = dict(self.dtypes)
+        for nm in names:
+            dts[nm] = np.dtype(np.int64)
+        return HipDataframe(parts, self._index, self.columns,
+                            self._row_lengths, pandas.Series(dts))
+
     def _shuffle_frame_by_key(self, by, with_pos: bool = False):
         """Re-shard the whole frame by the groupby key's sampled ranges
         (hf_shuffle_dest + exchange_column): afterwards every key range —

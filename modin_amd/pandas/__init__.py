@@ -744,31 +744,26 @@ class DataFrameGroupBy:
     def max(self):
         return self._agg("max")
 
-    def _need_dropna(self, what: str):
-        if not self._dropna:
-            raise lib.HfError(
-                f"groupby(dropna=False).{what} is a later round "
-                "(aggs sum/count/mean/min/max and the transform family "
-                "support dropna=False)")
+    def _tail_qc(self, fn_name: str, **kw):
+        """Route the non-reduce aggs through the query compiler with the
+        dropna flag (dropna=False rides the sentinel-NaN-key encoding)."""
+        return self._df._query_compiler.groupby_tail_agg(
+            self._by, fn_name, dropna=self._dropna, **kw)
 
     def var(self, ddof: int = 1):
-        self._need_dropna("var")
-        return DataFrame(query_compiler=self._df._query_compiler.groupby_var(
-            self._by, ddof=ddof))
+        return DataFrame(query_compiler=self._tail_qc("groupby_var",
+                                                      ddof=ddof))
 
     def std(self, ddof: int = 1):
-        self._need_dropna("std")
-        return DataFrame(query_compiler=self._df._query_compiler.groupby_std(
-            self._by, ddof=ddof))
+        return DataFrame(query_compiler=self._tail_qc("groupby_std",
+                                                      ddof=ddof))
 
     def median(self):
         return self._agg("median")
 
     def quantile(self, q: float = 0.5):
-        self._need_dropna("quantile")
         out = DataFrame(
-            query_compiler=self._df._query_compiler.groupby_quantile(
-                self._by, float(q)))
+            query_compiler=self._tail_qc("groupby_quantile", q=float(q)))
         if self._series_out and self._as_index:
             name = list(out._query_compiler._modin_frame.columns)[0]
             return Series(query_compiler=out._query_compiler, name=name)
@@ -777,8 +772,7 @@ class DataFrameGroupBy:
         return out
 
     def nunique(self):
-        self._need_dropna("nunique")
-        qc = self._df._query_compiler.groupby_nunique(self._by)
+        qc = self._tail_qc("groupby_nunique")
         out = DataFrame(query_compiler=qc)
         if self._series_out and self._as_index:
             name = list(qc._modin_frame.columns)[0]
@@ -858,8 +852,7 @@ class DataFrameGroupBy:
     def idxmax(self):
         """Original row label of each group's first max per column
         (all-NaN groups: NaN)."""
-        self._need_dropna("idxmax")
-        qc = self._df._query_compiler.groupby_idxmax(self._by)
+        qc = self._tail_qc("groupby_idxmax")
         out = DataFrame(query_compiler=qc)
         if self._series_out and self._as_index:
             name = list(qc._modin_frame.columns)[0]
@@ -867,8 +860,7 @@ class DataFrameGroupBy:
         return out
 
     def idxmin(self):
-        self._need_dropna("idxmin")
-        qc = self._df._query_compiler.groupby_idxmin(self._by)
+        qc = self._tail_qc("groupby_idxmin")
         out = DataFrame(query_compiler=qc)
         if self._series_out and self._as_index:
             name = list(qc._modin_frame.columns)[0]
@@ -890,10 +882,8 @@ class DataFrameGroupBy:
     def size(self):
         """pandas DataFrameGroupBy.size(): a Series of group row counts
         (NaN values included, NaN keys dropped)."""
-        self._need_dropna("size")
         out = DataFrame(
-            query_compiler=self._df._query_compiler.groupby_size(self._by)
-        ).to_pandas()
+            query_compiler=self._tail_qc("groupby_size")).to_pandas()
         return out["size"].rename(None)
 
     _AGGS = ("sum", "count", "mean", "min", "max", "var", "std",

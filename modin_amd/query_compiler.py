@@ -176,6 +176,42 @@ class HipQueryCompiler:
         vals = self._modin_frame.quantile_columns(list(qs))
         return pandas.DataFrame(vals, index=pandas.Index(list(qs)))
 
+    def groupby_tail_agg(self, by, fn_name: str, dropna: bool = True,
+                         **kw) -> "HipQueryCompiler":
+        """dropna=False for the non-reduce aggs (var/std/median/quantile/
+        nunique/size/first/last/idxmax): NaN float keys are encoded to the
+        order-preserving int64 form whose canonical-NaN sentinel sorts
+        LAST (exactly where pandas puts the NaN group), the agg runs with
+        ordinary int64 keys, and the result index decodes back to float
+        (sentinel -> NaN).  Single-key; string keys stay loud."""
+        fn = getattr(self, fn_name)
+        if dropna:
+            return fn(by, **kw)
+        frame = self._modin_frame
+        keys = [by] if isinstance(by, str) else list(by)
+        cats = (frame._partitions[0].block().cats
+                if frame._partitions else {})
+        if any(b in cats for b in keys):
+            raise lib.HfError(
+                f"groupby(dropna=False).{fn_name} with string keys is a "
+                "later round")
+        f64keys = [b for b in keys
+                   if frame.dtypes[b] == np.dtype(np.float64)]
+        if not f64keys:
+            return fn(by, **kw)  # int64 keys carry no NaN
+        if len(keys) > 1:
+            raise lib.HfError(
+                f"groupby(dropna=False).{fn_name} with multiple float "
+                "keys is a later round")
+        res = self.__constructor__(
+            frame.encode_keys_ordered(f64keys)).__getattribute__(
+                fn_name)(by, **kw)
+        rframe = res._modin_frame
+        ivals = np.asarray(rframe.index)
+        rframe._index = pandas.Index(
+            lib.ordered_to_f64_np(ivals.astype(np.int64)), name=keys[0])
+        return res
+
     def groupby_median(self, by) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.groupby_median(by))
 
@@ -331,8 +367,11 @@ class HipQueryCompiler:
         if agg in ("sum", "count", "mean", "min", "max"):
             return self._tag_key_index(fn(self, by, dropna=dropna), by)
         if not dropna:
-            raise lib.HfError(f"groupby(dropna=False).{agg} is a later "
-                              "round (sum/count/mean/min/max only)")
+            name = {"var": "groupby_var", "std": "groupby_std",
+                    "median": "groupby_median", "first": "groupby_first",
+                    "last": "groupby_last"}[agg]
+            return self._tag_key_index(
+                self.groupby_tail_agg(by, name, dropna=False), by)
         return self._tag_key_index(fn(self, by), by)
 
     # ---- comparisons (query_compiler gt/lt/eq bindings) -> int64 0/1 mask

@@ -2086,3 +2086,48 @@ def test_rank_na_option_vs_pandas(npartitions):
                                        exp[c].to_numpy(), rtol=0,
                                        equal_nan=True,
                                        err_msg=f"frame/{na}/{c}")
+
+
+def test_groupby_dropna_false_tail_aggs(npartitions):
+    """dropna=False for var/std/median/quantile/nunique/size/idxmax/first/
+    last (round-2 completion): NaN float keys form a real trailing group
+    via the sentinel-encoded key (reference pins these in
+    modin/tests/pandas/test_groupby.py)."""
+    rng = np.random.default_rng(55)
+    n = 30_000
+    k = rng.integers(0, 200, n).astype(np.float64)
+    k[rng.random(n) < 0.06] = np.nan
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.1] = np.nan
+    w = rng.integers(-50, 50, n).astype(np.int64)
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    gb = df.groupby("k", dropna=False)
+    egb = pdf.groupby("k", dropna=False)
+    for op in ("var", "std", "median", "nunique", "first", "last"):
+        got = getattr(gb, op)().to_pandas()
+        exp = getattr(egb, op)()
+        np.testing.assert_allclose(got.index.to_numpy(),
+                                   exp.index.to_numpy(), rtol=0,
+                                   equal_nan=True, err_msg=f"{op} keys")
+        for c in exp.columns:
+            np.testing.assert_allclose(
+                got[c].to_numpy(dtype=np.float64),
+                exp[c].to_numpy(dtype=np.float64), rtol=1e-12, atol=1e-9,
+                equal_nan=True, err_msg=f"{op}/{c}")
+    got = gb.quantile(0.25).to_pandas()
+    exp = egb.quantile(0.25)
+    for c in exp.columns:
+        np.testing.assert_allclose(got[c].to_numpy(), exp[c].to_numpy(),
+                                   rtol=1e-12, atol=1e-12, equal_nan=True,
+                                   err_msg=f"quantile/{c}")
+    got = gb.size()
+    exp = egb.size()
+    np.testing.assert_array_equal(np.asarray(got), exp.to_numpy())
+    got = gb.idxmax().to_pandas()
+    exp = egb.idxmax()
+    for c in exp.columns:
+        np.testing.assert_allclose(got[c].to_numpy(dtype=np.float64),
+                                   exp[c].to_numpy(dtype=np.float64),
+                                   rtol=0, equal_nan=True,
+                                   err_msg=f"idxmax/{c}")
