@@ -393,11 +393,12 @@ class HipDataframe:
         agg_op = lib.AGG_OP_OF[agg]
         blk_cats = (self._partitions[0].block().cats
                     if self._partitions else {})
-        bad = [v for v in val_names if v in blk_cats]
-        if bad:
+        dict_vals = [v for v in val_names if v in blk_cats]
+        if dict_vals and agg not in ("min", "max", "count"):
             raise lib.HfError(
-                f"groupby aggregation over string column(s) {bad}: select "
-                "numeric columns (string agg is a later round)")
+                f"groupby {agg} over string column(s) {dict_vals}: pandas "
+                "supports min/max/count on strings (sum concatenation is "
+                "a later round)")
         key_cats = blk_cats.get(by)
         parts = self._partitions
         float_key = (parts and key_cats is None
@@ -459,6 +460,21 @@ class HipDataframe:
                         fparts.append(HipDataframePartition(
                             DeviceBlock(cols, block.length, block.cats)))
                 parts = fparts
+        if dict_vals:
+            # sorted dictionaries: code order == lexicographic order, so
+            # string min/max/count ride the numeric path on codes with
+            # NaN (-1) masked to float-NaN (pandas skips NaN strings)
+            fparts2 = []
+            for p in parts:
+                b2 = p.block()
+                cols2 = dict(b2.columns)
+                for v in dict_vals:
+                    cols2[v] = lib.fixup_empty(
+                        lib.cast_f64(cols2[v]),
+                        lib.compare_scalar(lib.CMP_NE, cols2[v], -1.0))
+                fparts2.append(HipDataframePartition(
+                    DeviceBlock(cols2, b2.length, b2.cats)))
+            parts = fparts2
         keys, sums, counts, n, biases = \
             self._partition_mgr_cls.groupby_reduce(
                 parts, by, val_names, want_counts, agg_op)
@@ -499,9 +515,18 @@ class HipDataframe:
             # (int64 columns can't produce all-NaN groups, so the cast back
             # is safe)
             cols, dts = {}, {}
+            out_val_cats = {}
             for i, name in enumerate(val_names):
                 fixed = lib.fixup_empty(sums[i], counts[i])
-                cols[name], dts[name] = back_to_int(name, fixed)
+                if name in blk_cats:
+                    # decode: NaN -> code -1, back to int codes + the dict
+                    cols[name] = lib.map_scalar(
+                        lib.MAP_CAST_I64,
+                        lib.map_scalar(lib.MAP_FILLNA, fixed, -1.0), 0)
+                    dts[name] = np.dtype(object)
+                    out_val_cats[name] = blk_cats[name]
+                else:
+                    cols[name], dts[name] = back_to_int(name, fixed)
             dtypes = pandas.Series(dts)
         else:  # mean = sums / counts (GroupbyReduceImpl mean shape, groupby.py:87)
             cols = {}
@@ -509,7 +534,9 @@ class HipDataframe:
                 cnt_f = lib.cast_f64(counts[i])
                 cols[name] = lib.binary(lib.BIN_DIV, sums[i], cnt_f)
             dtypes = pandas.Series({n_: np.dtype(np.float64) for n_ in val_names})
-        block = DeviceBlock(cols, n)
+        block = DeviceBlock(cols, n,
+                            out_val_cats if agg in ("min", "max") and
+                            dict_vals else None)
         part = HipDataframePartition(block)
         if float_key:
             idx = pandas.Index(lib.ordered_to_f64_np(lib.get(keys)),

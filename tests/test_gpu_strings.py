@@ -362,3 +362,33 @@ def test_groupby_nunique_vs_golden(npartitions):
     out2 = df[["s", "v"]].groupby("s").nunique().to_pandas()
     assert_str_equal(out2.index.to_numpy(), g["out_sk_keys"], "sk keys")
     np.testing.assert_array_equal(out2["v"].to_numpy(), g["out_sk_v"])
+
+
+def test_string_value_groupby_min_max_count(gpu_ready):
+    """groupby min/max/count over STRING value columns (round-2 lift):
+    sorted dictionaries make code order == lex order, so the numeric
+    code path aggregates and decodes (NaN strings skipped like pandas)."""
+    rng = np.random.default_rng(42)
+    n = 20_000
+    k = rng.integers(0, 150, n).astype(np.int64)
+    words = np.array([f"w{i:04d}" for i in range(400)], dtype=object)
+    sv = words[rng.integers(0, 400, n)]
+    sv[rng.random(n) < 0.1] = np.nan
+    v = rng.random(n)
+    pdf = pandas.DataFrame({"k": k, "s": sv, "v": v})
+    df = mpd.DataFrame(pdf)
+    for op in ("min", "max", "count"):
+        got = getattr(df.groupby("k"), op)().to_pandas()
+        exp = getattr(pdf.groupby("k"), op)()
+        assert list(got.columns) == list(exp.columns)
+        np.testing.assert_array_equal(got.index.to_numpy(),
+                                      exp.index.to_numpy())
+        for c in exp.columns:
+            ge, ee = got[c].to_numpy(), exp[c].to_numpy()
+            if c == "s" and op != "count":
+                same = (pandas.isna(ge) & pandas.isna(ee)) | (ge == ee)
+                assert same.all(), f"{op}/s mismatch"
+            else:
+                np.testing.assert_allclose(ge.astype(np.float64),
+                                           ee.astype(np.float64),
+                                           rtol=1e-12, err_msg=f"{op}/{c}")
