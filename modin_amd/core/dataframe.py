@@ -152,7 +152,7 @@ class HipDataframe:
                 raise lib.HfError(
                     f"column {name!r}: timezone-aware datetimes are a "
                     "later round (convert with tz_localize(None))")
-            if np.issubdtype(dt, np.datetime64):
+            if isinstance(dt, np.dtype) and np.issubdtype(dt, np.datetime64):
                 # typed-column tag (SURVEY §8f.3 gateway): the device
                 # stores the int64 ns view; every sort/groupby/merge/
                 # compare path runs on int64, to_pandas restores the tag
@@ -484,7 +484,9 @@ class HipDataframe:
         # partition manager applied (added back INT-side here).
         int_vals = {n_ for n_ in val_names
                     if (self.dtypes[n_] == np.dtype(np.int64)
-                        or np.issubdtype(self.dtypes[n_], np.datetime64))}
+                        or (isinstance(self.dtypes[n_], np.dtype)
+                            and np.issubdtype(self.dtypes[n_],
+                                              np.datetime64)))}
 
         def back_to_int(name, col):
             if name not in int_vals:

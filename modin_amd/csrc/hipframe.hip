@@ -873,6 +873,31 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
   }
 }
 
+// Build P2 work items ON DEVICE from cursor snapshots (the overlapped
+// path): split every bucket's fresh region slice [lo[b], hi[b]) into
+// <= max_len-row items.  One block; item order is irrelevant.
+__global__ void k_gb_make_work(const unsigned* __restrict__ lo,
+                               const unsigned* __restrict__ hi, int nb,
+                               int max_len, GbWorkItem* __restrict__ out,
+                               int* __restrict__ count) {
+  if (blockIdx.x != 0) return;
+  __shared__ int s_cnt;
+  if (threadIdx.x == 0) s_cnt = 0;
+  __syncthreads();
+  for (int b = threadIdx.x; b < nb; b += blockDim.x) {
+    int64_t s0 = lo[b];
+    const int64_t s1 = hi[b];
+    while (s0 < s1) {
+      const int len = (int)min((int64_t)max_len, s1 - s0);
+      const int i = atomicAdd(&s_cnt, 1);
+      out[i] = GbWorkItem{s0, b, len};
+      s0 += len;
+    }
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) *count = s_cnt;
+}
+
 // P2: one work item per (bucket, chunk); one value column per launch.
 // ROWCNT: also mark group presence (first column / keys-only pass).  The
 // global rowcnt table is PRESENCE-ONLY everywhere (compaction tests >0;
@@ -881,16 +906,23 @@ __global__ void __launch_bounds__(BLK) k_gb_scatter(
 // CNT: also merge per-slot non-NaN counts for this column.
 // 2 rows/lane vectorized (regions are 64-row aligned; odd chunk tails are
 // only ever the last chunk of a bucket).
+// n_work (optional): device item count — the overlapped path launches an
+// upper-bound grid and builds the items on device from cursor snapshots
+// (k_gb_make_work), so extra blocks exit on the count.  Item starts may
+// have any parity there: scalar head/tail rows keep the 2-row vector
+// middle aligned.
 template <bool ROWCNT, bool CNT, bool HAVE_VAL, int RL, int AOP>
 __global__ void __launch_bounds__(512) k_gb_bucket_agg(
     const double* __restrict__ vals, const unsigned short* __restrict__ lowkeys,
-    const GbWorkItem* __restrict__ work, int64_t n_slots,
+    const GbWorkItem* __restrict__ work, const int* __restrict__ n_work,
+    int64_t n_slots,
     double* __restrict__ gsums, unsigned long long* __restrict__ growcnt,
     unsigned long long* __restrict__ gcounts) {
   constexpr int RANGE = 1 << RL;
   __shared__ double lsums[HAVE_VAL ? RANGE : 1];
   __shared__ unsigned lcnt[CNT ? RANGE : 1];
   __shared__ unsigned char ltouch[RANGE];
+  if (n_work && blockIdx.x >= (unsigned)*n_work) return;
   const GbWorkItem w = work[blockIdx.x];
   for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
     if (HAVE_VAL) lsums[s] = agg_identity<AOP>();
@@ -899,76 +931,6 @@ __global__ void __launch_bounds__(512) k_gb_bucket_agg(
   }
   __syncthreads();
   const int64_t end = w.start + w.len;
-  const int64_t npair = (int64_t)w.len >> 1;
-  const ushort2* k2 = reinterpret_cast<const ushort2*>(lowkeys + w.start);
-  const double2* v2 = reinterpret_cast<const double2*>(vals + w.start);
-  for (int64_t i = threadIdx.x; i < npair; i += blockDim.x) {
-    const ushort2 kk = k2[i];
-    ltouch[kk.x] = 1;
-    ltouch[kk.y] = 1;
-    if (HAVE_VAL) {
-      const double2 vv = v2[i];
-      if (vv.x == vv.x) {
-        lds_slot_agg<AOP>(&lsums[kk.x], vv.x);
-        if (CNT) atomicAdd(&lcnt[kk.x], 1u);
-      }
-      if (vv.y == vv.y) {
-        lds_slot_agg<AOP>(&lsums[kk.y], vv.y);
-        if (CNT) atomicAdd(&lcnt[kk.y], 1u);
-      }
-    }
-  }
-  if ((w.len & 1) && threadIdx.x == 0) {
-    const int slot = lowkeys[end - 1];
-    ltouch[slot] = 1;
-    if (HAVE_VAL) {
-      const double v = vals[end - 1];
-      if (v == v) {
-        lds_slot_agg<AOP>(&lsums[slot], v);
-        if (CNT) atomicAdd(&lcnt[slot], 1u);
-      }
-    }
-  }
-  __syncthreads();
-  const int64_t gbase = (int64_t)w.bucket << RL;
-  for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
-    if (!ltouch[s] || gbase + s >= n_slots) continue;
-    if (HAVE_VAL) glob_slot_agg<AOP>(&gsums[gbase + s], lsums[s]);
-    if (ROWCNT) atomicAdd(&growcnt[gbase + s], 1ULL);
-    if (CNT) atomicAdd(&gcounts[gbase + s], (unsigned long long)lcnt[s]);
-  }
-}
-
-// Overlapped-P2 owner form (round 2): block b owns bucket b for the whole
-// step and consumes the payload range [lo[b], hi[b]) — the region slice a
-// P1 scatter chunk just completed (lo/hi are device cursor snapshots, so
-// the host never syncs mid-step).  The merge into the global table is a
-// PLAIN read-modify-write: the sole writer of slice b is this block, and
-// all chunk launches serialize on the consumer stream.  Runs concurrently
-// with the next P1 chunk on the main stream (disjoint payload ranges).
-template <bool ROWCNT, bool CNT, bool HAVE_VAL, int RL, int AOP>
-__global__ void __launch_bounds__(512) k_gb_agg_range(
-    const double* __restrict__ vals, const unsigned short* __restrict__ lowkeys,
-    const unsigned* __restrict__ lo, const unsigned* __restrict__ hi,
-    int64_t n_slots, double* __restrict__ gsums,
-    unsigned long long* __restrict__ growcnt,
-    unsigned long long* __restrict__ gcounts) {
-  constexpr int RANGE = 1 << RL;
-  __shared__ double lsums[HAVE_VAL ? RANGE : 1];
-  __shared__ unsigned lcnt[CNT ? RANGE : 1];
-  __shared__ unsigned char ltouch[RANGE];
-  const int b = blockIdx.x;
-  const int64_t s0 = lo[b];
-  const int64_t s1 = hi[b];
-  if (s1 <= s0) return;
-  for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
-    if (HAVE_VAL) lsums[s] = agg_identity<AOP>();
-    ltouch[s] = 0;
-    if (CNT) lcnt[s] = 0;
-  }
-  __syncthreads();
-  // cursor snapshots have arbitrary parity: scalar head/tail rows keep the
-  // vectorized middle 2-row aligned
   auto one_row = [&](int64_t i) {
     const int slot = lowkeys[i];
     ltouch[slot] = 1;
@@ -980,10 +942,10 @@ __global__ void __launch_bounds__(512) k_gb_agg_range(
       }
     }
   };
-  int64_t a0 = s0;
+  int64_t a0 = w.start;
   if ((a0 & 1) && threadIdx.x == 0) one_row(a0);
   a0 += (a0 & 1);
-  const int64_t npair = (s1 - a0) >> 1;
+  const int64_t npair = (end - a0) >> 1;
   const ushort2* k2 = reinterpret_cast<const ushort2*>(lowkeys + a0);
   const double2* v2 = reinterpret_cast<const double2*>(vals + a0);
   for (int64_t i = threadIdx.x; i < npair; i += blockDim.x) {
@@ -1002,19 +964,14 @@ __global__ void __launch_bounds__(512) k_gb_agg_range(
       }
     }
   }
-  if (((s1 - a0) & 1) && threadIdx.x == 0) one_row(s1 - 1);
+  if (((end - a0) & 1) && threadIdx.x == 0) one_row(end - 1);
   __syncthreads();
-  const int64_t gbase = (int64_t)b << RL;
+  const int64_t gbase = (int64_t)w.bucket << RL;
   for (int s = threadIdx.x; s < RANGE; s += blockDim.x) {
     if (!ltouch[s] || gbase + s >= n_slots) continue;
-    if (HAVE_VAL) {
-      double* a = &gsums[gbase + s];
-      if (AOP == HF_AGG_SUM) *a += lsums[s];
-      else if (AOP == HF_AGG_MIN) *a = fmin(*a, lsums[s]);
-      else *a = fmax(*a, lsums[s]);
-    }
-    if (ROWCNT) growcnt[gbase + s] += 1ULL;
-    if (CNT) gcounts[gbase + s] += (unsigned long long)lcnt[s];
+    if (HAVE_VAL) glob_slot_agg<AOP>(&gsums[gbase + s], lsums[s]);
+    if (ROWCNT) atomicAdd(&growcnt[gbase + s], 1ULL);
+    if (CNT) atomicAdd(&gcounts[gbase + s], (unsigned long long)lcnt[s]);
   }
 }
 
@@ -2847,15 +2804,27 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
                            n_slots, (int)nb, d_cur, r0, r1, rk, d_err);
       });
     };
+    // per-chunk device work items (k_gb_make_work) + the regular
+    // atomic-merge P2 with an upper-bound grid gated by the device count
+    GbWorkItem* d_workc[NCH];
+    int* d_nwork[NCH];
+    uint32_t ub[NCH];
+    for (int c = 0; c < NCH; ++c) {
+      const int64_t rows_c = bounds[c + 1] - bounds[c];
+      ub[c] = (uint32_t)(nb + rows_c / AGG_CHUNK + 1);
+      HF_HIP("gb_radix",
+             dev_alloc((void**)&d_workc[c], ub[c] * sizeof(GbWorkItem),
+                       g.stream));
+      HF_HIP("gb_radix", dev_alloc((void**)&d_nwork[c], 4, g.stream));
+    }
     auto agg_chunk = [&](auto rTag, auto cTag, auto vTag, const double* v,
-                         double* gs, unsigned long long* gc,
-                         const unsigned* clo, const unsigned* chi) {
+                         double* gs, unsigned long long* gc, int c) {
       constexpr bool R = decltype(rTag)::value, C = decltype(cTag)::value,
                      V = decltype(vTag)::value;
       return timed_launch_on("gb_bucket_agg", g.stream2, [&] {
-        hipLaunchKernelGGL((k_gb_agg_range<R, C, V, RL, AOP>),
-                           dim3((uint32_t)nb), dim3(512), 0, g.stream2, v,
-                           rk, clo, chi, n_slots, gs,
+        hipLaunchKernelGGL((k_gb_bucket_agg<R, C, V, RL, AOP>),
+                           dim3(ub[c]), dim3(512), 0, g.stream2, v, rk,
+                           d_workc[c], d_nwork[c], n_slots, gs,
                            (unsigned long long*)rowcnt, gc);
       });
     };
@@ -2880,8 +2849,14 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
       const unsigned* clo =
           c == 0 ? (const unsigned*)keys->d_curinit : d_snap[c - 1];
       const unsigned* chi = d_snap[c];
+      rc = timed_launch_on("gb_make_work", g.stream2, [&] {
+        hipLaunchKernelGGL(k_gb_make_work, dim3(1), dim3(256), 0, g.stream2,
+                           clo, chi, (int)nb, (int)AGG_CHUNK, d_workc[c],
+                           d_nwork[c]);
+      });
+      if (rc != HF_OK) break;
       if (nvals == 0) {
-        rc = agg_chunk(T{}, F{}, F{}, nullptr, nullptr, nullptr, clo, chi);
+        rc = agg_chunk(T{}, F{}, F{}, nullptr, nullptr, nullptr, c);
       } else {
         for (int col = 0; col < nvals && rc == HF_OK; ++col) {
           const double* v = col == 0 ? r0 : r1;
@@ -2890,11 +2865,12 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
               cnt ? (unsigned long long*)counts + (int64_t)col * n_slots
                   : nullptr;
           if (col == 0 && with_rowcnt)
-            rc = cnt ? agg_chunk(T{}, T{}, T{}, v, gs, gc, clo, chi)
-                     : agg_chunk(T{}, F{}, T{}, v, gs, gc, clo, chi);
+            rc = cnt ? agg_chunk(T{}, T{}, T{}, v, gs, gc, c)
+                     : agg_chunk(T{}, F{}, T{}, v, gs, gc, c)
+            ;
           else
-            rc = cnt ? agg_chunk(F{}, T{}, T{}, v, gs, gc, clo, chi)
-                     : agg_chunk(F{}, F{}, T{}, v, gs, gc, clo, chi);
+            rc = cnt ? agg_chunk(F{}, T{}, T{}, v, gs, gc, c)
+                     : agg_chunk(F{}, F{}, T{}, v, gs, gc, c);
         }
       }
     }
@@ -2903,7 +2879,11 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
     // reuse argument intact)
     HF_HIP("gb_radix", hipEventRecord(g.ov_ev[NCH], g.stream2));
     HF_HIP("gb_radix", hipStreamWaitEvent(g.stream, g.ov_ev[NCH], 0));
-    for (int c = 0; c < NCH; ++c) dev_free(d_snap[c], g.stream);
+    for (int c = 0; c < NCH; ++c) {
+      dev_free(d_snap[c], g.stream);
+      dev_free(d_workc[c], g.stream);
+      dev_free(d_nwork[c], g.stream);
+    }
     if (r0) dev_free(r0, g.stream);
     if (r1) dev_free(r1, g.stream);
     dev_free(rk, g.stream);
@@ -2952,8 +2932,8 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
                    V = decltype(vTag)::value;
     return timed_launch("gb_bucket_agg", [&] {
       hipLaunchKernelGGL((k_gb_bucket_agg<R, C, V, RL, AOP>), dim3(agrid),
-                         dim3(512), 0, g.stream, v, rk, d_work, n_slots, gs,
-                         (unsigned long long*)rowcnt, gc);
+                         dim3(512), 0, g.stream, v, rk, d_work, nullptr,
+                         n_slots, gs, (unsigned long long*)rowcnt, gc);
     });
   };
   using T = std::true_type;

@@ -122,7 +122,8 @@ class HipQueryCompiler:
             by[0] if isinstance(by, (list, tuple)) and len(by) == 1 else None)
         if key is not None:
             dt = dict(self._modin_frame.dtypes).get(key)
-            if dt is not None and np.issubdtype(dt, np.datetime64):
+            if (dt is not None and isinstance(dt, np.dtype)
+                    and np.issubdtype(dt, np.datetime64)):
                 res._modin_frame._index_dtype = dt
         return res
 
