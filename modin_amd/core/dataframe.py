@@ -333,10 +333,13 @@ class HipDataframe:
         for sp in spans:
             total *= sp
             if total > (1 << 62):
-                raise lib.HfError(
-                    "groupby: combined key range of "
-                    f"{by_list} exceeds 2^62 (tuple-hash keys are a "
-                    "later round)")
+                # unbounded combined span: dense-rank fallback — NO tuple
+                # hash needed (the pinned sorted-heads composition,
+                # test_host_logic.py): per-column stable sort + OR'd run
+                # heads give each distinct key tuple a dense ordinal in
+                # lexicographic order
+                return self._combined_key_frame_sorted(by_list, parts,
+                                                       blk_cats)
         strides = [1] * len(by_list)
         for i in range(len(by_list) - 2, -1, -1):
             strides[i] = strides[i + 1] * spans[i + 1]
@@ -366,6 +369,72 @@ class HipDataframe:
             levels = []
             for b, mn, st, sp in zip(by_list, mins, strides, spans):
                 lv = (keys_np // st) % sp + mn
+                if b in blk_cats:
+                    lv = decode_dict(lv, blk_cats[b])
+                levels.append(lv)
+            return pandas.MultiIndex.from_arrays(levels, names=by_list)
+
+        return frame, decode
+
+    def _combined_key_frame_sorted(self, by_list, parts, blk_cats):
+        """Unbounded multi-key fold: dense group ordinals from one stable
+        multi-column sort — per-column effective keys compose the sort,
+        OR'd run-head flags delimit distinct tuples, cumsum(heads)-1 is
+        the dense lexicographic group id, scattered back to row order.
+        decode() gathers each original key column at the group's first
+        sorted occurrence.  Covers ANY int64/string key combination (the
+        pinned sorted-heads prototype); single-rank (group ids are
+        rank-local ordinals)."""
+        from ..distributed import is_active
+        if is_active():
+            raise lib.HfError(
+                "multi-key groupby beyond a 2^62 combined span at world>1 "
+                "is a later round")
+
+        def concat_col(name):
+            cs = [p.block().columns[name] for p in parts]
+            return cs[0] if len(cs) == 1 else lib.concat(cs)
+
+        n = sum(p.block().length for p in parts)
+        eff_keys = [self._effective_sort_key(concat_col(b), b in blk_cats,
+                                             True)
+                    for b in by_list]
+        perm = self._compose_sort_perm(eff_keys)
+        khead = None
+        for ekc, _ in eff_keys:
+            h = self._run_head_col(lib.gather(ekc, perm), n)
+            khead = h if khead is None else lib.binary(lib.BIN_ADD, khead,
+                                                       h)
+        if len(eff_keys) > 1:
+            khead = lib.compare_scalar(lib.CMP_GE, khead, 1.0)
+        gid_sorted = lib.map_scalar(lib.MAP_SUB, lib.cumsum(khead), 1)
+        gid = lib.scatter(gid_sorted, perm)
+        rep_sorted = lib.filter_iota(lib.filter_plan(khead), 0)
+        rep_orig = lib.gather(perm, rep_sorted)   # [ngroups] original rows
+        key_cols = {b: concat_col(b) for b in by_list}
+        new_parts = []
+        off = 0
+        for p in parts:
+            block = p.block()
+            cols = dict(block.columns)
+            cols[self.KEYCOL] = lib.col_slice(gid, off, block.length)
+            off += block.length
+            new_parts.append(HipDataframePartition(
+                DeviceBlock(cols, block.length, block.cats)))
+        columns = list(self.columns) + [self.KEYCOL]
+        dtypes = pandas.concat([self.dtypes, pandas.Series(
+            {self.KEYCOL: np.dtype(np.int64)})])
+        frame = HipDataframe(new_parts, self._index, columns,
+                             [p.block().length for p in new_parts]
+                             if new_parts else [0], dtypes)
+
+        def decode(keys_np):
+            from .partition import decode_dict
+            rows = lib.gather(rep_orig,
+                              lib.put(np.asarray(keys_np, dtype=np.int64)))
+            levels = []
+            for b in by_list:
+                lv = lib.get(lib.gather(key_cols[b], rows))
                 if b in blk_cats:
                     lv = decode_dict(lv, blk_cats[b])
                 levels.append(lv)

@@ -2763,13 +2763,20 @@ int gb_radix_path(hf_col* keys, const GbPtrs& ptrs, int nvals, int64_t key_min,
   HF_HIP("gb_radix", hipMemcpyAsync(d_cur, keys->d_curinit, nb * 4,
                                     hipMemcpyDeviceToDevice, g.stream));
 
-  // ---- overlapped path (round 2): scatter in 4 chunks on the main
-  // stream; after each chunk the consumer stream aggregates the region
-  // slices that chunk completed (owner-block k_gb_agg_range, plain RMW —
-  // cursor snapshots travel device-to-device).  The last chunk is small
-  // (10%) so the only non-overlapped P2 work is cheap; the rest of P2
-  // rides bandwidth the latency-bound scatter leaves unused.
-  const bool overlap = n >= (64LL << 20) && nvals <= 2 && nb <= 1024;
+  // ---- overlapped path: scatter in 4 chunks on the main stream; after
+  // each chunk the consumer stream aggregates the region slices that
+  // chunk completed (device-built work items from cursor snapshots).
+  // MEASURED NEGATIVE (round 2, profiles/r02): both kernels want >72 KB
+  // LDS, so concurrent scheduling PARTITIONS the CUs between them and
+  // each phase runs ~1.5x slower (uniform 9.2 ms vs 7.0 serial) — the
+  // overlap never recovers the loss.  Kept behind HF_GB_OVERLAP=1 for
+  // future experiments; default is the serial two-phase path.
+  static const bool overlap_env = [] {
+    const char* e = getenv("HF_GB_OVERLAP");
+    return e && e[0] == '1';
+  }();
+  const bool overlap =
+      overlap_env && n >= (64LL << 20) && nvals <= 2 && nb <= 1024;
   if (overlap) {
     const int NCH = 4;
     const double fracs[NCH] = {0.30, 0.30, 0.30, 0.10};

@@ -80,12 +80,6 @@ def install(monkeypatch):
         dt = np.int64 if dtype_code == HF_INT64 else np.float64
         return _reg(MockCol(np.zeros(length, dtype=dt)))
 
-    def fill_f64(dptr, value, n):
-        _by_ptr[dptr].arr[:n] = value
-
-    def fill_i64(dptr, value, n):
-        _by_ptr[dptr].arr[:n] = value
-
     def col_slice(col, start, length):
         return _reg(MockCol(col.arr[start:start + length].copy()))
 
@@ -289,6 +283,141 @@ def install(monkeypatch):
     def ensure_ready(gpu=None):
         pass
 
+    # ---- dense-table groupby engine (hf_alloc_raw / hf_groupby_accum /
+    # hf_groupby_compact and the hash/sorted variants) so the WHOLE
+    # groupby-reduce composition runs on CPU ----
+    _raw = {}
+    _raw_next = [1 << 40]
+
+    def alloc_raw(nbytes):
+        ptr = _raw_next[0]
+        _raw_next[0] += ((nbytes + 255) & ~255) + 256
+        _raw[ptr] = np.zeros(nbytes, dtype=np.uint8)
+        return ptr
+
+    def free_raw(ptr):
+        _raw.pop(ptr, None)
+
+    def memset_raw(ptr, value, nbytes):
+        _raw[ptr][:nbytes] = value
+
+    def _rawview(ptr, dtype, count):
+        return _raw[ptr][: count * 8].view(dtype)
+
+    def fill_f64(dptr, value, n):
+        if dptr in _raw:
+            _rawview(dptr, np.float64, n)[:n] = value
+        else:
+            _by_ptr[dptr].arr[:n] = value
+
+    def fill_i64(dptr, value, n):
+        if dptr in _raw:
+            _rawview(dptr, np.int64, n)[:n] = value
+        else:
+            _by_ptr[dptr].arr[:n] = value
+
+    def groupby_accum(keys, vals, agg_op, key_min, n_slots, sums, rowcnt,
+                      counts):
+        k = keys.arr - key_min
+        ok = (k >= 0) & (k < n_slots)
+        k = k[ok]
+        rc = _rawview(rowcnt, np.int64, n_slots)
+        rc += np.bincount(k, minlength=n_slots)
+        sv = _rawview(sums, np.float64, n_slots * max(len(vals), 1))
+        cv = (_rawview(counts, np.int64, n_slots * max(len(vals), 1))
+              if counts else None)
+        for c, vcol in enumerate(vals):
+            x = vcol.arr[ok].astype(np.float64)
+            m = ~np.isnan(x)
+            sl = sv[c * n_slots:(c + 1) * n_slots]
+            if agg_op == 0:
+                sl += np.bincount(k[m], weights=x[m], minlength=n_slots)
+            else:
+                fn2 = np.minimum if agg_op == 1 else np.maximum
+                np_fn = fn2.at
+                np_fn(sl, k[m], x[m])
+            if cv is not None:
+                cl = cv[c * n_slots:(c + 1) * n_slots]
+                cl += np.bincount(k[m], minlength=n_slots)
+
+    def groupby_compact(sums, rowcnt, counts, nvals, key_min, n_slots):
+        rc = _rawview(rowcnt, np.int64, n_slots)
+        present = np.nonzero(rc > 0)[0]
+        n = present.size
+        kcol = _reg(MockCol(present + key_min))
+        sv = _rawview(sums, np.float64, n_slots * max(nvals, 1))
+        scols = [_reg(MockCol(sv[c * n_slots:(c + 1) * n_slots]
+                              [present].copy())) for c in range(nvals)]
+        ccols = None
+        if counts:
+            cvv = _rawview(counts, np.int64, n_slots * max(nvals, 1))
+            ccols = [_reg(MockCol(cvv[c * n_slots:(c + 1) * n_slots]
+                                  [present].copy())) for c in range(nvals)]
+        return kcol, scols, ccols, n
+
+    def groupby_sorted(sorted_keys, vals, agg_op, want_counts):
+        k = sorted_keys.arr
+        if k.size == 0:
+            e = _reg(MockCol(np.empty(0, dtype=np.int64)))
+            return e, [], [] if want_counts else None, 0
+        uniq, inv = np.unique(k, return_inverse=True)
+        n = uniq.size
+        kcol = _reg(MockCol(uniq))
+        scols, ccols = [], []
+        for vcol in vals:
+            x = vcol.arr.astype(np.float64)
+            m = ~np.isnan(x)
+            if agg_op == 0:
+                sl = np.bincount(inv[m], weights=x[m], minlength=n)
+            else:
+                ident = np.inf if agg_op == 1 else -np.inf
+                sl = np.full(n, ident)
+                (np.minimum if agg_op == 1 else np.maximum).at(
+                    sl, inv[m], x[m])
+            scols.append(_reg(MockCol(sl)))
+            if want_counts:
+                ccols.append(_reg(MockCol(
+                    np.bincount(inv[m], minlength=n).astype(np.int64))))
+        return kcol, scols, (ccols if want_counts else None), n
+
+    _hash_store = {}
+
+    def groupby_hash_accum(keys, vals, agg_op, H, tkey, sums, rowcnt,
+                           counts):
+        st = _hash_store.setdefault(tkey, {
+            "k": [], "v": [[] for _ in vals], "agg": agg_op,
+        })
+        st["k"].append(keys.arr.copy())
+        for c, vcol in enumerate(vals):
+            st["v"][c].append(vcol.arr.astype(np.float64))
+
+    def groupby_hash_compact(tkey, sums, rowcnt, counts, nvals, H):
+        st = _hash_store.pop(tkey, {"k": [], "v": [[] for _ in range(nvals)],
+                                    "agg": 0})
+        k = (np.concatenate(st["k"]) if st["k"]
+             else np.empty(0, dtype=np.int64))
+        uniq, inv = np.unique(k, return_inverse=True)
+        n = uniq.size
+        agg = st["agg"]
+        scols, ccols = [], []
+        for c in range(nvals):
+            x = (np.concatenate(st["v"][c]) if st["v"][c]
+                 else np.empty(0, dtype=np.float64))
+            m = ~np.isnan(x)
+            if agg == 0:
+                sl = np.bincount(inv[m], weights=x[m], minlength=n)
+            else:
+                ident = np.inf if agg == 1 else -np.inf
+                sl = np.full(n, ident)
+                (np.minimum if agg == 1 else np.maximum).at(sl, inv[m],
+                                                            x[m])
+            scols.append(_reg(MockCol(sl)))
+            if counts:
+                ccols.append(_reg(MockCol(
+                    np.bincount(inv[m], minlength=n).astype(np.int64))))
+        return (_reg(MockCol(uniq)), scols,
+                (ccols if counts else None), n)
+
     for name, fn in [
         ("put", put), ("get", get), ("alloc", alloc),
         ("fill_f64", fill_f64), ("fill_i64", fill_i64),
@@ -302,5 +431,11 @@ def install(monkeypatch):
         ("fixup_empty", fixup_empty), ("ordered_i64", ordered_i64),
         ("search_sorted", search_sorted), ("cross_idx", cross_idx),
         ("sync", sync), ("ensure_ready", ensure_ready),
+        ("alloc_raw", alloc_raw), ("free_raw", free_raw),
+        ("memset_raw", memset_raw), ("groupby_accum", groupby_accum),
+        ("groupby_compact", groupby_compact),
+        ("groupby_sorted", groupby_sorted),
+        ("groupby_hash_accum", groupby_hash_accum),
+        ("groupby_hash_compact", groupby_hash_compact),
     ]:
         monkeypatch.setattr(lib, name, fn)

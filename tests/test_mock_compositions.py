@@ -295,3 +295,24 @@ def test_mock_rank_na_option(mlib):
                                        exp[c].to_numpy(), rtol=0,
                                        equal_nan=True,
                                        err_msg=f"frame/{na}/{c}")
+
+
+def test_mock_unbounded_multikey_groupby(mlib):
+    """Multi-key groupby whose combined span exceeds 2^62 takes the
+    sorted-heads dense-rank fold (no tuple hash) — vs pandas."""
+    rng = np.random.default_rng(17)
+    n = 5000
+    k1 = rng.integers(-2**61, 2**61, n)
+    k1[rng.random(n) < 0.3] = 77  # heavy dup runs
+    k2 = rng.integers(-2**61, 2**61, n)
+    k2[rng.random(n) < 0.3] = -5
+    v = rng.standard_normal(n)
+    pdf = pandas.DataFrame({"a": k1, "b": k2, "v": v})
+    df = mlib.DataFrame(pdf)
+    for agg in ("sum", "count", "mean"):
+        got = getattr(df.groupby(["a", "b"]), agg)().to_pandas()
+        exp = getattr(pdf.groupby(["a", "b"]), agg)()
+        assert list(got.index) == list(exp.index), f"{agg} keys"
+        np.testing.assert_allclose(got["v"].to_numpy(),
+                                   exp["v"].to_numpy(), rtol=1e-12,
+                                   err_msg=agg)
