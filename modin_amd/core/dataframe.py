@@ -2090,17 +2090,38 @@ class HipDataframe:
             return HipDataframe([part], res0._index, val_names, [0],
                                 pandas.Series({v: np.dtype(np.int64)
                                                for v in val_names}))
+        posmap = None
         if len(by_list) > 1:
-            # trailing-sentinel slicing below assumes NaN-key rows sort
-            # last, which only holds for a single key column
+            # the trailing-sentinel slicing below assumes NaN-key rows
+            # sort last, which only holds for one key column — multi-key
+            # NaN keys FILTER FIRST instead (the pinned prototype,
+            # test_host_logic.py:330), with original positions restored
+            # through the kept-row map afterwards
+            nan_masks = []
             for b in by_list:
                 c = concat_col(b)
-                bad = (lib.reduce(c).imn < 0) if b in blk_cats else (
-                    c.dtype_code == lib.HF_FLOAT64
-                    and lib.reduce(c).count != c.length)
-                if c.length and bad:
-                    raise lib.HfError("idxmax/idxmin: NaN in a multi-key "
-                                      "groupby key is a later round")
+                if not c.length:
+                    continue
+                if b in blk_cats:
+                    if lib.reduce(c).imn < 0:
+                        nan_masks.append(
+                            lib.compare_scalar(lib.CMP_GE, c, 0.0))
+                elif (c.dtype_code == lib.HF_FLOAT64
+                      and lib.reduce(c).count != c.length):
+                    nan_masks.append(
+                        lib.compare_scalar(lib.CMP_NOTNA, c, 0.0))
+            if nan_masks:
+                acc = nan_masks[0]
+                for m2 in nan_masks[1:]:
+                    acc = lib.binary(lib.BIN_MUL, acc, m2)
+                plan = lib.filter_plan(acc)
+                posmap = lib.filter_iota(plan, 0)
+                fcols = {name: lib.filter_apply(plan, concat_col(name))
+                         for name in list(by_list) + list(val_names)}
+                n = plan.n_kept
+
+                def concat_col(name, _f=fcols):  # noqa: F811
+                    return _f[name]
         eff_keys = [self._effective_sort_key(concat_col(b), b in blk_cats,
                                              True)
                     for b in by_list]
@@ -2149,6 +2170,8 @@ class HipDataframe:
             else:
                 ts = hp  # ascending sort: the min's first occurrence
             orig = lib.gather(perm, ts)               # [ng]
+            if posmap is not None:
+                orig = lib.gather(posmap, orig)  # back to original rows
             # keep only the valid (non-NaN-key) groups: the sentinel runs
             # sort last, so they are the trailing ng - ngv runs
             orig = lib.col_slice(orig, 0, ngv)

@@ -2131,3 +2131,45 @@ def test_groupby_dropna_false_tail_aggs(npartitions):
                                    exp[c].to_numpy(dtype=np.float64),
                                    rtol=0, equal_nan=True,
                                    err_msg=f"idxmax/{c}")
+
+
+def test_multikey_idx_nan_keys(npartitions):
+    """Multi-key idxmax/idxmin with NaN string keys (filter-first lift)."""
+    rng = np.random.default_rng(3)
+    n = 20_000
+    a = rng.choice(["x", "y", "z", None], n,
+                   p=[0.3, 0.3, 0.3, 0.1]).astype(object)
+    b = rng.integers(0, 10, n)
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.1] = np.nan
+    pdf = pandas.DataFrame({"a": a, "b": b, "v": v})
+    df = mpd.DataFrame(pdf)
+    for mx in (True, False):
+        got = (df.groupby(["a", "b"]).idxmax() if mx
+               else df.groupby(["a", "b"]).idxmin()).to_pandas()
+        exp = (pdf.groupby(["a", "b"]).idxmax() if mx
+               else pdf.groupby(["a", "b"]).idxmin())
+        assert list(got.index) == list(exp.index)
+        np.testing.assert_allclose(got["v"].to_numpy().astype(float),
+                                   exp["v"].to_numpy().astype(float),
+                                   rtol=0, equal_nan=True)
+
+
+def test_unbounded_multikey_groupby(npartitions):
+    """Combined key span beyond 2^62: the sorted-heads dense-rank fold."""
+    rng = np.random.default_rng(17)
+    n = 200_000
+    k1 = rng.integers(-2**61, 2**61, n)
+    k1[rng.random(n) < 0.3] = 77
+    k2 = rng.integers(-2**61, 2**61, n)
+    k2[rng.random(n) < 0.3] = -5
+    v = rng.standard_normal(n)
+    pdf = pandas.DataFrame({"a": k1, "b": k2, "v": v})
+    df = mpd.DataFrame(pdf)
+    for agg in ("sum", "count", "mean"):
+        got = getattr(df.groupby(["a", "b"]), agg)().to_pandas()
+        exp = getattr(pdf.groupby(["a", "b"]), agg)()
+        assert list(got.index) == list(exp.index), f"{agg} keys"
+        np.testing.assert_allclose(got["v"].to_numpy(),
+                                   exp["v"].to_numpy(), rtol=1e-12,
+                                   err_msg=agg)

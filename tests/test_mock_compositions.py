@@ -316,3 +316,26 @@ def test_mock_unbounded_multikey_groupby(mlib):
         np.testing.assert_allclose(got["v"].to_numpy(),
                                    exp["v"].to_numpy(), rtol=1e-12,
                                    err_msg=agg)
+
+
+def test_mock_multikey_idx_nan_keys(mlib):
+    """Multi-key idxmax/idxmin with NaN keys: filter-first composition
+    (pinned prototype lifted) vs pandas."""
+    rng = np.random.default_rng(3)
+    n = 8000
+    a = rng.choice(["x", "y", "z", None], n,
+                   p=[0.3, 0.3, 0.3, 0.1]).astype(object)
+    b = rng.integers(0, 10, n)
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.1] = np.nan
+    pdf = pandas.DataFrame({"a": a, "b": b, "v": v})
+    df = mlib.DataFrame(pdf)
+    for mx in (True, False):
+        got = (df.groupby(["a", "b"]).idxmax() if mx
+               else df.groupby(["a", "b"]).idxmin()).to_pandas()
+        exp = (pdf.groupby(["a", "b"]).idxmax() if mx
+               else pdf.groupby(["a", "b"]).idxmin())
+        assert list(got.index) == list(exp.index)
+        np.testing.assert_allclose(got["v"].to_numpy().astype(float),
+                                   exp["v"].to_numpy().astype(float),
+                                   rtol=0, equal_nan=True)
