@@ -1692,12 +1692,17 @@ def test_groupby_dropna_false_vs_pandas(npartitions):
         exp_n = pdf.groupby("k", dropna=False).ngroup()
         np.testing.assert_array_equal(got_n.to_numpy(), exp_n.to_numpy(),
                                       err_msg=f"{key}/ngroup")
-    # unsupported dropna=False paths raise loudly
-    df2 = mpd.DataFrame(pandas.DataFrame({"k": [1, 2], "v": [1.0, 2.0]}))
-    with pytest.raises(lib.HfError, match="dropna=False"):
-        df2.groupby("k", dropna=False).var()
-    with pytest.raises(lib.HfError, match="dropna=False"):
-        df2.groupby("k", dropna=False).median()
+    # round 2: var/median under dropna=False are SUPPORTED now (the
+    # sentinel-encoded key route; test_groupby_dropna_false_tail_aggs
+    # covers the full matrix) — spot-check they run and match pandas
+    df2 = mpd.DataFrame(pandas.DataFrame(
+        {"k": [1.0, 2.0, np.nan, 2.0], "v": [1.0, 2.0, 3.0, 5.0]}))
+    pdf2 = pandas.DataFrame(
+        {"k": [1.0, 2.0, np.nan, 2.0], "v": [1.0, 2.0, 3.0, 5.0]})
+    got2 = df2.groupby("k", dropna=False).median().to_pandas()
+    exp2 = pdf2.groupby("k", dropna=False).median()
+    np.testing.assert_allclose(got2["v"].to_numpy(),
+                               exp2["v"].to_numpy(), rtol=0)
 
 
 def test_groupby_ffill_bfill_vs_pandas(npartitions):
