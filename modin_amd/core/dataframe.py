@@ -1391,10 +1391,17 @@ class HipDataframe:
         f64 subtract for NaN-free values), filter to the distinct pairs
         and size-count them per key; groups whose values are all NaN
         report 0 (pandas dropna=True)."""
-        from ..distributed import is_active
-        if is_active():
-            raise lib.HfError("distributed groupby.nunique is a later "
-                              "round")
+        from .. import distributed as dist_mod
+        if dist_mod.is_active():
+            key = by if isinstance(by, str) else \
+                (by[0] if len(by) == 1 else None)
+            if key is None:
+                raise lib.HfError("distributed multi-key nunique is a "
+                                  "later round")
+            shuf = self._shuffle_frame_by_key(key)
+            with dist_mod.local_mode():
+                res = shuf.groupby_nunique(key)
+            return self._replicate_result_frame(res)
         if isinstance(by, (list, tuple)):
             if len(by) == 1:
                 by = by[0]
@@ -2060,10 +2067,39 @@ class HipDataframe:
         Composition: stable sort by (key, value) — the extreme's tie block
         start IS the first original occurrence; non-NaN counts locate the
         last non-NaN row (values sort NaN-last within each key run)."""
-        from ..distributed import is_active
-        if is_active():
-            raise lib.HfError("distributed groupby idxmax/idxmin is a "
-                              "later round")
+        from .. import distributed as dist_mod
+        if dist_mod.is_active():
+            key = by if isinstance(by, str) else \
+                (by[0] if len(by) == 1 else None)
+            if key is None:
+                raise lib.HfError("distributed multi-key idxmax/idxmin "
+                                  "is a later round")
+            # shuffle WITH global positions: local row order after the
+            # exchange is global-position-ascending, so the local pick is
+            # the right ROW; its label maps through the position column
+            shuf, pos_col = self._shuffle_frame_by_key(key, with_pos=True)
+            with dist_mod.local_mode():
+                res = shuf.groupby_idxminmax(key, maximum)
+            blk = res._partitions[0].block()
+            cols2 = {}
+            dts2 = {}
+            for nm, col in blk.columns.items():
+                if col.dtype_code == lib.HF_FLOAT64:
+                    ok = lib.compare_scalar(lib.CMP_NOTNA, col, 0.0)
+                    idx_i = lib.map_scalar(
+                        lib.MAP_CAST_I64,
+                        lib.map_scalar(lib.MAP_FILLNA, col, 0.0), 0)
+                    mapped = lib.gather(pos_col, idx_i)
+                    cols2[nm] = lib.fixup_empty(lib.cast_f64(mapped), ok)
+                    dts2[nm] = np.dtype(np.float64)
+                else:
+                    cols2[nm] = lib.gather(pos_col, col)
+                    dts2[nm] = np.dtype(np.int64)
+            res = HipDataframe(
+                [HipDataframePartition(DeviceBlock(cols2, blk.length))],
+                res._index, list(res.columns), [blk.length],
+                pandas.Series(dts2))
+            return self._replicate_result_frame(res)
         by_list = [by] if isinstance(by, str) else list(by)
         blk_cats = (self._partitions[0].block().cats
                     if self._partitions else {})

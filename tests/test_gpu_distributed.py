@@ -136,6 +136,24 @@ def _worker(rank, world, port, fail_q):
                     expect[c].to_numpy(dtype=np.float64), rtol=1e-12,
                     atol=1e-12, equal_nan=True, err_msg=f"{op}/{c}")
 
+        # ---- distributed nunique + idxmax/idxmin (round-2 lift) ----
+        out = qdf.groupby("k").nunique().to_pandas()
+        expect = qpdf.groupby("k").nunique()
+        for c in ("v", "w"):
+            np.testing.assert_array_equal(out[c].to_numpy(),
+                                          expect[c].to_numpy(),
+                                          err_msg=f"nunique/{c}")
+        for mx in (True, False):
+            out = (qdf.groupby("k").idxmax() if mx
+                   else qdf.groupby("k").idxmin()).to_pandas()
+            expect = (qpdf.groupby("k").idxmax() if mx
+                      else qpdf.groupby("k").idxmin())
+            for c in ("v", "w"):
+                np.testing.assert_allclose(
+                    out[c].to_numpy(dtype=np.float64),
+                    expect[c].to_numpy(dtype=np.float64), rtol=0,
+                    equal_nan=True, err_msg=f"idx mx={mx}/{c}")
+
         # ---- distributed transforms (shuffle + local + route-back):
         # row-aligned results on each rank's own shard ----
         tdf_p = qpdf.iloc[qlo:qhi].reset_index(drop=True)
@@ -221,5 +239,86 @@ def test_world2_groupby_dense_and_shuffle_on_gpu(gpu_ready):
     for p in procs:
         if p.is_alive():
             p.terminate()
+    assert not errs, "\n".join(errs)
+    assert all(p.exitcode == 0 for p in procs)
+
+
+def _nccl_worker(rank, world, port, fail_q):
+    """Exercise the RCCL branch itself (world-2 on ONE GPU): the dense
+    table all-reduce over nccl, the all_to_all_single exchange, the
+    zero-copy CAI send path of exchange_column, and the varlen gather —
+    the code that gloo's host-bounce fallback skips."""
+    try:
+        os.environ.update({
+            "RANK": str(rank), "WORLD_SIZE": str(world),
+            "LOCAL_RANK": "0",  # both ranks on the single visible GPU
+            "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        })
+        import torch
+        import modin_amd.distributed as dist_mod
+        from modin_amd.core import lib
+        assert dist_mod.init_from_env(backend="nccl", gpu=True)
+        assert dist_mod._state["backend"] == "nccl"
+        lib.ensure_ready(0)
+
+        # dense-table all-reduce on CUDA tensors (RCCL path)
+        n_slots = 1000
+        rng = np.random.default_rng(100 + rank)
+        sums = torch.tensor(rng.random(n_slots), dtype=torch.float64,
+                            device="cuda:0")
+        host_sums = sums.cpu().numpy().copy()
+        rowcnt = torch.ones(n_slots, dtype=torch.int64, device="cuda:0")
+
+        class T:
+            _torch_tensors = (sums, rowcnt, None)
+            agg_op = 0
+
+        dist_mod.maybe_allreduce_table(T)
+        both = (np.random.default_rng(100).random(n_slots)
+                + np.random.default_rng(101).random(n_slots))
+        np.testing.assert_allclose(T._torch_tensors[0].cpu().numpy(), both,
+                                   rtol=1e-12)
+        assert int(T._torch_tensors[1][0].item()) == world
+
+        # exchange_column: zero-copy CAI send + device receive
+        vals = np.arange(1000, dtype=np.float64) + 10_000 * rank
+        col = lib.put(vals)
+        out = dist_mod.exchange_column(col, [500, 500])
+        got = lib.get(out)
+        exp = np.concatenate([np.arange(500 * rank, 500 * (rank + 1),
+                                        dtype=np.float64) + 10_000 * src
+                              for src in range(world)])
+        np.testing.assert_array_equal(got, exp)
+
+        # varlen all-gather on the nccl device path
+        arrs = dist_mod.allgather_arrays(
+            [np.arange(rank + 3, dtype=np.int64)])
+        np.testing.assert_array_equal(
+            arrs[0], np.concatenate([np.arange(src + 3)
+                                     for src in range(world)]))
+        dist_mod.shutdown()
+    except Exception as e:  # pragma: no cover
+        import traceback
+        fail_q.put(f"rank {rank}: {e}\n{traceback.format_exc()}")
+
+
+@pytest.mark.timeout(240)
+def test_world2_nccl_branch_single_gpu(gpu_ready):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29571
+    procs = [ctx.Process(target=_nccl_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=200)
+    errs = []
+    while not q.empty():
+        errs.append(q.get())
+    for p in procs:
+        if p.is_alive():
+            p.terminate()
+            errs.append("nccl worker hung (RCCL world-2 on one GPU)")
     assert not errs, "\n".join(errs)
     assert all(p.exitcode == 0 for p in procs)
