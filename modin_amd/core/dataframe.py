@@ -266,7 +266,7 @@ class HipDataframe:
 
     KEYCOL = "\x00key\x00"  # internal combined multi-key column
 
-    def _combined_key_frame(self, by_list):
+    def _combined_key_frame(self, by_list, dropna: bool = True):
         """Multi-key groupby support: fold the key columns into ONE int64
         key (k1*span2*span3… + k2*span3… + …, mins subtracted — GLOBAL
         mins at world>1 so every rank combines identically), giving the
@@ -287,6 +287,7 @@ class HipDataframe:
                 raise lib.HfError(f"groupby: key column {b!r} missing")
         parts = self._partitions
         dict_keys = [b for b in by_list if b in blk_cats]
+        nan_dict_keys = set()
         if dict_keys:
             need = False
             for p in parts:
@@ -294,7 +295,8 @@ class HipDataframe:
                     c = p.block().columns[b]
                     if c.length and lib.reduce(c).imn < 0:
                         need = True
-            if need:
+                        nan_dict_keys.add(b)
+            if need and dropna:
                 fparts = []
                 for p in parts:
                     block = p.block()
@@ -309,6 +311,24 @@ class HipDataframe:
                             for m2, c in block.columns.items()}
                     fparts.append(HipDataframePartition(
                         DeviceBlock(cols, plan.n_kept, block.cats)))
+                parts = fparts
+            elif need:
+                # dropna=False: NaN string keys become one extra code at
+                # the TOP of the column's code space (sorts last per
+                # level, pandas' NaN placement); decode maps it back
+                fparts = []
+                for p in parts:
+                    block = p.block()
+                    cols = dict(block.columns)
+                    for b in nan_dict_keys:
+                        ncats = len(blk_cats[b])
+                        m = lib.compare_scalar(lib.CMP_GE, cols[b], 0.0)
+                        isna = lib.map_scalar(lib.MAP_RSUB, m, 1)
+                        cols[b] = lib.binary(
+                            lib.BIN_ADD, cols[b],
+                            lib.map_scalar(lib.MAP_MUL, isna, ncats + 1))
+                    fparts.append(HipDataframePartition(
+                        DeviceBlock(cols, block.length, block.cats)))
                 parts = fparts
         mins, spans = [], []
         for b in by_list:
@@ -370,6 +390,8 @@ class HipDataframe:
             for b, mn, st, sp in zip(by_list, mins, strides, spans):
                 lv = (keys_np // st) % sp + mn
                 if b in blk_cats:
+                    if b in nan_dict_keys and not dropna:
+                        lv = np.where(lv >= len(blk_cats[b]), -1, lv)
                     lv = decode_dict(lv, blk_cats[b])
                 levels.append(lv)
             return pandas.MultiIndex.from_arrays(levels, names=by_list)
@@ -449,10 +471,8 @@ class HipDataframe:
             if len(by) == 1:
                 by = by[0]
             else:
-                if not dropna:
-                    raise lib.HfError("multi-key groupby(dropna=False) is "
-                                      "a later round")
-                cf, decode = self._combined_key_frame(list(by))
+                cf, decode = self._combined_key_frame(list(by),
+                                                      dropna=dropna)
                 keep = [c for c in cf.columns if c not in by]
                 res = cf.take_columns(keep).groupby_reduce(self.KEYCOL, agg)
                 res._index = decode(lib.get(res._index.col))

@@ -2178,3 +2178,28 @@ def test_unbounded_multikey_groupby(npartitions):
         np.testing.assert_allclose(got["v"].to_numpy(),
                                    exp["v"].to_numpy(), rtol=1e-12,
                                    err_msg=agg)
+
+
+def test_multikey_dropna_false_vs_pandas(npartitions):
+    """Multi-key groupby(dropna=False) on the GPU path (string + int key)."""
+    rng = np.random.default_rng(8)
+    n = 25_000
+    a = rng.choice(["x", "y", None], n, p=[0.45, 0.45, 0.1]).astype(object)
+    b = rng.integers(0, 6, n)
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.1] = np.nan
+    pdf = pandas.DataFrame({"a": a, "b": b, "v": v})
+    df = mpd.DataFrame(pdf)
+    for agg in ("sum", "count", "mean", "min", "max"):
+        got = getattr(df.groupby(["a", "b"], dropna=False),
+                      agg)().to_pandas()
+        exp = getattr(pdf.groupby(["a", "b"], dropna=False), agg)()
+        assert len(got) == len(exp), agg
+        gi = [(x if isinstance(x, str) else "<NA>", y)
+              for x, y in got.index]
+        ei = [(x if isinstance(x, str) else "<NA>", y)
+              for x, y in exp.index]
+        assert gi == ei, f"{agg} keys"
+        np.testing.assert_allclose(got["v"].to_numpy().astype(float),
+                                   exp["v"].to_numpy().astype(float),
+                                   rtol=1e-12, atol=1e-9, equal_nan=True)

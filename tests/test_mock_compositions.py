@@ -339,3 +339,29 @@ def test_mock_multikey_idx_nan_keys(mlib):
         np.testing.assert_allclose(got["v"].to_numpy().astype(float),
                                    exp["v"].to_numpy().astype(float),
                                    rtol=0, equal_nan=True)
+
+
+def test_mock_multikey_dropna_false(mlib):
+    """Multi-key groupby(dropna=False): NaN string keys are an extra
+    top-of-code-space level per column (sorts last per level)."""
+    rng = np.random.default_rng(8)
+    n = 9000
+    a = rng.choice(["x", "y", None], n, p=[0.45, 0.45, 0.1]).astype(object)
+    b = rng.integers(0, 6, n)
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.1] = np.nan
+    pdf = pandas.DataFrame({"a": a, "b": b, "v": v})
+    df = mlib.DataFrame(pdf)
+    for agg in ("sum", "count", "mean", "min", "max"):
+        got = getattr(df.groupby(["a", "b"], dropna=False),
+                      agg)().to_pandas()
+        exp = getattr(pdf.groupby(["a", "b"], dropna=False), agg)()
+        assert len(got) == len(exp), agg
+        gi = [(x if isinstance(x, str) else "<NA>", y)
+              for x, y in got.index]
+        ei = [(x if isinstance(x, str) else "<NA>", y)
+              for x, y in exp.index]
+        assert gi == ei, f"{agg} keys"
+        np.testing.assert_allclose(got["v"].to_numpy().astype(float),
+                                   exp["v"].to_numpy().astype(float),
+                                   rtol=1e-12, atol=1e-9, equal_nan=True)
