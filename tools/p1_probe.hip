@@ -102,6 +102,55 @@ __global__ void __launch_bounds__(256) k_copyb(const float4* __restrict__ in,
   }
 }
 
+// nontemporal-store copy (does skipping the LC/L2 write path on the
+// streaming stores recover the guide's 6.29 TB/s?)
+template <int UNROLL>
+__global__ void __launch_bounds__(256) k_copyb_nt(
+    const float4* __restrict__ in, float4* __restrict__ out, int64_t n4) {
+  typedef float v4f __attribute__((ext_vector_type(4)));
+  const int64_t tile = (int64_t)blockDim.x * UNROLL;
+  const int64_t base = (int64_t)blockIdx.x * tile + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * tile;
+  for (int64_t i = base; i < n4; i += stride) {
+    v4f v[UNROLL];
+#pragma unroll
+    for (int u = 0; u < UNROLL; ++u) {
+      const int64_t j = i + (int64_t)u * blockDim.x;
+      if (j < n4)
+        v[u] = __builtin_nontemporal_load(
+            reinterpret_cast<const v4f*>(in) + j);
+    }
+#pragma unroll
+    for (int u = 0; u < UNROLL; ++u) {
+      const int64_t j = i + (int64_t)u * blockDim.x;
+      if (j < n4)
+        __builtin_nontemporal_store(v[u], reinterpret_cast<v4f*>(out) + j);
+    }
+  }
+}
+
+// wide-block variant (1024 threads)
+template <int UNROLL>
+__global__ void __launch_bounds__(1024) k_copyw(
+    const float4* __restrict__ in, float4* __restrict__ out, int64_t n4) {
+  const int64_t tile = (int64_t)blockDim.x * UNROLL;
+  const int64_t base = (int64_t)blockIdx.x * tile + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * tile;
+  for (int64_t i = base; i < n4; i += stride) {
+    float4 v[UNROLL];
+#pragma unroll
+    for (int u = 0; u < UNROLL; ++u) {
+      const int64_t j = i + (int64_t)u * blockDim.x;
+      if (j < n4) v[u] = in[j];
+    }
+#pragma unroll
+    for (int u = 0; u < UNROLL; ++u) {
+      const int64_t j = i + (int64_t)u * blockDim.x;
+      if (j < n4) out[j] = v[u];
+    }
+  }
+}
+
 // read-only (sum) and write-only anchors
 __global__ void __launch_bounds__(256) k_readbw(const float4* __restrict__ in,
                                                 int64_t n4,
@@ -346,6 +395,13 @@ int main(int argc, char** argv) {
     bw("copy u8 g1024", [&] { hipLaunchKernelGGL(k_copyb<8>, dim3(1024), dim3(256), 0, 0, a, b, n4); }, 2.0 * n4 * 16);
     bw("copy u16 g2048", [&] { hipLaunchKernelGGL(k_copyb<16>, dim3(2048), dim3(256), 0, 0, a, b, n4); }, 2.0 * n4 * 16);
     bw("read u8 g4096", [&] { hipLaunchKernelGGL(k_readbw, dim3(4096), dim3(256), 0, 0, a, n4, sink); }, 1.0 * n4 * 16);
+    bw("copy nt u4 g4096", [&] { hipLaunchKernelGGL(k_copyb_nt<4>, dim3(4096), dim3(256), 0, 0, a, b, n4); }, 2.0 * n4 * 16);
+    bw("copy nt u2 g4096", [&] { hipLaunchKernelGGL(k_copyb_nt<2>, dim3(4096), dim3(256), 0, 0, a, b, n4); }, 2.0 * n4 * 16);
+    bw("copy nt u8 g2048", [&] { hipLaunchKernelGGL(k_copyb_nt<8>, dim3(2048), dim3(256), 0, 0, a, b, n4); }, 2.0 * n4 * 16);
+    bw("copy w1024 u2 g1024", [&] { hipLaunchKernelGGL(k_copyw<2>, dim3(1024), dim3(1024), 0, 0, a, b, n4); }, 2.0 * n4 * 16);
+    bw("copy w1024 u4 g2048", [&] { hipLaunchKernelGGL(k_copyw<4>, dim3(2048), dim3(1024), 0, 0, a, b, n4); }, 2.0 * n4 * 16);
+    bw("copy u2 g8192", [&] { hipLaunchKernelGGL(k_copyb<2>, dim3(8192), dim3(256), 0, 0, a, b, n4); }, 2.0 * n4 * 16);
+    bw("copy u4 g8192", [&] { hipLaunchKernelGGL(k_copyb<4>, dim3(8192), dim3(256), 0, 0, a, b, n4); }, 2.0 * n4 * 16);
     CHECK(hipFree(a));
     CHECK(hipFree(b));
     CHECK(hipFree(sink));
