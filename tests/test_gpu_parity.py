@@ -726,13 +726,18 @@ def test_new_paths_empty_and_edge(npartitions):
     exp = pdf.groupby(["a"]).sum()
     np.testing.assert_array_equal(out["v"].to_numpy(),
                                   exp["v"].to_numpy())
-    # combined-range overflow is loud
-    big = mpd.DataFrame(pandas.DataFrame({
-        "a": np.array([0, 2**40], dtype=np.int64),
-        "b": np.array([0, 2**40], dtype=np.int64),
-        "v": np.array([1.0, 2.0])}))
-    with pytest.raises(lib.HfError, match="2\\^62"):
-        big.groupby(["a", "b"]).sum().to_pandas()
+    # combined-range beyond 2^62: the sorted-heads dense-rank fold now
+    # covers it (round 2) — must match pandas instead of raising
+    big_p = pandas.DataFrame({
+        "a": np.array([0, 2**40, 0, 2**40], dtype=np.int64),
+        "b": np.array([0, 2**40, 0, 5], dtype=np.int64),
+        "v": np.array([1.0, 2.0, 3.0, 4.0])})
+    big = mpd.DataFrame(big_p)
+    got_big = big.groupby(["a", "b"]).sum().to_pandas()
+    exp_big = big_p.groupby(["a", "b"]).sum()
+    assert list(got_big.index) == list(exp_big.index)
+    np.testing.assert_allclose(got_big["v"].to_numpy(),
+                               exp_big["v"].to_numpy(), rtol=0)
 
 
 def test_merge_unbounded_span_vs_golden(npartitions):
