@@ -176,7 +176,7 @@ __global__ void __launch_bounds__(256) k_readbw(const float4* __restrict__ in,
 // scatter variants.  VAR: 0 = production pipelined shape; 1 = scan||reserve
 // merged; 2 = u16 staged key + binary-search writeout (RPT may be larger).
 // ---------------------------------------------------------------------------
-template <int RPT, int BLK, int RL, int VAR>
+template <int RPT, int BLK, int RL, int VAR, bool NTW = false>
 __global__ void __launch_bounds__(BLK) k_scat(
     const unsigned* __restrict__ keys32, const double* __restrict__ v0,
     int64_t n, int nb, unsigned* __restrict__ cursors,
@@ -311,8 +311,15 @@ __global__ void __launch_bounds__(BLK) k_scat(
         b = skey32[p] >> 16;
       }
       const int64_t pos = (int64_t)it_gbase[b] + (p - it_off[b]);
-      rk[pos] = VAR == 2 ? skey16[p] : (unsigned short)(skey32[p] & 0xFFFF);
-      r0[pos] = sval0[p];
+      const unsigned short kv =
+          VAR == 2 ? skey16[p] : (unsigned short)(skey32[p] & 0xFFFF);
+      if (NTW) {
+        __builtin_nontemporal_store(kv, rk + pos);
+        __builtin_nontemporal_store(sval0[p], r0 + pos);
+      } else {
+        rk[pos] = kv;
+        r0[pos] = sval0[p];
+      }
     }
   }
 }
@@ -502,6 +509,12 @@ int main(int argc, char** argv) {
       false);
   run("v2 u16 12288 g2048",
       [&] { hipLaunchKernelGGL((k_scat<12, 1024, RL, 2>), dim3(2048), dim3(1024), 12288 * 10 + nb * 12 + 20, 0, keys32, vals, N, nb, d_cur, r0, rk); },
+      false);
+  run("v3 ntw g2048",
+      [&] { hipLaunchKernelGGL((k_scat<12, 1024, RL, 0, true>), dim3(2048), dim3(1024), lds12, 0, keys32, vals, N, nb, d_cur, r0, rk); },
+      false);
+  run("v3 ntw g1024",
+      [&] { hipLaunchKernelGGL((k_scat<12, 1024, RL, 0, true>), dim3(1024), dim3(1024), lds12, 0, keys32, vals, N, nb, d_cur, r0, rk); },
       false);
   printf("done\n");
   return 0;
