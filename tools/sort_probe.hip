@@ -235,6 +235,48 @@ __global__ void __launch_bounds__(64 * WPB) k_scat_staged2(
   }
 }
 
+// s4: bit-ballot ranking — the serial leader loop (one iteration per
+// DISTINCT digit in the wave, ~50 for 256 digits) becomes 8 ballots:
+// lanes sharing this lane's digit = AND over digit bits of
+// (bit set ? ballot : ~ballot).  Rank = popcount(below); the lowest lane
+// of each group advances the digit base.  O(8) per row, no divergent loop.
+__global__ void __launch_bounds__(BLOCK) k_scat_ballot(
+    const unsigned long long* __restrict__ pairs, int64_t n, int shift,
+    const unsigned long long* __restrict__ offs, int64_t ntiles,
+    unsigned long long* __restrict__ out) {
+  __shared__ unsigned long long base[SORT_WPB][256];
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const unsigned long long below = (1ull << lane) - 1ull;
+  for (int64_t tile = (int64_t)blockIdx.x * SORT_WPB + wave; tile < ntiles;
+       tile += (int64_t)gridDim.x * SORT_WPB) {
+    for (int d = lane; d < 256; d += 64)
+      base[wave][d] = offs[(int64_t)d * ntiles + tile];
+    __builtin_amdgcn_wave_barrier();
+    const int64_t t0 = tile * SORT_TILE;
+    for (int j = 0; j < SORT_RPT; ++j) {
+      const int64_t row = t0 + (int64_t)j * 64 + lane;
+      const bool valid = row < n;
+      const unsigned long long p = valid ? pairs[row] : 0;
+      const unsigned d = ((unsigned)p >> shift) & 255u;
+      unsigned long long same = __ballot(valid);
+#pragma unroll
+      for (int b = 0; b < 8; ++b) {
+        const unsigned long long m = __ballot((d >> b) & 1u);
+        same &= ((d >> b) & 1u) ? m : ~m;
+      }
+      if (valid) {
+        const unsigned rank = (unsigned)__popcll(same & below);
+        const unsigned long long pos = base[wave][d] + rank;
+        // lowest member advances the base for the next step
+        if (lane == __ffsll((long long)same) - 1)
+          base[wave][d] += __popcll(same);
+        out[pos] = p;
+      }
+      __builtin_amdgcn_wave_barrier();
+    }
+  }
+}
+
 __global__ void k_cmp64(const unsigned long long* a,
                         const unsigned long long* b, int64_t n,
                         unsigned long long* bad) {
@@ -312,6 +354,14 @@ int main(int argc, char** argv) {
   run("s3 staged wpb2", [&] {
     hipLaunchKernelGGL(k_scat_staged2<2>, dim3(4096), dim3(128), 0, 0,
                        pairs, N, shift, offs, C, ntiles, out);
+  }, true);
+  run("s4 bit-ballot", [&] {
+    hipLaunchKernelGGL(k_scat_ballot, dim3(2048), dim3(BLOCK), 0, 0, pairs,
+                       N, shift, offs, ntiles, out);
+  }, true);
+  run("s4 g4096", [&] {
+    hipLaunchKernelGGL(k_scat_ballot, dim3(4096), dim3(BLOCK), 0, 0, pairs,
+                       N, shift, offs, ntiles, out);
   }, true);
   printf("done\n");
   return 0;
