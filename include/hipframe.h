@@ -102,8 +102,12 @@ enum {
   HF_MAP_SQRT = 11, /* sqrt(x), f64 only (std = sqrt(var))                  */
   HF_MAP_MIN = 12,  /* fmin(x, s) — clip upper; NaN passes through (f64)    */
   HF_MAP_MAX = 13,  /* fmax(x, s) — clip lower; NaN passes through (f64)    */
-  HF_MAP_ROUND = 14 /* rint(x*s)/s, f64 only (pandas round(d): s = 10^d;
+  HF_MAP_ROUND = 14,/* rint(x*s)/s, f64 only (pandas round(d): s = 10^d;
                        half-even like numpy)                               */
+  HF_MAP_IDIV = 15, /* Python floordiv(x, s), int64 only — exact (the f64
+                       DIV path rounds beyond 2^53; datetime calendar math
+                       needs exactness)                                    */
+  HF_MAP_IMOD = 16  /* Python mod(x, s) (sign of s), int64 only           */
 };
 int hf_map_scalar(int op, const hf_col* in, double scalar, hf_col** out);
 /* i64 column with an exact int64 scalar (double cannot hold all int64). */

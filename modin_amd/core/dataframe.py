@@ -2069,6 +2069,85 @@ class HipDataframe:
                             pandas.Series({c: np.dtype(np.float64)
                                            for c in names}))
 
+    def dt_field(self, field: str) -> "HipDataframe":
+        """Series.dt.<field> over datetime64[ns] typed columns: exact
+        int64 calendar math on the ns view (Howard Hinnant's civil-from-
+        days algorithm composed from MAP_IDIV/IMOD + compares — f64 would
+        round modern-era ns values).  Reference surface: pandas
+        Series.dt accessors (the reference delegates to pandas)."""
+        fields = ("year", "month", "day", "hour", "minute", "second",
+                  "dayofweek")
+        if field not in fields:
+            raise lib.HfError(f"dt.{field} not supported ({fields})")
+        name = self.columns[0]
+        if not (isinstance(self.dtypes[name], np.dtype)
+                and np.issubdtype(self.dtypes[name], np.datetime64)):
+            raise lib.HfError(
+                f"dt accessor on non-datetime column {name!r}")
+
+        def concat_col():
+            cs = [p.block().columns[name] for p in self._partitions]
+            return cs[0] if len(cs) == 1 else lib.concat(cs)
+
+        ns = concat_col()
+        n = ns.length
+        DAY = 86_400 * 10**9
+
+        def idiv(c, k):
+            return lib.map_scalar(lib.MAP_IDIV, c, k)
+
+        def imod(c, k):
+            return lib.map_scalar(lib.MAP_IMOD, c, k)
+
+        def addc(c, k):
+            return lib.map_scalar(lib.MAP_ADD, c, k)
+
+        def mulc(c, k):
+            return lib.map_scalar(lib.MAP_MUL, c, k)
+
+        def sub(a, b):
+            return lib.binary(lib.BIN_SUB, a, b)
+
+        if field == "hour":
+            out = idiv(imod(ns, DAY), 3_600 * 10**9)
+        elif field == "minute":
+            out = idiv(imod(ns, 3_600 * 10**9), 60 * 10**9)
+        elif field == "second":
+            out = idiv(imod(ns, 60 * 10**9), 10**9)
+        elif field == "dayofweek":
+            out = imod(addc(idiv(ns, DAY), 3), 7)  # 1970-01-01 is a Thu
+        else:
+            days = idiv(ns, DAY)
+            z = addc(days, 719_468)
+            era = idiv(z, 146_097)
+            doe = sub(z, mulc(era, 146_097))           # [0, 146096]
+            # yoe = (doe - doe/1460 + doe/36524 - doe/146096) / 365
+            yoe = idiv(sub(lib.binary(lib.BIN_ADD,
+                                      sub(doe, idiv(doe, 1460)),
+                                      idiv(doe, 36_524)),
+                           idiv(doe, 146_096)), 365)
+            doy = sub(doe, sub(lib.binary(lib.BIN_ADD, mulc(yoe, 365),
+                                          idiv(yoe, 4)),
+                               idiv(yoe, 100)))
+            mp = idiv(addc(mulc(doy, 5), 2), 153)
+            if field == "day":
+                out = addc(sub(doy, idiv(addc(mulc(mp, 153), 2), 5)), 1)
+            else:
+                # m = mp + 3 - 12*(mp >= 10)
+                ge10 = lib.compare_scalar(lib.CMP_GE, mp, 10.0)
+                m = sub(addc(mp, 3), mulc(ge10, 12))
+                if field == "month":
+                    out = m
+                else:  # year = yoe + era*400 + (m <= 2)
+                    le2 = lib.compare_scalar(lib.CMP_LE, m, 2.0)
+                    out = lib.binary(
+                        lib.BIN_ADD,
+                        lib.binary(lib.BIN_ADD, yoe, mulc(era, 400)), le2)
+        part = HipDataframePartition(DeviceBlock({name: out}, n))
+        # pandas dt fields are int32
+        return HipDataframe([part], self._index, [name], [n],
+                            pandas.Series({name: np.dtype(np.int32)}))
+
     def rank_rows(self, ascending: bool = True, method: str = "average",
                   na_option: str = "keep") -> "HipDataframe":
         """Frame-level pandas rank(axis=0) over one constant-key group."""

@@ -46,7 +46,7 @@ HF_FLOAT64 = 1
 # map ops
 MAP_ADD, MAP_SUB, MAP_RSUB, MAP_MUL, MAP_DIV, MAP_RDIV, MAP_FILLNA, MAP_ABS, \
     MAP_NEG, MAP_CAST_F64, MAP_CAST_I64, MAP_SQRT, MAP_MIN, MAP_MAX, \
-    MAP_ROUND = range(15)
+    MAP_ROUND, MAP_IDIV, MAP_IMOD = range(17)
 # binary ops
 BIN_ADD, BIN_SUB, BIN_MUL, BIN_DIV, BIN_MIN, BIN_MAX = range(6)
 

@@ -309,6 +309,14 @@ __device__ __forceinline__ int64_t map1_i64(int64_t x, int64_t s) {
     case HF_MAP_MIN:  return x < s ? x : s;
     case HF_MAP_MAX:  return x > s ? x : s;
     case HF_MAP_NEG:  return -x;
+    case HF_MAP_IDIV: {  // Python floordiv: round toward -inf
+      int64_t q = x / s, r = x % s;
+      return (r != 0 && ((r < 0) != (s < 0))) ? q - 1 : q;
+    }
+    case HF_MAP_IMOD: {  // Python mod: result takes s's sign
+      int64_t r = x % s;
+      return (r != 0 && ((r < 0) != (s < 0))) ? r + s : r;
+    }
   }
   return x;
 }
@@ -2476,6 +2484,22 @@ int hf_map_scalar_i64(int op, const hf_col* in, int64_t scalar, hf_col** out) {
     case HF_MAP_NEG:  rc = launch_map_i64<HF_MAP_NEG>(in, scalar, *out); break;
     case HF_MAP_MIN:  rc = launch_map_i64<HF_MAP_MIN>(in, scalar, *out); break;
     case HF_MAP_MAX:  rc = launch_map_i64<HF_MAP_MAX>(in, scalar, *out); break;
+    case HF_MAP_IDIV:
+      if (scalar == 0) {
+        hf_col_free(*out);
+        *out = nullptr;
+        return set_err(HF_ERR_ARG, "hf_map_scalar_i64", "floordiv by zero");
+      }
+      rc = launch_map_i64<HF_MAP_IDIV>(in, scalar, *out);
+      break;
+    case HF_MAP_IMOD:
+      if (scalar == 0) {
+        hf_col_free(*out);
+        *out = nullptr;
+        return set_err(HF_ERR_ARG, "hf_map_scalar_i64", "mod by zero");
+      }
+      rc = launch_map_i64<HF_MAP_IMOD>(in, scalar, *out);
+      break;
     default:
       hf_col_free(*out);
       *out = nullptr;

@@ -525,6 +525,48 @@ class DataFrame(_HipPandasBase):
         return f"modin_amd.DataFrame({self.shape[0]}x{self.shape[1]} on device)"
 
 
+class _DtAccessor:
+    """Series.dt — calendar fields of datetime64[ns] typed columns,
+    computed on device with exact int64 calendar math (dataframe.dt_field)."""
+
+    def __init__(self, s: "Series"):
+        self._s = s
+
+    def _field(self, f: str) -> "Series":
+        return Series(query_compiler=self._s._query_compiler.dt_field(f),
+                      name=self._s.name)
+
+    @property
+    def year(self):
+        return self._field("year")
+
+    @property
+    def month(self):
+        return self._field("month")
+
+    @property
+    def day(self):
+        return self._field("day")
+
+    @property
+    def hour(self):
+        return self._field("hour")
+
+    @property
+    def minute(self):
+        return self._field("minute")
+
+    @property
+    def second(self):
+        return self._field("second")
+
+    @property
+    def dayofweek(self):
+        return self._field("dayofweek")
+
+    weekday = dayofweek
+
+
 class Series(_HipPandasBase):
     _bool_mask = False  # comparisons set this: to_pandas lowers int64->bool
 
@@ -539,6 +581,10 @@ class Series(_HipPandasBase):
         else:
             raise lib.HfError("Series accepts a pandas.Series")
         self._query_compiler = HipQueryCompiler.from_pandas(pdf)
+
+    @property
+    def dt(self) -> "_DtAccessor":
+        return _DtAccessor(self)
 
     def _rewrap(self, qc):
         return Series(query_compiler=qc, name=self.name)

@@ -251,6 +251,9 @@ class HipQueryCompiler:
         return self.__constructor__(
             self._modin_frame.groupby_idxminmax(by, maximum=False))
 
+    def dt_field(self, field: str) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.dt_field(field))
+
     def rank(self, method: str = "average", ascending: bool = True,
              na_option: str = "keep") -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.rank_rows(

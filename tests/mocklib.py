@@ -213,6 +213,10 @@ def install(monkeypatch):
                 r = -x
             elif op == lib.MAP_SQRT:
                 r = np.sqrt(x)
+            elif op == lib.MAP_IDIV:
+                r = x // s  # numpy int floordiv == Python semantics
+            elif op == lib.MAP_IMOD:
+                r = x % s
             elif op == lib.MAP_MIN:
                 r = np.where(np.isnan(x), x, np.fmin(x, s)) \
                     if not is_int else np.minimum(x, s)

@@ -2208,3 +2208,20 @@ def test_multikey_dropna_false_vs_pandas(npartitions):
         np.testing.assert_allclose(got["v"].to_numpy().astype(float),
                                    exp["v"].to_numpy().astype(float),
                                    rtol=1e-12, atol=1e-9, equal_nan=True)
+
+
+def test_dt_accessor_vs_pandas(npartitions):
+    """Series.dt calendar fields on device (exact int64 civil math)."""
+    rng = np.random.default_rng(5)
+    n = 60_000
+    ns = rng.integers(-2 * 10**18, 2 * 10**18, n)
+    t = pandas.Series(ns.astype("datetime64[ns]"), name="t")
+    pdf = pandas.DataFrame({"t": t})
+    df = mpd.DataFrame(pdf)
+    for f in ("year", "month", "day", "hour", "minute", "second",
+              "dayofweek"):
+        got = getattr(df["t"].dt, f).to_pandas()
+        exp = getattr(t.dt, f)
+        np.testing.assert_array_equal(np.asarray(got), exp.to_numpy(),
+                                      err_msg=f)
+        assert np.asarray(got).dtype == exp.to_numpy().dtype, f

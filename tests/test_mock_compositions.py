@@ -365,3 +365,20 @@ def test_mock_multikey_dropna_false(mlib):
         np.testing.assert_allclose(got["v"].to_numpy().astype(float),
                                    exp["v"].to_numpy().astype(float),
                                    rtol=1e-12, atol=1e-9, equal_nan=True)
+
+
+def test_mock_dt_accessor(mlib):
+    """Series.dt fields: exact int64 civil-calendar math vs pandas
+    (incl. pre-1970 dates — negative ns exercise floor div/mod)."""
+    rng = np.random.default_rng(5)
+    n = 4000
+    ns = rng.integers(-2 * 10**18, 2 * 10**18, n)
+    t = pandas.Series(ns.astype("datetime64[ns]"), name="t")
+    pdf = pandas.DataFrame({"t": t})
+    df = mlib.DataFrame(pdf)
+    for f in ("year", "month", "day", "hour", "minute", "second",
+              "dayofweek"):
+        got = getattr(df["t"].dt, f).to_pandas()
+        exp = getattr(t.dt, f)
+        np.testing.assert_array_equal(np.asarray(got), exp.to_numpy(),
+                                      err_msg=f)
