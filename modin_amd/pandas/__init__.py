@@ -525,6 +525,44 @@ class DataFrame(_HipPandasBase):
         return f"modin_amd.DataFrame({self.shape[0]}x{self.shape[1]} on device)"
 
 
+class _StrAccessor:
+    """Series.str — host-dictionary transforms + one device gather."""
+
+    def __init__(self, s: "Series"):
+        self._s = s
+
+    def _wrap(self, qc, bool_mask=False):
+        out = Series(query_compiler=qc, name=self._s.name)
+        out._bool_mask = bool_mask
+        return out
+
+    def len(self):  # noqa: A003
+        return self._wrap(self._s._query_compiler.str_op("len"))
+
+    def lower(self):
+        return self._wrap(self._s._query_compiler.str_op("lower"))
+
+    def upper(self):
+        return self._wrap(self._s._query_compiler.str_op("upper"))
+
+    def contains(self, pat, regex: bool = False, na=None):
+        if regex:
+            raise lib.HfError("str.contains(regex=True) is a later round")
+        return self._wrap(self._s._query_compiler.str_op("contains",
+                                                         pat=pat, na=na),
+                          bool_mask=na is not None)
+
+    def startswith(self, pat, na=None):
+        return self._wrap(self._s._query_compiler.str_op("startswith",
+                                                         pat=pat, na=na),
+                          bool_mask=na is not None)
+
+    def endswith(self, pat, na=None):
+        return self._wrap(self._s._query_compiler.str_op("endswith",
+                                                         pat=pat, na=na),
+                          bool_mask=na is not None)
+
+
 class _DtAccessor:
     """Series.dt — calendar fields of datetime64[ns] typed columns,
     computed on device with exact int64 calendar math (dataframe.dt_field)."""
@@ -585,6 +623,10 @@ class Series(_HipPandasBase):
     @property
     def dt(self) -> "_DtAccessor":
         return _DtAccessor(self)
+
+    @property
+    def str(self) -> "_StrAccessor":  # noqa: A003
+        return _StrAccessor(self)
 
     def _rewrap(self, qc):
         return Series(query_compiler=qc, name=self.name)

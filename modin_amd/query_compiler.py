@@ -251,6 +251,74 @@ class HipQueryCompiler:
         return self.__constructor__(
             self._modin_frame.groupby_idxminmax(by, maximum=False))
 
+    def str_op(self, op: str, pat: str = None,
+               na=None) -> "HipQueryCompiler":
+        """Series.str.<op> over a dictionary column: the transform runs on
+        the HOST DICTIONARY (O(#categories)) and reaches the rows with one
+        device gather — no string ever touches the GPU (SURVEY §8f.3
+        design).  ops: len, lower, upper, contains/startswith/endswith
+        (literal, na=True/False/None->NaN)."""
+        from modin_amd.core.dataframe import HipDataframe
+        from modin_amd.core.partition import DeviceBlock, \
+            HipDataframePartition
+        import pandas as pd
+        frame = self._modin_frame
+        name = frame.columns[0]
+        blk_cats = (frame._partitions[0].block().cats
+                    if frame._partitions else {})
+        if name not in blk_cats:
+            raise lib.HfError(f"str accessor on non-string column {name!r}")
+        cats = blk_cats[name]
+
+        def concat_col():
+            cs = [p.block().columns[name] for p in frame._partitions]
+            return cs[0] if len(cs) == 1 else lib.concat(cs)
+
+        codes = concat_col()
+        n = codes.length
+        shifted = lib.map_scalar(lib.MAP_ADD, codes, 1)  # NaN(-1) -> slot 0
+        if op in ("lower", "upper"):
+            vals = [getattr(c, op)() for c in cats.to_numpy(dtype=object)]
+            new_codes, new_cats = pd.factorize(pd.Series(vals), sort=True)
+            lut = np.empty(len(cats) + 1, dtype=np.int64)
+            lut[0] = -1
+            lut[1:] = new_codes
+            out = lib.gather(lib.put(lut), shifted)
+            blk = DeviceBlock({name: out}, n, {name: pd.Index(new_cats)})
+            dts = pd.Series({name: np.dtype(object)})
+        elif op == "len":
+            lut = np.empty(len(cats) + 1, dtype=np.float64)
+            lut[0] = np.nan
+            lut[1:] = [len(c) for c in cats.to_numpy(dtype=object)]
+            out = lib.gather(lib.put(lut), shifted)
+            blk = DeviceBlock({name: out}, n)
+            dts = pd.Series({name: np.dtype(np.float64)})
+        elif op in ("contains", "startswith", "endswith"):
+            test = {"contains": lambda c: pat in c,
+                    "startswith": lambda c: c.startswith(pat),
+                    "endswith": lambda c: c.endswith(pat)}[op]
+            hit = [bool(test(c)) for c in cats.to_numpy(dtype=object)]
+            if na is None:
+                # pandas returns object [True, False, NaN]; this backend
+                # returns float64 [1.0, 0.0, NaN] (documented deviation —
+                # pass na=True/False for a clean bool column)
+                lut = np.empty(len(cats) + 1, dtype=np.float64)
+                lut[0] = np.nan
+                lut[1:] = np.asarray(hit, dtype=np.float64)
+                dts = pd.Series({name: np.dtype(np.float64)})
+            else:
+                lut = np.empty(len(cats) + 1, dtype=np.int64)
+                lut[0] = 1 if na else 0
+                lut[1:] = np.asarray(hit, dtype=np.int64)
+                dts = pd.Series({name: np.dtype(bool)})
+            out = lib.gather(lib.put(lut), shifted)
+            blk = DeviceBlock({name: out}, n)
+        else:
+            raise lib.HfError(f"str.{op} not supported")
+        res = HipDataframe([HipDataframePartition(blk)], frame._index,
+                           [name], [n], dts)
+        return self.__constructor__(res)
+
     def dt_field(self, field: str) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.dt_field(field))
 

@@ -404,3 +404,39 @@ def test_string_value_groupby_min_max_count(gpu_ready):
                 np.testing.assert_allclose(ge.astype(np.float64),
                                            ee.astype(np.float64),
                                            rtol=1e-12, err_msg=f"{op}/{c}")
+
+
+def test_str_accessor_vs_pandas(npartitions):
+    """Series.str on device-backed dictionary columns (host LUT + gather)."""
+    rng = np.random.default_rng(9)
+    n = 40_000
+    words = np.array(["Apple", "beta", "Ba", "apple", "CAT", "ca t"],
+                     dtype=object)
+    sv = words[rng.integers(0, len(words), n)]
+    sv[rng.random(n) < 0.1] = np.nan
+    t = pandas.Series(sv, name="s")
+    df = mpd.DataFrame(pandas.DataFrame({"s": t}))
+    got = df["s"].str.len().to_pandas()
+    np.testing.assert_allclose(np.asarray(got, dtype=float),
+                               t.str.len().to_numpy(dtype=float),
+                               rtol=0, equal_nan=True)
+    for op in ("lower", "upper"):
+        got = getattr(df["s"].str, op)().to_pandas()
+        exp = getattr(t.str, op)()
+        same = (pandas.isna(np.asarray(got)) & pandas.isna(exp).to_numpy()
+                ) | (np.asarray(got) == exp.to_numpy())
+        assert same.all(), op
+    for op, pat in (("contains", "a"), ("startswith", "a"),
+                    ("endswith", "t")):
+        got = getattr(df["s"].str, op)(pat, na=False).to_pandas()
+        exp = getattr(t.str, op)(pat, na=False)
+        np.testing.assert_array_equal(
+            np.asarray(got, dtype=bool), exp.to_numpy(dtype=bool),
+            err_msg=op)
+    # masks compose with row selection
+    sel = df[df["s"].str.contains("a", na=False)].to_pandas()
+    exp_sel = pandas.DataFrame({"s": t})[t.str.contains("a", na=False)]
+    same = (pandas.isna(sel["s"].to_numpy()) &
+            pandas.isna(exp_sel["s"]).to_numpy()) | \
+        (sel["s"].to_numpy() == exp_sel["s"].to_numpy())
+    assert same.all()

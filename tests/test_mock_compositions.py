@@ -382,3 +382,38 @@ def test_mock_dt_accessor(mlib):
         exp = getattr(t.dt, f)
         np.testing.assert_array_equal(np.asarray(got), exp.to_numpy(),
                                       err_msg=f)
+
+
+def test_mock_str_accessor(mlib):
+    """Series.str len/lower/upper/contains/startswith/endswith: host
+    dictionary transform + device gather, vs pandas."""
+    rng = np.random.default_rng(9)
+    n = 4000
+    words = np.array(["Apple", "beta", "Ba", "apple", "CAT", "ca t"],
+                     dtype=object)
+    sv = words[rng.integers(0, len(words), n)]
+    sv[rng.random(n) < 0.1] = np.nan
+    t = pandas.Series(sv, name="s")
+    df = mlib.DataFrame(pandas.DataFrame({"s": t}))
+    got = df["s"].str.len().to_pandas()
+    np.testing.assert_allclose(np.asarray(got, dtype=float),
+                               t.str.len().to_numpy(dtype=float),
+                               rtol=0, equal_nan=True)
+    for op in ("lower", "upper"):
+        got = getattr(df["s"].str, op)().to_pandas()
+        exp = getattr(t.str, op)()
+        same = (pandas.isna(np.asarray(got)) & pandas.isna(exp).to_numpy()
+                ) | (np.asarray(got) == exp.to_numpy())
+        assert same.all(), op
+    for op, pat in (("contains", "a"), ("startswith", "a"),
+                    ("endswith", "t")):
+        got = getattr(df["s"].str, op)(pat, na=False).to_pandas()
+        exp = getattr(t.str, op)(pat, na=False)
+        np.testing.assert_array_equal(
+            np.asarray(got, dtype=bool), exp.to_numpy(dtype=bool),
+            err_msg=op)
+        got = getattr(df["s"].str, op)(pat).to_pandas()
+        exp2 = getattr(t.str, op)(pat)
+        np.testing.assert_allclose(
+            np.asarray(got, dtype=float), exp2.to_numpy(dtype=float),
+            rtol=0, equal_nan=True, err_msg=f"{op}/naNone")
