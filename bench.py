@@ -149,6 +149,9 @@ def cpu_baseline_leg(keys_card: int, unit: str, dist: str):
         "kind": "pandas",
         "sample": f"pandas {pd.__version__} groupby-sum, {sample} rows/pass x "
                   f"{done_rows // sample} passes in {dt:.1f}s",
+        "note": "Modin-on-CPU leg unavailable on the bench box (no modin "
+                "install there; BASELINE.md caveat) — plain pandas is the "
+                "reference backend's own per-partition engine",
         "oracle_secondary": {
             "value": passes * sample / dt_o,
             "kind": "port",
