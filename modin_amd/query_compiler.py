@@ -484,8 +484,13 @@ class HipQueryCompiler:
 
     # ---- boolean row mask (qc.getitem_array device form) ----
     def getitem_array(self, mask_qc: "HipQueryCompiler") -> "HipQueryCompiler":
-        return self.__constructor__(
-            self._modin_frame.filter_rows(mask_qc._modin_frame))
+        mask = mask_qc._modin_frame
+        frame = self._modin_frame
+        if list(mask._row_lengths) != list(frame._row_lengths):
+            # single-partition masks (str/dt accessor outputs) re-slice to
+            # the frame's partition boundaries
+            mask = mask.repartition_like(frame._row_lengths)
+        return self.__constructor__(frame.filter_rows(mask))
 
     def take_row_range(self, start: int, stop: int) -> "HipQueryCompiler":
         return self.__constructor__(
