@@ -2159,6 +2159,91 @@ class HipDataframe:
         """Frame-level pandas ffill/bfill over one constant-key group."""
         return self._with_const_key().groupby_transform(self.KEYCOL, how)
 
+    def groupby_prod(self, by) -> "HipDataframe":
+        """groupby().prod(): sort by key, segmented PRODUCT scan
+        (HF_AGG_PROD), take each run's value at its last non-NaN row
+        (pandas skipna; all-NaN groups give 1.0 — min_count=0).  int64
+        columns stay int64 with pandas' wrapping product."""
+        from .. import distributed as dist_mod
+        if dist_mod.is_active():
+            key = by if isinstance(by, str) else \
+                (by[0] if len(by) == 1 else None)
+            if key is None:
+                raise lib.HfError("distributed multi-key prod is a later "
+                                  "round")
+            shuf = self._shuffle_frame_by_key(key)
+            with dist_mod.local_mode():
+                res = shuf.groupby_prod(key)
+            return self._replicate_result_frame(res)
+        if isinstance(by, (list, tuple)):
+            if len(by) == 1:
+                by = by[0]
+            else:
+                cf, decode = self._combined_key_frame(list(by))
+                keep = [c for c in cf.columns if c not in by]
+                res = cf.take_columns(keep).groupby_prod(self.KEYCOL)
+                res._index = decode(lib.get(res._index.col))
+                return res
+        val_names = [c for c in self.columns if c != by]
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        bad = [v for v in val_names if v in blk_cats]
+        if bad:
+            raise lib.HfError(f"groupby prod over string column(s) {bad}")
+
+        def concat_col(name):
+            cols = [p.block().columns[name] for p in self._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        n = len(self)
+        res0 = self.groupby_size(by)
+        ngv = len(res0)
+        if n == 0 or ngv == 0:
+            part = HipDataframePartition(DeviceBlock(
+                {v: lib.alloc(0, lib.HF_FLOAT64) for v in val_names}, 0))
+            return HipDataframe([part], res0._index, val_names, [0],
+                                pandas.Series({v: np.dtype(np.float64)
+                                               for v in val_names}))
+        ekey, _ = self._effective_sort_key(concat_col(by), by in blk_cats,
+                                           True)
+        perm = self._compose_sort_perm([(ekey, True)])
+        khead = self._run_head_col(lib.gather(ekey, perm), n)
+        kplan = lib.filter_plan(khead)
+        ng = kplan.n_kept
+        hp = lib.filter_iota(kplan, 0)
+        ends = lib.map_scalar(
+            lib.MAP_SUB,
+            lib.concat([lib.col_slice(hp, 1, ng - 1),
+                        self._const_i64(n)]) if ng > 1
+            else self._const_i64(n), 1)
+        out_cols, dts = {}, {}
+        for v in val_names:
+            vc = concat_col(v)
+            sv = lib.gather(vc, perm)
+            seg = lib.seg_cumsum(sv, khead, lib.AGG_PROD)
+            if vc.dtype_code == lib.HF_INT64:
+                res = lib.col_slice(lib.gather(seg, ends), 0, ngv)
+                dts[v] = np.dtype(np.int64)
+            else:
+                m = lib.compare_scalar(lib.CMP_NOTNA, sv, 0.0)
+                segc = lib.seg_cumsum(m, khead, lib.AGG_SUM)
+                cnt = lib.gather(segc, ends)
+                pos = self._iota(n)
+                posv = lib.binary(
+                    lib.BIN_ADD, lib.binary(lib.BIN_MUL, pos, m),
+                    lib.map_scalar(lib.MAP_SUB, m, 1))
+                segmax = lib.seg_cumsum(posv, khead, lib.AGG_MAX)
+                lastp = lib.map_scalar(lib.MAP_MAX,
+                                       lib.gather(segmax, ends), 0)
+                resf = lib.fixup_empty(lib.gather(seg, lastp), cnt)
+                resf = lib.map_scalar(lib.MAP_FILLNA, resf, 1.0)
+                res = lib.col_slice(resf, 0, ngv)
+                dts[v] = np.dtype(np.float64)
+            out_cols[v] = res
+        part = HipDataframePartition(DeviceBlock(out_cols, ngv))
+        return HipDataframe([part], res0._index, val_names, [ngv],
+                            pandas.Series(dts))
+
     def groupby_idxminmax(self, by, maximum: bool) -> "HipDataframe":
         """groupby.idxmax/idxmin: per group and value column, the ORIGINAL
         row position (RangeIndex label) of the first occurrence of the

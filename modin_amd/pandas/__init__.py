@@ -869,6 +869,9 @@ class DataFrameGroupBy:
             return from_pandas(out.to_pandas().reset_index())
         return out
 
+    def prod(self):
+        return self._agg("prod")
+
     def first(self):
         return self._agg("first")
 
@@ -975,7 +978,7 @@ class DataFrameGroupBy:
         return out["size"].rename(None)
 
     _AGGS = ("sum", "count", "mean", "min", "max", "var", "std",
-             "median", "first", "last")
+             "median", "first", "last", "prod")
 
     def agg(self, how):
         """str, list-of-str (MultiIndex columns, pandas col-major order) or

@@ -220,6 +220,9 @@ class HipQueryCompiler:
         return self.__constructor__(
             self._modin_frame.groupby_quantile(by, q))
 
+    def groupby_prod(self, by) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.groupby_prod(by))
+
     def groupby_first(self, by) -> "HipQueryCompiler":
         return self.__constructor__(
             self._modin_frame.groupby_firstlast(by, last=False))
@@ -429,6 +432,7 @@ class HipQueryCompiler:
             "var": type(self).groupby_var,
             "std": type(self).groupby_std,
             "median": type(self).groupby_median,
+            "prod": type(self).groupby_prod,
             "first": type(self).groupby_first,
             "last": type(self).groupby_last,
         }.get(agg)
@@ -441,7 +445,8 @@ class HipQueryCompiler:
         if not dropna:
             name = {"var": "groupby_var", "std": "groupby_std",
                     "median": "groupby_median", "first": "groupby_first",
-                    "last": "groupby_last"}[agg]
+                    "last": "groupby_last",
+                    "prod": "groupby_prod"}[agg]
             return self._tag_key_index(
                 self.groupby_tail_agg(by, name, dropna=False), by)
         return self._tag_key_index(fn(self, by), by)

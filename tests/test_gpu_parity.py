@@ -2225,3 +2225,34 @@ def test_dt_accessor_vs_pandas(npartitions):
         np.testing.assert_array_equal(np.asarray(got), exp.to_numpy(),
                                       err_msg=f)
         assert np.asarray(got).dtype == exp.to_numpy().dtype, f
+
+
+def test_groupby_prod_vs_pandas(npartitions):
+    """groupby.prod on device: segmented PROD scan + last-valid pick."""
+    rng = np.random.default_rng(12)
+    n = 40_000
+    k = rng.integers(0, 300, n).astype(np.int64)
+    v = np.clip(rng.standard_normal(n), -1.4, 1.4)
+    v[rng.random(n) < 0.1] = np.nan
+    w = (rng.integers(0, 3, n) - 1).astype(np.int64)
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w})
+    df = mpd.DataFrame(pdf)
+    got = df.groupby("k").prod().to_pandas()
+    exp = pdf.groupby("k").prod()
+    for c in exp.columns:
+        np.testing.assert_allclose(got[c].to_numpy(dtype=float),
+                                   exp[c].to_numpy(dtype=float),
+                                   rtol=1e-12, atol=1e-300,
+                                   err_msg=f"prod/{c}")
+    assert list(got.dtypes) == list(exp.dtypes)
+    # float keys + dropna=False route
+    kf = k.astype(np.float64)
+    kf[rng.random(n) < 0.05] = np.nan
+    pdf2 = pandas.DataFrame({"k": kf, "v": v})
+    df2 = mpd.DataFrame(pdf2)
+    got = df2.groupby("k", dropna=False).prod().to_pandas()
+    exp = pdf2.groupby("k", dropna=False).prod()
+    np.testing.assert_allclose(got.index.to_numpy(), exp.index.to_numpy(),
+                               rtol=0, equal_nan=True)
+    np.testing.assert_allclose(got["v"].to_numpy(), exp["v"].to_numpy(),
+                               rtol=1e-12, atol=1e-300)

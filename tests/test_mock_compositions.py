@@ -417,3 +417,21 @@ def test_mock_str_accessor(mlib):
         np.testing.assert_allclose(
             np.asarray(got, dtype=float), exp2.to_numpy(dtype=float),
             rtol=0, equal_nan=True, err_msg=f"{op}/naNone")
+
+
+def test_mock_groupby_prod(mlib):
+    """groupby.prod: segmented PROD scan + last-valid pick; int64 stays
+    int64 (wrapping), NaN skipped, all-NaN groups give 1.0."""
+    rng = np.random.default_rng(12)
+    pdf = _frames(rng, n=3000)
+    pdf["v"] = np.clip(pdf["v"], -1.4, 1.4)
+    pdf["w"] = (pdf["w"] % 3) - 1
+    df = mlib.DataFrame(pdf)
+    got = df.groupby("k").prod().to_pandas()
+    exp = pdf.groupby("k").prod()
+    for c in exp.columns:
+        np.testing.assert_allclose(got[c].to_numpy(dtype=float),
+                                   exp[c].to_numpy(dtype=float),
+                                   rtol=1e-12, atol=1e-300,
+                                   err_msg=f"prod/{c}")
+    assert list(got.dtypes) == list(exp.dtypes)
