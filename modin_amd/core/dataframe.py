@@ -3172,6 +3172,30 @@ class HipDataframe:
         idx = self.index[start:stop]
         return HipDataframe(out_parts, idx, self.columns, lengths, self.dtypes)
 
+    def take_rows(self, positions: np.ndarray) -> "HipDataframe":
+        """Positional row selection (iloc list/array form): one device
+        gather per column over the concatenated frame; the result keeps
+        the selected original index labels."""
+        pos = np.asarray(positions, dtype=np.int64)
+        n = len(self)
+        if pos.size and (pos.min() < -n or pos.max() >= n):
+            raise lib.HfError("iloc: position out of bounds")
+        pos = np.where(pos < 0, pos + n, pos)
+
+        def concat_col(name):
+            cols = [p.block().columns[name] for p in self._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        gidx = lib.put(pos)
+        cols = {c: lib.gather(concat_col(c), gidx) for c in self.columns}
+        part = HipDataframePartition(
+            DeviceBlock(cols, int(pos.size), dict(blk_cats)))
+        idx = self.index[pos] if pos.size else self.index[:0]
+        return HipDataframe([part], idx, self.columns, [int(pos.size)],
+                            self.dtypes)
+
     # ---- astype over all columns ----
     def astype_all(self, dtype) -> "HipDataframe":
         if self._partitions and self._partitions[0].block().cats:

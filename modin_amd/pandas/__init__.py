@@ -326,6 +326,38 @@ def _unwrap(other):
     return other._query_compiler if isinstance(other, _HipPandasBase) else other
 
 
+class _ILocIndexer:
+    """df.iloc — positional rows: integer slices (step 1/None) ride the
+    device column slice, integer lists/arrays ride one device gather per
+    column, a bare int returns the row as a pandas Series."""
+
+    def __init__(self, df):
+        self._df = df
+
+    def __getitem__(self, key):
+        import numpy as _np
+        qc = self._df._query_compiler
+        n = len(qc)
+        if isinstance(key, slice):
+            if key.step not in (None, 1):
+                raise lib.HfError("iloc: slice step other than 1 is a "
+                                  "later round")
+            start, stop, _ = key.indices(n)
+            return self._df._rewrap(qc.take_row_range(start, stop))
+        if isinstance(key, (int, _np.integer)):
+            pos = int(key) + (n if key < 0 else 0)
+            row = DataFrame(
+                query_compiler=qc.take_rows([pos])).to_pandas()
+            if isinstance(self._df, Series):
+                return row.iloc[0, 0]
+            out = row.iloc[0]
+            out.name = row.index[0]
+            return out
+        if isinstance(key, (list, _np.ndarray, pandas.Index)):
+            return self._df._rewrap(qc.take_rows(_np.asarray(key)))
+        raise lib.HfError(f"iloc: unsupported selector {type(key)}")
+
+
 class DataFrame(_HipPandasBase):
     def __init__(self, data=None, query_compiler=None):
         if query_compiler is not None:
@@ -380,6 +412,10 @@ class DataFrame(_HipPandasBase):
                 query_compiler=self._query_compiler.getitem_column_array(list(key))
             )
         raise lib.HfError("only column selection / boolean masks are supported")
+
+    @property
+    def iloc(self):
+        return _ILocIndexer(self)
 
     def head(self, n: int = 5):
         total = len(self._query_compiler)
@@ -619,6 +655,10 @@ class Series(_HipPandasBase):
         else:
             raise lib.HfError("Series accepts a pandas.Series")
         self._query_compiler = HipQueryCompiler.from_pandas(pdf)
+
+    @property
+    def iloc(self):
+        return _ILocIndexer(self)
 
     @property
     def dt(self) -> "_DtAccessor":

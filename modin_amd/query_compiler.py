@@ -497,6 +497,9 @@ class HipQueryCompiler:
             mask = mask.repartition_like(frame._row_lengths)
         return self.__constructor__(frame.filter_rows(mask))
 
+    def take_rows(self, positions) -> "HipQueryCompiler":
+        return self.__constructor__(self._modin_frame.take_rows(positions))
+
     def take_row_range(self, start: int, stop: int) -> "HipQueryCompiler":
         return self.__constructor__(
             self._modin_frame.take_row_range(start, stop))

@@ -2256,3 +2256,18 @@ def test_groupby_prod_vs_pandas(npartitions):
                                rtol=0, equal_nan=True)
     np.testing.assert_allclose(got["v"].to_numpy(), exp["v"].to_numpy(),
                                rtol=1e-12, atol=1e-300)
+
+
+def test_iloc_vs_pandas(npartitions):
+    rng = np.random.default_rng(2)
+    n = 50_000
+    pdf = pandas.DataFrame({"k": rng.integers(0, 999, n),
+                            "v": rng.random(n)})
+    df = mpd.DataFrame(pdf)
+    pandas.testing.assert_frame_equal(df.iloc[1717:42000].to_pandas(),
+                                      pdf.iloc[1717:42000])
+    sel = rng.integers(-n, n, 5000).tolist()
+    pandas.testing.assert_frame_equal(df.iloc[sel].to_pandas(),
+                                      pdf.iloc[sel])
+    assert df.iloc[4242]["k"] == pdf.iloc[4242]["k"]
+    assert abs(df["v"].iloc[-3] - pdf["v"].iloc[-3]) < 1e-15

@@ -435,3 +435,17 @@ def test_mock_groupby_prod(mlib):
                                    rtol=1e-12, atol=1e-300,
                                    err_msg=f"prod/{c}")
     assert list(got.dtypes) == list(exp.dtypes)
+
+
+def test_mock_iloc(mlib):
+    rng = np.random.default_rng(2)
+    pdf = pandas.DataFrame({"k": rng.integers(0, 9, 500),
+                            "v": rng.random(500)})
+    df = mlib.DataFrame(pdf)
+    pandas.testing.assert_frame_equal(df.iloc[17:200].to_pandas(),
+                                      pdf.iloc[17:200])
+    sel = [3, 499, 0, -2, 77]
+    pandas.testing.assert_frame_equal(df.iloc[sel].to_pandas(),
+                                      pdf.iloc[sel])
+    assert df.iloc[42]["k"] == pdf.iloc[42]["k"]
+    assert abs(df["v"].iloc[-1] - pdf["v"].iloc[-1]) < 1e-15
