@@ -664,12 +664,15 @@ __global__ void __launch_bounds__(BLOCK) k_gb_accum(
 //   P0 k_gb_hist       : per-bucket row counts (8 B/row read; cached per key
 //                        column — immutable columns make this the device
 //                        analog of the reference's lazy metadata caches)
-//   P1 k_gb_scatter    : read 16 B/row, write 8 B val + 2 B lowkey into
-//                        exact per-bucket regions (LDS ranks, one global
-//                        cursor atomic per block-tile per bucket)
+//   P1 k_gb_scatter    : read 12 B/row (cached u32 key + f64 val), write
+//                        8 B val + 2 B lowkey into exact per-bucket regions
+//                        (LDS ranks, one global cursor atomic per
+//                        block-tile per bucket); software-pipelined across
+//                        tiles (round 2)
 //   P2 k_gb_bucket_agg : stream each bucket chunk into an LDS table
-//                        (ds_add_f64), merge non-empty slots into the global
-//                        dense table (atomics only per slot, not per row)
+//                        (ds_add_f64 behind a register run-accumulator for
+//                        skewed keys), merge non-empty slots into the
+//                        global dense table (atomics only per slot)
 // n_slots <= GB_RANGE skips P0/P1 entirely (k_gb_dense: one 16 B/row pass).
 // ---------------------------------------------------------------------------
 
