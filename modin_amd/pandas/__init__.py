@@ -417,6 +417,10 @@ class DataFrame(_HipPandasBase):
     def iloc(self):
         return _ILocIndexer(self)
 
+    def sort_index(self, ascending: bool = True):
+        return self._rewrap(
+            self._query_compiler.sort_index(bool(ascending)))
+
     def head(self, n: int = 5):
         total = len(self._query_compiler)
         stop = max(0, total + n) if n < 0 else n  # pandas head(-n)

@@ -497,6 +497,16 @@ class HipQueryCompiler:
             mask = mask.repartition_like(frame._row_lengths)
         return self.__constructor__(frame.filter_rows(mask))
 
+    def sort_index(self, ascending: bool = True) -> "HipQueryCompiler":
+        """pandas sort_index (host argsort of the index labels — the index
+        is host metadata — + one device gather per column)."""
+        idx = self._modin_frame.index
+        # Index.sort_values gives pandas' exact (stable, dup-safe) order
+        # for both directions — index labels are host metadata
+        _, order = idx.sort_values(return_indexer=True,
+                                   ascending=ascending)
+        return self.take_rows(order)
+
     def take_rows(self, positions) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.take_rows(positions))
 

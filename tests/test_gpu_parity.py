@@ -2271,3 +2271,15 @@ def test_iloc_vs_pandas(npartitions):
                                       pdf.iloc[sel])
     assert df.iloc[4242]["k"] == pdf.iloc[4242]["k"]
     assert abs(df["v"].iloc[-3] - pdf["v"].iloc[-3]) < 1e-15
+
+
+def test_sort_index_vs_pandas(npartitions):
+    rng = np.random.default_rng(4)
+    n = 30_000
+    pdf = pandas.DataFrame({"v": rng.random(n)},
+                           index=rng.integers(0, 500, n))
+    df = mpd.DataFrame(pdf)
+    for asc in (True, False):
+        got = df.sort_index(ascending=asc).to_pandas()
+        pandas.testing.assert_frame_equal(got,
+                                          pdf.sort_index(ascending=asc))

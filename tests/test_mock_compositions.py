@@ -449,3 +449,14 @@ def test_mock_iloc(mlib):
                                       pdf.iloc[sel])
     assert df.iloc[42]["k"] == pdf.iloc[42]["k"]
     assert abs(df["v"].iloc[-1] - pdf["v"].iloc[-1]) < 1e-15
+
+
+def test_mock_sort_index(mlib):
+    rng = np.random.default_rng(4)
+    pdf = pandas.DataFrame({"v": rng.random(300)},
+                           index=rng.integers(0, 40, 300))
+    df = mlib.DataFrame(pdf)
+    for asc in (True, False):
+        got = df.sort_index(ascending=asc).to_pandas()
+        pandas.testing.assert_frame_equal(got,
+                                          pdf.sort_index(ascending=asc))
