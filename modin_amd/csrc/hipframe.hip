@@ -1167,6 +1167,13 @@ __global__ void __launch_bounds__(BLOCK) k_fill_randcdf(
   }
 }
 
+__global__ void __launch_bounds__(BLOCK) k_i64_to_u32(
+    const int64_t* __restrict__ in, unsigned* __restrict__ out, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = (unsigned)in[i];
+}
+
 __global__ void __launch_bounds__(BLOCK) k_fill_i64(int64_t* __restrict__ p,
                                                     int64_t v, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -3964,16 +3971,43 @@ int hf_join_build(const hf_col* rkeys, const hf_col* const* rvals, int nr,
          hipMemcpyAsync(&h_err[1], d_fix_err, 8, hipMemcpyDeviceToHost,
                         g.stream));
   HF_HIP("hf_join_build", hipStreamSynchronize(g.stream));
-  if (h_err[0] || h_err[1]) {
+  if (h_err[0]) {
     hipMemsetAsync(d_hist_err, 0, 8, g.stream);
     hipMemsetAsync(d_fix_err, 0, 8, g.stream);
     hf_join_free(j);
     char buf[160];
     snprintf(buf, sizeof buf,
-             "%llu right keys outside [key_min, key_min+n_slots); %llu keys "
-             "exceeded the 4096-duplicate cap",
-             h_err[0], h_err[1]);
+             "%llu right keys outside [key_min, key_min+n_slots)",
+             h_err[0]);
     return set_err(HF_ERR_ARG, "hf_join_build", buf);
+  }
+  if (h_err[1]) {
+    // keys beyond the 4096-duplicate in-thread fixup (round 2): rebuild
+    // ORDER-CORRECT from a stable ascending key sort — within equal keys
+    // the stable sort preserves original right order, which is exactly
+    // the CSR layout (the offsets are fill-independent and already
+    // computed).  Removes the per-key multiplicity cap entirely.
+    hipMemsetAsync(d_fix_err, 0, 8, g.stream);
+    hf_col* perm = nullptr;
+    int r3 = hf_sort_perm(rkeys, 1, &perm);
+    if (r3 != HF_OK) { hf_join_free(j); return r3; }
+    r3 = timed_launch("join_sortfill", [&] {
+      hipLaunchKernelGGL(k_i64_to_u32, dim3((uint32_t)grid_for(n)),
+                         dim3(BLOCK), 0, g.stream,
+                         (const int64_t*)perm->dptr, j->d_jidx, n);
+    });
+    for (int c = 0; c < nr && r3 == HF_OK; ++c) {
+      hf_col* sv = nullptr;
+      r3 = hf_gather(rvals[c], perm, &sv);
+      if (r3 != HF_OK) break;
+      r3 = (hipMemcpyAsync(j->d_jval[c], sv->dptr, n * 8,
+                           hipMemcpyDeviceToDevice, g.stream) == hipSuccess)
+               ? HF_OK
+               : set_err(HF_ERR_HIP, "hf_join_build", "sortfill copy");
+      hf_col_free(sv);
+    }
+    hf_col_free(perm);
+    if (r3 != HF_OK) { hf_join_free(j); return r3; }
   }
   *out = j;
   return HF_OK;

@@ -2284,3 +2284,27 @@ def test_sort_index_vs_pandas(npartitions):
         got = df.sort_index(ascending=asc).to_pandas()
         pandas.testing.assert_frame_equal(got,
                                           pdf.sort_index(ascending=asc))
+
+
+def test_merge_big_duplicate_keys(npartitions):
+    """Per-key right multiplicity beyond the 4096 in-thread fixup: the
+    sorted-build fallback keeps pandas match order (round 2 — removes the
+    last r1 merge cap)."""
+    rng = np.random.default_rng(31)
+    nr = 60_000
+    rk = np.full(nr, 7, dtype=np.int64)
+    rk[rng.random(nr) < 0.2] = rng.integers(0, 50, nr)[rng.random(nr) < 0.2]
+    rv = rng.random(nr)
+    lk = np.array([7, 3, 7, 99, 12], dtype=np.int64)
+    lv = rng.random(5)
+    lpdf = pandas.DataFrame({"k": lk, "a": lv})
+    rpdf = pandas.DataFrame({"k": rk, "b": rv})
+    for how in ("inner", "left"):
+        got = mpd.DataFrame(lpdf).merge(mpd.DataFrame(rpdf), on="k",
+                                        how=how).to_pandas()
+        exp = lpdf.merge(rpdf, on="k", how=how)
+        assert len(got) == len(exp)
+        for c in ("k", "a", "b"):
+            np.testing.assert_allclose(
+                got[c].to_numpy(), exp[c].to_numpy(), rtol=0,
+                equal_nan=True, err_msg=f"{how}/{c}")
