@@ -2293,7 +2293,8 @@ def test_merge_big_duplicate_keys(npartitions):
     rng = np.random.default_rng(31)
     nr = 60_000
     rk = np.full(nr, 7, dtype=np.int64)
-    rk[rng.random(nr) < 0.2] = rng.integers(0, 50, nr)[rng.random(nr) < 0.2]
+    m = rng.random(nr) < 0.2
+    rk[m] = rng.integers(0, 50, int(m.sum()))
     rv = rng.random(nr)
     lk = np.array([7, 3, 7, 99, 12], dtype=np.int64)
     lv = rng.random(5)
