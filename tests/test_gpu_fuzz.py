@@ -30,7 +30,9 @@ def make_frame(rng, n):
     w = rng.integers(-1000, 1000, n).astype(np.int64)
     s = rng.choice(POOL[: rng.integers(2, len(POOL))], n).astype(object)
     s[rng.random(n) < rng.random() * 0.1] = np.nan
-    pdf = pandas.DataFrame({"k": k, "v": v, "w": w, "s": s})
+    t = (pandas.Timestamp("2000-01-01").value
+         + rng.integers(0, 10**18, n)).astype("datetime64[ns]")
+    pdf = pandas.DataFrame({"k": k, "v": v, "w": w, "s": s, "t": t})
     return pdf
 
 
@@ -46,6 +48,8 @@ def check(df, pdf, msg):
                 bn = isinstance(b, float) and np.isnan(b)
                 gn = isinstance(a, float) and np.isnan(a)
                 assert gn == bn and (gn or a == b), f"{msg}/{c}[{i}]"
+        elif np.issubdtype(e.dtype, np.datetime64):
+            np.testing.assert_array_equal(g, e, err_msg=f"{msg}/{c}")
         else:
             np.testing.assert_allclose(g, e, rtol=1e-12, atol=1e-9,
                                        equal_nan=True,
@@ -62,7 +66,7 @@ def test_fuzz_pipeline(seed):
     steps = rng.integers(2, 5)
     for si in range(steps):
         op = rng.choice(["filter", "sort", "head", "dropna", "arith",
-                         "round", "where", "dedup"])
+                         "round", "where", "dedup", "iloc", "strmask"])
         msg = f"seed {seed} step {si} op {op}"
         if op == "filter":
             thr = float(np.round(rng.standard_normal() * 10, 2))
@@ -92,12 +96,30 @@ def test_fuzz_pipeline(seed):
             pdf = pdf.reset_index(drop=True)
         elif op == "round":
             d = int(rng.integers(-1, 3))
-            df = df.round(d)
-            pdf = pdf.round(d)
+            df = df[["k", "v", "w"]].round(d)
+            pdf = pdf[["k", "v", "w"]].round(d)
         elif op == "where":
             thr = float(np.round(rng.standard_normal() * 5, 2))
-            df = df.where(df["v"] > thr)
-            pdf = pdf.where(pdf["v"] > thr)
+            sub2 = [c for c in ("k", "v", "w") if c in pdf.columns]
+            df = df[sub2].where(df["v"] > thr)
+            pdf = pdf[sub2].where(pdf["v"] > thr)
+        elif op == "iloc":
+            if rng.integers(0, 2):
+                a = int(rng.integers(0, max(len(pdf) - 1, 1)))
+                b = int(rng.integers(a, len(pdf) + 1))
+                df, pdf = df.iloc[a:b], pdf.iloc[a:b]
+            else:
+                sel = rng.integers(0, max(len(pdf), 1),
+                                   int(rng.integers(1, 2000))).tolist()
+                df, pdf = df.iloc[sel], pdf.iloc[sel]
+            df = mpd.DataFrame(df.to_pandas().reset_index(drop=True))
+            pdf = pdf.reset_index(drop=True)
+        elif op == "strmask":
+            pat = str(rng.choice(["a", "e", "ir", "d"]))
+            df = df[df["s"].str.contains(pat, na=False)]
+            pdf = pdf[pdf["s"].str.contains(pat, na=False)]
+            df = mpd.DataFrame(df.to_pandas().reset_index(drop=True))
+            pdf = pdf.reset_index(drop=True)
         elif op == "dedup":
             subs = [["k"], ["k", "s"], ["s", "w"], None][rng.integers(0, 4)]
             df = df.drop_duplicates(subs)
@@ -119,11 +141,19 @@ def test_fuzz_pipeline(seed):
         if len(pdf) == 0:
             break
 
+    # dt fields if the datetime column survived the pipeline
+    if len(pdf) and "t" in pdf.columns:
+        for f in ("year", "month", "dayofweek"):
+            g = np.asarray(getattr(df["t"].dt, f).to_pandas())
+            e = getattr(pdf["t"].dt, f).to_numpy()
+            np.testing.assert_array_equal(g, e,
+                                          err_msg=f"seed {seed} dt.{f}")
+
     # closing transform on whatever survived (original-row-order family)
-    if len(pdf):
+    if len(pdf) and "s" in pdf.columns:
         by = ["k", "s"][rng.integers(0, 2)]
         tr = ["cumsum", "cumcount", "shift", "rank", "tsum",
-              "tmean"][rng.integers(0, 6)]
+              "tmean", "cumprod"][rng.integers(0, 7)]
         sub = [by, "v", "w"]
         gb_g = df[sub].groupby(by)
         gb_p = pdf[sub].groupby(by)
@@ -151,10 +181,10 @@ def test_fuzz_pipeline(seed):
                     err_msg=f"seed {seed} {by}/{tr}/{c}")
 
     # closing aggregation on whatever survived
-    if len(pdf):
+    if len(pdf) and "s" in pdf.columns:
         by = ["k", "s"][rng.integers(0, 2)]
         agg = ["sum", "mean", "count", "min", "max", "var", "median",
-               "first", "last"][rng.integers(0, 9)]
+               "first", "last", "prod"][rng.integers(0, 10)]
         sub = [by, "v", "w"]  # numeric values only (string agg is loud)
         gout = getattr(df[sub].groupby(by), agg)().to_pandas()
         pout = getattr(pdf[sub].groupby(by), agg)()
