@@ -77,6 +77,7 @@ def test_fuzz_pipeline(seed):
         elif op == "sort":
             by = [["k"], ["s"], ["k", "w"], ["v"], ["s", "k"]][
                 rng.integers(0, 5)]
+            by = [b for b in by if b in pdf.columns] or ["v"]
             if "s" in by and pdf["s"].isna().any() and len(by) > 1 \
                     and by[0] != "s":
                 by = ["k", "w"]
@@ -96,8 +97,9 @@ def test_fuzz_pipeline(seed):
             pdf = pdf.reset_index(drop=True)
         elif op == "round":
             d = int(rng.integers(-1, 3))
-            df = df[["k", "v", "w"]].round(d)
-            pdf = pdf[["k", "v", "w"]].round(d)
+            sub2 = [c for c in ("k", "v", "w") if c in pdf.columns]
+            df = df[sub2].round(d)
+            pdf = pdf[sub2].round(d)
         elif op == "where":
             thr = float(np.round(rng.standard_normal() * 5, 2))
             sub2 = [c for c in ("k", "v", "w") if c in pdf.columns]
@@ -115,6 +117,8 @@ def test_fuzz_pipeline(seed):
             df = mpd.DataFrame(df.to_pandas().reset_index(drop=True))
             pdf = pdf.reset_index(drop=True)
         elif op == "strmask":
+            if "s" not in pdf.columns:
+                continue
             pat = str(rng.choice(["a", "e", "ir", "d"]))
             df = df[df["s"].str.contains(pat, na=False)]
             pdf = pdf[pdf["s"].str.contains(pat, na=False)]
@@ -122,6 +126,8 @@ def test_fuzz_pipeline(seed):
             pdf = pdf.reset_index(drop=True)
         elif op == "dedup":
             subs = [["k"], ["k", "s"], ["s", "w"], None][rng.integers(0, 4)]
+            if subs is not None:
+                subs = [c for c in subs if c in pdf.columns] or None
             df = df.drop_duplicates(subs)
             pdf = pdf.drop_duplicates(subset=subs)
             check(df, pdf, msg)
@@ -150,7 +156,7 @@ def test_fuzz_pipeline(seed):
                                           err_msg=f"seed {seed} dt.{f}")
 
     # closing transform on whatever survived (original-row-order family)
-    if len(pdf) and "s" in pdf.columns:
+    if len(pdf) and all(c in pdf.columns for c in ("s", "v", "w")):
         by = ["k", "s"][rng.integers(0, 2)]
         tr = ["cumsum", "cumcount", "shift", "rank", "tsum",
               "tmean", "cumprod"][rng.integers(0, 7)]
@@ -181,7 +187,7 @@ def test_fuzz_pipeline(seed):
                     err_msg=f"seed {seed} {by}/{tr}/{c}")
 
     # closing aggregation on whatever survived
-    if len(pdf) and "s" in pdf.columns:
+    if len(pdf) and all(c in pdf.columns for c in ("s", "v", "w")):
         by = ["k", "s"][rng.integers(0, 2)]
         agg = ["sum", "mean", "count", "min", "max", "var", "median",
                "first", "last", "prod"][rng.integers(0, 10)]
