@@ -1416,8 +1416,11 @@ class HipDataframe:
             key = by if isinstance(by, str) else \
                 (by[0] if len(by) == 1 else None)
             if key is None:
-                raise lib.HfError("distributed multi-key nunique is a "
-                                  "later round")
+                cf, decode = self._combined_key_frame(list(by))
+                keep = [c for c in cf.columns if c not in by]
+                res = cf.take_columns(keep).groupby_nunique(self.KEYCOL)
+                res._index = decode(np.asarray(res.index).astype(np.int64))
+                return res
             shuf = self._shuffle_frame_by_key(key)
             with dist_mod.local_mode():
                 res = shuf.groupby_nunique(key)
@@ -2169,8 +2172,11 @@ class HipDataframe:
             key = by if isinstance(by, str) else \
                 (by[0] if len(by) == 1 else None)
             if key is None:
-                raise lib.HfError("distributed multi-key prod is a later "
-                                  "round")
+                cf, decode = self._combined_key_frame(list(by))
+                keep = [c for c in cf.columns if c not in by]
+                res = cf.take_columns(keep).groupby_prod(self.KEYCOL)
+                res._index = decode(np.asarray(res.index).astype(np.int64))
+                return res
             shuf = self._shuffle_frame_by_key(key)
             with dist_mod.local_mode():
                 res = shuf.groupby_prod(key)
@@ -2256,8 +2262,12 @@ class HipDataframe:
             key = by if isinstance(by, str) else \
                 (by[0] if len(by) == 1 else None)
             if key is None:
-                raise lib.HfError("distributed multi-key idxmax/idxmin "
-                                  "is a later round")
+                cf, decode = self._combined_key_frame(list(by))
+                keep = [c for c in cf.columns if c not in by]
+                res = cf.take_columns(keep).groupby_idxminmax(
+                    self.KEYCOL, maximum)
+                res._index = decode(np.asarray(res.index).astype(np.int64))
+                return res
             # shuffle WITH global positions: local row order after the
             # exchange is global-position-ascending, so the local pick is
             # the right ROW; its label maps through the position column

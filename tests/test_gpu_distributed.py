@@ -154,6 +154,19 @@ def _worker(rank, world, port, fail_q):
                     expect[c].to_numpy(dtype=np.float64), rtol=0,
                     equal_nan=True, err_msg=f"idx mx={mx}/{c}")
 
+        # ---- distributed MULTI-KEY tail aggs (combined-key route) ----
+        qpdf2 = qpdf.assign(k2=(qpdf["k"] % 7))
+        qdf2 = mpd.DataFrame(qpdf2.iloc[qlo:qhi].reset_index(drop=True))
+        for op in ("median", "nunique"):
+            out = getattr(qdf2.groupby(["k", "k2"]), op)().to_pandas()
+            expect = getattr(qpdf2.groupby(["k", "k2"]), op)()
+            assert list(out.index) == list(expect.index), f"mk-{op} keys"
+            for c in ("v", "w"):
+                np.testing.assert_allclose(
+                    out[c].to_numpy(dtype=np.float64),
+                    expect[c].to_numpy(dtype=np.float64), rtol=1e-12,
+                    atol=1e-12, equal_nan=True, err_msg=f"mk-{op}/{c}")
+
         # ---- distributed transforms (shuffle + local + route-back):
         # row-aligned results on each rank's own shard ----
         tdf_p = qpdf.iloc[qlo:qhi].reset_index(drop=True)
