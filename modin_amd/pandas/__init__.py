@@ -326,6 +326,40 @@ def _unwrap(other):
     return other._query_compiler if isinstance(other, _HipPandasBase) else other
 
 
+class _LocIndexer:
+    """df.loc — the forms the hot path uses: boolean-mask rows (Series
+    mask), optionally with a column list/name; ':' rows with columns;
+    RangeIndex label slices (inclusive stop, pandas loc semantics)."""
+
+    def __init__(self, df):
+        self._df = df
+
+    def __getitem__(self, key):
+        rows, cols = key if isinstance(key, tuple) else (key, None)
+        out = self._df
+        if isinstance(rows, Series):
+            out = out[rows]
+        elif isinstance(rows, slice):
+            if rows.start is None and rows.stop is None:
+                pass
+            else:
+                idx = self._df.index
+                if not isinstance(idx, pandas.RangeIndex) or \
+                        idx.step != 1:
+                    raise lib.HfError("loc: label slices need a RangeIndex")
+                a = idx.start if rows.start is None else int(rows.start)
+                b = (idx.stop - 1) if rows.stop is None else int(rows.stop)
+                out = out._rewrap(out._query_compiler.take_row_range(
+                    a - idx.start, b - idx.start + 1))
+        else:
+            raise lib.HfError(f"loc: unsupported row selector {type(rows)}")
+        if cols is None:
+            return out
+        if isinstance(cols, str):
+            return out[cols]
+        return out[list(cols)]
+
+
 class _ILocIndexer:
     """df.iloc — positional rows: integer slices (step 1/None) ride the
     device column slice, integer lists/arrays ride one device gather per
@@ -416,6 +450,10 @@ class DataFrame(_HipPandasBase):
     @property
     def iloc(self):
         return _ILocIndexer(self)
+
+    @property
+    def loc(self):
+        return _LocIndexer(self)
 
     def sort_index(self, ascending: bool = True):
         return self._rewrap(

@@ -2309,3 +2309,20 @@ def test_merge_big_duplicate_keys(npartitions):
             np.testing.assert_allclose(
                 got[c].to_numpy(), exp[c].to_numpy(), rtol=0,
                 equal_nan=True, err_msg=f"{how}/{c}")
+
+
+def test_loc_vs_pandas(npartitions):
+    rng = np.random.default_rng(6)
+    n = 30_000
+    pdf = pandas.DataFrame({"k": rng.integers(0, 9, n),
+                            "v": rng.random(n),
+                            "w": rng.integers(-5, 5, n)})
+    df = mpd.DataFrame(pdf)
+    m = pdf["v"] > 0.5
+    pandas.testing.assert_frame_equal(
+        df.loc[df["v"] > 0.5].to_pandas(), pdf.loc[m])
+    pandas.testing.assert_frame_equal(
+        df.loc[df["v"] > 0.5, ["k", "w"]].to_pandas(),
+        pdf.loc[m, ["k", "w"]])
+    pandas.testing.assert_frame_equal(df.loc[100:2000].to_pandas(),
+                                      pdf.loc[100:2000])

@@ -460,3 +460,20 @@ def test_mock_sort_index(mlib):
         got = df.sort_index(ascending=asc).to_pandas()
         pandas.testing.assert_frame_equal(got,
                                           pdf.sort_index(ascending=asc))
+
+
+def test_mock_loc(mlib):
+    rng = np.random.default_rng(6)
+    pdf = pandas.DataFrame({"k": rng.integers(0, 9, 400),
+                            "v": rng.random(400),
+                            "w": rng.integers(-5, 5, 400)})
+    df = mlib.DataFrame(pdf)
+    m = pdf["v"] > 0.5
+    got = df.loc[df["v"] > 0.5].to_pandas()
+    pandas.testing.assert_frame_equal(got, pdf.loc[m])
+    got = df.loc[df["v"] > 0.5, ["k", "w"]].to_pandas()
+    pandas.testing.assert_frame_equal(got, pdf.loc[m, ["k", "w"]])
+    got = df.loc[10:20].to_pandas()
+    pandas.testing.assert_frame_equal(got, pdf.loc[10:20])
+    got = df.loc[:, ["v"]].to_pandas()
+    pandas.testing.assert_frame_equal(got, pdf.loc[:, ["v"]])
