@@ -41,6 +41,14 @@ def read_parquet(path, columns=None):
     return DataFrame(query_compiler=_rp(path, columns=columns))
 
 
+def read_csv(path, columns=None, **csv_kwargs):
+    """CSV ingestion straight to device via pyarrow.csv — the reference's
+    headline read_csv op, columnar, no per-row pandas materialization."""
+    from ..io import read_csv as _rc
+    return DataFrame(query_compiler=_rc(path, columns=columns,
+                                        **csv_kwargs))
+
+
 def concat(objs, ignore_index: bool = False):
     """pandas.concat(axis=0) over DataFrames with identical columns."""
     objs = list(objs)

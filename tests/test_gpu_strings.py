@@ -440,3 +440,32 @@ def test_str_accessor_vs_pandas(npartitions):
             pandas.isna(exp_sel["s"]).to_numpy()) | \
         (sel["s"].to_numpy() == exp_sel["s"].to_numpy())
     assert same.all()
+
+
+def test_read_csv_vs_pandas(tmp_path, npartitions):
+    """read_csv columnar ingestion: pyarrow.csv -> device columns; output
+    matches pandas.read_csv for int64/float64/string columns with nulls."""
+    rng = np.random.default_rng(73)
+    n = 20_000
+    sv = rng.choice(np.array(["red", "green", "blue", "x y"]),
+                    n).astype(object)
+    sv[rng.random(n) < 0.07] = None
+    v = rng.random(n).round(6)
+    v[rng.random(n) < 0.05] = np.nan
+    k = rng.integers(-1000, 1000, n)
+    pdf = pandas.DataFrame({"k": k, "v": v, "s": sv})
+    path = str(tmp_path / "t.csv")
+    pdf.to_csv(path, index=False)
+    exp = pandas.read_csv(path)
+    got = mpd.read_csv(path).to_pandas()
+    assert list(got.columns) == list(exp.columns)
+    np.testing.assert_array_equal(got["k"].to_numpy(), exp["k"].to_numpy())
+    np.testing.assert_allclose(got["v"].to_numpy(), exp["v"].to_numpy(),
+                               rtol=0, atol=1e-12, equal_nan=True)
+    ge, ee = got["s"].to_numpy(), exp["s"].to_numpy()
+    same = (pandas.isna(ge) & pandas.isna(ee)) | (ge == ee)
+    assert same.all()
+    # groupby straight off the ingested frame
+    g2 = mpd.read_csv(path).groupby("k").count().to_pandas()
+    e2 = exp.groupby("k").count()
+    np.testing.assert_array_equal(g2["v"].to_numpy(), e2["v"].to_numpy())
