@@ -3210,8 +3210,31 @@ class HipDataframe:
     def astype_all(self, dtype) -> "HipDataframe":
         if self._partitions and self._partitions[0].block().cats:
             raise lib.HfError("astype on string columns is a later round")
+        dt = np.dtype(dtype)
+        if np.issubdtype(dt, np.datetime64):
+            # int64 ns view -> datetime tag (device data unchanged; float
+            # sources cast to int64 ns first, pandas' rule)
+            def to_dt(block: DeviceBlock) -> DeviceBlock:
+                return DeviceBlock(
+                    {n: (c if c.dtype_code == lib.HF_INT64
+                         else lib.map_scalar(lib.MAP_CAST_I64, c, 0))
+                     for n, c in block.columns.items()}, block.length)
+            out = self.map(to_dt)
+            out.dtypes = pandas.Series(
+                {c: np.dtype("datetime64[ns]") for c in self.columns})
+            return out
+        if any(isinstance(d, np.dtype) and np.issubdtype(d, np.datetime64)
+               for d in self.dtypes) and dt == np.dtype(np.int64):
+            # datetime -> int64: drop the tag (device already int64 ns)
+            out = HipDataframe(self._partitions, self._index, self.columns,
+                               self._row_lengths,
+                               pandas.Series({c: np.dtype(np.int64)
+                                              for c in self.columns}))
+            return out
         code = {np.dtype(np.float64): lib.MAP_CAST_F64,
-                np.dtype(np.int64): lib.MAP_CAST_I64}[np.dtype(dtype)]
+                np.dtype(np.int64): lib.MAP_CAST_I64}.get(dt)
+        if code is None:
+            raise lib.HfError(f"astype to {dt} is a later round")
 
         def block_fn(block: DeviceBlock) -> DeviceBlock:
             return DeviceBlock({n: lib.map_scalar(code, c, 0)

@@ -2326,3 +2326,16 @@ def test_loc_vs_pandas(npartitions):
         pdf.loc[m, ["k", "w"]])
     pandas.testing.assert_frame_equal(df.loc[100:2000].to_pandas(),
                                       pdf.loc[100:2000])
+
+
+def test_astype_datetime_round_trip(npartitions):
+    rng = np.random.default_rng(44)
+    n = 20_000
+    ns = rng.integers(0, 2 * 10**18, n)
+    pdf = pandas.DataFrame({"x": ns})
+    df = mpd.DataFrame(pdf)
+    as_dt = df.astype("datetime64[ns]").to_pandas()
+    pandas.testing.assert_frame_equal(as_dt, pdf.astype("datetime64[ns]"))
+    t = pandas.DataFrame({"t": ns.astype("datetime64[ns]")})
+    back = mpd.DataFrame(t).astype("int64").to_pandas()
+    pandas.testing.assert_frame_equal(back, t.astype("int64"))
