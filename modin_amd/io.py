@@ -107,6 +107,11 @@ def read_csv(path, columns=None, **csv_kwargs):
 
     from .query_compiler import HipQueryCompiler
 
+    if "convert_options" not in csv_kwargs:
+        # pandas reads empty string fields as NaN; pyarrow's default
+        # keeps them as "" — align with pandas
+        csv_kwargs["convert_options"] = pacsv.ConvertOptions(
+            strings_can_be_null=True)
     table = pacsv.read_csv(path, **csv_kwargs)
     names = list(table.column_names)
     if columns is not None:
