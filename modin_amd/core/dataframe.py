@@ -2666,6 +2666,122 @@ class HipDataframe:
         part = HipDataframePartition(DeviceBlock(cols, n, cats))
         return HipDataframe([part], pandas.RangeIndex(n), names, [n], dts)
 
+    def merge_multi(self, other: "HipDataframe", on: list,
+                    how: str = "inner") -> "HipDataframe":
+        """Multi-key merge: fold the key columns of BOTH sides into ONE
+        int64 key with SHARED mins/strides (global over the union, so the
+        fold is consistent cross-frame; float keys ride the ordered
+        canonical-NaN transform first — NaN tuples match like pandas;
+        string keys recode into the left dictionary).  The right frame
+        drops its key columns (pandas on=[...] emits one set, from the
+        left), then the single-key engine runs on the fold and the
+        combined column drops from the result."""
+        from .. import distributed as dist_mod
+        KEY = "\x00mkey\x00"
+        lcats = (self._partitions[0].block().cats
+                 if self._partitions else {})
+        rcats = (other._partitions[0].block().cats
+                 if other._partitions else {})
+
+        def concat_col(frame, name):
+            cs = [p.block().columns[name] for p in frame._partitions]
+            return cs[0] if len(cs) == 1 else lib.concat(cs)
+
+        mins, spans, enc_l, enc_r = [], [], [], []
+        for b in on:
+            if b not in self.columns or b not in other.columns:
+                raise lib.HfError(f"merge: key column {b!r} missing")
+            if (b in lcats) != (b in rcats):
+                raise lib.HfError("merge: key column is a string on one "
+                                  "side only")
+            lc, rc = concat_col(self, b), concat_col(other, b)
+            if b in lcats:
+                if not rcats[b].equals(lcats[b]):
+                    rc = recode_dict_col(rc, rcats[b], lcats[b],
+                                         missing=-2)
+                # shift codes so -2/-1 fold like ordinary keys
+                lo = -2
+                lc2, rc2 = lc, rc
+            elif (self.dtypes[b] == np.dtype(np.float64)
+                  or other.dtypes[b] == np.dtype(np.float64)):
+                lc2 = lib.ordered_i64(lib.cast_f64(lc))
+                rc2 = lib.ordered_i64(lib.cast_f64(rc))
+                lo = None
+            else:
+                lc2, rc2 = lc, rc
+                lo = None
+            mn = mx = None
+            for c in (lc2, rc2):
+                if c.length:
+                    r = lib.reduce(c)
+                    mn = r.imn if mn is None else min(mn, r.imn)
+                    mx = r.imx if mx is None else max(mx, r.imx)
+            if dist_mod.is_active():
+                mn, mx = dist_mod.allreduce_minmax(mn, mx)
+            if mn is None:
+                mn, mx = 0, 0
+            if lo is not None:
+                mn = min(mn, lo)
+            if mx - mn + 1 > (1 << 31) and not dist_mod.is_active():
+                # wide span (float ordered keys span ~2^63): densify both
+                # sides through the sorted distinct UNION, fold on codes
+                both = lib.concat([lc2, rc2]) if rc2.length else lc2
+                perm = lib.sort_perm(both)
+                uniq, _u, _c, n_u = lib.groupby_sorted(
+                    lib.gather(both, perm), [], lib.AGG_SUM, False)
+                lc2 = lib.search_sorted(lc2, uniq)
+                rc2 = lib.search_sorted(rc2, uniq)
+                mn, mx = 0, max(int(n_u) - 1, 0)
+            mins.append(mn)
+            spans.append(mx - mn + 1)
+            enc_l.append(lc2)
+            enc_r.append(rc2)
+        total = 1
+        for sp in spans:
+            total *= sp
+            if total > (1 << 62):
+                raise lib.HfError(
+                    "merge: combined key range of "
+                    f"{on} exceeds 2^62 (hashed multi-key merge is a "
+                    "later round)")
+        strides = [1] * len(on)
+        for i in range(len(on) - 2, -1, -1):
+            strides[i] = strides[i + 1] * spans[i + 1]
+
+        def fold(cols):
+            comb = None
+            for c, mn, st in zip(cols, mins, strides):
+                t = lib.map_scalar(lib.MAP_SUB, c, mn)
+                if st != 1:
+                    t = lib.map_scalar(lib.MAP_MUL, t, st)
+                comb = t if comb is None else lib.binary(lib.BIN_ADD,
+                                                         comb, t)
+            return comb
+
+        def with_key(frame, comb, drop_keys):
+            names = [c for c in frame.columns
+                     if not (drop_keys and c in on)]
+            cats0 = (frame._partitions[0].block().cats
+                     if frame._partitions else {})
+            cols = {c: concat_col(frame, c) for c in names}
+            cols[KEY] = comb
+            n = comb.length
+            dts = {c: frame.dtypes[c] for c in names}
+            dts[KEY] = np.dtype(np.int64)
+            return HipDataframe(
+                [HipDataframePartition(DeviceBlock(
+                    cols, n, {c: v for c, v in cats0.items()
+                              if c in names}))],
+                pandas.RangeIndex(n), names + [KEY], [n],
+                pandas.Series(dts))
+
+        L = with_key(self, fold(enc_l), drop_keys=False)
+        R = with_key(other, fold(enc_r), drop_keys=True)
+        res = L.broadcast_join(R, KEY, how)
+        keep = [c for c in res.columns if c != KEY]
+        out = res.take_columns(keep)
+        return out
+
     def _binned_merge(self, other: "HipDataframe", on: str, how: str,
                       key_f64: bool, n_bins: int = 0) -> "HipDataframe":
         """Range-binned (co-shuffled) merge for giant right tables — the

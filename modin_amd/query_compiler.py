@@ -542,6 +542,17 @@ class HipQueryCompiler:
                 self._modin_frame.cross_join(right._modin_frame))
         if on is None:
             raise lib.HfError("merge: 'on' required (except how='cross')")
+        if isinstance(on, (list, tuple)):
+            if len(on) == 1:
+                on = on[0]
+            else:
+                if how not in ("inner", "left"):
+                    raise lib.HfError(
+                        f"multi-key merge how={how!r} is a later round "
+                        "(inner/left)")
+                return self.__constructor__(
+                    self._modin_frame.merge_multi(right._modin_frame,
+                                                  list(on), how))
         if how == "right":
             # pandas right join == swapped left join with the suffix roles
             # flipped back and columns restored to left-then-right order

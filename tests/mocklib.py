@@ -422,6 +422,34 @@ def install(monkeypatch):
         return (_reg(MockCol(uniq)), scols,
                 (ccols if counts else None), n)
 
+    class _MockJoin:
+        pass
+
+    def join_build(rkeys, rvals, key_min, n_slots):
+        j = _MockJoin()
+        j.k = rkeys.arr.copy()
+        j.v = [v.arr.copy() for v in rvals]
+        j.rdtypes = [v.dtype_code for v in rvals]
+        j.key_min = key_min
+        j.n_slots = n_slots
+        return j
+
+    def join_probe(j, lkeys):
+        import pandas as _pd
+        ldf = _pd.DataFrame({"k": lkeys.arr,
+                             "li": np.arange(lkeys.arr.size)})
+        rdf = _pd.DataFrame({"k": j.k, "ri": np.arange(j.k.size)})
+        m = ldf.merge(rdf, on="k", how="inner", sort=False)
+        keys = _reg(MockCol(m["k"].to_numpy().astype(np.int64)))
+        lidx = _reg(MockCol(m["li"].to_numpy().astype(np.int64)))
+        ri = m["ri"].to_numpy()
+        rcols = [_reg(MockCol(j.v[c][ri].copy()))
+                 for c in range(len(j.v))]
+        return keys, lidx, rcols, int(len(m))
+
+    def join_free(j):
+        pass
+
     for name, fn in [
         ("put", put), ("get", get), ("alloc", alloc),
         ("fill_f64", fill_f64), ("fill_i64", fill_i64),
@@ -441,5 +469,6 @@ def install(monkeypatch):
         ("groupby_sorted", groupby_sorted),
         ("groupby_hash_accum", groupby_hash_accum),
         ("groupby_hash_compact", groupby_hash_compact),
+        ("join_build", join_build), ("join_probe", join_probe),
     ]:
         monkeypatch.setattr(lib, name, fn)

@@ -477,3 +477,38 @@ def test_mock_loc(mlib):
     pandas.testing.assert_frame_equal(got, pdf.loc[10:20])
     got = df.loc[:, ["v"]].to_pandas()
     pandas.testing.assert_frame_equal(got, pdf.loc[:, ["v"]])
+
+
+def test_mock_multikey_merge(mlib):
+    """merge(on=[a,b]) inner/left: shared arithmetic fold over both
+    sides; float keys canonical-NaN match; string key recode."""
+    rng = np.random.default_rng(21)
+    nl, nr = 6000, 2500
+    lpdf = pandas.DataFrame({
+        "a": rng.integers(0, 40, nl),
+        "b": (rng.integers(-50, 50, nl) / 8.0),
+        "s": rng.choice(["u", "v", "w"], nl).astype(object),
+        "x": rng.random(nl)})
+    lpdf.loc[rng.random(nl) < 0.05, "b"] = np.nan
+    rpdf = pandas.DataFrame({
+        "a": rng.integers(0, 40, nr),
+        "b": (rng.integers(-50, 50, nr) / 8.0),
+        "s": rng.choice(["u", "v", "z"], nr).astype(object),
+        "y": rng.random(nr), "x": rng.random(nr)})
+    rpdf.loc[rng.random(nr) < 0.05, "b"] = np.nan
+    for keys in (["a", "b"], ["a", "s"], ["a", "b", "s"]):
+        for how in ("inner", "left"):
+            got = mlib.DataFrame(lpdf).merge(
+                mlib.DataFrame(rpdf), on=keys, how=how).to_pandas()
+            exp = lpdf.merge(rpdf, on=keys, how=how)
+            assert list(got.columns) == list(exp.columns), (keys, how)
+            assert len(got) == len(exp), (keys, how)
+            for c in exp.columns:
+                g, e = got[c].to_numpy(), exp[c].to_numpy()
+                if e.dtype == object:
+                    same = (pandas.isna(g) & pandas.isna(e)) | (g == e)
+                    assert same.all(), (keys, how, c)
+                else:
+                    np.testing.assert_allclose(
+                        g.astype(float), e.astype(float), rtol=0,
+                        equal_nan=True, err_msg=f"{keys}/{how}/{c}")
