@@ -7,6 +7,8 @@ the same chain.  Seeds are FIXED — failures reproduce exactly.  pandas is
 the arbiter the reference's own df_equals tests use.
 """
 
+import os
+
 import numpy as np
 import pandas
 import pytest
@@ -56,7 +58,8 @@ def check(df, pdf, msg):
                                        err_msg=f"{msg}/{c}")
 
 
-@pytest.mark.parametrize("seed", range(20))
+@pytest.mark.parametrize("seed", range(int(os.environ.get(
+    "HF_FUZZ_N", "20"))))
 def test_fuzz_pipeline(seed):
     rng = np.random.default_rng(1000 + seed)
     n = int(rng.integers(500, 40_000))
