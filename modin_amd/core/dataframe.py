@@ -2597,6 +2597,27 @@ class HipDataframe:
         return HipDataframe([part], self._index, list(cols), [n],
                             pandas.Series(dtypes))
 
+    def alias_column(self, name: str, new_name: str) -> "HipDataframe":
+        """Append ``new_name`` as a second reference to column ``name`` —
+        zero-copy (ColumnRef is refcounted, so both names share the device
+        buffer).  Used by the left_on/right_on merge rewrite."""
+        if new_name in self.columns:
+            raise lib.HfError(f"alias_column: {new_name!r} exists")
+        parts = []
+        for p, ln in zip(self._partitions, self._row_lengths):
+            b = p.block()
+            cols = dict(b.columns)
+            cols[new_name] = b.columns[name]
+            cats = dict(b.cats)
+            if name in cats:
+                cats[new_name] = cats[name]
+            parts.append(HipDataframePartition(DeviceBlock(cols, ln, cats)))
+        dtypes = self.dtypes.copy()
+        dtypes[new_name] = self.dtypes[name]
+        return HipDataframe(parts, self._index,
+                            list(self.columns) + [new_name],
+                            self._row_lengths, dtypes)
+
     # ---- broadcast inner join (MergeImpl.row_axis_merge device form,
     #      merge.py:104-178: combine() the right frame once, probe per left
     #      partition; pandas suffix rules "_x"/"_y" on collisions) ----

@@ -576,12 +576,14 @@ class DataFrame(_HipPandasBase):
                 by, ascending, na_position=na_position))
 
     def merge(self, other: "DataFrame", on: str = None,
-              how: str = "inner"):
+              how: str = "inner", left_on=None, right_on=None):
         """Merge on an int64/float64/string key column (modin/pandas API
         -> qc.merge -> broadcast-right device join); how='cross' takes no
-        key (cartesian product)."""
+        key (cartesian product); left_on/right_on keep both key columns
+        (pandas keep-keys rule)."""
         return DataFrame(query_compiler=self._query_compiler.merge(
-            other._query_compiler, on=on, how=how))
+            other._query_compiler, on=on, how=how,
+            left_on=left_on, right_on=right_on))
 
     def groupby(self, by, as_index: bool = True,
                 dropna: bool = True) -> "DataFrameGroupBy":

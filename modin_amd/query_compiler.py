@@ -534,7 +534,41 @@ class HipQueryCompiler:
     # ---- merge (query_compiler merge -> MergeImpl.row_axis_merge,
     #      storage_formats/pandas/merge.py:104) ----
     def merge(self, right: "HipQueryCompiler", on: str = None,
-              how: str = "inner") -> "HipQueryCompiler":
+              how: str = "inner", left_on=None,
+              right_on=None) -> "HipQueryCompiler":
+        if left_on is not None or right_on is not None:
+            # pandas left_on/right_on: both key columns survive in the
+            # output (merge.py keep_keys path).  Rewritten as an `on`
+            # merge over a zero-copy ALIAS of each key under one hidden
+            # name: the original key columns then ride the join as plain
+            # payload (suffixes, NaN fill for unmatched rows — exactly
+            # the pandas column semantics), and the hidden key is dropped.
+            if on is not None:
+                raise lib.HfError("merge: 'on' excludes left_on/right_on")
+            if left_on is None or right_on is None:
+                raise lib.HfError("merge: left_on and right_on must be "
+                                  "given together")
+            if isinstance(left_on, (list, tuple)):
+                if len(left_on) != 1:
+                    raise lib.HfError("merge: multi-key left_on/right_on "
+                                      "is a later round")
+                left_on = left_on[0]
+            if isinstance(right_on, (list, tuple)):
+                if len(right_on) != 1:
+                    raise lib.HfError("merge: multi-key left_on/right_on "
+                                      "is a later round")
+                right_on = right_on[0]
+            if left_on == right_on:
+                # pandas collapses same-named keys into the `on` form
+                return self.merge(right, on=left_on, how=how)
+            tmpk = "\x00lrk\x00"
+            l2 = self.__constructor__(
+                self._modin_frame.alias_column(left_on, tmpk))
+            r2 = self.__constructor__(
+                right._modin_frame.alias_column(right_on, tmpk))
+            merged = l2.merge(r2, on=tmpk, how=how)
+            return merged.getitem_column_array(
+                [c for c in merged.columns if c != tmpk])
         if how == "cross":
             if on is not None:
                 raise lib.HfError("merge: how='cross' forbids 'on'")

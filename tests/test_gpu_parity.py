@@ -2373,3 +2373,74 @@ def test_multikey_merge_vs_pandas(npartitions):
                     np.testing.assert_allclose(
                         g.astype(float), e.astype(float), rtol=0,
                         equal_nan=True, err_msg=f"{keys}/{how}/{c}")
+
+
+def test_merge_left_on_right_on_vs_pandas(npartitions):
+    """merge(left_on=, right_on=): rewrite over a zero-copy key alias —
+    both key columns survive as payload (pandas keep-keys rule), NaN
+    fills the unmatched side's key, suffixes on other collisions; all
+    four hows; int, float-NaN and dictionary (string) keys."""
+    rng = np.random.default_rng(123)
+    nl, nr = 30_000, 9_000
+    lpdf = pandas.DataFrame({
+        "a": rng.choice(np.r_[rng.standard_normal(200), np.nan], nl),
+        "v": rng.standard_normal(nl),
+        "c": rng.integers(0, 9, nl)})
+    rpdf = pandas.DataFrame({
+        "b": rng.choice(np.r_[rng.standard_normal(250), np.nan], nr),
+        "w": rng.standard_normal(nr),
+        "c": rng.integers(10, 19, nr)})
+    for how in ("inner", "left", "right", "outer"):
+        got = mpd.DataFrame(lpdf).merge(
+            mpd.DataFrame(rpdf), left_on="a", right_on="b",
+            how=how).to_pandas()
+        exp = lpdf.merge(rpdf, left_on="a", right_on="b", how=how)
+        assert list(got.columns) == list(exp.columns), how
+        assert len(got) == len(exp), how
+        order = ["a", "b", "v", "w", "c_x", "c_y"]
+        gs = got.sort_values(order, na_position="last").reset_index(drop=True)
+        es = exp.sort_values(order, na_position="last").reset_index(drop=True)
+        for c in exp.columns:
+            np.testing.assert_allclose(
+                gs[c].to_numpy().astype(float),
+                es[c].to_numpy().astype(float), rtol=0,
+                equal_nan=True, err_msg=f"{how}/{c}")
+    # dictionary keys under different names, mismatched dictionaries
+    lp = pandas.DataFrame({"s": rng.choice(["a", "b", "c", None], 4000),
+                           "v": rng.standard_normal(4000)})
+    rp = pandas.DataFrame({"t": rng.choice(["b", "c", "d", None], 1500),
+                           "w": rng.standard_normal(1500)})
+    for how in ("inner", "left"):
+        got = mpd.DataFrame(lp).merge(mpd.DataFrame(rp), left_on="s",
+                                      right_on="t", how=how).to_pandas()
+        exp = lp.merge(rp, left_on="s", right_on="t", how=how)
+        assert list(got.columns) == list(exp.columns)
+        assert len(got) == len(exp), how
+        gs = got.sort_values(["s", "t", "v", "w"]).reset_index(drop=True)
+        es = exp.sort_values(["s", "t", "v", "w"]).reset_index(drop=True)
+        for c in ("s", "t"):
+            np.testing.assert_array_equal(
+                gs[c].fillna("<NA>").to_numpy(),
+                es[c].fillna("<NA>").to_numpy(), err_msg=f"{how}/{c}")
+        for c in ("v", "w"):
+            np.testing.assert_allclose(gs[c].to_numpy(), es[c].to_numpy(),
+                                       rtol=0, equal_nan=True,
+                                       err_msg=f"{how}/{c}")
+    # int keys, len-1 list form, error surfaces
+    lp2 = pandas.DataFrame({"i": rng.integers(0, 40, 2000),
+                            "v": rng.standard_normal(2000)})
+    rp2 = pandas.DataFrame({"j": rng.integers(0, 40, 700),
+                            "w": rng.standard_normal(700)})
+    got = mpd.DataFrame(lp2).merge(mpd.DataFrame(rp2), left_on=["i"],
+                                   right_on=["j"]).to_pandas()
+    exp = lp2.merge(rp2, left_on="i", right_on="j")
+    assert len(got) == len(exp)
+    assert list(got.columns) == list(exp.columns)
+    np.testing.assert_allclose(np.sort(got["w"].to_numpy()),
+                               np.sort(exp["w"].to_numpy()), rtol=0)
+    import modin_amd.core.lib as hl
+    with pytest.raises(hl.HfError):
+        mpd.DataFrame(lp2).merge(mpd.DataFrame(rp2), left_on="i")
+    with pytest.raises(hl.HfError):
+        mpd.DataFrame(lp2).merge(mpd.DataFrame(rp2), on="i", left_on="i",
+                                 right_on="j")

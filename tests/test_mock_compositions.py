@@ -512,3 +512,43 @@ def test_mock_multikey_merge(mlib):
                     np.testing.assert_allclose(
                         g.astype(float), e.astype(float), rtol=0,
                         equal_nan=True, err_msg=f"{keys}/{how}/{c}")
+
+
+def test_mock_merge_left_on_right_on(mlib):
+    """merge(left_on=, right_on=): both key columns survive (pandas
+    keep-keys rule), suffixes over every other collision, NaN fill on
+    the unmatched side's key for left/right/outer."""
+    rng = np.random.default_rng(22)
+    nl, nr = 5000, 1800
+    lpdf = pandas.DataFrame({
+        "a": (rng.integers(0, 60, nl)).astype(np.float64),
+        "v": rng.random(nl),
+        "c": rng.integers(0, 9, nl)})
+    lpdf.loc[rng.random(nl) < 0.05, "a"] = np.nan
+    rpdf = pandas.DataFrame({
+        "b": (rng.integers(0, 80, nr)).astype(np.float64),
+        "w": rng.random(nr),
+        "c": rng.integers(10, 19, nr)})
+    rpdf.loc[rng.random(nr) < 0.05, "b"] = np.nan
+    for how in ("inner", "left", "right", "outer"):
+        got = mlib.DataFrame(lpdf).merge(
+            mlib.DataFrame(rpdf), left_on="a", right_on="b",
+            how=how).to_pandas()
+        exp = lpdf.merge(rpdf, left_on="a", right_on="b", how=how)
+        assert list(got.columns) == list(exp.columns), how
+        assert len(got) == len(exp), how
+        order = ["a", "b", "v", "w", "c_x", "c_y"]
+        gs = got.sort_values(order, na_position="last").reset_index(drop=True)
+        es = exp.sort_values(order, na_position="last").reset_index(drop=True)
+        for c in exp.columns:
+            np.testing.assert_allclose(
+                gs[c].to_numpy().astype(float),
+                es[c].to_numpy().astype(float), rtol=0,
+                equal_nan=True, err_msg=f"{how}/{c}")
+    # same-named keys collapse to the `on` form (one key column)
+    got = mlib.DataFrame(lpdf).merge(
+        mlib.DataFrame(rpdf.rename(columns={"b": "a"})),
+        left_on="a", right_on="a", how="inner").to_pandas()
+    exp = lpdf.merge(rpdf.rename(columns={"b": "a"}), on="a", how="inner")
+    assert list(got.columns) == list(exp.columns)
+    assert len(got) == len(exp)
