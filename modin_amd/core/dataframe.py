@@ -2618,6 +2618,88 @@ class HipDataframe:
                             list(self.columns) + [new_name],
                             self._row_lengths, dtypes)
 
+    def set_column(self, name: str, other: "HipDataframe") -> "HipDataframe":
+        """Replace-or-append column ``name`` with ``other``'s single column
+        (POSITIONAL assignment — the device form of the reference's
+        `__setitem__`/`insert` path, dataframe.py:2575 setitem_builder;
+        index alignment beyond equal length is the caller's concern).
+        ``other`` re-slices device-side to this frame's partitioning."""
+        if len(other) != len(self):
+            raise lib.HfError(
+                f"set_column: length mismatch ({len(other)} vs {len(self)})")
+        if len(other.columns) != 1:
+            raise lib.HfError("set_column: value must be a single column")
+        src = other.columns[0]
+        other = other.repartition_like(self._row_lengths)
+        parts = []
+        for p, q, ln in zip(self._partitions, other._partitions,
+                            self._row_lengths):
+            b, ob = p.block(), q.block()
+            cols = dict(b.columns)
+            cols[name] = ob.columns[src]
+            cats = dict(b.cats)
+            cats.pop(name, None)
+            if src in ob.cats:
+                cats[name] = ob.cats[src]
+            parts.append(HipDataframePartition(DeviceBlock(cols, ln, cats)))
+        columns = list(self.columns)
+        if name not in columns:
+            columns = columns + [name]
+        dtypes = self.dtypes.copy()
+        dtypes[name] = other.dtypes[src]
+        return HipDataframe(parts, self._index, columns,
+                            self._row_lengths, dtypes)
+
+    def set_scalar_column(self, name: str, value) -> "HipDataframe":
+        """Broadcast-scalar column assignment: df[name] = scalar.  Fills
+        device-side (no host array); strings become a one-entry
+        dictionary column; bools/ints land as int64 (our bool carrier),
+        floats/NaN/None as float64."""
+        parts, cats_add = [], None
+        if isinstance(value, str):
+            cats_add = pandas.Index([value])
+
+        def mk(ln):
+            if isinstance(value, str):
+                c = lib.alloc(ln, lib.HF_INT64)
+                lib.fill_i64(c.dptr(), 0, ln)
+                return c, np.dtype(object)
+            if value is None or (isinstance(value, float)
+                                 and np.isnan(value)):
+                c = lib.alloc(ln, lib.HF_FLOAT64)
+                lib.fill_f64(c.dptr(), float("nan"), ln)
+                return c, np.dtype(np.float64)
+            if isinstance(value, (bool, np.bool_, int, np.integer)):
+                c = lib.alloc(ln, lib.HF_INT64)
+                lib.fill_i64(c.dptr(), int(value), ln)
+                return c, np.dtype(np.int64)
+            c = lib.alloc(ln, lib.HF_FLOAT64)
+            lib.fill_f64(c.dptr(), float(value), ln)
+            return c, np.dtype(np.float64)
+
+        if isinstance(value, str):
+            dt = np.dtype(object)
+        elif isinstance(value, (bool, np.bool_, int, np.integer)):
+            dt = np.dtype(np.int64)
+        else:
+            dt = np.dtype(np.float64)
+        for p, ln in zip(self._partitions, self._row_lengths):
+            b = p.block()
+            cols = dict(b.columns)
+            cols[name], dt = mk(ln)
+            cats = dict(b.cats)
+            cats.pop(name, None)
+            if cats_add is not None:
+                cats[name] = cats_add
+            parts.append(HipDataframePartition(DeviceBlock(cols, ln, cats)))
+        columns = list(self.columns)
+        if name not in columns:
+            columns = columns + [name]
+        dtypes = self.dtypes.copy()
+        dtypes[name] = dt
+        return HipDataframe(parts, self._index, columns,
+                            self._row_lengths, dtypes)
+
     # ---- broadcast inner join (MergeImpl.row_axis_merge device form,
     #      merge.py:104-178: combine() the right frame once, probe per left
     #      partition; pandas suffix rules "_x"/"_y" on collisions) ----

@@ -455,6 +455,56 @@ class DataFrame(_HipPandasBase):
             )
         raise lib.HfError("only column selection / boolean masks are supported")
 
+    def _value_qc(self, value):
+        """Coerce a __setitem__/insert value to a 1-column qc (POSITIONAL
+        assignment; equal length enforced device-side).  Host arrays /
+        pandas Series upload through from_pandas (strings dictionary-
+        encode, datetimes tag)."""
+        if isinstance(value, Series):
+            return value._query_compiler
+        if isinstance(value, pandas.Series):
+            value = value.reset_index(drop=True)
+        elif isinstance(value, (list, np.ndarray, pandas.Index)):
+            value = pandas.Series(np.asarray(value))
+        else:
+            raise lib.HfError(
+                "setitem value must be a Series, array, list or scalar")
+        return HipQueryCompiler.from_pandas(value.to_frame(name="\x00v\x00"))
+
+    def __setitem__(self, key: str, value):
+        """df[col] = value — replace-or-append, POSITIONAL (length must
+        match; pandas' index-alignment beyond that is not replicated).
+        Scalars broadcast device-side with no host array; a boolean-mask
+        Series lands as int64 0/1 (our bool carrier)."""
+        if not isinstance(key, str):
+            raise lib.HfError("setitem key must be a column name")
+        if (np.isscalar(value) or value is None
+                or isinstance(value, (bool, np.bool_))):
+            self._query_compiler = \
+                self._query_compiler.write_scalar_column(key, value)
+            return
+        self._query_compiler = self._query_compiler.write_column(
+            key, self._value_qc(value))
+
+    def insert(self, loc: int, column: str, value):
+        """pandas DataFrame.insert: new column at position ``loc``."""
+        if column in list(self.columns):
+            raise lib.HfError(f"insert: column {column!r} already exists")
+        self[column] = value
+        cols = [c for c in self.columns if c != column]
+        cols.insert(int(loc), column)
+        self._query_compiler = \
+            self._query_compiler.getitem_column_array(cols)
+
+    def assign(self, **kwargs) -> "DataFrame":
+        """pandas DataFrame.assign over non-callable values."""
+        out = DataFrame(query_compiler=self._query_compiler)
+        for k, v in kwargs.items():
+            if callable(v):
+                v = v(out)
+            out[k] = v
+        return out
+
     @property
     def iloc(self):
         return _ILocIndexer(self)

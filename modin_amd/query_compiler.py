@@ -621,6 +621,17 @@ class HipQueryCompiler:
             self._modin_frame.broadcast_join(right._modin_frame, on, how)
         )
 
+    # ---- column assignment (reference qc.setitem / insert,
+    #      storage_formats/pandas/query_compiler.py setitem_builder) ----
+    def write_column(self, name: str,
+                     value: "HipQueryCompiler") -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.set_column(name, value._modin_frame))
+
+    def write_scalar_column(self, name: str, value) -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.set_scalar_column(name, value))
+
     # ---- projection ----
     def getitem_column_array(self, names) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.take_columns(list(names)))

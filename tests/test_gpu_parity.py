@@ -2444,3 +2444,58 @@ def test_merge_left_on_right_on_vs_pandas(npartitions):
     with pytest.raises(hl.HfError):
         mpd.DataFrame(lp2).merge(mpd.DataFrame(rp2), on="i", left_on="i",
                                  right_on="j")
+
+
+def test_setitem_insert_assign_vs_pandas(npartitions):
+    """Column assignment on device: derived columns (zero host traffic),
+    scalar broadcasts via device fill, host arrays (strings dictionary-
+    encode, datetimes tag), insert-at-position, assign."""
+    rng = np.random.default_rng(124)
+    n = 50_000
+    pdf = pandas.DataFrame({"a": rng.integers(0, 100, n),
+                            "v": rng.standard_normal(n)})
+    df = mpd.DataFrame(pdf)
+    exp = pdf.copy()
+    df["v"] = df["v"] * 3.0
+    exp["v"] = exp["v"] * 3.0
+    df["w"] = df["a"] - df["v"]
+    exp["w"] = exp["a"] - exp["v"]
+    df["k7"] = 7
+    exp["k7"] = 7
+    df["nn"] = np.nan
+    exp["nn"] = np.nan
+    df["s"] = "zz"
+    exp["s"] = "zz"
+    strs = rng.choice(["p", "q", None], n)
+    df["t"] = strs
+    exp["t"] = strs
+    ts = pandas.Series(
+        pandas.to_datetime("2024-01-01")
+        + pandas.to_timedelta(rng.integers(0, 10_000, n), unit="m"))
+    df["d"] = ts
+    exp["d"] = ts
+    got = df.to_pandas()
+    assert list(got.columns) == list(exp.columns)
+    assert got["d"].dtype == exp["d"].dtype
+    for c in exp.columns:
+        g, e = got[c].to_numpy(), exp[c].to_numpy()
+        if e.dtype == object:
+            same = (pandas.isna(g) & pandas.isna(e)) | (g == e)
+            assert same.all(), c
+        elif c == "d":
+            np.testing.assert_array_equal(g, e)
+        else:
+            np.testing.assert_allclose(g.astype(float), e.astype(float),
+                                       rtol=0, equal_nan=True, err_msg=c)
+    # derived column keeps working downstream: groupby over the new key
+    g1 = df.groupby("k7").sum().to_pandas()
+    assert len(g1) == 1
+    # insert + assign
+    df.insert(0, "z", df["a"] * 2)
+    exp.insert(0, "z", exp["a"] * 2)
+    assert list(df.columns) == list(exp.columns)
+    out = df.assign(q=lambda d: d["z"] + 1)
+    expq = exp.assign(q=lambda d: d["z"] + 1)
+    np.testing.assert_allclose(
+        out.to_pandas()["q"].to_numpy().astype(float),
+        expq["q"].to_numpy().astype(float), rtol=0)

@@ -15,6 +15,7 @@ import pandas
 import pytest
 
 from tests import mocklib
+from modin_amd.core.lib import HfError as _HfErr
 
 
 @pytest.fixture()
@@ -552,3 +553,69 @@ def test_mock_merge_left_on_right_on(mlib):
     exp = lpdf.merge(rpdf.rename(columns={"b": "a"}), on="a", how="inner")
     assert list(got.columns) == list(exp.columns)
     assert len(got) == len(exp)
+
+
+def test_mock_setitem_insert_assign(mlib):
+    """df[c] = series/array/scalar, insert(loc), assign: replace-or-
+    append positional assignment; scalar broadcast; multi-partition
+    value re-slice."""
+    rng = np.random.default_rng(23)
+    n = 3000
+    pdf = pandas.DataFrame({"a": rng.integers(0, 50, n),
+                            "v": rng.standard_normal(n)})
+    df = mlib.DataFrame(pdf)
+    exp = pdf.copy()
+    # derived-column assignment (replace + append)
+    df["v"] = df["v"] * 2.0
+    exp["v"] = exp["v"] * 2.0
+    df["w"] = df["a"] + df["v"]
+    exp["w"] = exp["a"] + exp["v"]
+    # scalar broadcasts (int / float / NaN / string)
+    df["k"] = 7
+    exp["k"] = 7
+    df["f"] = 2.5
+    exp["f"] = 2.5
+    df["nn"] = np.nan
+    exp["nn"] = np.nan
+    df["s"] = "hello"
+    exp["s"] = "hello"
+    # host array / pandas Series values
+    arr = rng.standard_normal(n)
+    df["h"] = arr
+    exp["h"] = arr
+    ps = pandas.Series(rng.integers(-9, 9, n))
+    df["i"] = ps
+    exp["i"] = ps
+    got = df.to_pandas()
+    assert list(got.columns) == list(exp.columns)
+    for c in exp.columns:
+        g, e = got[c].to_numpy(), exp[c].to_numpy()
+        if e.dtype == object:
+            np.testing.assert_array_equal(g, e, err_msg=c)
+        else:
+            np.testing.assert_allclose(g.astype(float), e.astype(float),
+                                       rtol=0, equal_nan=True, err_msg=c)
+    # insert at a position
+    df2 = mlib.DataFrame(pdf)
+    exp2 = pdf.copy()
+    df2.insert(1, "z", df2["a"] * 10)
+    exp2.insert(1, "z", exp2["a"] * 10)
+    got2 = df2.to_pandas()
+    assert list(got2.columns) == list(exp2.columns)
+    np.testing.assert_allclose(got2["z"].to_numpy(),
+                               exp2["z"].to_numpy(), rtol=0)
+    with pytest.raises(_HfErr):
+        df2.insert(0, "z", 1)  # duplicate
+    # assign (incl. callable) leaves the original untouched
+    df3 = mlib.DataFrame(pdf)
+    out = df3.assign(q=lambda d: d["v"] - 1.0, r=5)
+    assert list(df3.columns) == list(pdf.columns)
+    expq = pdf.assign(q=lambda d: d["v"] - 1.0, r=5)
+    gotq = out.to_pandas()
+    np.testing.assert_allclose(gotq["q"].to_numpy(),
+                               expq["q"].to_numpy(), rtol=1e-15)
+    np.testing.assert_array_equal(gotq["r"].to_numpy(),
+                                  expq["r"].to_numpy())
+    # length mismatch is loud
+    with pytest.raises(_HfErr):
+        df3["bad"] = np.zeros(n - 1)
