@@ -29,7 +29,7 @@ from ..core import lib
 from ..core.lib import HfError as HfErrorProxy
 from ..query_compiler import HipQueryCompiler
 
-__all__ = ["DataFrame", "Series", "concat", "from_pandas"]
+__all__ = ["DataFrame", "Series", "concat", "merge", "from_pandas"]
 
 
 def read_parquet(path, columns=None):
@@ -60,6 +60,11 @@ def concat(objs, ignore_index: bool = False):
         out._query_compiler._modin_frame._index = __import__(
             "pandas").RangeIndex(len(out))
     return out
+
+
+def merge(left: "DataFrame", right: "DataFrame", **kwargs) -> "DataFrame":
+    """Module-level pandas.merge form."""
+    return left.merge(right, **kwargs)
 
 
 def from_pandas(df: pandas.DataFrame) -> "DataFrame":
@@ -793,9 +798,27 @@ class Series(_HipPandasBase):
         return Series(query_compiler=qc.getitem_array(qc.notna()),
                       name=self.name)
 
-    def replace(self, to_replace, value) -> "Series":
+    def map(self, arg) -> "Series":  # noqa: A003
+        """pandas Series.map(dict): unmapped values -> NaN.  Dictionary
+        columns remap host-side + one gather; int64 columns via device
+        binary search over the key set."""
+        if isinstance(arg, pandas.Series):
+            arg = arg.to_dict()
+        if not isinstance(arg, dict):
+            raise lib.HfError("Series.map accepts a dict (callables are "
+                              "host-bound; a later round)")
+        return self._rewrap(self._query_compiler.map_dict(
+            arg, keep_missing=False))
+
+    def replace(self, to_replace, value=None) -> "Series":
         """pandas Series.replace(scalar, scalar): where(self != a, b) —
-        NaN rows survive untouched (NaN != a is True)."""
+        NaN rows survive untouched (NaN != a is True).  replace(dict):
+        unmapped values keep their value (device LUT path)."""
+        if isinstance(to_replace, dict):
+            if value is not None:
+                raise lib.HfError("replace(dict) takes no value")
+            return self._rewrap(self._query_compiler.map_dict(
+                to_replace, keep_missing=True))
         if not (np.isscalar(to_replace) and np.isscalar(value)):
             raise lib.HfError("replace: scalar to_replace/value only "
                               "this round")

@@ -322,6 +322,118 @@ class HipQueryCompiler:
                            [name], [n], dts)
         return self.__constructor__(res)
 
+    def map_dict(self, mapping: dict,
+                 keep_missing: bool) -> "HipQueryCompiler":
+        """Series.map(dict) (keep_missing=False: unmapped -> NaN, pandas
+        map rule) and Series.replace(dict) (keep_missing=True: unmapped
+        keep their value).  Dictionary columns remap on the HOST
+        DICTIONARY (O(#categories)) + one device gather; int64 columns
+        match exactly via hf_search_sorted over the sorted key set + LUT
+        gather.  float64 source columns are a later round (loud)."""
+        from modin_amd.core.dataframe import HipDataframe
+        from modin_amd.core.partition import DeviceBlock, \
+            HipDataframePartition
+        import pandas as pd
+        frame = self._modin_frame
+        name = frame.columns[0]
+        blk_cats = (frame._partitions[0].block().cats
+                    if frame._partitions else {})
+
+        def concat_col():
+            cs = [p.block().columns[name] for p in frame._partitions]
+            return cs[0] if len(cs) == 1 else lib.concat(cs)
+
+        n = len(frame)
+        _MISS = object()
+        if name in blk_cats:
+            cats = blk_cats[name].to_numpy(dtype=object)
+            vals = [mapping.get(c, _MISS) for c in cats]
+            if keep_missing:
+                vals = [c if v is _MISS else v
+                        for c, v in zip(cats, vals)]
+            codes = concat_col()
+            shifted = lib.map_scalar(lib.MAP_ADD, codes, 1)
+            if all(v is _MISS or isinstance(v, str) for v in vals):
+                vv = pd.Series([None if v is _MISS else v for v in vals],
+                               dtype=object)
+                new_codes, new_cats = pd.factorize(vv, sort=True)
+                lut = np.empty(len(cats) + 1, dtype=np.int64)
+                lut[0] = -1
+                lut[1:] = new_codes
+                out = lib.gather(lib.put(lut), shifted)
+                blk = DeviceBlock({name: out}, n,
+                                  {name: pd.Index(new_cats)})
+                dts = pd.Series({name: np.dtype(object)})
+            elif all(v is _MISS
+                     or isinstance(v, (int, float, np.integer, np.floating))
+                     for v in vals):
+                lut = np.empty(len(cats) + 1, dtype=np.float64)
+                lut[0] = np.nan
+                lut[1:] = [np.nan if v is _MISS else float(v)
+                           for v in vals]
+                out = lib.gather(lib.put(lut), shifted)
+                blk = DeviceBlock({name: out}, n)
+                dts = pd.Series({name: np.dtype(np.float64)})
+            else:
+                raise lib.HfError(
+                    "map/replace over a string column needs all-string "
+                    "or all-numeric mapped values")
+            res = HipDataframe([HipDataframePartition(blk)], frame._index,
+                               [name], [n], dts)
+            return self.__constructor__(res)
+        if frame.dtypes[name] != np.dtype(np.int64):
+            raise lib.HfError("map/replace(dict) over float64 columns is "
+                              "a later round (int64/string sources only)")
+        if not all(isinstance(k, (int, np.integer)) for k in mapping):
+            raise lib.HfError("map/replace(dict) on an int64 column needs "
+                              "int keys")
+        ks = np.array(sorted(mapping), dtype=np.int64)
+        vs = [mapping[int(k)] for k in ks]
+        if any(isinstance(v, float) and np.isnan(v) for v in vs) \
+                and keep_missing:
+            raise lib.HfError("replace(dict) with NaN values is a later "
+                              "round")
+        int_vals = all(isinstance(v, (bool, int, np.integer)) for v in vs)
+        col = concat_col()
+        pos = lib.search_sorted(col, lib.put(ks))
+        shifted = lib.map_scalar(lib.MAP_ADD, pos, 1)
+        if keep_missing:
+            m = lib.compare_scalar(lib.CMP_GE, pos, 0)
+            inv = lib.map_scalar(lib.MAP_RSUB, m, 1)
+            if int_vals:
+                lut = np.r_[np.int64(0), np.array(vs, dtype=np.int64)]
+                mapped = lib.gather(lib.put(lut), shifted)
+                out = lib.binary(lib.BIN_ADD,
+                                 lib.binary(lib.BIN_MUL, mapped, m),
+                                 lib.binary(lib.BIN_MUL, col, inv))
+                dt = np.dtype(np.int64)
+            else:
+                lut = np.r_[np.float64(0.0),
+                            np.array(vs, dtype=np.float64)]
+                mapped = lib.gather(lib.put(lut), shifted)
+                out = lib.binary(
+                    lib.BIN_ADD,
+                    lib.binary(lib.BIN_MUL, mapped, lib.cast_f64(m)),
+                    lib.binary(lib.BIN_MUL, lib.cast_f64(col),
+                               lib.cast_f64(inv)))
+                dt = np.dtype(np.float64)
+        else:
+            covered = n == 0 or lib.reduce(pos).imn >= 0
+            if covered and int_vals:
+                lut = np.r_[np.int64(0), np.array(vs, dtype=np.int64)]
+                out = lib.gather(lib.put(lut), shifted)
+                dt = np.dtype(np.int64)
+            else:
+                lut = np.empty(len(ks) + 1, dtype=np.float64)
+                lut[0] = np.nan
+                lut[1:] = np.array(vs, dtype=np.float64)
+                out = lib.gather(lib.put(lut), shifted)
+                dt = np.dtype(np.float64)
+        blk = DeviceBlock({name: out}, n)
+        res = HipDataframe([HipDataframePartition(blk)], frame._index,
+                           [name], [n], pd.Series({name: dt}))
+        return self.__constructor__(res)
+
     def dt_field(self, field: str) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.dt_field(field))
 

@@ -2488,8 +2488,10 @@ def test_setitem_insert_assign_vs_pandas(npartitions):
             np.testing.assert_allclose(g.astype(float), e.astype(float),
                                        rtol=0, equal_nan=True, err_msg=c)
     # derived column keeps working downstream: groupby over the new key
-    g1 = df.groupby("k7").sum().to_pandas()
+    g1 = df[["k7", "a", "v"]].groupby("k7").sum().to_pandas()
+    e1 = exp[["k7", "a", "v"]].groupby("k7").sum()
     assert len(g1) == 1
+    np.testing.assert_allclose(g1.to_numpy(), e1.to_numpy(), rtol=1e-12)
     # insert + assign
     df.insert(0, "z", df["a"] * 2)
     exp.insert(0, "z", exp["a"] * 2)
@@ -2499,3 +2501,43 @@ def test_setitem_insert_assign_vs_pandas(npartitions):
     np.testing.assert_allclose(
         out.to_pandas()["q"].to_numpy().astype(float),
         expq["q"].to_numpy().astype(float), rtol=0)
+
+
+def test_series_map_replace_dict_vs_pandas(npartitions):
+    """Series.map(dict)/replace(dict): int64 via hf_search_sorted LUT
+    (coverage-dependent result dtype), strings via host-dictionary remap
+    + one gather."""
+    rng = np.random.default_rng(125)
+    n = 80_000
+    pdf = pandas.DataFrame({"i": rng.integers(-50, 50, n),
+                            "s": rng.choice(["aa", "bb", "cc", None], n)})
+    df = mpd.DataFrame(pdf)
+    full = {k: int(k) * 3 - 7 for k in range(-50, 50)}
+    got = df["i"].map(full).to_pandas()
+    exp = pdf["i"].map(full)
+    assert got.dtype == exp.dtype
+    np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy())
+    part = {k: float(k) / 4 for k in range(-10, 10)}
+    got = df["i"].map(part).to_pandas()
+    exp = pdf["i"].map(part)
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy().astype(float),
+                               rtol=0, equal_nan=True)
+    rep = {0: 1000, -7: 7000, 13: -13000}
+    got = df["i"].replace(rep).to_pandas()
+    exp = pdf["i"].replace(rep)
+    assert got.dtype == exp.dtype
+    np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy())
+    got = df["s"].map({"aa": "x", "cc": "y"}).to_pandas()
+    exp = pdf["s"].map({"aa": "x", "cc": "y"})
+    g, e = got.to_numpy(), exp.to_numpy()
+    same = (pandas.isna(g) & pandas.isna(e)) | (g == e)
+    assert same.all()
+    got = df["s"].replace({"bb": "BB"}).to_pandas()
+    exp = pdf["s"].replace({"bb": "BB"})
+    g, e = got.to_numpy(), exp.to_numpy()
+    same = (pandas.isna(g) & pandas.isna(e)) | (g == e)
+    assert same.all()
+    got = df["s"].map({"aa": 1.5, "bb": 2, "cc": 3}).to_pandas()
+    exp = pdf["s"].map({"aa": 1.5, "bb": 2, "cc": 3})
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy().astype(float),
+                               rtol=0, equal_nan=True)

@@ -619,3 +619,56 @@ def test_mock_setitem_insert_assign(mlib):
     # length mismatch is loud
     with pytest.raises(_HfErr):
         df3["bad"] = np.zeros(n - 1)
+
+
+def test_mock_series_map_replace_dict(mlib):
+    """Series.map(dict) / replace(dict): int64 LUT via device binary
+    search; string columns remap on the host dictionary."""
+    rng = np.random.default_rng(24)
+    n = 4000
+    pdf = pandas.DataFrame({"i": rng.integers(0, 10, n),
+                            "s": rng.choice(["a", "b", "c", None], n)})
+    df = mlib.DataFrame(pdf)
+    # int map, full coverage + int values -> int64
+    full = {k: k * 10 for k in range(10)}
+    got = df["i"].map(full).to_pandas()
+    exp = pdf["i"].map(full)
+    assert got.dtype == exp.dtype == np.int64
+    np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy())
+    # int map, partial coverage -> float64 + NaN
+    part = {1: 100, 3: 300.5}
+    got = df["i"].map(part).to_pandas()
+    exp = pdf["i"].map(part)
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy().astype(float),
+                               rtol=0, equal_nan=True)
+    # int replace: unmapped keep their value, int64 preserved
+    got = df["i"].replace({2: -2, 5: -5}).to_pandas()
+    exp = pdf["i"].replace({2: -2, 5: -5})
+    assert got.dtype == exp.dtype
+    np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy())
+    # int replace with float values -> float64
+    got = df["i"].replace({2: 2.5}).to_pandas()
+    exp = pdf["i"].replace({2: 2.5})
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy().astype(float),
+                               rtol=0)
+    # string map -> string (missing cat -> NaN)
+    got = df["s"].map({"a": "X", "b": "Y"}).to_pandas()
+    exp = pdf["s"].map({"a": "X", "b": "Y"})
+    ge = got.to_numpy(), exp.to_numpy()
+    same = (pandas.isna(ge[0]) & pandas.isna(ge[1])) | (ge[0] == ge[1])
+    assert same.all()
+    # string map -> numeric
+    got = df["s"].map({"a": 1, "b": 2, "c": 3}).to_pandas()
+    exp = pdf["s"].map({"a": 1, "b": 2, "c": 3})
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy().astype(float),
+                               rtol=0, equal_nan=True)
+    # string replace: unmapped strings keep their value, NaN stays NaN
+    got = df["s"].replace({"a": "Q"}).to_pandas()
+    exp = pdf["s"].replace({"a": "Q"})
+    g, e = got.to_numpy(), exp.to_numpy()
+    same = (pandas.isna(g) & pandas.isna(e)) | (g == e)
+    assert same.all()
+    # module-level merge form
+    l = mlib.DataFrame({"k": [1, 2], "v": [1.0, 2.0]})
+    r = mlib.DataFrame({"k": [2, 3], "w": [5.0, 6.0]})
+    assert len(mlib.merge(l, r, on="k")) == 1
