@@ -640,6 +640,79 @@ class DataFrame(_HipPandasBase):
             other._query_compiler, on=on, how=how,
             left_on=left_on, right_on=right_on))
 
+    def melt(self, id_vars=None, value_vars=None, var_name=None,
+             value_name: str = "value") -> "DataFrame":
+        """pandas melt: wide -> long, entirely on device — per value
+        column a projection + zero-copy value alias + device-filled
+        `variable` column, then one concat (dictionary union recodes the
+        variable column's single-entry cats).  Mixed int64/float64 value
+        columns promote to float64 (pandas concat rule); mixing numeric
+        with string/datetime value columns is loud."""
+        qc = self._query_compiler
+        cols = list(self.columns)
+        if id_vars is None:
+            id_vars = []
+        elif isinstance(id_vars, str):
+            id_vars = [id_vars]
+        else:
+            id_vars = list(id_vars)
+        if value_vars is None:
+            value_vars = [c for c in cols if c not in id_vars]
+        elif isinstance(value_vars, str):
+            value_vars = [value_vars]
+        else:
+            value_vars = list(value_vars)
+        var_name = var_name or "variable"
+        if not value_vars:
+            raise lib.HfError("melt: no value columns")
+        for c in id_vars + value_vars:
+            if c not in cols:
+                raise lib.HfError(f"melt: unknown column {c!r}")
+        if value_name in id_vars or var_name in id_vars \
+                or var_name == value_name:
+            raise lib.HfError("melt: var_name/value_name collide with "
+                              "id_vars")
+        dts = self.dtypes
+        vset = {str(dts[v]) for v in value_vars}
+        cast_f64 = False
+        if len(vset) > 1:
+            if vset <= {"int64", "float64"}:
+                cast_f64 = True
+            else:
+                raise lib.HfError(
+                    f"melt: value columns mix incompatible dtypes {vset}")
+        pieces = []
+        for v in value_vars:
+            sub = qc.getitem_column_array(id_vars + [v])
+            if cast_f64 and str(dts[v]) == "int64":
+                sub = sub.write_column(
+                    v, qc.getitem_column_array([v]).astype(np.float64))
+            sub = sub.rename_columns({v: value_name})
+            sub = sub.write_scalar_column(var_name, v)
+            pieces.append(sub.getitem_column_array(
+                id_vars + [var_name, value_name]))
+        out = pieces[0].concat(pieces[1:]) if len(pieces) > 1 else pieces[0]
+        out._modin_frame._index = pandas.RangeIndex(len(out))
+        return DataFrame(query_compiler=out)
+
+    def pivot_table(self, values=None, index=None, columns=None,
+                    aggfunc: str = "mean", fill_value=None) -> "DataFrame":
+        """pandas pivot_table over single index/columns/values names:
+        the aggregation runs as a device multi-key groupby reduce; only
+        the REDUCED ngroups-sized table is reshaped host-side
+        (unstack)."""
+        for arg, nm in ((values, "values"), (index, "index"),
+                        (columns, "columns")):
+            if not isinstance(arg, str) or arg not in list(self.columns):
+                raise lib.HfError(
+                    f"pivot_table: {nm} must name one column")
+        sub = self[[index, columns, values]]
+        red = sub.groupby([index, columns]).agg(aggfunc).to_pandas()
+        wide = red[values].unstack(level=-1)
+        if fill_value is not None:
+            wide = wide.fillna(fill_value)
+        return DataFrame(wide)
+
     def groupby(self, by, as_index: bool = True,
                 dropna: bool = True) -> "DataFrameGroupBy":
         bys = list(by) if isinstance(by, (list, tuple)) else [by]

@@ -2541,3 +2541,42 @@ def test_series_map_replace_dict_vs_pandas(npartitions):
     exp = pdf["s"].map({"aa": 1.5, "bb": 2, "cc": 3})
     np.testing.assert_allclose(got.to_numpy(), exp.to_numpy().astype(float),
                                rtol=0, equal_nan=True)
+
+
+def test_melt_pivot_table_vs_pandas(npartitions):
+    """melt on device (value alias + variable fill + concat) and
+    pivot_table (device multi-key groupby + host unstack of the reduced
+    table), NaN values included."""
+    rng = np.random.default_rng(126)
+    n = 60_000
+    pdf = pandas.DataFrame({
+        "id": rng.integers(0, 50, n),
+        "g": rng.choice(["r", "s", "t", "u"], n),
+        "x": rng.standard_normal(n),
+        "y": rng.integers(-5, 5, n),
+        "z": rng.standard_normal(n)})
+    pdf.loc[rng.random(n) < 0.1, "x"] = np.nan
+    df = mpd.DataFrame(pdf)
+    got = df.melt(id_vars=["id", "g"]).to_pandas()
+    exp = pdf.melt(id_vars=["id", "g"])
+    assert list(got.columns) == list(exp.columns)
+    assert len(got) == len(exp)
+    for c in exp.columns:
+        g, e = got[c].to_numpy(), exp[c].to_numpy()
+        if e.dtype == object:
+            np.testing.assert_array_equal(g, e, err_msg=c)
+        else:
+            np.testing.assert_allclose(g.astype(float), e.astype(float),
+                                       rtol=0, equal_nan=True, err_msg=c)
+    for aggfunc in ("mean", "sum", "min", "max"):
+        got = df.pivot_table(values="x", index="id", columns="g",
+                             aggfunc=aggfunc).to_pandas()
+        exp = pdf.pivot_table(values="x", index="id", columns="g",
+                              aggfunc=aggfunc)
+        assert list(got.columns) == list(exp.columns), aggfunc
+        np.testing.assert_array_equal(np.asarray(got.index),
+                                      np.asarray(exp.index))
+        np.testing.assert_allclose(got.to_numpy().astype(float),
+                                   exp.to_numpy().astype(float),
+                                   rtol=1e-12, equal_nan=True,
+                                   err_msg=aggfunc)

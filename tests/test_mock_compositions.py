@@ -672,3 +672,52 @@ def test_mock_series_map_replace_dict(mlib):
     l = mlib.DataFrame({"k": [1, 2], "v": [1.0, 2.0]})
     r = mlib.DataFrame({"k": [2, 3], "w": [5.0, 6.0]})
     assert len(mlib.merge(l, r, on="k")) == 1
+
+
+def test_mock_melt_pivot_table(mlib):
+    """melt (device-wide->long) and pivot_table (device groupby +
+    host reshape of the reduced table)."""
+    rng = np.random.default_rng(25)
+    n = 2000
+    pdf = pandas.DataFrame({
+        "id": rng.integers(0, 6, n),
+        "g": rng.choice(["r", "s", "t"], n),
+        "x": rng.standard_normal(n),
+        "y": rng.integers(-5, 5, n),
+        "z": rng.standard_normal(n)})
+    df = mlib.DataFrame(pdf)
+    got = df.melt(id_vars=["id", "g"]).to_pandas()
+    exp = pdf.melt(id_vars=["id", "g"])
+    assert list(got.columns) == list(exp.columns)
+    assert len(got) == len(exp)
+    for c in exp.columns:
+        g, e = got[c].to_numpy(), exp[c].to_numpy()
+        if e.dtype == object:
+            np.testing.assert_array_equal(g, e, err_msg=c)
+        else:
+            np.testing.assert_allclose(g.astype(float),
+                                       e.astype(float), rtol=0,
+                                       equal_nan=True, err_msg=c)
+    # subset value_vars, custom names, single value dtype (no cast)
+    got = df.melt(id_vars="id", value_vars=["x", "z"], var_name="V",
+                  value_name="W").to_pandas()
+    exp = pdf.melt(id_vars="id", value_vars=["x", "z"], var_name="V",
+                   value_name="W")
+    assert list(got.columns) == list(exp.columns)
+    np.testing.assert_allclose(got["W"].to_numpy(), exp["W"].to_numpy(),
+                               rtol=0)
+    np.testing.assert_array_equal(got["V"].to_numpy(),
+                                  exp["V"].to_numpy())
+    # pivot_table
+    for aggfunc in ("mean", "sum", "count"):
+        got = df.pivot_table(values="x", index="id", columns="g",
+                             aggfunc=aggfunc).to_pandas()
+        exp = pdf.pivot_table(values="x", index="id", columns="g",
+                              aggfunc=aggfunc)
+        assert list(got.columns) == list(exp.columns), aggfunc
+        np.testing.assert_array_equal(np.asarray(got.index),
+                                      np.asarray(exp.index))
+        np.testing.assert_allclose(got.to_numpy().astype(float),
+                                   exp.to_numpy().astype(float),
+                                   rtol=1e-12, equal_nan=True,
+                                   err_msg=aggfunc)
