@@ -2700,6 +2700,32 @@ class HipDataframe:
         return HipDataframe(parts, self._index, columns,
                             self._row_lengths, dtypes)
 
+    def sample_rows(self, n: int, seed: int) -> "HipDataframe":
+        """Uniform sample WITHOUT replacement, fully device-side: one
+        splitmix64 uniform key per row (hf_fill_randf64), the n smallest
+        keys win (hf_sort_perm + gather) — the draw never leaves HBM;
+        only the n selected index labels come host-side.  Row order is
+        the random draw order (pandas sample also permutes)."""
+        total = len(self)
+        if not 0 <= n <= total:
+            raise lib.HfError(f"sample: n={n} out of range 0..{total}")
+
+        def concat_col(name):
+            cols = [p.block().columns[name] for p in self._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        keys = lib.fill_randf64(total, seed)
+        perm = lib.col_slice(lib.sort_perm(keys), 0, n)
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        out_cols = {c: lib.gather(concat_col(c), perm)
+                    for c in self.columns}
+        idx = pandas.Index(np.asarray(self.index))[lib.get(perm)]
+        part = HipDataframePartition(
+            DeviceBlock(out_cols, n, dict(blk_cats)))
+        return HipDataframe([part], idx, list(self.columns), [n],
+                            self.dtypes.copy())
+
     # ---- broadcast inner join (MergeImpl.row_axis_merge device form,
     #      merge.py:104-178: combine() the right frame once, probe per left
     #      partition; pandas suffix rules "_x"/"_y" on collisions) ----

@@ -722,6 +722,21 @@ class DataFrame(_HipPandasBase):
                     "groupby(by=<column name> | [column names]) only")
         return DataFrameGroupBy(self, by, as_index=as_index, dropna=dropna)
 
+    def sample(self, n: int = None, frac: float = None,
+               random_state=None) -> "DataFrame":
+        """pandas sample(replace=False): device-side draw (one uniform
+        key per row; the n smallest win) — the row VALUES never leave
+        the GPU.  The permutation differs from pandas' MT19937 stream
+        for a given random_state (documented deviation; the sample is
+        still uniform without replacement)."""
+        if (n is None) == (frac is None):
+            raise lib.HfError("sample: exactly one of n/frac")
+        if frac is not None:
+            n = int(round(frac * len(self)))
+        seed = 0x5A11 if random_state is None else int(random_state)
+        return DataFrame(
+            query_compiler=self._query_compiler.sample_rows(n, seed))
+
     def nlargest(self, n: int, columns: str):
         """pandas nlargest(keep='first'): stable descending NaN-last sort
         + head(n) (NaN rows only appear once n exceeds the non-NaN

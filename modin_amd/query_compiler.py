@@ -744,6 +744,10 @@ class HipQueryCompiler:
         return self.__constructor__(
             self._modin_frame.set_scalar_column(name, value))
 
+    def sample_rows(self, n: int, seed: int) -> "HipQueryCompiler":
+        return self.__constructor__(
+            self._modin_frame.sample_rows(int(n), int(seed)))
+
     # ---- projection ----
     def getitem_column_array(self, names) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.take_columns(list(names)))

@@ -103,6 +103,10 @@ def install(monkeypatch):
                                        if keys.arr.size else keys.arr,
                                        kind="stable")))
 
+    def fill_randf64(n, seed):
+        from oracle import ops as _o
+        return _reg(MockCol(_o.rand_f64(seed, n)))
+
     def cumsum(col, agg_op=0):
         x = col.arr
         if col.dtype_code == HF_INT64:
@@ -455,6 +459,7 @@ def install(monkeypatch):
         ("fill_f64", fill_f64), ("fill_i64", fill_i64),
         ("col_slice", col_slice), ("concat", concat), ("gather", gather),
         ("scatter", scatter), ("sort_perm", sort_perm),
+        ("fill_randf64", fill_randf64),
         ("cumsum", cumsum), ("seg_cumsum", seg_cumsum),
         ("filter_plan", filter_plan), ("filter_apply", filter_apply),
         ("filter_iota", filter_iota), ("compare_scalar", compare_scalar),

@@ -2580,3 +2580,30 @@ def test_melt_pivot_table_vs_pandas(npartitions):
                                    exp.to_numpy().astype(float),
                                    rtol=1e-12, equal_nan=True,
                                    err_msg=aggfunc)
+
+
+def test_sample_device_draw(npartitions):
+    """sample(n) on device: exact count, no duplicates, values/labels
+    consistent with the source, seed-reproducible, and the mock tier's
+    oracle-RNG mirror agrees with the device draw."""
+    rng = np.random.default_rng(127)
+    n = 200_000
+    pdf = pandas.DataFrame({"a": np.arange(n),
+                            "v": rng.standard_normal(n)})
+    df = mpd.DataFrame(pdf)
+    got = df.sample(n=1000, random_state=11).to_pandas()
+    assert len(got) == 1000
+    assert got["a"].is_unique
+    np.testing.assert_allclose(
+        got["v"].to_numpy(), pdf.loc[got["a"].to_numpy(), "v"].to_numpy(),
+        rtol=0)
+    got2 = df.sample(n=1000, random_state=11).to_pandas()
+    np.testing.assert_array_equal(got["a"].to_numpy(), got2["a"].to_numpy())
+    # device RNG == oracle mirror: the selected set is the argsort of
+    # the oracle's uniform stream
+    from oracle import ops as oops
+    keys = oops.rand_f64(11, n)
+    exp_sel = np.argsort(keys, kind="stable")[:1000]
+    np.testing.assert_array_equal(np.sort(got["a"].to_numpy()),
+                                  np.sort(exp_sel))
+    assert len(df.sample(frac=0.5)) == n // 2

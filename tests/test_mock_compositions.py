@@ -721,3 +721,30 @@ def test_mock_melt_pivot_table(mlib):
                                    exp.to_numpy().astype(float),
                                    rtol=1e-12, equal_nan=True,
                                    err_msg=aggfunc)
+
+
+def test_mock_sample(mlib):
+    """sample(n/frac): exact count, no duplicates, rows drawn from the
+    original (values + index labels consistent), seeded reproducible."""
+    rng = np.random.default_rng(26)
+    n = 5000
+    pdf = pandas.DataFrame({"a": np.arange(n),
+                            "v": rng.standard_normal(n)})
+    df = mlib.DataFrame(pdf)
+    got = df.sample(n=500, random_state=7).to_pandas()
+    assert len(got) == 500
+    assert got["a"].is_unique
+    # each sampled row matches the original at its label
+    np.testing.assert_allclose(
+        got["v"].to_numpy(), pdf.loc[got["a"].to_numpy(), "v"].to_numpy(),
+        rtol=0)
+    np.testing.assert_array_equal(np.asarray(got.index),
+                                  got["a"].to_numpy())
+    got2 = df.sample(n=500, random_state=7).to_pandas()
+    np.testing.assert_array_equal(got["a"].to_numpy(),
+                                  got2["a"].to_numpy())
+    assert len(df.sample(frac=0.25)) == n // 4
+    with pytest.raises(_HfErr):
+        df.sample(n=100, frac=0.5)
+    with pytest.raises(_HfErr):
+        df.sample(n=n + 1)
