@@ -244,6 +244,34 @@ class _HipPandasBase:
         return self._rewrap(type(self._query_compiler).rtruediv(self._query_compiler,
                                                                 _unwrap(other)))
 
+    def _int_scalar_op(self, qc_name, other, opname):
+        if not isinstance(other, (int, np.integer)) or other == 0:
+            raise lib.HfError(f"{opname}: nonzero int scalar only "
+                              "(float floor ops are a later round)")
+        bad = [c for c, dt in self._query_compiler.dtypes.items()
+               if dt != np.dtype(np.int64)]
+        if bad:
+            raise lib.HfError(f"{opname}: int64 columns only (got "
+                              f"{bad})")
+        return self._rewrap(getattr(type(self._query_compiler), qc_name)(
+            self._query_compiler, int(other)))
+
+    def __floordiv__(self, other):
+        """// with an int scalar over int64 columns: HF_MAP_IDIV (Python
+        floor semantics, exact toward -inf)."""
+        return self._int_scalar_op("floordiv_int", other, "floordiv")
+
+    def __mod__(self, other):
+        """% with an int scalar over int64 columns: HF_MAP_IMOD (Python
+        sign rule)."""
+        return self._int_scalar_op("mod_int", other, "mod")
+
+    def floordiv(self, other):
+        return self.__floordiv__(other)
+
+    def mod(self, other):
+        return self.__mod__(other)
+
     def add(self, other):
         return self.__add__(other)
 

@@ -64,6 +64,9 @@ class HipQueryCompiler:
     mul = Binary.register(lib.MAP_MUL, lib.BIN_MUL)
     truediv = Binary.register(lib.MAP_DIV, lib.BIN_DIV)
     rtruediv = Binary.register(lib.MAP_RDIV, lib.BIN_DIV)
+    # scalar-only int ops (HF_MAP_IDIV/IMOD: Python floor/mod semantics)
+    floordiv_int = Map.register(lib.MAP_IDIV)
+    mod_int = Map.register(lib.MAP_IMOD)
 
     # ---- TreeReduce (query_compiler.py:984 sum etc.) ----
     sum = TreeReduce.register("sum")

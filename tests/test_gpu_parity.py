@@ -2607,3 +2607,18 @@ def test_sample_device_draw(npartitions):
     np.testing.assert_array_equal(np.sort(got["a"].to_numpy()),
                                   np.sort(exp_sel))
     assert len(df.sample(frac=0.5)) == n // 2
+
+
+def test_floordiv_mod_vs_pandas(npartitions):
+    """int64 // and % with int scalars: HF_MAP_IDIV/IMOD Python floor
+    semantics, exact over negatives."""
+    rng = np.random.default_rng(128)
+    pdf = pandas.DataFrame({"a": rng.integers(-10**12, 10**12, 100_000)})
+    df = mpd.DataFrame(pdf)
+    for k in (7, -7, 86_400_000_000_000):
+        np.testing.assert_array_equal(
+            (df["a"] // k).to_pandas().to_numpy(),
+            (pdf["a"] // k).to_numpy(), err_msg=f"//{k}")
+        np.testing.assert_array_equal(
+            (df["a"] % k).to_pandas().to_numpy(),
+            (pdf["a"] % k).to_numpy(), err_msg=f"%{k}")

@@ -748,3 +748,20 @@ def test_mock_sample(mlib):
         df.sample(n=100, frac=0.5)
     with pytest.raises(_HfErr):
         df.sample(n=n + 1)
+
+
+def test_mock_floordiv_mod(mlib):
+    rng = np.random.default_rng(27)
+    pdf = pandas.DataFrame({"a": rng.integers(-100, 100, 3000)})
+    df = mlib.DataFrame(pdf)
+    for k in (7, -7, 3):
+        np.testing.assert_array_equal(
+            (df["a"] // k).to_pandas().to_numpy(),
+            (pdf["a"] // k).to_numpy(), err_msg=f"//{k}")
+        np.testing.assert_array_equal(
+            (df["a"] % k).to_pandas().to_numpy(),
+            (pdf["a"] % k).to_numpy(), err_msg=f"%{k}")
+    with pytest.raises(_HfErr):
+        df["a"] // 0
+    with pytest.raises(_HfErr):
+        df["a"] // 2.5
