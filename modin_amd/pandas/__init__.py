@@ -7,8 +7,10 @@ elementwise arithmetic and comparisons, map ops (fillna/abs/round/clip/
 where/mask), reductions (sum/mean/count/min/max/var/std/median/quantile/
 idxmax/idxmin, axis 0 and 1), sort_values (multi-key, na_position),
 filter/dropna/duplicated/drop_duplicates/nlargest, merge
-(inner/left/right/outer/cross, int64/float64/string keys incl. NaN
-matching), concat, and the groupby family: the reduce aggs + agg forms
+(inner/left/right/outer/cross, on/left_on+right_on, int64/float64/string
+keys incl. NaN matching), concat, column assignment
+(setitem/insert/assign), map/replace(dict), melt/pivot_table, sample,
+and the groupby family: the reduce aggs + agg forms
 (modin/pandas/dataframe.py:2188 sum; modin/pandas/groupby.py:1330
 DataFrameGroupBy.sum -> _wrap_aggregation), size/nunique/first/last/
 median/quantile/idxmax/idxmin, dropna=False, and the same-length
