@@ -2714,7 +2714,7 @@ class HipDataframe:
             cols = [p.block().columns[name] for p in self._partitions]
             return cols[0] if len(cols) == 1 else lib.concat(cols)
 
-        keys = lib.fill_randf64(total, seed)
+        keys = lib.ordered_i64(lib.fill_randf64(total, seed))
         perm = lib.col_slice(lib.sort_perm(keys), 0, n)
         blk_cats = (self._partitions[0].block().cats
                     if self._partitions else {})
