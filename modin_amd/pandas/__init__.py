@@ -670,6 +670,15 @@ class DataFrame(_HipPandasBase):
             other._query_compiler, on=on, how=how,
             left_on=left_on, right_on=right_on))
 
+    def corr(self) -> pandas.DataFrame:
+        """pandas corr (Pearson, pairwise-complete rows): masked moments
+        from NaN-propagating device passes; k x k host combine."""
+        return self._query_compiler.corr()
+
+    def cov(self, ddof: int = 1) -> pandas.DataFrame:
+        """pandas cov (pairwise-complete rows, ddof=1)."""
+        return self._query_compiler.cov(ddof=ddof)
+
     def melt(self, id_vars=None, value_vars=None, var_name=None,
              value_name: str = "value") -> "DataFrame":
         """pandas melt: wide -> long, entirely on device — per value

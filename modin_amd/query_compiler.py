@@ -118,6 +118,72 @@ class HipQueryCompiler:
         return pandas.Series(vals, index=pandas.Index(list(frame.columns)),
                              dtype=np.float64)
 
+    def corr(self) -> pandas.DataFrame:
+        return self._pairwise_moments(corr=True)
+
+    def cov(self, ddof: int = 1) -> pandas.DataFrame:
+        return self._pairwise_moments(corr=False, ddof=ddof)
+
+    def _pairwise_moments(self, corr: bool, ddof: int = 1):
+        """Pearson corr / cov over PAIRWISE-COMPLETE rows (pandas rule):
+        per pair, z = x + 0*y propagates NaN from either side and the
+        NaN-skipping f64 reduce then yields the masked n/Σ/Σ² in one
+        pass each — no mask materialization, 6 device passes per pair,
+        k×k scalars combined on host."""
+        import math
+        frame = self._modin_frame
+        blk_cats = (frame._partitions[0].block().cats
+                    if frame._partitions else {})
+        names = list(frame.columns)
+        bad = [c for c in names
+               if c in blk_cats or not (
+                   isinstance(frame.dtypes[c], np.dtype)
+                   and frame.dtypes[c] in (np.dtype(np.int64),
+                                           np.dtype(np.float64)))]
+        if bad:
+            raise lib.HfError(f"corr/cov: non-numeric columns {bad} — "
+                              "select numeric columns")
+        from .distributed import is_active, world_size
+        if is_active() and world_size() > 1:
+            raise lib.HfError("corr/cov at world>1 is a later round")
+
+        def concat_col(name):
+            cs = [p.block().columns[name] for p in frame._partitions]
+            c = cs[0] if len(cs) == 1 else lib.concat(cs)
+            return lib.cast_f64(c)
+
+        cols = {c: concat_col(c) for c in names}
+        k = len(names)
+        out = np.full((k, k), np.nan)
+        for i in range(k):
+            for j in range(i, k):
+                x, y = cols[names[i]], cols[names[j]]
+                zx = lib.binary(lib.BIN_ADD, x,
+                                lib.map_scalar(lib.MAP_MUL, y, 0.0))
+                zy = lib.binary(lib.BIN_ADD, y,
+                                lib.map_scalar(lib.MAP_MUL, x, 0.0))
+                rx, ry = lib.reduce(zx), lib.reduce(zy)
+                n = rx.count
+                sxy = lib.reduce(lib.binary(lib.BIN_MUL, zx, zy)).sum
+                sxx = lib.reduce(lib.binary(lib.BIN_MUL, zx, zx)).sum
+                syy = lib.reduce(lib.binary(lib.BIN_MUL, zy, zy)).sum
+                if corr:
+                    if n < 2:
+                        continue
+                    den = ((sxx - rx.sum ** 2 / n)
+                           * (syy - ry.sum ** 2 / n))
+                    if den <= 0:
+                        continue
+                    v = (sxy - rx.sum * ry.sum / n) / math.sqrt(den)
+                    v = max(-1.0, min(1.0, v))
+                else:
+                    if n - ddof <= 0:
+                        continue
+                    v = (sxy - rx.sum * ry.sum / n) / (n - ddof)
+                out[i, j] = out[j, i] = v
+        idx = pandas.Index(names)
+        return pandas.DataFrame(out, index=idx, columns=idx)
+
     def _tag_key_index(self, res: "HipQueryCompiler", by):
         """Groupby by a datetime64 key: the result index materializes back
         to the tagged dtype (the int64-ns typed-column layer)."""

@@ -2622,3 +2622,24 @@ def test_floordiv_mod_vs_pandas(npartitions):
         np.testing.assert_array_equal(
             (df["a"] % k).to_pandas().to_numpy(),
             (pdf["a"] % k).to_numpy(), err_msg=f"%{k}")
+
+
+def test_corr_cov_vs_pandas(npartitions):
+    """corr/cov: pairwise-complete masked moments on device (NaN-
+    propagating z = x + 0*y + NaN-skipping reduce)."""
+    rng = np.random.default_rng(129)
+    n = 150_000
+    pdf = pandas.DataFrame({"x": rng.standard_normal(n),
+                            "y": rng.standard_normal(n),
+                            "z": rng.standard_normal(n),
+                            "w": rng.integers(-50, 50, n)})
+    pdf["y"] += 0.7 * pdf["x"]
+    pdf.loc[rng.random(n) < 0.15, "x"] = np.nan
+    pdf.loc[rng.random(n) < 0.15, "y"] = np.nan
+    df = mpd.DataFrame(pdf)
+    np.testing.assert_allclose(df.corr().to_numpy(),
+                               pdf.corr().to_numpy(), rtol=1e-9,
+                               equal_nan=True)
+    np.testing.assert_allclose(df.cov().to_numpy(),
+                               pdf.cov().to_numpy(), rtol=1e-9,
+                               equal_nan=True)

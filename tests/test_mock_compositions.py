@@ -765,3 +765,22 @@ def test_mock_floordiv_mod(mlib):
         df["a"] // 0
     with pytest.raises(_HfErr):
         df["a"] // 2.5
+
+
+def test_mock_corr_cov(mlib):
+    rng = np.random.default_rng(28)
+    n = 4000
+    pdf = pandas.DataFrame({"x": rng.standard_normal(n),
+                            "y": rng.standard_normal(n),
+                            "w": rng.integers(-5, 5, n)})
+    pdf["y"] += 0.5 * pdf["x"]
+    pdf.loc[rng.random(n) < 0.1, "x"] = np.nan
+    pdf.loc[rng.random(n) < 0.1, "y"] = np.nan
+    df = mlib.DataFrame(pdf)
+    got, exp = df.corr(), pdf.corr()
+    assert list(got.columns) == list(exp.columns)
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(),
+                               rtol=1e-10, equal_nan=True)
+    got, exp = df.cov(), pdf.cov()
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(),
+                               rtol=1e-10, equal_nan=True)
