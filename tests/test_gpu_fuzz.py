@@ -69,7 +69,8 @@ def test_fuzz_pipeline(seed):
     steps = rng.integers(2, 5)
     for si in range(steps):
         op = rng.choice(["filter", "sort", "head", "dropna", "arith",
-                         "round", "where", "dedup", "iloc", "strmask"])
+                         "round", "where", "dedup", "iloc", "strmask",
+                         "assign", "mapdict"])
         msg = f"seed {seed} step {si} op {op}"
         if op == "filter":
             thr = float(np.round(rng.standard_normal() * 10, 2))
@@ -127,6 +128,25 @@ def test_fuzz_pipeline(seed):
             pdf = pdf[pdf["s"].str.contains(pat, na=False)]
             df = mpd.DataFrame(df.to_pandas().reset_index(drop=True))
             pdf = pdf.reset_index(drop=True)
+        elif op == "assign":
+            if not all(c in pdf.columns for c in ("v", "w")):
+                continue
+            c = float(np.round(rng.standard_normal() * 3, 3))
+            pdf = pdf.copy()
+            df["q"] = df["v"] * c + df["w"]
+            pdf["q"] = pdf["v"] * c + pdf["w"]
+            sc = int(rng.integers(-9, 9))
+            df["qc"] = sc
+            pdf["qc"] = sc
+        elif op == "mapdict":
+            if "w" not in pdf.columns or pdf["w"].dtype != np.int64:
+                continue
+            pdf = pdf.copy()
+            keys = rng.integers(-1000, 1000, 8)
+            rep = {int(a): int(b)
+                   for a, b in zip(keys, rng.integers(-5, 5, 8))}
+            df["w"] = df["w"].replace(rep)
+            pdf["w"] = pdf["w"].replace(rep)
         elif op == "dedup":
             subs = [["k"], ["k", "s"], ["s", "w"], None][rng.integers(0, 4)]
             if subs is not None:
