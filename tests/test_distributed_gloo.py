@@ -421,3 +421,101 @@ def test_gloo_concat_global_order(world):
         errs.append(q.get())
     assert not errs, "\n".join(errs)
     assert all(p.exitcode == 0 for p in procs)
+
+
+def _melt_worker(rank, world, port, fail_q):
+    """World>1 melt + column assignment through the REAL composition code
+    on the numpy mock: melt's concat must land in GLOBAL pandas melt
+    order (value column 0 over all ranks, then column 1, ...), with the
+    dictionary-encoded `variable` column surviving the exchange."""
+    try:
+        os.environ.update({
+            "RANK": str(rank), "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        })
+        import pandas
+        import modin_amd.distributed as dist_mod
+        from tests import mocklib
+
+        class _RawPatch:
+            def setattr(self, obj, name, fn):
+                setattr(obj, name, fn)
+
+        mocklib.install(_RawPatch())
+        assert dist_mod.init_from_env(backend="gloo", gpu=False)
+        import modin_amd.pandas as mpd
+        from modin_amd.core import lib
+        from modin_amd.core.dataframe import HipDataframe
+        from modin_amd.core.partition import (DeviceBlock,
+                                              HipDataframePartition)
+        from modin_amd.query_compiler import HipQueryCompiler
+
+        rng = np.random.default_rng(77)  # same stream on all ranks
+        n = 1200
+        gid = rng.integers(0, 9, n).astype(np.int64)
+        gx = rng.random(n)
+        gy = rng.random(n)
+        counts = oracle.split_row_counts(n, world, 1)
+        offs = np.cumsum([0] + counts)
+        sl = slice(offs[rank], offs[rank + 1])
+        nl = counts[rank]
+        block = DeviceBlock({"id": lib.put(gid[sl]), "x": lib.put(gx[sl]),
+                             "y": lib.put(gy[sl])}, nl)
+        frame = HipDataframe(
+            [HipDataframePartition(block)],
+            pandas.RangeIndex(offs[rank], offs[rank + 1]),
+            ["id", "x", "y"], [nl],
+            pandas.Series({"id": np.dtype(np.int64),
+                           "x": np.dtype(np.float64),
+                           "y": np.dtype(np.float64)}))
+        df = mpd.DataFrame(query_compiler=HipQueryCompiler(frame))
+
+        # column assignment stays rank-local (co-sharded operands)
+        df["z"] = df["x"] - df["y"]
+        got_z = np.concatenate(dist_mod._gather_np_varlen(
+            lib.get(df._query_compiler._modin_frame._partitions[0]
+                    .block().columns["z"])))
+        np.testing.assert_allclose(got_z, gx - gy, rtol=0)
+
+        out = df[["id", "x", "y"]].melt(id_vars="id")._query_compiler\
+            ._modin_frame
+        blk = out._partitions[0].block()
+        got_id = np.concatenate(dist_mod._gather_np_varlen(
+            lib.get(blk.columns["id"])))
+        got_var = np.concatenate(dist_mod._gather_np_varlen(
+            lib.get(blk.columns["variable"])))
+        got_val = np.concatenate(dist_mod._gather_np_varlen(
+            lib.get(blk.columns["value"])))
+        exp = pandas.DataFrame({"id": gid, "x": gx, "y": gy}).melt(
+            id_vars="id")
+        np.testing.assert_array_equal(got_id, exp["id"].to_numpy())
+        cats = blk.cats["variable"]
+        np.testing.assert_array_equal(
+            cats.to_numpy(dtype=object)[got_var],
+            exp["variable"].to_numpy())
+        np.testing.assert_allclose(got_val, exp["value"].to_numpy(),
+                                   rtol=0)
+        dist_mod.shutdown()
+    except Exception:  # noqa: BLE001
+        import traceback
+        fail_q.put(f"rank {rank}:\n{traceback.format_exc()}")
+        raise SystemExit(1)
+
+
+@pytest.mark.parametrize("world", [2])
+def test_gloo_melt_setitem_global_order(world):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29555
+    procs = [ctx.Process(target=_melt_worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=110)
+    errs = []
+    while not q.empty():
+        errs.append(q.get())
+    assert not errs, "\n".join(errs)
+    assert all(p.exitcode == 0 for p in procs)
