@@ -52,15 +52,60 @@ def read_csv(path, columns=None, **csv_kwargs):
 
 
 def concat(objs, ignore_index: bool = False):
-    """pandas.concat(axis=0) over DataFrames with identical columns."""
+    """pandas.concat(axis=0).  Mismatched columns align with NaN fills
+    (pandas outer-join rule: first frame's columns, then new names in
+    appearance order; an int64 column missing anywhere promotes to
+    float64 — the pandas dtype rule).  Missing datetime columns raise
+    (NaT is a later round)."""
     objs = list(objs)
     if not objs:
         raise HfErrorProxy("concat of empty list")
-    out = DataFrame(query_compiler=objs[0]._query_compiler.concat(
-        [o._query_compiler for o in objs[1:]]))
+    union = []
+    for o in objs:
+        for c in o.columns:
+            if c not in union:
+                union.append(c)
+    if any(list(o.columns) != union for o in objs):
+        dts = {}
+        for o in objs:
+            for c in o.columns:
+                dts.setdefault(c, o.dtypes[c])
+        aligned = []
+        for o in objs:
+            qc = o._query_compiler
+            have = set(o.columns)
+            for c in union:
+                if c in have:
+                    continue
+                dt = dts[c]
+                if isinstance(dt, np.dtype) and np.issubdtype(
+                        dt, np.datetime64):
+                    raise HfErrorProxy(
+                        f"concat: datetime column {c!r} missing in one "
+                        "frame (NaT fill is a later round)")
+                if dt == np.dtype(object):
+                    qc = qc.write_column(
+                        c, HipQueryCompiler.from_pandas(
+                            pandas.DataFrame(
+                                {c: pandas.Series(
+                                    [None] * len(o), dtype=object)})))
+                else:
+                    qc = qc.write_scalar_column(c, np.nan)
+            # a NaN-filled int64 column promotes EVERY frame's copy
+            for c in union:
+                if (c in have and dts[c] == np.dtype(np.int64)
+                        and any(c not in set(x.columns) for x in objs)):
+                    qc = qc.write_column(
+                        c, qc.getitem_column_array([c]).astype(
+                            np.float64))
+            aligned.append(qc.getitem_column_array(union))
+        out = DataFrame(query_compiler=aligned[0].concat(aligned[1:]))
+    else:
+        out = DataFrame(query_compiler=objs[0]._query_compiler.concat(
+            [o._query_compiler for o in objs[1:]]))
     if ignore_index:
-        out._query_compiler._modin_frame._index = __import__(
-            "pandas").RangeIndex(len(out))
+        out._query_compiler._modin_frame._index = pandas.RangeIndex(
+            len(out))
     return out
 
 

@@ -2643,3 +2643,33 @@ def test_corr_cov_vs_pandas(npartitions):
     np.testing.assert_allclose(df.cov().to_numpy(),
                                pdf.cov().to_numpy(), rtol=1e-9,
                                equal_nan=True)
+
+
+def test_concat_column_alignment_vs_pandas(npartitions):
+    """concat with mismatched columns: NaN fills, appearance-order
+    union, int->float promotion (pandas outer-align rules)."""
+    rng = np.random.default_rng(130)
+    p1 = pandas.DataFrame({"a": rng.integers(0, 90, 20_000),
+                           "v": rng.standard_normal(20_000),
+                           "s": rng.choice(["x", "y", None], 20_000)})
+    p2 = pandas.DataFrame({"a": rng.integers(0, 90, 9_000),
+                           "w": rng.integers(-50, 50, 9_000)})
+    got = mpd.concat([mpd.DataFrame(p1), mpd.DataFrame(p2)],
+                     ignore_index=True).to_pandas()
+    exp = pandas.concat([p1, p2], ignore_index=True)
+    assert list(got.columns) == list(exp.columns)
+    assert list(got.dtypes) == list(exp.dtypes)
+    for c in exp.columns:
+        g, e = got[c].to_numpy(), exp[c].to_numpy()
+        if e.dtype == object:
+            same = (pandas.isna(g) & pandas.isna(e)) | (g == e)
+            assert same.all(), c
+        else:
+            np.testing.assert_allclose(g.astype(float), e.astype(float),
+                                       rtol=0, equal_nan=True, err_msg=c)
+    # downstream groupby over the aligned result still runs on device
+    r = got.groupby("a")["w"].count()
+    gdf = mpd.concat([mpd.DataFrame(p1), mpd.DataFrame(p2)],
+                     ignore_index=True)
+    rg = gdf.groupby("a")["w"].count().to_pandas()
+    np.testing.assert_array_equal(rg.to_numpy(), r.to_numpy())

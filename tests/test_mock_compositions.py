@@ -784,3 +784,37 @@ def test_mock_corr_cov(mlib):
     got, exp = df.cov(), pdf.cov()
     np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(),
                                rtol=1e-10, equal_nan=True)
+
+
+def test_mock_concat_column_alignment(mlib):
+    """concat over frames with DIFFERENT columns: NaN-fill alignment,
+    appearance-order column union, int64 -> float64 promotion where a
+    column is missing anywhere (pandas rules)."""
+    rng = np.random.default_rng(29)
+    p1 = pandas.DataFrame({"a": rng.integers(0, 9, 500),
+                           "v": rng.random(500),
+                           "s": rng.choice(["x", "y"], 500)})
+    p2 = pandas.DataFrame({"a": rng.integers(0, 9, 300),
+                           "w": rng.integers(-5, 5, 300)})
+    p3 = pandas.DataFrame({"w": rng.integers(-5, 5, 200),
+                           "s": rng.choice(["y", "z"], 200),
+                           "a": rng.integers(0, 9, 200)})
+    got = mlib.concat([mlib.DataFrame(p) for p in (p1, p2, p3)],
+                      ignore_index=True).to_pandas()
+    exp = pandas.concat([p1, p2, p3], ignore_index=True)
+    assert list(got.columns) == list(exp.columns)
+    assert list(got.dtypes) == list(exp.dtypes)
+    for c in exp.columns:
+        g, e = got[c].to_numpy(), exp[c].to_numpy()
+        if e.dtype == object:
+            same = (pandas.isna(g) & pandas.isna(e)) | (g == e)
+            assert same.all(), c
+        else:
+            np.testing.assert_allclose(g.astype(float), e.astype(float),
+                                       rtol=0, equal_nan=True, err_msg=c)
+    # identical columns keep the fast path + int dtype
+    got2 = mlib.concat([mlib.DataFrame(p2), mlib.DataFrame(p2)],
+                       ignore_index=True).to_pandas()
+    exp2 = pandas.concat([p2, p2], ignore_index=True)
+    assert list(got2.dtypes) == list(exp2.dtypes)
+    np.testing.assert_array_equal(got2.to_numpy(), exp2.to_numpy())
