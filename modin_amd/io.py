@@ -134,12 +134,10 @@ def read_csv(path, columns=None, **csv_kwargs):
     return HipQueryCompiler(frame)
 
 
-def write_parquet(qc, path):
-    """Device columns -> pyarrow table -> parquet.  Dictionary columns
-    rebuild as pyarrow DictionaryArrays straight from the codes (−1 ->
-    null); numeric columns transfer as numpy."""
+def _qc_to_arrow(qc):
+    """Device columns -> pyarrow table (dictionary columns rebuild as
+    DictionaryArrays straight from the codes, −1 -> null)."""
     import pyarrow as pa
-    import pyarrow.parquet as pq
 
     frame = qc._modin_frame
     arrays, names = [], []
@@ -156,7 +154,27 @@ def write_parquet(qc, path):
             arr = pa.DictionaryArray.from_arrays(
                 idx, pa.array(cats.to_numpy(dtype=object).tolist()))
         else:
+            dt = dict(frame.dtypes).get(name)
+            if (isinstance(dt, np.dtype)
+                    and np.issubdtype(dt, np.datetime64)):
+                merged = merged.view("datetime64[ns]")
             arr = pa.array(merged)
         arrays.append(arr)
         names.append(name)
-    pq.write_table(pa.table(dict(zip(names, arrays))), path)
+    return pa.table(dict(zip(names, arrays)))
+
+
+def write_csv(qc, path):
+    """Device columns -> pyarrow -> CSV via the multithreaded C++ writer
+    (symmetric with read_csv; dictionary columns decode to strings in the
+    writer, never per-row in Python)."""
+    import pyarrow.csv as pacsv
+
+    pacsv.write_csv(_qc_to_arrow(qc), path)
+
+
+def write_parquet(qc, path):
+    """Device columns -> pyarrow table -> parquet (via _qc_to_arrow)."""
+    import pyarrow.parquet as pq
+
+    pq.write_table(_qc_to_arrow(qc), path)

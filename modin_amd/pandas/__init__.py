@@ -611,6 +611,16 @@ class DataFrame(_HipPandasBase):
                                                                total))
 
     def astype(self, dtype):
+        """Scalar dtype (all columns) or {column: dtype} (per-column
+        device casts through write_column)."""
+        if isinstance(dtype, dict):
+            qc = self._query_compiler
+            for c, dt in dtype.items():
+                if c not in list(self.columns):
+                    raise lib.HfError(f"astype: unknown column {c!r}")
+                qc = qc.write_column(
+                    c, qc.getitem_column_array([c]).astype(dt))
+            return DataFrame(query_compiler=qc)
         return DataFrame(query_compiler=self._query_compiler.astype(dtype))
 
     def quantile(self, q=0.5):
@@ -664,6 +674,12 @@ class DataFrame(_HipPandasBase):
         Python objects either direction)."""
         from ..io import write_parquet
         write_parquet(self._query_compiler, path)
+
+    def to_csv(self, path):
+        """Columnar CSV write through pyarrow.csv (the C++ writer), same
+        device->arrow path as to_parquet."""
+        from ..io import write_csv
+        write_csv(self._query_compiler, path)
 
     def rename(self, columns: dict):
         qc = self._query_compiler

@@ -469,3 +469,37 @@ def test_read_csv_vs_pandas(tmp_path, npartitions):
     g2 = mpd.read_csv(path).groupby("k").count().to_pandas()
     e2 = exp.groupby("k").count()
     np.testing.assert_array_equal(g2["v"].to_numpy(), e2["v"].to_numpy())
+
+
+def test_to_csv_roundtrip(tmp_path, npartitions):
+    """to_csv -> read_csv round trip (pyarrow C++ writer/reader), incl.
+    strings with NaN and datetimes."""
+    rng = np.random.default_rng(131)
+    n = 5000
+    pdf = pandas.DataFrame({
+        "a": rng.integers(-100, 100, n),
+        "v": np.round(rng.standard_normal(n), 6),
+        "s": rng.choice(["aa", "bb", None], n)})
+    df = mpd.DataFrame(pdf)
+    p = str(tmp_path / "t.csv")
+    df.to_csv(p)
+    back = mpd.read_csv(p).to_pandas()
+    np.testing.assert_array_equal(back["a"].to_numpy(),
+                                  pdf["a"].to_numpy())
+    np.testing.assert_allclose(back["v"].to_numpy(), pdf["v"].to_numpy(),
+                               rtol=0, atol=1e-12)
+    g, e = back["s"].to_numpy(), pdf["s"].to_numpy()
+    same = (pandas.isna(g) & pandas.isna(e)) | (g == e)
+    assert same.all()
+
+
+def test_astype_dict(npartitions):
+    rng = np.random.default_rng(132)
+    pdf = pandas.DataFrame({"a": rng.integers(0, 50, 3000),
+                            "v": np.round(rng.standard_normal(3000), 3),
+                            "w": rng.integers(-9, 9, 3000)})
+    df = mpd.DataFrame(pdf)
+    got = df.astype({"a": np.float64, "v": np.int64}).to_pandas()
+    exp = pdf.astype({"a": np.float64, "v": np.int64})
+    assert list(got.dtypes) == list(exp.dtypes)
+    np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(), rtol=0)
