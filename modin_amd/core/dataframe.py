@@ -108,6 +108,9 @@ def union_cats(a, b):
         return u
 
 
+INAT = np.iinfo(np.int64).min  # pandas' own NaT bit pattern in the ns view
+
+
 class HipDataframe:
     _partition_mgr_cls = HipDataframePartitionManager
 
@@ -117,6 +120,34 @@ class HipDataframe:
         self.columns = pandas.Index(columns)
         self._row_lengths = row_lengths
         self.dtypes = dtypes                   # pandas.Series name -> np.dtype
+
+    # ---- datetime/NaT helpers ----
+    def _dt_cols(self) -> set:
+        """Names of datetime64[ns]-tagged columns (int64 ns on device)."""
+        return {c for c in self.columns
+                if isinstance(self.dtypes[c], np.dtype)
+                and np.issubdtype(self.dtypes[c], np.datetime64)}
+
+    def _col_has_nat(self, name: str) -> bool:
+        """True if the tagged column holds any NaT (iNaT = INT64_MIN).
+        One cached device reduce per partition column — O(1) after the
+        first call on the same buffers."""
+        for p in self._partitions:
+            col = p.block().columns[name]
+            if col.length and lib.reduce(col).imn == INAT:
+                return True
+        return False
+
+    def _guard_nat(self, ctx: str, cols=None) -> None:
+        """Loud gate for ops whose kernels would treat iNaT as a huge
+        negative int (DESIGN.md NaT scope): raises if any datetime
+        column in `cols` (default all) holds NaT."""
+        dtc = self._dt_cols()
+        for c in (cols if cols is not None else self.columns):
+            if c in dtc and self._col_has_nat(c):
+                raise lib.HfError(
+                    f"{ctx}: column {c!r} holds NaT — this op does not "
+                    "support NaT yet (dropna() it first)")
 
     # ---- metadata ----
     @property
@@ -154,11 +185,9 @@ class HipDataframe:
                     "later round (convert with tz_localize(None))")
             if isinstance(dt, np.dtype) and np.issubdtype(dt, np.datetime64):
                 # typed-column tag (SURVEY §8f.3 gateway): the device
-                # stores the int64 ns view; every sort/groupby/merge/
+                # stores the int64 ns view (NaT = iNaT = INT64_MIN, the
+                # same bits pandas uses); every sort/groupby/merge/
                 # compare path runs on int64, to_pandas restores the tag
-                if df[name].isna().any():
-                    raise lib.HfError(
-                        f"column {name!r}: NaT values are a later round")
                 if edf is df:
                     edf = df.copy()
                 edf[name] = df[name].to_numpy().astype(
@@ -783,6 +812,7 @@ class HipDataframe:
         value_counts/nunique drop it)."""
         blk_cats = (self._partitions[0].block().cats
                     if self._partitions else {})
+        self._guard_nat("unique/value_counts/nunique", [name])
         POS = "\x00pos\x00"
         parts2 = []
         base = 0
@@ -1240,21 +1270,29 @@ class HipDataframe:
             cols = [p.block().columns[name] for p in self._partitions]
             return cols[0] if len(cols) == 1 else lib.concat(cols)
 
+        dtc = self._dt_cols()
         out_cols, dtypes = {}, {}
         for name in self.columns:
             col = concat_col(name)
-            if col.dtype_code == lib.HF_INT64:
-                col = lib.cast_f64(col)
-            nanb = lib.alloc(k, lib.HF_FLOAT64)
-            if k:
-                lib.fill_f64(nanb.dptr(), float("nan"), k)
+            if name in dtc:
+                # pandas shift on datetime keeps the dtype, filling NaT
+                nanb = lib.alloc(k, lib.HF_INT64)
+                if k:
+                    lib.fill_i64(nanb.dptr(), INAT, k)
+                dtypes[name] = self.dtypes[name]
+            else:
+                if col.dtype_code == lib.HF_INT64:
+                    col = lib.cast_f64(col)
+                nanb = lib.alloc(k, lib.HF_FLOAT64)
+                if k:
+                    lib.fill_f64(nanb.dptr(), float("nan"), k)
+                dtypes[name] = np.dtype(np.float64)
             if periods > 0:
                 kept = lib.col_slice(col, 0, n - k)
                 out_cols[name] = lib.concat([nanb, kept])
             else:
                 kept = lib.col_slice(col, k, n - k)
                 out_cols[name] = lib.concat([kept, nanb])
-            dtypes[name] = np.dtype(np.float64)
         part = HipDataframePartition(DeviceBlock(out_cols, n))
         return HipDataframe([part], self._index, self.columns, [n],
                             pandas.Series(dtypes))
@@ -1266,6 +1304,9 @@ class HipDataframe:
                     if self._partitions else {})
         if blk_cats:
             raise lib.HfError("diff over string columns is a later round")
+        if self._dt_cols():
+            raise lib.HfError("diff over datetime columns yields "
+                              "timedelta64 — a later round")
         shifted = self.shift_rows(periods) if periods else None
         n = len(self)
         out_cols = {}
@@ -1290,6 +1331,13 @@ class HipDataframe:
                     if self._partitions else {})
         if blk_cats:
             raise lib.HfError("cumulative ops over string columns")
+        dtc = self._dt_cols()
+        if dtc:
+            if agg_op not in (lib.AGG_MIN, lib.AGG_MAX):
+                raise lib.HfError(
+                    f"cumsum/cumprod over datetime columns {sorted(dtc)} "
+                    "(pandas raises too — cummin/cummax are supported)")
+            self._guard_nat("cummin/cummax", dtc)
         n = len(self)
         out_cols, dtypes = {}, {}
         for name in self.columns:
@@ -1920,6 +1968,9 @@ class HipDataframe:
                     if self._partitions else {})
         if blk_cats:
             raise lib.HfError("rolling over string columns")
+        if self._dt_cols():
+            raise lib.HfError("rolling over datetime columns (pandas is "
+                              "numeric-only here too)")
         if not isinstance(self._index, pandas.RangeIndex) or \
                 self._index.start != 0 or self._index.step != 1:
             raise lib.HfError("rolling: only RangeIndex frames this round")
@@ -2024,6 +2075,9 @@ class HipDataframe:
         minp = 1 if min_periods is None else int(min_periods)
         if minp < 0:
             raise lib.HfError("expanding: min_periods >= 0")
+        if self._dt_cols():
+            raise lib.HfError("expanding over datetime columns (pandas "
+                              "is numeric-only here too)")
         if op not in ("sum", "mean", "count", "min", "max"):
             raise lib.HfError(f"expanding.{op} not supported")
         if (self._partitions and self._partitions[0].block().cats):
@@ -2146,6 +2200,14 @@ class HipDataframe:
                     out = lib.binary(
                         lib.BIN_ADD,
                         lib.binary(lib.BIN_ADD, yoe, mulc(era, 400)), le2)
+        if n and lib.reduce(ns).imn == INAT:
+            # NaT rows -> NaN, float64 result (the pandas dtype rule when
+            # NaT is present)
+            notnat = lib.compare_scalar(lib.CMP_NE, ns, float(INAT))
+            out = lib.fixup_empty(lib.cast_f64(out), notnat)
+            part = HipDataframePartition(DeviceBlock({name: out}, n))
+            return HipDataframe([part], self._index, [name], [n],
+                                pandas.Series({name: np.dtype(np.float64)}))
         part = HipDataframePartition(DeviceBlock({name: out}, n))
         # pandas dt fields are int32
         return HipDataframe([part], self._index, [name], [n],
@@ -2500,6 +2562,10 @@ class HipDataframe:
             b = p.block()
             mcols.append(b.columns[list(b.columns)[0]])
         m = mcols[0] if len(mcols) == 1 else lib.concat(mcols)
+        if self._dt_cols():
+            raise lib.HfError("where/mask over datetime columns needs NaT "
+                              "fills — a later round (select other "
+                              "columns)")
         blk_cats = (self._partitions[0].block().cats
                     if self._partitions else {})
 
@@ -3056,6 +3122,14 @@ class HipDataframe:
             raise lib.HfError(
                 f"merge how={how!r} not implemented (inner/left/outer/"
                 "right this round)")
+        fill_dtc = (other._dt_cols() if how == "left"
+                    else (self._dt_cols() | other._dt_cols()
+                          if how == "outer" else set()))
+        if fill_dtc - {on} or (how == "outer" and on in fill_dtc):
+            raise lib.HfError(
+                "merge: unmatched rows would need NaT fills in datetime "
+                f"column(s) {sorted(fill_dtc)} — a later round (inner "
+                "joins support datetime)")
         left_names = [c for c in self.columns if c != on]
         right_names = [c for c in other.columns if c != on]
         common = set(left_names) & set(right_names)
@@ -3385,6 +3459,8 @@ class HipDataframe:
 
     # ---- comparison map (mask column) ----
     def compare_scalar(self, op_code: int, scalar) -> "HipDataframe":
+        dtc = self._dt_cols()
+
         def block_fn(block: DeviceBlock) -> DeviceBlock:
             out = {}
             for name, col in block.columns.items():
@@ -3396,8 +3472,18 @@ class HipDataframe:
                         raise lib.HfError(
                             f"comparing numeric column {name!r} to a "
                             "string scalar")
-                    out[name] = lib.compare_scalar(op_code, col,
-                                                   float(scalar))
+                    m = lib.compare_scalar(op_code, col, float(scalar))
+                    if name in dtc:
+                        # pandas NaT semantics: every ordered compare is
+                        # False on NaT rows (NE stays True — IEEE-NaN
+                        # style); NOTNA is exactly (!= iNaT)
+                        notnat = lib.compare_scalar(lib.CMP_NE, col,
+                                                    float(INAT))
+                        if op_code == lib.CMP_NOTNA:
+                            m = notnat
+                        elif op_code != lib.CMP_NE:
+                            m = lib.binary(lib.BIN_MUL, m, notnat)
+                    out[name] = m
             return DeviceBlock(out, block.length)
         return self.map(block_fn)
 
@@ -3470,7 +3556,9 @@ class HipDataframe:
             return out
         if any(isinstance(d, np.dtype) and np.issubdtype(d, np.datetime64)
                for d in self.dtypes) and dt == np.dtype(np.int64):
-            # datetime -> int64: drop the tag (device already int64 ns)
+            # datetime -> int64: drop the tag (device already int64 ns);
+            # pandas raises on NaT -> int64 and so do we
+            self._guard_nat("astype(int64)", self._dt_cols())
             out = HipDataframe(self._partitions, self._index, self.columns,
                                self._row_lengths,
                                pandas.Series({c: np.dtype(np.int64)
@@ -3603,7 +3691,8 @@ class HipDataframe:
     # ---- sort (PandasDataframe.sort_by device form, dataframe.py:2742;
     #      SURVEY §8f.2): stable radix permutation + column gathers ----
     @staticmethod
-    def _effective_sort_key(col, is_dict, ascending, na_first=False):
+    def _effective_sort_key(col, is_dict, ascending, na_first=False,
+                            is_dt=False):
         """(key col, ascending) -> (int64 key, ascending') whose stable
         ASCENDING' radix sort realizes pandas order.  NaNs — dictionary
         code −1 or float NaN — sort LAST for both directions by default
@@ -3631,6 +3720,20 @@ class HipDataframe:
             return lib.binary(lib.BIN_ADD,
                               lib.binary(lib.BIN_MUL, base, notna),
                               isna_big), True
+        if is_dt and col.length and lib.reduce(col).imn == INAT:
+            # datetime key with NaT: same sentinel scheme as float NaN —
+            # NaT rows jump above (na last) / below (na first) every
+            # valid ns value, valid keys keep or negate their order
+            NANKEY = ((1 << 63) - 1) * (-1 if na_first else 1)
+            notnat = lib.compare_scalar(lib.CMP_NE, col, float(INAT))
+            nat_big = lib.map_scalar(
+                lib.MAP_MUL, lib.map_scalar(lib.MAP_RSUB, notnat, 1),
+                NANKEY)
+            base = col if ascending else lib.map_scalar(lib.MAP_NEG,
+                                                        col, 0)
+            return lib.binary(lib.BIN_ADD,
+                              lib.binary(lib.BIN_MUL, base, notnat),
+                              nat_big), True
         if not is_dict or not col.length or lib.reduce(col).imn >= 0:
             return col, ascending
         if ascending and na_first:
@@ -3698,8 +3801,9 @@ class HipDataframe:
                 cache[name] = concat_col(name)
             return cache[name]
 
+        dtc = self._dt_cols()
         eff = [self._effective_sort_key(cat_col(b), b in blk_cats, a,
-                                        na_first)
+                                        na_first, is_dt=b in dtc)
                for b, a in zip(by_list, asc_list)]
         perm = self._compose_sort_perm(eff)
         cols = {name: lib.gather(cat_col(name), perm)
@@ -3724,9 +3828,11 @@ class HipDataframe:
         import numpy as np
         from .. import distributed as dist_mod
         P = dist_mod.world_size()
+        dtc = self._dt_cols()
         k0 = concat_col(by_list[0])
         ek0, ea0 = self._effective_sort_key(k0, by_list[0] in blk_cats,
-                                            asc_list[0], na_first)
+                                            asc_list[0], na_first,
+                                            is_dt=by_list[0] in dtc)
         n = ek0.length
         S = min(n, 4096)
         if S:
@@ -3755,7 +3861,7 @@ class HipDataframe:
                 for m in names}
         rpos = dist_mod.exchange_column(lib.concat(send_pos), send_counts)
         eff = [self._effective_sort_key(recv[b], b in blk_cats, a,
-                                        na_first)
+                                        na_first, is_dt=b in dtc)
                for b, a in zip(by_list, asc_list)]
         perm = self._compose_sort_perm(eff)
         out_cols = {m: lib.gather(recv[m], perm) for m in names}
@@ -3767,11 +3873,15 @@ class HipDataframe:
 
     # ---- dropna mask: AND of per-column notna (pandas dropna(how="any")) ----
     def notna_all_mask(self) -> "HipDataframe":
+        dtc = self._dt_cols()
+
         def block_fn(block: DeviceBlock) -> DeviceBlock:
             acc = None
             for name, col in block.columns.items():
                 if name in block.cats:  # dict-encoded: NaN is code −1
                     m = lib.compare_scalar(lib.CMP_NE, col, -1.0)
+                elif name in dtc:  # datetime: NaT is iNaT
+                    m = lib.compare_scalar(lib.CMP_NE, col, float(INAT))
                 else:
                     m = lib.compare_scalar(lib.CMP_NOTNA, col, 0.0)
                 acc = m if acc is None else lib.binary(lib.BIN_MUL, acc, m)

@@ -332,7 +332,18 @@ class _HipPandasBase:
         return self.__truediv__(other)
 
     def fillna(self, value):
-        """pandas fillna: scalar, or {column: scalar} dict."""
+        """pandas fillna: scalar, {column: scalar} dict, or a Timestamp
+        for datetime Series (NaT -> value, exact int64 ns replace)."""
+        import datetime as _dtm
+        if isinstance(value, (pandas.Timestamp, np.datetime64,
+                              _dtm.datetime)):
+            qc = self._query_compiler
+            if len(qc.columns) != 1:
+                raise lib.HfError("fillna(Timestamp): Series only")
+            from ..core.dataframe import INAT
+            return self._rewrap(qc.map_dict(
+                {int(INAT): int(pandas.Timestamp(value).value)},
+                keep_missing=True))
         if isinstance(value, dict):
             return self._rewrap(
                 self._query_compiler.fillna_dict(
@@ -1157,9 +1168,33 @@ class DataFrameGroupBy:
                                 series_out=isinstance(key, str),
                                 dropna=self._dropna)
 
+    def _nat_handled_df(self) -> "DataFrame":
+        """NaT groupby keys: dropna=True drops the NaT rows (a device
+        filter — pandas' NaN-group rule applied to the iNaT key);
+        dropna=False with NaT keys is loud.  Datetime VALUE columns with
+        NaT are loud (the int64 agg kernels would treat iNaT as a huge
+        negative ns value)."""
+        bys = (list(self._by) if isinstance(self._by, (list, tuple))
+               else [self._by])
+        qc = self._df._query_compiler
+        frame = qc._modin_frame
+        dtc = frame._dt_cols()
+        frame._guard_nat("groupby values",
+                         [c for c in frame.columns
+                          if c in dtc and c not in bys])
+        natkeys = [b for b in bys
+                   if b in dtc and frame._col_has_nat(b)]
+        if not natkeys:
+            return self._df
+        if not self._dropna:
+            raise lib.HfError("groupby(dropna=False) with NaT keys is a "
+                              "later round")
+        mask = qc.getitem_column_array(natkeys).dropna_mask()
+        return DataFrame(query_compiler=qc.getitem_array(mask))
+
     def _agg(self, how: str) -> DataFrame:
-        qc = self._df._query_compiler.groupby_agg(self._by, how,
-                                                  dropna=self._dropna)
+        qc = self._nat_handled_df()._query_compiler.groupby_agg(
+            self._by, how, dropna=self._dropna)
         out = DataFrame(query_compiler=qc)
         if self._series_out and self._as_index:
             name = list(qc._modin_frame.columns)[0]
@@ -1188,7 +1223,7 @@ class DataFrameGroupBy:
     def _tail_qc(self, fn_name: str, **kw):
         """Route the non-reduce aggs through the query compiler with the
         dropna flag (dropna=False rides the sentinel-NaN-key encoding)."""
-        return self._df._query_compiler.groupby_tail_agg(
+        return self._nat_handled_df()._query_compiler.groupby_tail_agg(
             self._by, fn_name, dropna=self._dropna, **kw)
 
     def var(self, ddof: int = 1):
@@ -1235,6 +1270,10 @@ class DataFrameGroupBy:
         """Same-length transforms in original row order (pandas
         DataFrameGroupBy.cumsum/cummin/cummax/cumcount/rank).  as_index is
         irrelevant (pandas keeps the caller's index for transforms)."""
+        bys = (list(self._by) if isinstance(self._by, (list, tuple))
+               else [self._by])
+        frame = self._df._query_compiler._modin_frame
+        frame._guard_nat("groupby transform", bys)
         qc = self._df._query_compiler.groupby_transform(
             self._by, how, dropna=self._dropna, **kw)
         if self._series_out or how in ("cumcount", "ngroup"):
@@ -1342,6 +1381,7 @@ class DataFrameGroupBy:
             return self._agg(how)
         if isinstance(how, dict):
             qcs = []
+            base = self._nat_handled_df()
             for col, a in how.items():
                 if not isinstance(a, str):
                     raise lib.HfError(
@@ -1349,7 +1389,7 @@ class DataFrameGroupBy:
                         "this round")
                 bys = (list(self._by) if isinstance(self._by, (list, tuple))
                        else [self._by])
-                sub = self._df[[*bys, col]]
+                sub = base[[*bys, col]]
                 qcs.append(sub._query_compiler.groupby_agg(
                     self._by, a, dropna=self._dropna))
             return DataFrame(query_compiler=qcs[0].hconcat(qcs[1:]))
@@ -1358,12 +1398,13 @@ class DataFrameGroupBy:
                    else [self._by])
             val_cols = [c for c in self._df.columns if c not in bys]
             qcs = []
+            base = self._nat_handled_df()
             for col in val_cols:  # pandas order: per column, per agg
                 for a in how:
                     if not isinstance(a, str):
                         raise lib.HfError("groupby.agg list entries must "
                                           "be agg names")
-                    sub = self._df[[*bys, col]]
+                    sub = base[[*bys, col]]
                     qc = sub._query_compiler.groupby_agg(
                         self._by, a, dropna=self._dropna)
                     qcs.append(qc.rename_columns({col: (col, a)}))

@@ -450,7 +450,10 @@ class HipQueryCompiler:
             res = HipDataframe([HipDataframePartition(blk)], frame._index,
                                [name], [n], dts)
             return self.__constructor__(res)
-        if frame.dtypes[name] != np.dtype(np.int64):
+        src_dt = frame.dtypes[name]
+        src_is_dt = (isinstance(src_dt, np.dtype)
+                     and np.issubdtype(src_dt, np.datetime64))
+        if src_dt != np.dtype(np.int64) and not src_is_dt:
             raise lib.HfError("map/replace(dict) over float64 columns is "
                               "a later round (int64/string sources only)")
         if not all(isinstance(k, (int, np.integer)) for k in mapping):
@@ -466,6 +469,9 @@ class HipQueryCompiler:
         col = concat_col()
         pos = lib.search_sorted(col, lib.put(ks))
         shifted = lib.map_scalar(lib.MAP_ADD, pos, 1)
+        if src_is_dt and not (keep_missing and int_vals):
+            raise lib.HfError("map over datetime columns is a later "
+                              "round (replace with int ns values works)")
         if keep_missing:
             m = lib.compare_scalar(lib.CMP_GE, pos, 0)
             inv = lib.map_scalar(lib.MAP_RSUB, m, 1)
@@ -475,7 +481,7 @@ class HipQueryCompiler:
                 out = lib.binary(lib.BIN_ADD,
                                  lib.binary(lib.BIN_MUL, mapped, m),
                                  lib.binary(lib.BIN_MUL, col, inv))
-                dt = np.dtype(np.int64)
+                dt = src_dt if src_is_dt else np.dtype(np.int64)
             else:
                 lut = np.r_[np.float64(0.0),
                             np.array(vs, dtype=np.float64)]
@@ -635,6 +641,10 @@ class HipQueryCompiler:
     # ---- comparisons (query_compiler gt/lt/eq bindings) -> int64 0/1 mask
     def _compare(self, op_code, other):
         import datetime
+        if other is None or (isinstance(other, float) and np.isnan(other)) \
+                or other is pandas.NaT:
+            raise lib.HfError("comparisons against NaN/NaT: pandas "
+                              "returns all-False — compare explicitly")
         if isinstance(other, (pandas.Timestamp, np.datetime64,
                               datetime.datetime)):
             # datetime operand vs the int64-ns typed column

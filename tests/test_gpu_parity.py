@@ -2035,11 +2035,12 @@ def test_datetime64_typed_columns(npartitions):
     np.testing.assert_allclose(got["b"].to_numpy(), exp["b"].to_numpy(),
                                rtol=0)
 
-    # NaT and tz-aware are loud errors
-    bad = pandas.DataFrame({"t": pandas.to_datetime(
+    # NaT round-trips (iNaT bits); tz-aware stays a loud error
+    nat = pandas.DataFrame({"t": pandas.to_datetime(
         ["2020-01-01", None])})
-    with pytest.raises(lib.HfError, match="NaT"):
-        mpd.DataFrame(bad)
+    back = mpd.DataFrame(nat).to_pandas()
+    np.testing.assert_array_equal(back["t"].to_numpy(),
+                                  nat["t"].to_numpy())
 
 
 def test_cumprod_vs_pandas(npartitions):
@@ -2673,3 +2674,88 @@ def test_concat_column_alignment_vs_pandas(npartitions):
                      ignore_index=True)
     rg = gdf.groupby("a")["w"].count().to_pandas()
     np.testing.assert_array_equal(rg.to_numpy(), r.to_numpy())
+
+
+def test_nat_semantics_vs_pandas(npartitions):
+    """NaT on device (iNaT ns bits): round trip, masks, compares, dt
+    fields, sort na_position, groupby-key drop, shift fill,
+    fillna(Timestamp), inner merge NaT==NaT, loud guards."""
+    rng = np.random.default_rng(133)
+    n = 40_000
+    t = pandas.Series(pandas.to_datetime("2021-03-01")
+                      + pandas.to_timedelta(
+                          rng.integers(0, 10**6, n), unit="min"))
+    t[rng.random(n) < 0.15] = pandas.NaT
+    pdf = pandas.DataFrame({"t": t, "v": rng.standard_normal(n),
+                            "k": rng.integers(0, 70, n)})
+    df = mpd.DataFrame(pdf)
+    back = df.to_pandas()
+    np.testing.assert_array_equal(back["t"].to_numpy(),
+                                  pdf["t"].to_numpy())
+    np.testing.assert_array_equal(df["t"].notna().to_pandas().to_numpy(),
+                                  pdf["t"].notna().to_numpy())
+    got = df.dropna().to_pandas()
+    exp = pdf.dropna()
+    assert len(got) == len(exp)
+    np.testing.assert_array_equal(got["t"].to_numpy(),
+                                  exp["t"].to_numpy())
+    ts = pandas.Timestamp("2021-06-01")
+    for op in ("__gt__", "__ge__", "__lt__", "__le__", "__eq__",
+               "__ne__"):
+        g = getattr(df["t"], op)(ts).to_pandas().to_numpy()
+        e = getattr(pdf["t"], op)(ts).to_numpy()
+        np.testing.assert_array_equal(g.astype(bool), e, err_msg=op)
+    for f in ("year", "month", "day", "dayofweek", "hour", "minute",
+              "second"):
+        g = getattr(df["t"].dt, f).to_pandas()
+        e = getattr(pdf["t"].dt, f)
+        np.testing.assert_allclose(g.to_numpy().astype(float),
+                                   e.to_numpy().astype(float), rtol=0,
+                                   equal_nan=True, err_msg=f)
+    for asc in (True, False):
+        for nap in ("last", "first"):
+            g = df.sort_values("t", ascending=asc,
+                               na_position=nap).to_pandas()
+            e = pdf.sort_values("t", ascending=asc, na_position=nap,
+                                kind="stable")
+            np.testing.assert_array_equal(g["t"].to_numpy(),
+                                          e["t"].to_numpy(),
+                                          err_msg=f"{asc}/{nap}")
+            np.testing.assert_array_equal(np.asarray(g.index),
+                                          e.index.to_numpy())
+    g = df.groupby("t").sum().to_pandas()
+    e = pdf.groupby("t").sum()
+    assert len(g) == len(e)
+    assert g.index.dtype == e.index.dtype
+    np.testing.assert_array_equal(g.index.to_numpy(), e.index.to_numpy())
+    np.testing.assert_allclose(g["v"].to_numpy(), e["v"].to_numpy(),
+                               rtol=1e-12)
+    g = df.groupby("t").size()
+    e = pdf.groupby("t").size()
+    np.testing.assert_array_equal(np.asarray(g), e.to_numpy())
+    with pytest.raises(lib.HfError):
+        df.groupby("t", dropna=False).sum()
+    with pytest.raises(lib.HfError):
+        df.groupby("k").min()  # NaT in a VALUE column
+    g = df[["t"]].shift(-3).to_pandas()
+    e = pdf[["t"]].shift(-3)
+    assert g["t"].dtype == e["t"].dtype
+    np.testing.assert_array_equal(g["t"].to_numpy(), e["t"].to_numpy())
+    fv = pandas.Timestamp("1999-12-31 23:59:59.123456789")
+    g = df["t"].fillna(fv).to_pandas()
+    e = pdf["t"].fillna(fv)
+    assert g.dtype == e.dtype
+    np.testing.assert_array_equal(g.to_numpy(), e.to_numpy())
+    # inner merge on a NaT-bearing datetime key: NaT==NaT matches
+    rp = pandas.DataFrame({"t": pandas.concat(
+        [t.iloc[:300], pandas.Series([pandas.NaT])],
+        ignore_index=True), "b": rng.random(301)})
+    g = df.merge(mpd.DataFrame(rp), on="t", how="inner").to_pandas()
+    e = pdf.merge(rp, on="t", how="inner")
+    assert len(g) == len(e)
+    assert g["t"].dtype == e["t"].dtype
+    np.testing.assert_allclose(np.sort(g["b"].to_numpy()),
+                               np.sort(e["b"].to_numpy()), rtol=0)
+    # non-inner merges needing NaT fills are loud
+    with pytest.raises(lib.HfError):
+        df.merge(mpd.DataFrame(rp), on="t", how="left")

@@ -818,3 +818,86 @@ def test_mock_concat_column_alignment(mlib):
     exp2 = pandas.concat([p2, p2], ignore_index=True)
     assert list(got2.dtypes) == list(exp2.dtypes)
     np.testing.assert_array_equal(got2.to_numpy(), exp2.to_numpy())
+
+
+def test_mock_nat_semantics(mlib):
+    """NaT through the typed-column layer: iNaT bits on device; masks,
+    compares, dt fields, sort na_position, groupby-key drop, shift NaT
+    fill, fillna(Timestamp) — pandas semantics; unsupported ops loud."""
+    rng = np.random.default_rng(30)
+    n = 3000
+    t = pandas.Series(pandas.to_datetime("2021-03-01")
+                      + pandas.to_timedelta(
+                          rng.integers(0, 10**6, n), unit="min"))
+    t[rng.random(n) < 0.15] = pandas.NaT
+    pdf = pandas.DataFrame({"t": t, "v": rng.standard_normal(n),
+                            "k": rng.integers(0, 7, n)})
+    df = mlib.DataFrame(pdf)
+    # round trip
+    back = df.to_pandas()
+    np.testing.assert_array_equal(back["t"].to_numpy(),
+                                  pdf["t"].to_numpy())
+    # isna / notna / dropna
+    np.testing.assert_array_equal(
+        df["t"].isna().to_pandas().to_numpy(),
+        pdf["t"].isna().to_numpy())
+    got = df.dropna().to_pandas()
+    exp = pdf.dropna()
+    assert len(got) == len(exp)
+    np.testing.assert_array_equal(got["t"].to_numpy(),
+                                  exp["t"].to_numpy())
+    # ordered compares are False on NaT rows; NE is True
+    ts = pandas.Timestamp("2021-06-01")
+    for op in ("__gt__", "__le__", "__eq__", "__ne__"):
+        g = getattr(df["t"], op)(ts).to_pandas().to_numpy()
+        e = getattr(pdf["t"], op)(ts).to_numpy()
+        np.testing.assert_array_equal(g.astype(bool), e, err_msg=op)
+    # dt fields: NaT -> NaN, float64 (pandas rule)
+    for f in ("year", "month", "dayofweek", "hour"):
+        g = getattr(df["t"].dt, f).to_pandas()
+        e = getattr(pdf["t"].dt, f)
+        np.testing.assert_allclose(g.to_numpy().astype(float),
+                                   e.to_numpy().astype(float), rtol=0,
+                                   equal_nan=True, err_msg=f)
+    # sort: NaT last (default) and first, both directions
+    for asc in (True, False):
+        for nap in ("last", "first"):
+            g = df.sort_values("t", ascending=asc,
+                               na_position=nap).to_pandas()
+            e = pdf.sort_values("t", ascending=asc, na_position=nap,
+                                kind="stable")
+            np.testing.assert_array_equal(g["t"].to_numpy(),
+                                          e["t"].to_numpy(),
+                                          err_msg=f"{asc}/{nap}")
+            np.testing.assert_array_equal(np.asarray(g.index),
+                                          e.index.to_numpy())
+    # groupby by the datetime key: NaT group dropped (pandas dropna=True)
+    g = df.groupby("t").sum().to_pandas()
+    e = pdf.groupby("t").sum()
+    assert len(g) == len(e)
+    np.testing.assert_array_equal(g.index.to_numpy(), e.index.to_numpy())
+    np.testing.assert_allclose(g["v"].to_numpy(), e["v"].to_numpy(),
+                               rtol=1e-12)
+    with pytest.raises(_HfErr, match="dropna"):
+        df.groupby("t", dropna=False).sum()
+    # datetime VALUE column with NaT under an agg is loud
+    with pytest.raises(_HfErr, match="NaT"):
+        df.groupby("k").min()
+    # shift keeps dtype, fills NaT
+    g = df[["t"]].shift(2).to_pandas()
+    e = pdf[["t"]].shift(2)
+    np.testing.assert_array_equal(g["t"].to_numpy(), e["t"].to_numpy())
+    # fillna(Timestamp): exact ns replace
+    fv = pandas.Timestamp("1999-12-31 23:59:59.123456789")
+    g = df["t"].fillna(fv).to_pandas()
+    e = pdf["t"].fillna(fv)
+    assert g.dtype == e.dtype
+    np.testing.assert_array_equal(g.to_numpy(), e.to_numpy())
+    # unsupported ops are loud, not wrong
+    for fn in (lambda: df[["t", "v"]].cumsum(),
+               lambda: df[["t"]].diff(),
+               lambda: df["t"].value_counts(),
+               lambda: df[["t", "v"]].where(df["v"] > 0),
+               lambda: df[["t"]].astype(np.int64)):
+        with pytest.raises(_HfErr):
+            fn()
