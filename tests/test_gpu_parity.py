@@ -2756,6 +2756,17 @@ def test_nat_semantics_vs_pandas(npartitions):
     assert g["t"].dtype == e["t"].dtype
     np.testing.assert_allclose(np.sort(g["b"].to_numpy()),
                                np.sort(e["b"].to_numpy()), rtol=0)
-    # non-inner merges needing NaT fills are loud
+    # left merge on the datetime KEY is fine (key never fills; right
+    # payload fills NaN)
+    g = df.merge(mpd.DataFrame(rp), on="t", how="left").to_pandas()
+    e = pdf.merge(rp, on="t", how="left")
+    assert len(g) == len(e)
+    assert g["t"].dtype == e["t"].dtype
+    np.testing.assert_allclose(
+        np.sort(g["b"].to_numpy()), np.sort(e["b"].to_numpy()),
+        rtol=0, equal_nan=True)
+    # ... but a datetime PAYLOAD column that would need NaT fills is loud
+    rp2 = rp.copy()
+    rp2["d2"] = pandas.Timestamp("2000-01-01")
     with pytest.raises(lib.HfError):
-        df.merge(mpd.DataFrame(rp), on="t", how="left")
+        df.merge(mpd.DataFrame(rp2), on="t", how="left")
