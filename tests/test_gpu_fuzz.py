@@ -34,6 +34,7 @@ def make_frame(rng, n):
     s[rng.random(n) < rng.random() * 0.1] = np.nan
     t = (pandas.Timestamp("2000-01-01").value
          + rng.integers(0, 10**18, n)).astype("datetime64[ns]")
+    t[rng.random(n) < rng.random() * 0.1] = np.datetime64("NaT")
     pdf = pandas.DataFrame({"k": k, "v": v, "w": w, "s": s, "t": t})
     return pdf
 
