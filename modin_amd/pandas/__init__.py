@@ -114,6 +114,17 @@ def merge(left: "DataFrame", right: "DataFrame", **kwargs) -> "DataFrame":
     return left.merge(right, **kwargs)
 
 
+def to_datetime(arg, format=None, errors: str = "raise"):  # noqa: A002
+    """pandas.to_datetime over a string Series: parses the HOST
+    DICTIONARY once per distinct value + one device gather (string NaN
+    -> NaT); datetime Series pass through."""
+    if not isinstance(arg, Series):
+        raise HfErrorProxy("to_datetime takes a Series")
+    return Series(
+        query_compiler=arg._query_compiler.to_datetime_from_strings(
+            format=format, errors=errors), name=arg.name)
+
+
 def from_pandas(df: pandas.DataFrame) -> "DataFrame":
     return DataFrame(query_compiler=HipQueryCompiler.from_pandas(df))
 

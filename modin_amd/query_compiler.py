@@ -509,6 +509,51 @@ class HipQueryCompiler:
                            [name], [n], pd.Series({name: dt}))
         return self.__constructor__(res)
 
+    def to_datetime_from_strings(self, format=None,
+                                 errors: str = "raise"
+                                 ) -> "HipQueryCompiler":
+        """pandas.to_datetime over a string (dictionary) column: the
+        parse runs ONCE PER CATEGORY on the host (pandas' own parser —
+        exact format semantics), then one device LUT gather maps every
+        row's code to its int64 ns; string NaN (code −1) -> NaT.  No
+        per-row Python object is ever created."""
+        from modin_amd.core.dataframe import HipDataframe, INAT
+        from modin_amd.core.partition import DeviceBlock, \
+            HipDataframePartition
+        import pandas as pd
+        frame = self._modin_frame
+        name = frame.columns[0]
+        dt = frame.dtypes[name]
+        if isinstance(dt, np.dtype) and np.issubdtype(dt, np.datetime64):
+            return self  # already datetime
+        blk_cats = (frame._partitions[0].block().cats
+                    if frame._partitions else {})
+        if name not in blk_cats:
+            raise lib.HfError("to_datetime: string (or datetime) Series "
+                              "only — int64 ns columns use "
+                              "astype('datetime64[ns]')")
+        cats = blk_cats[name]
+        parsed = pd.to_datetime(pd.Series(cats.to_numpy(dtype=object)),
+                                format=format, errors=errors)
+        lut = np.empty(len(cats) + 1, dtype=np.int64)
+        lut[0] = INAT
+        lut[1:] = parsed.to_numpy().astype("datetime64[ns]").view(np.int64)
+
+        def concat_col():
+            cs = [p.block().columns[name] for p in frame._partitions]
+            return cs[0] if len(cs) == 1 else lib.concat(cs)
+
+        codes = concat_col()
+        n = codes.length
+        shifted = lib.map_scalar(lib.MAP_ADD, codes, 1)
+        out = lib.gather(lib.put(lut), shifted)
+        blk = DeviceBlock({name: out}, n)
+        res = HipDataframe([HipDataframePartition(blk)], frame._index,
+                           [name], [n],
+                           pandas.Series({name: np.dtype(
+                               "datetime64[ns]")}))
+        return self.__constructor__(res)
+
     def dt_field(self, field: str) -> "HipQueryCompiler":
         return self.__constructor__(self._modin_frame.dt_field(field))
 

@@ -503,3 +503,33 @@ def test_astype_dict(npartitions):
     exp = pdf.astype({"a": np.float64, "v": np.int64})
     assert list(got.dtypes) == list(exp.dtypes)
     np.testing.assert_allclose(got.to_numpy(), exp.to_numpy(), rtol=0)
+
+
+def test_to_datetime_vs_pandas(npartitions):
+    """to_datetime: host-dictionary parse + one device LUT gather;
+    NaT for string NaN / coerced failures; downstream dt fields."""
+    rng = np.random.default_rng(134)
+    n = 30_000
+    days = rng.integers(0, 2000, 40)
+    cats = [(pandas.Timestamp("2019-01-01")
+             + pandas.Timedelta(days=int(d))).strftime("%Y-%m-%d")
+            for d in days] + [None]
+    sv = rng.choice(np.array(cats, dtype=object), n)
+    pdf = pandas.DataFrame({"s": sv})
+    df = mpd.DataFrame(pdf)
+    got = mpd.to_datetime(df["s"]).to_pandas()
+    exp = pandas.to_datetime(pdf["s"])
+    assert got.dtype == exp.dtype
+    np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy())
+    g = mpd.to_datetime(df["s"]).dt.dayofweek.to_pandas()
+    e = pandas.to_datetime(pdf["s"]).dt.dayofweek
+    np.testing.assert_allclose(g.to_numpy().astype(float),
+                               e.to_numpy().astype(float), rtol=0,
+                               equal_nan=True)
+    # sorting the parsed column: NaT last
+    d2 = mpd.DataFrame(query_compiler=mpd.to_datetime(
+        df["s"])._query_compiler)
+    gs = d2.sort_values("s").to_pandas()["s"]
+    es = pandas.DataFrame({"s": exp}).sort_values(
+        "s", kind="stable")["s"]
+    np.testing.assert_array_equal(gs.to_numpy(), es.to_numpy())

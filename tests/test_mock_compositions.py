@@ -901,3 +901,30 @@ def test_mock_nat_semantics(mlib):
                lambda: df[["t"]].astype(np.int64)):
         with pytest.raises(_HfErr):
             fn()
+
+
+def test_mock_to_datetime(mlib):
+    rng = np.random.default_rng(31)
+    dates = ["2021-01-05", "1999-12-31", "2024-02-29", None]
+    sv = rng.choice(np.array(dates, dtype=object), 2000)
+    pdf = pandas.DataFrame({"s": sv})
+    df = mlib.DataFrame(pdf)
+    got = mlib.to_datetime(df["s"]).to_pandas()
+    exp = pandas.to_datetime(pdf["s"])
+    assert got.dtype == exp.dtype
+    np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy())
+    # format + errors='coerce' (unparseable -> NaT)
+    pdf2 = pandas.DataFrame(
+        {"s": rng.choice(np.array(["05/01/2021", "31/12/1999", "oops"],
+                                  dtype=object), 800)})
+    got = mlib.to_datetime(mlib.DataFrame(pdf2)["s"],
+                           format="%d/%m/%Y", errors="coerce").to_pandas()
+    exp = pandas.to_datetime(pdf2["s"], format="%d/%m/%Y",
+                             errors="coerce")
+    np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy())
+    # downstream: dt fields after parsing (NaT -> NaN)
+    g = mlib.to_datetime(df["s"]).dt.year.to_pandas()
+    e = pandas.to_datetime(pdf["s"]).dt.year
+    np.testing.assert_allclose(g.to_numpy().astype(float),
+                               e.to_numpy().astype(float), rtol=0,
+                               equal_nan=True)
