@@ -1068,6 +1068,64 @@ class Series(_HipPandasBase):
             query_compiler=self._query_compiler.take_row_range(0, stop),
             name=self.name)
 
+    def tail(self, n: int = 5):
+        total = len(self._query_compiler)
+        start = min(total, -n) if n < 0 else total - n  # pandas tail(-n)
+        return Series(
+            query_compiler=self._query_compiler.take_row_range(start,
+                                                               total),
+            name=self.name)
+
+    def to_frame(self, name=None) -> "DataFrame":
+        """pandas Series.to_frame: 1-column DataFrame (zero copy)."""
+        qc = self._query_compiler
+        cur = list(qc._modin_frame.columns)[0]
+        want = name if name is not None else (
+            self.name if self.name is not None else 0)
+        if want != cur:
+            qc = qc.rename_columns({cur: want})
+        return DataFrame(query_compiler=qc)
+
+    def astype(self, dtype) -> "Series":
+        return Series(query_compiler=self._query_compiler.astype(dtype),
+                      name=self.name)
+
+    def quantile(self, q=0.5):
+        """pandas Series.quantile: scalar for scalar q, Series for a
+        list (linear interpolation, NaN skipped)."""
+        qs = [q] if np.isscalar(q) else list(q)
+        out = self._query_compiler.quantile(qs)
+        col = out.columns[0]
+        return float(out[col].iloc[0]) if np.isscalar(q) else out[col]
+
+    def any(self) -> bool:  # noqa: A003
+        """pandas Series.any (skipna): non-NaN nonzero exists =
+        count(notna) - count(== 0) > 0 (NaN == 0 is False on device, so
+        the zero count never includes NaN rows)."""
+        nz = (self == 0)._query_compiler.sum().iloc[0]
+        nn = self._query_compiler.notna().sum().iloc[0]
+        return bool(nn - nz > 0)
+
+    def all(self) -> bool:  # noqa: A003
+        """pandas Series.all over the 0/1 mask (NaN counts truthy, the
+        pandas rule)."""
+        qc = (self == 0)._query_compiler
+        s = qc.sum()
+        return bool(s.iloc[0] == 0)
+
+    def mode(self) -> "Series":
+        """pandas Series.mode: every value at the max multiplicity,
+        sorted — composed from the device value_counts (host reshape of
+        the ngroups-sized result only)."""
+        vc = self.value_counts()
+        vals = np.asarray(vc.index)
+        cnts = np.asarray(vc)
+        if not len(cnts):
+            return Series(pandas.Series([], dtype=np.float64,
+                                        name=self.name))
+        best = vals[cnts == cnts.max()]
+        return Series(pandas.Series(np.sort(best), name=self.name))
+
     def nlargest(self, n: int = 5):
         """pandas Series.nlargest(keep='first'): stable descending
         NaN-last sort + head(n)."""

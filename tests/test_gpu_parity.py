@@ -2770,3 +2770,23 @@ def test_nat_semantics_vs_pandas(npartitions):
     rp2["d2"] = pandas.Timestamp("2000-01-01")
     with pytest.raises(lib.HfError):
         df.merge(mpd.DataFrame(rp2), on="t", how="left")
+
+
+def test_series_extras_vs_pandas(npartitions):
+    """Series tail/to_frame/astype/quantile/any/all/mode on device."""
+    rng = np.random.default_rng(135)
+    n = 30_000
+    pdf = pandas.DataFrame({"a": rng.integers(0, 50, n),
+                            "v": rng.standard_normal(n)})
+    pdf.loc[rng.random(n) < 0.1, "v"] = np.nan
+    df = mpd.DataFrame(pdf)
+    pandas.testing.assert_series_equal(df["v"].tail(9).to_pandas(),
+                                       pdf["v"].tail(9))
+    assert abs(df["v"].quantile(0.75) - pdf["v"].quantile(0.75)) < 1e-12
+    assert df["v"].any() == pdf["v"].any()
+    assert df["v"].all() == pdf["v"].all()
+    assert (df["v"] > 100).any() == (pdf["v"] > 100).any()
+    got, exp = df["a"].mode().to_pandas(), pdf["a"].mode()
+    np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy())
+    g = df["a"].astype(np.float64).to_pandas()
+    assert g.dtype == np.float64

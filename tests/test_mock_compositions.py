@@ -928,3 +928,34 @@ def test_mock_to_datetime(mlib):
     np.testing.assert_allclose(g.to_numpy().astype(float),
                                e.to_numpy().astype(float), rtol=0,
                                equal_nan=True)
+
+
+def test_mock_series_extras(mlib):
+    """Series tail/to_frame/astype/quantile/any/all/mode."""
+    rng = np.random.default_rng(32)
+    pdf = pandas.DataFrame({"a": rng.integers(0, 6, 400),
+                            "v": rng.standard_normal(400)})
+    pdf.loc[rng.random(400) < 0.1, "v"] = np.nan
+    df = mlib.DataFrame(pdf)
+    pandas.testing.assert_series_equal(
+        df["v"].tail(7).to_pandas(), pdf["v"].tail(7))
+    f = df["v"].to_frame("x").to_pandas()
+    pandas.testing.assert_frame_equal(f, pdf["v"].to_frame("x"))
+    g = df["a"].astype(np.float64).to_pandas()
+    assert g.dtype == np.float64
+    np.testing.assert_allclose(g.to_numpy(),
+                               pdf["a"].to_numpy().astype(float), rtol=0)
+    assert abs(df["v"].quantile(0.3) - pdf["v"].quantile(0.3)) < 1e-12
+    np.testing.assert_allclose(
+        np.asarray(df["v"].quantile([0.1, 0.9])),
+        pdf["v"].quantile([0.1, 0.9]).to_numpy(), rtol=1e-12)
+    # any/all incl. NaN-skip rule
+    for data in ([0.0, np.nan], [0.0, 1.0], [np.nan], [1.0],
+                 [0, 0], [2, 3]):
+        ps = pandas.Series(data)
+        ms = mlib.DataFrame({"x": ps})["x"]
+        assert ms.any() == ps.any(), data
+        assert ms.all() == ps.all(), data
+    got = df["a"].mode().to_pandas()
+    exp = pdf["a"].mode()
+    np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy())
