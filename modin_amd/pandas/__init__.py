@@ -1514,6 +1514,40 @@ class Rolling:
     def max(self):
         return self._agg("max")
 
+    def var(self, ddof: int = 1):
+        return _window_var(
+            self._obj,
+            lambda o: Rolling(o, self._window, self._min_periods),
+            ddof, sqrt_=False)
+
+    def std(self, ddof: int = 1):
+        return _window_var(
+            self._obj,
+            lambda o: Rolling(o, self._window, self._min_periods),
+            ddof, sqrt_=True)
+
+
+def _window_var(obj, mk, ddof, sqrt_):
+    """rolling/expanding var/std from the existing window sums:
+    (Σx² − (Σx)²/n) / (n − ddof) over the window's non-NaN
+    observations, NaN when n <= ddof — the same prefix-scan kernels, no
+    new device code.  min_periods gating rides Σx (NaN propagates)."""
+    if isinstance(obj, DataFrame):
+        qcs = []
+        for c in obj.columns:
+            qcs.append(_window_var(obj[c], mk, ddof,
+                                   sqrt_)._query_compiler)
+        return DataFrame(query_compiler=qcs[0].hconcat(qcs[1:]))
+    s1 = mk(obj).sum()
+    s2 = mk(obj * obj).sum()
+    n = mk(obj).count()
+    num = (s2 - s1 * s1 / n).clip(lower=0.0)
+    v = (num / (n - ddof)).where(n > ddof)
+    if sqrt_:
+        v = v._rewrap(type(v._query_compiler).sqrt(
+            v._query_compiler, 0.0))
+    return v
+
 
 class Expanding:
     """pandas Expanding (growing windows) for sum/mean/count/min/max."""
@@ -1526,6 +1560,16 @@ class Expanding:
         qc = self._obj._query_compiler.expanding_agg(self._min_periods,
                                                      op)
         return self._obj._rewrap(qc)
+
+    def var(self, ddof: int = 1):
+        return _window_var(self._obj,
+                           lambda o: Expanding(o, self._min_periods),
+                           ddof, sqrt_=False)
+
+    def std(self, ddof: int = 1):
+        return _window_var(self._obj,
+                           lambda o: Expanding(o, self._min_periods),
+                           ddof, sqrt_=True)
 
     def sum(self):
         return self._agg("sum")

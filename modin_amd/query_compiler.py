@@ -67,6 +67,7 @@ class HipQueryCompiler:
     # scalar-only int ops (HF_MAP_IDIV/IMOD: Python floor/mod semantics)
     floordiv_int = Map.register(lib.MAP_IDIV)
     mod_int = Map.register(lib.MAP_IMOD)
+    sqrt = Map.register(lib.MAP_SQRT, f64_only=True)
 
     # ---- TreeReduce (query_compiler.py:984 sum etc.) ----
     sum = TreeReduce.register("sum")

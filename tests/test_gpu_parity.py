@@ -2790,3 +2790,27 @@ def test_series_extras_vs_pandas(npartitions):
     np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy())
     g = df["a"].astype(np.float64).to_pandas()
     assert g.dtype == np.float64
+
+
+def test_window_var_std_vs_pandas(npartitions):
+    """rolling/expanding var/std composed from the window prefix-scan
+    sums — no new kernels."""
+    rng = np.random.default_rng(136)
+    n = 60_000
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.12] = np.nan
+    pdf = pandas.DataFrame({"v": v})
+    df = mpd.DataFrame(pdf)
+    for w_, mp in ((16, None), (7, 3)):
+        for op in ("var", "std"):
+            g = getattr(df["v"].rolling(w_, min_periods=mp),
+                        op)().to_pandas()
+            e = getattr(pdf["v"].rolling(w_, min_periods=mp), op)()
+            np.testing.assert_allclose(g.to_numpy(), e.to_numpy(),
+                                       rtol=1e-7, atol=1e-9,
+                                       equal_nan=True,
+                                       err_msg=f"{w_}/{mp}/{op}")
+    g = df["v"].expanding(2).std().to_pandas()
+    e = pdf["v"].expanding(2).std()
+    np.testing.assert_allclose(g.to_numpy(), e.to_numpy(), rtol=1e-7,
+                               atol=1e-9, equal_nan=True)

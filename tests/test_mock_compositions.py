@@ -959,3 +959,32 @@ def test_mock_series_extras(mlib):
     got = df["a"].mode().to_pandas()
     exp = pdf["a"].mode()
     np.testing.assert_array_equal(got.to_numpy(), exp.to_numpy())
+
+
+def test_mock_window_var_std(mlib):
+    rng = np.random.default_rng(33)
+    n = 2000
+    v = rng.standard_normal(n)
+    v[rng.random(n) < 0.12] = np.nan
+    pdf = pandas.DataFrame({"v": v, "w": rng.standard_normal(n)})
+    df = mlib.DataFrame(pdf)
+    for w_, mp in ((8, None), (5, 2), (10, 1)):
+        for op in ("var", "std"):
+            g = getattr(df["v"].rolling(w_, min_periods=mp), op)()
+            e = getattr(pdf["v"].rolling(w_, min_periods=mp), op)()
+            np.testing.assert_allclose(g.to_pandas().to_numpy(),
+                                       e.to_numpy(), rtol=1e-9,
+                                       atol=1e-9, equal_nan=True,
+                                       err_msg=f"{w_}/{mp}/{op}")
+    for mp in (1, 3):
+        g = df["v"].expanding(mp).var().to_pandas()
+        e = pdf["v"].expanding(mp).var()
+        np.testing.assert_allclose(g.to_numpy(), e.to_numpy(),
+                                   rtol=1e-9, atol=1e-9, equal_nan=True)
+    # frame-wide rolling std
+    g = df.rolling(6).std().to_pandas()
+    e = pdf.rolling(6).std()
+    for c in e.columns:
+        np.testing.assert_allclose(g[c].to_numpy(), e[c].to_numpy(),
+                                   rtol=1e-9, atol=1e-9, equal_nan=True,
+                                   err_msg=c)
