@@ -2766,6 +2766,30 @@ class HipDataframe:
         return HipDataframe(parts, self._index, columns,
                             self._row_lengths, dtypes)
 
+    def reverse_rows(self) -> "HipDataframe":
+        """Row reversal, device-side: reversed iota (cumsum of ones,
+        RSUB n) + one gather per column.  Used by duplicated(keep=
+        'last') — pandas computes keep='last' as keep='first' over the
+        reversed rows."""
+        n = len(self)
+
+        def concat_col(name):
+            cols = [p.block().columns[name] for p in self._partitions]
+            return cols[0] if len(cols) == 1 else lib.concat(cols)
+
+        ones = lib.alloc(n, lib.HF_INT64)
+        if n:
+            lib.fill_i64(ones.dptr(), 1, n)
+        perm = lib.map_scalar(lib.MAP_RSUB, lib.cumsum(ones, lib.AGG_SUM),
+                              n)  # n-1, n-2, …, 0
+        blk_cats = (self._partitions[0].block().cats
+                    if self._partitions else {})
+        cols = {c: lib.gather(concat_col(c), perm) for c in self.columns}
+        idx = pandas.Index(np.asarray(self.index)[::-1])
+        part = HipDataframePartition(DeviceBlock(cols, n, dict(blk_cats)))
+        return HipDataframe([part], idx, list(self.columns), [n],
+                            self.dtypes.copy())
+
     def sample_rows(self, n: int, seed: int) -> "HipDataframe":
         """Uniform sample WITHOUT replacement, fully device-side: one
         splitmix64 uniform key per row (hf_fill_randf64), the n smallest

@@ -209,18 +209,20 @@ class _HipPandasBase:
             raise lib.HfError("where/mask take a boolean Series condition")
         return self.where(~cond, other)
 
-    def duplicated(self, subset=None):
-        """pandas duplicated(keep='first'): boolean Series; NaN keys
-        compare equal (pandas semantics)."""
-        qc = self._query_compiler.duplicated(subset)
+    def duplicated(self, subset=None, keep="first"):
+        """pandas duplicated(keep='first'|'last'|False): boolean Series;
+        NaN keys compare equal (pandas semantics); 'last' rides the
+        device row reversal."""
+        qc = self._query_compiler.duplicated(subset, keep=keep)
         out = Series(query_compiler=qc, name=None)
         out._bool_mask = True
         return out
 
-    def drop_duplicates(self, subset=None):
-        """pandas drop_duplicates(keep='first'): original index labels of
-        the kept rows."""
-        return self._rewrap(self._query_compiler.drop_duplicates(subset))
+    def drop_duplicates(self, subset=None, keep="first"):
+        """pandas drop_duplicates: original index labels of the kept
+        rows (keep='first'|'last'|False)."""
+        return self._rewrap(self._query_compiler.drop_duplicates(
+            subset, keep=keep))
 
     def cumsum(self):
         return self._rewrap(self._query_compiler.cumsum())

@@ -599,18 +599,42 @@ class HipQueryCompiler:
             return DeviceBlock(out, block.length, block.cats)
         return self.__constructor__(frame.map(block_fn))
 
-    def duplicated(self, subset=None) -> "HipQueryCompiler":
-        """Row-duplicate mask, keep='first' (pandas duplicated): cumcount
-        over ALL subset columns with dropna=False (NaN==NaN, the
-        canonical-NaN effective key) > 0."""
+    def duplicated(self, subset=None, keep="first") -> "HipQueryCompiler":
+        """Row-duplicate mask (pandas duplicated): cumcount over ALL
+        subset columns with dropna=False (NaN==NaN, the canonical-NaN
+        effective key) > 0.  keep='last' runs keep='first' over the
+        device-reversed rows (pandas' own reduction); keep=False ORs the
+        two masks."""
+        if keep == "last":
+            rev = self._modin_frame.reverse_rows()
+            rev._index = pandas.RangeIndex(len(rev))  # positional pass
+            d = self.__constructor__(rev).duplicated(subset)
+            m = d._modin_frame.reverse_rows()
+            m._index = pandas.RangeIndex(len(m))
+            return self.__constructor__(m)
+        if keep is False:
+            f = self.duplicated(subset)._modin_frame
+            l_ = self.duplicated(subset, keep="last")._modin_frame
+
+            def orzip(lb, rb):
+                from modin_amd.core.partition import DeviceBlock
+                (ln, lc), = lb.columns.items()
+                (rc,) = rb.columns.values()
+                return DeviceBlock(
+                    {ln: lib.binary(lib.BIN_MAX, lc, rc)}, lb.length)
+
+            return self.__constructor__(f.n_ary_op(orzip, l_))
+        if keep != "first":
+            raise lib.HfError(f"duplicated: bad keep {keep!r}")
         by = (list(self._modin_frame.columns) if subset is None
               else ([subset] if isinstance(subset, str) else list(subset)))
         cc = self._modin_frame.groupby_transform(by, "cumcount",
                                                  dropna=False)
         return self.__constructor__(cc.compare_scalar(lib.CMP_GE, 1.0))
 
-    def drop_duplicates(self, subset=None) -> "HipQueryCompiler":
-        dup = self.duplicated(subset)
+    def drop_duplicates(self, subset=None,
+                        keep="first") -> "HipQueryCompiler":
+        dup = self.duplicated(subset, keep=keep)
         inv = dup._modin_frame.compare_scalar(lib.CMP_EQ, 0.0)
         # the transform result is one partition; re-slice the mask to the
         # frame's partition lengths so filter_rows stays co-partitioned

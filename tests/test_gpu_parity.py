@@ -2814,3 +2814,26 @@ def test_window_var_std_vs_pandas(npartitions):
     e = pdf["v"].expanding(2).std()
     np.testing.assert_allclose(g.to_numpy(), e.to_numpy(), rtol=1e-7,
                                atol=1e-9, equal_nan=True)
+
+
+def test_duplicated_keep_vs_pandas(npartitions):
+    """duplicated/drop_duplicates keep='last'/False via device row
+    reversal + mask OR."""
+    rng = np.random.default_rng(137)
+    n = 50_000
+    pdf = pandas.DataFrame({
+        "k": rng.integers(0, 900, n),
+        "s": rng.choice(["a", "b", "c", None], n),
+        "v": rng.standard_normal(n)})
+    df = mpd.DataFrame(pdf)
+    for keep in ("first", "last", False):
+        for subs in (["k"], ["k", "s"], None):
+            g = df.duplicated(subs, keep=keep).to_pandas().to_numpy()
+            e = pdf.duplicated(subset=subs, keep=keep).to_numpy()
+            np.testing.assert_array_equal(g.astype(bool), e,
+                                          err_msg=f"{keep}/{subs}")
+            gd = df.drop_duplicates(subs, keep=keep).to_pandas()
+            ed = pdf.drop_duplicates(subset=subs, keep=keep)
+            assert len(gd) == len(ed), (keep, subs)
+            np.testing.assert_array_equal(np.asarray(gd.index),
+                                          ed.index.to_numpy())

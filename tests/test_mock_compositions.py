@@ -988,3 +988,24 @@ def test_mock_window_var_std(mlib):
         np.testing.assert_allclose(g[c].to_numpy(), e[c].to_numpy(),
                                    rtol=1e-9, atol=1e-9, equal_nan=True,
                                    err_msg=c)
+
+
+def test_mock_duplicated_keep(mlib):
+    rng = np.random.default_rng(34)
+    n = 3000
+    pdf = pandas.DataFrame({
+        "k": rng.integers(0, 60, n),
+        "s": rng.choice(["a", "b", None], n),
+        "v": rng.standard_normal(n)})
+    df = mlib.DataFrame(pdf)
+    for keep in ("first", "last", False):
+        for subs in (["k"], ["k", "s"], None):
+            g = df.duplicated(subs, keep=keep).to_pandas().to_numpy()
+            e = pdf.duplicated(subset=subs, keep=keep).to_numpy()
+            np.testing.assert_array_equal(g.astype(bool), e,
+                                          err_msg=f"{keep}/{subs}")
+            gd = df.drop_duplicates(subs, keep=keep).to_pandas()
+            ed = pdf.drop_duplicates(subset=subs, keep=keep)
+            assert len(gd) == len(ed), (keep, subs)
+            np.testing.assert_array_equal(np.asarray(gd.index),
+                                          ed.index.to_numpy())
