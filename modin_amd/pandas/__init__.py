@@ -127,6 +127,91 @@ def to_datetime(arg, format=None, errors: str = "raise"):  # noqa: A002
             format=format, errors=errors), name=arg.name)
 
 
+def _round_frac(x, precision):
+    # pandas core/reshape/tile.py _round_frac: round to `precision`
+    # SIGNIFICANT fractional digits for sub-1 magnitudes
+    if not np.isfinite(x) or x == 0:
+        return x
+    frac, whole = np.modf(x)
+    if whole == 0:
+        digits = -int(np.floor(np.log10(abs(frac)))) - 1 + precision
+    else:
+        digits = precision
+    return np.around(x, digits)
+
+
+def _infer_precision(base, edges):
+    # pandas tile.py _infer_precision: smallest precision keeping the
+    # rounded breaks unique
+    for precision in range(base, 20):
+        lv = np.asarray([_round_frac(b, precision) for b in edges])
+        if np.unique(lv).size == edges.size:
+            return precision
+    return base
+
+
+def cut(x: "Series", bins, right: bool = True, labels=None) -> "Series":
+    """pandas.cut: device range binning (one hf_shuffle_dest pass in the
+    order-isomorphic int64 space).  labels=None -> Interval values as a
+    dictionary column (groupby/value_counts behave observed=True —
+    empty bins don't appear; documented deviation from pandas'
+    Categorical); labels=False -> float64 bin codes (NaN unbinned),
+    exactly pandas."""
+    if not isinstance(x, Series):
+        raise HfErrorProxy("cut takes a Series")
+    if labels not in (None, False):
+        raise HfErrorProxy("cut: labels=None|False this round")
+    if isinstance(bins, (int, np.integer)):
+        mn = x.min()
+        mx = x.max()
+        if np.isnan(mn):
+            raise HfErrorProxy("cut: all-NaN input")
+        if mn == mx:  # pandas widens a zero-range by 0.1%
+            mn = mn - 0.001 * abs(mn) if mn != 0 else mn - 0.001
+            mx = mx + 0.001 * abs(mx) if mx != 0 else mx + 0.001
+            edges = np.linspace(mn, mx, int(bins) + 1)
+        else:
+            edges = np.linspace(mn, mx, int(bins) + 1)
+            adj = (mx - mn) * 0.001  # pandas includes the boundary point
+            if right:
+                edges[0] -= adj
+            else:
+                edges[-1] += adj
+        # pandas labels round the breaks to an inferred precision while
+        # the BINNING itself uses the unrounded edges
+        prec = _infer_precision(3, edges)
+        lab = np.asarray([_round_frac(b, prec) for b in edges])
+        return Series(query_compiler=x._query_compiler.cut_codes(
+            edges, right=right, as_codes=labels is False,
+            label_edges=lab), name=x.name)
+    edges = np.asarray(bins, dtype=np.float64)
+    return Series(query_compiler=x._query_compiler.cut_codes(
+        edges, right=right, as_codes=labels is False), name=x.name)
+
+
+def qcut(x: "Series", q, labels=None,
+         duplicates: str = "raise") -> "Series":
+    """pandas.qcut: quantile edges from the device sorted-column
+    quantiles, then the cut engine.  labels=False matches pandas
+    exactly; labels=None yields Interval values from the raw quantile
+    edges (pandas' display rounding is not replicated)."""
+    if not isinstance(x, Series):
+        raise HfErrorProxy("qcut takes a Series")
+    qs = (np.linspace(0, 1, int(q) + 1) if isinstance(q, (int, np.integer))
+          else np.asarray(q, dtype=np.float64))
+    edges = np.asarray(x.quantile(list(qs)))
+    if duplicates == "drop":
+        edges = np.unique(edges)
+    elif (np.diff(edges) <= 0).any():
+        raise HfErrorProxy("qcut: duplicate bin edges (pass "
+                           "duplicates='drop')")
+    edges = edges.copy()
+    # pandas qcut includes the minimum: widen the first edge a hair
+    edges[0] -= (abs(edges[0]) * 0.001 if edges[0] != 0 else 0.001)
+    return Series(query_compiler=x._query_compiler.cut_codes(
+        edges, right=True, as_codes=labels is False), name=x.name)
+
+
 def from_pandas(df: pandas.DataFrame) -> "DataFrame":
     return DataFrame(query_compiler=HipQueryCompiler.from_pandas(df))
 

@@ -510,6 +510,69 @@ class HipQueryCompiler:
                            [name], [n], pd.Series({name: dt}))
         return self.__constructor__(res)
 
+    def cut_codes(self, edges, right: bool = True,
+                  as_codes: bool = False,
+                  label_edges=None) -> "HipQueryCompiler":
+        """pd.cut engine: device range binning — ONE hf_shuffle_dest
+        pass over the ordered-transformed edges (#{edge < x} in the
+        order-isomorphic int64 space), codes −1 for NaN/out-of-range.
+        as_codes=False -> dictionary column over an IntervalIndex (the
+        categorical form); as_codes=True -> float64 codes (pandas
+        labels=False, NaN for unbinned)."""
+        from modin_amd.core.dataframe import HipDataframe
+        from modin_amd.core.partition import DeviceBlock, \
+            HipDataframePartition
+        import pandas as pd
+        frame = self._modin_frame
+        name = frame.columns[0]
+        blk_cats = (frame._partitions[0].block().cats
+                    if frame._partitions else {})
+        dt = frame.dtypes[name]
+        if name in blk_cats or (isinstance(dt, np.dtype)
+                                and np.issubdtype(dt, np.datetime64)):
+            raise lib.HfError("cut: numeric columns only this round")
+        edges = np.asarray(edges, dtype=np.float64)
+        if len(edges) < 2 or not (np.diff(edges) > 0).all():
+            raise lib.HfError("cut: edges must be increasing (duplicate "
+                              "quantile edges: qcut duplicates='drop')")
+        nbins = len(edges) - 1
+
+        def concat_col():
+            cs = [p.block().columns[name] for p in frame._partitions]
+            return cs[0] if len(cs) == 1 else lib.concat(cs)
+
+        col = concat_col()
+        n = col.length
+        ox = lib.ordered_i64(lib.cast_f64(col))
+        oe = lib.get(lib.ordered_i64(lib.put(edges)))
+        dest = lib.shuffle_dest(ox, oe + 1 if right else oe)
+        c = lib.map_scalar(lib.MAP_SUB, dest, 1)
+        ok = lib.binary(lib.BIN_MUL,
+                        lib.compare_scalar(lib.CMP_GE, c, 0.0),
+                        lib.compare_scalar(lib.CMP_LE, c,
+                                           float(nbins - 1)))
+        codes = lib.map_scalar(
+            lib.MAP_SUB,
+            lib.binary(lib.BIN_MUL,
+                       lib.map_scalar(lib.MAP_ADD, c, 1), ok), 1)
+        if as_codes:
+            # pandas labels=False: float64 with NaN for unbinned rows
+            out = lib.fixup_empty(lib.cast_f64(codes), ok)
+            blk = DeviceBlock({name: out}, n)
+            dts = pd.Series({name: np.dtype(np.float64)})
+            cats = {}
+        else:
+            cats_idx = pd.IntervalIndex.from_breaks(
+                edges if label_edges is None else label_edges,
+                closed="right" if right else "left")
+            blk = DeviceBlock({name: codes}, n,
+                              {name: pd.Index(cats_idx)})
+            dts = pd.Series({name: np.dtype(object)})
+            cats = {name: pd.Index(cats_idx)}
+        res = HipDataframe([HipDataframePartition(blk)], frame._index,
+                           [name], [n], dts)
+        return self.__constructor__(res)
+
     def to_datetime_from_strings(self, format=None,
                                  errors: str = "raise"
                                  ) -> "HipQueryCompiler":

@@ -103,6 +103,11 @@ def install(monkeypatch):
                                        if keys.arr.size else keys.arr,
                                        kind="stable")))
 
+    def shuffle_dest(keys, splitters):
+        spl = np.asarray(splitters, dtype=np.int64)
+        return _reg(MockCol(np.searchsorted(spl, keys.arr,
+                                            side="right").astype(np.int64)))
+
     def fill_randf64(n, seed):
         from oracle import ops as _o
         return _reg(MockCol(_o.rand_f64(seed, n)))
@@ -459,7 +464,7 @@ def install(monkeypatch):
         ("fill_f64", fill_f64), ("fill_i64", fill_i64),
         ("col_slice", col_slice), ("concat", concat), ("gather", gather),
         ("scatter", scatter), ("sort_perm", sort_perm),
-        ("fill_randf64", fill_randf64),
+        ("fill_randf64", fill_randf64), ("shuffle_dest", shuffle_dest),
         ("cumsum", cumsum), ("seg_cumsum", seg_cumsum),
         ("filter_plan", filter_plan), ("filter_apply", filter_apply),
         ("filter_iota", filter_iota), ("compare_scalar", compare_scalar),

@@ -1009,3 +1009,49 @@ def test_mock_duplicated_keep(mlib):
             assert len(gd) == len(ed), (keep, subs)
             np.testing.assert_array_equal(np.asarray(gd.index),
                                           ed.index.to_numpy())
+
+
+def test_mock_cut_qcut(mlib):
+    rng = np.random.default_rng(35)
+    n = 4000
+    v = rng.standard_normal(n) * 10
+    v[rng.random(n) < 0.1] = np.nan
+    pdf = pandas.DataFrame({"v": v})
+    df = mlib.DataFrame(pdf)
+    # labels=False: exact code parity incl. NaN
+    for bins in (4, 7, [-30.0, -5.0, 0.0, 5.0, 30.0]):
+        for right in (True, False):
+            g = mlib.cut(df["v"], bins, right=right,
+                         labels=False).to_pandas()
+            e = pandas.cut(pdf["v"], bins, right=right, labels=False)
+            np.testing.assert_allclose(
+                g.to_numpy(), e.to_numpy().astype(float), rtol=0,
+                equal_nan=True, err_msg=f"{bins}/{right}")
+    # labels=None: Interval values equal pandas' astype(object)
+    g = mlib.cut(df["v"], [-30.0, -5.0, 0.0, 5.0, 30.0]).to_pandas()
+    e = pandas.cut(pdf["v"], [-30.0, -5.0, 0.0, 5.0, 30.0]).astype(object)
+    same = (pandas.isna(g.to_numpy()) & pandas.isna(e.to_numpy())) \
+        | (g.to_numpy() == e.to_numpy())
+    assert same.all()
+    # int-bins Interval values match pandas' computed edges
+    g = mlib.cut(df["v"], 5).to_pandas()
+    e = pandas.cut(pdf["v"], 5).astype(object)
+    same = (pandas.isna(g.to_numpy()) & pandas.isna(e.to_numpy())) \
+        | (g.to_numpy() == e.to_numpy())
+    assert same.all()
+    # qcut codes
+    for q in (4, 10):
+        g = mlib.qcut(df["v"], q, labels=False).to_pandas()
+        e = pandas.qcut(pdf["v"], q, labels=False)
+        np.testing.assert_allclose(g.to_numpy(),
+                                   e.to_numpy().astype(float), rtol=0,
+                                   equal_nan=True, err_msg=str(q))
+    # groupby over a cut column (the common binning pattern)
+    pdf2 = pandas.DataFrame({"v": v, "w": rng.random(n)})
+    df2 = mlib.DataFrame(pdf2)
+    df2["bin"] = mlib.cut(df2["v"], [-30.0, 0.0, 30.0])
+    got = df2[["bin", "w"]].groupby("bin").sum().to_pandas()
+    pdf2["bin"] = pandas.cut(pdf2["v"], [-30.0, 0.0, 30.0])
+    exp = pdf2[["bin", "w"]].groupby("bin", observed=True).sum()
+    np.testing.assert_allclose(got["w"].to_numpy(), exp["w"].to_numpy(),
+                               rtol=1e-12)
