@@ -813,15 +813,54 @@ class DataFrame(_HipPandasBase):
         return DataFrame(query_compiler=type(qc)(nf))
 
     def reset_index(self, drop: bool = False):
-        if not drop:
-            raise lib.HfError("reset_index(drop=False) keeps the old index "
-                              "as a column — later round")
         qc = self._query_compiler
         frame = qc._modin_frame
         from ..core.dataframe import HipDataframe
+        if not drop:
+            # pandas: the old index becomes the leading column
+            idx = self.index
+            name = idx.name if idx.name is not None else "index"
+            if name in list(self.columns):
+                raise lib.HfError(
+                    f"reset_index: column {name!r} already exists")
+            out = DataFrame(query_compiler=qc)
+            vals = np.asarray(idx)
+            if vals.dtype == object:
+                out[name] = pandas.Series(vals, dtype=object)
+            else:
+                out[name] = vals
+            cols = [name] + [c for c in out.columns if c != name]
+            qc2 = out._query_compiler.getitem_column_array(cols)
+            qc2._modin_frame._index = pandas.RangeIndex(len(frame))
+            return DataFrame(query_compiler=qc2)
         nf = HipDataframe(frame._partitions, pandas.RangeIndex(len(frame)),
                           frame.columns, frame._row_lengths, frame.dtypes)
         return DataFrame(query_compiler=type(qc)(nf))
+
+    def set_index(self, keys: str, drop: bool = True):
+        """pandas set_index(column): the column BECOMES the index as a
+        lazy DeviceIndex (no host materialization until the index is
+        read); dictionary columns materialize through their cats,
+        datetime columns keep the dtype tag."""
+        if not isinstance(keys, str) or keys not in list(self.columns):
+            raise lib.HfError("set_index: one existing column name")
+        from ..core.dataframe import DeviceIndex, HipDataframe
+        frame = self._query_compiler._modin_frame
+        cols = [p.block().columns[keys] for p in frame._partitions]
+        col = cols[0] if len(cols) == 1 else lib.concat(cols)
+        cats = (frame._partitions[0].block().cats.get(keys)
+                if frame._partitions else None)
+        didx = DeviceIndex(col, name=keys, cats=cats)
+        names = ([c for c in frame.columns if c != keys] if drop
+                 else list(frame.columns))
+        nf = frame.take_columns(names)
+        nf = HipDataframe(nf._partitions, didx, names, nf._row_lengths,
+                          nf.dtypes)
+        dt = frame.dtypes[keys]
+        if isinstance(dt, np.dtype) and np.issubdtype(dt, np.datetime64):
+            nf._index_dtype = dt
+        return DataFrame(
+            query_compiler=type(self._query_compiler)(nf))
 
     def sort_values(self, by: str, ascending: bool = True,
                     kind: str = "stable", na_position: str = "last"):

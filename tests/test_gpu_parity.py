@@ -2874,3 +2874,28 @@ def test_cut_qcut_vs_pandas(npartitions):
     exp = pdf[["bin", "w"]].groupby("bin", observed=True).sum()
     np.testing.assert_allclose(got["w"].to_numpy(), exp["w"].to_numpy(),
                                rtol=1e-12)
+
+
+def test_set_reset_index_vs_pandas(npartitions):
+    """set_index (lazy DeviceIndex, datetime tag kept) and
+    reset_index(drop=False)."""
+    rng = np.random.default_rng(139)
+    n = 40_000
+    t = pandas.Series(pandas.to_datetime("2022-01-01")
+                      + pandas.to_timedelta(rng.integers(0, 10**5, n),
+                                            unit="s"))
+    pdf = pandas.DataFrame({"k": rng.integers(0, 900, n),
+                            "t": t, "v": rng.standard_normal(n)})
+    df = mpd.DataFrame(pdf)
+    got = df.set_index("t").to_pandas()
+    exp = pdf.set_index("t")
+    assert got.index.dtype == exp.index.dtype
+    np.testing.assert_array_equal(got.index.to_numpy(),
+                                  exp.index.to_numpy())
+    assert list(got.columns) == list(exp.columns)
+    g2 = df.set_index("k").reset_index().to_pandas()
+    e2 = pdf.set_index("k").reset_index()
+    assert list(g2.columns) == list(e2.columns)
+    np.testing.assert_array_equal(g2["k"].to_numpy(), e2["k"].to_numpy())
+    np.testing.assert_allclose(g2["v"].to_numpy(), e2["v"].to_numpy(),
+                               rtol=0, equal_nan=True)

@@ -1055,3 +1055,36 @@ def test_mock_cut_qcut(mlib):
     exp = pdf2[["bin", "w"]].groupby("bin", observed=True).sum()
     np.testing.assert_allclose(got["w"].to_numpy(), exp["w"].to_numpy(),
                                rtol=1e-12)
+
+
+def test_mock_set_reset_index(mlib):
+    rng = np.random.default_rng(36)
+    n = 1500
+    pdf = pandas.DataFrame({"k": rng.integers(0, 40, n),
+                            "s": rng.choice(["a", "b", None], n),
+                            "v": rng.standard_normal(n)})
+    df = mlib.DataFrame(pdf)
+    for key in ("k", "s", "v"):
+        got = df.set_index(key).to_pandas()
+        exp = pdf.set_index(key)
+        assert list(got.columns) == list(exp.columns), key
+        gi = got.index.to_numpy(dtype=object)
+        ei = exp.index.to_numpy(dtype=object)
+        same = (pandas.isna(gi) & pandas.isna(ei)) | (gi == ei)
+        assert same.all(), key
+    # drop=False keeps the column
+    got = df.set_index("k", drop=False).to_pandas()
+    exp = pdf.set_index("k", drop=False)
+    assert list(got.columns) == list(exp.columns)
+    # reset_index(drop=False): old index becomes the leading column
+    g2 = df.set_index("k").reset_index().to_pandas()
+    e2 = pdf.set_index("k").reset_index()
+    assert list(g2.columns) == list(e2.columns)
+    np.testing.assert_array_equal(g2["k"].to_numpy(), e2["k"].to_numpy())
+    pandas.testing.assert_index_equal(g2.index, e2.index)
+    # unnamed RangeIndex -> "index" column
+    g3 = df.reset_index().to_pandas()
+    e3 = pdf.reset_index()
+    assert list(g3.columns) == list(e3.columns)
+    np.testing.assert_array_equal(g3["index"].to_numpy(),
+                                  e3["index"].to_numpy())
