@@ -2771,6 +2771,10 @@ class HipDataframe:
         RSUB n) + one gather per column.  Used by duplicated(keep=
         'last') — pandas computes keep='last' as keep='first' over the
         reversed rows."""
+        from ..distributed import is_active, world_size
+        if is_active() and world_size() > 1:
+            raise lib.HfError("duplicated(keep='last'/False) at world>1 "
+                              "is a later round")
         n = len(self)
 
         def concat_col(name):
@@ -2799,6 +2803,9 @@ class HipDataframe:
         total = len(self)
         if not 0 <= n <= total:
             raise lib.HfError(f"sample: n={n} out of range 0..{total}")
+        from ..distributed import is_active, world_size
+        if is_active() and world_size() > 1:
+            raise lib.HfError("sample at world>1 is a later round")
 
         def concat_col(name):
             cols = [p.block().columns[name] for p in self._partitions]
