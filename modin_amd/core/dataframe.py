@@ -2562,10 +2562,6 @@ class HipDataframe:
             b = p.block()
             mcols.append(b.columns[list(b.columns)[0]])
         m = mcols[0] if len(mcols) == 1 else lib.concat(mcols)
-        if self._dt_cols():
-            raise lib.HfError("where/mask over datetime columns needs NaT "
-                              "fills — a later round (select other "
-                              "columns)")
         blk_cats = (self._partitions[0].block().cats
                     if self._partitions else {})
 
@@ -2576,9 +2572,22 @@ class HipDataframe:
         n = len(self)
         if m.length != n:
             raise lib.HfError("where/mask: condition length mismatch")
+        # pandas only upcasts int64 when a fill actually happens: an
+        # all-true cond keeps the column untouched (one cached reduce)
+        all_true = n == 0 or lib.reduce(m).imn >= 1
+        if self._dt_cols() and not all_true:
+            raise lib.HfError("where/mask over datetime columns needs NaT "
+                              "fills — a later round (select other "
+                              "columns)")
         out_cols, dts, cats = {}, {}, {}
         for c in self.columns:
             col = concat_col(c)
+            if all_true:
+                out_cols[c] = col
+                dts[c] = self.dtypes[c]
+                if c in blk_cats:
+                    cats[c] = blk_cats[c]
+                continue
             if c in blk_cats:
                 if other is not None:
                     raise lib.HfError("where/mask with a fill value over "

@@ -1088,3 +1088,11 @@ def test_mock_set_reset_index(mlib):
     assert list(g3.columns) == list(e3.columns)
     np.testing.assert_array_equal(g3["index"].to_numpy(),
                                   e3["index"].to_numpy())
+
+
+@pytest.mark.parametrize("seed", range(500, 540))
+def test_mock_fuzz_pipeline(mlib, seed):
+    """The GPU fuzz harness body over the numpy mock — a free CPU-tier
+    sweep on a DIFFERENT seed range than the GPU tier runs."""
+    from tests.test_gpu_fuzz import test_fuzz_pipeline
+    test_fuzz_pipeline(seed)
