@@ -97,11 +97,18 @@ def install(monkeypatch):
     def sort_perm(keys, ascending=True):
         if ascending:
             return _reg(MockCol(np.argsort(keys.arr, kind="stable")))
-        # stable descending == stable ascending on the negated key
-        # (hf_sort_perm sorts key_max - key)
-        return _reg(MockCol(np.argsort(keys.arr.max() - keys.arr
-                                       if keys.arr.size else keys.arr,
-                                       kind="stable")))
+        if not keys.arr.size:
+            return _reg(MockCol(np.argsort(keys.arr, kind="stable")))
+        if keys.dtype_code == HF_FLOAT64:
+            return _reg(MockCol(np.argsort(-keys.arr, kind="stable")))
+        # stable descending: mirror the kernel's mod-2^64 shifted keys
+        # (span - (key - key_min)) — plain `max - key` overflows int64
+        # for mixed-sign ordered keys
+        u = keys.arr.astype(np.int64).view(np.uint64)
+        mn = np.array([keys.arr.min()], dtype=np.int64).view(np.uint64)[0]
+        shifted = u - mn
+        return _reg(MockCol(np.argsort(shifted.max() - shifted,
+                                       kind="stable").astype(np.int64)))
 
     def shuffle_dest(keys, splitters):
         spl = np.asarray(splitters, dtype=np.int64)
