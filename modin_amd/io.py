@@ -50,6 +50,16 @@ def _column_to_device_ready(col, name):
         if isinstance(d, pa.ChunkedArray):
             d = d.combine_chunks()
         return _column_to_device_ready(d, name)
+    if pa.types.is_timestamp(t):
+        if t.tz is not None:
+            raise lib.HfError(
+                f"read: column {name!r} is tz-aware (convert with "
+                "tz_localize(None))")
+        out = arr.cast(pa.timestamp("ns")).to_numpy(zero_copy_only=False)
+        return np.ascontiguousarray(out, dtype="datetime64[ns]"), None
+    if pa.types.is_date(t):
+        out = arr.cast(pa.timestamp("ns")).to_numpy(zero_copy_only=False)
+        return np.ascontiguousarray(out, dtype="datetime64[ns]"), None
     if pa.types.is_floating(t):
         out = arr.cast(pa.float64()).to_numpy(zero_copy_only=False)
         return np.ascontiguousarray(out, dtype=np.float64), None
@@ -79,11 +89,16 @@ def read_parquet(path, columns=None):
     arrays, cats_map, dtypes = {}, {}, {}
     for name in names:
         arr, cats = _column_to_device_ready(table.column(name), name)
-        arrays[name] = arr
         if cats is not None:
+            arrays[name] = arr
             cats_map[name] = cats
             dtypes[name] = np.dtype(object)
+        elif np.issubdtype(arr.dtype, np.datetime64):
+            # tagged int64 ns view (NaT = iNaT bits)
+            arrays[name] = arr.view(np.int64)
+            dtypes[name] = np.dtype("datetime64[ns]")
         else:
+            arrays[name] = arr
             dtypes[name] = arr.dtype
     n = table.num_rows
     edf = pandas.DataFrame(arrays, index=pandas.RangeIndex(n), copy=False)
@@ -119,11 +134,16 @@ def read_csv(path, columns=None, **csv_kwargs):
     arrays, cats_map, dtypes = {}, {}, {}
     for name in names:
         arr, cats = _column_to_device_ready(table.column(name), name)
-        arrays[name] = arr
         if cats is not None:
+            arrays[name] = arr
             cats_map[name] = cats
             dtypes[name] = np.dtype(object)
+        elif np.issubdtype(arr.dtype, np.datetime64):
+            # tagged int64 ns view (NaT = iNaT bits)
+            arrays[name] = arr.view(np.int64)
+            dtypes[name] = np.dtype("datetime64[ns]")
         else:
+            arrays[name] = arr
             dtypes[name] = arr.dtype
     n = table.num_rows
     edf = pandas.DataFrame(arrays, index=pandas.RangeIndex(n), copy=False)

@@ -533,3 +533,41 @@ def test_to_datetime_vs_pandas(npartitions):
     es = pandas.DataFrame({"s": exp}).sort_values(
         "s", kind="stable")["s"]
     np.testing.assert_array_equal(gs.to_numpy(), es.to_numpy())
+
+
+def test_parquet_csv_datetime_roundtrip(tmp_path, npartitions):
+    """Parquet and CSV readers carry timestamp columns to the tagged
+    int64-ns device form (NaT included); to_parquet writes them back."""
+    rng = np.random.default_rng(140)
+    n = 4000
+    t = pandas.Series(pandas.to_datetime("2020-06-01")
+                      + pandas.to_timedelta(rng.integers(0, 10**6, n),
+                                            unit="s"))
+    t[rng.random(n) < 0.1] = pandas.NaT
+    pdf = pandas.DataFrame({"t": t, "v": rng.standard_normal(n)})
+    p = str(tmp_path / "d.parquet")
+    pdf.to_parquet(p)
+    got = mpd.read_parquet(p)
+    assert got.dtypes["t"] == np.dtype("datetime64[ns]")
+    back = got.to_pandas()
+    np.testing.assert_array_equal(back["t"].to_numpy(),
+                                  pdf["t"].to_numpy())
+    # write side
+    p2 = str(tmp_path / "d2.parquet")
+    got.to_parquet(p2)
+    again = pandas.read_parquet(p2)
+    np.testing.assert_array_equal(again["t"].to_numpy(),
+                                  pdf["t"].to_numpy())
+    # CSV: pyarrow infers ISO timestamps
+    p3 = str(tmp_path / "d.csv")
+    pdf.to_csv(p3, index=False)
+    gcsv = mpd.read_csv(p3)
+    assert gcsv.dtypes["t"] == np.dtype("datetime64[ns]")
+    np.testing.assert_array_equal(gcsv.to_pandas()["t"].to_numpy(),
+                                  pdf["t"].to_numpy())
+    # dt accessor straight off the ingested column
+    g = gcsv["t"].dt.hour.to_pandas()
+    e = pdf["t"].dt.hour
+    np.testing.assert_allclose(g.to_numpy().astype(float),
+                               e.to_numpy().astype(float), rtol=0,
+                               equal_nan=True)
