@@ -116,8 +116,8 @@ def read_csv(path, columns=None, **csv_kwargs):
     csv_dispatcher.py) without any per-row pandas materialization.
 
     Mirrors pandas.read_csv defaults for the supported column types
-    (int64 / float64 / strings; nullable ints -> float64+NaN).  Datetime
-    parsing stays with the caller (read as strings, convert)."""
+    (int64 / float64 / strings / ISO timestamps; nullable ints ->
+    float64+NaN; timestamp nulls -> NaT)."""
     import pyarrow.csv as pacsv
 
     from .query_compiler import HipQueryCompiler
