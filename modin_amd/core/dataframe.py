@@ -1261,8 +1261,6 @@ class HipDataframe:
             return self
         blk_cats = (self._partitions[0].block().cats
                     if self._partitions else {})
-        if blk_cats:
-            raise lib.HfError("shift over string columns is a later round")
         n = len(self)
         k = min(abs(periods), n)
 
@@ -1271,10 +1269,18 @@ class HipDataframe:
             return cols[0] if len(cols) == 1 else lib.concat(cols)
 
         dtc = self._dt_cols()
-        out_cols, dtypes = {}, {}
+        out_cols, dtypes, out_cats = {}, {}, {}
         for name in self.columns:
             col = concat_col(name)
-            if name in dtc:
+            if name in blk_cats:
+                # dictionary codes shift with a -1 (NaN) fill; the
+                # dictionary itself is untouched
+                nanb = lib.alloc(k, lib.HF_INT64)
+                if k:
+                    lib.fill_i64(nanb.dptr(), -1, k)
+                dtypes[name] = self.dtypes[name]
+                out_cats[name] = blk_cats[name]
+            elif name in dtc:
                 # pandas shift on datetime keeps the dtype, filling NaT
                 nanb = lib.alloc(k, lib.HF_INT64)
                 if k:
@@ -1293,7 +1299,7 @@ class HipDataframe:
             else:
                 kept = lib.col_slice(col, k, n - k)
                 out_cols[name] = lib.concat([kept, nanb])
-        part = HipDataframePartition(DeviceBlock(out_cols, n))
+        part = HipDataframePartition(DeviceBlock(out_cols, n, out_cats))
         return HipDataframe([part], self._index, self.columns, [n],
                             pandas.Series(dtypes))
 
@@ -3579,9 +3585,51 @@ class HipDataframe:
 
     # ---- astype over all columns ----
     def astype_all(self, dtype) -> "HipDataframe":
-        if self._partitions and self._partitions[0].block().cats:
-            raise lib.HfError("astype on string columns is a later round")
         dt = np.dtype(dtype)
+        blk_cats0 = (self._partitions[0].block().cats
+                     if self._partitions else {})
+        if blk_cats0:
+            if dt == np.dtype(object) or dt.kind in ("U", "S"):
+                return self  # string -> str is the identity
+            if dt not in (np.dtype(np.float64), np.dtype(np.int64)):
+                raise lib.HfError(f"astype({dt}) on string columns is a "
+                                  "later round")
+            # numeric parse of the HOST DICTIONARY + one device gather
+            # (pandas astype semantics: unparseable or NaN->int raise)
+            def parse_block(block: DeviceBlock) -> DeviceBlock:
+                out = {}
+                for n, c in block.columns.items():
+                    if n not in block.cats:
+                        out[n] = lib.map_scalar(
+                            lib.MAP_CAST_F64 if dt == np.dtype(np.float64)
+                            else lib.MAP_CAST_I64, c, 0)
+                        continue
+                    cats = block.cats[n].to_numpy(dtype=object)
+                    try:
+                        parsed = [float(x) for x in cats]
+                    except (TypeError, ValueError) as e:
+                        raise lib.HfError(
+                            f"astype({dt}): column {n!r}: {e}")
+                    if dt == np.dtype(np.int64):
+                        if any(p != int(p) for p in parsed):
+                            raise lib.HfError(
+                                f"astype(int64): column {n!r} holds "
+                                "non-integral strings")
+                        lut = np.empty(len(cats) + 1, dtype=np.int64)
+                        lut[1:] = [int(p) for p in parsed]
+                        lut[0] = 0
+                        if lib.reduce(c).imn < 0:
+                            raise lib.HfError(
+                                "astype(int64): NaN strings cannot "
+                                "convert (pandas raises too)")
+                    else:
+                        lut = np.empty(len(cats) + 1, dtype=np.float64)
+                        lut[0] = np.nan
+                        lut[1:] = parsed
+                    shifted = lib.map_scalar(lib.MAP_ADD, c, 1)
+                    out[n] = lib.gather(lib.put(lut), shifted)
+                return DeviceBlock(out, block.length)
+            return self.map(parse_block)
         if np.issubdtype(dt, np.datetime64):
             # int64 ns view -> datetime tag (device data unchanged; float
             # sources cast to int64 ns first, pandas' rule)

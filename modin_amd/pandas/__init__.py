@@ -1027,11 +1027,43 @@ class _StrAccessor:
         return self._wrap(self._s._query_compiler.str_op("upper"))
 
     def contains(self, pat, regex: bool = False, na=None):
-        if regex:
-            raise lib.HfError("str.contains(regex=True) is a later round")
-        return self._wrap(self._s._query_compiler.str_op("contains",
+        return self._wrap(self._s._query_compiler.str_op(
+            "contains", pat=pat, na=na, regex=regex),
+            bool_mask=na is not None)
+
+    def match(self, pat, na=None):
+        return self._wrap(self._s._query_compiler.str_op("match",
                                                          pat=pat, na=na),
                           bool_mask=na is not None)
+
+    def fullmatch(self, pat, na=None):
+        return self._wrap(self._s._query_compiler.str_op("fullmatch",
+                                                         pat=pat, na=na),
+                          bool_mask=na is not None)
+
+    def replace(self, pat, repl, regex: bool = True):
+        """pandas Series.str.replace (regex default True, pandas 2)."""
+        return self._wrap(self._s._query_compiler.str_op(
+            "replace", pat=pat, repl=repl, regex=regex))
+
+    def strip(self):
+        return self._wrap(self._s._query_compiler.str_op("strip"))
+
+    def lstrip(self):
+        return self._wrap(self._s._query_compiler.str_op("lstrip"))
+
+    def rstrip(self):
+        return self._wrap(self._s._query_compiler.str_op("rstrip"))
+
+    def title(self):
+        return self._wrap(self._s._query_compiler.str_op("title"))
+
+    def capitalize(self):
+        return self._wrap(self._s._query_compiler.str_op("capitalize"))
+
+    def zfill(self, width: int):
+        return self._wrap(self._s._query_compiler.str_op("zfill",
+                                                         width=width))
 
     def startswith(self, pat, na=None):
         return self._wrap(self._s._query_compiler.str_op("startswith",
@@ -1300,17 +1332,21 @@ class Series(_HipPandasBase):
         values = list(values)
         if len(values) > 64:
             raise lib.HfError("isin supports up to 64 values this round")
-        for v in values:
-            if isinstance(v, float) and v != v:
-                raise lib.HfError(
-                    "isin with NaN in the value list is a later round")
+        has_nan = any(isinstance(v, float) and v != v for v in values)
+        values = [v for v in values
+                  if not (isinstance(v, float) and v != v)]
         qc = self._query_compiler
-        if not values:
+        if not values and not has_nan:
             acc = qc.eq(float("inf"))  # all False (NaN == inf is False too)
         else:
             acc = None
-            for v in values:
-                m = qc.eq(v)
+            masks = [qc.eq(v) for v in values]
+            if has_nan:
+                # pandas isin: NaN in the value list matches NaN rows
+                notna = qc.notna()
+                one = type(notna).mul(notna, -1)
+                masks.append(type(one).add(one, 1))  # 1 - notna
+            for m in masks:
                 if acc is None:
                     acc = m
                 else:  # OR of 0/1 masks: a + b - a*b

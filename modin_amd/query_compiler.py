@@ -325,12 +325,15 @@ class HipQueryCompiler:
             self._modin_frame.groupby_idxminmax(by, maximum=False))
 
     def str_op(self, op: str, pat: str = None,
-               na=None) -> "HipQueryCompiler":
+               na=None, repl: str = None, regex: bool = False,
+               width: int = None) -> "HipQueryCompiler":
         """Series.str.<op> over a dictionary column: the transform runs on
         the HOST DICTIONARY (O(#categories)) and reaches the rows with one
         device gather — no string ever touches the GPU (SURVEY §8f.3
-        design).  ops: len, lower, upper, contains/startswith/endswith
-        (literal, na=True/False/None->NaN)."""
+        design).  ops: len, lower, upper, strip/lstrip/rstrip, title,
+        capitalize, zfill, replace (literal or regex),
+        contains (literal or regex)/startswith/endswith/match/fullmatch
+        (na=True/False/None->NaN)."""
         from modin_amd.core.dataframe import HipDataframe
         from modin_amd.core.partition import DeviceBlock, \
             HipDataframePartition
@@ -350,8 +353,20 @@ class HipQueryCompiler:
         codes = concat_col()
         n = codes.length
         shifted = lib.map_scalar(lib.MAP_ADD, codes, 1)  # NaN(-1) -> slot 0
-        if op in ("lower", "upper"):
-            vals = [getattr(c, op)() for c in cats.to_numpy(dtype=object)]
+        if op in ("lower", "upper", "strip", "lstrip", "rstrip", "title",
+                  "capitalize", "replace", "zfill"):
+            import re
+            if op == "replace":
+                if regex:
+                    rx = re.compile(pat)
+                    fn = lambda c: rx.sub(repl, c)  # noqa: E731
+                else:
+                    fn = lambda c: c.replace(pat, repl)  # noqa: E731
+            elif op == "zfill":
+                fn = lambda c: c.zfill(int(width))  # noqa: E731
+            else:
+                fn = lambda c: getattr(c, op)()  # noqa: E731
+            vals = [fn(c) for c in cats.to_numpy(dtype=object)]
             new_codes, new_cats = pd.factorize(pd.Series(vals), sort=True)
             lut = np.empty(len(cats) + 1, dtype=np.int64)
             lut[0] = -1
@@ -366,10 +381,23 @@ class HipQueryCompiler:
             out = lib.gather(lib.put(lut), shifted)
             blk = DeviceBlock({name: out}, n)
             dts = pd.Series({name: np.dtype(np.float64)})
-        elif op in ("contains", "startswith", "endswith"):
-            test = {"contains": lambda c: pat in c,
-                    "startswith": lambda c: c.startswith(pat),
-                    "endswith": lambda c: c.endswith(pat)}[op]
+        elif op in ("contains", "startswith", "endswith", "match",
+                    "fullmatch"):
+            import re
+            if op == "contains" and regex:
+                rx = re.compile(pat)
+                test = lambda c: rx.search(c) is not None  # noqa: E731
+            elif op == "match":
+                rx = re.compile(pat)
+                test = lambda c: rx.match(c) is not None  # noqa: E731
+            elif op == "fullmatch":
+                rx = re.compile(pat)
+                test = (lambda c:  # noqa: E731
+                        rx.fullmatch(c) is not None)
+            else:
+                test = {"contains": lambda c: pat in c,
+                        "startswith": lambda c: c.startswith(pat),
+                        "endswith": lambda c: c.endswith(pat)}[op]
             hit = [bool(test(c)) for c in cats.to_numpy(dtype=object)]
             if na is None:
                 # pandas returns object [True, False, NaN]; this backend

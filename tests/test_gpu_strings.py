@@ -571,3 +571,49 @@ def test_parquet_csv_datetime_roundtrip(tmp_path, npartitions):
     np.testing.assert_allclose(g.to_numpy().astype(float),
                                e.to_numpy().astype(float), rtol=0,
                                equal_nan=True)
+
+
+def test_str_extras_vs_pandas(npartitions):
+    """Regex contains/match/fullmatch, replace, strip/title/capitalize/
+    zfill, string shift, string astype, isin(NaN)."""
+    rng = np.random.default_rng(141)
+    n = 30_000
+    s = rng.choice(["  Alpha ", "beta42", "Gamma", "7.5", "x9y", None], n)
+    num = rng.choice(["1", "2.5", "-3"], n)
+    pdf = pandas.DataFrame({"s": s, "num": num,
+                            "v": rng.standard_normal(n)})
+    df = mpd.DataFrame(pdf)
+    for pat in (r"a\d+", r"^[Gx]"):
+        g = df["s"].str.contains(pat, regex=True, na=False).to_pandas()
+        e = pdf["s"].str.contains(pat, regex=True, na=False)
+        np.testing.assert_array_equal(g.to_numpy().astype(bool),
+                                      e.to_numpy(), err_msg=pat)
+    g = df["s"].str.fullmatch(r"\w+", na=False).to_pandas()
+    e = pdf["s"].str.fullmatch(r"\w+", na=False)
+    np.testing.assert_array_equal(g.to_numpy().astype(bool), e.to_numpy())
+    for op in ("strip", "lstrip", "rstrip", "title", "capitalize"):
+        g = getattr(df["s"].str, op)().to_pandas().to_numpy()
+        e = getattr(pdf["s"].str, op)().to_numpy()
+        same = (pandas.isna(g) & pandas.isna(e)) | (g == e)
+        assert same.all(), op
+    g = df["num"].str.zfill(5).to_pandas().to_numpy()
+    e = pdf["num"].str.zfill(5).to_numpy()
+    np.testing.assert_array_equal(g, e)
+    g = df["s"].str.replace("a", "_", regex=False).to_pandas().to_numpy()
+    e = pdf["s"].str.replace("a", "_", regex=False).to_numpy()
+    same = (pandas.isna(g) & pandas.isna(e)) | (g == e)
+    assert same.all()
+    g = df[["s", "v"]].shift(-2).to_pandas()
+    e = pdf[["s", "v"]].shift(-2)
+    gs, es = g["s"].to_numpy(), e["s"].to_numpy()
+    same = (pandas.isna(gs) & pandas.isna(es)) | (gs == es)
+    assert same.all()
+    g = df["num"].astype(np.float64).to_pandas()
+    np.testing.assert_allclose(g.to_numpy(),
+                               pdf["num"].astype(np.float64).to_numpy(),
+                               rtol=0)
+    g = df["num"].astype(np.float64).to_pandas()
+    m = df["v"].where(df["v"] > 1).isin([0.5, np.nan]).to_pandas()
+    em = pdf["v"].where(pdf["v"] > 1).isin([0.5, np.nan])
+    np.testing.assert_array_equal(m.to_numpy().astype(bool),
+                                  em.to_numpy())

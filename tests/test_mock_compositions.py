@@ -1096,3 +1096,55 @@ def test_mock_fuzz_pipeline(mlib, seed):
     sweep on a DIFFERENT seed range than the GPU tier runs."""
     from tests.test_gpu_fuzz import test_fuzz_pipeline
     test_fuzz_pipeline(seed)
+
+
+def test_mock_str_extras_shift_astype_isin(mlib):
+    """str regex/transform ops, string-column shift, string astype
+    (host-dictionary parse), isin with NaN."""
+    rng = np.random.default_rng(37)
+    n = 2000
+    s = rng.choice(["  Alpha ", "beta42", "Gamma", "7.5", None], n)
+    num = rng.choice(["1", "2.5", "-3", None], n)
+    pdf = pandas.DataFrame({"s": s, "num": num,
+                            "v": rng.standard_normal(n)})
+    df = mlib.DataFrame(pdf)
+    # regex contains / match / fullmatch
+    for pat in (r"a\d+", r"[GA]a?m"):
+        g = df["s"].str.contains(pat, regex=True, na=False).to_pandas()
+        e = pdf["s"].str.contains(pat, regex=True, na=False)
+        np.testing.assert_array_equal(g.to_numpy().astype(bool),
+                                      e.to_numpy(), err_msg=pat)
+        g = df["s"].str.match(pat, na=False).to_pandas()
+        e = pdf["s"].str.match(pat, na=False)
+        np.testing.assert_array_equal(g.to_numpy().astype(bool),
+                                      e.to_numpy(), err_msg=pat)
+    # transforms
+    for op, eop in (("strip", "strip"), ("title", "title"),
+                    ("capitalize", "capitalize")):
+        g = getattr(df["s"].str, op)().to_pandas().to_numpy()
+        e = getattr(pdf["s"].str, op)().to_numpy()
+        same = (pandas.isna(g) & pandas.isna(e)) | (g == e)
+        assert same.all(), op
+    g = df["s"].str.replace(r"\d+", "#", regex=True).to_pandas().to_numpy()
+    e = pdf["s"].str.replace(r"\d+", "#", regex=True).to_numpy()
+    same = (pandas.isna(g) & pandas.isna(e)) | (g == e)
+    assert same.all()
+    # string shift keeps dtype + NaN fill
+    g = df[["s", "v"]].shift(3).to_pandas()
+    e = pdf[["s", "v"]].shift(3)
+    gs, es = g["s"].to_numpy(), e["s"].to_numpy()
+    same = (pandas.isna(gs) & pandas.isna(es)) | (gs == es)
+    assert same.all()
+    # string astype float (NaN-safe) and int (loud on NaN)
+    g = df["num"].astype(np.float64).to_pandas()
+    e = pdf["num"].astype(np.float64)
+    np.testing.assert_allclose(g.to_numpy(), e.to_numpy(), rtol=0,
+                               equal_nan=True)
+    with pytest.raises(_HfErr):
+        df["num"].astype(np.int64)  # NaN strings
+    with pytest.raises(_HfErr):
+        df["s"].astype(np.float64)  # unparseable
+    # isin with NaN matches NaN rows
+    g = df["v"].where(df["v"] > 0).isin([np.nan]).to_pandas()
+    e = pdf["v"].where(pdf["v"] > 0).isin([np.nan])
+    np.testing.assert_array_equal(g.to_numpy().astype(bool), e.to_numpy())
