@@ -931,21 +931,23 @@ class HipQueryCompiler:
         if isinstance(on, (list, tuple)):
             if len(on) == 1:
                 on = on[0]
-            else:
-                if how not in ("inner", "left"):
-                    raise lib.HfError(
-                        f"multi-key merge how={how!r} is a later round "
-                        "(inner/left)")
+            elif how not in ("inner", "left", "right"):
+                raise lib.HfError(
+                    f"multi-key merge how={how!r} is a later round "
+                    "(inner/left/right)")
+            elif how != "right":
                 return self.__constructor__(
                     self._modin_frame.merge_multi(right._modin_frame,
                                                   list(on), how))
         if how == "right":
             # pandas right join == swapped left join with the suffix roles
             # flipped back and columns restored to left-then-right order
+            # (single- and multi-key: `keyset` carries both forms)
+            keyset = {on} if isinstance(on, str) else set(on)
             swapped = right.merge(self, on=on, how="left")
             frame = swapped._modin_frame
-            lcols = [c for c in self.columns if c != on]
-            rcols = [c for c in right.columns if c != on]
+            lcols = [c for c in self.columns if c not in keyset]
+            rcols = [c for c in right.columns if c not in keyset]
             common = set(lcols) & set(rcols)
             # in the swapped join, OUR columns got "_y" and right's "_x"
             ren = {}
@@ -961,9 +963,9 @@ class HipQueryCompiler:
             if ren:
                 out = out.rename_columns(
                     {("\x00tmp\x00" + k): v for k, v in ren.items()})
-            # pandas puts the key at its left-frame position; ours keeps
-            # the left column order with the key in place
-            left_order = [on if c == on else
+            # pandas puts the keys at their left-frame positions; ours
+            # keeps the left column order with the keys in place
+            left_order = [c if c in keyset else
                           (c + "_x" if c in common else c)
                           for c in self.columns]
             order = left_order + [(c + "_y" if c in common else c)

@@ -2899,3 +2899,30 @@ def test_set_reset_index_vs_pandas(npartitions):
     np.testing.assert_array_equal(g2["k"].to_numpy(), e2["k"].to_numpy())
     np.testing.assert_allclose(g2["v"].to_numpy(), e2["v"].to_numpy(),
                                rtol=0, equal_nan=True)
+
+
+def test_multikey_merge_right_vs_pandas(npartitions):
+    """merge(on=[a,b], how='right'): swapped-left composition over the
+    multi-key fold."""
+    rng = np.random.default_rng(142)
+    nl, nr = 20_000, 7_000
+    lpdf = pandas.DataFrame({
+        "a": rng.integers(0, 60, nl),
+        "b": (rng.integers(-20, 20, nl) / 4.0),
+        "x": rng.standard_normal(nl), "c": rng.integers(0, 5, nl)})
+    rpdf = pandas.DataFrame({
+        "a": rng.integers(0, 60, nr),
+        "b": (rng.integers(-20, 20, nr) / 4.0),
+        "y": rng.standard_normal(nr), "c": rng.integers(5, 9, nr)})
+    got = mpd.DataFrame(lpdf).merge(mpd.DataFrame(rpdf),
+                                    on=["a", "b"], how="right").to_pandas()
+    exp = lpdf.merge(rpdf, on=["a", "b"], how="right")
+    assert list(got.columns) == list(exp.columns)
+    assert len(got) == len(exp)
+    order = ["a", "b", "x", "y", "c_x", "c_y"]
+    gs = got.sort_values(order, na_position="last").reset_index(drop=True)
+    es = exp.sort_values(order, na_position="last").reset_index(drop=True)
+    for c in exp.columns:
+        np.testing.assert_allclose(gs[c].to_numpy().astype(float),
+                                   es[c].to_numpy().astype(float),
+                                   rtol=0, equal_nan=True, err_msg=c)

@@ -1148,3 +1148,28 @@ def test_mock_str_extras_shift_astype_isin(mlib):
     g = df["v"].where(df["v"] > 0).isin([np.nan]).to_pandas()
     e = pdf["v"].where(pdf["v"] > 0).isin([np.nan])
     np.testing.assert_array_equal(g.to_numpy().astype(bool), e.to_numpy())
+
+
+def test_mock_multikey_merge_right(mlib):
+    rng = np.random.default_rng(38)
+    nl, nr = 4000, 1500
+    lpdf = pandas.DataFrame({
+        "a": rng.integers(0, 30, nl),
+        "b": rng.integers(0, 6, nl),
+        "x": rng.random(nl), "c": rng.integers(0, 5, nl)})
+    rpdf = pandas.DataFrame({
+        "a": rng.integers(0, 30, nr),
+        "b": rng.integers(0, 6, nr),
+        "y": rng.random(nr), "c": rng.integers(5, 9, nr)})
+    got = mlib.DataFrame(lpdf).merge(mlib.DataFrame(rpdf),
+                                     on=["a", "b"], how="right").to_pandas()
+    exp = lpdf.merge(rpdf, on=["a", "b"], how="right")
+    assert list(got.columns) == list(exp.columns)
+    assert len(got) == len(exp)
+    order = ["a", "b", "x", "y", "c_x", "c_y"]
+    gs = got.sort_values(order, na_position="last").reset_index(drop=True)
+    es = exp.sort_values(order, na_position="last").reset_index(drop=True)
+    for c in exp.columns:
+        np.testing.assert_allclose(gs[c].to_numpy().astype(float),
+                                   es[c].to_numpy().astype(float),
+                                   rtol=0, equal_nan=True, err_msg=c)
