@@ -2580,6 +2580,9 @@ class HipDataframe:
             raise lib.HfError("where/mask: condition length mismatch")
         # pandas only upcasts int64 when a fill actually happens: an
         # all-true cond keeps the column untouched (one cached reduce)
+        if other is not None and isinstance(other, float) \
+                and np.isnan(other):
+            other = None  # NaN fill == the default fill
         all_true = n == 0 or lib.reduce(m).imn >= 1
         if self._dt_cols() and not all_true:
             raise lib.HfError("where/mask over datetime columns needs NaT "
@@ -2595,17 +2598,38 @@ class HipDataframe:
                     cats[c] = blk_cats[c]
                 continue
             if c in blk_cats:
-                if other is not None:
-                    raise lib.HfError("where/mask with a fill value over "
-                                      "string columns is a later round")
-                # codes: (code+1)*m - 1 -> untouched codes / −1 (NaN)
-                t = lib.map_scalar(
-                    lib.MAP_SUB,
-                    lib.binary(lib.BIN_MUL,
-                               lib.map_scalar(lib.MAP_ADD, col, 1), m), 1)
+                if other is not None and not isinstance(other, str):
+                    raise lib.HfError("where/mask over string columns: "
+                                      "the fill value must be a string")
+                ccats = blk_cats[c]
+                if other is None:
+                    # codes: (code+1)*m - 1 -> untouched codes / −1 (NaN)
+                    t = lib.map_scalar(
+                        lib.MAP_SUB,
+                        lib.binary(lib.BIN_MUL,
+                                   lib.map_scalar(lib.MAP_ADD, col, 1),
+                                   m), 1)
+                else:
+                    # string fill: union the fill value into the
+                    # dictionary (host), recode, then blend codes:
+                    # (code+1)*m + (fc+1)*(1-m) - 1
+                    ucats = union_cats(ccats, pandas.Index([other]))
+                    if not ucats.equals(ccats):
+                        col = recode_dict_col(col, ccats, ucats)
+                        ccats = ucats
+                    fc = int(ccats.get_loc(other))
+                    keep = lib.binary(
+                        lib.BIN_MUL,
+                        lib.map_scalar(lib.MAP_ADD, col, 1), m)
+                    fill = lib.map_scalar(
+                        lib.MAP_MUL, lib.map_scalar(lib.MAP_RSUB, m, 1),
+                        fc + 1)
+                    t = lib.map_scalar(
+                        lib.MAP_SUB, lib.binary(lib.BIN_ADD, keep, fill),
+                        1)
                 out_cols[c] = t
                 dts[c] = self.dtypes[c]
-                cats[c] = blk_cats[c]
+                cats[c] = ccats
             elif (col.dtype_code == lib.HF_INT64 and other is not None
                     and isinstance(other, (int, np.integer))):
                 t1 = lib.binary(lib.BIN_MUL, col, m)
@@ -2615,6 +2639,10 @@ class HipDataframe:
                 out_cols[c] = lib.binary(lib.BIN_ADD, t1, t2)
                 dts[c] = np.dtype(np.int64)
             else:
+                if isinstance(other, str):
+                    raise lib.HfError(
+                        f"where/mask: string fill over numeric column "
+                        f"{c!r} (pandas would upcast to object)")
                 cf = lib.cast_f64(col)
                 t = lib.fixup_empty(cf, m)
                 if other is not None:

@@ -617,3 +617,32 @@ def test_str_extras_vs_pandas(npartitions):
     em = pdf["v"].where(pdf["v"] > 1).isin([0.5, np.nan])
     np.testing.assert_array_equal(m.to_numpy().astype(bool),
                                   em.to_numpy())
+
+
+def test_where_string_fill_vs_pandas(npartitions):
+    """where/mask with a string fill over dictionary columns: dictionary
+    union + device code blend."""
+    rng = np.random.default_rng(143)
+    n = 40_000
+    pdf = pandas.DataFrame({"s": rng.choice(["aa", "bb", "cc", None], n),
+                            "v": rng.standard_normal(n)})
+    df = mpd.DataFrame(pdf)
+    m, pm = df["v"] > 0, pdf["v"] > 0
+    for fill in ("bb", "new"):
+        g = df[["s"]].where(m, fill).to_pandas()
+        e = pdf[["s"]].where(pm, fill)
+        gs, es = g["s"].to_numpy(), e["s"].to_numpy()
+        same = (pandas.isna(gs) & pandas.isna(es)) | (gs == es)
+        assert same.all(), fill
+    # downstream groupby over the filled column
+    d2 = mpd.DataFrame(query_compiler=df[["s"]].where(
+        m, "zzz")._query_compiler)
+    d2["v"] = df["v"]
+    got = d2.groupby("s").count().to_pandas()
+    p2 = pdf[["s"]].where(pm, "zzz")
+    p2["v"] = pdf["v"]
+    exp = p2.groupby("s").count()
+    np.testing.assert_array_equal(got.index.to_numpy(),
+                                  exp.index.to_numpy())
+    np.testing.assert_array_equal(got["v"].to_numpy(),
+                                  exp["v"].to_numpy())

@@ -1173,3 +1173,28 @@ def test_mock_multikey_merge_right(mlib):
         np.testing.assert_allclose(gs[c].to_numpy().astype(float),
                                    es[c].to_numpy().astype(float),
                                    rtol=0, equal_nan=True, err_msg=c)
+
+
+def test_mock_where_string_fill(mlib):
+    rng = np.random.default_rng(39)
+    n = 2000
+    pdf = pandas.DataFrame({"s": rng.choice(["a", "b", None], n),
+                            "v": rng.standard_normal(n)})
+    df = mlib.DataFrame(pdf)
+    m = df["v"] > 0
+    pm = pdf["v"] > 0
+    # existing-category fill and new-category fill
+    for fill in ("a", "zz"):
+        g = df[["s"]].where(m, fill).to_pandas()
+        e = pdf[["s"]].where(pm, fill)
+        gs, es = g["s"].to_numpy(), e["s"].to_numpy()
+        same = (pandas.isna(gs) & pandas.isna(es)) | (gs == es)
+        assert same.all(), fill
+    # mask() form too
+    g = df[["s"]].mask(m, "q").to_pandas()
+    e = pdf[["s"]].mask(pm, "q")
+    gs, es = g["s"].to_numpy(), e["s"].to_numpy()
+    same = (pandas.isna(gs) & pandas.isna(es)) | (gs == es)
+    assert same.all()
+    with pytest.raises(_HfErr):
+        df[["v"]].where(m, "x")
