@@ -130,7 +130,8 @@ class HipQueryCompiler:
         per pair, z = x + 0*y propagates NaN from either side and the
         NaN-skipping f64 reduce then yields the masked n/Σ/Σ² in one
         pass each — no mask materialization, 6 device passes per pair,
-        k×k scalars combined on host."""
+        k×k scalars combined on host; world>1 SUM-all-reduces the six
+        scalars per pair (moments are linear over shards)."""
         import math
         frame = self._modin_frame
         blk_cats = (frame._partitions[0].block().cats
@@ -144,9 +145,7 @@ class HipQueryCompiler:
         if bad:
             raise lib.HfError(f"corr/cov: non-numeric columns {bad} — "
                               "select numeric columns")
-        from .distributed import is_active, world_size
-        if is_active() and world_size() > 1:
-            raise lib.HfError("corr/cov at world>1 is a later round")
+        from . import distributed as dist_mod
 
         def concat_col(name):
             cs = [p.block().columns[name] for p in frame._partitions]
@@ -155,7 +154,9 @@ class HipQueryCompiler:
 
         cols = {c: concat_col(c) for c in names}
         k = len(names)
-        out = np.full((k, k), np.nan)
+        # shard-local masked moments per (i, j) pair — sums are linear,
+        # so world>1 just SUM-all-reduces the 6 scalars per pair
+        moments = {}
         for i in range(k):
             for j in range(i, k):
                 x, y = cols[names[i]], cols[names[j]]
@@ -164,23 +165,34 @@ class HipQueryCompiler:
                 zy = lib.binary(lib.BIN_ADD, y,
                                 lib.map_scalar(lib.MAP_MUL, x, 0.0))
                 rx, ry = lib.reduce(zx), lib.reduce(zy)
-                n = rx.count
-                sxy = lib.reduce(lib.binary(lib.BIN_MUL, zx, zy)).sum
-                sxx = lib.reduce(lib.binary(lib.BIN_MUL, zx, zx)).sum
-                syy = lib.reduce(lib.binary(lib.BIN_MUL, zy, zy)).sum
+                moments[(i, j)] = [
+                    float(rx.count), rx.sum, ry.sum,
+                    lib.reduce(lib.binary(lib.BIN_MUL, zx, zy)).sum,
+                    lib.reduce(lib.binary(lib.BIN_MUL, zx, zx)).sum,
+                    lib.reduce(lib.binary(lib.BIN_MUL, zy, zy)).sum]
+        if dist_mod.is_active() and dist_mod.world_size() > 1:
+            keys_ = sorted(moments)
+            flat = [v for p in keys_ for v in moments[p]]
+            flat = dist_mod.allreduce_scalars(flat)
+            for a, p in enumerate(keys_):
+                moments[p] = flat[a * 6:(a + 1) * 6]
+        out = np.full((k, k), np.nan)
+        for i in range(k):
+            for j in range(i, k):
+                n, sx, sy, sxy, sxx, syy = moments[(i, j)]
                 if corr:
                     if n < 2:
                         continue
-                    den = ((sxx - rx.sum ** 2 / n)
-                           * (syy - ry.sum ** 2 / n))
+                    den = ((sxx - sx ** 2 / n)
+                           * (syy - sy ** 2 / n))
                     if den <= 0:
                         continue
-                    v = (sxy - rx.sum * ry.sum / n) / math.sqrt(den)
+                    v = (sxy - sx * sy / n) / math.sqrt(den)
                     v = max(-1.0, min(1.0, v))
                 else:
                     if n - ddof <= 0:
                         continue
-                    v = (sxy - rx.sum * ry.sum / n) / (n - ddof)
+                    v = (sxy - sx * sy / n) / (n - ddof)
                 out[i, j] = out[j, i] = v
         idx = pandas.Index(names)
         return pandas.DataFrame(out, index=idx, columns=idx)

@@ -519,3 +519,78 @@ def test_gloo_melt_setitem_global_order(world):
         errs.append(q.get())
     assert not errs, "\n".join(errs)
     assert all(p.exitcode == 0 for p in procs)
+
+
+def _corr_worker(rank, world, port, fail_q):
+    """World>1 corr/cov: the six masked moments per pair SUM-all-reduce
+    across shards; result must equal pandas on the GLOBAL frame."""
+    try:
+        os.environ.update({
+            "RANK": str(rank), "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        })
+        import pandas
+        import modin_amd.distributed as dist_mod
+        from tests import mocklib
+
+        class _RawPatch:
+            def setattr(self, obj, name, fn):
+                setattr(obj, name, fn)
+
+        mocklib.install(_RawPatch())
+        assert dist_mod.init_from_env(backend="gloo", gpu=False)
+        import modin_amd.pandas as mpd
+        from modin_amd.core import lib
+        from modin_amd.core.dataframe import HipDataframe
+        from modin_amd.core.partition import (DeviceBlock,
+                                              HipDataframePartition)
+        from modin_amd.query_compiler import HipQueryCompiler
+
+        rng = np.random.default_rng(88)  # same stream on all ranks
+        n = 3000
+        gx = rng.standard_normal(n)
+        gy = 0.6 * gx + rng.standard_normal(n)
+        gx[rng.random(n) < 0.1] = np.nan
+        gy[rng.random(n) < 0.1] = np.nan
+        counts = oracle.split_row_counts(n, world, 1)
+        offs = np.cumsum([0] + counts)
+        sl = slice(offs[rank], offs[rank + 1])
+        nl = counts[rank]
+        block = DeviceBlock({"x": lib.put(gx[sl]),
+                             "y": lib.put(gy[sl])}, nl)
+        frame = HipDataframe(
+            [HipDataframePartition(block)],
+            pandas.RangeIndex(offs[rank], offs[rank + 1]),
+            ["x", "y"], [nl],
+            pandas.Series({"x": np.dtype(np.float64),
+                           "y": np.dtype(np.float64)}))
+        df = mpd.DataFrame(query_compiler=HipQueryCompiler(frame))
+        pdf = pandas.DataFrame({"x": gx, "y": gy})
+        np.testing.assert_allclose(df.corr().to_numpy(),
+                                   pdf.corr().to_numpy(), rtol=1e-10)
+        np.testing.assert_allclose(df.cov().to_numpy(),
+                                   pdf.cov().to_numpy(), rtol=1e-10)
+        dist_mod.shutdown()
+    except Exception:  # noqa: BLE001
+        import traceback
+        fail_q.put(f"rank {rank}:\n{traceback.format_exc()}")
+        raise SystemExit(1)
+
+
+@pytest.mark.parametrize("world", [2])
+def test_gloo_corr_allreduce(world):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29557
+    procs = [ctx.Process(target=_corr_worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=110)
+    errs = []
+    while not q.empty():
+        errs.append(q.get())
+    assert not errs, "\n".join(errs)
+    assert all(p.exitcode == 0 for p in procs)
