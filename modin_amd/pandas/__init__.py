@@ -1605,11 +1605,26 @@ class DataFrameGroupBy:
     _AGGS = ("sum", "count", "mean", "min", "max", "var", "std",
              "median", "first", "last", "prod")
 
-    def agg(self, how):
-        """str, list-of-str (MultiIndex columns, pandas col-major order) or
-        dict {column: agg} — composed from the single-agg kernels and a
-        device-side horizontal concat (no data copies; columns re-label
-        lazily)."""
+    def agg(self, how=None, **named):
+        """str, list-of-str (MultiIndex columns, pandas col-major order),
+        dict {column: agg}, or pandas NAMED aggregation
+        (out=("col", "agg")) — composed from the single-agg kernels and
+        a device-side horizontal concat (no data copies; columns
+        re-label lazily)."""
+        if how is None and named:
+            bys = (list(self._by) if isinstance(self._by, (list, tuple))
+                   else [self._by])
+            base = self._nat_handled_df()
+            qcs = []
+            for out_name, spec in named.items():
+                if not (isinstance(spec, tuple) and len(spec) == 2):
+                    raise lib.HfError("named agg takes out=(column, agg)")
+                col, a = spec
+                sub = base[[*bys, col]]
+                qc = sub._query_compiler.groupby_agg(
+                    self._by, a, dropna=self._dropna)
+                qcs.append(qc.rename_columns({col: out_name}))
+            return DataFrame(query_compiler=qcs[0].hconcat(qcs[1:]))
         if isinstance(how, str):
             return self._agg(how)
         if isinstance(how, dict):

@@ -2926,3 +2926,23 @@ def test_multikey_merge_right_vs_pandas(npartitions):
         np.testing.assert_allclose(gs[c].to_numpy().astype(float),
                                    es[c].to_numpy().astype(float),
                                    rtol=0, equal_nan=True, err_msg=c)
+
+
+def test_named_agg_vs_pandas(npartitions):
+    rng = np.random.default_rng(144)
+    n = 50_000
+    pdf = pandas.DataFrame({"k": rng.integers(0, 500, n),
+                            "v": rng.standard_normal(n),
+                            "w": rng.integers(-9, 9, n)})
+    df = mpd.DataFrame(pdf)
+    got = df.groupby("k").agg(total=("v", "sum"), hi=("w", "max"),
+                              m=("v", "mean")).to_pandas()
+    exp = pdf.groupby("k").agg(total=("v", "sum"), hi=("w", "max"),
+                               m=("v", "mean"))
+    assert list(got.columns) == list(exp.columns)
+    np.testing.assert_array_equal(got.index.to_numpy(),
+                                  exp.index.to_numpy())
+    for c in exp.columns:
+        np.testing.assert_allclose(got[c].to_numpy().astype(float),
+                                   exp[c].to_numpy().astype(float),
+                                   rtol=1e-12, err_msg=c)

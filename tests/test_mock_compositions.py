@@ -1198,3 +1198,23 @@ def test_mock_where_string_fill(mlib):
     assert same.all()
     with pytest.raises(_HfErr):
         df[["v"]].where(m, "x")
+
+
+def test_mock_named_agg(mlib):
+    rng = np.random.default_rng(40)
+    n = 3000
+    pdf = pandas.DataFrame({"k": rng.integers(0, 25, n),
+                            "v": rng.standard_normal(n),
+                            "w": rng.integers(-9, 9, n)})
+    df = mlib.DataFrame(pdf)
+    got = df.groupby("k").agg(total=("v", "sum"), lo=("w", "min"),
+                              n=("v", "count")).to_pandas()
+    exp = pdf.groupby("k").agg(total=("v", "sum"), lo=("w", "min"),
+                               n=("v", "count"))
+    assert list(got.columns) == list(exp.columns)
+    np.testing.assert_array_equal(got.index.to_numpy(),
+                                  exp.index.to_numpy())
+    for c in exp.columns:
+        np.testing.assert_allclose(got[c].to_numpy().astype(float),
+                                   exp[c].to_numpy().astype(float),
+                                   rtol=1e-12, err_msg=c)
