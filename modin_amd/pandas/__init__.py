@@ -505,10 +505,35 @@ class _HipPandasBase:
         out._bool_mask = True
         return out
 
-    def dropna(self):
-        """pandas dropna(how='any', axis=0): keep rows with no NaN."""
-        mask_qc = self._query_compiler.dropna_mask()
-        return self._rewrap(self._query_compiler.getitem_array(mask_qc))
+    def dropna(self, how: str = "any", subset=None):
+        """pandas dropna(axis=0): how='any' keeps rows with no NaN in
+        the subset, how='all' drops rows where EVERY subset column is
+        NaN (notna-OR mask)."""
+        qc = self._query_compiler
+        sel = qc
+        if subset is not None:
+            subset = [subset] if isinstance(subset, str) else list(subset)
+            for c in subset:
+                if c not in list(qc.columns):
+                    raise lib.HfError(f"dropna: unknown column {c!r}")
+            sel = qc.getitem_column_array(subset)
+        if how == "any":
+            mask_qc = sel.dropna_mask()
+        elif how == "all":
+            masks = sel.notna()  # per-column 0/1 frame
+            acc = None
+            for c in masks.columns:
+                m = masks.getitem_column_array([c])
+                if acc is None:
+                    acc = m
+                else:
+                    t = type(acc).add(acc, m)
+                    u = type(acc).mul(acc, m)
+                    acc = type(t).sub(t, u)  # OR
+            mask_qc = acc
+        else:
+            raise lib.HfError(f"dropna: bad how {how!r}")
+        return self._rewrap(qc.getitem_array(mask_qc))
 
     def abs(self):
         return self._rewrap(type(self._query_compiler).abs(self._query_compiler,
@@ -1307,19 +1332,25 @@ class Series(_HipPandasBase):
             return np.array(vals, dtype=np.int64)
         return np.array(vals, dtype=object)
 
-    def value_counts(self):
+    def value_counts(self, normalize: bool = False):
         """pandas Series.value_counts: the device groupby supplies distinct
         values + counts + first-appearance positions; the final count-desc
         sort over the (small) distinct set is delegated to pandas
         sort_values on the appearance-ordered counts — bit-identical tie
         order with pandas (which starts from its hashtable's appearance
-        order and quicksorts)."""
+        order and quicksorts).  normalize=True divides by the non-NaN
+        total (pandas 'proportion')."""
         st = self._query_compiler.distinct_stats()
         order = np.argsort(st["firstpos"], kind="stable")
         idx = pandas.Index(np.asarray(st["values"], dtype=object)[order],
                            name=self.name)
-        pre = pandas.Series(st["counts"][order].astype(np.int64),
-                            index=idx, name="count")
+        counts = st["counts"][order].astype(np.int64)
+        if normalize:
+            tot = counts.sum()
+            pre = pandas.Series(counts / tot if tot else counts * 0.0,
+                                index=idx, name="proportion")
+        else:
+            pre = pandas.Series(counts, index=idx, name="count")
         return pre.sort_values(ascending=False)
 
     def nunique(self) -> int:
@@ -1328,7 +1359,8 @@ class Series(_HipPandasBase):
     def isin(self, values) -> "Series":
         """Membership mask composed from EQ compares, OR-folded as
         a+b-a*b over the 0/1 masks (<= 64 values; string Series translate
-        through the dictionary; NaN values rejected loudly)."""
+        through the dictionary; a NaN entry matches NaN rows via an
+        isna() mask — the pandas rule)."""
         values = list(values)
         if len(values) > 64:
             raise lib.HfError("isin supports up to 64 values this round")

@@ -1218,3 +1218,27 @@ def test_mock_named_agg(mlib):
         np.testing.assert_allclose(got[c].to_numpy().astype(float),
                                    exp[c].to_numpy().astype(float),
                                    rtol=1e-12, err_msg=c)
+
+
+def test_mock_dropna_how_subset_vc_normalize(mlib):
+    rng = np.random.default_rng(41)
+    n = 2500
+    pdf = pandas.DataFrame({"a": rng.standard_normal(n),
+                            "b": rng.standard_normal(n),
+                            "s": rng.choice(["x", "y", None], n)})
+    pdf.loc[rng.random(n) < 0.3, "a"] = np.nan
+    pdf.loc[rng.random(n) < 0.3, "b"] = np.nan
+    df = mlib.DataFrame(pdf)
+    for how in ("any", "all"):
+        for subset in (None, ["a"], ["a", "b"]):
+            g = df.dropna(how=how, subset=subset).to_pandas()
+            e = pdf.dropna(how=how, subset=subset)
+            assert len(g) == len(e), (how, subset)
+            np.testing.assert_array_equal(np.asarray(g.index),
+                                          e.index.to_numpy(),
+                                          err_msg=f"{how}/{subset}")
+    g = df["s"].value_counts(normalize=True)
+    e = pdf["s"].value_counts(normalize=True)
+    np.testing.assert_array_equal(np.asarray(g.index),
+                                  e.index.to_numpy())
+    np.testing.assert_allclose(np.asarray(g), e.to_numpy(), rtol=1e-12)
