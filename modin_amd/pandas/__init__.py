@@ -997,6 +997,75 @@ class DataFrame(_HipPandasBase):
                     "groupby(by=<column name> | [column names]) only")
         return DataFrameGroupBy(self, by, as_index=as_index, dropna=dropna)
 
+    @property
+    def empty(self) -> bool:
+        return len(self) == 0 or len(self.columns) == 0
+
+    @property
+    def size(self) -> int:
+        return len(self) * len(self.columns)
+
+    @property
+    def ndim(self) -> int:
+        return 2
+
+    @property
+    def values(self) -> np.ndarray:
+        return self.to_pandas().to_numpy()
+
+    def to_numpy(self) -> np.ndarray:
+        return self.to_pandas().to_numpy()
+
+    def copy(self) -> "DataFrame":
+        """Columns are immutable device buffers; a copy is a new frame
+        over the same ColumnRefs (copy-on-write by construction)."""
+        frame = self._query_compiler._modin_frame
+        from ..core.dataframe import HipDataframe
+        nf = HipDataframe(list(frame._partitions), frame._index,
+                          list(frame.columns), list(frame._row_lengths),
+                          frame.dtypes.copy())
+        return DataFrame(query_compiler=type(self._query_compiler)(nf))
+
+    def equals(self, other) -> bool:
+        if not isinstance(other, DataFrame):
+            return False
+        a, b = self.to_pandas(), other.to_pandas()
+        return a.equals(b)
+
+    def keys(self):
+        return self.columns
+
+    def items(self):
+        for c in self.columns:
+            yield c, self[c]
+
+    def take(self, indices) -> "DataFrame":
+        """pandas take(axis=0): positional row gather."""
+        return self.iloc[list(indices)]
+
+    def add_prefix(self, prefix: str) -> "DataFrame":
+        return self.rename(columns={c: f"{prefix}{c}"
+                                    for c in self.columns})
+
+    def add_suffix(self, suffix: str) -> "DataFrame":
+        return self.rename(columns={c: f"{c}{suffix}"
+                                    for c in self.columns})
+
+    def pop(self, col: str) -> "Series":
+        """pandas pop: return the column and drop it IN PLACE."""
+        out = self[col]
+        self._query_compiler = self._query_compiler.getitem_column_array(
+            [c for c in self.columns if c != col])
+        return out
+
+    def get(self, key, default=None):
+        return self[key] if key in list(self.columns) else default
+
+    def squeeze(self):
+        if len(self.columns) == 1:
+            return self[self.columns[0]]
+        return self
+
     def sample(self, n: int = None, frac: float = None,
                random_state=None) -> "DataFrame":
         """pandas sample(replace=False): device-side draw (one uniform
@@ -1388,6 +1457,34 @@ class Series(_HipPandasBase):
         out = Series(query_compiler=acc, name=self.name)
         out._bool_mask = True
         return out
+
+    @property
+    def empty(self) -> bool:
+        return len(self) == 0
+
+    @property
+    def size(self) -> int:
+        return len(self)
+
+    @property
+    def ndim(self) -> int:
+        return 1
+
+    @property
+    def values(self) -> np.ndarray:
+        return self.to_pandas().to_numpy()
+
+    def to_numpy(self) -> np.ndarray:
+        return self.to_pandas().to_numpy()
+
+    def copy(self) -> "Series":
+        return Series(query_compiler=self._query_compiler,
+                      name=self.name)
+
+    def equals(self, other) -> bool:
+        if not isinstance(other, Series):
+            return False
+        return self.to_pandas().equals(other.to_pandas())
 
     def to_pandas(self) -> pandas.Series:
         df = self._query_compiler.to_pandas()

@@ -1242,3 +1242,30 @@ def test_mock_dropna_how_subset_vc_normalize(mlib):
     np.testing.assert_array_equal(np.asarray(g.index),
                                   e.index.to_numpy())
     np.testing.assert_allclose(np.asarray(g), e.to_numpy(), rtol=1e-12)
+
+
+def test_mock_misc_surface(mlib):
+    rng = np.random.default_rng(42)
+    pdf = pandas.DataFrame({"a": rng.integers(0, 9, 300),
+                            "v": rng.standard_normal(300)})
+    df = mlib.DataFrame(pdf)
+    assert not df.empty and df.size == 600 and df.ndim == 2
+    np.testing.assert_allclose(df.to_numpy(), pdf.to_numpy(), rtol=0)
+    assert df.equals(df.copy())
+    assert list(df.keys()) == list(pdf.keys())
+    assert dict((k, len(v)) for k, v in df.items()) == \
+        {"a": 300, "v": 300}
+    g = df.take([5, 1, 7]).to_pandas()
+    e = pdf.take([5, 1, 7]).reset_index(drop=True)
+    np.testing.assert_allclose(g.to_numpy(), e.to_numpy(), rtol=0)
+    assert list(df.add_prefix("p_").columns) == ["p_a", "p_v"]
+    assert list(df.add_suffix("_s").columns) == ["a_s", "v_s"]
+    d2 = df.copy()
+    s = d2.pop("a")
+    assert list(d2.columns) == ["v"] and s.name == "a"
+    assert d2.get("missing") is None
+    assert d2.squeeze().name == "v"
+    sv = df["v"]
+    assert sv.size == 300 and sv.ndim == 1 and not sv.empty
+    assert sv.equals(sv.copy())
+    np.testing.assert_allclose(sv.values, pdf["v"].to_numpy(), rtol=0)
