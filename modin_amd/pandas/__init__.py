@@ -997,6 +997,71 @@ class DataFrame(_HipPandasBase):
                     "groupby(by=<column name> | [column names]) only")
         return DataFrameGroupBy(self, by, as_index=as_index, dropna=dropna)
 
+    def query(self, expr: str) -> "DataFrame":
+        """pandas query over the supported mask algebra: comparisons of
+        columns against literals or other columns (col-col rides the
+        subtract-compare composition), &/|/~ and and/or/not,
+        parentheses.  Everything else raises loudly."""
+        import ast
+        try:
+            tree = ast.parse(expr, mode="eval").body
+        except SyntaxError as e:
+            raise lib.HfError(f"query: cannot parse {expr!r}: {e}")
+        OPS = {ast.Gt: "__gt__", ast.GtE: "__ge__", ast.Lt: "__lt__",
+               ast.LtE: "__le__", ast.Eq: "__eq__", ast.NotEq: "__ne__"}
+        FLIP = {"__gt__": "__lt__", "__ge__": "__le__",
+                "__lt__": "__gt__", "__le__": "__ge__",
+                "__eq__": "__eq__", "__ne__": "__ne__"}
+
+        def value(n):
+            if isinstance(n, ast.Name):
+                if n.id not in list(self.columns):
+                    raise lib.HfError(f"query: unknown column {n.id!r}")
+                return self[n.id]
+            if isinstance(n, ast.Constant):
+                return n.value
+            if isinstance(n, ast.UnaryOp) \
+                    and isinstance(n.op, ast.USub) \
+                    and isinstance(n.operand, ast.Constant):
+                return -n.operand.value
+            raise lib.HfError("query: operands must be column names or "
+                              "literals")
+
+        def build(node):
+            if isinstance(node, ast.BoolOp):
+                ms = [build(v) for v in node.values]
+                acc = ms[0]
+                for m in ms[1:]:
+                    acc = (acc & m if isinstance(node.op, ast.And)
+                           else acc | m)
+                return acc
+            if isinstance(node, ast.UnaryOp) \
+                    and isinstance(node.op, ast.Not):
+                return ~build(node.operand)
+            if isinstance(node, ast.BinOp) \
+                    and isinstance(node.op, (ast.BitAnd, ast.BitOr)):
+                a, b = build(node.left), build(node.right)
+                return a & b if isinstance(node.op, ast.BitAnd) else a | b
+            if isinstance(node, ast.Compare):
+                if len(node.ops) != 1:
+                    raise lib.HfError("query: chained comparisons are a "
+                                      "later round")
+                opn = OPS.get(type(node.ops[0]))
+                if opn is None:
+                    raise lib.HfError("query: unsupported comparison")
+                a, b = value(node.left), value(node.comparators[0])
+                if isinstance(a, Series) and isinstance(b, Series):
+                    # col-col: compare the difference against 0 (NaN
+                    # rows propagate to False, the pandas rule)
+                    return getattr(a - b, opn)(0)
+                if isinstance(b, Series):
+                    a, b, opn = b, a, FLIP[opn]
+                return getattr(a, opn)(b)
+            raise lib.HfError("query: unsupported expression "
+                              f"{ast.dump(node)[:60]}")
+
+        return self[build(tree)]
+
     @property
     def empty(self) -> bool:
         return len(self) == 0 or len(self.columns) == 0

@@ -2970,3 +2970,23 @@ def test_dropna_how_subset_vc_normalize(npartitions):
     np.testing.assert_array_equal(np.asarray(g.index).astype(np.int64),
                                   e.index.to_numpy())
     np.testing.assert_allclose(np.asarray(g), e.to_numpy(), rtol=1e-12)
+
+
+def test_query_vs_pandas(npartitions):
+    """query: mask-algebra composition from parsed expressions
+    (comparisons, col-col via subtract, and/or/not, strings)."""
+    rng = np.random.default_rng(146)
+    n = 60_000
+    pdf = pandas.DataFrame({"a": rng.integers(0, 100, n),
+                            "b": rng.integers(0, 100, n),
+                            "v": rng.standard_normal(n),
+                            "s": rng.choice(["x", "y", "z"], n)})
+    pdf.loc[rng.random(n) < 0.1, "v"] = np.nan
+    df = mpd.DataFrame(pdf)
+    for expr in ("a > 50", "v <= 0.5 and a != 3", "a > b",
+                 "(a > 5 or b < 2) and not v > 0", "s != 'y'"):
+        g = df.query(expr).to_pandas()
+        e = pdf.query(expr)
+        assert len(g) == len(e), expr
+        np.testing.assert_array_equal(np.asarray(g.index),
+                                      e.index.to_numpy(), err_msg=expr)

@@ -1269,3 +1269,26 @@ def test_mock_misc_surface(mlib):
     assert sv.size == 300 and sv.ndim == 1 and not sv.empty
     assert sv.equals(sv.copy())
     np.testing.assert_allclose(sv.values, pdf["v"].to_numpy(), rtol=0)
+
+
+def test_mock_query(mlib):
+    rng = np.random.default_rng(43)
+    n = 3000
+    pdf = pandas.DataFrame({"a": rng.integers(0, 20, n),
+                            "b": rng.integers(0, 20, n),
+                            "v": rng.standard_normal(n),
+                            "s": rng.choice(["x", "y", "z"], n)})
+    pdf.loc[rng.random(n) < 0.1, "v"] = np.nan
+    df = mlib.DataFrame(pdf)
+    for expr in ("a > 10", "v <= 0.5 and a != 3", "a > b",
+                 "(a > 5 or b < 2) and not v > 0",
+                 "s == 'x'", "a == b or v > 1.5"):
+        g = df.query(expr).to_pandas()
+        e = pdf.query(expr)
+        assert len(g) == len(e), expr
+        np.testing.assert_array_equal(np.asarray(g.index),
+                                      e.index.to_numpy(), err_msg=expr)
+    with pytest.raises(_HfErr):
+        df.query("c > 1")
+    with pytest.raises(_HfErr):
+        df.query("a + b > 2")
