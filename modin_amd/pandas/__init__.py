@@ -997,6 +997,56 @@ class DataFrame(_HipPandasBase):
                     "groupby(by=<column name> | [column names]) only")
         return DataFrameGroupBy(self, by, as_index=as_index, dropna=dropna)
 
+    def filter(self, items=None, like: str = None,  # noqa: A003
+               regex: str = None) -> "DataFrame":
+        """pandas filter(axis=1): column selection by list, substring or
+        regex (metadata-only)."""
+        import re
+        given = sum(x is not None for x in (items, like, regex))
+        if given != 1:
+            raise lib.HfError("filter: exactly one of items/like/regex")
+        if items is not None:
+            keep = [c for c in self.columns if c in set(items)]
+        elif like is not None:
+            keep = [c for c in self.columns if like in str(c)]
+        else:
+            rx = re.compile(regex)
+            keep = [c for c in self.columns if rx.search(str(c))]
+        return DataFrame(
+            query_compiler=self._query_compiler.getitem_column_array(keep))
+
+    def select_dtypes(self, include=None, exclude=None) -> "DataFrame":
+        """pandas select_dtypes over this backend's dtype set (int64,
+        float64, object/strings, datetime64[ns])."""
+        def norm(spec):
+            if spec is None:
+                return None
+            spec = [spec] if not isinstance(spec, (list, tuple)) else spec
+            out = set()
+            for x in spec:
+                if x in (object, "object", str, "str"):
+                    out.add("object")
+                elif x in ("number", np.number):
+                    out.update(("int64", "float64"))
+                elif x in ("datetime", "datetime64", "datetime64[ns]",
+                           np.datetime64):
+                    out.add("datetime64[ns]")
+                else:
+                    out.add(str(np.dtype(x)))
+            return out
+
+        inc, exc = norm(include), norm(exclude)
+        keep = []
+        for c in self.columns:
+            dt = str(self.dtypes[c])
+            if inc is not None and dt not in inc:
+                continue
+            if exc is not None and dt in exc:
+                continue
+            keep.append(c)
+        return DataFrame(
+            query_compiler=self._query_compiler.getitem_column_array(keep))
+
     def query(self, expr: str) -> "DataFrame":
         """pandas query over the supported mask algebra: comparisons of
         columns against literals or other columns (col-col rides the
@@ -1245,6 +1295,24 @@ class _DtAccessor:
     def _field(self, f: str) -> "Series":
         return Series(query_compiler=self._s._query_compiler.dt_field(f),
                       name=self._s.name)
+
+    _FREQ_NS = {"D": 86_400 * 10**9, "h": 3_600 * 10**9, "H": 3_600 * 10**9,
+                "min": 60 * 10**9, "T": 60 * 10**9, "s": 10**9,
+                "S": 10**9, "ms": 10**6, "us": 10**3, "ns": 1}
+
+    def floor(self, freq: str) -> "Series":
+        """pandas Series.dt.floor: truncate to the unit (D/h/min/s/ms/
+        us; NaT passes through)."""
+        unit = self._FREQ_NS.get(freq)
+        if unit is None:
+            raise lib.HfError(f"dt.floor: unsupported freq {freq!r}")
+        return Series(
+            query_compiler=self._s._query_compiler.dt_floor(unit),
+            name=self._s.name)
+
+    def normalize(self) -> "Series":
+        """pandas Series.dt.normalize == floor('D')."""
+        return self.floor("D")
 
     @property
     def year(self):

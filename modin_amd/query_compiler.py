@@ -613,6 +613,39 @@ class HipQueryCompiler:
                            [name], [n], dts)
         return self.__constructor__(res)
 
+    def dt_floor(self, unit_ns: int) -> "HipQueryCompiler":
+        """Series.dt.floor / normalize: ns − (ns mod unit) — HF_MAP_IMOD
+        carries Python's sign rule, so pre-1970 values floor correctly;
+        NaT rows keep iNaT through an arithmetic blend."""
+        from modin_amd.core.dataframe import HipDataframe, INAT
+        from modin_amd.core.partition import DeviceBlock, \
+            HipDataframePartition
+        frame = self._modin_frame
+        name = frame.columns[0]
+        dt = frame.dtypes[name]
+        if not (isinstance(dt, np.dtype) and np.issubdtype(
+                dt, np.datetime64)):
+            raise lib.HfError("dt.floor on a non-datetime column")
+
+        def concat_col():
+            cs = [p.block().columns[name] for p in frame._partitions]
+            return cs[0] if len(cs) == 1 else lib.concat(cs)
+
+        ns = concat_col()
+        n = ns.length
+        r = lib.binary(lib.BIN_SUB, ns,
+                       lib.map_scalar(lib.MAP_IMOD, ns, int(unit_ns)))
+        if n and lib.reduce(ns).imn == INAT:
+            m = lib.compare_scalar(lib.CMP_NE, ns, float(INAT))
+            inv = lib.map_scalar(lib.MAP_RSUB, m, 1)
+            nat = lib.map_scalar(lib.MAP_MUL, inv, INAT)
+            r = lib.binary(lib.BIN_ADD,
+                           lib.binary(lib.BIN_MUL, r, m), nat)
+        blk = DeviceBlock({name: r}, n)
+        res = HipDataframe([HipDataframePartition(blk)], frame._index,
+                           [name], [n], pandas.Series({name: dt}))
+        return self.__constructor__(res)
+
     def to_datetime_from_strings(self, format=None,
                                  errors: str = "raise"
                                  ) -> "HipQueryCompiler":

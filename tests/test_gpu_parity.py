@@ -2990,3 +2990,34 @@ def test_query_vs_pandas(npartitions):
         assert len(g) == len(e), expr
         np.testing.assert_array_equal(np.asarray(g.index),
                                       e.index.to_numpy(), err_msg=expr)
+
+
+def test_filter_selectdtypes_dtfloor(npartitions):
+    rng = np.random.default_rng(147)
+    n = 30_000
+    t = pandas.Series(pandas.to_datetime("1965-01-01")
+                      + pandas.to_timedelta(
+                          rng.integers(0, 10**7, n), unit="min"))
+    t[rng.random(n) < 0.1] = pandas.NaT
+    pdf = pandas.DataFrame({"aa": rng.integers(0, 5, n),
+                            "ab": rng.standard_normal(n), "t": t})
+    df = mpd.DataFrame(pdf)
+    assert list(df.filter(like="a").columns) == ["aa", "ab"]
+    assert list(df.select_dtypes(include="number").columns) == \
+        list(pdf.select_dtypes(include="number").columns)
+    for freq in ("D", "h", "min", "s"):
+        g = df["t"].dt.floor(freq).to_pandas()
+        e = pdf["t"].dt.floor(freq)
+        assert g.dtype == e.dtype
+        np.testing.assert_array_equal(g.to_numpy(), e.to_numpy(),
+                                      err_msg=freq)
+    # normalized column groups by calendar day
+    d2 = mpd.DataFrame(query_compiler=df["t"].dt.normalize()
+                       ._query_compiler)
+    d2["v"] = df["ab"]
+    g = d2.groupby("t").count().to_pandas()
+    p2 = pandas.DataFrame({"t": pdf["t"].dt.normalize(),
+                           "v": pdf["ab"]})
+    e = p2.groupby("t").count()
+    np.testing.assert_array_equal(g.index.to_numpy(), e.index.to_numpy())
+    np.testing.assert_array_equal(g["v"].to_numpy(), e["v"].to_numpy())

@@ -1292,3 +1292,34 @@ def test_mock_query(mlib):
         df.query("c > 1")
     with pytest.raises(_HfErr):
         df.query("a + b > 2")
+
+
+def test_mock_filter_selectdtypes_dtfloor(mlib):
+    rng = np.random.default_rng(44)
+    n = 1000
+    t = pandas.Series(pandas.to_datetime("1965-01-01")
+                      + pandas.to_timedelta(
+                          rng.integers(0, 10**7, n), unit="min"))
+    t[rng.random(n) < 0.1] = pandas.NaT
+    pdf = pandas.DataFrame({"aa": rng.integers(0, 5, n),
+                            "ab": rng.standard_normal(n),
+                            "s": rng.choice(["x", "y"], n), "t": t})
+    df = mlib.DataFrame(pdf)
+    assert list(df.filter(items=["ab", "s"]).columns) == ["ab", "s"]
+    assert list(df.filter(like="a").columns) == ["aa", "ab"]
+    assert list(df.filter(regex="^a.$").columns) == ["aa", "ab"]
+    assert list(df.select_dtypes(include="number").columns) == \
+        list(pdf.select_dtypes(include="number").columns)
+    assert list(df.select_dtypes(include=object).columns) == ["s"]
+    assert list(df.select_dtypes(exclude=["datetime64[ns]"]).columns) == \
+        ["aa", "ab", "s"]
+    # dt.floor / normalize incl. pre-1970 values and NaT
+    for freq in ("D", "h", "min"):
+        g = df["t"].dt.floor(freq).to_pandas()
+        e = pdf["t"].dt.floor(freq)
+        assert g.dtype == e.dtype
+        np.testing.assert_array_equal(g.to_numpy(), e.to_numpy(),
+                                      err_msg=freq)
+    g = df["t"].dt.normalize().to_pandas()
+    e = pdf["t"].dt.normalize()
+    np.testing.assert_array_equal(g.to_numpy(), e.to_numpy())
