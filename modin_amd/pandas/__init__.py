@@ -212,6 +212,30 @@ def qcut(x: "Series", q, labels=None,
         edges, right=True, as_codes=labels is False), name=x.name)
 
 
+def get_dummies(data: "Series", prefix: str = None) -> "DataFrame":
+    """pandas.get_dummies over a string Series: one 0/1 device column
+    per category (EQ masks over the dictionary codes; NaN rows are 0 in
+    every column, the pandas dummy_na=False default).  Columns land as
+    int64 0/1 (this backend's bool carrier — pandas emits bool)."""
+    if not isinstance(data, Series):
+        raise HfErrorProxy("get_dummies takes a Series")
+    qc = data._query_compiler
+    frame = qc._modin_frame
+    name = frame.columns[0]
+    blk_cats = (frame._partitions[0].block().cats
+                if frame._partitions else {})
+    if name not in blk_cats:
+        raise HfErrorProxy("get_dummies: string Series only this round")
+    cats = list(blk_cats[name].to_numpy(dtype=object))
+    if len(cats) > 64:
+        raise HfErrorProxy("get_dummies: > 64 categories")
+    names = [c if prefix is None else f"{prefix}_{c}" for c in cats]
+    acc = qc.eq(cats[0]).rename_columns({name: names[0]})
+    for c, out_name in zip(cats[1:], names[1:]):
+        acc = acc.write_column(out_name, qc.eq(c))
+    return DataFrame(query_compiler=acc)
+
+
 def from_pandas(df: pandas.DataFrame) -> "DataFrame":
     return DataFrame(query_compiler=HipQueryCompiler.from_pandas(df))
 

@@ -646,3 +646,20 @@ def test_where_string_fill_vs_pandas(npartitions):
                                   exp.index.to_numpy())
     np.testing.assert_array_equal(got["v"].to_numpy(),
                                   exp["v"].to_numpy())
+
+
+def test_get_dummies_vs_pandas(npartitions):
+    rng = np.random.default_rng(148)
+    n = 40_000
+    pdf = pandas.DataFrame({"s": rng.choice(
+        ["aa", "bb", "cc", "dd", None], n)})
+    df = mpd.DataFrame(pdf)
+    got = mpd.get_dummies(df["s"]).to_pandas()
+    exp = pandas.get_dummies(pdf["s"])
+    assert list(got.columns) == list(exp.columns)
+    np.testing.assert_array_equal(got.to_numpy().astype(bool),
+                                  exp.to_numpy().astype(bool))
+    # dummies sum per row == notna
+    s = mpd.get_dummies(df["s"]).sum(axis=1).to_pandas()
+    np.testing.assert_array_equal(s.to_numpy().astype(int),
+                                  pdf["s"].notna().to_numpy().astype(int))

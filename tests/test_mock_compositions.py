@@ -1323,3 +1323,20 @@ def test_mock_filter_selectdtypes_dtfloor(mlib):
     g = df["t"].dt.normalize().to_pandas()
     e = pdf["t"].dt.normalize()
     np.testing.assert_array_equal(g.to_numpy(), e.to_numpy())
+
+
+def test_mock_get_dummies(mlib):
+    rng = np.random.default_rng(45)
+    n = 1500
+    pdf = pandas.DataFrame({"s": rng.choice(["a", "b", "c", None], n)})
+    df = mlib.DataFrame(pdf)
+    got = mlib.get_dummies(df["s"]).to_pandas()
+    exp = pandas.get_dummies(pdf["s"])
+    assert list(got.columns) == list(exp.columns)
+    np.testing.assert_array_equal(got.to_numpy().astype(bool),
+                                  exp.to_numpy().astype(bool))
+    got = mlib.get_dummies(df["s"], prefix="p").to_pandas()
+    exp = pandas.get_dummies(pdf["s"], prefix="p")
+    assert list(got.columns) == list(exp.columns)
+    np.testing.assert_array_equal(got.to_numpy().astype(bool),
+                                  exp.to_numpy().astype(bool))
