@@ -10,8 +10,10 @@ filter/dropna/duplicated/drop_duplicates/nlargest, merge
 (inner/left/right/outer/cross, on/left_on+right_on, int64/float64/string
 keys incl. NaN matching), concat, column assignment
 (setitem/insert/assign), map/replace(dict), melt/pivot_table, sample,
-corr/cov, to_datetime + NaT semantics (DESIGN.md NaT scope), floordiv/
-mod, astype(dict), read_csv/to_csv, concat column alignment,
+corr/cov (incl. world>1), to_datetime + NaT semantics (DESIGN.md NaT
+scope), floordiv/mod, astype(dict), read_csv/to_csv, concat column
+alignment, query, cut/qcut, get_dummies, sample, set_index,
+filter/select_dtypes, dt.floor/normalize, named aggregation,
 and the groupby family: the reduce aggs + agg forms
 (modin/pandas/dataframe.py:2188 sum; modin/pandas/groupby.py:1330
 DataFrameGroupBy.sum -> _wrap_aggregation), size/nunique/first/last/
