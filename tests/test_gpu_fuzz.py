@@ -71,7 +71,7 @@ def test_fuzz_pipeline(seed):
     for si in range(steps):
         op = rng.choice(["filter", "sort", "head", "dropna", "arith",
                          "round", "where", "dedup", "iloc", "strmask",
-                         "assign", "mapdict"])
+                         "assign", "mapdict", "query"])
         msg = f"seed {seed} step {si} op {op}"
         if op == "filter":
             thr = float(np.round(rng.standard_normal() * 10, 2))
@@ -148,6 +148,18 @@ def test_fuzz_pipeline(seed):
                    for a, b in zip(keys, rng.integers(-5, 5, 8))}
             df["w"] = df["w"].replace(rep)
             pdf["w"] = pdf["w"].replace(rep)
+        elif op == "query":
+            if not all(c in pdf.columns for c in ("k", "v", "w")):
+                continue
+            thr = int(rng.integers(-500, 500))
+            expr = [f"w > {thr}", f"k < 200 and w != {thr}",
+                    f"w > k or v < 0.5",
+                    f"not (w > {thr} or v > 1.0)"][rng.integers(0, 4)]
+            df = df.query(expr)
+            pdf = pdf.query(expr)
+            check(df, pdf, msg)
+            df = mpd.DataFrame(df.to_pandas().reset_index(drop=True))
+            pdf = pdf.reset_index(drop=True)
         elif op == "dedup":
             subs = [["k"], ["k", "s"], ["s", "w"], None][rng.integers(0, 4)]
             if subs is not None:
