@@ -2584,13 +2584,33 @@ class HipDataframe:
                 and np.isnan(other):
             other = None  # NaN fill == the default fill
         all_true = n == 0 or lib.reduce(m).imn >= 1
-        if self._dt_cols() and not all_true:
-            raise lib.HfError("where/mask over datetime columns needs NaT "
-                              "fills — a later round (select other "
-                              "columns)")
+        dtc = self._dt_cols()
+        if dtc and other is not None and not all_true:
+            import datetime as _dtm
+            if not isinstance(other, (pandas.Timestamp, np.datetime64,
+                                      _dtm.datetime)):
+                raise lib.HfError(
+                    "where/mask over datetime columns: the fill must be "
+                    "a Timestamp (or None -> NaT)")
         out_cols, dts, cats = {}, {}, {}
         for c in self.columns:
             col = concat_col(c)
+            if c in dtc and not all_true:
+                # datetime: false rows fill NaT (or the Timestamp's ns);
+                # dtype kept — an int64 blend, no f64 round trip
+                fill = (INAT if other is None
+                        else int(pandas.Timestamp(other).value))
+                inv = lib.map_scalar(lib.MAP_RSUB, m, 1)
+                out_cols[c] = lib.binary(
+                    lib.BIN_ADD, lib.binary(lib.BIN_MUL, col, m),
+                    lib.map_scalar(lib.MAP_MUL, inv, fill))
+                dts[c] = self.dtypes[c]
+                continue
+            if c in dtc and other is not None:
+                # all-true: values untouched, dtype kept
+                out_cols[c] = col
+                dts[c] = self.dtypes[c]
+                continue
             if all_true:
                 out_cols[c] = col
                 dts[c] = self.dtypes[c]
@@ -2639,10 +2659,12 @@ class HipDataframe:
                 out_cols[c] = lib.binary(lib.BIN_ADD, t1, t2)
                 dts[c] = np.dtype(np.int64)
             else:
-                if isinstance(other, str):
+                if other is not None and not isinstance(
+                        other, (int, float, np.integer, np.floating)):
                     raise lib.HfError(
-                        f"where/mask: string fill over numeric column "
-                        f"{c!r} (pandas would upcast to object)")
+                        f"where/mask: fill {type(other).__name__} over "
+                        f"numeric column {c!r} (pandas would upcast to "
+                        "object)")
                 cf = lib.cast_f64(col)
                 t = lib.fixup_empty(cf, m)
                 if other is not None:

@@ -3021,3 +3021,27 @@ def test_filter_selectdtypes_dtfloor(npartitions):
     e = p2.groupby("t").count()
     np.testing.assert_array_equal(g.index.to_numpy(), e.index.to_numpy())
     np.testing.assert_array_equal(g["v"].to_numpy(), e["v"].to_numpy())
+
+
+def test_where_datetime_nat_fill(npartitions):
+    """where/mask over datetime columns: NaT (or Timestamp) fills via an
+    int64 blend, dtype kept."""
+    rng = np.random.default_rng(149)
+    n = 30_000
+    t = pandas.Series(pandas.to_datetime("2020-01-01")
+                      + pandas.to_timedelta(
+                          rng.integers(0, 10**6, n), unit="min"))
+    t[rng.random(n) < 0.1] = pandas.NaT
+    pdf = pandas.DataFrame({"t": t, "v": rng.standard_normal(n)})
+    df = mpd.DataFrame(pdf)
+    m, pm = df["v"] > 0, pdf["v"] > 0
+    g = df[["t"]].where(m).to_pandas()
+    e = pdf[["t"]].where(pm)
+    assert g["t"].dtype == e["t"].dtype
+    np.testing.assert_array_equal(g["t"].to_numpy(), e["t"].to_numpy())
+    fv = pandas.Timestamp("1999-01-01 03:04:05.000000006")
+    g = df[["t"]].mask(m, fv).to_pandas()
+    e = pdf[["t"]].mask(pm, fv)
+    np.testing.assert_array_equal(g["t"].to_numpy(), e["t"].to_numpy())
+    with pytest.raises(lib.HfError):
+        df[["v"]].where(m, fv)

@@ -893,11 +893,15 @@ def test_mock_nat_semantics(mlib):
     e = pdf["t"].fillna(fv)
     assert g.dtype == e.dtype
     np.testing.assert_array_equal(g.to_numpy(), e.to_numpy())
+    # where over datetime now fills NaT (dtype kept)
+    gw = df[["t", "v"]].where(df["v"] > 0).to_pandas()
+    ew = pdf[["t", "v"]].where(pdf["v"] > 0)
+    assert gw["t"].dtype == ew["t"].dtype
+    np.testing.assert_array_equal(gw["t"].to_numpy(), ew["t"].to_numpy())
     # unsupported ops are loud, not wrong
     for fn in (lambda: df[["t", "v"]].cumsum(),
                lambda: df[["t"]].diff(),
                lambda: df["t"].value_counts(),
-               lambda: df[["t", "v"]].where(df["v"] > 0),
                lambda: df[["t"]].astype(np.int64)):
         with pytest.raises(_HfErr):
             fn()
@@ -1340,3 +1344,25 @@ def test_mock_get_dummies(mlib):
     assert list(got.columns) == list(exp.columns)
     np.testing.assert_array_equal(got.to_numpy().astype(bool),
                                   exp.to_numpy().astype(bool))
+
+
+def test_mock_where_datetime(mlib):
+    rng = np.random.default_rng(46)
+    n = 1500
+    t = pandas.Series(pandas.to_datetime("2020-01-01")
+                      + pandas.to_timedelta(
+                          rng.integers(0, 10**6, n), unit="min"))
+    t[rng.random(n) < 0.1] = pandas.NaT
+    pdf = pandas.DataFrame({"t": t, "v": rng.standard_normal(n)})
+    df = mlib.DataFrame(pdf)
+    m, pm = df["v"] > 0, pdf["v"] > 0
+    g = df[["t"]].where(m).to_pandas()
+    e = pdf[["t"]].where(pm)
+    assert g["t"].dtype == e["t"].dtype
+    np.testing.assert_array_equal(g["t"].to_numpy(), e["t"].to_numpy())
+    fv = pandas.Timestamp("1999-01-01 03:04:05")
+    g = df[["t"]].mask(m, fv).to_pandas()
+    e = pdf[["t"]].mask(pm, fv)
+    np.testing.assert_array_equal(g["t"].to_numpy(), e["t"].to_numpy())
+    with pytest.raises(_HfErr):
+        df[["v"]].where(m, fv)
