@@ -55,8 +55,10 @@ def read_csv(path, columns=None, **csv_kwargs):
                                         **csv_kwargs))
 
 
-def concat(objs, ignore_index: bool = False):
-    """pandas.concat(axis=0).  Mismatched columns align with NaN fills
+def concat(objs, ignore_index: bool = False, axis=0):
+    """pandas.concat.  axis=1 composes columns positionally through the
+    zero-copy set_column path (equal lengths enforced device-side).
+    axis=0: mismatched columns align with NaN fills
     (pandas outer-join rule: first frame's columns, then new names in
     appearance order; an int64 column missing anywhere promotes to
     float64 — the pandas dtype rule).  Missing datetime columns raise
@@ -64,6 +66,17 @@ def concat(objs, ignore_index: bool = False):
     objs = list(objs)
     if not objs:
         raise HfErrorProxy("concat of empty list")
+    if axis in (1, "columns"):
+        frames = [o.to_frame() if isinstance(o, Series) else o
+                  for o in objs]
+        out = frames[0].copy()
+        for o in frames[1:]:
+            for c in o.columns:
+                if c in list(out.columns):
+                    raise HfErrorProxy(
+                        f"concat(axis=1): duplicate column {c!r}")
+                out[c] = o[c]
+        return out
     union = []
     for o in objs:
         for c in o.columns:
@@ -248,11 +261,18 @@ class _HipPandasBase:
     # ---- reductions ----
     def _reduce(self, name, **kwargs):
         axis = kwargs.pop("axis", 0)
+        numeric_only = kwargs.pop("numeric_only", False)
+        qc = self._query_compiler
+        if numeric_only:
+            keep = [c for c in qc.columns
+                    if isinstance(qc.dtypes[c], np.dtype)
+                    and qc.dtypes[c] in (np.dtype(np.int64),
+                                         np.dtype(np.float64))]
+            qc = qc.getitem_column_array(keep)
         if axis in (1, "columns"):
-            qc = self._query_compiler.reduce_axis1(name)
-            out = Series(query_compiler=qc, name=None)
+            out = Series(query_compiler=qc.reduce_axis1(name), name=None)
             return out
-        return self._lower(getattr(self._query_compiler, name)(**kwargs))
+        return self._lower(getattr(qc, name)(**kwargs))
 
     def sum(self, **kwargs):
         return self._reduce("sum", **kwargs)

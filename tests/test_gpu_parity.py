@@ -3045,3 +3045,22 @@ def test_where_datetime_nat_fill(npartitions):
     np.testing.assert_array_equal(g["t"].to_numpy(), e["t"].to_numpy())
     with pytest.raises(lib.HfError):
         df[["v"]].where(m, fv)
+
+
+def test_concat_axis1_numeric_only(npartitions):
+    rng = np.random.default_rng(150)
+    n = 30_000
+    p1 = pandas.DataFrame({"a": rng.integers(0, 90, n),
+                           "s": rng.choice(["x", "y"], n)})
+    p2 = pandas.DataFrame({"b": rng.standard_normal(n)})
+    got = mpd.concat([mpd.DataFrame(p1), mpd.DataFrame(p2)],
+                     axis=1).to_pandas()
+    exp = pandas.concat([p1, p2], axis=1)
+    assert list(got.columns) == list(exp.columns)
+    np.testing.assert_allclose(got["b"].to_numpy(), exp["b"].to_numpy(),
+                               rtol=0)
+    df = mpd.DataFrame(p1)
+    g = df.sum(numeric_only=True)
+    e = p1.sum(numeric_only=True)
+    np.testing.assert_allclose(np.asarray(g),
+                               e.to_numpy().astype(float), rtol=0)

@@ -1366,3 +1366,31 @@ def test_mock_where_datetime(mlib):
     np.testing.assert_array_equal(g["t"].to_numpy(), e["t"].to_numpy())
     with pytest.raises(_HfErr):
         df[["v"]].where(m, fv)
+
+
+def test_mock_concat_axis1_numeric_only(mlib):
+    rng = np.random.default_rng(47)
+    n = 800
+    p1 = pandas.DataFrame({"a": rng.integers(0, 9, n),
+                           "s": rng.choice(["x", "y"], n)})
+    p2 = pandas.DataFrame({"b": rng.random(n)})
+    got = mlib.concat([mlib.DataFrame(p1), mlib.DataFrame(p2)],
+                      axis=1).to_pandas()
+    exp = pandas.concat([p1, p2], axis=1)
+    assert list(got.columns) == list(exp.columns)
+    np.testing.assert_allclose(got["b"].to_numpy(), exp["b"].to_numpy(),
+                               rtol=0)
+    # Series operand
+    got = mlib.concat([mlib.DataFrame(p2),
+                       mlib.DataFrame(p1)["a"].to_frame("a2")],
+                      axis=1).to_pandas()
+    assert list(got.columns) == ["b", "a2"]
+    # numeric_only reductions skip string columns
+    df = mlib.DataFrame(p1)
+    g = df.sum(numeric_only=True)
+    e = p1.sum(numeric_only=True)
+    np.testing.assert_allclose(np.asarray(g), e.to_numpy().astype(float),
+                               rtol=0)
+    g = df.mean(numeric_only=True)
+    e = p1.mean(numeric_only=True)
+    np.testing.assert_allclose(np.asarray(g), e.to_numpy(), rtol=1e-12)
