@@ -594,3 +594,91 @@ def test_gloo_corr_allreduce(world):
         errs.append(q.get())
     assert not errs, "\n".join(errs)
     assert all(p.exitcode == 0 for p in procs)
+
+
+def _natsort_worker(rank, world, port, fail_q):
+    """World>1 sort_values by a NaT-bearing datetime key: the sentinel
+    scheme must survive the range shuffle (NaT rows land LAST on the
+    last rank for na_position='last')."""
+    try:
+        os.environ.update({
+            "RANK": str(rank), "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        })
+        import pandas
+        import modin_amd.distributed as dist_mod
+        from tests import mocklib
+
+        class _RawPatch:
+            def setattr(self, obj, name, fn):
+                setattr(obj, name, fn)
+
+        mocklib.install(_RawPatch())
+        assert dist_mod.init_from_env(backend="gloo", gpu=False)
+        import modin_amd.pandas as mpd
+        from modin_amd.core import lib
+        from modin_amd.core.dataframe import HipDataframe
+        from modin_amd.core.partition import (DeviceBlock,
+                                              HipDataframePartition)
+        from modin_amd.query_compiler import HipQueryCompiler
+
+        rng = np.random.default_rng(99)  # same stream on all ranks
+        n = 4000
+        base = pandas.Timestamp("2019-06-01").value
+        tns = base + rng.integers(0, 10**15, n)
+        tns[rng.random(n) < 0.12] = np.iinfo(np.int64).min  # NaT
+        gv = rng.standard_normal(n)
+        counts = oracle.split_row_counts(n, world, 1)
+        offs = np.cumsum([0] + counts)
+        sl = slice(offs[rank], offs[rank + 1])
+        nl = counts[rank]
+        block = DeviceBlock({"t": lib.put(tns[sl]), "v": lib.put(gv[sl])},
+                            nl)
+        frame = HipDataframe(
+            [HipDataframePartition(block)],
+            pandas.RangeIndex(nl),
+            ["t", "v"], [nl],
+            pandas.Series({"t": np.dtype("datetime64[ns]"),
+                           "v": np.dtype(np.float64)}))
+        df = mpd.DataFrame(query_compiler=HipQueryCompiler(frame))
+        for asc, nap in ((True, "last"), (False, "last"),
+                         (True, "first")):
+            out = df.sort_values("t", ascending=asc, na_position=nap)
+            of = out._query_compiler._modin_frame
+            got_t = np.concatenate(dist_mod._gather_np_varlen(
+                lib.get(of._partitions[0].block().columns["t"])))
+            got_pos = np.concatenate(dist_mod._gather_np_varlen(
+                np.asarray(of.index).astype(np.int64)))
+            pdf = pandas.DataFrame(
+                {"t": tns.view("datetime64[ns]"), "v": gv})
+            exp = pdf.sort_values("t", ascending=asc, na_position=nap,
+                                  kind="stable")
+            np.testing.assert_array_equal(
+                got_t.view("datetime64[ns]"), exp["t"].to_numpy(),
+                err_msg=f"{asc}/{nap}")
+            np.testing.assert_array_equal(got_pos, exp.index.to_numpy(),
+                                          err_msg=f"{asc}/{nap}")
+        dist_mod.shutdown()
+    except Exception:  # noqa: BLE001
+        import traceback
+        fail_q.put(f"rank {rank}:\n{traceback.format_exc()}")
+        raise SystemExit(1)
+
+
+@pytest.mark.parametrize("world", [2])
+def test_gloo_natsort_global_order(world):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29559
+    procs = [ctx.Process(target=_natsort_worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=110)
+    errs = []
+    while not q.empty():
+        errs.append(q.get())
+    assert not errs, "\n".join(errs)
+    assert all(p.exitcode == 0 for p in procs)
