@@ -251,6 +251,25 @@ def get_dummies(data: "Series", prefix: str = None) -> "DataFrame":
     return DataFrame(query_compiler=acc)
 
 
+def isna(obj):
+    """Module-level pandas.isna over this backend's Series."""
+    if isinstance(obj, Series):
+        return obj.isna()
+    return pandas.isna(obj)
+
+
+def notna(obj):
+    if isinstance(obj, Series):
+        return obj.notna()
+    return pandas.notna(obj)
+
+
+def unique(s: "Series"):
+    if not isinstance(s, Series):
+        raise HfErrorProxy("unique takes a Series")
+    return s.unique()
+
+
 def from_pandas(df: pandas.DataFrame) -> "DataFrame":
     return DataFrame(query_compiler=HipQueryCompiler.from_pandas(df))
 
