@@ -251,3 +251,82 @@ def test_fuzz_pipeline(seed):
             np.testing.assert_allclose(
                 g.astype(float), e.astype(float), rtol=1e-9, atol=1e-9,
                 equal_nan=True, err_msg=f"seed {seed} {by}/{agg}/{c}")
+
+
+@pytest.mark.parametrize("seed", range(int(os.environ.get(
+    "HF_MERGE_FUZZ_N", "15"))))
+def test_fuzz_merge(seed):
+    """Randomized merge matrix vs pandas: key dtype (small/wide int,
+    float+NaN, string, datetime, two-key), how, overlapping payload
+    names, empty-overlap cases — sorted-multiset comparison."""
+    rng = np.random.default_rng(3000 + seed)
+    nl = int(rng.integers(200, 20_000))
+    nr = int(rng.integers(100, 8_000))
+    kind = rng.choice(["int", "wide", "float", "str", "dt", "multi"])
+    how = str(rng.choice({
+        "int": ["inner", "left", "right", "outer"],
+        "wide": ["inner", "left", "right", "outer"],
+        "float": ["inner", "left", "right", "outer"],
+        "str": ["inner", "left"],
+        "dt": ["inner"],
+        "multi": ["inner", "left", "right"],
+    }[kind]))
+
+    def keys(n):
+        if kind == "int":
+            return rng.integers(0, rng.integers(5, 400), n)
+        if kind == "wide":
+            return rng.integers(-10**14, 10**14, n)
+        if kind == "float":
+            pool = np.r_[rng.standard_normal(50), np.nan]
+            return rng.choice(pool, n)
+        if kind == "str":
+            return rng.choice(np.array(
+                ["a", "b", "c", "d", None], dtype=object), n)
+        if kind == "dt":
+            return (pandas.Timestamp("2020-01-01").value
+                    + rng.integers(0, 50, n) * 86_400 * 10**9
+                    ).astype("datetime64[ns]")
+        return None
+
+    if kind == "multi":
+        lpdf = pandas.DataFrame({
+            "a": rng.integers(0, 40, nl), "b": rng.integers(0, 7, nl),
+            "x": rng.standard_normal(nl), "c": rng.integers(0, 5, nl)})
+        rpdf = pandas.DataFrame({
+            "a": rng.integers(0, 40, nr), "b": rng.integers(0, 7, nr),
+            "y": rng.standard_normal(nr), "c": rng.integers(5, 9, nr)})
+        on = ["a", "b"]
+        sort_cols = ["a", "b", "x", "y", "c_x", "c_y"]
+    else:
+        lpdf = pandas.DataFrame({"k": keys(nl),
+                                 "x": rng.standard_normal(nl),
+                                 "c": rng.integers(0, 5, nl)})
+        rpdf = pandas.DataFrame({"k": keys(nr),
+                                 "y": rng.standard_normal(nr),
+                                 "c": rng.integers(5, 9, nr)})
+        on = "k"
+        sort_cols = ["k", "x", "y", "c_x", "c_y"]
+    got = mpd.DataFrame(lpdf).merge(mpd.DataFrame(rpdf), on=on,
+                                    how=how).to_pandas()
+    exp = lpdf.merge(rpdf, on=on, how=how)
+    msg = f"seed {seed} {kind}/{how}"
+    assert list(got.columns) == list(exp.columns), msg
+    assert len(got) == len(exp), msg
+    if not len(exp):
+        return
+    gs = got.sort_values(sort_cols, na_position="last").reset_index(
+        drop=True)
+    es = exp.sort_values(sort_cols, na_position="last").reset_index(
+        drop=True)
+    for c in exp.columns:
+        g, e = gs[c].to_numpy(), es[c].to_numpy()
+        if e.dtype == object:
+            same = (pandas.isna(g) & pandas.isna(e)) | (g == e)
+            assert same.all(), f"{msg}/{c}"
+        elif np.issubdtype(e.dtype, np.datetime64):
+            np.testing.assert_array_equal(g, e, err_msg=f"{msg}/{c}")
+        else:
+            np.testing.assert_allclose(g.astype(float), e.astype(float),
+                                       rtol=0, equal_nan=True,
+                                       err_msg=f"{msg}/{c}")
