@@ -62,8 +62,18 @@ def check(df, pdf, msg):
 @pytest.mark.parametrize("seed", range(int(os.environ.get(
     "HF_FUZZ_N", "20"))))
 def test_fuzz_pipeline(seed):
+    import modin_amd.config as _cfg
     rng = np.random.default_rng(1000 + seed)
     n = int(rng.integers(500, 40_000))
+    old_np = _cfg.NPartitions.get()
+    _cfg.NPartitions.put(int(rng.integers(1, 5)))
+    try:
+        _fuzz_pipeline_body(rng, n, seed)
+    finally:
+        _cfg.NPartitions.put(old_np)
+
+
+def _fuzz_pipeline_body(rng, n, seed):
     pdf = make_frame(rng, n)
     df = mpd.DataFrame(pdf)
 
