@@ -1394,3 +1394,52 @@ def test_mock_concat_axis1_numeric_only(mlib):
     g = df.mean(numeric_only=True)
     e = p1.mean(numeric_only=True)
     np.testing.assert_allclose(np.asarray(g), e.to_numpy(), rtol=1e-12)
+
+
+@pytest.mark.parametrize("n", [0, 1])
+def test_mock_edge_lengths(mlib, n):
+    """Empty and single-row frames through the whole op surface —
+    results (columns, lengths, values) match pandas."""
+    pdf = pandas.DataFrame({"k": np.arange(n, dtype=np.int64),
+                            "v": np.ones(n)})
+    df = mlib.DataFrame(pdf)
+    checks = [
+        ("gbsum", lambda: df.groupby("k").sum().to_pandas(),
+         lambda: pdf.groupby("k").sum()),
+        ("sort", lambda: df.sort_values("k").to_pandas(),
+         lambda: pdf.sort_values("k")),
+        ("cumsum", lambda: df.cumsum().to_pandas(),
+         lambda: pdf.cumsum()),
+        ("dedup", lambda: df.drop_duplicates(["k"]).to_pandas(),
+         lambda: pdf.drop_duplicates(subset=["k"])),
+        ("merge", lambda: df.merge(
+            mlib.DataFrame(pdf.rename(columns={"v": "w"})),
+            on="k").to_pandas(),
+         lambda: pdf.merge(pdf.rename(columns={"v": "w"}), on="k")),
+        ("where", lambda: df.where(df["v"] > 2).to_pandas(),
+         lambda: pdf.where(pdf["v"] > 2)),
+        ("shift", lambda: df.shift(1).to_pandas(),
+         lambda: pdf.shift(1)),
+        ("rank", lambda: df.rank().to_pandas(), lambda: pdf.rank()),
+        ("filter", lambda: df[df["v"] > 0].to_pandas(),
+         lambda: pdf[pdf["v"] > 0]),
+        ("concat", lambda: mlib.concat([df, df],
+                                       ignore_index=True).to_pandas(),
+         lambda: pandas.concat([pdf, pdf], ignore_index=True)),
+        ("melt", lambda: df.melt(id_vars="k").to_pandas(),
+         lambda: pdf.melt(id_vars="k")),
+    ]
+    for name, gf, ef in checks:
+        g, e = gf(), ef()
+        assert list(g.columns) == list(e.columns), (n, name)
+        assert len(g) == len(e), (n, name)
+        if len(e):
+            num = e.select_dtypes(include="number").columns
+            np.testing.assert_allclose(
+                g[num].to_numpy().astype(float),
+                e[num].to_numpy().astype(float), rtol=0,
+                equal_nan=True, err_msg=f"{n}/{name}")
+    np.testing.assert_allclose(np.asarray(df.sum(), dtype=float),
+                               pdf.sum().to_numpy().astype(float),
+                               rtol=0)
+    assert df["v"].nunique() == pdf["v"].nunique()
