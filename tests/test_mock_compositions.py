@@ -1443,3 +1443,11 @@ def test_mock_edge_lengths(mlib, n):
                                pdf.sum().to_numpy().astype(float),
                                rtol=0)
     assert df["v"].nunique() == pdf["v"].nunique()
+
+
+@pytest.mark.parametrize("seed", range(600, 615))
+def test_mock_fuzz_merge(mlib, seed):
+    """The GPU merge fuzzer body over the numpy mock (different seed
+    range than the GPU tier)."""
+    from tests.test_gpu_fuzz import test_fuzz_merge
+    test_fuzz_merge(seed)
