@@ -1445,7 +1445,7 @@ def test_mock_edge_lengths(mlib, n):
     assert df["v"].nunique() == pdf["v"].nunique()
 
 
-@pytest.mark.parametrize("seed", range(600, 615))
+@pytest.mark.parametrize("seed", range(600, 620))
 def test_mock_fuzz_merge(mlib, seed):
     """The GPU merge fuzzer body over the numpy mock (different seed
     range than the GPU tier)."""
