@@ -1451,3 +1451,28 @@ def test_mock_fuzz_merge(mlib, seed):
     range than the GPU tier)."""
     from tests.test_gpu_fuzz import test_fuzz_merge
     test_fuzz_merge(seed)
+
+
+def test_mock_left_on_key_name_collision(mlib):
+    """left_on/right_on where the LEFT key name also exists as a RIGHT
+    payload column: pandas suffixes the key itself (a_x / a_y)."""
+    rng = np.random.default_rng(48)
+    l = pandas.DataFrame({"a": rng.integers(0, 10, 500),
+                          "v": rng.random(500)})
+    r = pandas.DataFrame({"b": rng.integers(0, 10, 200),
+                          "a": rng.integers(50, 60, 200),
+                          "w": rng.random(200)})
+    for how in ("inner", "left", "right", "outer"):
+        g = mlib.DataFrame(l).merge(mlib.DataFrame(r), left_on="a",
+                                    right_on="b", how=how).to_pandas()
+        e = l.merge(r, left_on="a", right_on="b", how=how)
+        assert list(g.columns) == list(e.columns), how
+        assert len(g) == len(e), how
+        order = list(e.columns)
+        gs = g.sort_values(order, na_position="last").reset_index(
+            drop=True)
+        es = e.sort_values(order, na_position="last").reset_index(
+            drop=True)
+        np.testing.assert_allclose(gs.to_numpy().astype(float),
+                                   es.to_numpy().astype(float), rtol=0,
+                                   equal_nan=True, err_msg=how)
