@@ -332,6 +332,14 @@ class _HipPandasBase:
     def var(self, ddof: int = 1):
         return self._lower(self._query_compiler.var(ddof=ddof))
 
+    def sem(self, ddof: int = 1):
+        """pandas sem: std/sqrt(count), composed from the one-pass
+        reduce partials."""
+        qc = self._query_compiler
+        std = qc.std(ddof=ddof)
+        cnt = qc.count()
+        return self._lower(std / np.sqrt(cnt.astype(float)))
+
     def rank(self, method: str = "average", ascending: bool = True,
              na_option: str = "keep"):
         """pandas rank(axis=0): always float64."""
@@ -1796,6 +1804,25 @@ class DataFrameGroupBy:
 
     def median(self):
         return self._agg("median")
+
+    def sem(self, ddof: int = 1):
+        """pandas DataFrameGroupBy.sem: std/sqrt(count) from the
+        existing groupby partials (frame-wise device divide + sqrt)."""
+        std = DataFrame(query_compiler=self._tail_qc("groupby_std",
+                                                     ddof=ddof))
+        cnt = self._agg("count")
+        if isinstance(cnt, Series):
+            cnt = cnt.to_frame(std.columns[0])
+            cnt = DataFrame(query_compiler=cnt._query_compiler)
+        # std / sqrt(cnt): sqrt via the map kernel, then frame divide
+        sq = DataFrame(query_compiler=type(
+            cnt._query_compiler).sqrt(
+            cnt._query_compiler.astype(np.float64), 0.0))
+        out = std / sq
+        if self._series_out and self._as_index:
+            name = list(out._query_compiler._modin_frame.columns)[0]
+            return Series(query_compiler=out._query_compiler, name=name)
+        return out
 
     def quantile(self, q: float = 0.5):
         out = DataFrame(

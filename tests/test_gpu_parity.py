@@ -3064,3 +3064,20 @@ def test_concat_axis1_numeric_only(npartitions):
     e = p1.sum(numeric_only=True)
     np.testing.assert_allclose(np.asarray(g),
                                e.to_numpy().astype(float), rtol=0)
+
+
+def test_sem_vs_pandas(npartitions):
+    rng = np.random.default_rng(151)
+    n = 40_000
+    pdf = pandas.DataFrame({"k": rng.integers(0, 300, n),
+                            "v": rng.standard_normal(n),
+                            "w": rng.standard_normal(n)})
+    pdf.loc[rng.random(n) < 0.1, "v"] = np.nan
+    df = mpd.DataFrame(pdf)
+    np.testing.assert_allclose(np.asarray(df.sem()),
+                               pdf.sem().to_numpy(), rtol=1e-10)
+    g = df.groupby("k").sem().to_pandas()
+    e = pdf.groupby("k").sem()
+    for c in e.columns:
+        np.testing.assert_allclose(g[c].to_numpy(), e[c].to_numpy(),
+                                   rtol=1e-9, equal_nan=True, err_msg=c)

@@ -1476,3 +1476,25 @@ def test_mock_left_on_key_name_collision(mlib):
         np.testing.assert_allclose(gs.to_numpy().astype(float),
                                    es.to_numpy().astype(float), rtol=0,
                                    equal_nan=True, err_msg=how)
+
+
+def test_mock_sem(mlib):
+    rng = np.random.default_rng(49)
+    n = 3000
+    pdf = pandas.DataFrame({"k": rng.integers(0, 30, n),
+                            "v": rng.standard_normal(n),
+                            "w": rng.standard_normal(n)})
+    pdf.loc[rng.random(n) < 0.1, "v"] = np.nan
+    df = mlib.DataFrame(pdf)
+    np.testing.assert_allclose(np.asarray(df.sem()),
+                               pdf.sem().to_numpy(), rtol=1e-10)
+    g = df.groupby("k").sem().to_pandas()
+    e = pdf.groupby("k").sem()
+    np.testing.assert_array_equal(g.index.to_numpy(), e.index.to_numpy())
+    for c in e.columns:
+        np.testing.assert_allclose(g[c].to_numpy(), e[c].to_numpy(),
+                                   rtol=1e-9, equal_nan=True, err_msg=c)
+    s = df.groupby("k")["v"].sem().to_pandas()
+    es = pdf.groupby("k")["v"].sem()
+    np.testing.assert_allclose(s.to_numpy(), es.to_numpy(), rtol=1e-9,
+                               equal_nan=True)
