@@ -249,7 +249,7 @@ def _fuzz_pipeline_body(rng, n, seed):
     if len(pdf) and all(c in pdf.columns for c in ("s", "v", "w")):
         by = ["k", "s"][rng.integers(0, 2)]
         agg = ["sum", "mean", "count", "min", "max", "var", "median",
-               "first", "last", "prod"][rng.integers(0, 10)]
+               "first", "last", "prod", "sem"][rng.integers(0, 11)]
         sub = [by, "v", "w"]  # numeric values only (string agg is loud)
         gout = getattr(df[sub].groupby(by), agg)().to_pandas()
         pout = getattr(pdf[sub].groupby(by), agg)()
